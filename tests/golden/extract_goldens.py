@@ -1,0 +1,133 @@
+#!/usr/bin/env python3
+"""Extract golden parity vectors from the reference checkout into JSON fixtures.
+
+Runs ONLY in the build container, where /root/reference (quickwit-oss/quickwit,
+workspace 0.9.0) is mounted read-only. The GPU box never sees /root/reference;
+tests consume the committed JSON files in this directory instead.
+
+Sources:
+  - BM25 exact scores: quickwit/quickwit-search/src/tests.rs:600-691
+    (test_sort_bm25) — transcribed below with the corpus and expected
+    (score_f32, doc_id) lists, since Rust cannot be parsed/compiled here.
+  - Aggregation buckets: quickwit/rest-api-tests/scenarii/aggregations/
+    {_setup.quickwit.yaml, 0001-aggregations.yaml} — parsed from YAML.
+"""
+import json
+import os
+import sys
+
+REF = "/root/reference/quickwit"
+OUT = os.path.dirname(os.path.abspath(__file__))
+
+
+def bm25_golden():
+    # Transcription of test_sort_bm25 (tests.rs:600-691).
+    # Doc mapping (tests.rs:602-617): title text record=freq fieldnorms,
+    # nofreq text record=basic fieldnorms. Default search fields include both.
+    docs = [
+        {"title": "one pad", "nofreq": "two pad"},  # doc 0
+        {"title": "one", "nofreq": "two"},          # doc 1
+        {"title": "one one", "nofreq": "two two"},  # doc 2
+    ]
+    cases = [
+        # (query_ast, expected [(score_f32, doc_id)] sorted desc)  tests.rs:670-688
+        {
+            "name": "term_with_freq",
+            "query_ast": {"type": "term", "field": "title", "value": "one"},
+            "expected": [[0.1738279, 2], [0.15965714, 1], [0.12343242, 0]],
+        },
+        {
+            "name": "term_record_basic",
+            "query_ast": {"type": "term", "field": "nofreq", "value": "two"},
+            "expected": [[0.15965714, 1], [0.12343242, 2], [0.12343242, 0]],
+        },
+        {
+            "name": "two_term_disjunction",
+            "query_ast": {
+                "type": "bool",
+                "should": [
+                    {"type": "term", "field": "title", "value": "one"},
+                    {"type": "term", "field": "nofreq", "value": "two"},
+                ],
+            },
+            "expected": [[0.31931427, 1], [0.2972603, 2], [0.24686484, 0]],
+        },
+    ]
+    return {
+        "source": "quickwit/quickwit-search/src/tests.rs:600-691 (test_sort_bm25)",
+        "sort": ["_score desc"],
+        "schema": [
+            {"name": "title", "type": "text", "record": "freq", "fieldnorms": True},
+            {"name": "nofreq", "type": "text", "record": "basic", "fieldnorms": True},
+        ],
+        "docs": docs,
+        "cases": cases,
+    }
+
+
+def agg_golden():
+    import yaml
+
+    setup_path = os.path.join(REF, "rest-api-tests/scenarii/aggregations/_setup.quickwit.yaml")
+    tests_path = os.path.join(REF, "rest-api-tests/scenarii/aggregations/0001-aggregations.yaml")
+    setup_steps = list(yaml.safe_load_all(open(setup_path)))
+    # The two ingest steps carry the corpus, one per split (commit: force).
+    splits = []
+    for step in setup_steps:
+        if step and step.get("endpoint") == "aggregations/ingest":
+            splits.append(step["ndjson"])
+    assert len(splits) == 2 and len(splits[0]) == 5 and len(splits[1]) == 5
+
+    wanted = {
+        # scenario index (0-based across the YAML docs) -> fixture name; only
+        # cases within the round-1 subset (single-valued fields, no
+        # percentiles/cardinality/split_size error bounds).
+        "date_histogram_basic": 0,
+        "date_histogram_extended_bounds": 1,
+        "date_histogram_stats_subagg": 2,
+        "date_histogram_stats_subagg_exists_filter": 3,
+        "terms_full": 5,
+        "histogram_interval50": 11,
+    }
+    steps = list(yaml.safe_load_all(open(tests_path)))
+    cases = {}
+    for name, idx in wanted.items():
+        step = steps[idx]
+        cases[name] = {
+            "request": step["json"],
+            "expected": step["expected"]["aggregations"],
+        }
+    # terms_full also asserts the multi-valued "tags" field; strip it (multi-
+    # valued fast fields are a later round) but keep hosts + note the removal.
+    tf = cases["terms_full"]
+    tf["request"]["aggs"].pop("tags", None)
+    tf["expected"].pop("tags", None)
+    tf["note"] = "tags (multi-valued) case removed: multi-valued fast fields not in r1 scope"
+    return {
+        "source": "quickwit/rest-api-tests/scenarii/aggregations/{_setup.quickwit.yaml,0001-aggregations.yaml}",
+        "schema": [
+            {"name": "date", "type": "datetime", "fast": True, "nullable": True},
+            {"name": "response", "type": "u64", "fast": True, "nullable": True},
+            {"name": "name", "type": "str_fast", "fast": True, "nullable": True},
+            {"name": "host", "type": "str_fast", "fast": True, "nullable": True},
+            {"name": "id", "type": "u64", "fast": True, "nullable": True},
+        ],
+        "splits": splits,
+        "cases": cases,
+    }
+
+
+def main():
+    if not os.path.isdir(REF):
+        sys.exit("reference checkout not present; fixtures must already be committed")
+    with open(os.path.join(OUT, "bm25_sort.json"), "w") as f:
+        json.dump(bm25_golden(), f, indent=1, sort_keys=True)
+        f.write("\n")
+    with open(os.path.join(OUT, "aggregations.json"), "w") as f:
+        json.dump(agg_golden(), f, indent=1, sort_keys=True)
+        f.write("\n")
+    print("wrote bm25_sort.json, aggregations.json")
+
+
+if __name__ == "__main__":
+    main()
