@@ -1,0 +1,466 @@
+"""QWA1 split container writer + synthetic corpus generator.
+
+Format spec: DESIGN.md §3. This is indexing-side tooling (the reference's
+counterpart is the quickwit-indexing pipeline + tantivy's segment writer,
+which are out of scope as services — SURVEY.md §2): the build needs only a
+deterministic split writer for tests and benchmarks.
+
+Layout is GPU-first: 128-doc bitpacked posting blocks with 16-byte skip
+entries, raw fixed-width fast-field columns, u8 fieldnorms, all sections
+64-byte aligned so the file image can be uploaded to HBM as-is and indexed by
+offset (tantivy-*shaped*: same structures the reference's hot loops walk —
+SURVEY.md §8a rows "posting block decode" / "fast-field predicate" — with our
+own encodings, since tantivy's are unpinned).
+"""
+import datetime as _dt
+import json
+
+import numpy as np
+
+from .fieldnorm import norm_to_id
+
+MAGIC = b"QWAMDSP1"
+FOOTER_MAGIC = b"QWA1FOOT"
+ALIGN = 64
+BLOCK = 128
+
+SKIP_DTYPE = np.dtype(
+    [
+        ("first_doc", "<u4"),
+        ("last_doc", "<u4"),
+        ("word_off", "<u4"),  # u32-word offset into the field's payload section
+        ("id_bits", "u1"),
+        ("tf_bits", "u1"),  # 0 = record:basic (no tf section)
+        ("count", "<u2"),
+    ]
+)
+assert SKIP_DTYPE.itemsize == 16
+
+
+# ---------------------------------------------------------------- tokenizers
+def tokenize(text: str, tokenizer: str):
+    """'raw' = whole string, verbatim (tantivy raw tokenizer). 'default' =
+    alphanumeric runs, lowercased, tokens longer than 40 chars dropped
+    (tantivy default = SimpleTokenizer + RemoveLongFilter(40) + LowerCaser)."""
+    if text is None:
+        return []
+    if tokenizer == "raw":
+        return [text] if text else []
+    toks, cur = [], []
+    for ch in text:
+        if ch.isalnum():
+            cur.append(ch.lower())
+        elif cur:
+            toks.append("".join(cur))
+            cur = []
+    if cur:
+        toks.append("".join(cur))
+    return [t for t in toks if len(t) <= 40]
+
+
+# ------------------------------------------------------------- bit packing
+def _bit_width(vals: np.ndarray) -> np.ndarray:
+    """Per-row max bit width (>=1) of a (nblk, BLOCK) uint64 matrix."""
+    m = vals.max(axis=1)
+    bits = np.ones(len(m), dtype=np.uint8)
+    for k in range(1, 33):
+        bits += (m >= (np.uint64(1) << np.uint64(k))).astype(np.uint8)
+    return bits
+
+
+def _pack_group(vals: np.ndarray, w: int) -> np.ndarray:
+    """Pack (nb, BLOCK) uint64 values of width w into (nb, ceil(BLOCK*w/64))
+    u64 words, LSB-first."""
+    nb = vals.shape[0]
+    nw = (BLOCK * w + 63) // 64
+    packed = np.zeros((nb, nw), dtype=np.uint64)
+    for j in range(BLOCK):
+        p = j * w
+        wd, sh = p >> 6, p & 63
+        packed[:, wd] |= vals[:, j] << np.uint64(sh)
+        if sh + w > 64:
+            packed[:, wd + 1] |= vals[:, j] >> np.uint64(64 - sh)
+    return packed
+
+
+def _pack_blocks(vals: np.ndarray, widths: np.ndarray, words64_off: np.ndarray, out: np.ndarray):
+    """Scatter-pack each block i (vals[i], width widths[i]) into
+    out[words64_off[i] : words64_off[i] + nwords64(widths[i])]."""
+    for w in np.unique(widths):
+        sel = np.nonzero(widths == w)[0]
+        packed = _pack_group(vals[sel], int(w))
+        nw = packed.shape[1]
+        idx = (words64_off[sel][:, None] + np.arange(nw, dtype=np.int64)[None, :]).ravel()
+        out[idx] = packed.ravel()
+
+
+# ------------------------------------------------------------- the writer
+class _Sections:
+    def __init__(self):
+        self.chunks = [MAGIC, b"\0" * (ALIGN - len(MAGIC))]
+        self.pos = ALIGN
+
+    def add(self, data) -> list:
+        b = data.tobytes() if isinstance(data, np.ndarray) else bytes(data)
+        off = self.pos
+        self.chunks.append(b)
+        self.pos += len(b)
+        pad = (-self.pos) % ALIGN
+        if pad:
+            self.chunks.append(b"\0" * pad)
+            self.pos += pad
+        return [off, len(b)]
+
+
+def _build_text_field(sec, fspec, num_docs, term_ids, doc_ids, tfs, vocab, doc_lengths):
+    """term_ids/doc_ids/tfs: postings sorted by (term, doc). vocab: list of
+    term strings ordered by term id == sorted order. Returns meta dict."""
+    record_freq = fspec.get("record", "basic") == "freq"
+    nterms = len(vocab)
+    df = np.bincount(term_ids, minlength=nterms).astype(np.int64)
+    term_start = np.concatenate([[0], np.cumsum(df)])
+    nblk_t = (df + BLOCK - 1) // BLOCK
+    total_blocks = int(nblk_t.sum())
+    npost = len(doc_ids)
+
+    if total_blocks:
+        blk_term = np.repeat(np.arange(nterms), nblk_t)
+        blk_local = np.arange(total_blocks) - np.repeat(
+            np.concatenate([[0], np.cumsum(nblk_t)[:-1]]), nblk_t
+        )
+        blk_start = term_start[blk_term] + blk_local * BLOCK
+        blk_count = np.minimum(df[blk_term] - blk_local * BLOCK, BLOCK).astype(np.int64)
+        first_doc = doc_ids[blk_start]
+        last_doc = doc_ids[blk_start + blk_count - 1]
+
+        gaps = np.diff(doc_ids.astype(np.int64), prepend=0)
+        gaps[blk_start] = 0
+        blk_of_p = np.repeat(np.arange(total_blocks), blk_count)
+        col_of_p = np.arange(npost) - blk_start[blk_of_p]
+        vals = np.zeros((total_blocks, BLOCK), dtype=np.uint64)
+        vals[blk_of_p, col_of_p] = gaps.astype(np.uint64)
+        id_bits = _bit_width(vals)
+        id_w64 = (BLOCK * id_bits.astype(np.int64) + 63) // 64
+
+        if record_freq:
+            tvals = np.zeros((total_blocks, BLOCK), dtype=np.uint64)
+            tvals[blk_of_p, col_of_p] = (tfs - 1).astype(np.uint64)
+            tf_bits = _bit_width(tvals)
+            tf_w64 = (BLOCK * tf_bits.astype(np.int64) + 63) // 64
+        else:
+            tf_bits = np.zeros(total_blocks, dtype=np.uint8)
+            tf_w64 = np.zeros(total_blocks, dtype=np.int64)
+
+        blk_w64 = id_w64 + tf_w64
+        blk_off64 = np.concatenate([[0], np.cumsum(blk_w64)[:-1]])
+        payload = np.zeros(int(blk_w64.sum()), dtype=np.uint64)
+        _pack_blocks(vals, id_bits, blk_off64, payload)
+        if record_freq:
+            _pack_blocks(tvals, tf_bits, blk_off64 + id_w64, payload)
+
+        skip = np.zeros(total_blocks, dtype=SKIP_DTYPE)
+        skip["first_doc"] = first_doc
+        skip["last_doc"] = last_doc
+        skip["word_off"] = (blk_off64 * 2).astype(np.uint32)  # u64 -> u32 words
+        skip["id_bits"] = id_bits
+        skip["tf_bits"] = tf_bits
+        skip["count"] = blk_count.astype(np.uint16)
+
+        term_first_blk = np.concatenate([[0], np.cumsum(nblk_t)[:-1]])
+        posting_off = (blk_off64[np.minimum(term_first_blk, total_blocks - 1)] * 8).astype(
+            np.uint64
+        )
+        posting_off[df == 0] = 0
+        skip_off = (term_first_blk * 16).astype(np.uint64)
+    else:
+        payload = np.zeros(0, dtype=np.uint64)
+        skip = np.zeros(0, dtype=SKIP_DTYPE)
+        posting_off = np.zeros(nterms, dtype=np.uint64)
+        skip_off = np.zeros(nterms, dtype=np.uint64)
+
+    term_bytes = "".join(vocab).encode("utf-8")
+    term_offsets = np.zeros(nterms + 1, dtype=np.uint32)
+    np.cumsum([len(t.encode("utf-8")) for t in vocab], out=term_offsets[1:])
+
+    norms = norm_to_id(doc_lengths) if fspec.get("fieldnorms", True) else None
+
+    meta = {
+        "name": fspec["name"],
+        "type": "text",
+        "tokenizer": fspec.get("tokenizer", "default"),
+        "record": "freq" if record_freq else "basic",
+        "fieldnorms": norms is not None,
+        "total_tokens": int(doc_lengths.sum()),
+        "num_terms": nterms,
+        "sec": {
+            "term_offsets": sec.add(term_offsets),
+            "term_bytes": sec.add(np.frombuffer(term_bytes, dtype=np.uint8)),
+            "posting_off": sec.add(posting_off),
+            "doc_freq": sec.add(df.astype(np.uint32)),
+            "n_blocks": sec.add(nblk_t.astype(np.uint32)),
+            "skip_off": sec.add(skip_off),
+            "skip": sec.add(skip),
+            "payload": sec.add(payload),
+        },
+    }
+    if norms is not None:
+        meta["sec"]["fieldnorms"] = sec.add(norms)
+    return meta
+
+
+def _build_fast_field(sec, fspec, num_docs, values, present):
+    """values: list aligned to docs (None where absent) or np array.
+    present: bool array or None (all present)."""
+    ftype = fspec["type"]
+    meta = {"name": fspec["name"], "type": ftype, "nullable": False, "sec": {}}
+    if present is None:
+        present = np.ones(num_docs, dtype=bool)
+    else:
+        present = np.asarray(present, dtype=bool)
+    nullable = not present.all()
+    meta["nullable"] = bool(nullable)
+
+    if ftype == "str":
+        if isinstance(values, tuple) and values[0] == "ords":
+            # bulk path: precomputed (ords ndarray, sorted vocab list)
+            _, raw_ords, strs = values
+            assert list(strs) == sorted(strs)
+            card = len(strs)
+            width = 1 if card <= 0xFF else (2 if card <= 0xFFFF else 4)
+            ords = np.asarray(raw_ords).astype(
+                {1: np.uint8, 2: np.uint16, 4: np.uint32}[width]
+            )
+        else:
+            strs = sorted({v for v, p in zip(values, present) if p})
+            dct = {s: i for i, s in enumerate(strs)}
+            card = len(strs)
+            width = 1 if card <= 0xFF else (2 if card <= 0xFFFF else 4)
+            ords = np.zeros(num_docs, dtype={1: np.uint8, 2: np.uint16, 4: np.uint32}[width])
+            for d, (v, p) in enumerate(zip(values, present)):
+                if p:
+                    ords[d] = dct[v]
+        dict_bytes = "".join(strs).encode("utf-8")
+        dict_offsets = np.zeros(card + 1, dtype=np.uint32)
+        np.cumsum([len(s.encode("utf-8")) for s in strs], out=dict_offsets[1:])
+        meta.update({"cardinality": card, "ord_width": width})
+        meta["sec"]["values"] = sec.add(ords)
+        meta["sec"]["dict_offsets"] = sec.add(dict_offsets)
+        meta["sec"]["dict_bytes"] = sec.add(np.frombuffer(dict_bytes, dtype=np.uint8))
+    else:
+        dt = np.uint64 if ftype == "u64" else np.int64  # i64/datetime -> i64
+        col = np.zeros(num_docs, dtype=dt)
+        vv = np.asarray(
+            [v if p else 0 for v, p in zip(values, present)]
+            if not isinstance(values, np.ndarray)
+            else np.where(present, values, 0),
+            dtype=dt,
+        )
+        col[:] = vv
+        pv = col[present]
+        meta["min_value"] = int(pv.min()) if len(pv) else 0
+        meta["max_value"] = int(pv.max()) if len(pv) else 0
+        meta["sec"]["values"] = sec.add(col)
+    if nullable:
+        bits = np.zeros((num_docs + 63) // 64, dtype=np.uint64)
+        idx = np.nonzero(present)[0]
+        np.bitwise_or.at(bits, idx >> 6, np.uint64(1) << (idx & 63).astype(np.uint64))
+        meta["sec"]["nulls"] = sec.add(bits)
+    return meta
+
+
+def parse_datetime_ms(v) -> int:
+    """Accept RFC3339 strings or unix seconds/ms ints (hdfs-logs uses
+    unix_timestamp input; the agg golden corpus uses rfc3339)."""
+    if isinstance(v, str):
+        d = _dt.datetime.fromisoformat(v.replace("Z", "+00:00"))
+        return int(d.timestamp() * 1000)
+    v = int(v)
+    # unix seconds unless the magnitude says millis (quickwit's
+    # unix_timestamp input format guesses the unit the same way)
+    return v * 1000 if abs(v) < 10**12 else v
+
+
+class SplitWriter:
+    """Document-at-a-time writer for small corpora (golden tests).
+    Bulk columnar path: build_split_from_columns."""
+
+    def __init__(self, schema: dict, split_id: str):
+        self.schema = schema
+        self.split_id = split_id
+        self.docs = []
+
+    def add_documents(self, docs):
+        self.docs.extend(docs)
+
+    def finalize(self) -> bytes:
+        num_docs = len(self.docs)
+        text_inputs = {}
+        fast_inputs = {}
+        for fspec in self.schema["fields"]:
+            name = fspec["name"]
+            if fspec["type"] == "text":
+                tok = fspec.get("tokenizer", "default")
+                per_doc = [tokenize(d.get(name), tok) if name in d else [] for d in self.docs]
+                text_inputs[name] = per_doc
+            elif fspec.get("fast", False):
+                vals, pres = [], []
+                for d in self.docs:
+                    v = d.get(name)
+                    pres.append(v is not None)
+                    if fspec["type"] == "datetime" and v is not None:
+                        v = parse_datetime_ms(v)
+                    vals.append(v if v is not None else 0)
+                fast_inputs[name] = (vals, np.array(pres, dtype=bool))
+        return _assemble(self.schema, self.split_id, num_docs, text_inputs, fast_inputs)
+
+
+def _postings_from_tokens(per_doc_tokens):
+    """per_doc_tokens: list of token lists -> (vocab, term_ids, doc_ids, tfs)
+    sorted by (term, doc), plus doc token counts."""
+    vocab = sorted({t for toks in per_doc_tokens for t in toks})
+    tid = {t: i for i, t in enumerate(vocab)}
+    tuples = []
+    for d, toks in enumerate(per_doc_tokens):
+        counts = {}
+        for t in toks:
+            counts[t] = counts.get(t, 0) + 1
+        for t, c in counts.items():
+            tuples.append((tid[t], d, c))
+    tuples.sort()
+    if tuples:
+        arr = np.array(tuples, dtype=np.int64)
+        term_ids, doc_ids, tfs = arr[:, 0], arr[:, 1].astype(np.uint32), arr[:, 2].astype(np.uint32)
+    else:
+        term_ids = np.zeros(0, np.int64)
+        doc_ids = np.zeros(0, np.uint32)
+        tfs = np.zeros(0, np.uint32)
+    lengths = np.array([len(t) for t in per_doc_tokens], dtype=np.int64)
+    return vocab, term_ids, doc_ids, tfs, lengths
+
+
+def _assemble(schema, split_id, num_docs, text_inputs, fast_inputs, precomputed_text=None):
+    sec = _Sections()
+    fields_meta = []
+    for fspec in schema["fields"]:
+        name = fspec["name"]
+        if fspec["type"] == "text":
+            if precomputed_text and name in precomputed_text:
+                vocab, term_ids, doc_ids, tfs, lengths = precomputed_text[name]
+            else:
+                vocab, term_ids, doc_ids, tfs, lengths = _postings_from_tokens(
+                    text_inputs[name]
+                )
+            fields_meta.append(
+                _build_text_field(sec, fspec, num_docs, term_ids, doc_ids, tfs, vocab, lengths)
+            )
+        elif fspec.get("fast", False):
+            vals, pres = fast_inputs[name]
+            fields_meta.append(_build_fast_field(sec, fspec, num_docs, vals, pres))
+
+    meta = {
+        "format": "QWA1",
+        "version": 1,
+        "split_id": split_id,
+        "num_docs": int(num_docs),
+        "timestamp_field": schema.get("timestamp_field"),
+        "fields": fields_meta,
+    }
+    meta_b = json.dumps(meta, sort_keys=True).encode("utf-8")
+    meta_off = sec.pos
+    sec.chunks.append(meta_b)
+    sec.pos += len(meta_b)
+    footer = np.array([meta_off, len(meta_b)], dtype="<u8").tobytes() + FOOTER_MAGIC
+    sec.chunks.append(footer)
+    return b"".join(sec.chunks)
+
+
+def build_split_from_columns(schema, split_id, num_docs, text_token_ids, text_vocabs, fast_columns):
+    """Bulk path for synthetic corpora.
+
+    text_token_ids[name]: int32 (num_docs, L) matrix of ids into
+    text_vocabs[name] (a sorted list of token strings); -1 = empty slot.
+    fast_columns[name]: (values ndarray, present ndarray or None).
+    """
+    precomputed = {}
+    for name, mat in text_token_ids.items():
+        vocab = text_vocabs[name]
+        assert list(vocab) == sorted(vocab)
+        n, L = mat.shape
+        doc_of = np.broadcast_to(np.arange(n, dtype=np.int64)[:, None], mat.shape)
+        mask = mat >= 0
+        terms = mat[mask].astype(np.int64)
+        docs = doc_of[mask]
+        lengths = mask.sum(axis=1).astype(np.int64)
+        key = terms * np.int64(num_docs) + docs
+        ukey, tfs = np.unique(key, return_counts=True)
+        term_ids = ukey // num_docs
+        doc_ids = (ukey % num_docs).astype(np.uint32)
+        precomputed[name] = (list(vocab), term_ids, doc_ids, tfs.astype(np.uint32), lengths)
+    fast_inputs = {k: (v[0], v[1]) for k, v in fast_columns.items()}
+    return _assemble(schema, split_id, num_docs, {}, fast_inputs, precomputed_text=precomputed)
+
+
+# ------------------------------------------------------- synthetic corpus
+HDFS_SCHEMA = {
+    "timestamp_field": "timestamp",
+    "fields": [
+        {"name": "timestamp", "type": "datetime", "fast": True},
+        {"name": "tenant_id", "type": "u64", "fast": True},
+        {"name": "tenant_name", "type": "str", "fast": True},
+        {"name": "severity_text", "type": "text", "tokenizer": "raw", "record": "basic",
+         "fieldnorms": True},
+        {"name": "body", "type": "text", "tokenizer": "default", "record": "freq",
+         "fieldnorms": True},
+    ],
+}
+
+SEVERITIES = ["DEBUG", "INFO", "WARN", "ERROR", "FATAL"]
+SEVERITY_P = [0.50, 0.40, 0.07, 0.029, 0.001]
+BODY_VOCAB_SIZE = 10_000
+BODY_TOKENS_PER_DOC = 10
+N_TENANTS = 1_000
+T0_EPOCH_S = 1_700_000_000  # synthetic 30-day window start
+
+
+def body_vocab():
+    return ["w%05d" % i for i in range(BODY_VOCAB_SIZE)]  # already sorted
+
+
+def zipf_probs(n, s=1.0):
+    p = 1.0 / np.arange(1, n + 1) ** s
+    return p / p.sum()
+
+
+def generate_split(split_ord: int, num_docs: int, seed: int = 42) -> bytes:
+    """Seeded hdfs-logs-style synthetic split: timestamp
+    uniform over 30 days, tenant_id Zipf over 1k tenants, severity_text from
+    the fixed distribution, body = 10 tokens/doc Zipf over a 10k vocabulary.
+    Token ids are assigned so that df rank follows the Zipf draw (vocab is
+    sorted lexically; id order == lexical order by construction of w%05d)."""
+    rng = np.random.default_rng(np.random.SeedSequence([seed, split_ord]))
+    ts_ms = (
+        T0_EPOCH_S + rng.integers(0, 30 * 86400, size=num_docs, dtype=np.int64)
+    ) * 1000
+    tenants = rng.choice(N_TENANTS, size=num_docs, p=zipf_probs(N_TENANTS)).astype(np.uint64)
+    sev = rng.choice(len(SEVERITIES), size=num_docs, p=SEVERITY_P).astype(np.int32)
+    body = rng.choice(
+        BODY_VOCAB_SIZE, size=(num_docs, BODY_TOKENS_PER_DOC), p=zipf_probs(BODY_VOCAB_SIZE)
+    ).astype(np.int32)
+
+    sev_vocab = sorted(SEVERITIES)
+    sev_map = np.array([sev_vocab.index(s) for s in SEVERITIES], dtype=np.int32)
+    sev_ids = sev_map[sev][:, None]  # (N,1) single raw token per doc
+
+    tenant_names = ["t%04d" % i for i in range(N_TENANTS)]
+    return build_split_from_columns(
+        HDFS_SCHEMA,
+        f"synthetic-{seed}-{split_ord:04d}",
+        num_docs,
+        {"severity_text": sev_ids, "body": body},
+        {"severity_text": sev_vocab, "body": body_vocab()},
+        {
+            "timestamp": (ts_ms, None),
+            "tenant_id": (tenants, None),
+            "tenant_name": (("ords", tenants, tenant_names), None),
+        },
+    )
