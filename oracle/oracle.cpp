@@ -1,0 +1,798 @@
+// ============================================================================
+// ORACLE — TEST INFRASTRUCTURE ONLY.
+//
+// CPU restatement of Quickwit's per-split leaf-search semantics
+// (quickwit/quickwit-search/src/leaf.rs:655-970 and the tantivy 0.27 @
+// 86641f7 execution it delegates to — tantivy is a non-vendored dependency,
+// so its algorithms are restated from its published behavior and pinned by
+// the reference's own golden vectors, tests/golden/*).
+//
+// Only tests/, __graft_entry__.smoke() and bench.py's cpu_baseline leg may
+// call into this library. The product path (libquickwit_amd.so) never routes
+// through it and fails loudly without a GPU.
+//
+// What is restated where:
+//  - posting decode: our QWA1 blocks, scalar loop here;
+//  - boolean combination: tantivy BooleanQuery/union/intersection semantics
+//    (tantivy_query_ast.rs:153-374 for construction; minimum_should_match
+//    default: >=1 should when no must/filter clause);
+//  - BM25: Lucene BM25 with the (k1+1) factor, k1=1.2 b=0.75,
+//    idf = ln(1+(N-df+0.5)/(df+0.5)), fieldnorm-quantized |d| — pinned by
+//    tests.rs:600-691 golden scores (tests/golden/bm25_sort.json);
+//  - top-K + tie-breaks: top_k_collector.rs + docs/internals/sorting.md:14-26
+//    ((sort_value, sort_value2), None last, GlobalDocId tie in order1);
+//  - aggregations: tantivy aggregation semantics pinned by
+//    rest-api-tests/scenarii/aggregations (tests/golden/aggregations.json);
+//  - per-request orchestration: leaf_search_single_split → one result per
+//    split, merged with IncrementalCollector semantics (collector.rs:1198).
+//
+// OpenMP parallelism is across splits only (one thread per split), mirroring
+// the reference's rayon pool with single-threaded per-split closures
+// (quickwit-search/src/lib.rs:100-104). This parallel variant is the timed
+// CPU baseline of bench.py (cpu_baseline.kind = "port").
+// ============================================================================
+#include <omp.h>
+
+#include <algorithm>
+#include <chrono>
+#include <cmath>
+#include <cstdint>
+#include <cstdlib>
+#include <cstring>
+#include <map>
+#include <memory>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#include "../quickwit_amd/csrc/fieldnorm.h"
+#include "../quickwit_amd/csrc/minijson.h"
+#include "../quickwit_amd/csrc/pb.h"
+#include "../quickwit_amd/csrc/qagg_format.h"
+#include "../quickwit_amd/csrc/qast.h"
+#include "../quickwit_amd/csrc/qsplit.h"
+#include "../quickwit_amd/csrc/sortkey.h"
+
+namespace qw {
+namespace oracle {
+
+// --------------------------------------------------------------- decode
+// Scalar bit-unpack of one posting block (QWA1).
+static void decode_block(const uint32_t* payload, const SkipEntry& e, uint32_t* docs,
+                         uint32_t* tfs) {
+    const uint32_t* base = payload + e.word_off;
+    uint32_t w = e.id_bits;
+    uint64_t mask = w >= 64 ? ~0ULL : ((1ULL << w) - 1);
+    uint32_t doc = e.first_doc;
+    for (uint32_t j = 0; j < e.count; ++j) {
+        uint64_t bitpos = uint64_t(j) * w;
+        uint64_t word = bitpos >> 5;
+        uint32_t sh = uint32_t(bitpos & 31);
+        uint64_t v = (uint64_t(base[word]) | (uint64_t(base[word + 1]) << 32)) >> sh;
+        if (sh + w > 64) v |= uint64_t(base[word + 2]) << (64 - sh);
+        doc += uint32_t(v & mask);
+        docs[j] = doc;
+    }
+    if (e.tf_bits == 0) {
+        for (uint32_t j = 0; j < e.count; ++j) tfs[j] = 1;
+        return;
+    }
+    const uint32_t* tbase = base + 2 * ((128 * uint32_t(e.id_bits) + 63) / 64);
+    w = e.tf_bits;
+    mask = (1ULL << w) - 1;
+    for (uint32_t j = 0; j < e.count; ++j) {
+        uint64_t bitpos = uint64_t(j) * w;
+        uint64_t word = bitpos >> 5;
+        uint32_t sh = uint32_t(bitpos & 31);
+        uint64_t v = (uint64_t(tbase[word]) | (uint64_t(tbase[word + 1]) << 32)) >> sh;
+        if (sh + w > 64) v |= uint64_t(tbase[word + 2]) << (64 - sh);
+        tfs[j] = uint32_t(v & mask) + 1;
+    }
+}
+
+struct Postings {
+    std::vector<uint32_t> docs;
+    std::vector<uint32_t> tfs;
+};
+
+static Postings decode_term(const TextFieldView& f, int64_t tid) {
+    Postings p;
+    uint32_t df = f.h_doc_freq[tid];
+    uint32_t nblk = f.h_n_blocks[tid];
+    p.docs.resize(df);
+    p.tfs.resize(df);
+    const SkipEntry* skip = f.h_skip + f.h_skip_off[tid] / 16;
+    uint32_t out = 0;
+    for (uint32_t b = 0; b < nblk; ++b) {
+        decode_block(f.h_payload, skip[b], p.docs.data() + out, p.tfs.data() + out);
+        out += skip[b].count;
+    }
+    return p;
+}
+
+// --------------------------------------------------------------- match algebra
+struct Match {
+    bool all = false;  // matches every doc (scores empty)
+    std::vector<uint32_t> docs;
+    std::vector<float> scores;  // empty when not scored
+    size_t size(uint32_t num_docs) const { return all ? num_docs : docs.size(); }
+};
+
+static Match intersect(const Match& a, const Match& b) {
+    if (a.all) return b;
+    if (b.all) return a;
+    Match r;
+    bool sa = !a.scores.empty(), sb = !b.scores.empty();
+    if (sa || sb) r.scores.reserve(std::min(a.docs.size(), b.docs.size()));
+    r.docs.reserve(std::min(a.docs.size(), b.docs.size()));
+    size_t i = 0, j = 0;
+    while (i < a.docs.size() && j < b.docs.size()) {
+        if (a.docs[i] < b.docs[j]) ++i;
+        else if (b.docs[j] < a.docs[i]) ++j;
+        else {
+            r.docs.push_back(a.docs[i]);
+            if (sa || sb)
+                r.scores.push_back((sa ? a.scores[i] : 0.f) + (sb ? b.scores[j] : 0.f));
+            ++i;
+            ++j;
+        }
+    }
+    return r;
+}
+
+static Match subtract(Match a, const Match& b, uint32_t num_docs) {
+    if (b.all) return Match{};
+    if (b.docs.empty()) return a;
+    if (a.all) {
+        Match r;
+        r.docs.reserve(num_docs - b.docs.size());
+        size_t j = 0;
+        for (uint32_t d = 0; d < num_docs; ++d) {
+            while (j < b.docs.size() && b.docs[j] < d) ++j;
+            if (j < b.docs.size() && b.docs[j] == d) continue;
+            r.docs.push_back(d);
+        }
+        return r;
+    }
+    Match r;
+    r.docs.reserve(a.docs.size());
+    bool sa = !a.scores.empty();
+    if (sa) r.scores.reserve(a.docs.size());
+    size_t j = 0;
+    for (size_t i = 0; i < a.docs.size(); ++i) {
+        uint32_t d = a.docs[i];
+        while (j < b.docs.size() && b.docs[j] < d) ++j;
+        if (j < b.docs.size() && b.docs[j] == d) continue;
+        r.docs.push_back(d);
+        if (sa) r.scores.push_back(a.scores[i]);
+    }
+    return r;
+}
+
+// k-way union; keeps docs matched by >= msm clauses; sums scores of matching
+// scoring clauses (tantivy BufferedUnionScorer semantics)
+static Match union_n(const std::vector<Match>& ms, size_t msm, uint32_t num_docs) {
+    (void)num_docs;
+    Match r;
+    if (msm == 0) msm = 1;
+    for (auto& m : ms)
+        if (m.all)
+            throw std::runtime_error("union over a match_all clause not supported here");
+    size_t k = ms.size();
+    std::vector<size_t> idx(k, 0);
+    bool scored = false;
+    for (auto& m : ms) scored |= !m.scores.empty();
+    while (true) {
+        uint32_t best = UINT32_MAX;
+        for (size_t c = 0; c < k; ++c) {
+            const Match& m = ms[c];
+            if (idx[c] < m.docs.size()) best = std::min(best, m.docs[idx[c]]);
+        }
+        if (best == UINT32_MAX) break;
+        size_t cnt = 0;
+        float sc = 0;
+        for (size_t c = 0; c < k; ++c) {
+            const Match& m = ms[c];
+            if (idx[c] < m.docs.size() && m.docs[idx[c]] == best) {
+                ++cnt;
+                if (!m.scores.empty()) sc += m.scores[idx[c]];
+                ++idx[c];
+            }
+        }
+        if (cnt >= msm) {
+            r.docs.push_back(best);
+            if (scored) r.scores.push_back(sc);
+        }
+    }
+    return r;
+}
+
+// --------------------------------------------------------------- evaluator
+struct SplitSearcher {
+    const SplitView& sv;
+    const Schema& schema;
+    bool scoring;
+
+    Match eval(const PlanNode& n) const {
+        switch (n.kind) {
+            case PlanNode::MATCH_ALL: {
+                Match m;
+                m.all = true;
+                return m;
+            }
+            case PlanNode::MATCH_NONE:
+                return Match{};
+            case PlanNode::TERM:
+                return eval_term(n);
+            case PlanNode::RANGE:
+                return eval_range(n);
+            case PlanNode::FIELD_PRESENCE:
+                return eval_presence(n);
+            case PlanNode::BOOL:
+                return eval_bool(n);
+        }
+        return Match{};
+    }
+
+    Match eval_term(const PlanNode& n) const {
+        Match m;
+        const TextFieldView* f = sv.text_field(n.field);
+        if (!f) return m;
+        int64_t tid = f->find_term(n.value.data(), n.value.size());
+        if (tid < 0) return m;
+        Postings p = decode_term(*f, tid);
+        m.docs = std::move(p.docs);
+        if (scoring) {
+            // BM25 (restated; golden-pinned — see file header). Weight in
+            // f64, per-posting arithmetic in f32 exactly as the GPU kernel
+            // computes it: score = W * tf / (tf + K[normid]).
+            double N = double(sv.num_docs);
+            double df = double(f->h_doc_freq[tid]);
+            double idf = std::log(1.0 + (N - df + 0.5) / (df + 0.5));
+            const double K1 = 1.2, B = 0.75;
+            float W = float(idf * (1.0 + K1) * double(n.boost));
+            double avgdl = f->total_tokens > 0 && sv.num_docs > 0
+                               ? double(f->total_tokens) / double(sv.num_docs)
+                               : 0.0;
+            float Ktab[256];
+            for (int i = 0; i < 256; ++i) {
+                double fn = double(FIELDNORM_TABLE.v[i]);
+                Ktab[i] = float(K1 * (1.0 - B + B * fn / (avgdl > 0 ? avgdl : 1.0)));
+            }
+            m.scores.resize(m.docs.size());
+            const uint8_t* norms = f->h_fieldnorms;
+            for (size_t i = 0; i < m.docs.size(); ++i) {
+                float tf = float(p.tfs[i]);
+                float K = norms ? Ktab[norms[m.docs[i]]] : Ktab[fieldnorm_encode(1)];
+                m.scores[i] = W * (tf / (tf + K));
+            }
+        }
+        return m;
+    }
+
+    bool range_test(const FastFieldView& f, uint32_t d, const Bound& lo,
+                    const Bound& hi) const {
+        if (!f.present(d)) return false;
+        if (f.type == FastFieldView::U64) {
+            uint64_t v = f.u64(d);
+            if (lo.kind != Bound::UNBOUNDED) {
+                uint64_t b = uint64_t(lo.ival);
+                if (lo.kind == Bound::INCLUDED ? v < b : v <= b) return false;
+            }
+            if (hi.kind != Bound::UNBOUNDED) {
+                uint64_t b = uint64_t(hi.ival);
+                if (hi.kind == Bound::INCLUDED ? v > b : v >= b) return false;
+            }
+            return true;
+        }
+        int64_t v = f.i64(d);
+        if (lo.kind != Bound::UNBOUNDED &&
+            (lo.kind == Bound::INCLUDED ? v < lo.ival : v <= lo.ival))
+            return false;
+        if (hi.kind != Bound::UNBOUNDED &&
+            (hi.kind == Bound::INCLUDED ? v > hi.ival : v >= hi.ival))
+            return false;
+        return true;
+    }
+
+    Match eval_range(const PlanNode& n) const {
+        Match m;
+        const FastFieldView* f = sv.fast_field(n.field);
+        if (!f) return m;
+        for (uint32_t d = 0; d < sv.num_docs; ++d)
+            if (range_test(*f, d, n.lo, n.hi)) m.docs.push_back(d);
+        return m;
+    }
+
+    Match eval_presence(const PlanNode& n) const {
+        Match m;
+        const FastFieldView* f = sv.fast_field(n.field);
+        if (!f) return m;
+        if (!f->nullable) {
+            m.all = true;
+            return m;
+        }
+        for (uint32_t d = 0; d < sv.num_docs; ++d)
+            if (f->present(d)) m.docs.push_back(d);
+        return m;
+    }
+
+    Match eval_bool(const PlanNode& n) const {
+        std::vector<Match> req;
+        for (auto& c : n.must) req.push_back(eval(c));
+        std::vector<Match> filt;
+        for (auto& c : n.filter) {
+            Match m = eval(c);
+            m.scores.clear();  // filter clauses never score
+            filt.push_back(std::move(m));
+        }
+        std::vector<Match> shoulds;
+        for (auto& c : n.should) shoulds.push_back(eval(c));
+
+        Match base;
+        bool has_req = !req.empty() || !filt.empty();
+        if (has_req) {
+            base.all = true;
+            // sum must scores: intersect accumulates scores
+            for (auto& m : req) base = intersect(base, m);
+            for (auto& m : filt) base = intersect(base, m);
+            int64_t msm = n.minimum_should_match < 0 ? 0 : n.minimum_should_match;
+            if (!shoulds.empty()) {
+                bool should_all = false;
+                for (auto& s : shoulds) should_all |= s.all;
+                if (msm > 0) {
+                    if (should_all && shoulds.size() == 1) {
+                        // match_all should: always satisfied
+                    } else {
+                        Match su = union_n(shoulds, size_t(msm), sv.num_docs);
+                        Match su_noscore = su;
+                        su_noscore.scores.clear();
+                        base = intersect(base, su_noscore);
+                    }
+                }
+                if (scoring) {
+                    // add should scores for docs already in base (merge-join)
+                    Match su = union_n(shoulds, 1, sv.num_docs);
+                    if (!su.scores.empty()) {
+                        if (base.scores.empty()) base.scores.resize(base.docs.size(), 0.f);
+                        size_t j = 0;
+                        for (size_t i = 0; i < base.docs.size(); ++i) {
+                            while (j < su.docs.size() && su.docs[j] < base.docs[i]) ++j;
+                            if (j < su.docs.size() && su.docs[j] == base.docs[i])
+                                base.scores[i] += su.scores[j];
+                        }
+                    }
+                }
+            }
+        } else {
+            size_t msm = n.minimum_should_match < 0 ? 1 : size_t(n.minimum_should_match);
+            bool any_all = false;
+            for (auto& s : shoulds) any_all |= s.all;
+            if (any_all) {
+                if (shoulds.size() == 1) base = shoulds[0];
+                else {
+                    // union with a match_all clause = match_all (+ scores of
+                    // the others); represent as all + no scores unless scoring
+                    base.all = true;
+                    if (scoring) throw std::runtime_error(
+                        "scored union with match_all clause not supported");
+                }
+            } else {
+                base = union_n(shoulds, msm, sv.num_docs);
+            }
+            if (shoulds.empty()) base = Match{};
+        }
+        if (!n.must_not.empty()) {
+            std::vector<Match> nots;
+            bool not_all = false;
+            for (auto& c : n.must_not) {
+                Match m = eval(c);
+                not_all |= m.all;
+                nots.push_back(std::move(m));
+            }
+            if (not_all) return Match{};  // must_not match_all excludes everything
+            Match nu = union_n(nots, 1, sv.num_docs);
+            base = subtract(std::move(base), nu, sv.num_docs);
+        }
+        if (n.boost != 1.0f && !base.scores.empty())
+            for (auto& s : base.scores) s *= n.boost;
+        return base;
+    }
+};
+
+// --------------------------------------------------------------- sort values
+struct SortSpec {
+    enum Comp { DOC_ID, SCORE, FAST_FIELD } comp = DOC_ID;
+    const FastFieldView* ff = nullptr;
+    int order = 1;  // 0 asc 1 desc
+};
+
+static pb::SortByValue sort_value_of(const SortSpec& s, const SplitView& sv, uint32_t doc,
+                                     float score) {
+    pb::SortByValue v;
+    switch (s.comp) {
+        case SortSpec::DOC_ID:
+            break;
+        case SortSpec::SCORE:
+            v.kind = pb::SortByValue::F64;
+            v.f64 = double(score);
+            break;
+        case SortSpec::FAST_FIELD: {
+            const FastFieldView* f = s.ff;
+            if (!f || !f->present(doc)) break;
+            if (f->type == FastFieldView::U64) {
+                v.kind = pb::SortByValue::U64;
+                v.u64 = f->u64(doc);
+            } else if (f->type == FastFieldView::DATETIME) {
+                v.kind = pb::SortByValue::I64;
+                v.i64 = f->i64(doc) * 1000000;  // ms -> ns (sorting.md default
+                                                // output unix_timestamp_nanos)
+            } else if (f->type == FastFieldView::STR) {
+                v.kind = pb::SortByValue::U64;
+                v.u64 = f->ord(doc);
+            } else {
+                v.kind = pb::SortByValue::I64;
+                v.i64 = f->i64(doc);
+            }
+            break;
+        }
+    }
+    return v;
+}
+
+// --------------------------------------------------------------- aggregation
+static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
+                                           const SplitView& sv, const Match& m) {
+    IntermediateAggResults out;
+    for (const AggDef& d : defs) {
+        AggResult a;
+        a.name = d.name;
+        for (auto& s : d.sub) a.sub_names.push_back(s.name);
+        const FastFieldView* f = sv.fast_field(d.field);
+        std::vector<const FastFieldView*> subf;
+        for (auto& s : d.sub) subf.push_back(sv.fast_field(s.field));
+        auto sub_value = [&](size_t si, uint32_t doc, double* v) -> bool {
+            const FastFieldView* sf = subf[si];
+            if (!sf || !sf->present(doc)) return false;
+            switch (sf->type) {
+                case FastFieldView::U64: *v = double(sf->u64(doc)); return true;
+                case FastFieldView::STR: return false;
+                default: *v = double(sf->i64(doc)); return true;
+            }
+        };
+        if (d.kind == AggDef::TERMS) {
+            a.kind = 3;
+            if (f && f->type == FastFieldView::STR) {
+                std::vector<uint64_t> counts(f->cardinality, 0);
+                auto visit = [&](uint32_t doc) {
+                    if (!f->present(doc)) return;
+                    counts[f->ord(doc)]++;
+                    a.terms_matched_docs++;
+                };
+                if (m.all)
+                    for (uint32_t doc = 0; doc < sv.num_docs; ++doc) visit(doc);
+                else
+                    for (uint32_t doc : m.docs) visit(doc);
+                for (uint32_t o = 0; o < f->cardinality; ++o)
+                    if (counts[o]) a.term_counts.emplace_back(f->dict_entry(o), counts[o]);
+                // dict order is lexicographic => term_counts sorted by key
+            }
+            out.aggs.push_back(std::move(a));
+            continue;
+        }
+        a.kind = d.kind == AggDef::DATE_HISTOGRAM ? 1 : 2;
+        std::map<int64_t, AggBucket> buckets;  // key quantized to bucket index
+        auto visit = [&](uint32_t doc) {
+            if (!f || !f->present(doc)) return;
+            double v;
+            if (f->type == FastFieldView::U64) v = double(f->u64(doc));
+            else if (f->type == FastFieldView::STR) return;
+            else v = double(f->i64(doc));
+            int64_t bi = int64_t(std::floor((v - d.offset) / d.interval));
+            AggBucket& b = buckets[bi];
+            b.doc_count++;
+            if (!d.sub.empty() && b.sub.empty()) b.sub.resize(d.sub.size());
+            for (size_t si = 0; si < d.sub.size(); ++si) {
+                double sval;
+                if (sub_value(si, doc, &sval)) {
+                    StatsPayload& sp = b.sub[si];
+                    sp.count++;
+                    sp.sum += sval;
+                    sp.min = std::min(sp.min, sval);
+                    sp.max = std::max(sp.max, sval);
+                }
+            }
+        };
+        if (m.all)
+            for (uint32_t doc = 0; doc < sv.num_docs; ++doc) visit(doc);
+        else
+            for (uint32_t doc : m.docs) visit(doc);
+        for (auto& kv : buckets) {
+            AggBucket b = kv.second;
+            b.key = double(kv.first) * d.interval + d.offset;
+            if (b.sub.empty() && !d.sub.empty()) b.sub.resize(d.sub.size());
+            a.buckets.push_back(std::move(b));
+        }
+        out.aggs.push_back(std::move(a));
+    }
+    return out;
+}
+
+// --------------------------------------------------------------- per split
+struct SplitResult {
+    uint64_t num_hits = 0;
+    std::vector<pb::PartialHit> hits;
+    IntermediateAggResults aggs;
+    bool has_aggs = false;
+    uint64_t cpu_micros = 0;
+    std::string error;
+};
+
+static SplitResult search_split(const SplitView& sv, const pb::SearchRequest& req,
+                                const Schema& schema) {
+    SplitResult out;
+    auto t0 = std::chrono::steady_clock::now();
+    PlanNode plan = parse_query_ast(req.query_ast, schema);
+    // timestamp pruning filter: [start, end) in seconds
+    // (leaf.rs:977 rewrite_request applies it via the timestamp fast field)
+    if ((req.start_timestamp || req.end_timestamp) && !schema.timestamp_field.empty()) {
+        PlanNode ts;
+        ts.kind = PlanNode::RANGE;
+        ts.field = schema.timestamp_field;
+        if (req.start_timestamp) {
+            ts.lo.kind = Bound::INCLUDED;
+            ts.lo.ival = *req.start_timestamp * 1000;
+        }
+        if (req.end_timestamp) {
+            ts.hi.kind = Bound::EXCLUDED;
+            ts.hi.ival = *req.end_timestamp * 1000;
+        }
+        PlanNode b;
+        b.kind = PlanNode::BOOL;
+        b.filter.push_back(std::move(ts));
+        b.must.push_back(std::move(plan));
+        plan = std::move(b);
+    }
+
+    // sort specs (max 2, like the reference — search.proto:269)
+    std::vector<SortSpec> specs;
+    for (auto& sf : req.sort_fields) {
+        SortSpec s;
+        s.order = sf.sort_order;
+        if (sf.field_name == "_score") s.comp = SortSpec::SCORE;
+        else {
+            s.comp = SortSpec::FAST_FIELD;
+            s.ff = sv.fast_field(sf.field_name);
+        }
+        specs.push_back(s);
+    }
+    bool scoring = false;
+    for (auto& s : specs) scoring |= s.comp == SortSpec::SCORE;
+
+    SplitSearcher searcher{sv, schema, scoring};
+    Match m = searcher.eval(plan);
+    out.num_hits = m.size(sv.num_docs);
+
+    uint64_t leaf_max_hits = req.max_hits + req.start_offset;
+    if (leaf_max_hits > 0 && out.num_hits > 0) {
+        int order1 = specs.empty() ? 1 : specs[0].order;
+        int order2 = specs.size() > 1 ? specs[1].order : 1;
+        auto mk_hit = [&](uint32_t doc, float score) {
+            pb::PartialHit h;
+            h.split_id = sv.split_id;
+            h.segment_ord = 0;
+            h.doc_id = doc;
+            if (!specs.empty()) h.sort_value = sort_value_of(specs[0], sv, doc, score);
+            if (specs.size() > 1) h.sort_value2 = sort_value_of(specs[1], sv, doc, score);
+            return h;
+        };
+        std::vector<pb::PartialHit> cand;
+        if (m.all) {
+            // no clause work: hits are just doc ids; default order doc desc
+            cand.reserve(std::min<uint64_t>(sv.num_docs, leaf_max_hits * 4 + 16));
+            if (specs.empty()) {
+                // doc-id sort: take head/tail directly
+                uint64_t k = std::min<uint64_t>(leaf_max_hits, sv.num_docs);
+                for (uint64_t i = 0; i < k; ++i) {
+                    uint32_t doc = order1 == 1 ? uint32_t(sv.num_docs - 1 - i) : uint32_t(i);
+                    cand.push_back(mk_hit(doc, 0.f));
+                }
+            } else {
+                for (uint32_t doc = 0; doc < sv.num_docs; ++doc)
+                    cand.push_back(mk_hit(doc, 0.f));
+            }
+        } else if (specs.empty()) {
+            uint64_t k = std::min<uint64_t>(leaf_max_hits, m.docs.size());
+            for (uint64_t i = 0; i < k; ++i) {
+                uint32_t doc = order1 == 1 ? m.docs[m.docs.size() - 1 - i] : m.docs[i];
+                cand.push_back(mk_hit(doc, 0.f));
+            }
+        } else {
+            cand.reserve(m.docs.size());
+            for (size_t i = 0; i < m.docs.size(); ++i)
+                cand.push_back(mk_hit(m.docs[i], m.scores.empty() ? 0.f : m.scores[i]));
+        }
+        auto cmp = [&](const pb::PartialHit& a, const pb::PartialHit& b) {
+            return hit_before(a, b, order1, order2);
+        };
+        size_t k = std::min<size_t>(leaf_max_hits, cand.size());
+        std::partial_sort(cand.begin(), cand.begin() + k, cand.end(), cmp);
+        cand.resize(k);
+        out.hits = std::move(cand);
+    }
+
+    if (req.aggregation_request) {
+        std::vector<AggDef> defs = parse_agg_request(*req.aggregation_request);
+        out.aggs = collect_aggs(defs, sv, m);
+        out.has_aggs = true;
+    }
+    out.cpu_micros = uint64_t(std::chrono::duration_cast<std::chrono::microseconds>(
+                                  std::chrono::steady_clock::now() - t0)
+                                  .count());
+    return out;
+}
+
+}  // namespace oracle
+}  // namespace qw
+
+// ============================================================== C ABI
+extern "C" {
+
+typedef struct qw_oracle_ctx qw_oracle_ctx;
+
+struct qw_oracle_ctx {
+    struct SplitHolder {
+        std::vector<uint8_t> data;
+        qw::SplitView view;
+    };
+    std::map<std::string, std::unique_ptr<SplitHolder>> splits;
+    std::string last_error;
+};
+
+typedef struct {
+    uint8_t* data;
+    size_t len;
+} qw_oracle_buf;
+
+qw_oracle_ctx* qw_oracle_create() { return new qw_oracle_ctx(); }
+void qw_oracle_free(qw_oracle_ctx* ctx) { delete ctx; }
+const char* qw_oracle_last_error(qw_oracle_ctx* ctx) { return ctx->last_error.c_str(); }
+void qw_oracle_buf_free(qw_oracle_buf* b) {
+    free(b->data);
+    b->data = nullptr;
+    b->len = 0;
+}
+
+int qw_oracle_add_split(qw_oracle_ctx* ctx, const char* split_id, const uint8_t* data,
+                        size_t len) {
+    try {
+        auto h = std::make_unique<qw_oracle_ctx::SplitHolder>();
+        h->data.assign(data, data + len);
+        h->view.parse(h->data.data(), h->data.size());
+        ctx->splits[split_id] = std::move(h);
+        return 0;
+    } catch (const std::exception& e) {
+        ctx->last_error = e.what();
+        return -6;
+    }
+}
+
+static void fill_buf(qw_oracle_buf* out, const std::string& s) {
+    out->data = (uint8_t*)malloc(s.size());
+    memcpy(out->data, s.data(), s.size());
+    out->len = s.size();
+}
+
+// LeafSearchRequest pb in -> LeafSearchResponse pb out. Splits run OpenMP-
+// parallel (one thread each), then merge with IncrementalCollector semantics.
+int qw_oracle_leaf_search(qw_oracle_ctx* ctx, const uint8_t* req_pb, size_t req_len,
+                          qw_oracle_buf* out) {
+    using namespace qw;
+    using namespace qw::oracle;
+    try {
+        pb::LeafSearchRequest lreq = pb::LeafSearchRequest::decode(req_pb, req_len);
+        const pb::SearchRequest& req = lreq.search_request;
+        // one doc mapper (schema json) per index; r1 handles one index per call
+        if (lreq.doc_mappers.empty()) throw std::runtime_error("missing doc_mapper");
+        Schema schema = Schema::parse(lreq.doc_mappers[0]);
+
+        struct Task {
+            const qw_oracle_ctx::SplitHolder* holder;
+            std::string split_id;
+        };
+        std::vector<Task> tasks;
+        for (auto& lr : lreq.leaf_requests)
+            for (auto& so : lr.split_offsets) {
+                auto it = ctx->splits.find(so.split_id);
+                if (it == ctx->splits.end())
+                    throw std::runtime_error("unknown split: " + so.split_id);
+                tasks.push_back({it->second.get(), so.split_id});
+            }
+
+        std::vector<SplitResult> results(tasks.size());
+#pragma omp parallel for schedule(dynamic)
+        for (size_t i = 0; i < tasks.size(); ++i) {
+            try {
+                results[i] = search_split(tasks[i].holder->view, req, schema);
+            } catch (const std::exception& e) {
+                results[i].error = e.what();
+            }
+        }
+
+        pb::LeafSearchResponse resp;
+        resp.num_attempted_splits = tasks.size();
+        IntermediateAggResults merged_aggs;
+        bool any_aggs = false;
+        std::vector<pb::PartialHit> all_hits;
+        pb::LeafResourceStats rstats;
+        rstats.search_pool_cpu_threads = uint64_t(omp_get_max_threads());
+        uint64_t worst_key = 0;
+        for (size_t i = 0; i < tasks.size(); ++i) {
+            SplitResult& r = results[i];
+            if (!r.error.empty()) {
+                pb::SplitSearchError e;
+                e.error = r.error;
+                e.split_id = tasks[i].split_id;
+                e.retryable_error = true;
+                resp.failed_splits.push_back(std::move(e));
+                continue;
+            }
+            resp.num_successful_splits++;
+            resp.num_hits += r.num_hits;
+            for (auto& h : r.hits) all_hits.push_back(std::move(h));
+            if (r.has_aggs) {
+                if (!any_aggs) {
+                    merged_aggs = std::move(r.aggs);
+                    any_aggs = true;
+                } else merged_aggs.merge(r.aggs);
+            }
+            pb::SplitResourceStats ss;
+            ss.split_num_docs = tasks[i].holder->view.num_docs;
+            ss.matched_num_docs = r.num_hits;
+            ss.cpu_search_microsecs = r.cpu_micros;
+            rstats.localexec_num_splits++;
+            rstats.localexec_num_docs += ss.split_num_docs;
+            rstats.split_resources_sum.split_num_docs += ss.split_num_docs;
+            rstats.split_resources_sum.matched_num_docs += ss.matched_num_docs;
+            rstats.split_resources_sum.cpu_search_microsecs += ss.cpu_search_microsecs;
+            if (ss.cpu_search_microsecs >= worst_key) {
+                worst_key = ss.cpu_search_microsecs;
+                rstats.split_resources_worst = ss;
+            }
+        }
+        // cross-split merge (merge_fruits semantics, collector.rs:832-861)
+        int order1 = req.sort_fields.empty() ? 1 : req.sort_fields[0].sort_order;
+        int order2 = req.sort_fields.size() > 1 ? req.sort_fields[1].sort_order : 1;
+        auto cmp = [&](const pb::PartialHit& a, const pb::PartialHit& b) {
+            return qw::hit_before(a, b, order1, order2);
+        };
+        size_t k = std::min<size_t>(req.max_hits + req.start_offset, all_hits.size());
+        std::partial_sort(all_hits.begin(), all_hits.begin() + k, all_hits.end(), cmp);
+        all_hits.resize(k);
+        resp.partial_hits = std::move(all_hits);
+        if (any_aggs) resp.intermediate_aggregation_result = merged_aggs.encode();
+        resp.resource_stats = rstats;
+        fill_buf(out, resp.encode());
+        return 0;
+    } catch (const std::exception& e) {
+        ctx->last_error = e.what();
+        return -4;
+    }
+}
+
+// finalize an intermediate agg blob to ES-shaped JSON (same code path the
+// product exports as qw_finalize_agg_to_json — shared qagg_format.h)
+int qw_oracle_finalize_agg(const uint8_t* blob, size_t len, const char* agg_req_json,
+                           qw_oracle_buf* out) {
+    try {
+        qw::IntermediateAggResults ir = qw::IntermediateAggResults::decode(blob, len);
+        std::vector<qw::AggDef> defs = qw::parse_agg_request(agg_req_json);
+        fill_buf(out, qw::finalize_aggs_json(ir, defs));
+        return 0;
+    } catch (const std::exception& e) {
+        (void)e;
+        return -4;
+    }
+}
+
+}  // extern "C"

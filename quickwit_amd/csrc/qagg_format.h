@@ -1,0 +1,411 @@
+// QAGG1 — this build's intermediate aggregation result blob, carried in
+// LeafSearchResponse.intermediate_aggregation_result (search.proto:640).
+// Replaces the reference's postcard-serialized tantivy
+// IntermediateAggregationResults (an unpinned tantivy-internal encoding —
+// declared deviation). Merge semantics = sum-by-key, exactly
+// what tantivy's intermediate merge does for histogram/terms counts; finalize
+// produces the ES-shaped JSON the golden scenarios assert
+// (rest-api-tests/scenarii/aggregations/0001-aggregations.yaml).
+//
+// Layout (little-endian):
+//   u32 magic 'QAG1' (0x31474151), u16 version=1, u16 n_aggs
+//   per agg:
+//     u16 name_len, name bytes
+//     u8 kind: 1=date_histogram 2=histogram 3=terms
+//     u16 n_sub (metric sub-aggs); per sub: u16 name_len, name
+//     histos: u32 n_buckets; per bucket:
+//        f64 key, u64 doc_count, per sub {u64 count, f64 sum, f64 min, f64 max}
+//       (buckets sorted by key, sparse: only non-empty buckets pre-merge)
+//     terms: u64 matched_docs_with_value, u32 n_entries; per entry:
+//        u16 key_len, key bytes, u64 doc_count   (sorted by key bytes)
+#pragma once
+#include <algorithm>
+#include <cstdint>
+#include <cstring>
+#include <limits>
+#include <map>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#include "qast.h"
+
+namespace qw {
+
+struct StatsPayload {
+    uint64_t count = 0;
+    double sum = 0;
+    double min = std::numeric_limits<double>::infinity();
+    double max = -std::numeric_limits<double>::infinity();
+    void merge(const StatsPayload& o) {
+        count += o.count;
+        sum += o.sum;
+        min = std::min(min, o.min);
+        max = std::max(max, o.max);
+    }
+};
+
+struct AggBucket {
+    double key = 0;
+    uint64_t doc_count = 0;
+    std::vector<StatsPayload> sub;
+};
+
+struct AggResult {
+    std::string name;
+    uint8_t kind = 1;  // 1=date_histogram 2=histogram 3=terms
+    std::vector<std::string> sub_names;
+    std::vector<AggBucket> buckets;                            // histos (sorted by key)
+    std::vector<std::pair<std::string, uint64_t>> term_counts; // terms (sorted by key)
+    uint64_t terms_matched_docs = 0;
+};
+
+struct IntermediateAggResults {
+    std::vector<AggResult> aggs;
+
+    std::string encode() const {
+        std::string o;
+        auto put = [&](const void* p, size_t n) { o.append((const char*)p, n); };
+        uint32_t magic = 0x31474151;
+        uint16_t ver = 1, n = uint16_t(aggs.size());
+        put(&magic, 4);
+        put(&ver, 2);
+        put(&n, 2);
+        for (const AggResult& a : aggs) {
+            uint16_t nl = uint16_t(a.name.size());
+            put(&nl, 2);
+            put(a.name.data(), nl);
+            put(&a.kind, 1);
+            uint16_t ns = uint16_t(a.sub_names.size());
+            put(&ns, 2);
+            for (auto& s : a.sub_names) {
+                uint16_t sl = uint16_t(s.size());
+                put(&sl, 2);
+                put(s.data(), sl);
+            }
+            if (a.kind == 3) {
+                put(&a.terms_matched_docs, 8);
+                uint32_t ne = uint32_t(a.term_counts.size());
+                put(&ne, 4);
+                for (auto& kv : a.term_counts) {
+                    uint16_t kl = uint16_t(kv.first.size());
+                    put(&kl, 2);
+                    put(kv.first.data(), kl);
+                    put(&kv.second, 8);
+                }
+            } else {
+                uint32_t nb = uint32_t(a.buckets.size());
+                put(&nb, 4);
+                for (auto& b : a.buckets) {
+                    put(&b.key, 8);
+                    put(&b.doc_count, 8);
+                    for (size_t s = 0; s < a.sub_names.size(); ++s) {
+                        const StatsPayload& sp =
+                            s < b.sub.size() ? b.sub[s] : StatsPayload{};
+                        put(&sp.count, 8);
+                        put(&sp.sum, 8);
+                        put(&sp.min, 8);
+                        put(&sp.max, 8);
+                    }
+                }
+            }
+        }
+        return o;
+    }
+
+    static IntermediateAggResults decode(const uint8_t* p, size_t len) {
+        IntermediateAggResults r;
+        const uint8_t* end = p + len;
+        auto need = [&](size_t n) {
+            if (size_t(end - p) < n) throw std::runtime_error("QAGG1: truncated");
+        };
+        auto get = [&](void* d, size_t n) {
+            need(n);
+            memcpy(d, p, n);
+            p += n;
+        };
+        uint32_t magic;
+        uint16_t ver, n;
+        get(&magic, 4);
+        get(&ver, 2);
+        get(&n, 2);
+        if (magic != 0x31474151 || ver != 1) throw std::runtime_error("QAGG1: bad header");
+        for (int i = 0; i < n; ++i) {
+            AggResult a;
+            uint16_t nl;
+            get(&nl, 2);
+            need(nl);
+            a.name.assign((const char*)p, nl);
+            p += nl;
+            get(&a.kind, 1);
+            uint16_t ns;
+            get(&ns, 2);
+            for (int s = 0; s < ns; ++s) {
+                uint16_t sl;
+                get(&sl, 2);
+                need(sl);
+                a.sub_names.emplace_back((const char*)p, sl);
+                p += sl;
+            }
+            if (a.kind == 3) {
+                get(&a.terms_matched_docs, 8);
+                uint32_t ne;
+                get(&ne, 4);
+                a.term_counts.reserve(ne);
+                for (uint32_t e = 0; e < ne; ++e) {
+                    uint16_t kl;
+                    get(&kl, 2);
+                    need(kl);
+                    std::string k((const char*)p, kl);
+                    p += kl;
+                    uint64_t c;
+                    get(&c, 8);
+                    a.term_counts.emplace_back(std::move(k), c);
+                }
+            } else {
+                uint32_t nb;
+                get(&nb, 4);
+                a.buckets.reserve(nb);
+                for (uint32_t bi = 0; bi < nb; ++bi) {
+                    AggBucket b;
+                    get(&b.key, 8);
+                    get(&b.doc_count, 8);
+                    b.sub.resize(a.sub_names.size());
+                    for (auto& sp : b.sub) {
+                        get(&sp.count, 8);
+                        get(&sp.sum, 8);
+                        get(&sp.min, 8);
+                        get(&sp.max, 8);
+                    }
+                    a.buckets.push_back(std::move(b));
+                }
+            }
+            r.aggs.push_back(std::move(a));
+        }
+        return r;
+    }
+
+    // merge-by-key (both inputs sorted); mirrors the reference's intermediate
+    // aggregation merge (collector.rs:867+, tantivy aggregation merge)
+    void merge(const IntermediateAggResults& o) {
+        if (aggs.empty()) {
+            aggs = o.aggs;
+            return;
+        }
+        if (aggs.size() != o.aggs.size()) throw std::runtime_error("QAGG1: merge shape");
+        for (size_t i = 0; i < aggs.size(); ++i) {
+            AggResult& a = aggs[i];
+            const AggResult& b = o.aggs[i];
+            if (a.name != b.name || a.kind != b.kind)
+                throw std::runtime_error("QAGG1: merge mismatch");
+            if (a.kind == 3) {
+                std::vector<std::pair<std::string, uint64_t>> merged;
+                merged.reserve(a.term_counts.size() + b.term_counts.size());
+                size_t x = 0, y = 0;
+                while (x < a.term_counts.size() || y < b.term_counts.size()) {
+                    if (y >= b.term_counts.size() ||
+                        (x < a.term_counts.size() &&
+                         a.term_counts[x].first < b.term_counts[y].first))
+                        merged.push_back(a.term_counts[x++]);
+                    else if (x >= a.term_counts.size() ||
+                             b.term_counts[y].first < a.term_counts[x].first)
+                        merged.push_back(b.term_counts[y++]);
+                    else {
+                        merged.emplace_back(a.term_counts[x].first,
+                                            a.term_counts[x].second +
+                                                b.term_counts[y].second);
+                        ++x;
+                        ++y;
+                    }
+                }
+                a.term_counts = std::move(merged);
+                a.terms_matched_docs += b.terms_matched_docs;
+            } else {
+                std::vector<AggBucket> merged;
+                merged.reserve(a.buckets.size() + b.buckets.size());
+                size_t x = 0, y = 0;
+                while (x < a.buckets.size() || y < b.buckets.size()) {
+                    if (y >= b.buckets.size() ||
+                        (x < a.buckets.size() && a.buckets[x].key < b.buckets[y].key))
+                        merged.push_back(a.buckets[x++]);
+                    else if (x >= a.buckets.size() || b.buckets[y].key < a.buckets[x].key)
+                        merged.push_back(b.buckets[y++]);
+                    else {
+                        AggBucket m = a.buckets[x++];
+                        const AggBucket& n = b.buckets[y++];
+                        m.doc_count += n.doc_count;
+                        m.sub.resize(a.sub_names.size());
+                        for (size_t s = 0; s < m.sub.size() && s < n.sub.size(); ++s)
+                            m.sub[s].merge(n.sub[s]);
+                        merged.push_back(std::move(m));
+                    }
+                }
+                a.buckets = std::move(merged);
+            }
+        }
+    }
+};
+
+// ---- finalize to ES-shaped JSON ("aggregations" object body)
+inline void stats_to_json(std::string& o, const MetricAgg& m, const StatsPayload& s) {
+    auto num = [&](double d) { mj::num_to(o, d); };
+    bool empty = s.count == 0;
+    switch (m.kind) {
+        case MetricAgg::STATS:
+            o += "{\"avg\":";
+            if (empty) o += "null";
+            else num(s.sum / double(s.count));
+            o += ",\"count\":";
+            {
+                char buf[24];
+                snprintf(buf, sizeof buf, "%llu", (unsigned long long)s.count);
+                o += buf;
+            }
+            o += ",\"max\":";
+            if (empty) o += "null";
+            else num(s.max);
+            o += ",\"min\":";
+            if (empty) o += "null";
+            else num(s.min);
+            o += ",\"sum\":";
+            num(s.sum);
+            o += "}";
+            break;
+        case MetricAgg::AVG:
+            o += "{\"value\":";
+            if (empty) o += "null";
+            else num(s.sum / double(s.count));
+            o += "}";
+            break;
+        case MetricAgg::SUM:
+            o += "{\"value\":";
+            num(s.sum);
+            o += "}";
+            break;
+        case MetricAgg::MIN:
+            o += "{\"value\":";
+            if (empty) o += "null";
+            else num(s.min);
+            o += "}";
+            break;
+        case MetricAgg::MAX:
+            o += "{\"value\":";
+            if (empty) o += "null";
+            else num(s.max);
+            o += "}";
+            break;
+        case MetricAgg::COUNT:
+            o += "{\"value\":";
+            num(double(s.count));
+            o += "}";
+            break;
+    }
+}
+
+inline std::string finalize_aggs_json(const IntermediateAggResults& ir,
+                                      const std::vector<AggDef>& defs) {
+    if (ir.aggs.size() != defs.size())
+        throw std::runtime_error("finalize: agg count mismatch");
+    std::string o = "{";
+    for (size_t i = 0; i < defs.size(); ++i) {
+        const AggDef& d = defs[i];
+        const AggResult& a = ir.aggs[i];
+        if (i) o += ",";
+        mj::escape_to(o, d.name);
+        o += ":";
+        if (d.kind == AggDef::TERMS) {
+            // order: doc_count desc, then key asc (ES/tantivy default);
+            // truncate to size; sum_other = matched - shown
+            std::vector<std::pair<std::string, uint64_t>> ordered = a.term_counts;
+            std::stable_sort(ordered.begin(), ordered.end(),
+                             [](const auto& x, const auto& y) {
+                                 if (x.second != y.second) return x.second > y.second;
+                                 return x.first < y.first;
+                             });
+            uint64_t shown_docs = 0;
+            size_t nshow = std::min(size_t(d.size), ordered.size());
+            o += "{\"buckets\":[";
+            for (size_t b = 0; b < nshow; ++b) {
+                if (b) o += ",";
+                o += "{\"doc_count\":";
+                char buf[24];
+                snprintf(buf, sizeof buf, "%llu", (unsigned long long)ordered[b].second);
+                o += buf;
+                shown_docs += ordered[b].second;
+                o += ",\"key\":";
+                mj::escape_to(o, ordered[b].first);
+                o += "}";
+            }
+            o += "],\"doc_count_error_upper_bound\":0,\"sum_other_doc_count\":";
+            char buf[24];
+            snprintf(buf, sizeof buf, "%llu",
+                     (unsigned long long)(a.terms_matched_docs - shown_docs));
+            o += buf;
+            o += "}";
+        } else {
+            // gap-fill between min and max key (min_doc_count=0 default), and
+            // extend to extended_bounds when given
+            std::vector<AggBucket> bs = a.buckets;
+            double lo = 0, hi = -1;
+            if (!bs.empty()) {
+                lo = bs.front().key;
+                hi = bs.back().key;
+            }
+            auto bucket_key = [&](double v) {
+                return floor((v - d.offset) / d.interval) * d.interval + d.offset;
+            };
+            if (d.has_bounds) {
+                double blo = bucket_key(d.bmin), bhi = bucket_key(d.bmax);
+                if (bs.empty()) {
+                    lo = blo;
+                    hi = bhi;
+                } else {
+                    lo = std::min(lo, blo);
+                    hi = std::max(hi, bhi);
+                }
+            }
+            o += "{\"buckets\":[";
+            bool first = true;
+            size_t bi = 0;
+            if (hi >= lo && (!bs.empty() || d.has_bounds)) {
+                long long nb = (long long)llround((hi - lo) / d.interval);
+                for (long long k = 0; k <= nb; ++k) {
+                    double key = lo + double(k) * d.interval;
+                    const AggBucket* b = nullptr;
+                    while (bi < bs.size() && bs[bi].key < key - d.interval * 0.5) ++bi;
+                    if (bi < bs.size() && fabs(bs[bi].key - key) < d.interval * 0.5)
+                        b = &bs[bi];
+                    uint64_t dc = b ? b->doc_count : 0;
+                    if (!b && d.min_doc_count > 0) continue;
+                    if ((long long)dc < d.min_doc_count) continue;
+                    if (!first) o += ",";
+                    first = false;
+                    o += "{\"doc_count\":";
+                    char buf[24];
+                    snprintf(buf, sizeof buf, "%llu", (unsigned long long)dc);
+                    o += buf;
+                    o += ",\"key\":";
+                    mj::num_to(o, key);
+                    if (d.kind == AggDef::DATE_HISTOGRAM) {
+                        o += ",\"key_as_string\":";
+                        mj::escape_to(o, ms_to_rfc3339(int64_t(key)));
+                    }
+                    for (size_t s = 0; s < d.sub.size(); ++s) {
+                        o += ",";
+                        mj::escape_to(o, d.sub[s].name);
+                        o += ":";
+                        StatsPayload sp;
+                        if (b && s < b->sub.size()) sp = b->sub[s];
+                        stats_to_json(o, d.sub[s], sp);
+                    }
+                    o += "}";
+                }
+            }
+            o += "]}";
+        }
+    }
+    o += "}";
+    return o;
+}
+
+}  // namespace qw

@@ -1,0 +1,497 @@
+// QueryAst JSON -> normalized query plan, + doc-mapper schema parse, +
+// aggregation-request parse (ES-style aggs JSON subset).
+//
+// Semantics restated from the reference:
+//  - QueryAst serde model: quickwit-query/src/query_ast/mod.rs:57-77
+//    (#[serde(tag="type", rename_all="snake_case")]).
+//  - full_text -> per-token term disjunction/conjunction with
+//    IndexRecordOption::WithFreqs: full_text_query.rs:103-137.
+//  - bool -> TantivyBoolQuery: tantivy_query_ast.rs:153-374; default
+//    minimum_should_match semantics: with no must/filter clause at least one
+//    should must match; an explicit minimum_should_match overrides
+//    (bool_query.rs:34-45, tantivy_query_ast.rs:372).
+//  - range over fast fields: range_query.rs:30-158 (Bound<JsonLiteral> with
+//    Rust serde form {"included":..}/{"excluded":..}/"unbounded" — serde
+//    renames Bound variants to snake_case inside quickwit's JSON).
+//  - tokenizers: "raw" = one verbatim token; "default" = alphanumeric runs,
+//    lowercased, tokens > 40 chars removed (tantivy default analyzer).
+// The shared location (product + oracle both build plans from it) is a
+// declared plumbing exception — results are pinned end-to-end by the golden
+// suites.
+#pragma once
+#include <algorithm>
+#include <cctype>
+#include <cstdint>
+#include <memory>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#include "minijson.h"
+
+namespace qw {
+
+// ---------------------------------------------------------------- schema
+struct SchemaField {
+    std::string name;
+    std::string type;       // text | u64 | i64 | datetime | str
+    std::string tokenizer;  // text only
+    bool record_freq = false;
+    bool fast = false;
+};
+
+struct Schema {
+    std::vector<SchemaField> fields;
+    std::vector<std::string> default_search_fields;
+    std::string timestamp_field;
+
+    const SchemaField* field(const std::string& n) const {
+        for (auto& f : fields)
+            if (f.name == n) return &f;
+        return nullptr;
+    }
+
+    // Accepts this repo's schema JSON (splitgen.HDFS_SCHEMA shape):
+    // {"timestamp_field": ..., "default_search_fields": [...]?, "fields":
+    //  [{"name","type","tokenizer"?,"record"?,"fast"?}...]}
+    static Schema parse(const std::string& json) {
+        Schema s;
+        mj::ValuePtr root = mj::parse(json);
+        const mj::Value* tf = root->get("timestamp_field");
+        if (tf && !tf->is_null()) s.timestamp_field = tf->s;
+        const mj::Value* dsf = root->get("default_search_fields");
+        if (dsf)
+            for (auto& v : dsf->arr) s.default_search_fields.push_back(v->s);
+        for (auto& fv : root->at("fields")->arr) {
+            SchemaField f;
+            f.name = fv->at("name")->s;
+            f.type = fv->at("type")->s;
+            const mj::Value* tok = fv->get("tokenizer");
+            f.tokenizer = tok ? tok->s : "default";
+            const mj::Value* rec = fv->get("record");
+            f.record_freq = rec && rec->s == "freq";
+            const mj::Value* fast = fv->get("fast");
+            f.fast = fast && fast->b;
+            s.fields.push_back(std::move(f));
+        }
+        return s;
+    }
+};
+
+inline std::vector<std::string> tokenize(const std::string& text,
+                                         const std::string& tokenizer) {
+    if (tokenizer == "raw") {
+        if (text.empty()) return {};
+        return {text};
+    }
+    std::vector<std::string> out;
+    std::string cur;
+    for (unsigned char c : text) {
+        if (std::isalnum(c)) cur.push_back(char(std::tolower(c)));
+        else if (!cur.empty()) {
+            if (cur.size() <= 40) out.push_back(cur);
+            cur.clear();
+        }
+    }
+    if (!cur.empty() && cur.size() <= 40) out.push_back(cur);
+    return out;
+}
+
+// ---------------------------------------------------------------- plan
+struct Bound {
+    enum Kind { UNBOUNDED, INCLUDED, EXCLUDED } kind = UNBOUNDED;
+    int64_t ival = 0;    // canonical value for u64/i64/datetime(ms) columns
+    bool from_f64 = false;
+    double fval = 0;
+};
+
+struct PlanNode {
+    enum Kind { MATCH_ALL, MATCH_NONE, TERM, BOOL, RANGE, FIELD_PRESENCE } kind = MATCH_ALL;
+    // TERM
+    std::string field;
+    std::string value;
+    // BOOL
+    std::vector<PlanNode> must, must_not, should, filter;
+    int64_t minimum_should_match = -1;  // -1 = unset
+    // RANGE / FIELD_PRESENCE
+    Bound lo, hi;
+    float boost = 1.0f;
+};
+
+inline int64_t parse_datetime_ms(const mj::Value* lit);
+
+// RFC3339 (subset: YYYY-MM-DD['T'HH:MM[:SS[.fff]]][Z|±hh:mm]) -> epoch ms
+inline int64_t rfc3339_to_ms(const std::string& s);
+
+inline Bound parse_bound(const mj::Value* v, const SchemaField& f) {
+    Bound b;
+    if (!v) return b;
+    // serde's built-in std::ops::Bound impl: {"Included": lit} /
+    // {"Excluded": lit} / "Unbounded" (capitalized; lowercase also accepted)
+    const mj::Value* lit = nullptr;
+    if (v->kind == mj::Value::STR && (v->s == "Unbounded" || v->s == "unbounded")) return b;
+    const mj::Value* inc = v->get("Included");
+    if (!inc) inc = v->get("included");
+    const mj::Value* exc = v->get("Excluded");
+    if (!exc) exc = v->get("excluded");
+    if (inc) {
+        b.kind = Bound::INCLUDED;
+        lit = inc;
+    } else if (exc) {
+        b.kind = Bound::EXCLUDED;
+        lit = exc;
+    } else throw std::runtime_error("bad range bound");
+    if (f.type == "datetime") {
+        b.ival = parse_datetime_ms(lit);
+    } else if (lit->kind == mj::Value::INT) {
+        b.ival = lit->i;
+    } else if (lit->kind == mj::Value::DBL) {
+        b.from_f64 = true;
+        b.fval = lit->d;
+        b.ival = int64_t(lit->d);
+    } else if (lit->kind == mj::Value::STR) {
+        b.ival = strtoll(lit->s.c_str(), nullptr, 10);
+    } else throw std::runtime_error("bad range literal");
+    return b;
+}
+
+inline PlanNode build_plan(const mj::Value* ast, const Schema& schema);
+
+inline PlanNode full_text_plan(const std::string& field, const std::string& text,
+                               const std::string& op, const Schema& schema) {
+    const SchemaField* f = schema.field(field);
+    if (!f || f->type != "text")
+        throw std::runtime_error("full_text on unknown/non-text field: " + field);
+    std::vector<std::string> toks = tokenize(text, f->tokenizer);
+    if (toks.empty()) {
+        PlanNode n;
+        n.kind = PlanNode::MATCH_NONE;  // zero_terms_query default MatchNone
+        return n;
+    }
+    if (toks.size() == 1) {
+        PlanNode n;
+        n.kind = PlanNode::TERM;
+        n.field = field;
+        n.value = toks[0];
+        return n;
+    }
+    PlanNode b;
+    b.kind = PlanNode::BOOL;
+    for (auto& t : toks) {
+        PlanNode n;
+        n.kind = PlanNode::TERM;
+        n.field = field;
+        n.value = t;
+        (op == "and" ? b.must : b.should).push_back(std::move(n));
+    }
+    return b;
+}
+
+// minimal user_input support: whitespace-separated clauses of `field:token`
+// or bare tokens over default fields, implicit OR (the subset quickwit's
+// query parser covers for plain log searches; anything else -> error =
+// SearchError::InvalidQuery, never a silent wrong answer)
+inline PlanNode user_input_plan(const std::string& text,
+                                const std::vector<std::string>& default_fields,
+                                const Schema& schema) {
+    std::vector<PlanNode> clauses;
+    size_t i = 0;
+    while (i < text.size()) {
+        while (i < text.size() && text[i] == ' ') ++i;
+        size_t j = text.find(' ', i);
+        if (j == std::string::npos) j = text.size();
+        if (j > i) {
+            std::string tok = text.substr(i, j - i);
+            if (tok.find('"') != std::string::npos || tok.find('(') != std::string::npos ||
+                tok == "AND" || tok == "OR" || tok == "NOT" || tok[0] == '-' ||
+                tok[0] == '+')
+                throw std::runtime_error("user_input syntax not supported: " + tok);
+            size_t c = tok.find(':');
+            if (c != std::string::npos) {
+                clauses.push_back(
+                    full_text_plan(tok.substr(0, c), tok.substr(c + 1), "or", schema));
+            } else {
+                const std::vector<std::string>& dfs =
+                    default_fields.empty() ? schema.default_search_fields : default_fields;
+                if (dfs.empty()) throw std::runtime_error("no default search fields");
+                PlanNode b;
+                b.kind = PlanNode::BOOL;
+                for (auto& df : dfs) b.should.push_back(full_text_plan(df, tok, "or", schema));
+                clauses.push_back(dfs.size() == 1 ? std::move(b.should[0]) : std::move(b));
+            }
+        }
+        i = j + 1;
+    }
+    if (clauses.empty()) {
+        PlanNode n;
+        n.kind = PlanNode::MATCH_ALL;
+        return n;
+    }
+    if (clauses.size() == 1) return clauses[0];
+    PlanNode b;
+    b.kind = PlanNode::BOOL;
+    b.should = std::move(clauses);
+    return b;
+}
+
+inline PlanNode build_plan(const mj::Value* ast, const Schema& schema) {
+    std::string ty = ast->at("type")->s;
+    PlanNode n;
+    if (ty == "match_all") {
+        n.kind = PlanNode::MATCH_ALL;
+    } else if (ty == "match_none") {
+        n.kind = PlanNode::MATCH_NONE;
+    } else if (ty == "term") {
+        n.kind = PlanNode::TERM;
+        n.field = ast->at("field")->s;
+        n.value = ast->at("value")->s;
+        if (!schema.field(n.field))
+            throw std::runtime_error("term on unknown field: " + n.field);
+    } else if (ty == "full_text") {
+        std::string op = "or";
+        const mj::Value* params = ast->get("params");
+        if (params) {
+            const mj::Value* mode = params->get("mode");
+            if (mode) {
+                // FullTextMode serde: {"type":"bool","operator":"or"|"and"} or
+                // {"type":"phrase",...} (full_text_query.rs:172)
+                const mj::Value* mt = mode->get("type");
+                if (mt && mt->s != "bool" && mt->s != "bool_prefix")
+                    throw std::runtime_error("full_text mode not supported: " + mt->s);
+                const mj::Value* o = mode->get("operator");
+                if (o) op = o->s;
+                std::transform(op.begin(), op.end(), op.begin(), ::tolower);
+            }
+        }
+        return full_text_plan(ast->at("field")->s, ast->at("text")->s, op, schema);
+    } else if (ty == "bool") {
+        n.kind = PlanNode::BOOL;
+        auto fill = [&](const char* key, std::vector<PlanNode>& out) {
+            const mj::Value* arr = ast->get(key);
+            if (arr)
+                for (auto& c : arr->arr) out.push_back(build_plan(c.get(), schema));
+        };
+        fill("must", n.must);
+        fill("must_not", n.must_not);
+        fill("should", n.should);
+        fill("filter", n.filter);
+        const mj::Value* msm = ast->get("minimum_should_match");
+        if (msm && !msm->is_null()) n.minimum_should_match = msm->as_i64();
+    } else if (ty == "range") {
+        n.kind = PlanNode::RANGE;
+        n.field = ast->at("field")->s;
+        const SchemaField* f = schema.field(n.field);
+        if (!f || !f->fast)
+            throw std::runtime_error("range on non-fast field: " + n.field);
+        n.lo = parse_bound(ast->get("lower_bound"), *f);
+        n.hi = parse_bound(ast->get("upper_bound"), *f);
+    } else if (ty == "field_presence") {
+        n.kind = PlanNode::FIELD_PRESENCE;
+        n.field = ast->at("field")->s;
+    } else if (ty == "boost") {
+        PlanNode inner = build_plan(ast->at("underlying"), schema);
+        inner.boost *= float(ast->at("boost")->num());
+        return inner;
+    } else if (ty == "user_input") {
+        std::vector<std::string> dfs;
+        const mj::Value* d = ast->get("default_fields");
+        if (d && !d->is_null())
+            for (auto& v : d->arr) dfs.push_back(v->s);
+        return user_input_plan(ast->at("user_text")->s, dfs, schema);
+    } else {
+        throw std::runtime_error("query ast type not supported: " + ty);
+    }
+    return n;
+}
+
+inline PlanNode parse_query_ast(const std::string& json, const Schema& schema) {
+    return build_plan(mj::parse(json).get(), schema);
+}
+
+// ---------------------------------------------------------------- rfc3339
+inline int64_t days_from_civil(int64_t y, unsigned m, unsigned d) {
+    y -= m <= 2;
+    const int64_t era = (y >= 0 ? y : y - 399) / 400;
+    const unsigned yoe = unsigned(y - era * 400);
+    const unsigned doy = (153 * (m + (m > 2 ? -3 : 9)) + 2) / 5 + d - 1;
+    const unsigned doe = yoe * 365 + yoe / 4 - yoe / 100 + doy;
+    return era * 146097 + int64_t(doe) - 719468;
+}
+
+inline int64_t rfc3339_to_ms(const std::string& s) {
+    int y = 0, mo = 0, d = 0, h = 0, mi = 0;
+    double sec = 0;
+    int off_sign = 0, off_h = 0, off_m = 0;
+    if (s.size() < 10) throw std::runtime_error("bad datetime: " + s);
+    y = atoi(s.substr(0, 4).c_str());
+    mo = atoi(s.substr(5, 2).c_str());
+    d = atoi(s.substr(8, 2).c_str());
+    size_t i = 10;
+    if (i < s.size() && (s[i] == 'T' || s[i] == 't' || s[i] == ' ')) {
+        h = atoi(s.substr(i + 1, 2).c_str());
+        mi = atoi(s.substr(i + 4, 2).c_str());
+        i += 6;
+        if (i < s.size() && s[i] == ':') {
+            size_t j = i + 1;
+            while (j < s.size() && (isdigit((unsigned char)s[j]) || s[j] == '.')) ++j;
+            sec = atof(s.substr(i + 1, j - i - 1).c_str());
+            i = j;
+        }
+        if (i < s.size() && (s[i] == '+' || s[i] == '-')) {
+            off_sign = s[i] == '+' ? 1 : -1;
+            off_h = atoi(s.substr(i + 1, 2).c_str());
+            off_m = atoi(s.substr(i + 4, 2).c_str());
+        }
+    }
+    int64_t days = days_from_civil(y, unsigned(mo), unsigned(d));
+    double t = double(days) * 86400.0 + h * 3600.0 + mi * 60.0 + sec -
+               off_sign * (off_h * 3600.0 + off_m * 60.0);
+    return int64_t(t * 1000.0 + (t >= 0 ? 0.5 : -0.5));
+}
+
+inline int64_t parse_datetime_ms(const mj::Value* lit) {
+    if (lit->kind == mj::Value::STR) {
+        // numeric string or rfc3339
+        char* endp = nullptr;
+        long long v = strtoll(lit->s.c_str(), &endp, 10);
+        if (endp && *endp == 0) {
+            long long a = v < 0 ? -v : v;
+            return a < 1000000000000LL ? v * 1000 : v;  // s vs ms, like splitgen
+        }
+        return rfc3339_to_ms(lit->s);
+    }
+    int64_t v = lit->as_i64();
+    int64_t a = v < 0 ? -v : v;
+    return a < 1000000000000LL ? v * 1000 : v;
+}
+
+inline std::string ms_to_rfc3339(int64_t ms) {
+    int64_t days = ms / 86400000;
+    int64_t rem = ms % 86400000;
+    if (rem < 0) {
+        rem += 86400000;
+        days -= 1;
+    }
+    // civil_from_days (Howard Hinnant)
+    int64_t z = days + 719468;
+    const int64_t era = (z >= 0 ? z : z - 146096) / 146097;
+    const unsigned doe = unsigned(z - era * 146097);
+    const unsigned yoe = (doe - doe / 1460 + doe / 36524 - doe / 146096) / 365;
+    const int64_t y = int64_t(yoe) + era * 400;
+    const unsigned doy = doe - (365 * yoe + yoe / 4 - yoe / 100);
+    const unsigned mp = (5 * doy + 2) / 153;
+    const unsigned d = doy - (153 * mp + 2) / 5 + 1;
+    const unsigned m = mp + (mp < 10 ? 3 : -9);
+    int64_t yy = y + (m <= 2);
+    int sec_of_day = int(rem / 1000);
+    int msec = int(rem % 1000);
+    char buf[40];
+    if (msec)
+        snprintf(buf, sizeof buf, "%04lld-%02u-%02uT%02d:%02d:%02d.%03dZ", (long long)yy, m,
+                 d, sec_of_day / 3600, (sec_of_day / 60) % 60, sec_of_day % 60, msec);
+    else
+        snprintf(buf, sizeof buf, "%04lld-%02u-%02uT%02d:%02d:%02dZ", (long long)yy, m, d,
+                 sec_of_day / 3600, (sec_of_day / 60) % 60, sec_of_day % 60);
+    return buf;
+}
+
+// ---------------------------------------------------------------- agg plan
+struct MetricAgg {
+    std::string name;
+    enum Kind { STATS, AVG, SUM, MIN, MAX, COUNT } kind = STATS;
+    std::string field;
+};
+
+struct AggDef {
+    std::string name;
+    enum Kind { DATE_HISTOGRAM, HISTOGRAM, TERMS } kind = DATE_HISTOGRAM;
+    std::string field;
+    double interval = 0;  // ms for date_histogram; raw units for histogram
+    double offset = 0;
+    bool has_bounds = false;
+    double bmin = 0, bmax = 0;
+    uint32_t size = 10;          // terms
+    int64_t min_doc_count = -1;  // -1 = default (0 for histos, 1 for terms)
+    std::vector<MetricAgg> sub;
+};
+
+inline double parse_duration_ms(const std::string& s) {
+    size_t i = 0;
+    while (i < s.size() && (isdigit((unsigned char)s[i]) || s[i] == '.' || s[i] == '-' ||
+                            s[i] == '+'))
+        ++i;
+    double v = atof(s.substr(0, i).c_str());
+    std::string u = s.substr(i);
+    if (u == "ms") return v;
+    if (u == "s") return v * 1000;
+    if (u == "m") return v * 60000;
+    if (u == "h") return v * 3600000;
+    if (u == "d") return v * 86400000;
+    throw std::runtime_error("bad interval: " + s);
+}
+
+inline std::vector<AggDef> parse_agg_request(const std::string& json) {
+    std::vector<AggDef> out;
+    mj::ValuePtr root = mj::parse(json);
+    for (auto& kv : root->obj) {
+        AggDef a;
+        a.name = kv.first;
+        const mj::Value* body = kv.second.get();
+        const mj::Value* spec = nullptr;
+        if ((spec = body->get("date_histogram"))) {
+            a.kind = AggDef::DATE_HISTOGRAM;
+            a.field = spec->at("field")->s;
+            const mj::Value* fi = spec->get("fixed_interval");
+            if (!fi) throw std::runtime_error("date_histogram requires fixed_interval");
+            a.interval = parse_duration_ms(fi->s);
+            if (const mj::Value* off = spec->get("offset"))
+                a.offset = off->kind == mj::Value::STR ? parse_duration_ms(off->s)
+                                                       : off->num();
+            if (const mj::Value* eb = spec->get("extended_bounds")) {
+                a.has_bounds = true;
+                a.bmin = eb->at("min")->num();
+                a.bmax = eb->at("max")->num();
+            }
+        } else if ((spec = body->get("histogram"))) {
+            a.kind = AggDef::HISTOGRAM;
+            a.field = spec->at("field")->s;
+            a.interval = spec->at("interval")->num();
+            if (const mj::Value* off = spec->get("offset")) a.offset = off->num();
+            if (const mj::Value* eb = spec->get("extended_bounds")) {
+                a.has_bounds = true;
+                a.bmin = eb->at("min")->num();
+                a.bmax = eb->at("max")->num();
+            }
+        } else if ((spec = body->get("terms"))) {
+            a.kind = AggDef::TERMS;
+            a.field = spec->at("field")->s;
+            if (const mj::Value* sz = spec->get("size")) a.size = uint32_t(sz->as_i64());
+            if (const mj::Value* mdc = spec->get("min_doc_count"))
+                a.min_doc_count = mdc->as_i64();
+        } else {
+            throw std::runtime_error("aggregation not supported: " + a.name);
+        }
+        if (a.min_doc_count < 0) a.min_doc_count = (a.kind == AggDef::TERMS) ? 1 : 0;
+        if (const mj::Value* subs = body->get("aggs")) {
+            for (auto& skv : subs->obj) {
+                MetricAgg m;
+                m.name = skv.first;
+                const mj::Value* sb = skv.second.get();
+                const mj::Value* ms = nullptr;
+                if ((ms = sb->get("stats"))) m.kind = MetricAgg::STATS;
+                else if ((ms = sb->get("avg"))) m.kind = MetricAgg::AVG;
+                else if ((ms = sb->get("sum"))) m.kind = MetricAgg::SUM;
+                else if ((ms = sb->get("min"))) m.kind = MetricAgg::MIN;
+                else if ((ms = sb->get("max"))) m.kind = MetricAgg::MAX;
+                else if ((ms = sb->get("value_count"))) m.kind = MetricAgg::COUNT;
+                else throw std::runtime_error("sub-aggregation not supported: " + m.name);
+                m.field = ms->at("field")->s;
+                a.sub.push_back(std::move(m));
+            }
+        }
+        out.push_back(std::move(a));
+    }
+    return out;
+}
+
+}  // namespace qw

@@ -1,0 +1,206 @@
+// QWA1 split container view: footer/meta parse + typed section pointers.
+// Format spec: DESIGN.md §3 (written by quickwit_amd/splitgen.py). This is
+// layout plumbing only — posting decode lives in the HIP kernels (product)
+// and in oracle/oracle.cpp (restatement), independently.
+#pragma once
+#include <cstdint>
+#include <cstring>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#include "minijson.h"
+
+namespace qw {
+
+#pragma pack(push, 1)
+struct SkipEntry {  // 16 B, matches splitgen.SKIP_DTYPE
+    uint32_t first_doc;
+    uint32_t last_doc;
+    uint32_t word_off;  // u32-word offset into the field's payload section
+    uint8_t id_bits;
+    uint8_t tf_bits;  // 0 = no tf section (record: basic)
+    uint16_t count;
+};
+#pragma pack(pop)
+static_assert(sizeof(SkipEntry) == 16, "skip entry layout");
+
+struct Section {
+    uint64_t off = 0, len = 0;
+};
+
+struct TextFieldView {
+    std::string name;
+    std::string tokenizer;  // "raw" | "default"
+    bool record_freq = false;
+    bool has_norms = false;
+    uint64_t total_tokens = 0;
+    uint32_t num_terms = 0;
+    Section term_offsets, term_bytes, posting_off, doc_freq, n_blocks, skip_off, skip,
+        payload, fieldnorms;
+    // host views
+    const uint32_t* h_term_offsets = nullptr;
+    const uint8_t* h_term_bytes = nullptr;
+    const uint64_t* h_posting_off = nullptr;
+    const uint32_t* h_doc_freq = nullptr;
+    const uint32_t* h_n_blocks = nullptr;
+    const uint64_t* h_skip_off = nullptr;
+    const SkipEntry* h_skip = nullptr;
+    const uint32_t* h_payload = nullptr;
+    const uint8_t* h_fieldnorms = nullptr;
+
+    // binary search the sorted term dictionary; -1 if absent
+    int64_t find_term(const char* t, size_t tlen) const {
+        int64_t lo = 0, hi = int64_t(num_terms) - 1;
+        while (lo <= hi) {
+            int64_t mid = (lo + hi) >> 1;
+            const char* s = (const char*)h_term_bytes + h_term_offsets[mid];
+            size_t slen = h_term_offsets[mid + 1] - h_term_offsets[mid];
+            int c = memcmp(s, t, slen < tlen ? slen : tlen);
+            if (c == 0) c = (slen < tlen) ? -1 : (slen > tlen ? 1 : 0);
+            if (c == 0) return mid;
+            if (c < 0) lo = mid + 1;
+            else hi = mid - 1;
+        }
+        return -1;
+    }
+};
+
+struct FastFieldView {
+    std::string name;
+    enum Type { U64, I64, DATETIME, STR } type = U64;
+    bool nullable = false;
+    uint32_t cardinality = 0;  // str
+    int ord_width = 0;         // str: 1/2/4
+    int64_t min_value = 0, max_value = 0;
+    Section values, nulls, dict_offsets, dict_bytes;
+    const void* h_values = nullptr;
+    const uint64_t* h_nulls = nullptr;  // bit d set = doc d has a value
+    const uint32_t* h_dict_offsets = nullptr;
+    const uint8_t* h_dict_bytes = nullptr;
+
+    bool present(uint32_t doc) const {
+        return !nullable || ((h_nulls[doc >> 6] >> (doc & 63)) & 1);
+    }
+    uint64_t ord(uint32_t doc) const {
+        switch (ord_width) {
+            case 1: return ((const uint8_t*)h_values)[doc];
+            case 2: return ((const uint16_t*)h_values)[doc];
+            default: return ((const uint32_t*)h_values)[doc];
+        }
+    }
+    int64_t i64(uint32_t doc) const { return ((const int64_t*)h_values)[doc]; }
+    uint64_t u64(uint32_t doc) const { return ((const uint64_t*)h_values)[doc]; }
+    std::string dict_entry(uint64_t o) const {
+        return std::string((const char*)h_dict_bytes + h_dict_offsets[o],
+                           h_dict_offsets[o + 1] - h_dict_offsets[o]);
+    }
+};
+
+struct SplitView {
+    std::string split_id;
+    uint32_t num_docs = 0;
+    std::string timestamp_field;
+    std::vector<TextFieldView> text_fields;
+    std::vector<FastFieldView> fast_fields;
+    const uint8_t* base = nullptr;  // host copy of the file image
+    size_t file_len = 0;
+
+    const TextFieldView* text_field(const std::string& n) const {
+        for (auto& f : text_fields)
+            if (f.name == n) return &f;
+        return nullptr;
+    }
+    const FastFieldView* fast_field(const std::string& n) const {
+        for (auto& f : fast_fields)
+            if (f.name == n) return &f;
+        return nullptr;
+    }
+
+    static Section sec(const mj::Value* m, const char* key, bool required = true) {
+        const mj::Value* v = m->get(key);
+        if (!v || v->is_null()) {
+            if (required) throw std::runtime_error(std::string("missing section ") + key);
+            return {};
+        }
+        return {uint64_t(v->arr.at(0)->as_i64()), uint64_t(v->arr.at(1)->as_i64())};
+    }
+
+    void parse(const uint8_t* data, size_t len) {
+        base = data;
+        file_len = len;
+        if (len < 64 + 24 || memcmp(data, "QWAMDSP1", 8) != 0 ||
+            memcmp(data + len - 8, "QWA1FOOT", 8) != 0)
+            throw std::runtime_error("bad QWA1 container");
+        uint64_t meta_off, meta_len;
+        memcpy(&meta_off, data + len - 24, 8);
+        memcpy(&meta_len, data + len - 16, 8);
+        if (meta_off + meta_len > len) throw std::runtime_error("bad QWA1 footer");
+        mj::ValuePtr meta = mj::parse((const char*)data + meta_off, meta_len);
+        split_id = meta->at("split_id")->s;
+        num_docs = uint32_t(meta->at("num_docs")->as_i64());
+        const mj::Value* tsf = meta->get("timestamp_field");
+        if (tsf && !tsf->is_null()) timestamp_field = tsf->s;
+        for (auto& fv : meta->at("fields")->arr) {
+            const mj::Value* f = fv.get();
+            std::string ty = f->at("type")->s;
+            const mj::Value* s = f->at("sec");
+            if (ty == "text") {
+                TextFieldView t;
+                t.name = f->at("name")->s;
+                t.tokenizer = f->at("tokenizer")->s;
+                t.record_freq = f->at("record")->s == "freq";
+                t.has_norms = f->at("fieldnorms")->b;
+                t.total_tokens = uint64_t(f->at("total_tokens")->as_i64());
+                t.num_terms = uint32_t(f->at("num_terms")->as_i64());
+                t.term_offsets = sec(s, "term_offsets");
+                t.term_bytes = sec(s, "term_bytes");
+                t.posting_off = sec(s, "posting_off");
+                t.doc_freq = sec(s, "doc_freq");
+                t.n_blocks = sec(s, "n_blocks");
+                t.skip_off = sec(s, "skip_off");
+                t.skip = sec(s, "skip");
+                t.payload = sec(s, "payload");
+                if (t.has_norms) t.fieldnorms = sec(s, "fieldnorms");
+                t.h_term_offsets = (const uint32_t*)(data + t.term_offsets.off);
+                t.h_term_bytes = data + t.term_bytes.off;
+                t.h_posting_off = (const uint64_t*)(data + t.posting_off.off);
+                t.h_doc_freq = (const uint32_t*)(data + t.doc_freq.off);
+                t.h_n_blocks = (const uint32_t*)(data + t.n_blocks.off);
+                t.h_skip_off = (const uint64_t*)(data + t.skip_off.off);
+                t.h_skip = (const SkipEntry*)(data + t.skip.off);
+                t.h_payload = (const uint32_t*)(data + t.payload.off);
+                if (t.has_norms) t.h_fieldnorms = data + t.fieldnorms.off;
+                text_fields.push_back(std::move(t));
+            } else {
+                FastFieldView ff;
+                ff.name = f->at("name")->s;
+                ff.type = ty == "u64"   ? FastFieldView::U64
+                          : ty == "i64" ? FastFieldView::I64
+                          : ty == "datetime" ? FastFieldView::DATETIME
+                                             : FastFieldView::STR;
+                ff.nullable = f->at("nullable")->b;
+                ff.values = sec(s, "values");
+                if (ff.nullable) ff.nulls = sec(s, "nulls");
+                if (ff.type == FastFieldView::STR) {
+                    ff.cardinality = uint32_t(f->at("cardinality")->as_i64());
+                    ff.ord_width = int(f->at("ord_width")->as_i64());
+                    ff.dict_offsets = sec(s, "dict_offsets");
+                    ff.dict_bytes = sec(s, "dict_bytes");
+                    ff.h_dict_offsets = (const uint32_t*)(data + ff.dict_offsets.off);
+                    ff.h_dict_bytes = data + ff.dict_bytes.off;
+                } else {
+                    const mj::Value* mn = f->get("min_value");
+                    const mj::Value* mx = f->get("max_value");
+                    if (mn) ff.min_value = mn->as_i64();
+                    if (mx) ff.max_value = mx->as_i64();
+                }
+                ff.h_values = data + ff.values.off;
+                if (ff.nullable) ff.h_nulls = (const uint64_t*)(data + ff.nulls.off);
+                fast_fields.push_back(std::move(ff));
+            }
+        }
+    }
+};
+
+}  // namespace qw

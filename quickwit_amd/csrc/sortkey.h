@@ -1,0 +1,78 @@
+// Monotonic sort-value <-> u64 maps and the PartialHit comparator.
+// Restates tantivy-common's MonotonicallyMappableToU64 (f64_to_u64 /
+// i64_to_u64) and the quickwit ordering rules of
+// docs/internals/sorting.md:14-26 + quickwit-search/src/top_k_collector.rs:
+//   - keys compared in the requested SortOrder, None (missing) always last;
+//   - ties broken by GlobalDocId (split_id, segment_ord, doc_id) in the
+//     order of the first sort field (Desc by default).
+#pragma once
+#include <cstdint>
+#include <cstring>
+#include <string>
+
+#include "pb.h"
+
+namespace qw {
+
+inline uint64_t f64_to_u64(double v) {
+    uint64_t bits;
+    memcpy(&bits, &v, 8);
+    return (bits & (1ULL << 63)) ? ~bits : bits | (1ULL << 63);
+}
+inline double u64_to_f64(uint64_t u) {
+    uint64_t bits = (u & (1ULL << 63)) ? u & ~(1ULL << 63) : ~u;
+    double d;
+    memcpy(&d, &bits, 8);
+    return d;
+}
+inline uint64_t i64_to_u64(int64_t v) { return uint64_t(v) ^ (1ULL << 63); }
+inline int64_t u64_to_i64(uint64_t u) { return int64_t(u ^ (1ULL << 63)); }
+
+// SortByValue -> comparable u64 key (same typed mapping the reference applies
+// in collector.rs:180-205). has=false means None.
+struct SortKey {
+    bool has = false;
+    uint64_t key = 0;
+};
+
+inline SortKey sort_key_of(const pb::SortByValue& v) {
+    SortKey k;
+    switch (v.kind) {
+        case pb::SortByValue::NONE: break;
+        case pb::SortByValue::U64: k = {true, v.u64}; break;
+        case pb::SortByValue::I64: k = {true, i64_to_u64(v.i64)}; break;
+        case pb::SortByValue::F64: k = {true, f64_to_u64(v.f64)}; break;
+        case pb::SortByValue::BOOL: k = {true, v.boolean ? 1ULL : 0ULL}; break;
+    }
+    return k;
+}
+
+// returns -1/0/+1: a before/equal/after b under (order1, order2); order 0=Asc
+// 1=Desc (search.proto:295). Missing values sort last under either order
+// (sorting.md:19-21).
+inline int cmp_key(const SortKey& a, const SortKey& b, int order) {
+    if (a.has != b.has) return a.has ? -1 : 1;  // None last
+    if (!a.has) return 0;
+    if (a.key == b.key) return 0;
+    bool a_first = (order == 1) ? (a.key > b.key) : (a.key < b.key);
+    return a_first ? -1 : 1;
+}
+
+// Full PartialHit comparator for leaf top-K and cross-split merge.
+inline bool hit_before(const pb::PartialHit& a, const pb::PartialHit& b, int order1,
+                       int order2) {
+    int c = cmp_key(sort_key_of(a.sort_value), sort_key_of(b.sort_value), order1);
+    if (c) return c < 0;
+    c = cmp_key(sort_key_of(a.sort_value2), sort_key_of(b.sort_value2), order2);
+    if (c) return c < 0;
+    // GlobalDocId tie-break in the first field's order (sorting.md:14-17)
+    int sc = a.split_id.compare(b.split_id);
+    if (sc) return order1 == 1 ? sc > 0 : sc < 0;
+    if (a.segment_ord != b.segment_ord)
+        return order1 == 1 ? a.segment_ord > b.segment_ord : a.segment_ord < b.segment_ord;
+    if (a.doc_id != b.doc_id)
+        return order1 == 1 ? a.doc_id > b.doc_id : a.doc_id < b.doc_id;
+    return false;
+}
+
+}  // namespace qw

@@ -108,8 +108,8 @@ def agg_golden():
         "schema": [
             {"name": "date", "type": "datetime", "fast": True, "nullable": True},
             {"name": "response", "type": "u64", "fast": True, "nullable": True},
-            {"name": "name", "type": "str_fast", "fast": True, "nullable": True},
-            {"name": "host", "type": "str_fast", "fast": True, "nullable": True},
+            {"name": "name", "type": "str", "fast": True, "nullable": True},
+            {"name": "host", "type": "str", "fast": True, "nullable": True},
             {"name": "id", "type": "u64", "fast": True, "nullable": True},
         ],
         "splits": splits,

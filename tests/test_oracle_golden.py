@@ -1,0 +1,145 @@
+"""Oracle vs the reference's golden vectors.
+
+- BM25 exact scores: tests/golden/bm25_sort.json (quickwit tests.rs:600-691).
+- Aggregation buckets: tests/golden/aggregations.json (rest-api-tests
+  aggregations scenarios, exact expected JSON).
+CPU-only: exercises oracle/liboracle.so through ctypes.
+"""
+import json
+import math
+import os
+import subprocess
+
+import pytest
+
+from quickwit_amd import splitgen
+from quickwit_amd.api import OracleSearcher, make_leaf_request
+
+GOLDEN = os.path.join(os.path.dirname(__file__), "golden")
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+@pytest.fixture(scope="module", autouse=True)
+def build_oracle():
+    subprocess.run(["make", "-s", "-C", os.path.join(REPO, "oracle")], check=True)
+
+
+def load(name):
+    with open(os.path.join(GOLDEN, name)) as f:
+        return json.load(f)
+
+
+# ------------------------------------------------------------------ BM25
+def bm25_schema(g):
+    return {"timestamp_field": None, "fields": [
+        {"name": f["name"], "type": "text", "tokenizer": "default",
+         "record": f["record"], "fieldnorms": f["fieldnorms"]}
+        for f in g["schema"]]}
+
+
+def test_bm25_golden_scores():
+    g = load("bm25_sort.json")
+    schema = bm25_schema(g)
+    w = splitgen.SplitWriter(schema, "split-bm25")
+    w.add_documents(g["docs"])
+    data = w.finalize()
+
+    s = OracleSearcher()
+    s.add_split("split-bm25", data)
+    for case in g["cases"]:
+        req = make_leaf_request(
+            case["query_ast"], schema, [("split-bm25", len(g["docs"]))],
+            max_hits=1000,
+            sort_fields=[{"field_name": "_score", "sort_order": 1}])
+        resp = s.leaf_search(req)
+        got = [(h["sort_value"]["f64"], h.get("doc_id", 0)) for h in resp["partial_hits"]]
+        exp = case["expected"]
+        assert len(got) == len(exp), case["name"]
+        for (gs, gd), (es, ed) in zip(got, exp):
+            assert gd == ed, (case["name"], got, exp)
+            assert math.isclose(gs, es, rel_tol=1e-6), (case["name"], gs, es)
+        assert resp["num_hits"] == len(exp)
+
+
+def test_bm25_tie_break_doc_desc():
+    # nofreq:two ties at 0.12343242 between docs 2 and 0 -> doc 2 first
+    # (GlobalDocId tie-break follows the Desc sort: sorting.md:14-17)
+    g = load("bm25_sort.json")
+    case = g["cases"][1]
+    assert case["expected"][1][1] == 2 and case["expected"][2][1] == 0
+
+
+# ------------------------------------------------------------- aggregations
+def es_query_to_ast(q):
+    if "match_all" in q:
+        return {"type": "match_all"}
+    if "bool" in q:
+        out = {"type": "bool"}
+        for clause in ("must", "must_not", "should", "filter"):
+            if clause in q["bool"]:
+                items = q["bool"][clause]
+                if isinstance(items, dict):
+                    items = [items]
+                out[clause] = [es_query_to_ast(i) for i in items]
+        return out
+    if "exists" in q:
+        return {"type": "field_presence", "field": q["exists"]["field"]}
+    raise ValueError(f"unsupported es query: {q}")
+
+
+def agg_schema(g):
+    type_map = {"str_fast": "str"}
+    return {"timestamp_field": None, "fields": [
+        {"name": f["name"], "type": type_map.get(f["type"], f["type"]), "fast": True}
+        for f in g["schema"]]}
+
+
+def deep_approx(got, exp, path=""):
+    if isinstance(exp, dict):
+        assert isinstance(got, dict), path
+        for k, v in exp.items():
+            assert k in got, f"{path}.{k} missing in {got}"
+            deep_approx(got[k], v, f"{path}.{k}")
+    elif isinstance(exp, list):
+        assert isinstance(got, list) and len(got) == len(exp), (path, got, exp)
+        for i, (gv, ev) in enumerate(zip(got, exp)):
+            deep_approx(gv, ev, f"{path}[{i}]")
+    elif isinstance(exp, float) or isinstance(exp, int) and not isinstance(exp, bool):
+        assert math.isclose(float(got), float(exp), rel_tol=1e-9, abs_tol=1e-9), \
+            (path, got, exp)
+    else:
+        assert got == exp, (path, got, exp)
+
+
+@pytest.fixture(scope="module")
+def agg_searcher():
+    g = load("aggregations.json")
+    schema = agg_schema(g)
+    s = OracleSearcher()
+    splits = []
+    for i, docs in enumerate(g["splits"]):
+        w = splitgen.SplitWriter(schema, f"agg-split-{i}")
+        w.add_documents(docs)
+        s.add_split(f"agg-split-{i}", w.finalize())
+        splits.append((f"agg-split-{i}", len(docs)))
+    return g, schema, s, splits
+
+
+@pytest.mark.parametrize("case_name", [
+    "date_histogram_basic",
+    "date_histogram_extended_bounds",
+    "date_histogram_stats_subagg",
+    "date_histogram_stats_subagg_exists_filter",
+    "terms_full",
+    "histogram_interval50",
+])
+def test_agg_golden(agg_searcher, case_name):
+    g, schema, s, splits = agg_searcher
+    case = g["cases"][case_name]
+    aggs = case["request"]["aggs"]
+    ast = es_query_to_ast(case["request"]["query"])
+    req = make_leaf_request(ast, schema, splits, max_hits=0, aggregation=aggs)
+    resp = s.leaf_search(req)
+    blob = resp["intermediate_aggregation_result"]
+    got = s.finalize_agg_json(blob, aggs)
+    deep_approx(got, case["expected"], case_name)
