@@ -1,0 +1,105 @@
+// Device-visible descriptors shared between the host orchestration
+// (product.cpp) and the HIP kernels (kernels.hip). All offsets are BYTE
+// offsets into the split's device image (the QWA1 file uploaded verbatim to
+// HBM — DESIGN.md §3/§5).
+#pragma once
+#include <cstdint>
+
+namespace qw {
+
+constexpr uint32_t TILE_DOCS = 8192;   // docs per workgroup tile
+constexpr uint32_t TILE_THREADS = 256; // 4 waves
+constexpr uint32_t AGG_LDS_BUCKETS = 2048;
+constexpr uint32_t TOPK_BINS = 4096;
+
+// role of a term in the flattened boolean
+enum TermRole : uint32_t { ROLE_SHOULD = 0, ROLE_MUST = 1, ROLE_MUST_NOT = 2 };
+
+struct TermDev {
+    uint32_t role;
+    uint32_t n_blocks;
+    uint64_t skip_off;     // byte offset of this term's first SkipEntry
+    uint64_t payload_off;  // byte offset of the FIELD's payload section
+    uint64_t norms_off;    // byte offset of the field's fieldnorms (0 = none)
+    float weight;          // BM25 W = idf*(1+k1)*boost (0 when not scoring)
+    uint32_t ktab_idx;     // index into the query's K tables (per field)
+    uint64_t ranges_off;   // byte offset into query scratch: u32 lo[ntiles],
+                           // u32 hi[ntiles] block-index ranges for this term
+};
+
+enum PredType : uint32_t {
+    PRED_RANGE_U64 = 0,
+    PRED_RANGE_I64 = 1,  // i64 / datetime(ms)
+    PRED_PRESENCE = 2,
+};
+enum PredFlags : uint32_t {
+    PRED_NEGATED = 1,       // must_not predicate
+    PRED_LO_INCLUDED = 2,
+    PRED_LO_EXCLUDED = 4,
+    PRED_HI_INCLUDED = 8,
+    PRED_HI_EXCLUDED = 16,
+};
+
+struct PredDev {
+    uint32_t type;
+    uint32_t flags;
+    int64_t lo, hi;        // bound values (u64 preds reinterpret the bits)
+    uint64_t values_off;   // column byte offset (0 for pure presence w/o col)
+    uint64_t nulls_off;    // null bitmap byte offset (0 = non-nullable)
+    uint32_t value_width;  // 8 for u64/i64; 1/2/4 for str ords
+};
+
+enum AggKindDev : uint32_t { AGGD_HISTO = 0, AGGD_TERMS = 1 };
+
+struct AggDev {
+    uint32_t kind;
+    uint32_t n_buckets;     // histo: bucket count from [lo_idx..hi_idx];
+                            // terms: cardinality
+    int64_t base_index;     // histo: floor((min-offset)/interval) at bucket 0
+    double interval;        // histo (ms for date_histogram)
+    double offset;          // histo
+    uint64_t values_off;    // column byte offset
+    uint64_t nulls_off;
+    uint32_t value_width;   // 8, or ord width for terms
+    uint32_t value_is_i64;  // 1 = signed column
+    uint64_t counts_out;    // byte offset into result scratch: u64[n_buckets]
+    uint64_t matched_out;   // terms: u64 counter of docs-with-value
+    // one optional stats sub-agg set per bucket: {u64 cnt, f64 sum, u64 min_s,
+    // u64 max_s (sortable-mapped)} × n_sub, laid out bucket-major
+    uint32_t n_sub;
+    uint64_t sub_out;        // byte offset: n_buckets * n_sub * 32 bytes
+    uint64_t sub_values_off[4];  // sub-agg source columns (<=4)
+    uint64_t sub_nulls_off[4];
+    uint32_t sub_width[4];
+    uint32_t sub_is_i64[4];
+};
+
+struct QueryDev {
+    const uint8_t* split;    // device base of the split image
+    const uint8_t* scratch;  // device base of the query scratch (ranges etc.)
+    uint8_t* results;        // device base of the results scratch
+    uint32_t num_docs;
+    uint32_t n_tiles;
+    uint32_t n_terms;
+    uint32_t n_must;      // count of ROLE_MUST terms
+    uint32_t n_must_not;  // count of ROLE_MUST_NOT terms
+    uint32_t n_preds;
+    uint32_t n_aggs;
+    uint32_t msm;           // minimum should match (0 = no should clauses)
+    uint32_t scoring;       // 1 = BM25 scores wanted
+    uint32_t match_all;     // 1 = base matches everything (no should clauses)
+    uint32_t collect_hits;  // 1 = write candidates
+    uint32_t sort_asc;      // 1 = ascending _score sort (flip candidate keys)
+    uint64_t terms_off;     // scratch offsets of descriptor arrays
+    uint64_t preds_off;
+    uint64_t aggs_off;
+    uint64_t ktabs_off;     // scratch: n_fields * 256 f32 BM25 K tables
+    // result offsets (all in `results`)
+    uint64_t tile_counts_off;  // u32[n_tiles]
+    uint64_t cand_count_off;   // u32 (atomic)
+    uint64_t cand_off;         // {u32 key, u32 doc}[cap]
+    uint64_t cand_cap;
+    uint64_t hist_off;         // u32[TOPK_BINS] histogram of candidate keys
+};
+
+}  // namespace qw

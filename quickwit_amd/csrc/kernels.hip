@@ -1,0 +1,418 @@
+// HIP/CDNA4 (gfx950) kernels for the leaf-search hot path — DESIGN.md §5.
+//
+// All kernels are HBM-bound integer/byte work (posting decode, boolean
+// combination, BM25 FMA, column predicate scans, bucket histograms); MFMA is
+// deliberately unused (nothing here is a dense contraction — BASELINE.json
+// north star). Shapes restated from the reference hot loops:
+//   - 128-doc bitpacked posting blocks + skip data (tantivy `bitpacking`,
+//     SURVEY.md §8a "posting block decode");
+//   - union/intersection combination (tantivy boolean scorers);
+//   - BM25 fused with decode (leaf.rs:929 searcher.search inner loop);
+//   - collect_block counting + top-K threshold (collector.rs:523-562,
+//     top_k_collector.rs) — here: per-tile ballot popcount + exact u64
+//     (sortkey, doc-tiebreak) selection;
+//   - fast-field predicate + aggregation column scans (range_query.rs:86,
+//     tantivy aggregation) — coalesced column gathers + LDS histograms.
+//
+// Execution geometry: one workgroup (256 threads = 4 wave64) owns a tile of
+// TILE_DOCS=8192 consecutive doc ids with an LDS f32 accumulator; posting
+// blocks overlapping the tile are decoded wave-per-block with funnel-shift
+// bit extraction and a 64-lane shuffle prefix-sum for the doc-id deltas.
+#include <hip/hip_runtime.h>
+
+#include "gpu_types.h"
+
+namespace qw {
+
+#pragma pack(push, 1)
+struct SkipEntryDev {  // mirrors qsplit.h SkipEntry (device-safe copy)
+    uint32_t first_doc;
+    uint32_t last_doc;
+    uint32_t word_off;
+    uint8_t id_bits;
+    uint8_t tf_bits;
+    uint16_t count;
+};
+#pragma pack(pop)
+
+// ---------------------------------------------------------------- helpers
+__device__ __forceinline__ uint32_t lane_id() { return threadIdx.x & 63u; }
+
+__device__ __forceinline__ uint64_t extract_bits(const uint32_t* base, uint64_t bitpos,
+                                                 uint32_t w) {
+    uint64_t word = bitpos >> 5;
+    uint32_t sh = uint32_t(bitpos & 31);
+    uint64_t v = (uint64_t(base[word]) | (uint64_t(base[word + 1]) << 32)) >> sh;
+    return v & ((1ull << w) - 1ull);  // w <= 32
+}
+
+__device__ __forceinline__ uint32_t wave_incl_scan_u32(uint32_t v) {
+    #pragma unroll
+    for (int d = 1; d < 64; d <<= 1) {
+        uint32_t n = __shfl_up(v, d, 64);
+        if ((threadIdx.x & 63u) >= uint32_t(d)) v += n;
+    }
+    return v;
+}
+
+__device__ __forceinline__ uint32_t f32_sortable(float f) {
+    uint32_t b = __float_as_uint(f);
+    return (b & 0x80000000u) ? ~b : (b | 0x80000000u);
+}
+
+__device__ __forceinline__ uint64_t f64_sortable(double d) {
+    uint64_t b = __double_as_longlong(d);
+    return (b & 0x8000000000000000ull) ? ~b : (b | 0x8000000000000000ull);
+}
+
+// ---------------------------------------------------------------- LDS state
+struct TileLds {
+    float score[TILE_DOCS];            // BM25 sums, or should-match counters
+    uint32_t bits_acc[TILE_DOCS / 32]; // must intersection
+    uint32_t bits_not[TILE_DOCS / 32]; // must_not union
+    uint32_t agg_hist[AGG_LDS_BUCKETS];
+    uint32_t agg_terms[AGG_LDS_BUCKETS];
+    uint32_t wave_base[4];
+};
+
+// decode every block of `t` overlapping the tile; accumulate into score[] /
+// set bits. mode: 0 = add score (or +1 count), 1 = set bits_acc-candidate
+// (bits_not reused as temp by caller), 2 = set bits_not
+__device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t tile,
+                                 uint32_t tile_lo, uint32_t tile_hi, TileLds& lds,
+                                 uint32_t* bitset, int add_score) {
+    const uint32_t* ranges = (const uint32_t*)(q.scratch + t.ranges_off);
+    uint32_t blo = ranges[tile], bhi = ranges[q.n_tiles + tile];
+    if (blo >= bhi) return;
+    const SkipEntryDev* skip = (const SkipEntryDev*)(q.split + t.skip_off);
+    const uint32_t* payload = (const uint32_t*)(q.split + t.payload_off);
+    const uint8_t* norms = t.norms_off ? q.split + t.norms_off : nullptr;
+    const float* ktab = (const float*)(q.scratch + q.ktabs_off) + 256 * t.ktab_idx;
+    uint32_t wave = threadIdx.x >> 6;
+    uint32_t lane = lane_id();
+    for (uint32_t blk = blo + wave; blk < bhi; blk += TILE_THREADS / 64) {
+        SkipEntryDev e = skip[blk];
+        if (e.first_doc >= tile_hi || e.last_doc < tile_lo) continue;
+        const uint32_t* idbase = payload + e.word_off;
+        uint32_t w = e.id_bits;
+        // lane handles elements 2*lane and 2*lane+1
+        uint32_t j0 = 2 * lane, j1 = 2 * lane + 1;
+        uint32_t g0 = uint32_t(extract_bits(idbase, uint64_t(j0) * w, w));
+        uint32_t g1 = uint32_t(extract_bits(idbase, uint64_t(j1) * w, w));
+        uint32_t incl = wave_incl_scan_u32(g0 + g1);
+        uint32_t doc1 = e.first_doc + incl;
+        uint32_t doc0 = doc1 - g1;
+        float s0 = 1.0f, s1 = 1.0f;
+        if (add_score && q.scoring) {
+            uint32_t tf0 = 1, tf1 = 1;
+            if (e.tf_bits) {
+                const uint32_t* tfbase =
+                    idbase + 2 * ((128u * e.id_bits + 63u) / 64u);
+                tf0 = uint32_t(extract_bits(tfbase, uint64_t(j0) * e.tf_bits,
+                                            e.tf_bits)) + 1u;
+                tf1 = uint32_t(extract_bits(tfbase, uint64_t(j1) * e.tf_bits,
+                                            e.tf_bits)) + 1u;
+            }
+            // BM25: W * tf / (tf + K[normid]) — same op order as the oracle
+            float K0 = ktab[norms ? norms[min(doc0, q.num_docs - 1)] : 1];
+            float K1 = ktab[norms ? norms[min(doc1, q.num_docs - 1)] : 1];
+            s0 = t.weight * (float(tf0) / (float(tf0) + K0));
+            s1 = t.weight * (float(tf1) / (float(tf1) + K1));
+        }
+        if (j0 < e.count && doc0 >= tile_lo && doc0 < tile_hi) {
+            uint32_t li = doc0 - tile_lo;
+            if (add_score) atomicAdd(&lds.score[li], s0);
+            else atomicOr(&bitset[li >> 5], 1u << (li & 31));
+        }
+        if (j1 < e.count && doc1 >= tile_lo && doc1 < tile_hi) {
+            uint32_t li = doc1 - tile_lo;
+            if (add_score) atomicAdd(&lds.score[li], s1);
+            else atomicOr(&bitset[li >> 5], 1u << (li & 31));
+        }
+    }
+}
+
+__device__ __forceinline__ bool eval_pred(const QueryDev& q, const PredDev& p,
+                                          uint32_t doc) {
+    bool ok = true;
+    if (p.nulls_off) {
+        const uint64_t* nulls = (const uint64_t*)(q.split + p.nulls_off);
+        ok = (nulls[doc >> 6] >> (doc & 63)) & 1;
+    }
+    if (ok && p.type != PRED_PRESENCE) {
+        if (p.type == PRED_RANGE_U64) {
+            uint64_t v = ((const uint64_t*)(q.split + p.values_off))[doc];
+            if (p.flags & PRED_LO_INCLUDED) ok &= v >= uint64_t(p.lo);
+            if (p.flags & PRED_LO_EXCLUDED) ok &= v > uint64_t(p.lo);
+            if (p.flags & PRED_HI_INCLUDED) ok &= v <= uint64_t(p.hi);
+            if (p.flags & PRED_HI_EXCLUDED) ok &= v < uint64_t(p.hi);
+        } else {
+            int64_t v = ((const int64_t*)(q.split + p.values_off))[doc];
+            if (p.flags & PRED_LO_INCLUDED) ok &= v >= p.lo;
+            if (p.flags & PRED_LO_EXCLUDED) ok &= v > p.lo;
+            if (p.flags & PRED_HI_INCLUDED) ok &= v <= p.hi;
+            if (p.flags & PRED_HI_EXCLUDED) ok &= v < p.hi;
+        }
+    }
+    return (p.flags & PRED_NEGATED) ? !ok : ok;
+}
+
+__device__ __forceinline__ double agg_value(const QueryDev& q, uint64_t values_off,
+                                            uint32_t width, uint32_t is_i64,
+                                            uint32_t doc) {
+    const uint8_t* col = q.split + values_off;
+    switch (width) {
+        case 1: return double(col[doc]);
+        case 2: return double(((const uint16_t*)col)[doc]);
+        case 4: return double(((const uint32_t*)col)[doc]);
+        default:
+            return is_i64 ? double(((const int64_t*)col)[doc])
+                          : double(((const uint64_t*)col)[doc]);
+    }
+}
+
+__device__ __forceinline__ uint64_t agg_ord(const QueryDev& q, uint64_t values_off,
+                                            uint32_t width, uint32_t doc) {
+    const uint8_t* col = q.split + values_off;
+    switch (width) {
+        case 1: return col[doc];
+        case 2: return ((const uint16_t*)col)[doc];
+        default: return ((const uint32_t*)col)[doc];
+    }
+}
+
+// ---------------------------------------------------------------- main kernel
+extern "C" __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile(QueryDev q,
+                                                                       uint32_t tile_base,
+                                                                       uint32_t tile_end,
+                                                                       uint32_t do_aggs,
+                                                                       uint32_t do_count,
+                                                                       uint32_t collect) {
+    __shared__ TileLds lds;
+    const TermDev* terms = (const TermDev*)(q.scratch + q.terms_off);
+    const PredDev* preds = (const PredDev*)(q.scratch + q.preds_off);
+    const AggDev* aggs = (const AggDev*)(q.scratch + q.aggs_off);
+    uint32_t* tile_counts = (uint32_t*)(q.results + q.tile_counts_off);
+    uint32_t* cand_count = (uint32_t*)(q.results + q.cand_count_off);
+    uint64_t* cand = (uint64_t*)(q.results + q.cand_off);
+
+    for (uint32_t tile = tile_base + blockIdx.x; tile < tile_end; tile += gridDim.x) {
+        uint32_t tile_lo = tile * TILE_DOCS;
+        uint32_t tile_hi = min(tile_lo + TILE_DOCS, q.num_docs);
+        bool have_should = q.msm > 0;
+        // ---- zero LDS
+        if (have_should)
+            for (uint32_t i = threadIdx.x; i < TILE_DOCS; i += TILE_THREADS)
+                lds.score[i] = 0.f;
+        if (q.n_must || q.n_must_not)
+            for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS) {
+                lds.bits_acc[i] = 0;
+                lds.bits_not[i] = 0;
+            }
+        if (do_aggs && q.n_aggs)
+            for (uint32_t i = threadIdx.x; i < AGG_LDS_BUCKETS; i += TILE_THREADS) {
+                lds.agg_hist[i] = 0;
+                lds.agg_terms[i] = 0;
+            }
+        __syncthreads();
+
+        // ---- should terms: score / count union
+        for (uint32_t t = 0; t < q.n_terms; ++t)
+            if (terms[t].role == ROLE_SHOULD)
+                decode_term_tile(q, terms[t], tile, tile_lo, tile_hi, lds, nullptr, 1);
+        __syncthreads();
+
+        // ---- must terms: intersection via per-term bitset
+        for (uint32_t t = 0, mi = 0; t < q.n_terms; ++t) {
+            if (terms[t].role != ROLE_MUST) continue;
+            // temp bitset lives in bits_not while must_nots are not yet done
+            uint32_t* tmp = lds.bits_not;
+            for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS)
+                tmp[i] = 0;
+            __syncthreads();
+            decode_term_tile(q, terms[t], tile, tile_lo, tile_hi, lds, tmp, 0);
+            if (q.scoring)  // musts contribute to the score too
+                decode_term_tile(q, terms[t], tile, tile_lo, tile_hi, lds, nullptr, 1);
+            __syncthreads();
+            for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS)
+                lds.bits_acc[i] = mi == 0 ? tmp[i] : (lds.bits_acc[i] & tmp[i]);
+            ++mi;
+            __syncthreads();
+        }
+        if (q.n_must_not) {
+            for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS)
+                lds.bits_not[i] = 0;
+            __syncthreads();
+            for (uint32_t t = 0; t < q.n_terms; ++t)
+                if (terms[t].role == ROLE_MUST_NOT)
+                    decode_term_tile(q, terms[t], tile, tile_lo, tile_hi, lds,
+                                     lds.bits_not, 0);
+            __syncthreads();
+        }
+
+        // ---- epilogue: matched test, count, candidates, aggregations
+        uint32_t local_count = 0;
+        for (uint32_t d = tile_lo + threadIdx.x; d < tile_hi; d += TILE_THREADS) {
+            uint32_t li = d - tile_lo;
+            bool m = true;
+            float sc = 0.f;
+            if (!q.match_all) {
+                if (have_should) {
+                    if (q.scoring) {
+                        sc = lds.score[li];
+                        m = sc > 0.f;
+                    } else m = lds.score[li] >= float(q.msm);
+                } else m = q.n_must > 0;
+                if (m && q.n_must) m = (lds.bits_acc[li >> 5] >> (li & 31)) & 1;
+                if (q.scoring && m && !have_should) sc = lds.score[li];
+            }
+            if (m && q.n_must_not) m = !((lds.bits_not[li >> 5] >> (li & 31)) & 1);
+            for (uint32_t p = 0; m && p < q.n_preds; ++p) m = eval_pred(q, preds[p], d);
+
+            if (m) {
+                ++local_count;
+                if (do_aggs)
+                    for (uint32_t ai = 0; ai < q.n_aggs; ++ai) {
+                        const AggDev& a = aggs[ai];
+                        if (a.nulls_off) {
+                            const uint64_t* nulls = (const uint64_t*)(q.split + a.nulls_off);
+                            if (!((nulls[d >> 6] >> (d & 63)) & 1)) continue;
+                        }
+                        if (a.kind == AGGD_TERMS) {
+                            uint64_t o = agg_ord(q, a.values_off, a.value_width, d);
+                            if (ai == 1 && a.n_buckets <= AGG_LDS_BUCKETS)
+                                atomicAdd(&lds.agg_terms[o], 1u);
+                            else
+                                atomicAdd((unsigned long long*)(q.results + a.counts_out) + o,
+                                          1ull);
+                            atomicAdd((unsigned long long*)(q.results + a.matched_out), 1ull);
+                        } else {
+                            double v = agg_value(q, a.values_off, a.value_width,
+                                                 a.value_is_i64, d);
+                            int64_t idx =
+                                int64_t(floor((v - a.offset) / a.interval)) - a.base_index;
+                            if (idx < 0 || idx >= int64_t(a.n_buckets)) continue;
+                            if (ai == 0 && a.n_buckets <= AGG_LDS_BUCKETS)
+                                atomicAdd(&lds.agg_hist[uint32_t(idx)], 1u);
+                            else
+                                atomicAdd((unsigned long long*)(q.results + a.counts_out) +
+                                              idx, 1ull);
+                            for (uint32_t si = 0; si < a.n_sub; ++si) {
+                                if (a.sub_nulls_off[si]) {
+                                    const uint64_t* sn =
+                                        (const uint64_t*)(q.split + a.sub_nulls_off[si]);
+                                    if (!((sn[d >> 6] >> (d & 63)) & 1)) continue;
+                                }
+                                double sv = agg_value(q, a.sub_values_off[si],
+                                                      a.sub_width[si], a.sub_is_i64[si], d);
+                                uint8_t* slot = q.results + a.sub_out +
+                                                (uint64_t(idx) * a.n_sub + si) * 32;
+                                atomicAdd((unsigned long long*)slot, 1ull);
+                                atomicAdd((double*)(slot + 8), sv);
+                                atomicMin((unsigned long long*)(slot + 16),
+                                          (unsigned long long)f64_sortable(sv));
+                                atomicMax((unsigned long long*)(slot + 24),
+                                          (unsigned long long)f64_sortable(sv));
+                            }
+                        }
+                    }
+            }
+            // candidate append (wave-aggregated). combined exact key:
+            // high 32 = sortable score (flipped for asc), low 32 = doc
+            // tie-break in the order1 direction (sorting.md:14-17)
+            if (collect) {
+                bool write = m;
+                uint64_t key = 0;
+                if (write) {
+                    uint32_t kh = q.scoring ? f32_sortable(sc) : 0u;
+                    if (q.sort_asc) kh = ~kh;
+                    uint32_t kl = q.sort_asc ? ~d : d;
+                    key = (uint64_t(kh) << 32) | kl;
+                }
+                uint64_t mask = __ballot(write);
+                uint32_t nw = __popcll(mask);
+                if (nw) {
+                    uint32_t base;
+                    if (lane_id() == __ffsll((unsigned long long)mask) - 1)
+                        base = atomicAdd(cand_count, nw);
+                    base = __shfl(base, __ffsll((unsigned long long)mask) - 1, 64);
+                    if (write) {
+                        uint32_t off =
+                            __popcll(mask & ((1ull << lane_id()) - 1ull));
+                        if (base + off < q.cand_cap) cand[base + off] = key;
+                    }
+                }
+            }
+        }
+        // ---- flush LDS agg arrays
+        if (do_aggs)
+            for (uint32_t ai = 0; ai < q.n_aggs && ai < 2; ++ai) {
+                const AggDev& a = aggs[ai];
+                if (a.n_buckets > AGG_LDS_BUCKETS) continue;
+                uint32_t* src = a.kind == AGGD_TERMS ? lds.agg_terms : lds.agg_hist;
+                if ((a.kind == AGGD_TERMS) != (ai == 1)) continue;
+                __syncthreads();
+                for (uint32_t i = threadIdx.x; i < a.n_buckets; i += TILE_THREADS)
+                    if (src[i])
+                        atomicAdd((unsigned long long*)(q.results + a.counts_out) + i,
+                                  (unsigned long long)src[i]);
+            }
+        // ---- per-tile count
+        if (do_count) {
+            __syncthreads();
+            uint32_t* wb = lds.wave_base;
+            #pragma unroll
+            for (int dlt = 32; dlt; dlt >>= 1) local_count += __shfl_down(local_count, dlt, 64);
+            if (lane_id() == 0) wb[threadIdx.x >> 6] = local_count;
+            __syncthreads();
+            if (threadIdx.x == 0)
+                tile_counts[tile] = wb[0] + wb[1] + wb[2] + wb[3];
+        }
+        __syncthreads();
+    }
+}
+
+// ----------------------------------------------------- top-K selection passes
+// histogram of the top 12 bits (after `shift`) of candidates matching
+// (key >> prefix_shift) == prefix
+extern "C" __global__ void k_cand_hist(const uint64_t* cand, uint32_t n, uint64_t prefix,
+                                       uint32_t prefix_bits, uint32_t* hist) {
+    __shared__ uint32_t lh[TOPK_BINS];
+    for (uint32_t i = threadIdx.x; i < TOPK_BINS; i += blockDim.x) lh[i] = 0;
+    __syncthreads();
+    uint32_t shift = 64 - prefix_bits - 12;
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * blockDim.x) {
+        uint64_t k = cand[i];
+        if (prefix_bits && (k >> (64 - prefix_bits)) != prefix) continue;
+        atomicAdd(&lh[(k >> shift) & (TOPK_BINS - 1)], 1u);
+    }
+    __syncthreads();
+    for (uint32_t i = threadIdx.x; i < TOPK_BINS; i += blockDim.x)
+        if (lh[i]) atomicAdd(&hist[i], lh[i]);
+}
+
+// compact candidates with key >= floor_key into out (bounded by cap)
+extern "C" __global__ void k_cand_compact(const uint64_t* cand, uint32_t n,
+                                          uint64_t floor_key, uint64_t ceil_key,
+                                          uint64_t* out, uint32_t* out_count,
+                                          uint32_t cap) {
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * blockDim.x) {
+        uint64_t k = cand[i];
+        bool take = k >= floor_key && k <= ceil_key;
+        uint64_t mask = __ballot(take);
+        uint32_t nw = __popcll(mask);
+        if (!nw) continue;
+        uint32_t leader = __ffsll((unsigned long long)mask) - 1;
+        uint32_t base;
+        if (lane_id() == leader) base = atomicAdd(out_count, nw);
+        base = __shfl(base, leader, 64);
+        if (take) {
+            uint32_t off = __popcll(mask & ((1ull << lane_id()) - 1ull));
+            if (base + off < cap) out[base + off] = k;
+        }
+    }
+}
+
+}  // namespace qw
