@@ -1,0 +1,215 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: Quickwit leaf-search hot path on MI355X.
+
+Workload (BASELINE.json configs[1], the single-GPU configuration the metric
+is quoted on): 3-term BM25 disjunction (OR) over 10M synthetic log docs
+(hdfs-logs schema, seeded generator — SURVEY.md §8d), 1 split per GPU,
+max_hits=10, sorted by _score desc. A "step" is one full leaf_search call
+over the split batch, inputs already resident in HBM (the reference's warm
+state that cpu_search_microsecs times, leaf.rs:905-946) + the cross-rank
+top-K merge when N>1.
+
+Contract: python bench.py --gpus N --steps K --warmup W
+  N>1 is launched by the driver via torch.distributed.run, one rank per GPU
+  over RCCL; splits shard one-batch-per-GPU ("scaling": "weak");
+  the only exchange is the response allgather + rank-0 merge.
+Rank 0 prints ONE JSON line with metric/value plus:
+  roofline: dominant kernel (union_bm25) algorithmic-bytes/launch over its
+    HIP-event launch time, vs 8 TB/s HBM3E peak (MI355X_MICROARCH.md).
+    Algorithmic bytes = posting payload+skip bytes of the 3 query terms
+    + 1B fieldnorm + 8B candidate record per union hit + per-tile metadata
+     — counted from the generated index, not from DRAM traffic.
+  cpu_baseline: the OpenMP oracle (reference restatement, kind "port") timed
+    on the same split on this box's host cores (bounded sample).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+HBM_PEAK_GBS = 8000.0  # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
+
+
+def build_query(terms):
+    return {"type": "bool", "should": [
+        {"type": "term", "field": "body", "value": t} for t in terms]}
+
+
+def algo_bytes_per_launch(split_bytes, terms, num_union, n_tiles):
+    """Algorithmic HBM bytes one k_leaf_tile launch must move."""
+    from quickwit_amd import splitread
+    sp = splitread.Split(split_bytes)
+    f = sp.fields["body"]
+    posting_off = sp._sec("body", "posting_off", "<u8")
+    n_blocks = sp._sec("body", "n_blocks", "<u4")
+    total = 0
+    for t in terms:
+        tid = sp.term_id("body", t)
+        if tid is None:
+            continue
+        total += int(posting_off[tid + 1] - posting_off[tid])  # packed payload
+        total += int(n_blocks[tid]) * 16                       # skip entries
+        total += 2 * 4 * n_tiles                               # block ranges
+    total += num_union * (1 + 8)  # fieldnorm gather + candidate record write
+    total += n_tiles * 4          # tile counts
+    return total
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--docs", type=int, default=10_000_000,
+                    help="docs per split (one split per GPU)")
+    ap.add_argument("--max-hits", type=int, default=10)
+    ap.add_argument("--cpu-baseline-steps", type=int, default=2)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    import torch
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        dist.init_process_group(backend)
+        if torch.cuda.is_available():
+            torch.cuda.set_device(local_rank)
+
+    from quickwit_amd import proto, splitgen
+    from quickwit_amd.api import GpuSearcher, OracleSearcher, make_leaf_request
+    from quickwit_amd.merge import merge_leaf_responses
+
+    t_gen = time.perf_counter()
+    split_bytes = splitgen.generate_split(rank, args.docs, seed=42)
+    sid = f"synthetic-42-{rank:04d}"
+    gen_s = time.perf_counter() - t_gen
+
+    searcher = GpuSearcher(device=local_rank)
+    searcher.add_split(sid, split_bytes)
+
+    terms = ["w%05d" % i for i in range(3)]  # highest-df Zipf terms (df≈10%)
+    req = make_leaf_request(
+        build_query(terms), splitgen.HDFS_SCHEMA, [(sid, args.docs)],
+        max_hits=args.max_hits,
+        sort_fields=[{"field_name": "_score", "sort_order": 1}])
+    req_pb = proto.encode("LeafSearchRequest", req)
+    sreq_pb = proto.encode("SearchRequest", req["search_request"])
+
+    def one_step():
+        resp_pb = searcher.leaf_search_raw(req_pb)
+        if world > 1:
+            gathered = [None] * world
+            dist.all_gather_object(gathered, resp_pb)
+            if rank == 0:
+                return merge_leaf_responses(sreq_pb, gathered)
+        return resp_pb
+
+    for _ in range(args.warmup):
+        last = one_step()
+    searcher.kernel_stats_reset()
+
+    if dist:
+        dist.barrier()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    step_times = []
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        ts = time.perf_counter()
+        last = one_step()
+        step_times.append(time.perf_counter() - ts)
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    if dist:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+        dist.barrier()
+
+    if rank != 0:
+        return
+
+    resp = proto.decode("LeafSearchResponse", last)
+    num_union = resp["num_hits"] // world if world > 1 else resp["num_hits"]
+    total_docs = args.docs * world
+    value = args.steps * total_docs / elapsed
+    ms_per_step = elapsed / args.steps * 1e3
+    p50_ms = sorted(step_times)[len(step_times) // 2] * 1e3
+
+    # roofline of the dominant kernel
+    kms, launches = searcher.kernel_stats("union_bm25")
+    roofline = None
+    if launches:
+        ms_per_launch = kms / launches
+        n_tiles = (args.docs + 8192 - 1) // 8192
+        ab = algo_bytes_per_launch(split_bytes, terms, num_union, n_tiles)
+        achieved = ab / (ms_per_launch / 1e3) / 1e9
+        roofline = {"bound": "hbm", "achieved": round(achieved, 1),
+                    "peak": HBM_PEAK_GBS, "unit": "GB/s",
+                    "frac": round(achieved / HBM_PEAK_GBS, 4),
+                    "traffic": None,  # filled from rocprofv3 --pmc (profiles/)
+                    "kernel": "union_bm25",
+                    "algo_bytes_per_launch": ab,
+                    "ms_per_launch": round(ms_per_launch, 4)}
+
+    # CPU baseline: the oracle restatement on this box's host cores (rank 0,
+    # N=1 only; bounded sample). OpenMP parallelism is across splits; with one
+    # split the per-split closure is single-threaded like the reference's.
+    cpu_baseline = None
+    if world == 1 and args.cpu_baseline_steps > 0:
+        cpu = OracleSearcher()
+        cpu.add_split(sid, split_bytes)
+        cpu.leaf_search_raw(req_pb)  # warm
+        tc = time.perf_counter()
+        for _ in range(args.cpu_baseline_steps):
+            cpu.leaf_search_raw(req_pb)
+        tcpu = (time.perf_counter() - tc) / args.cpu_baseline_steps
+        cpu_baseline = {
+            "value": round(args.docs / tcpu, 1), "unit": "docs/s", "cores": 1,
+            "kind": "port",
+            "sample": f"{args.cpu_baseline_steps} leaf_search calls over the same "
+                      f"{args.docs}-doc split ({tcpu:.2f}s each)"}
+
+    out = {
+        "metric": "leaf_search_docs_per_sec",
+        "value": round(value, 1),
+        "unit": "docs/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(ms_per_step, 3),
+        "p50_ms": round(p50_ms, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,  # no published leaf-search number
+        "dtype": "u32+f32",
+        "data": "synthetic",
+        "config": {
+            "workload": "3term_bm25_or_10M" if args.docs == 10_000_000
+                        else f"3term_bm25_or_{args.docs}",
+            "query": " OR ".join(f"body:{t}" for t in terms),
+            "docs_per_gpu": args.docs,
+            "splits_per_gpu": 1,
+            "max_hits": args.max_hits,
+            "num_union_hits_per_split": num_union,
+            "parallelism": f"split-dp{world}",
+            "gen_seconds": round(gen_s, 1),
+        },
+        "roofline": roofline,
+        "cpu_baseline": cpu_baseline,
+    }
+    print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

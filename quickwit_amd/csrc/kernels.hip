@@ -199,9 +199,10 @@ extern "C" __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile(QueryDev 
     for (uint32_t tile = tile_base + blockIdx.x; tile < tile_end; tile += gridDim.x) {
         uint32_t tile_lo = tile * TILE_DOCS;
         uint32_t tile_hi = min(tile_lo + TILE_DOCS, q.num_docs);
-        bool have_should = q.msm > 0;
+        bool have_should = q.msm > 0;  // shoulds REQUIRED (msm>=1); msm==0 =>
+                                       // shoulds optional, only add scores
         // ---- zero LDS
-        if (have_should)
+        if (q.n_terms)
             for (uint32_t i = threadIdx.x; i < TILE_DOCS; i += TILE_THREADS)
                 lds.score[i] = 0.f;
         if (q.n_must || q.n_must_not)
@@ -254,17 +255,23 @@ extern "C" __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile(QueryDev 
         uint32_t local_count = 0;
         for (uint32_t d = tile_lo + threadIdx.x; d < tile_hi; d += TILE_THREADS) {
             uint32_t li = d - tile_lo;
-            bool m = true;
+            bool m;
             float sc = 0.f;
-            if (!q.match_all) {
-                if (have_should) {
-                    if (q.scoring) {
-                        sc = lds.score[li];
-                        m = sc > 0.f;
-                    } else m = lds.score[li] >= float(q.msm);
-                } else m = q.n_must > 0;
+            if (have_should) {
+                // shoulds required: scoring path assumes msm==1 and all
+                // weights > 0 (host rejects anything else — product.cpp)
+                if (q.scoring) {
+                    sc = lds.score[li];
+                    m = sc > 0.f;
+                } else m = lds.score[li] >= float(q.msm);
                 if (m && q.n_must) m = (lds.bits_acc[li >> 5] >> (li & 31)) & 1;
-                if (q.scoring && m && !have_should) sc = lds.score[li];
+            } else {
+                // base = match_all (no must terms; preds filter below) or the
+                // must-term intersection; optional shoulds only add scores
+                m = q.match_all ? true
+                                : (q.n_must > 0 &&
+                                   ((lds.bits_acc[li >> 5] >> (li & 31)) & 1));
+                if (q.scoring && m) sc = lds.score[li];
             }
             if (m && q.n_must_not) m = !((lds.bits_not[li >> 5] >> (li & 31)) & 1);
             for (uint32_t p = 0; m && p < q.n_preds; ++p) m = eval_pred(q, preds[p], d);

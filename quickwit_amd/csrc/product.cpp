@@ -1,0 +1,1181 @@
+// ============================================================================
+// libquickwit_amd — MI355X-native host orchestration for the Quickwit
+// leaf-search hot path (the PRODUCT; C-ABI in include/quickwit_amd.h).
+//
+// This is the work the reference ships to its rayon pool inside
+// leaf_search_single_split (quickwit/quickwit-search/src/leaf.rs:899-959):
+// query build (doc_mapper_impl.rs:637 -> tantivy_query_ast.rs:153-374),
+// posting decode + boolean combination + BM25 + collection
+// (searcher.search at leaf.rs:929, collector.rs:523-594), fast-field
+// predicates (range_query.rs:86-158) and aggregations, top-K with the
+// reference tie-breaks (top_k_collector.rs, docs/internals/sorting.md:14-26),
+// and the cross-split merge (collector.rs:1198, :832-861).
+//
+// All index arithmetic runs in HIP kernels on gfx950 (kernels.hip, included
+// below as one translation unit — no -fgpu-rdc needed). There is NO CPU
+// fallback: without a HIP device every search returns QW_ERR_NO_GPU
+// (DESIGN.md §1). The CPU restatement lives in oracle/ (test infra only).
+// ============================================================================
+#include <hip/hip_runtime.h>
+
+#include <algorithm>
+#include <chrono>
+#include <cmath>
+#include <cstdint>
+#include <cstdlib>
+#include <cstring>
+#include <map>
+#include <memory>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+#include "../../include/quickwit_amd.h"
+#include "fieldnorm.h"
+#include "minijson.h"
+#include "pb.h"
+#include "qagg_format.h"
+#include "qast.h"
+#include "qsplit.h"
+#include "sortkey.h"
+
+// device code: single TU (kernels + host) so no relocatable device link
+#include "kernels.hip"
+
+namespace qw {
+
+#define HIP_CHECK(expr)                                                        \
+    do {                                                                       \
+        hipError_t _e = (expr);                                                \
+        if (_e != hipSuccess)                                                  \
+            throw std::runtime_error(std::string("HIP error: ") +              \
+                                     hipGetErrorString(_e) + " at " #expr);    \
+    } while (0)
+
+struct DeviceSplit {
+    std::vector<uint8_t> host;  // host copy: term-dict lookups + meta parse
+    SplitView view;             // views into `host`
+    uint8_t* d_image = nullptr; // full QWA1 file image resident in HBM
+    size_t len = 0;
+};
+
+struct KernelTimer {
+    double total_ms = 0;
+    uint64_t launches = 0;
+};
+
+struct DevBuf {  // grow-only device scratch
+    uint8_t* p = nullptr;
+    size_t cap = 0;
+    void ensure(size_t n) {
+        if (n <= cap) return;
+        if (p) (void)hipFree(p);
+        p = nullptr;
+        cap = 0;
+        HIP_CHECK(hipMalloc(&p, n));
+        cap = n;
+    }
+    ~DevBuf() {
+        if (p) (void)hipFree(p);
+    }
+};
+
+}  // namespace qw
+
+struct qw_ctx {
+    int device = 0;
+    bool device_ready = false;
+    int64_t agg_bucket_limit = 65000;
+    std::map<std::string, std::unique_ptr<qw::DeviceSplit>> splits;
+    std::string last_error;
+    hipStream_t stream = nullptr;
+    hipEvent_t ev_start = nullptr, ev_stop = nullptr;
+    qw::DevBuf d_scratch, d_results, d_survivors;
+    std::map<std::string, qw::KernelTimer> timers;
+};
+
+static void ctx_ensure_device(qw_ctx* ctx) {
+    using namespace qw;
+    if (ctx->device_ready) return;
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess || n == 0)
+        throw std::runtime_error("no HIP device visible");
+    if (ctx->device >= n) throw std::runtime_error("bad device ordinal");
+    HIP_CHECK(hipSetDevice(ctx->device));
+    HIP_CHECK(hipStreamCreateWithFlags(&ctx->stream, hipStreamNonBlocking));
+    HIP_CHECK(hipEventCreate(&ctx->ev_start));
+    HIP_CHECK(hipEventCreate(&ctx->ev_stop));
+    ctx->device_ready = true;
+}
+
+namespace qw {
+
+// ------------------------------------------------------------ flattening
+// The kernel executes ONE flattened boolean level (should/must/must_not term
+// lists + fast-field predicates). Nested plans that cannot be flattened are
+// rejected with InvalidQuery — never silently mis-answered. This covers every
+// BASELINE.json config and the golden-suite query shapes (term, full_text
+// and/or, bool with term/range clauses, range, match_all).
+struct FlatQuery {
+    bool match_none = false;
+    bool match_all = false;  // positive base = all docs (preds still filter)
+    bool scoring = false;
+    uint32_t msm = 0;  // >0: shoulds required (count >= msm)
+    struct FTerm {
+        const TextFieldView* f;
+        int64_t tid;
+        uint32_t role;
+        float weight;  // BM25 W (0 for unscored filter terms)
+    };
+    std::vector<FTerm> terms;
+    std::vector<PredDev> preds;
+    std::vector<const TextFieldView*> ktab_fields;  // ktab_idx -> field
+
+    uint32_t ktab_for(const TextFieldView* f) {
+        for (size_t i = 0; i < ktab_fields.size(); ++i)
+            if (ktab_fields[i] == f) return uint32_t(i);
+        ktab_fields.push_back(f);
+        return uint32_t(ktab_fields.size() - 1);
+    }
+};
+
+static float bm25_weight(const SplitView& sv, const TextFieldView& f, int64_t tid,
+                         float boost) {
+    double N = double(sv.num_docs);
+    double df = double(f.h_doc_freq[tid]);
+    double idf = std::log(1.0 + (N - df + 0.5) / (df + 0.5));
+    return float(idf * (1.0 + 1.2) * double(boost));
+}
+
+static void add_range_pred(FlatQuery& fq, const SplitView& sv, const PlanNode& n,
+                           bool negated) {
+    const FastFieldView* f = sv.fast_field(n.field);
+    PredDev p{};
+    if (!f) {
+        // unknown fast field: matches nothing (oracle eval_range -> empty)
+        if (negated) return;  // must_not nothing = no-op
+        fq.match_none = true;
+        return;
+    }
+    p.type = f->type == FastFieldView::U64 ? PRED_RANGE_U64 : PRED_RANGE_I64;
+    p.flags = negated ? PRED_NEGATED : 0;
+    if (n.kind == PlanNode::FIELD_PRESENCE) {
+        if (!f->nullable && !negated) return;  // non-nullable: always present
+        if (!f->nullable && negated) {
+            fq.match_none = true;
+            return;
+        }
+        p.type = PRED_PRESENCE;
+    } else {
+        if (n.lo.kind == Bound::INCLUDED) p.flags |= PRED_LO_INCLUDED;
+        if (n.lo.kind == Bound::EXCLUDED) p.flags |= PRED_LO_EXCLUDED;
+        if (n.hi.kind == Bound::INCLUDED) p.flags |= PRED_HI_INCLUDED;
+        if (n.hi.kind == Bound::EXCLUDED) p.flags |= PRED_HI_EXCLUDED;
+        p.lo = n.lo.ival;
+        p.hi = n.hi.ival;
+        if (f->type == FastFieldView::STR)
+            throw std::runtime_error("range over str fast field not supported");
+    }
+    p.values_off = f->values.off;
+    p.nulls_off = f->nullable ? f->nulls.off : 0;
+    p.value_width = 8;
+    fq.preds.push_back(p);
+}
+
+static void add_term(FlatQuery& fq, const SplitView& sv, const std::string& field,
+                     const std::string& value, uint32_t role, float boost,
+                     bool scored) {
+    const TextFieldView* f = sv.text_field(field);
+    int64_t tid = f ? f->find_term(value.data(), value.size()) : -1;
+    if (tid < 0) {
+        // absent term: MUST -> split matches nothing; SHOULD/MUST_NOT -> skip
+        // (skipped shoulds can never satisfy msm — handled by caller count)
+        if (role == ROLE_MUST) fq.match_none = true;
+        return;
+    }
+    FlatQuery::FTerm t;
+    t.f = f;
+    t.tid = tid;
+    t.role = role;
+    t.weight = scored ? bm25_weight(sv, *f, tid, boost) : 0.f;
+    fq.terms.push_back(t);
+}
+
+static bool is_pure_should_terms(const PlanNode& n) {
+    if (n.kind != PlanNode::BOOL) return false;
+    if (!n.must.empty() || !n.must_not.empty() || !n.filter.empty()) return false;
+    if (n.minimum_should_match > 1) return false;
+    for (auto& c : n.should)
+        if (c.kind != PlanNode::TERM) return false;
+    return true;
+}
+
+static bool is_pure_must_terms(const PlanNode& n) {
+    if (n.kind != PlanNode::BOOL) return false;
+    if (!n.should.empty() || !n.must_not.empty() || !n.filter.empty()) return false;
+    for (auto& c : n.must)
+        if (c.kind != PlanNode::TERM) return false;
+    return true;
+}
+
+static void add_positive(FlatQuery& fq, const SplitView& sv, const PlanNode& c,
+                         bool scored, float boost);
+
+static void flatten_bool(FlatQuery& fq, const SplitView& sv, const PlanNode& n,
+                         float boost);
+
+// a MUST/FILTER clause
+static void add_positive(FlatQuery& fq, const SplitView& sv, const PlanNode& c,
+                         bool scored, float boost) {
+    switch (c.kind) {
+        case PlanNode::TERM:
+            add_term(fq, sv, c.field, c.value, ROLE_MUST, boost * c.boost, scored);
+            break;
+        case PlanNode::RANGE:
+        case PlanNode::FIELD_PRESENCE:
+            add_range_pred(fq, sv, c, false);
+            break;
+        case PlanNode::MATCH_ALL:
+            break;  // no constraint
+        case PlanNode::MATCH_NONE:
+            fq.match_none = true;
+            break;
+        case PlanNode::BOOL:
+            if (is_pure_must_terms(c)) {
+                for (auto& m : c.must)
+                    add_term(fq, sv, m.field, m.value, ROLE_MUST,
+                             boost * c.boost * m.boost, scored);
+            } else if (is_pure_should_terms(c) && c.should.size() == 1) {
+                add_term(fq, sv, c.should[0].field, c.should[0].value, ROLE_MUST,
+                         boost * c.boost * c.should[0].boost, scored);
+            } else {
+                throw std::runtime_error(
+                    "nested boolean inside must/filter not flattenable (GPU r1)");
+            }
+            break;
+    }
+}
+
+static void flatten_bool(FlatQuery& fq, const SplitView& sv, const PlanNode& n,
+                         float boost) {
+    boost *= n.boost;
+    bool has_req = !n.must.empty() || !n.filter.empty();
+    for (auto& c : n.must) add_positive(fq, sv, c, fq.scoring, boost);
+    for (auto& c : n.filter) add_positive(fq, sv, c, false, boost);
+
+    // shoulds: flatten TERM / pure-OR bool children into SHOULD terms
+    size_t should_clauses = 0;
+    for (auto& c : n.should) {
+        if (c.kind == PlanNode::TERM) {
+            add_term(fq, sv, c.field, c.value, ROLE_SHOULD, boost * c.boost,
+                     fq.scoring);
+            ++should_clauses;
+        } else if (is_pure_should_terms(c)) {
+            // OR-of-terms child: each term is an independent SHOULD here; one
+            // clause satisfied iff any term matches — valid for msm<=1 only
+            for (auto& t : c.should)
+                add_term(fq, sv, t.field, t.value, ROLE_SHOULD,
+                         boost * c.boost * t.boost, fq.scoring);
+            ++should_clauses;
+        } else if (c.kind == PlanNode::MATCH_NONE) {
+            // contributes nothing
+        } else {
+            throw std::runtime_error("should clause not flattenable (GPU r1)");
+        }
+    }
+
+    for (auto& c : n.must_not) {
+        if (c.kind == PlanNode::TERM)
+            add_term(fq, sv, c.field, c.value, ROLE_MUST_NOT, 1.f, false);
+        else if (c.kind == PlanNode::RANGE || c.kind == PlanNode::FIELD_PRESENCE)
+            add_range_pred(fq, sv, c, true);
+        else if (is_pure_should_terms(c))
+            // not(a or b) = not a and not b: must_not is a union-then-subtract
+            for (auto& t : c.should)
+                add_term(fq, sv, t.field, t.value, ROLE_MUST_NOT, 1.f, false);
+        else if (c.kind == PlanNode::MATCH_NONE) {
+        } else if (c.kind == PlanNode::MATCH_ALL) {
+            fq.match_none = true;
+        } else
+            throw std::runtime_error("must_not clause not flattenable (GPU r1)");
+    }
+
+    int64_t msm = n.minimum_should_match;
+    if (msm < 0) msm = has_req ? 0 : 1;
+    bool nested_or = false;
+    for (auto& c : n.should) nested_or |= c.kind != PlanNode::TERM;
+    if (msm > 1 && nested_or)
+        throw std::runtime_error("minimum_should_match>1 over nested clauses");
+    if (should_clauses == 0 && msm > 0 && !n.should.empty()) {
+        fq.match_none = true;  // all required shoulds were absent terms
+        return;
+    }
+    if (!has_req && n.should.empty()) {
+        fq.match_none = true;  // no positive clause matches nothing
+        return;
+    }
+    fq.msm = uint32_t(msm);
+    uint32_t n_must_terms = 0, n_should_terms = 0;
+    for (auto& t : fq.terms) {
+        n_must_terms += t.role == ROLE_MUST;
+        n_should_terms += t.role == ROLE_SHOULD;
+    }
+    if (msm > 0 && n_should_terms == 0) {
+        fq.match_none = true;  // required shoulds but every term absent
+        return;
+    }
+    if (fq.scoring && msm > 0 && n_must_terms > 0)
+        throw std::runtime_error("scored msm>0 with must terms not supported (GPU r1)");
+    if (fq.scoring && msm > 1)
+        throw std::runtime_error("scored minimum_should_match>1 not supported (GPU r1)");
+    // base: match_all when no must terms (pred-only/should-optional base)
+    fq.match_all = n_must_terms == 0 && (has_req || msm == 0);
+    if (!has_req && msm == 0 && n.should.empty()) fq.match_none = true;
+}
+
+static FlatQuery flatten(const SplitView& sv, const PlanNode& plan, bool scoring) {
+    FlatQuery fq;
+    fq.scoring = scoring;
+    switch (plan.kind) {
+        case PlanNode::MATCH_ALL:
+            fq.match_all = true;
+            break;
+        case PlanNode::MATCH_NONE:
+            fq.match_none = true;
+            break;
+        case PlanNode::TERM:
+            add_term(fq, sv, plan.field, plan.value, ROLE_SHOULD, plan.boost, scoring);
+            if (fq.terms.empty()) fq.match_none = true;
+            fq.msm = 1;
+            break;
+        case PlanNode::RANGE:
+        case PlanNode::FIELD_PRESENCE:
+            fq.match_all = true;
+            add_range_pred(fq, sv, plan, false);
+            break;
+        case PlanNode::BOOL:
+            flatten_bool(fq, sv, plan, 1.f);
+            break;
+    }
+    if (fq.scoring)
+        for (auto& t : fq.terms)
+            if (t.role == ROLE_SHOULD && fq.msm > 0 && t.weight <= 0.f)
+                throw std::runtime_error("non-positive BM25 weight (boost<=0?)");
+    return fq;
+}
+
+// ------------------------------------------------------------ agg planning
+struct AggPlan {
+    std::vector<AggDef> defs;
+    std::vector<AggDev> devs;            // one per def
+    std::vector<const FastFieldView*> fields;
+    size_t out_bytes = 0;                // total result bytes (counts + subs)
+    std::vector<uint8_t> init;           // initial contents of the out region
+};
+
+static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
+                         int64_t bucket_limit, uint64_t out_base) {
+    AggPlan ap;
+    ap.defs = parse_agg_request(agg_json);
+    uint64_t off = out_base;
+    for (const AggDef& d : ap.defs) {
+        const FastFieldView* f = sv.fast_field(d.field);
+        ap.fields.push_back(f);
+        AggDev a{};
+        if (d.kind == AggDef::TERMS) {
+            a.kind = AGGD_TERMS;
+            if (f && f->type == FastFieldView::STR) {
+                a.n_buckets = f->cardinality;
+                a.values_off = f->values.off;
+                a.nulls_off = f->nullable ? f->nulls.off : 0;
+                a.value_width = uint32_t(f->ord_width);
+            } else {
+                a.n_buckets = 0;  // no such column: zero buckets
+            }
+        } else {
+            a.kind = AGGD_HISTO;
+            if (f && f->type != FastFieldView::STR) {
+                a.interval = d.interval;
+                a.offset = d.offset;
+                a.values_off = f->values.off;
+                a.nulls_off = f->nullable ? f->nulls.off : 0;
+                a.value_width = 8;
+                a.value_is_i64 = f->type != FastFieldView::U64;
+                double mn = double(f->min_value), mx = double(f->max_value);
+                if (f->type == FastFieldView::U64) {
+                    mn = double(uint64_t(f->min_value));
+                    mx = double(uint64_t(f->max_value));
+                }
+                int64_t b0 = int64_t(std::floor((mn - d.offset) / d.interval));
+                int64_t b1 = int64_t(std::floor((mx - d.offset) / d.interval));
+                a.base_index = b0;
+                int64_t nb = b1 - b0 + 1;
+                if (nb < 0 || nb > bucket_limit)
+                    throw std::runtime_error(
+                        "aggregation bucket limit exceeded (AggregationLimitsGuard)");
+                a.n_buckets = uint32_t(nb);
+                a.n_sub = uint32_t(std::min<size_t>(d.sub.size(), 4));
+                if (d.sub.size() > 4)
+                    throw std::runtime_error(">4 metric sub-aggregations");
+                for (size_t si = 0; si < d.sub.size(); ++si) {
+                    const FastFieldView* sf = sv.fast_field(d.sub[si].field);
+                    if (sf && sf->type != FastFieldView::STR) {
+                        a.sub_values_off[si] = sf->values.off;
+                        a.sub_nulls_off[si] = sf->nullable ? sf->nulls.off : 0;
+                        a.sub_width[si] = 8;
+                        a.sub_is_i64[si] = sf->type != FastFieldView::U64;
+                    } else {
+                        a.sub_values_off[si] = 0;  // missing column: no values
+                        a.sub_width[si] = 0;
+                    }
+                }
+            } else {
+                a.n_buckets = 0;
+            }
+        }
+        // layout: counts u64[n_buckets] | matched u64 | subs n_buckets*n_sub*32
+        a.counts_out = off;
+        off += uint64_t(a.n_buckets) * 8;
+        a.matched_out = off;
+        off += 8;
+        a.sub_out = off;
+        off += uint64_t(a.n_buckets) * a.n_sub * 32;
+        ap.devs.push_back(a);
+    }
+    ap.out_bytes = off - out_base;
+    // init pattern: zeros except sub min slots = f64_sortable(+inf) pattern max
+    ap.init.assign(ap.out_bytes, 0);
+    for (size_t i = 0; i < ap.devs.size(); ++i) {
+        const AggDev& a = ap.devs[i];
+        for (uint64_t b = 0; b < a.n_buckets; ++b)
+            for (uint32_t s = 0; s < a.n_sub; ++s) {
+                uint64_t slot = a.sub_out - out_base + (b * a.n_sub + s) * 32;
+                uint64_t ones = ~0ull;
+                memcpy(&ap.init[slot + 16], &ones, 8);  // min slot: max sortable
+                // max slot stays 0 (= most negative sortable)
+            }
+    }
+    return ap;
+}
+
+// ------------------------------------------------------------ per-split search
+struct SplitResult {
+    uint64_t num_hits = 0;
+    std::vector<pb::PartialHit> hits;
+    IntermediateAggResults aggs;
+    bool has_aggs = false;
+    uint64_t micros = 0;
+    std::string error;
+};
+
+struct SortPlan {
+    bool by_score = false;  // else doc-id order
+    int order1 = 1, order2 = 1;
+};
+
+static void record_kernel_time(qw_ctx* ctx, const char* name, float ms) {
+    KernelTimer& t = ctx->timers[name];
+    t.total_ms += ms;
+    t.launches += 1;
+}
+
+static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
+                                    const pb::SearchRequest& req, const Schema& schema) {
+    SplitResult out;
+    auto t0 = std::chrono::steady_clock::now();
+    const SplitView& sv = ds.view;
+
+    PlanNode plan = parse_query_ast(req.query_ast, schema);
+    if ((req.start_timestamp || req.end_timestamp) && !schema.timestamp_field.empty()) {
+        // leaf.rs:977 rewrite_request: [start,end) seconds over the ts column
+        PlanNode ts;
+        ts.kind = PlanNode::RANGE;
+        ts.field = schema.timestamp_field;
+        if (req.start_timestamp) {
+            ts.lo.kind = Bound::INCLUDED;
+            ts.lo.ival = *req.start_timestamp * 1000;
+        }
+        if (req.end_timestamp) {
+            ts.hi.kind = Bound::EXCLUDED;
+            ts.hi.ival = *req.end_timestamp * 1000;
+        }
+        PlanNode b;
+        b.kind = PlanNode::BOOL;
+        b.filter.push_back(std::move(ts));
+        b.must.push_back(std::move(plan));
+        plan = std::move(b);
+    }
+
+    SortPlan sp;
+    if (!req.sort_fields.empty()) {
+        sp.order1 = req.sort_fields[0].sort_order;
+        if (req.sort_fields.size() > 1) sp.order2 = req.sort_fields[1].sort_order;
+        if (req.sort_fields[0].field_name == "_score") sp.by_score = true;
+        else
+            throw std::runtime_error(
+                "sort by fast field not supported on GPU path in r1 (use _score or "
+                "default doc order)");
+        if (req.sort_fields.size() > 1)
+            throw std::runtime_error("two sort fields not supported on GPU path in r1");
+    }
+
+    FlatQuery fq = flatten(sv, plan, sp.by_score);
+    uint64_t leaf_max_hits = req.max_hits + req.start_offset;
+
+    if (fq.match_none) {
+        if (req.aggregation_request) {
+            // empty-but-shaped blob, like the oracle over an empty match set
+            for (const AggDef& d : parse_agg_request(*req.aggregation_request)) {
+                AggResult r;
+                r.name = d.name;
+                r.kind = d.kind == AggDef::TERMS      ? 3
+                         : d.kind == AggDef::HISTOGRAM ? 2
+                                                       : 1;
+                for (auto& s : d.sub) r.sub_names.push_back(s.name);
+                out.aggs.aggs.push_back(std::move(r));
+            }
+            out.has_aggs = true;
+        }
+        out.micros = uint64_t(std::chrono::duration_cast<std::chrono::microseconds>(
+                                  std::chrono::steady_clock::now() - t0)
+                                  .count());
+        return out;
+    }
+
+    AggPlan ap;
+    bool do_aggs = req.aggregation_request.has_value();
+
+    uint32_t n_tiles = (sv.num_docs + TILE_DOCS - 1) / TILE_DOCS;
+    bool pure_match_all = fq.match_all && fq.preds.empty() && fq.terms.empty();
+    bool need_kernel = !pure_match_all || do_aggs;
+    // candidate collection: needed unless hits are trivially enumerable
+    bool collect = leaf_max_hits > 0 && !pure_match_all && !fq.match_none;
+
+    // ---- scratch assembly (descriptors + ktabs + block ranges), one H2D
+    std::vector<TermDev> terms(fq.terms.size());
+    uint32_t n_must = 0, n_must_not = 0;
+    size_t ranges_bytes = 0;
+    for (size_t i = 0; i < fq.terms.size(); ++i) ranges_bytes += 2ull * n_tiles * 4;
+    std::vector<float> ktabs(fq.ktab_fields.size() * 0);  // filled below
+
+    // K tables per field actually referenced by scored terms
+    std::vector<const TextFieldView*> ktab_fields;
+    auto ktab_idx_of = [&](const TextFieldView* f) {
+        for (size_t i = 0; i < ktab_fields.size(); ++i)
+            if (ktab_fields[i] == f) return uint32_t(i);
+        ktab_fields.push_back(f);
+        return uint32_t(ktab_fields.size() - 1);
+    };
+    for (size_t i = 0; i < fq.terms.size(); ++i) {
+        const FlatQuery::FTerm& t = fq.terms[i];
+        TermDev& d = terms[i];
+        d.role = t.role;
+        d.n_blocks = t.f->h_n_blocks[t.tid];
+        d.skip_off = t.f->skip.off + t.f->h_skip_off[t.tid];
+        d.payload_off = t.f->payload.off;
+        d.norms_off = t.f->has_norms ? t.f->fieldnorms.off : 0;
+        d.weight = t.weight;
+        d.ktab_idx = fq.scoring ? ktab_idx_of(t.f) : 0;
+        n_must += t.role == ROLE_MUST;
+        n_must_not += t.role == ROLE_MUST_NOT;
+    }
+    ktabs.resize(std::max<size_t>(1, ktab_fields.size()) * 256, 0.f);
+    for (size_t k = 0; k < ktab_fields.size(); ++k) {
+        const TextFieldView* f = ktab_fields[k];
+        double avgdl = f->total_tokens > 0 && sv.num_docs > 0
+                           ? double(f->total_tokens) / double(sv.num_docs)
+                           : 0.0;
+        const double K1 = 1.2, B = 0.75;
+        for (int i = 0; i < 256; ++i) {
+            double fn = double(FIELDNORM_TABLE.v[i]);
+            ktabs[k * 256 + i] =
+                float(K1 * (1.0 - B + B * fn / (avgdl > 0 ? avgdl : 1.0)));
+        }
+    }
+
+    // scratch layout
+    size_t off_terms = 0;
+    size_t off_preds = off_terms + terms.size() * sizeof(TermDev);
+    size_t off_aggs = off_preds + fq.preds.size() * sizeof(PredDev);
+    // aggs descriptors appended after planning (below)
+    // results layout
+    size_t r_tile_counts = 0;
+    size_t r_cand_count = r_tile_counts + size_t(n_tiles) * 4;
+    r_cand_count = (r_cand_count + 63) & ~size_t(63);
+    size_t r_hist = r_cand_count + 64;  // cand_count u32 + survivors count u32
+    size_t r_agg = r_hist + TOPK_BINS * 4;
+    r_agg = (r_agg + 63) & ~size_t(63);
+
+    if (do_aggs)
+        ap = plan_aggs(sv, *req.aggregation_request, ctx->agg_bucket_limit, r_agg);
+    size_t r_cand = r_agg + ((ap.out_bytes + 63) & ~size_t(63));
+    size_t cand_cap = collect ? sv.num_docs : 0;
+    size_t results_bytes = r_cand + cand_cap * 8;
+
+    size_t off_ktabs = off_aggs + ap.devs.size() * sizeof(AggDev);
+    size_t off_ranges = off_ktabs + ktabs.size() * 4;
+    off_ranges = (off_ranges + 63) & ~size_t(63);
+    size_t scratch_bytes = off_ranges + ranges_bytes;
+
+    // block ranges per term per tile (host two-pointer walk over skip entries)
+    std::vector<uint32_t> ranges(ranges_bytes / 4, 0);
+    for (size_t i = 0; i < fq.terms.size(); ++i) {
+        const FlatQuery::FTerm& t = fq.terms[i];
+        const SkipEntry* sk = t.f->h_skip + t.f->h_skip_off[t.tid] / 16;
+        uint32_t nb = t.f->h_n_blocks[t.tid];
+        uint32_t* lo = ranges.data() + (i * 2ull * n_tiles);
+        uint32_t* hi = lo + n_tiles;
+        uint32_t b_lo = 0, b_hi = 0;
+        for (uint32_t tile = 0; tile < n_tiles; ++tile) {
+            uint32_t tlo = tile * TILE_DOCS;
+            uint32_t thi = std::min<uint32_t>(tlo + TILE_DOCS, sv.num_docs);
+            while (b_lo < nb && sk[b_lo].last_doc < tlo) ++b_lo;
+            if (b_hi < b_lo) b_hi = b_lo;
+            while (b_hi < nb && sk[b_hi].first_doc < thi) ++b_hi;
+            lo[tile] = b_lo;
+            hi[tile] = b_hi;
+        }
+        terms[i].ranges_off = off_ranges + i * 2ull * n_tiles * 4;
+    }
+
+    std::vector<uint8_t> scratch(scratch_bytes, 0);
+    memcpy(scratch.data() + off_terms, terms.data(), terms.size() * sizeof(TermDev));
+    memcpy(scratch.data() + off_preds, fq.preds.data(),
+           fq.preds.size() * sizeof(PredDev));
+    memcpy(scratch.data() + off_aggs, ap.devs.data(), ap.devs.size() * sizeof(AggDev));
+    memcpy(scratch.data() + off_ktabs, ktabs.data(), ktabs.size() * 4);
+    memcpy(scratch.data() + off_ranges, ranges.data(), ranges_bytes);
+
+    uint64_t matched = 0;
+    std::vector<uint64_t> top_keys;  // survivors, sorted best-first
+
+    if (need_kernel) {
+        ctx->d_scratch.ensure(scratch_bytes);
+        ctx->d_results.ensure(results_bytes);
+        HIP_CHECK(hipMemcpyAsync(ctx->d_scratch.p, scratch.data(), scratch_bytes,
+                                 hipMemcpyHostToDevice, ctx->stream));
+        HIP_CHECK(hipMemsetAsync(ctx->d_results.p, 0, r_agg, ctx->stream));
+        if (do_aggs && ap.out_bytes)
+            HIP_CHECK(hipMemcpyAsync(ctx->d_results.p + r_agg, ap.init.data(),
+                                     ap.out_bytes, hipMemcpyHostToDevice, ctx->stream));
+
+        QueryDev q{};
+        q.split = ds.d_image;
+        q.scratch = ctx->d_scratch.p;
+        q.results = ctx->d_results.p;
+        q.num_docs = sv.num_docs;
+        q.n_tiles = n_tiles;
+        q.n_terms = uint32_t(terms.size());
+        q.n_must = n_must;
+        q.n_must_not = n_must_not;
+        q.n_preds = uint32_t(fq.preds.size());
+        q.n_aggs = uint32_t(ap.devs.size());
+        q.msm = fq.msm;
+        q.scoring = fq.scoring ? 1 : 0;
+        q.match_all = fq.match_all ? 1 : 0;
+        q.collect_hits = collect ? 1 : 0;
+        q.sort_asc = (sp.by_score && sp.order1 == 0) ? 1 : 0;
+        q.terms_off = off_terms;
+        q.preds_off = off_preds;
+        q.aggs_off = off_aggs;
+        q.ktabs_off = off_ktabs;
+        q.tile_counts_off = r_tile_counts;
+        q.cand_count_off = r_cand_count;
+        q.cand_off = r_cand;
+        q.cand_cap = cand_cap;
+        q.hist_off = r_hist;
+
+        uint32_t grid = n_tiles;  // one workgroup per tile; gfx950 has 256 CUs
+        const char* kname = !fq.terms.empty() ? "union_bm25"
+                            : do_aggs         ? "column_agg"
+                                              : "range_filter";
+        HIP_CHECK(hipEventRecord(ctx->ev_start, ctx->stream));
+        hipLaunchKernelGGL(k_leaf_tile, dim3(grid), dim3(TILE_THREADS), 0, ctx->stream,
+                           q, 0u, n_tiles, uint32_t(do_aggs ? 1 : 0), 1u,
+                           uint32_t(collect ? 1 : 0));
+        HIP_CHECK(hipEventRecord(ctx->ev_stop, ctx->stream));
+        HIP_CHECK(hipGetLastError());
+
+        // ---- download counts
+        std::vector<uint32_t> tile_counts(n_tiles);
+        HIP_CHECK(hipMemcpyAsync(tile_counts.data(), ctx->d_results.p + r_tile_counts,
+                                 size_t(n_tiles) * 4, hipMemcpyDeviceToHost,
+                                 ctx->stream));
+        uint32_t cand_n = 0;
+        if (collect)
+            HIP_CHECK(hipMemcpyAsync(&cand_n, ctx->d_results.p + r_cand_count, 4,
+                                     hipMemcpyDeviceToHost, ctx->stream));
+        HIP_CHECK(hipStreamSynchronize(ctx->stream));
+        float ms = 0;
+        HIP_CHECK(hipEventElapsedTime(&ms, ctx->ev_start, ctx->ev_stop));
+        record_kernel_time(ctx, kname, ms);
+        for (uint32_t c : tile_counts) matched += c;
+
+        // ---- top-K selection over candidates (device histogram refinement
+        // + host exact sort of the survivors; top_k_collector.rs semantics)
+        if (collect && cand_n > 0) {
+            uint64_t K = std::min<uint64_t>(leaf_max_hits, cand_n);
+            uint64_t* d_cand = (uint64_t*)(ctx->d_results.p + r_cand);
+            uint32_t* d_hist = (uint32_t*)(ctx->d_results.p + r_hist);
+            uint32_t* d_scount = (uint32_t*)(ctx->d_results.p + r_cand_count) + 1;
+            uint64_t prefix = 0;
+            uint32_t prefix_bits = 0;
+            uint64_t survivors = cand_n, above = 0, Krem = K;
+            uint64_t floor_key = 0;
+            std::vector<uint32_t> hist(TOPK_BINS);
+            HIP_CHECK(hipEventRecord(ctx->ev_start, ctx->stream));
+            while (true) {
+                if (survivors <= std::max<uint64_t>(4 * K, 65536) || prefix_bits >= 48) {
+                    if (prefix_bits == 0) floor_key = 0;
+                    break;
+                }
+                HIP_CHECK(hipMemsetAsync(d_hist, 0, TOPK_BINS * 4, ctx->stream));
+                uint32_t hgrid = std::min<uint32_t>(2048, (cand_n + 255) / 256);
+                hipLaunchKernelGGL(k_cand_hist, dim3(hgrid), dim3(256), 0, ctx->stream,
+                                   d_cand, cand_n, prefix, prefix_bits, d_hist);
+                HIP_CHECK(hipMemcpyAsync(hist.data(), d_hist, TOPK_BINS * 4,
+                                         hipMemcpyDeviceToHost, ctx->stream));
+                HIP_CHECK(hipStreamSynchronize(ctx->stream));
+                uint64_t cum = 0;
+                int b = TOPK_BINS - 1;
+                for (; b >= 0; --b) {
+                    if (cum + hist[b] >= Krem) break;
+                    cum += hist[b];
+                }
+                if (b < 0) b = 0;
+                uint32_t shift = 64 - prefix_bits - 12;
+                floor_key = (prefix_bits ? (prefix << (64 - prefix_bits)) : 0) |
+                            (uint64_t(uint32_t(b)) << shift);
+                above += cum;
+                survivors = above + hist[b];
+                Krem = Krem - cum;
+                prefix = (prefix << 12) | uint64_t(uint32_t(b));
+                prefix_bits += 12;
+            }
+            ctx->d_survivors.ensure(survivors * 8 + 8);
+            HIP_CHECK(hipMemsetAsync(d_scount, 0, 4, ctx->stream));
+            uint32_t cgrid = std::min<uint32_t>(2048, (cand_n + 255) / 256);
+            hipLaunchKernelGGL(k_cand_compact, dim3(cgrid), dim3(256), 0, ctx->stream,
+                               d_cand, cand_n, floor_key, ~0ull,
+                               (uint64_t*)ctx->d_survivors.p, d_scount,
+                               uint32_t(survivors));
+            uint32_t sn = 0;
+            HIP_CHECK(hipMemcpyAsync(&sn, d_scount, 4, hipMemcpyDeviceToHost,
+                                     ctx->stream));
+            HIP_CHECK(hipStreamSynchronize(ctx->stream));
+            sn = std::min<uint32_t>(sn, uint32_t(survivors));
+            top_keys.resize(sn);
+            HIP_CHECK(hipMemcpy(top_keys.data(), ctx->d_survivors.p, size_t(sn) * 8,
+                                hipMemcpyDeviceToHost));
+            HIP_CHECK(hipEventRecord(ctx->ev_stop, ctx->stream));
+            HIP_CHECK(hipStreamSynchronize(ctx->stream));
+            float tms = 0;
+            HIP_CHECK(hipEventElapsedTime(&tms, ctx->ev_start, ctx->ev_stop));
+            record_kernel_time(ctx, "topk_select", tms);
+            std::sort(top_keys.begin(), top_keys.end(), std::greater<uint64_t>());
+            if (top_keys.size() > K) top_keys.resize(K);
+        }
+    } else {
+        matched = sv.num_docs;  // pure match_all, no aggs
+    }
+    if (pure_match_all) matched = sv.num_docs;
+
+    out.num_hits = matched;
+
+    // ---- build PartialHits
+    auto mk_hit = [&](uint32_t doc, float score) {
+        pb::PartialHit h;
+        h.split_id = sv.split_id;
+        h.segment_ord = 0;
+        h.doc_id = doc;
+        if (sp.by_score) {
+            h.sort_value.kind = pb::SortByValue::F64;
+            h.sort_value.f64 = double(score);
+        }
+        return h;
+    };
+    if (leaf_max_hits > 0) {
+        if (pure_match_all) {
+            // hits enumerable without the kernel: doc-id order (desc default)
+            uint64_t k = std::min<uint64_t>(leaf_max_hits, sv.num_docs);
+            for (uint64_t i = 0; i < k; ++i) {
+                uint32_t doc = sp.order1 == 1 ? uint32_t(sv.num_docs - 1 - i)
+                                              : uint32_t(i);
+                out.hits.push_back(mk_hit(doc, 0.f));
+            }
+        } else {
+            bool asc = sp.by_score && sp.order1 == 0;
+            for (uint64_t key : top_keys) {
+                uint32_t kh = uint32_t(key >> 32), kl = uint32_t(key);
+                if (asc) {
+                    kh = ~kh;
+                    kl = ~kl;
+                }
+                float score = 0.f;
+                if (fq.scoring) {
+                    uint32_t b = (kh & 0x80000000u) ? (kh & ~0x80000000u) : ~kh;
+                    memcpy(&score, &b, 4);
+                }
+                out.hits.push_back(mk_hit(kl, score));
+            }
+        }
+    }
+
+    // ---- aggregation download + assembly (QAGG1 intermediate)
+    if (do_aggs) {
+        std::vector<uint8_t> agg_out(ap.out_bytes);
+        if (need_kernel && ap.out_bytes) {
+            HIP_CHECK(hipMemcpy(agg_out.data(), ctx->d_results.p + r_agg, ap.out_bytes,
+                                hipMemcpyDeviceToHost));
+        }
+        for (size_t i = 0; i < ap.defs.size(); ++i) {
+            const AggDef& d = ap.defs[i];
+            const AggDev& a = ap.devs[i];
+            const FastFieldView* f = ap.fields[i];
+            AggResult r;
+            r.name = d.name;
+            for (auto& s : d.sub) r.sub_names.push_back(s.name);
+            const uint8_t* base = agg_out.data() + (a.counts_out - r_agg);
+            const uint64_t* counts = (const uint64_t*)base;
+            if (d.kind == AggDef::TERMS) {
+                r.kind = 3;
+                if (a.n_buckets) {
+                    uint64_t m = 0;
+                    memcpy(&m, agg_out.data() + (a.matched_out - r_agg), 8);
+                    r.terms_matched_docs = m;
+                    for (uint32_t o = 0; o < a.n_buckets; ++o)
+                        if (counts[o])
+                            r.term_counts.emplace_back(f->dict_entry(o), counts[o]);
+                }
+            } else {
+                r.kind = d.kind == AggDef::DATE_HISTOGRAM ? 1 : 2;
+                const uint8_t* subs = agg_out.data() + (a.sub_out - r_agg);
+                for (uint32_t bi = 0; bi < a.n_buckets; ++bi) {
+                    if (!counts[bi]) continue;
+                    AggBucket b;
+                    b.key = double(a.base_index + int64_t(bi)) * d.interval + d.offset;
+                    b.doc_count = counts[bi];
+                    b.sub.resize(d.sub.size());
+                    for (uint32_t s = 0; s < a.n_sub; ++s) {
+                        const uint8_t* slot = subs + (uint64_t(bi) * a.n_sub + s) * 32;
+                        StatsPayload sp2;
+                        uint64_t mn, mx;
+                        memcpy(&sp2.count, slot, 8);
+                        memcpy(&sp2.sum, slot + 8, 8);
+                        memcpy(&mn, slot + 16, 8);
+                        memcpy(&mx, slot + 24, 8);
+                        if (sp2.count) {
+                            sp2.min = u64_to_f64(mn);
+                            sp2.max = u64_to_f64(mx);
+                        }
+                        b.sub[s] = sp2;
+                    }
+                    r.buckets.push_back(std::move(b));
+                }
+            }
+            out.aggs.aggs.push_back(std::move(r));
+        }
+        out.has_aggs = true;
+    }
+
+    out.micros = uint64_t(std::chrono::duration_cast<std::chrono::microseconds>(
+                              std::chrono::steady_clock::now() - t0)
+                              .count());
+    return out;
+}
+
+}  // namespace qw
+
+// ============================================================== C ABI
+static thread_local std::string g_tls_error;
+
+static void set_err(qw_ctx* ctx, const std::string& msg) {
+    if (ctx) ctx->last_error = msg;
+    else g_tls_error = msg;
+}
+
+static void fill_buf(qw_buf* out, const std::string& s) {
+    out->data = (uint8_t*)malloc(s.size() ? s.size() : 1);
+    memcpy(out->data, s.data(), s.size());
+    out->len = s.size();
+}
+
+extern "C" {
+
+const char* qw_version(void) { return "quickwit_amd 0.1 gfx950"; }
+
+qw_ctx* qw_ctx_create(const char* config_json) {
+    auto* ctx = new qw_ctx();
+    try {
+        if (config_json && *config_json) {
+            mj::ValuePtr cfg = mj::parse(config_json);
+            if (const mj::Value* d = cfg->get("device")) ctx->device = int(d->as_i64());
+            if (const mj::Value* b = cfg->get("aggregation_bucket_limit"))
+                ctx->agg_bucket_limit = b->as_i64();
+        }
+        return ctx;
+    } catch (const std::exception& e) {
+        g_tls_error = e.what();
+        delete ctx;
+        return nullptr;
+    }
+}
+
+void qw_ctx_free(qw_ctx* ctx) {
+    if (!ctx) return;
+    if (ctx->device_ready) {
+        (void)hipSetDevice(ctx->device);
+        for (auto& kv : ctx->splits)
+            if (kv.second->d_image) (void)hipFree(kv.second->d_image);
+        if (ctx->ev_start) (void)hipEventDestroy(ctx->ev_start);
+        if (ctx->ev_stop) (void)hipEventDestroy(ctx->ev_stop);
+        if (ctx->stream) (void)hipStreamDestroy(ctx->stream);
+    }
+    delete ctx;
+}
+
+const char* qw_last_error(const qw_ctx* ctx) {
+    return ctx ? ctx->last_error.c_str() : g_tls_error.c_str();
+}
+
+int32_t qw_ctx_add_split(qw_ctx* ctx, const char* split_id, const uint8_t* data,
+                         size_t len) {
+    using namespace qw;
+    try {
+        ctx_ensure_device(ctx);
+    } catch (const std::exception& e) {
+        set_err(ctx, e.what());
+        return QW_ERR_NO_GPU;
+    }
+    try {
+        auto ds = std::make_unique<DeviceSplit>();
+        ds->host.assign(data, data + len);
+        ds->view.parse(ds->host.data(), len);
+        ds->len = len;
+        HIP_CHECK(hipSetDevice(ctx->device));
+        HIP_CHECK(hipMalloc(&ds->d_image, len + 64));  // +64: decode overread pad
+        HIP_CHECK(hipMemcpy(ds->d_image, data, len, hipMemcpyHostToDevice));
+        ctx->splits[split_id] = std::move(ds);
+        return QW_OK;
+    } catch (const std::exception& e) {
+        set_err(ctx, e.what());
+        return QW_ERR_BAD_SPLIT;
+    }
+}
+
+int32_t qw_ctx_remove_split(qw_ctx* ctx, const char* split_id) {
+    auto it = ctx->splits.find(split_id);
+    if (it == ctx->splits.end()) return QW_ERR_NOT_FOUND;
+    if (it->second->d_image) (void)hipFree(it->second->d_image);
+    ctx->splits.erase(it);
+    return QW_OK;
+}
+
+int32_t qw_leaf_search(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
+                       qw_buf* out) {
+    using namespace qw;
+    try {
+        ctx_ensure_device(ctx);
+    } catch (const std::exception& e) {
+        set_err(ctx, e.what());
+        return QW_ERR_NO_GPU;
+    }
+    try {
+        HIP_CHECK(hipSetDevice(ctx->device));
+        pb::LeafSearchRequest lreq = pb::LeafSearchRequest::decode(req_pb, req_len);
+        const pb::SearchRequest& req = lreq.search_request;
+        if (lreq.doc_mappers.empty()) throw std::runtime_error("missing doc_mapper");
+        Schema schema = Schema::parse(lreq.doc_mappers[0]);
+
+        std::vector<std::pair<const DeviceSplit*, std::string>> tasks;
+        for (auto& lr : lreq.leaf_requests)
+            for (auto& so : lr.split_offsets) {
+                auto it = ctx->splits.find(so.split_id);
+                if (it == ctx->splits.end()) {
+                    set_err(ctx, "unknown split: " + so.split_id);
+                    return QW_ERR_NOT_FOUND;
+                }
+                tasks.push_back({it->second.get(), so.split_id});
+            }
+
+        pb::LeafSearchResponse resp;
+        resp.num_attempted_splits = tasks.size();
+        IntermediateAggResults merged_aggs;
+        bool any_aggs = false;
+        std::vector<pb::PartialHit> all_hits;
+        pb::LeafResourceStats rstats;
+        rstats.search_pool_cpu_threads = 1;  // one host thread per GPU
+        uint64_t worst = 0;
+        for (auto& [ds, sid] : tasks) {
+            SplitResult r;
+            try {
+                r = search_split_gpu(ctx, *ds, req, schema);
+            } catch (const std::exception& e) {
+                // per-split failure is data, not an exception (leaf.rs:2143)
+                pb::SplitSearchError se;
+                se.error = e.what();
+                se.split_id = sid;
+                se.retryable_error = true;
+                resp.failed_splits.push_back(std::move(se));
+                continue;
+            }
+            resp.num_successful_splits++;
+            resp.num_hits += r.num_hits;
+            for (auto& h : r.hits) all_hits.push_back(std::move(h));
+            if (r.has_aggs) {
+                if (!any_aggs) {
+                    merged_aggs = std::move(r.aggs);
+                    any_aggs = true;
+                } else merged_aggs.merge(r.aggs);
+            }
+            pb::SplitResourceStats ss;
+            ss.split_num_docs = ds->view.num_docs;
+            ss.matched_num_docs = r.num_hits;
+            ss.cpu_search_microsecs = r.micros;  // GPU wall (cpu_search analog)
+            rstats.localexec_num_splits++;
+            rstats.localexec_num_docs += ss.split_num_docs;
+            rstats.split_resources_sum.split_num_docs += ss.split_num_docs;
+            rstats.split_resources_sum.matched_num_docs += ss.matched_num_docs;
+            rstats.split_resources_sum.cpu_search_microsecs += ss.cpu_search_microsecs;
+            if (ss.cpu_search_microsecs >= worst) {
+                worst = ss.cpu_search_microsecs;
+                rstats.split_resources_worst = ss;
+            }
+        }
+        int order1 = req.sort_fields.empty() ? 1 : req.sort_fields[0].sort_order;
+        int order2 = req.sort_fields.size() > 1 ? req.sort_fields[1].sort_order : 1;
+        auto cmp = [&](const pb::PartialHit& a, const pb::PartialHit& b) {
+            return hit_before(a, b, order1, order2);
+        };
+        size_t k = std::min<size_t>(req.max_hits + req.start_offset, all_hits.size());
+        std::partial_sort(all_hits.begin(), all_hits.begin() + k, all_hits.end(), cmp);
+        all_hits.resize(k);
+        resp.partial_hits = std::move(all_hits);
+        if (any_aggs) resp.intermediate_aggregation_result = merged_aggs.encode();
+        resp.resource_stats = rstats;
+        fill_buf(out, resp.encode());
+        return QW_OK;
+    } catch (const std::exception& e) {
+        set_err(ctx, e.what());
+        std::string msg = e.what();
+        if (msg.find("not supported") != std::string::npos ||
+            msg.find("not flattenable") != std::string::npos)
+            return QW_ERR_INVALID_QUERY;
+        return QW_ERR_INTERNAL;
+    }
+}
+
+void qw_buf_free(qw_buf* buf) {
+    if (!buf) return;
+    free(buf->data);
+    buf->data = nullptr;
+    buf->len = 0;
+}
+
+int32_t qw_merge_leaf_responses(const uint8_t* search_request_pb,
+                                size_t search_request_len, const uint8_t** response_pbs,
+                                const size_t* response_lens, size_t n,
+                                qw_buf* merged_out) {
+    using namespace qw;
+    try {
+        pb::SearchRequest req = pb::SearchRequest::decode(
+            pb::Reader(search_request_pb, search_request_len));
+        pb::LeafSearchResponse m;
+        IntermediateAggResults aggs;
+        bool any_aggs = false;
+        std::vector<pb::PartialHit> all_hits;
+        for (size_t i = 0; i < n; ++i) {
+            pb::LeafSearchResponse r =
+                pb::LeafSearchResponse::decode(response_pbs[i], response_lens[i]);
+            m.num_hits += r.num_hits;
+            m.num_attempted_splits += r.num_attempted_splits;
+            m.num_successful_splits += r.num_successful_splits;
+            for (auto& h : r.partial_hits) all_hits.push_back(std::move(h));
+            for (auto& f : r.failed_splits) m.failed_splits.push_back(std::move(f));
+            if (r.intermediate_aggregation_result) {
+                IntermediateAggResults ir = IntermediateAggResults::decode(
+                    (const uint8_t*)r.intermediate_aggregation_result->data(),
+                    r.intermediate_aggregation_result->size());
+                if (!any_aggs) {
+                    aggs = std::move(ir);
+                    any_aggs = true;
+                } else aggs.merge(ir);
+            }
+            if (r.resource_stats) {
+                if (!m.resource_stats) m.resource_stats = pb::LeafResourceStats{};
+                m.resource_stats->localexec_num_splits +=
+                    r.resource_stats->localexec_num_splits;
+                m.resource_stats->localexec_num_docs +=
+                    r.resource_stats->localexec_num_docs;
+                m.resource_stats->split_resources_sum.cpu_search_microsecs +=
+                    r.resource_stats->split_resources_sum.cpu_search_microsecs;
+                m.resource_stats->split_resources_sum.split_num_docs +=
+                    r.resource_stats->split_resources_sum.split_num_docs;
+                m.resource_stats->split_resources_sum.matched_num_docs +=
+                    r.resource_stats->split_resources_sum.matched_num_docs;
+            }
+        }
+        int order1 = req.sort_fields.empty() ? 1 : req.sort_fields[0].sort_order;
+        int order2 = req.sort_fields.size() > 1 ? req.sort_fields[1].sort_order : 1;
+        auto cmp = [&](const pb::PartialHit& a, const pb::PartialHit& b) {
+            return hit_before(a, b, order1, order2);
+        };
+        size_t k = std::min<size_t>(req.max_hits + req.start_offset, all_hits.size());
+        std::partial_sort(all_hits.begin(), all_hits.begin() + k, all_hits.end(), cmp);
+        all_hits.resize(k);
+        m.partial_hits = std::move(all_hits);
+        if (any_aggs) m.intermediate_aggregation_result = aggs.encode();
+        fill_buf(merged_out, m.encode());
+        return QW_OK;
+    } catch (const std::exception& e) {
+        set_err(nullptr, e.what());
+        return QW_ERR_INTERNAL;
+    }
+}
+
+int32_t qw_finalize_agg_to_json(const uint8_t* blob, size_t len,
+                                const char* agg_request_json, qw_buf* json_out) {
+    using namespace qw;
+    try {
+        IntermediateAggResults ir = IntermediateAggResults::decode(blob, len);
+        std::vector<AggDef> defs = parse_agg_request(agg_request_json);
+        fill_buf(json_out, finalize_aggs_json(ir, defs));
+        return QW_OK;
+    } catch (const std::exception& e) {
+        set_err(nullptr, e.what());
+        return QW_ERR_INTERNAL;
+    }
+}
+
+int32_t qw_kernel_stats(qw_ctx* ctx, const char* kernel_name, double* total_ms,
+                        uint64_t* launches) {
+    auto it = ctx->timers.find(kernel_name);
+    if (it == ctx->timers.end()) {
+        static const char* known[] = {"union_bm25", "range_filter", "column_agg",
+                                      "topk_select"};
+        bool ok = false;
+        for (const char* k : known) ok |= kernel_name == std::string(k);
+        if (!ok) return QW_ERR_NOT_FOUND;
+        *total_ms = 0;
+        *launches = 0;
+        return QW_OK;
+    }
+    *total_ms = it->second.total_ms;
+    *launches = it->second.launches;
+    return QW_OK;
+}
+
+void qw_kernel_stats_reset(qw_ctx* ctx) { ctx->timers.clear(); }
+
+int32_t qw_ctx_device_sync(qw_ctx* ctx) {
+    try {
+        ctx_ensure_device(ctx);
+        HIP_CHECK(hipDeviceSynchronize());
+        return QW_OK;
+    } catch (const std::exception& e) {
+        set_err(ctx, e.what());
+        return QW_ERR_NO_GPU;
+    }
+}
+
+}  // extern "C"

@@ -1,0 +1,135 @@
+"""C-ABI surface checks that need no GPU (the library must load and export
+every symbol include/quickwit_amd.h declares; the product path must fail
+LOUDLY without a GPU — QW_ERR_NO_GPU, never a CPU fallback), plus CPU-side
+parity of the ctx-less entry points (merge, agg finalize) against the oracle:
+qw_merge_leaf_responses must reproduce the oracle's cross-split merge
+(reference merge_fruits semantics, collector.rs:832-861)."""
+import ctypes
+import json
+import os
+import re
+import subprocess
+
+import pytest
+
+from quickwit_amd import proto, splitgen
+from quickwit_amd.api import OracleSearcher, make_leaf_request
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+LIB = os.path.join(REPO, "libquickwit_amd.so")
+
+
+@pytest.fixture(scope="module", autouse=True)
+def build_all():
+    import __graft_entry__
+    __graft_entry__.build()
+
+
+def header_symbols():
+    hdr = open(os.path.join(REPO, "include", "quickwit_amd.h")).read()
+    return sorted(set(re.findall(r"\b(qw_[a-z_0-9]+)\s*\(", hdr)) - {"qw_buf"})
+
+
+def test_exports_every_header_symbol():
+    lib = ctypes.CDLL(LIB)
+    missing = [s for s in header_symbols() if not hasattr(lib, s)]
+    assert not missing, f"symbols declared in quickwit_amd.h but not exported: {missing}"
+
+
+def test_version_string():
+    lib = ctypes.CDLL(LIB)
+    lib.qw_version.restype = ctypes.c_char_p
+    v = lib.qw_version().decode()
+    assert "gfx950" in v
+
+
+def test_no_gpu_fails_loudly():
+    """the product never falls back to CPU."""
+    import torch
+    if torch.cuda.is_available():
+        pytest.skip("GPU present; covered by gpu-marked tests")
+    lib = ctypes.CDLL(LIB)
+    lib.qw_ctx_create.restype = ctypes.c_void_p
+    lib.qw_ctx_create.argtypes = [ctypes.c_char_p]
+    ctx = lib.qw_ctx_create(b'{"device": 0}')
+    assert ctx
+    lib.qw_ctx_add_split.argtypes = [ctypes.c_void_p, ctypes.c_char_p,
+                                     ctypes.c_char_p, ctypes.c_size_t]
+    rc = lib.qw_ctx_add_split(ctx, b"s", b"x" * 128, 128)
+    assert rc == -5, f"expected QW_ERR_NO_GPU (-5), got {rc}"
+    lib.qw_ctx_free.argtypes = [ctypes.c_void_p]
+    lib.qw_ctx_free(ctx)
+
+
+class _Buf(ctypes.Structure):
+    _fields_ = [("data", ctypes.POINTER(ctypes.c_uint8)), ("len", ctypes.c_size_t)]
+
+
+def _merge(lib, sreq_pb, resp_pbs):
+    lib.qw_merge_leaf_responses.argtypes = [
+        ctypes.c_char_p, ctypes.c_size_t,
+        ctypes.POINTER(ctypes.c_char_p), ctypes.POINTER(ctypes.c_size_t),
+        ctypes.c_size_t, ctypes.POINTER(_Buf)]
+    arr = (ctypes.c_char_p * len(resp_pbs))(*resp_pbs)
+    lens = (ctypes.c_size_t * len(resp_pbs))(*[len(r) for r in resp_pbs])
+    buf = _Buf()
+    rc = lib.qw_merge_leaf_responses(sreq_pb, len(sreq_pb), arr, lens,
+                                     len(resp_pbs), ctypes.byref(buf))
+    assert rc == 0
+    out = ctypes.string_at(buf.data, buf.len)
+    lib.qw_buf_free.argtypes = [ctypes.POINTER(_Buf)]
+    lib.qw_buf_free(ctypes.byref(buf))
+    return proto.decode("LeafSearchResponse", out)
+
+
+def test_merge_matches_oracle_cross_split_merge():
+    """Merging per-split responses with the product's qw_merge_leaf_responses
+    must equal the oracle searching both splits in one call."""
+    lib = ctypes.CDLL(LIB)
+    schema = splitgen.HDFS_SCHEMA
+    datas = [splitgen.generate_split(i, 5_000, seed=7) for i in range(2)]
+    sids = [f"synthetic-7-{i:04d}" for i in range(2)]
+
+    both = OracleSearcher()
+    singles = [OracleSearcher() for _ in range(2)]
+    for i in range(2):
+        both.add_split(sids[i], datas[i])
+        singles[i].add_split(sids[i], datas[i])
+
+    query = {"type": "bool", "should": [
+        {"type": "term", "field": "body", "value": "w%05d" % i} for i in range(3)]}
+    aggs = {"per_tenant": {"terms": {"field": "tenant_name", "size": 5}},
+            "per_hour": {"date_histogram":
+                         {"field": "timestamp", "fixed_interval": "86400000ms"}}}
+    kw = dict(max_hits=7, aggregation=aggs,
+              sort_fields=[{"field_name": "_score", "sort_order": 1}])
+    req_both = make_leaf_request(query, schema,
+                                 [(sids[i], 5_000) for i in range(2)], **kw)
+    expected = both.leaf_search(req_both)
+
+    resp_pbs = []
+    for i in range(2):
+        r = make_leaf_request(query, schema, [(sids[i], 5_000)], **kw)
+        resp_pbs.append(singles[i].leaf_search_raw(proto.encode("LeafSearchRequest", r)))
+    sreq_pb = proto.encode("SearchRequest", req_both["search_request"])
+    merged = _merge(lib, sreq_pb, resp_pbs)
+
+    assert merged["num_hits"] == expected["num_hits"]
+    assert ([(h["split_id"], h["doc_id"]) for h in merged["partial_hits"]] ==
+            [(h["split_id"], h["doc_id"]) for h in expected["partial_hits"]])
+    # aggregation blobs merge to the same finalized JSON
+    fin_m = _finalize(lib, merged["intermediate_aggregation_result"], aggs)
+    fin_e = both.finalize_agg_json(expected["intermediate_aggregation_result"], aggs)
+    assert fin_m == fin_e
+
+
+def _finalize(lib, blob, aggs):
+    lib.qw_finalize_agg_to_json.argtypes = [
+        ctypes.c_char_p, ctypes.c_size_t, ctypes.c_char_p, ctypes.POINTER(_Buf)]
+    buf = _Buf()
+    rc = lib.qw_finalize_agg_to_json(blob, len(blob), json.dumps(aggs).encode(),
+                                     ctypes.byref(buf))
+    assert rc == 0
+    out = ctypes.string_at(buf.data, buf.len)
+    lib.qw_buf_free(ctypes.byref(buf))
+    return json.loads(out)
