@@ -20,19 +20,20 @@ class _Buf(ctypes.Structure):
     _fields_ = [("data", ctypes.POINTER(ctypes.c_uint8)), ("len", ctypes.c_size_t)]
 
 
-def _bind(lib, prefix):
-    getattr(lib, prefix + "create").restype = ctypes.c_void_p
-    getattr(lib, prefix + "add_split").argtypes = [
+def _bind(lib, syms):
+    """syms: dict role -> exported symbol name."""
+    getattr(lib, syms["create"]).restype = ctypes.c_void_p
+    getattr(lib, syms["add_split"]).argtypes = [
         ctypes.c_void_p, ctypes.c_char_p, ctypes.c_char_p, ctypes.c_size_t]
-    getattr(lib, prefix + "leaf_search").argtypes = [
+    getattr(lib, syms["leaf_search"]).argtypes = [
         ctypes.c_void_p, ctypes.c_char_p, ctypes.c_size_t, ctypes.POINTER(_Buf)]
-    getattr(lib, prefix + "last_error").restype = ctypes.c_char_p
-    getattr(lib, prefix + "last_error").argtypes = [ctypes.c_void_p]
-    getattr(lib, prefix + "buf_free").argtypes = [ctypes.POINTER(_Buf)]
+    getattr(lib, syms["last_error"]).restype = ctypes.c_char_p
+    getattr(lib, syms["last_error"]).argtypes = [ctypes.c_void_p]
+    getattr(lib, syms["buf_free"]).argtypes = [ctypes.POINTER(_Buf)]
 
 
 class _BaseSearcher:
-    _prefix = None
+    _syms = None
     _libname = None
 
     def __init__(self, create_arg=None):
@@ -41,33 +42,34 @@ class _BaseSearcher:
             raise FileNotFoundError(
                 f"{path} not built — run __graft_entry__.build() first")
         self._lib = ctypes.CDLL(path)
-        _bind(self._lib, self._prefix)
+        _bind(self._lib, self._syms)
+        create = getattr(self._lib, self._syms["create"])
         if create_arg is None:
-            self._ctx = getattr(self._lib, self._prefix + "create")()
+            self._ctx = create()
         else:
-            getattr(self._lib, self._prefix + "create").argtypes = [ctypes.c_char_p]
-            self._ctx = getattr(self._lib, self._prefix + "create")(create_arg)
+            create.argtypes = [ctypes.c_char_p]
+            self._ctx = create(create_arg)
         if not self._ctx:
             raise RuntimeError("ctx creation failed")
 
     def _err(self):
-        msg = getattr(self._lib, self._prefix + "last_error")(self._ctx)
+        msg = getattr(self._lib, self._syms["last_error"])(self._ctx)
         return msg.decode() if msg else "unknown error"
 
     def add_split(self, split_id: str, data: bytes):
-        rc = getattr(self._lib, self._prefix + "add_split")(
+        rc = getattr(self._lib, self._syms["add_split"])(
             self._ctx, split_id.encode(), data, len(data))
         if rc != 0:
             raise RuntimeError(f"add_split failed ({rc}): {self._err()}")
 
     def leaf_search_raw(self, req_pb: bytes) -> bytes:
         buf = _Buf()
-        rc = getattr(self._lib, self._prefix + "leaf_search")(
+        rc = getattr(self._lib, self._syms["leaf_search"])(
             self._ctx, req_pb, len(req_pb), ctypes.byref(buf))
         if rc != 0:
             raise RuntimeError(f"leaf_search failed ({rc}): {self._err()}")
         out = ctypes.string_at(buf.data, buf.len)
-        getattr(self._lib, self._prefix + "buf_free")(ctypes.byref(buf))
+        getattr(self._lib, self._syms["buf_free"])(ctypes.byref(buf))
         return out
 
     def leaf_search(self, leaf_req: dict) -> dict:
@@ -78,7 +80,9 @@ class _BaseSearcher:
 class OracleSearcher(_BaseSearcher):
     """CPU restatement (oracle) — tests/bench cpu_baseline only."""
 
-    _prefix = "qw_oracle_"
+    _syms = {"create": "qw_oracle_create", "add_split": "qw_oracle_add_split",
+             "leaf_search": "qw_oracle_leaf_search",
+             "last_error": "qw_oracle_last_error", "buf_free": "qw_oracle_buf_free"}
     _libname = "oracle/liboracle.so"
 
     def finalize_agg_json(self, blob: bytes, agg_request: dict) -> dict:
@@ -98,7 +102,9 @@ class GpuSearcher(_BaseSearcher):
     """The product path. Fails loudly when no MI355X is visible
     (QW_ERR_NO_GPU) — there is no CPU fallback."""
 
-    _prefix = "qw_"
+    _syms = {"create": "qw_ctx_create", "add_split": "qw_ctx_add_split",
+             "leaf_search": "qw_leaf_search", "last_error": "qw_last_error",
+             "buf_free": "qw_buf_free"}
     _libname = "libquickwit_amd.so"
 
     def __init__(self, device: int = 0, config: dict | None = None):

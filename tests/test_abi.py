@@ -133,3 +133,15 @@ def _finalize(lib, blob, aggs):
     out = ctypes.string_at(buf.data, buf.len)
     lib.qw_buf_free(ctypes.byref(buf))
     return json.loads(out)
+
+
+def test_gpu_searcher_binding_resolves():
+    """GpuSearcher's ctypes binding must resolve every product symbol even
+    without a GPU (ctx creation is lazy; add_split then fails with
+    QW_ERR_NO_GPU — covered above)."""
+    from quickwit_amd.api import GpuSearcher
+    import torch
+    s = GpuSearcher(device=0)  # binds all symbols; no device touch
+    if not torch.cuda.is_available():
+        with pytest.raises(RuntimeError, match="-5"):
+            s.add_split("s", b"x" * 128)
