@@ -286,3 +286,54 @@ def test_kernel_stats_populated(searchers):
     gpu.leaf_search(req)
     ms, n = gpu.kernel_stats("union_bm25")
     assert n == 1 and ms > 0
+
+
+# ------------------------------------------------- golden agg scenarios (GPU)
+def _es_query_to_ast(q):
+    if "match_all" in q:
+        return {"type": "match_all"}
+    if "bool" in q:
+        out = {"type": "bool"}
+        for clause in ("must", "must_not", "should", "filter"):
+            if clause in q["bool"]:
+                items = q["bool"][clause]
+                if isinstance(items, dict):
+                    items = [items]
+                out[clause] = [_es_query_to_ast(i) for i in items]
+        return out
+    if "exists" in q:
+        return {"type": "field_presence", "field": q["exists"]["field"]}
+    raise ValueError(f"unsupported es query: {q}")
+
+
+@pytest.mark.parametrize("case_name", [
+    "date_histogram_basic",
+    "date_histogram_extended_bounds",
+    "date_histogram_stats_subagg",
+    "date_histogram_stats_subagg_exists_filter",
+    "terms_full",
+    "histogram_interval50",
+])
+def test_agg_golden_scenarios_on_gpu(case_name):
+    """The reference's rest-api-tests aggregation scenarios (exact expected
+    JSON) through the HIP kernels — same suite that pins the oracle."""
+    with open(os.path.join(REPO, "tests", "golden", "aggregations.json")) as f:
+        g = json.load(f)
+    type_map = {"str_fast": "str"}
+    schema = {"timestamp_field": None, "fields": [
+        {"name": f["name"], "type": type_map.get(f["type"], f["type"]), "fast": True}
+        for f in g["schema"]]}
+    s = GpuSearcher(device=0)
+    splits = []
+    for i, docs in enumerate(g["splits"]):
+        w = splitgen.SplitWriter(schema, f"agg-split-{i}")
+        w.add_documents(docs)
+        s.add_split(f"agg-split-{i}", w.finalize())
+        splits.append((f"agg-split-{i}", len(docs)))
+    case = g["cases"][case_name]
+    aggs = case["request"]["aggs"]
+    ast = _es_query_to_ast(case["request"]["query"])
+    req = make_leaf_request(ast, schema, splits, max_hits=0, aggregation=aggs)
+    resp = s.leaf_search(req)
+    got = s.finalize_agg_json(resp["intermediate_aggregation_result"], aggs)
+    approx_json(got, case["expected"], case_name)
