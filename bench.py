@@ -96,7 +96,10 @@ def main():
     searcher = GpuSearcher(device=local_rank)
     searcher.add_split(sid, split_bytes)
 
-    terms = ["w%05d" % i for i in range(3)]  # highest-df Zipf terms (df≈10%)
+    # Zipf ranks 10-12: P(doc contains term) = 1-(1-p_r)^10 ≈ 10% each,
+    # matching SURVEY §8d config 2 ("chosen at df≈10% each"). The top ranks
+    # (w00000..) sit at df≈66%/doc — a much heavier union than the config.
+    terms = ["w%05d" % i for i in (9, 10, 11)]
     req = make_leaf_request(
         build_query(terms), splitgen.HDFS_SCHEMA, [(sid, args.docs)],
         max_hits=args.max_hits,
