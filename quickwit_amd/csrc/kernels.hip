@@ -70,9 +70,13 @@ struct TileLds {
     float score[TILE_DOCS];            // BM25 sums, or should-match counters
     uint32_t bits_acc[TILE_DOCS / 32]; // must intersection
     uint32_t bits_not[TILE_DOCS / 32]; // must_not union
+    uint32_t bits_m[TILE_DOCS / 32];   // final per-doc match (epilogue phase A)
+    uint32_t word_pref[TILE_DOCS / 32]; // exclusive match-rank prefix per word
     uint32_t agg_hist[AGG_LDS_BUCKETS];
     uint32_t agg_terms[AGG_LDS_BUCKETS];
+    uint32_t agg_matched[4];  // terms-agg docs-with-value, flushed per tile
     uint32_t wave_base[4];
+    uint32_t cand_base;
 };
 
 // decode every block of `t` overlapping the tile; accumulate into score[] /
@@ -210,11 +214,16 @@ extern "C" __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile(QueryDev 
                 lds.bits_acc[i] = 0;
                 lds.bits_not[i] = 0;
             }
-        if (do_aggs && q.n_aggs)
+        if (collect)
+            for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS)
+                lds.bits_m[i] = 0;
+        if (do_aggs && q.n_aggs) {
             for (uint32_t i = threadIdx.x; i < AGG_LDS_BUCKETS; i += TILE_THREADS) {
                 lds.agg_hist[i] = 0;
                 lds.agg_terms[i] = 0;
             }
+            if (threadIdx.x < 4) lds.agg_matched[threadIdx.x] = 0;
+        }
         __syncthreads();
 
         // ---- should terms: score / count union
@@ -292,7 +301,11 @@ extern "C" __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile(QueryDev 
                             else
                                 atomicAdd((unsigned long long*)(q.results + a.counts_out) + o,
                                           1ull);
-                            atomicAdd((unsigned long long*)(q.results + a.matched_out), 1ull);
+                            if (ai < 4)  // per-tile LDS count, flushed below
+                                atomicAdd(&lds.agg_matched[ai], 1u);
+                            else
+                                atomicAdd((unsigned long long*)(q.results + a.matched_out),
+                                          1ull);
                         } else {
                             double v = agg_value(q, a.values_off, a.value_width,
                                                  a.value_is_i64, d);
@@ -324,35 +337,18 @@ extern "C" __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile(QueryDev 
                         }
                     }
             }
-            // candidate append (wave-aggregated). combined exact key:
-            // high 32 = sortable score (flipped for asc), low 32 = doc
-            // tie-break in the order1 direction (sorting.md:14-17)
+            // phase A of collection: record matches in an LDS bitset (no
+            // global traffic; one ballot + direct word store per wave —
+            // Guideline 12: never a per-wave global atomic in the hot loop)
             if (collect) {
-                bool write = m;
-                uint64_t key = 0;
-                if (write) {
-                    uint32_t kh = q.scoring ? f32_sortable(sc) : 0u;
-                    if (q.sort_asc) kh = ~kh;
-                    uint32_t kl = q.sort_asc ? ~d : d;
-                    key = (uint64_t(kh) << 32) | kl;
-                }
-                uint64_t mask = __ballot(write);
-                uint32_t nw = __popcll(mask);
-                if (nw) {
-                    uint32_t base;
-                    if (lane_id() == __ffsll((unsigned long long)mask) - 1)
-                        base = atomicAdd(cand_count, nw);
-                    base = __shfl(base, __ffsll((unsigned long long)mask) - 1, 64);
-                    if (write) {
-                        uint32_t off =
-                            __popcll(mask & ((1ull << lane_id()) - 1ull));
-                        if (base + off < q.cand_cap) cand[base + off] = key;
-                    }
-                }
+                uint64_t mb = __ballot(m);
+                uint32_t wbase = (li - lane_id()) >> 5;
+                if (lane_id() == 0) lds.bits_m[wbase] = uint32_t(mb);
+                else if (lane_id() == 32) lds.bits_m[wbase + 1] = uint32_t(mb >> 32);
             }
         }
-        // ---- flush LDS agg arrays
-        if (do_aggs)
+        // ---- flush LDS agg arrays + matched counters (per-tile, once)
+        if (do_aggs) {
             for (uint32_t ai = 0; ai < q.n_aggs && ai < 2; ++ai) {
                 const AggDev& a = aggs[ai];
                 if (a.n_buckets > AGG_LDS_BUCKETS) continue;
@@ -364,8 +360,50 @@ extern "C" __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile(QueryDev 
                         atomicAdd((unsigned long long*)(q.results + a.counts_out) + i,
                                   (unsigned long long)src[i]);
             }
-        // ---- per-tile count
-        if (do_count) {
+            __syncthreads();
+            if (threadIdx.x < q.n_aggs && threadIdx.x < 4) {
+                const AggDev& a = aggs[threadIdx.x];
+                if (a.kind == AGGD_TERMS && lds.agg_matched[threadIdx.x])
+                    atomicAdd((unsigned long long*)(q.results + a.matched_out),
+                              (unsigned long long)lds.agg_matched[threadIdx.x]);
+            }
+        }
+        // ---- per-tile count + phase B of collection (one global atomic per
+        // TILE reserves contiguous cand space; ranks from an LDS bitset
+        // prefix-scan; writes come out doc-ordered => coalesced)
+        if (collect) {
+            __syncthreads();  // bits_m/score final
+            uint32_t w = threadIdx.x;  // one 32-doc word per thread (256 words)
+            uint32_t cnt = __popc(lds.bits_m[w]);
+            uint32_t incl = wave_incl_scan_u32(cnt);
+            uint32_t wv = threadIdx.x >> 6;
+            if (lane_id() == 63) lds.wave_base[wv] = incl;
+            __syncthreads();
+            uint32_t wave_off = 0;
+            for (uint32_t i = 0; i < wv; ++i) wave_off += lds.wave_base[i];
+            lds.word_pref[w] = wave_off + incl - cnt;
+            uint32_t tile_total = lds.wave_base[0] + lds.wave_base[1] +
+                                  lds.wave_base[2] + lds.wave_base[3];
+            if (threadIdx.x == 0) {
+                lds.cand_base = atomicAdd(cand_count, tile_total);
+                if (do_count) tile_counts[tile] = tile_total;
+            }
+            __syncthreads();
+            uint64_t base = lds.cand_base;
+            for (uint32_t d = tile_lo + threadIdx.x; d < tile_hi; d += TILE_THREADS) {
+                uint32_t li = d - tile_lo;
+                uint32_t bits = lds.bits_m[li >> 5];
+                if (!((bits >> (li & 31)) & 1)) continue;
+                uint64_t rank = lds.word_pref[li >> 5] +
+                                __popc(bits & ((1u << (li & 31)) - 1u));
+                float sc = q.scoring ? lds.score[li] : 0.f;
+                uint32_t kh = q.scoring ? f32_sortable(sc) : 0u;
+                if (q.sort_asc) kh = ~kh;
+                uint32_t kl = q.sort_asc ? ~d : d;
+                if (base + rank < q.cand_cap)
+                    cand[base + rank] = (uint64_t(kh) << 32) | kl;
+            }
+        } else if (do_count) {
             __syncthreads();
             uint32_t* wb = lds.wave_base;
             #pragma unroll
