@@ -752,6 +752,8 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                 prefix = (prefix << 12) | uint64_t(uint32_t(b));
                 prefix_bits += 12;
             }
+            // #keys >= floor_key is known EXACTLY from the histogram walk, so
+            // no round trip for the compact count: one async D2H of survivors
             ctx->d_survivors.ensure(survivors * 8 + 8);
             HIP_CHECK(hipMemsetAsync(d_scount, 0, 4, ctx->stream));
             uint32_t cgrid = std::min<uint32_t>(2048, (cand_n + 255) / 256);
@@ -759,14 +761,10 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                                d_cand, cand_n, floor_key, ~0ull,
                                (uint64_t*)ctx->d_survivors.p, d_scount,
                                uint32_t(survivors));
-            uint32_t sn = 0;
-            HIP_CHECK(hipMemcpyAsync(&sn, d_scount, 4, hipMemcpyDeviceToHost,
+            top_keys.resize(survivors);
+            HIP_CHECK(hipMemcpyAsync(top_keys.data(), ctx->d_survivors.p,
+                                     survivors * 8, hipMemcpyDeviceToHost,
                                      ctx->stream));
-            HIP_CHECK(hipStreamSynchronize(ctx->stream));
-            sn = std::min<uint32_t>(sn, uint32_t(survivors));
-            top_keys.resize(sn);
-            HIP_CHECK(hipMemcpy(top_keys.data(), ctx->d_survivors.p, size_t(sn) * 8,
-                                hipMemcpyDeviceToHost));
             HIP_CHECK(hipEventRecord(ctx->ev_stop, ctx->stream));
             HIP_CHECK(hipStreamSynchronize(ctx->stream));
             float tms = 0;
