@@ -1,62 +1,132 @@
 #!/usr/bin/env python3
 """Flagship benchmark: Quickwit leaf-search hot path on MI355X.
 
-Workload (BASELINE.json configs[1], the single-GPU configuration the metric
-is quoted on): 3-term BM25 disjunction (OR) over 10M synthetic log docs
-(hdfs-logs schema, seeded generator — SURVEY.md §8d), 1 split per GPU,
-max_hits=10, sorted by _score desc. A "step" is one full leaf_search call
-over the split batch, inputs already resident in HBM (the reference's warm
-state that cpu_search_microsecs times, leaf.rs:905-946) + the cross-rank
-top-K merge when N>1.
+Default workload (BASELINE.json configs[1], the single-GPU configuration the
+metric is quoted on): 3-term BM25 disjunction (OR) over 10M synthetic log
+docs (hdfs-logs schema, seeded generator — SURVEY.md §8d config 2), 1 split
+per GPU, max_hits=10, sorted by _score desc. A "step" is one full
+leaf_search call over the split batch, inputs already resident in HBM (the
+reference's warm state that cpu_search_microsecs times, leaf.rs:905-946) +
+the cross-rank top-K merge when N>1.
 
 Contract: python bench.py --gpus N --steps K --warmup W
   N>1 is launched by the driver via torch.distributed.run, one rank per GPU
   over RCCL; splits shard one-batch-per-GPU ("scaling": "weak");
   the only exchange is the response allgather + rank-0 merge.
 Rank 0 prints ONE JSON line with metric/value plus:
-  roofline: dominant kernel (union_bm25) algorithmic-bytes/launch over its
-    HIP-event launch time, vs 8 TB/s HBM3E peak (MI355X_MICROARCH.md).
-    Algorithmic bytes = posting payload+skip bytes of the 3 query terms
-    + 1B fieldnorm + 8B candidate record per union hit + per-tile metadata
-     — counted from the generated index, not from DRAM traffic.
-  cpu_baseline: the OpenMP oracle (reference restatement, kind "port") timed
-    on the same split on this box's host cores (bounded sample).
+  roofline: dominant kernel algorithmic-bytes/launch over its HIP-event
+    launch time, vs 8 TB/s HBM3E peak (MI355X_MICROARCH.md). Algorithmic
+    bytes are counted from the generated index, not from DRAM
+    traffic. traffic stays null here; PMC evidence lives under profiles/.
+  cpu_baseline: the oracle (reference restatement, kind "port") timed on this
+    box's host cores on a bounded sample (--cpu-baseline-docs).
+
+Extra workloads (NOT the default the driver runs):
+  --workload range  bool must severity_text:INFO + u64 tenant_id range
+  --workload agg    date_histogram(1h) + terms(tenant) under match_all
 """
 import argparse
 import json
 import os
 import sys
+import tempfile
 import time
 
 REPO = os.path.dirname(os.path.abspath(__file__))
 sys.path.insert(0, REPO)
 
 HBM_PEAK_GBS = 8000.0  # MI355X HBM3E spec peak (MI355X_MICROARCH.md)
+TILE_DOCS = 8192
+
+BM25_TERMS = ["w%05d" % i for i in (9, 10, 11)]
+# Zipf ranks 10-12: P(doc contains term) = 1-(1-p_r)^10 ≈ 10% each, matching
+# SURVEY §8d config 2 ("chosen at df≈10% each"); the top ranks sit at ≈66%.
+
+AGGS = {"per_hour": {"date_histogram": {"field": "timestamp",
+                                        "fixed_interval": "3600000ms"}},
+        "per_tenant": {"terms": {"field": "tenant_name", "size": 10}}}
 
 
-def build_query(terms):
-    return {"type": "bool", "should": [
-        {"type": "term", "field": "body", "value": t} for t in terms]}
+def _posting_bytes(sp, field, term, n_tiles):
+    posting_off = sp._sec(field, "posting_off", "<u8")
+    n_blocks = sp._sec(field, "n_blocks", "<u4")
+    tid = sp.term_id(field, term)
+    if tid is None:
+        return 0
+    return (int(posting_off[tid + 1] - posting_off[tid]) +
+            int(n_blocks[tid]) * 16 + 2 * 4 * n_tiles)
 
 
-def algo_bytes_per_launch(split_bytes, terms, num_union, n_tiles):
-    """Algorithmic HBM bytes one k_leaf_tile launch must move."""
-    from quickwit_amd import splitread
-    sp = splitread.Split(split_bytes)
-    f = sp.fields["body"]
-    posting_off = sp._sec("body", "posting_off", "<u8")
-    n_blocks = sp._sec("body", "n_blocks", "<u4")
-    total = 0
-    for t in terms:
-        tid = sp.term_id("body", t)
-        if tid is None:
-            continue
-        total += int(posting_off[tid + 1] - posting_off[tid])  # packed payload
-        total += int(n_blocks[tid]) * 16                       # skip entries
-        total += 2 * 4 * n_tiles                               # block ranges
-    total += num_union * (1 + 8)  # fieldnorm gather + candidate record write
-    total += n_tiles * 4          # tile counts
-    return total
+def make_workload(name, docs, max_hits):
+    """-> dict(query, sort, aggregation, max_hits, kernel, algo_bytes(fn),
+    label, query_str)."""
+    if name == "bm25":
+        return {
+            "query": {"type": "bool", "should": [
+                {"type": "term", "field": "body", "value": t}
+                for t in BM25_TERMS]},
+            "sort": [{"field_name": "_score", "sort_order": 1}],
+            "aggregation": None, "max_hits": max_hits,
+            "kernel": "union_bm25",
+            "label": f"3term_bm25_or_{docs}",
+            "query_str": " OR ".join(f"body:{t}" for t in BM25_TERMS),
+            # postings+skip+ranges of 3 terms + 1B fieldnorm + 8B candidate
+            # record per union hit + tile counts
+            "algo_bytes": lambda sp, nh, nt: (
+                sum(_posting_bytes(sp, "body", t, nt) for t in BM25_TERMS)
+                + nh * (1 + 8) + nt * 4),
+        }
+    if name == "range":
+        q = {"type": "bool",
+             "must": [{"type": "term", "field": "severity_text",
+                       "value": "INFO"}],
+             "filter": [{"type": "range", "field": "tenant_id",
+                         "lower_bound": {"included": 100},
+                         "upper_bound": {"excluded": 300}}]}
+        return {
+            "query": q, "sort": None, "aggregation": None,
+            "max_hits": max_hits, "kernel": "union_bm25",
+            "label": f"must_info_tenant_range_{docs}",
+            "query_str": "severity_text:INFO AND tenant_id:[100..300)",
+            # must postings + 8B u64 column per candidate doc (pred evaluated
+            # on must-matched docs only... kernel reads col per matching doc;
+            # count full col: preds are gather-per-doc -> matched*8) + writes
+            "algo_bytes": lambda sp, nh, nt: (
+                _posting_bytes(sp, "severity_text", "INFO", nt)
+                + int(0.40 * nt * TILE_DOCS) * 8  # col gather on INFO docs
+                + nh * 8 + nt * 4),
+        }
+    if name == "agg":
+        return {
+            "query": {"type": "match_all"}, "sort": None,
+            "aggregation": AGGS, "max_hits": 0,
+            "kernel": "column_agg",
+            "label": f"datehisto_terms_agg_{docs}",
+            "query_str": "* with per_hour date_histogram + per_tenant terms",
+            # 8B ts + 2B tenant ord per doc + per-tile LDS flush
+            # ((720+1000) u64 slots per tile)
+            "algo_bytes": lambda sp, nh, nt: (
+                nt * TILE_DOCS * (8 + 2) + nt * (720 + 1000) * 8),
+        }
+    raise SystemExit(f"unknown workload {name}")
+
+
+def cached_split(rank, docs, seed=42):
+    from quickwit_amd import splitgen
+    path = os.path.join(tempfile.gettempdir(),
+                        f"qwa1-{seed}-{rank}-{docs}.bin")
+    if os.path.exists(path) and os.path.getsize(path) > 0:
+        with open(path, "rb") as f:
+            return f.read()
+    data = splitgen.generate_split(rank, docs, seed=seed)
+    tmp = path + ".tmp"
+    try:
+        with open(tmp, "wb") as f:
+            f.write(data)
+        os.rename(tmp, path)
+    except OSError:
+        pass
+    return data
 
 
 def main():
@@ -67,7 +137,11 @@ def main():
     ap.add_argument("--docs", type=int, default=10_000_000,
                     help="docs per split (one split per GPU)")
     ap.add_argument("--max-hits", type=int, default=10)
+    ap.add_argument("--workload", default="bm25",
+                    choices=["bm25", "range", "agg"])
     ap.add_argument("--cpu-baseline-steps", type=int, default=2)
+    ap.add_argument("--cpu-baseline-docs", type=int, default=0,
+                    help="oracle sample size (0 = min(docs, 10M))")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -84,26 +158,24 @@ def main():
         if torch.cuda.is_available():
             torch.cuda.set_device(local_rank)
 
-    from quickwit_amd import proto, splitgen
+    from quickwit_amd import proto, splitgen, splitread
     from quickwit_amd.api import GpuSearcher, OracleSearcher, make_leaf_request
     from quickwit_amd.merge import merge_leaf_responses
 
+    wl = make_workload(args.workload, args.docs, args.max_hits)
+
     t_gen = time.perf_counter()
-    split_bytes = splitgen.generate_split(rank, args.docs, seed=42)
+    split_bytes = cached_split(rank, args.docs)
     sid = f"synthetic-42-{rank:04d}"
     gen_s = time.perf_counter() - t_gen
 
     searcher = GpuSearcher(device=local_rank)
     searcher.add_split(sid, split_bytes)
 
-    # Zipf ranks 10-12: P(doc contains term) = 1-(1-p_r)^10 ≈ 10% each,
-    # matching SURVEY §8d config 2 ("chosen at df≈10% each"). The top ranks
-    # (w00000..) sit at df≈66%/doc — a much heavier union than the config.
-    terms = ["w%05d" % i for i in (9, 10, 11)]
     req = make_leaf_request(
-        build_query(terms), splitgen.HDFS_SCHEMA, [(sid, args.docs)],
-        max_hits=args.max_hits,
-        sort_fields=[{"field_name": "_score", "sort_order": 1}])
+        wl["query"], splitgen.HDFS_SCHEMA, [(sid, args.docs)],
+        max_hits=wl["max_hits"], sort_fields=wl["sort"],
+        aggregation=wl["aggregation"])
     req_pb = proto.encode("LeafSearchRequest", req)
     sreq_pb = proto.encode("SearchRequest", req["search_request"])
 
@@ -143,45 +215,55 @@ def main():
         return
 
     resp = proto.decode("LeafSearchResponse", last)
-    num_union = resp["num_hits"] // world if world > 1 else resp["num_hits"]
+    num_hits = resp.get("num_hits", 0)
+    per_split_hits = num_hits // world if world > 1 else num_hits
     total_docs = args.docs * world
     value = args.steps * total_docs / elapsed
     ms_per_step = elapsed / args.steps * 1e3
     p50_ms = sorted(step_times)[len(step_times) // 2] * 1e3
 
     # roofline of the dominant kernel
-    kms, launches = searcher.kernel_stats("union_bm25")
+    kms, launches = searcher.kernel_stats(wl["kernel"])
     roofline = None
     if launches:
         ms_per_launch = kms / launches
-        n_tiles = (args.docs + 8192 - 1) // 8192
-        ab = algo_bytes_per_launch(split_bytes, terms, num_union, n_tiles)
+        n_tiles = (args.docs + TILE_DOCS - 1) // TILE_DOCS
+        sp = splitread.Split(split_bytes)
+        ab = int(wl["algo_bytes"](sp, per_split_hits, n_tiles))
         achieved = ab / (ms_per_launch / 1e3) / 1e9
         roofline = {"bound": "hbm", "achieved": round(achieved, 1),
                     "peak": HBM_PEAK_GBS, "unit": "GB/s",
                     "frac": round(achieved / HBM_PEAK_GBS, 4),
-                    "traffic": None,  # filled from rocprofv3 --pmc (profiles/)
-                    "kernel": "union_bm25",
+                    "traffic": None,  # PMC evidence committed under profiles/
+                    "kernel": wl["kernel"],
                     "algo_bytes_per_launch": ab,
                     "ms_per_launch": round(ms_per_launch, 4)}
 
-    # CPU baseline: the oracle restatement on this box's host cores (rank 0,
-    # N=1 only; bounded sample). OpenMP parallelism is across splits; with one
-    # split the per-split closure is single-threaded like the reference's.
+    # CPU baseline: the oracle restatement (kind "port") on this box's host
+    # cores; bounded sample so the default run finishes in minutes. OpenMP
+    # parallelism is across splits; one split => single-threaded per-split
+    # closure exactly like the reference's rayon model.
     cpu_baseline = None
     if world == 1 and args.cpu_baseline_steps > 0:
+        bdocs = args.cpu_baseline_docs or min(args.docs, 10_000_000)
+        bdata = split_bytes if bdocs == args.docs else cached_split(rank, bdocs)
+        breq = make_leaf_request(
+            wl["query"], splitgen.HDFS_SCHEMA, [(f"synthetic-42-{rank:04d}", bdocs)],
+            max_hits=wl["max_hits"], sort_fields=wl["sort"],
+            aggregation=wl["aggregation"])
+        breq_pb = proto.encode("LeafSearchRequest", breq)
         cpu = OracleSearcher()
-        cpu.add_split(sid, split_bytes)
-        cpu.leaf_search_raw(req_pb)  # warm
+        cpu.add_split(f"synthetic-42-{rank:04d}", bdata)
+        cpu.leaf_search_raw(breq_pb)  # warm
         tc = time.perf_counter()
         for _ in range(args.cpu_baseline_steps):
-            cpu.leaf_search_raw(req_pb)
+            cpu.leaf_search_raw(breq_pb)
         tcpu = (time.perf_counter() - tc) / args.cpu_baseline_steps
         cpu_baseline = {
-            "value": round(args.docs / tcpu, 1), "unit": "docs/s", "cores": 1,
+            "value": round(bdocs / tcpu, 1), "unit": "docs/s", "cores": 1,
             "kind": "port",
-            "sample": f"{args.cpu_baseline_steps} leaf_search calls over the same "
-                      f"{args.docs}-doc split ({tcpu:.2f}s each)"}
+            "sample": f"{args.cpu_baseline_steps} leaf_search calls over a "
+                      f"{bdocs}-doc split ({tcpu:.2f}s each)"}
 
     out = {
         "metric": "leaf_search_docs_per_sec",
@@ -198,13 +280,12 @@ def main():
         "dtype": "u32+f32",
         "data": "synthetic",
         "config": {
-            "workload": "3term_bm25_or_10M" if args.docs == 10_000_000
-                        else f"3term_bm25_or_{args.docs}",
-            "query": " OR ".join(f"body:{t}" for t in terms),
+            "workload": wl["label"],
+            "query": wl["query_str"],
             "docs_per_gpu": args.docs,
             "splits_per_gpu": 1,
-            "max_hits": args.max_hits,
-            "num_union_hits_per_split": num_union,
+            "max_hits": wl["max_hits"],
+            "num_hits_per_split": per_split_hits,
             "parallelism": f"split-dp{world}",
             "gen_seconds": round(gen_s, 1),
         },

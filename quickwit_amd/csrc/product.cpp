@@ -730,7 +730,9 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     break;
                 }
                 HIP_CHECK(hipMemsetAsync(d_hist, 0, TOPK_BINS * 4, ctx->stream));
-                uint32_t hgrid = std::min<uint32_t>(2048, (cand_n + 255) / 256);
+                // grid small enough that the per-WG 4096-bin LDS flush (global
+                // atomics ∝ grid) stays cheap, big enough to fill the chip
+                uint32_t hgrid = std::min<uint32_t>(512, (cand_n + 4095) / 4096);
                 hipLaunchKernelGGL(k_cand_hist, dim3(hgrid), dim3(256), 0, ctx->stream,
                                    d_cand, cand_n, prefix, prefix_bits, d_hist);
                 HIP_CHECK(hipMemcpyAsync(hist.data(), d_hist, TOPK_BINS * 4,
