@@ -1,0 +1,156 @@
+"""Edge-case parity, reference-style (empty and ragged inputs,
+nulls, boundary sizes — the cases quickwit's own tests pin in tests.rs and the
+rest-api scenarios). GPU path vs oracle on the same hand-built splits."""
+import math
+import os
+
+import pytest
+
+from quickwit_amd import splitgen
+from quickwit_amd.api import GpuSearcher, OracleSearcher, make_leaf_request
+
+pytestmark = pytest.mark.gpu
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+SCHEMA = splitgen.HDFS_SCHEMA
+
+
+@pytest.fixture(scope="module", autouse=True)
+def build_all():
+    import __graft_entry__
+    __graft_entry__.build()
+
+
+def hid(h):
+    return (h.get("split_id", ""), h.get("doc_id", 0))
+
+
+def pair(data, sid, ndocs):
+    gpu, cpu = GpuSearcher(device=0), OracleSearcher()
+    gpu.add_split(sid, data)
+    cpu.add_split(sid, data)
+
+    def run(query, **kw):
+        req = make_leaf_request(query, SCHEMA, [(sid, ndocs)], **kw)
+        return gpu.leaf_search(req), cpu.leaf_search(req)
+    return run
+
+
+RAGGED_DOCS = [
+    {"timestamp": 1700000000, "tenant_id": 7, "severity_text": "INFO",
+     "body": ""},                                     # empty body (norm id 0)
+    {"timestamp": 1700000001, "severity_text": "ERROR",
+     "body": "hello world"},                          # missing tenant_id
+    {"severity_text": "ERROR", "body": "hello"},      # missing timestamp
+    {"timestamp": 1700000500, "tenant_id": 9, "severity_text": "WARN",
+     "body": "hello hello hello " + " ".join(f"tok{i}" for i in range(60))},
+]
+
+
+@pytest.fixture(scope="module")
+def ragged():
+    w = splitgen.SplitWriter(SCHEMA, "ragged")
+    w.add_documents(RAGGED_DOCS)
+    return pair(w.finalize(), "ragged", len(RAGGED_DOCS))
+
+
+def same_hits(got, exp):
+    assert got.get("num_hits", 0) == exp.get("num_hits", 0)
+    assert [hid(h) for h in got.get("partial_hits", [])] == \
+           [hid(h) for h in exp.get("partial_hits", [])]
+
+
+def test_empty_split():
+    w = splitgen.SplitWriter(SCHEMA, "empty")
+    w.add_documents([])
+    run = pair(w.finalize(), "empty", 0)
+    got, exp = run({"type": "match_all"}, max_hits=5)
+    same_hits(got, exp)
+    assert got.get("num_hits", 0) == 0
+
+
+def test_ragged_term_scored(ragged):
+    got, exp = ragged({"type": "term", "field": "body", "value": "hello"},
+                      max_hits=10,
+                      sort_fields=[{"field_name": "_score", "sort_order": 1}])
+    assert got.get("num_hits") == exp.get("num_hits") == 3
+    for gh, eh in zip(got["partial_hits"], exp["partial_hits"]):
+        assert hid(gh) == hid(eh)
+        gs = gh["sort_value"]["f64"]
+        es = eh["sort_value"]["f64"]
+        assert math.isclose(gs, es, rel_tol=1e-5), (gs, es)
+
+
+def test_missing_fast_field_range(ragged):
+    # doc 1 has no tenant_id -> excluded from the range (null bitmap path)
+    got, exp = ragged({"type": "range", "field": "tenant_id",
+                       "lower_bound": {"included": 0},
+                       "upper_bound": {"included": 100}}, max_hits=10)
+    same_hits(got, exp)
+    assert got.get("num_hits") == 2
+
+
+def test_field_presence(ragged):
+    got, exp = ragged({"type": "field_presence", "field": "timestamp"},
+                      max_hits=10)
+    same_hits(got, exp)
+    assert got.get("num_hits") == 3
+
+
+def test_must_not_only_matches_nothing(ragged):
+    got, exp = ragged({"type": "bool", "must_not": [
+        {"type": "term", "field": "body", "value": "hello"}]}, max_hits=10)
+    same_hits(got, exp)
+    assert got.get("num_hits", 0) == 0
+
+
+def test_msm_exceeds_clause_count(ragged):
+    got, exp = ragged({"type": "bool", "should": [
+        {"type": "term", "field": "body", "value": "hello"},
+        {"type": "term", "field": "body", "value": "world"}],
+        "minimum_should_match": 3}, max_hits=10)
+    same_hits(got, exp)
+    assert got.get("num_hits", 0) == 0
+
+
+def test_max_hits_exceeds_num_docs(ragged):
+    got, exp = ragged({"type": "match_all"}, max_hits=1000)
+    same_hits(got, exp)
+    assert len(got["partial_hits"]) == len(RAGGED_DOCS)
+
+
+def test_start_offset_beyond_hits(ragged):
+    got, exp = ragged({"type": "term", "field": "body", "value": "hello"},
+                      max_hits=10, start_offset=50)
+    assert got.get("num_hits") == exp.get("num_hits") == 3
+    # leaf returns start_offset+max_hits best; both sides identical
+    assert [hid(h) for h in got.get("partial_hits", [])] == \
+           [hid(h) for h in exp.get("partial_hits", [])]
+
+
+def test_match_none_with_aggregation(ragged):
+    aggs = {"h": {"date_histogram": {"field": "timestamp",
+                                     "fixed_interval": "86400000ms"}}}
+    got, exp = ragged({"type": "match_none"}, max_hits=0, aggregation=aggs)
+    assert got.get("num_hits", 0) == exp.get("num_hits", 0) == 0
+    assert ("intermediate_aggregation_result" in got) == \
+           ("intermediate_aggregation_result" in exp)
+
+
+def test_timestamp_window_excludes_missing(ragged):
+    # doc 2 has no timestamp; the [start,end) filter must drop it
+    got, exp = ragged({"type": "match_all"}, max_hits=10,
+                      start_timestamp=1700000000, end_timestamp=1700001000)
+    same_hits(got, exp)
+    assert got.get("num_hits") == 3
+
+
+def test_unknown_field_term(ragged):
+    # per-split failures are data, not exceptions (leaf.rs:2143-2147):
+    # both sides report the bad split inside failed_splits, 0 hits
+    got, exp = ragged({"type": "term", "field": "nosuch", "value": "x"},
+                      max_hits=5)
+    assert got.get("num_hits", 0) == 0 and exp.get("num_hits", 0) == 0
+    assert len(got.get("failed_splits", [])) == 1
+    assert len(exp.get("failed_splits", [])) == 1
+    assert "unknown field" in got["failed_splits"][0]["error"]
