@@ -58,6 +58,14 @@ struct AggDev {
     int64_t base_index;     // histo: floor((min-offset)/interval) at bucket 0
     double interval;        // histo (ms for date_histogram)
     double offset;          // histo
+    // exact integer fast path (set by plan_aggs when interval/offset are
+    // integral and the column range makes floor-div provably bit-equal to
+    // the double path — see histo_bucket in kernels.hip)
+    uint32_t int_fast;
+    uint32_t lds_slot;      // 0 = agg_hist, 1 = agg_terms, 0xFF = global atomics
+    int64_t i_interval;
+    int64_t i_offset;
+    double inv_interval;    // 1.0 / i_interval
     uint64_t values_off;    // column byte offset
     uint64_t nulls_off;
     uint32_t value_width;   // 8, or ord width for terms

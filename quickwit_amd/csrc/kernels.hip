@@ -15,9 +15,17 @@
 //     tantivy aggregation) — coalesced column gathers + LDS histograms.
 //
 // Execution geometry: one workgroup (256 threads = 4 wave64) owns a tile of
-// TILE_DOCS=8192 consecutive doc ids with an LDS f32 accumulator; posting
-// blocks overlapping the tile are decoded wave-per-block with funnel-shift
-// bit extraction and a 64-lane shuffle prefix-sum for the doc-id deltas.
+// TILE_DOCS=8192 consecutive doc ids; posting blocks overlapping the tile are
+// decoded wave-per-block with funnel-shift bit extraction and a 64-lane
+// shuffle prefix-sum for the doc-id deltas.
+//
+// The main kernel is TEMPLATE-SPECIALIZED over the query shape
+// (NS = score/should-count array, NB = must/must_not bitsets, NA = LDS agg
+// arrays, NC = candidate collection) with DYNAMIC LDS sized to exactly the
+// sections the shape uses. LDS is the occupancy limiter on this path
+// (160 KiB/CU): the monolithic struct was ~53 KiB => 3 workgroups/CU for
+// every workload; specialized shapes run 4 (BM25+collect, 34 KiB) to 9+
+// (agg-only, 17 KiB) workgroups/CU, which is what hides the gather latency.
 #include <hip/hip_runtime.h>
 
 #include "gpu_types.h"
@@ -65,33 +73,45 @@ __device__ __forceinline__ uint64_t f64_sortable(double d) {
     return (b & 0x8000000000000000ull) ? ~b : (b | 0x8000000000000000ull);
 }
 
-// ---------------------------------------------------------------- LDS state
-struct TileLds {
-    float score[TILE_DOCS];            // BM25 sums, or should-match counters
-    uint32_t bits_acc[TILE_DOCS / 32]; // must intersection
-    uint32_t bits_not[TILE_DOCS / 32]; // must_not union
-    uint32_t bits_m[TILE_DOCS / 32];   // final per-doc match (epilogue phase A)
-    uint32_t word_pref[TILE_DOCS / 32]; // exclusive match-rank prefix per word
-    uint32_t agg_hist[AGG_LDS_BUCKETS];
-    uint32_t agg_terms[AGG_LDS_BUCKETS];
-    uint32_t agg_matched[4];  // terms-agg docs-with-value, flushed per tile
-    uint32_t wave_base[4];
-    uint32_t cand_base;
+// ------------------------------------------------------------- LDS layout
+// Byte offsets of the sections a given query shape uses; total is the
+// dynamic-LDS size the host passes at launch (keep tile_lds_bytes in sync).
+template <bool NS, bool NB, bool NA, bool NC>
+struct SmemMap {
+    static constexpr uint32_t score_off = 0;                       // f32[TILE_DOCS]
+    static constexpr uint32_t score_end = NS ? TILE_DOCS * 4 : 0;
+    static constexpr uint32_t bits_acc_off = score_end;            // u32[256]
+    static constexpr uint32_t bits_not_off = bits_acc_off + (NB ? TILE_DOCS / 8 : 0);
+    static constexpr uint32_t bits_end = bits_not_off + (NB ? TILE_DOCS / 8 : 0);
+    static constexpr uint32_t bits_m_off = bits_end;               // u32[256]
+    static constexpr uint32_t word_pref_off = bits_m_off + (NC ? TILE_DOCS / 8 : 0);
+    static constexpr uint32_t coll_end = word_pref_off + (NC ? TILE_DOCS / 8 : 0);
+    static constexpr uint32_t agg_hist_off = coll_end;             // u32[2048] x2
+    static constexpr uint32_t agg_terms_off = agg_hist_off + (NA ? AGG_LDS_BUCKETS * 4 : 0);
+    static constexpr uint32_t agg_end = agg_terms_off + (NA ? AGG_LDS_BUCKETS * 4 : 0);
+    static constexpr uint32_t tail_off = agg_end;  // agg_matched[4] wave_base[4] cand_base
+    static constexpr uint32_t total = tail_off + 4 * 4 + 4 * 4 + 4;
 };
 
+constexpr uint32_t tile_lds_bytes(bool ns, bool nb, bool na, bool nc) {
+    return (ns ? TILE_DOCS * 4 : 0) + (nb ? 2 * TILE_DOCS / 8 : 0) +
+           (nc ? 2 * TILE_DOCS / 8 : 0) + (na ? 2 * AGG_LDS_BUCKETS * 4 : 0) + 36;
+}
+
 // decode every block of `t` overlapping the tile; accumulate into score[] /
-// set bits. mode: 0 = add score (or +1 count), 1 = set bits_acc-candidate
-// (bits_not reused as temp by caller), 2 = set bits_not
+// set bits in `bitset`. SCORE: add BM25 weight (or +1 count) into score.
+template <bool SCORE>
 __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t tile,
-                                 uint32_t tile_lo, uint32_t tile_hi, TileLds& lds,
-                                 uint32_t* bitset, int add_score) {
+                                 uint32_t tile_lo, uint32_t tile_hi, float* score,
+                                 uint32_t* bitset, bool scoring) {
     const uint32_t* ranges = (const uint32_t*)(q.scratch + t.ranges_off);
     uint32_t blo = ranges[tile], bhi = ranges[q.n_tiles + tile];
     if (blo >= bhi) return;
-    const SkipEntryDev* skip = (const SkipEntryDev*)(q.split + t.skip_off);
-    const uint32_t* payload = (const uint32_t*)(q.split + t.payload_off);
-    const uint8_t* norms = t.norms_off ? q.split + t.norms_off : nullptr;
-    const float* ktab = (const float*)(q.scratch + q.ktabs_off) + 256 * t.ktab_idx;
+    const SkipEntryDev* __restrict__ skip = (const SkipEntryDev*)(q.split + t.skip_off);
+    const uint32_t* __restrict__ payload = (const uint32_t*)(q.split + t.payload_off);
+    const uint8_t* __restrict__ norms = t.norms_off ? q.split + t.norms_off : nullptr;
+    const float* __restrict__ ktab =
+        (const float*)(q.scratch + q.ktabs_off) + 256 * t.ktab_idx;
     uint32_t wave = threadIdx.x >> 6;
     uint32_t lane = lane_id();
     for (uint32_t blk = blo + wave; blk < bhi; blk += TILE_THREADS / 64) {
@@ -107,7 +127,7 @@ __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t t
         uint32_t doc1 = e.first_doc + incl;
         uint32_t doc0 = doc1 - g1;
         float s0 = 1.0f, s1 = 1.0f;
-        if (add_score && q.scoring) {
+        if (SCORE && scoring) {
             uint32_t tf0 = 1, tf1 = 1;
             if (e.tf_bits) {
                 const uint32_t* tfbase =
@@ -125,12 +145,12 @@ __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t t
         }
         if (j0 < e.count && doc0 >= tile_lo && doc0 < tile_hi) {
             uint32_t li = doc0 - tile_lo;
-            if (add_score) atomicAdd(&lds.score[li], s0);
+            if (SCORE) atomicAdd(&score[li], s0);
             else atomicOr(&bitset[li >> 5], 1u << (li & 31));
         }
         if (j1 < e.count && doc1 >= tile_lo && doc1 < tile_hi) {
             uint32_t li = doc1 - tile_lo;
-            if (add_score) atomicAdd(&lds.score[li], s1);
+            if (SCORE) atomicAdd(&score[li], s1);
             else atomicOr(&bitset[li >> 5], 1u << (li & 31));
         }
     }
@@ -185,17 +205,50 @@ __device__ __forceinline__ uint64_t agg_ord(const QueryDev& q, uint64_t values_o
     }
 }
 
+// histogram bucket index relative to base_index, or -1 when out of range /
+// null. int_fast: exact 64-bit floor-division (host proved it bit-equal to
+// the oracle's floor((v-offset)/interval) double path for this column's
+// value range — product.cpp plan_aggs); else the double path verbatim.
+__device__ __forceinline__ int64_t histo_bucket(const QueryDev& q, const AggDev& a,
+                                                uint32_t doc) {
+    if (a.int_fast) {
+        const int64_t* col = (const int64_t*)(q.split + a.values_off);
+        int64_t num = col[doc] - a.i_offset;
+        // floor division by positive i_interval via reciprocal multiply +
+        // exact integer correction (cheaper than i64 div on CDNA4)
+        int64_t idx = int64_t(floor(double(num) * a.inv_interval));
+        if (idx * a.i_interval > num) --idx;
+        else if ((idx + 1) * a.i_interval <= num) ++idx;
+        return idx - a.base_index;
+    }
+    double v = agg_value(q, a.values_off, a.value_width, a.value_is_i64, doc);
+    return int64_t(floor((v - a.offset) / a.interval)) - a.base_index;
+}
+
 // ---------------------------------------------------------------- main kernel
-extern "C" __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile(QueryDev q,
-                                                                       uint32_t tile_base,
-                                                                       uint32_t tile_end,
-                                                                       uint32_t do_aggs,
-                                                                       uint32_t do_count,
-                                                                       uint32_t collect) {
-    __shared__ TileLds lds;
-    const TermDev* terms = (const TermDev*)(q.scratch + q.terms_off);
-    const PredDev* preds = (const PredDev*)(q.scratch + q.preds_off);
-    const AggDev* aggs = (const AggDev*)(q.scratch + q.aggs_off);
+// One instantiation per query shape; LDS sections per SmemMap. do_aggs /
+// collect of the old interface became NA / NC.
+template <bool NS, bool NB, bool NA, bool NC>
+__global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
+                                                              uint32_t tile_base,
+                                                              uint32_t tile_end,
+                                                              uint32_t do_count) {
+    using M = SmemMap<NS, NB, NA, NC>;
+    extern __shared__ uint8_t smem[];
+    float* sc_score = (float*)(smem + M::score_off);
+    uint32_t* sc_bits_acc = (uint32_t*)(smem + M::bits_acc_off);
+    uint32_t* sc_bits_not = (uint32_t*)(smem + M::bits_not_off);
+    uint32_t* sc_bits_m = (uint32_t*)(smem + M::bits_m_off);
+    uint32_t* sc_word_pref = (uint32_t*)(smem + M::word_pref_off);
+    uint32_t* sc_agg_hist = (uint32_t*)(smem + M::agg_hist_off);
+    uint32_t* sc_agg_terms = (uint32_t*)(smem + M::agg_terms_off);
+    uint32_t* sc_agg_matched = (uint32_t*)(smem + M::tail_off);
+    uint32_t* sc_wave_base = sc_agg_matched + 4;
+    uint32_t* sc_cand_base = sc_wave_base + 4;
+
+    const TermDev* __restrict__ terms = (const TermDev*)(q.scratch + q.terms_off);
+    const PredDev* __restrict__ preds = (const PredDev*)(q.scratch + q.preds_off);
+    const AggDev* __restrict__ aggs = (const AggDev*)(q.scratch + q.aggs_off);
     uint32_t* tile_counts = (uint32_t*)(q.results + q.tile_counts_off);
     uint32_t* cand_count = (uint32_t*)(q.results + q.cand_count_off);
     uint64_t* cand = (uint64_t*)(q.results + q.cand_off);
@@ -206,58 +259,64 @@ extern "C" __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile(QueryDev 
         bool have_should = q.msm > 0;  // shoulds REQUIRED (msm>=1); msm==0 =>
                                        // shoulds optional, only add scores
         // ---- zero LDS
-        if (q.n_terms)
+        if (NS)
             for (uint32_t i = threadIdx.x; i < TILE_DOCS; i += TILE_THREADS)
-                lds.score[i] = 0.f;
-        if (q.n_must || q.n_must_not)
+                sc_score[i] = 0.f;
+        if (NB)
             for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS) {
-                lds.bits_acc[i] = 0;
-                lds.bits_not[i] = 0;
+                sc_bits_acc[i] = 0;
+                sc_bits_not[i] = 0;
             }
-        if (collect)
+        if (NC)
             for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS)
-                lds.bits_m[i] = 0;
-        if (do_aggs && q.n_aggs) {
+                sc_bits_m[i] = 0;
+        if (NA && q.n_aggs) {
             for (uint32_t i = threadIdx.x; i < AGG_LDS_BUCKETS; i += TILE_THREADS) {
-                lds.agg_hist[i] = 0;
-                lds.agg_terms[i] = 0;
+                sc_agg_hist[i] = 0;
+                sc_agg_terms[i] = 0;
             }
-            if (threadIdx.x < 4) lds.agg_matched[threadIdx.x] = 0;
+            if (threadIdx.x < 4) sc_agg_matched[threadIdx.x] = 0;
         }
         __syncthreads();
 
         // ---- should terms: score / count union
-        for (uint32_t t = 0; t < q.n_terms; ++t)
-            if (terms[t].role == ROLE_SHOULD)
-                decode_term_tile(q, terms[t], tile, tile_lo, tile_hi, lds, nullptr, 1);
+        if (NS)
+            for (uint32_t t = 0; t < q.n_terms; ++t)
+                if (terms[t].role == ROLE_SHOULD)
+                    decode_term_tile<true>(q, terms[t], tile, tile_lo, tile_hi,
+                                           sc_score, nullptr, q.scoring);
         __syncthreads();
 
         // ---- must terms: intersection via per-term bitset
-        for (uint32_t t = 0, mi = 0; t < q.n_terms; ++t) {
-            if (terms[t].role != ROLE_MUST) continue;
-            // temp bitset lives in bits_not while must_nots are not yet done
-            uint32_t* tmp = lds.bits_not;
-            for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS)
-                tmp[i] = 0;
-            __syncthreads();
-            decode_term_tile(q, terms[t], tile, tile_lo, tile_hi, lds, tmp, 0);
-            if (q.scoring)  // musts contribute to the score too
-                decode_term_tile(q, terms[t], tile, tile_lo, tile_hi, lds, nullptr, 1);
-            __syncthreads();
-            for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS)
-                lds.bits_acc[i] = mi == 0 ? tmp[i] : (lds.bits_acc[i] & tmp[i]);
-            ++mi;
-            __syncthreads();
-        }
-        if (q.n_must_not) {
-            for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS)
-                lds.bits_not[i] = 0;
-            __syncthreads();
-            for (uint32_t t = 0; t < q.n_terms; ++t)
-                if (terms[t].role == ROLE_MUST_NOT)
-                    decode_term_tile(q, terms[t], tile, tile_lo, tile_hi, lds,
-                                     lds.bits_not, 0);
-            __syncthreads();
+        if (NB) {
+            for (uint32_t t = 0, mi = 0; t < q.n_terms; ++t) {
+                if (terms[t].role != ROLE_MUST) continue;
+                // temp bitset lives in bits_not while must_nots are not yet done
+                uint32_t* tmp = sc_bits_not;
+                for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS)
+                    tmp[i] = 0;
+                __syncthreads();
+                decode_term_tile<false>(q, terms[t], tile, tile_lo, tile_hi, nullptr,
+                                        tmp, false);
+                if (NS && q.scoring)  // musts contribute to the score too
+                    decode_term_tile<true>(q, terms[t], tile, tile_lo, tile_hi,
+                                           sc_score, nullptr, true);
+                __syncthreads();
+                for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS)
+                    sc_bits_acc[i] = mi == 0 ? tmp[i] : (sc_bits_acc[i] & tmp[i]);
+                ++mi;
+                __syncthreads();
+            }
+            if (q.n_must_not) {
+                for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS)
+                    sc_bits_not[i] = 0;
+                __syncthreads();
+                for (uint32_t t = 0; t < q.n_terms; ++t)
+                    if (terms[t].role == ROLE_MUST_NOT)
+                        decode_term_tile<false>(q, terms[t], tile, tile_lo, tile_hi,
+                                                nullptr, sc_bits_not, false);
+                __syncthreads();
+            }
         }
 
         // ---- epilogue: matched test, count, candidates, aggregations
@@ -266,28 +325,28 @@ extern "C" __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile(QueryDev 
             uint32_t li = d - tile_lo;
             bool m;
             float sc = 0.f;
-            if (have_should) {
+            if (NS && have_should) {
                 // shoulds required: scoring path assumes msm==1 and all
                 // weights > 0 (host rejects anything else — product.cpp)
                 if (q.scoring) {
-                    sc = lds.score[li];
+                    sc = sc_score[li];
                     m = sc > 0.f;
-                } else m = lds.score[li] >= float(q.msm);
-                if (m && q.n_must) m = (lds.bits_acc[li >> 5] >> (li & 31)) & 1;
+                } else m = sc_score[li] >= float(q.msm);
+                if (NB && m && q.n_must) m = (sc_bits_acc[li >> 5] >> (li & 31)) & 1;
             } else {
                 // base = match_all (no must terms; preds filter below) or the
                 // must-term intersection; optional shoulds only add scores
                 m = q.match_all ? true
-                                : (q.n_must > 0 &&
-                                   ((lds.bits_acc[li >> 5] >> (li & 31)) & 1));
-                if (q.scoring && m) sc = lds.score[li];
+                                : (NB && q.n_must > 0 &&
+                                   ((sc_bits_acc[li >> 5] >> (li & 31)) & 1));
+                if (NS && q.scoring && m) sc = sc_score[li];
             }
-            if (m && q.n_must_not) m = !((lds.bits_not[li >> 5] >> (li & 31)) & 1);
+            if (NB && m && q.n_must_not) m = !((sc_bits_not[li >> 5] >> (li & 31)) & 1);
             for (uint32_t p = 0; m && p < q.n_preds; ++p) m = eval_pred(q, preds[p], d);
 
             if (m) {
                 ++local_count;
-                if (do_aggs)
+                if (NA)
                     for (uint32_t ai = 0; ai < q.n_aggs; ++ai) {
                         const AggDev& a = aggs[ai];
                         if (a.nulls_off) {
@@ -295,25 +354,23 @@ extern "C" __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile(QueryDev 
                             if (!((nulls[d >> 6] >> (d & 63)) & 1)) continue;
                         }
                         if (a.kind == AGGD_TERMS) {
+                            if (!a.n_buckets) continue;  // missing/non-str column
                             uint64_t o = agg_ord(q, a.values_off, a.value_width, d);
-                            if (ai == 1 && a.n_buckets <= AGG_LDS_BUCKETS)
-                                atomicAdd(&lds.agg_terms[o], 1u);
+                            if (a.lds_slot == 1)
+                                atomicAdd(&sc_agg_terms[o], 1u);
                             else
                                 atomicAdd((unsigned long long*)(q.results + a.counts_out) + o,
                                           1ull);
                             if (ai < 4)  // per-tile LDS count, flushed below
-                                atomicAdd(&lds.agg_matched[ai], 1u);
+                                atomicAdd(&sc_agg_matched[ai], 1u);
                             else
                                 atomicAdd((unsigned long long*)(q.results + a.matched_out),
                                           1ull);
                         } else {
-                            double v = agg_value(q, a.values_off, a.value_width,
-                                                 a.value_is_i64, d);
-                            int64_t idx =
-                                int64_t(floor((v - a.offset) / a.interval)) - a.base_index;
+                            int64_t idx = histo_bucket(q, a, d);
                             if (idx < 0 || idx >= int64_t(a.n_buckets)) continue;
-                            if (ai == 0 && a.n_buckets <= AGG_LDS_BUCKETS)
-                                atomicAdd(&lds.agg_hist[uint32_t(idx)], 1u);
+                            if (a.lds_slot == 0)
+                                atomicAdd(&sc_agg_hist[uint32_t(idx)], 1u);
                             else
                                 atomicAdd((unsigned long long*)(q.results + a.counts_out) +
                                               idx, 1ull);
@@ -340,21 +397,20 @@ extern "C" __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile(QueryDev 
             // phase A of collection: record matches in an LDS bitset (no
             // global traffic; one ballot + direct word store per wave —
             // Guideline 12: never a per-wave global atomic in the hot loop)
-            if (collect) {
+            if (NC) {
                 uint64_t mb = __ballot(m);
                 uint32_t wbase = (li - lane_id()) >> 5;
-                if (lane_id() == 0) lds.bits_m[wbase] = uint32_t(mb);
-                else if (lane_id() == 32) lds.bits_m[wbase + 1] = uint32_t(mb >> 32);
+                if (lane_id() == 0) sc_bits_m[wbase] = uint32_t(mb);
+                else if (lane_id() == 32) sc_bits_m[wbase + 1] = uint32_t(mb >> 32);
             }
         }
         // ---- flush LDS agg arrays + matched counters (per-tile, once)
-        if (do_aggs) {
-            for (uint32_t ai = 0; ai < q.n_aggs && ai < 2; ++ai) {
+        if (NA) {
+            __syncthreads();
+            for (uint32_t ai = 0; ai < q.n_aggs; ++ai) {
                 const AggDev& a = aggs[ai];
-                if (a.n_buckets > AGG_LDS_BUCKETS) continue;
-                uint32_t* src = a.kind == AGGD_TERMS ? lds.agg_terms : lds.agg_hist;
-                if ((a.kind == AGGD_TERMS) != (ai == 1)) continue;
-                __syncthreads();
+                if (a.lds_slot > 1) continue;
+                uint32_t* src = a.lds_slot == 1 ? sc_agg_terms : sc_agg_hist;
                 for (uint32_t i = threadIdx.x; i < a.n_buckets; i += TILE_THREADS)
                     if (src[i])
                         atomicAdd((unsigned long long*)(q.results + a.counts_out) + i,
@@ -363,40 +419,40 @@ extern "C" __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile(QueryDev 
             __syncthreads();
             if (threadIdx.x < q.n_aggs && threadIdx.x < 4) {
                 const AggDev& a = aggs[threadIdx.x];
-                if (a.kind == AGGD_TERMS && lds.agg_matched[threadIdx.x])
+                if (a.kind == AGGD_TERMS && sc_agg_matched[threadIdx.x])
                     atomicAdd((unsigned long long*)(q.results + a.matched_out),
-                              (unsigned long long)lds.agg_matched[threadIdx.x]);
+                              (unsigned long long)sc_agg_matched[threadIdx.x]);
             }
         }
         // ---- per-tile count + phase B of collection (one global atomic per
         // TILE reserves contiguous cand space; ranks from an LDS bitset
         // prefix-scan; writes come out doc-ordered => coalesced)
-        if (collect) {
+        if (NC) {
             __syncthreads();  // bits_m/score final
             uint32_t w = threadIdx.x;  // one 32-doc word per thread (256 words)
-            uint32_t cnt = __popc(lds.bits_m[w]);
+            uint32_t cnt = __popc(sc_bits_m[w]);
             uint32_t incl = wave_incl_scan_u32(cnt);
             uint32_t wv = threadIdx.x >> 6;
-            if (lane_id() == 63) lds.wave_base[wv] = incl;
+            if (lane_id() == 63) sc_wave_base[wv] = incl;
             __syncthreads();
             uint32_t wave_off = 0;
-            for (uint32_t i = 0; i < wv; ++i) wave_off += lds.wave_base[i];
-            lds.word_pref[w] = wave_off + incl - cnt;
-            uint32_t tile_total = lds.wave_base[0] + lds.wave_base[1] +
-                                  lds.wave_base[2] + lds.wave_base[3];
+            for (uint32_t i = 0; i < wv; ++i) wave_off += sc_wave_base[i];
+            sc_word_pref[w] = wave_off + incl - cnt;
+            uint32_t tile_total = sc_wave_base[0] + sc_wave_base[1] +
+                                  sc_wave_base[2] + sc_wave_base[3];
             if (threadIdx.x == 0) {
-                lds.cand_base = atomicAdd(cand_count, tile_total);
+                *sc_cand_base = atomicAdd(cand_count, tile_total);
                 if (do_count) tile_counts[tile] = tile_total;
             }
             __syncthreads();
-            uint64_t base = lds.cand_base;
+            uint64_t base = *sc_cand_base;
             for (uint32_t d = tile_lo + threadIdx.x; d < tile_hi; d += TILE_THREADS) {
                 uint32_t li = d - tile_lo;
-                uint32_t bits = lds.bits_m[li >> 5];
+                uint32_t bits = sc_bits_m[li >> 5];
                 if (!((bits >> (li & 31)) & 1)) continue;
-                uint64_t rank = lds.word_pref[li >> 5] +
+                uint64_t rank = sc_word_pref[li >> 5] +
                                 __popc(bits & ((1u << (li & 31)) - 1u));
-                float sc = q.scoring ? lds.score[li] : 0.f;
+                float sc = (NS && q.scoring) ? sc_score[li] : 0.f;
                 uint32_t kh = q.scoring ? f32_sortable(sc) : 0u;
                 if (q.sort_asc) kh = ~kh;
                 uint32_t kl = q.sort_asc ? ~d : d;
@@ -405,16 +461,50 @@ extern "C" __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile(QueryDev 
             }
         } else if (do_count) {
             __syncthreads();
-            uint32_t* wb = lds.wave_base;
             #pragma unroll
             for (int dlt = 32; dlt; dlt >>= 1) local_count += __shfl_down(local_count, dlt, 64);
-            if (lane_id() == 0) wb[threadIdx.x >> 6] = local_count;
+            if (lane_id() == 0) sc_wave_base[threadIdx.x >> 6] = local_count;
             __syncthreads();
             if (threadIdx.x == 0)
-                tile_counts[tile] = wb[0] + wb[1] + wb[2] + wb[3];
+                tile_counts[tile] = sc_wave_base[0] + sc_wave_base[1] +
+                                    sc_wave_base[2] + sc_wave_base[3];
         }
         __syncthreads();
     }
+}
+
+// host-side dispatcher over the 16 shape instantiations; returns the dynamic
+// LDS bytes it launched with (for occupancy reporting)
+inline uint32_t launch_leaf_tile(bool ns, bool nb, bool na, bool nc, dim3 grid,
+                                 hipStream_t stream, const QueryDev& q,
+                                 uint32_t tile_base, uint32_t tile_end,
+                                 uint32_t do_count) {
+    uint32_t lds = tile_lds_bytes(ns, nb, na, nc);
+    #define QW_CASE(NS_, NB_, NA_, NC_)                                           \
+        if (ns == NS_ && nb == NB_ && na == NA_ && nc == NC_) {                   \
+            hipLaunchKernelGGL((k_leaf_tile_t<NS_, NB_, NA_, NC_>), grid,         \
+                               dim3(TILE_THREADS), lds, stream, q, tile_base,     \
+                               tile_end, do_count);                               \
+            return lds;                                                           \
+        }
+    QW_CASE(false, false, false, false)
+    QW_CASE(false, false, false, true)
+    QW_CASE(false, false, true, false)
+    QW_CASE(false, false, true, true)
+    QW_CASE(false, true, false, false)
+    QW_CASE(false, true, false, true)
+    QW_CASE(false, true, true, false)
+    QW_CASE(false, true, true, true)
+    QW_CASE(true, false, false, false)
+    QW_CASE(true, false, false, true)
+    QW_CASE(true, false, true, false)
+    QW_CASE(true, false, true, true)
+    QW_CASE(true, true, false, false)
+    QW_CASE(true, true, false, true)
+    QW_CASE(true, true, true, false)
+    QW_CASE(true, true, true, true)
+    #undef QW_CASE
+    return 0;  // unreachable
 }
 
 // ----------------------------------------------------- top-K selection passes

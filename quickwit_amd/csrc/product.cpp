@@ -382,6 +382,7 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
         const FastFieldView* f = sv.fast_field(d.field);
         ap.fields.push_back(f);
         AggDev a{};
+        a.lds_slot = 0xFF;
         if (d.kind == AggDef::TERMS) {
             a.kind = AGGD_TERMS;
             if (f && f->type == FastFieldView::STR) {
@@ -409,6 +410,29 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
                 int64_t b0 = int64_t(std::floor((mn - d.offset) / d.interval));
                 int64_t b1 = int64_t(std::floor((mx - d.offset) / d.interval));
                 a.base_index = b0;
+                // exact integer fast path (kernels.hip histo_bucket): valid
+                // when interval/offset are integral, the i64 load reads the
+                // true value (u64 columns must fit i64), and |v-offset| stays
+                // below 2^53 so the double division the oracle does can never
+                // land within one rounding error of an integer other than at
+                // exact multiples (error <= |v-off|*2^-53 < 1 < 1/interval
+                // distance to the nearest non-multiple boundary).
+                {
+                    double iv = d.interval, of = d.offset;
+                    bool integral = iv >= 1.0 && std::floor(iv) == iv &&
+                                    std::floor(of) == of;
+                    bool u64_ok = f->type != FastFieldView::U64 ||
+                                  uint64_t(f->max_value) < (1ull << 63);
+                    double span = std::max(std::fabs(mn - of), std::fabs(mx - of));
+                    if (integral && u64_ok && span < 9.0e15 && iv < 9.0e15 &&
+                        std::fabs(of) < 9.0e15 && std::fabs(mn) < 9.0e15 &&
+                        std::fabs(mx) < 9.0e15) {
+                        a.int_fast = 1;
+                        a.i_interval = int64_t(iv);
+                        a.i_offset = int64_t(of);
+                        a.inv_interval = 1.0 / iv;
+                    }
+                }
                 int64_t nb = b1 - b0 + 1;
                 if (nb < 0 || nb > bucket_limit)
                     throw std::runtime_error(
@@ -441,6 +465,21 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
         a.sub_out = off;
         off += uint64_t(a.n_buckets) * a.n_sub * 32;
         ap.devs.push_back(a);
+    }
+    // LDS slot assignment: the first histogram agg that fits gets the LDS
+    // histogram array, the first terms agg that fits gets the LDS count
+    // table; everything else accumulates with global atomics.
+    bool slot0 = false, slot1 = false;
+    for (AggDev& a : ap.devs) {
+        if (a.kind == AGGD_HISTO && !slot0 && a.n_buckets &&
+            a.n_buckets <= AGG_LDS_BUCKETS) {
+            a.lds_slot = 0;
+            slot0 = true;
+        } else if (a.kind == AGGD_TERMS && !slot1 && a.n_buckets &&
+                   a.n_buckets <= AGG_LDS_BUCKETS) {
+            a.lds_slot = 1;
+            slot1 = true;
+        }
     }
     ap.out_bytes = off - out_base;
     // init pattern: zeros except sub min slots = f64_sortable(+inf) pattern max
@@ -689,10 +728,15 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         const char* kname = !fq.terms.empty() ? "union_bm25"
                             : do_aggs         ? "column_agg"
                                               : "range_filter";
+        // query-shape flags select the template instantiation whose dynamic
+        // LDS holds exactly the sections this shape touches (kernels.hip)
+        uint32_t n_should = uint32_t(terms.size()) - n_must - n_must_not;
+        bool ns = n_should > 0 || (fq.scoring && !terms.empty());
+        bool nb = n_must > 0 || n_must_not > 0;
+        bool na = do_aggs && !ap.devs.empty();
         HIP_CHECK(hipEventRecord(ctx->ev_start, ctx->stream));
-        hipLaunchKernelGGL(k_leaf_tile, dim3(grid), dim3(TILE_THREADS), 0, ctx->stream,
-                           q, 0u, n_tiles, uint32_t(do_aggs ? 1 : 0), 1u,
-                           uint32_t(collect ? 1 : 0));
+        launch_leaf_tile(ns, nb, na, collect, dim3(grid), ctx->stream, q, 0u,
+                         n_tiles, 1u);
         HIP_CHECK(hipEventRecord(ctx->ev_stop, ctx->stream));
         HIP_CHECK(hipGetLastError());
 
