@@ -114,9 +114,11 @@ __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t t
         (const float*)(q.scratch + q.ktabs_off) + 256 * t.ktab_idx;
     uint32_t wave = threadIdx.x >> 6;
     uint32_t lane = lane_id();
-    for (uint32_t blk = blo + wave; blk < bhi; blk += TILE_THREADS / 64) {
+    // decode one 128-doc block within a wave: funnel-shift gap extraction, a
+    // 64-lane shuffle prefix-sum for the deltas, fused BM25 FMA
+    auto do_block = [&](uint32_t blk) {
         SkipEntryDev e = skip[blk];
-        if (e.first_doc >= tile_hi || e.last_doc < tile_lo) continue;
+        if (e.first_doc >= tile_hi || e.last_doc < tile_lo) return;
         const uint32_t* idbase = payload + e.word_off;
         uint32_t w = e.id_bits;
         // lane handles elements 2*lane and 2*lane+1
@@ -153,6 +155,12 @@ __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t t
             if (SCORE) atomicAdd(&score[li], s1);
             else atomicOr(&bitset[li >> 5], 1u << (li & 31));
         }
+    };
+    // two blocks per wave-iteration: the prefix-scan chains of the pair are
+    // independent, so their shuffle/latency chains overlap (ILP)
+    for (uint32_t blk = blo + 2 * wave; blk < bhi; blk += 2 * (TILE_THREADS / 64)) {
+        do_block(blk);
+        if (blk + 1 < bhi) do_block(blk + 1);
     }
 }
 
@@ -253,6 +261,18 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
     uint32_t* cand_count = (uint32_t*)(q.results + q.cand_count_off);
     uint64_t* cand = (uint64_t*)(q.results + q.cand_off);
 
+    // agg LDS arrays accumulate across ALL tiles this workgroup owns and
+    // flush ONCE at the end (global flush atomics scale with gridDim, not
+    // n_tiles — the host caps the agg grid so a workgroup owns several
+    // tiles). u32 counts cannot overflow: a workgroup owns < 2^31 docs.
+    if (NA && q.n_aggs) {
+        for (uint32_t i = threadIdx.x; i < AGG_LDS_BUCKETS; i += TILE_THREADS) {
+            sc_agg_hist[i] = 0;
+            sc_agg_terms[i] = 0;
+        }
+        if (threadIdx.x < 4) sc_agg_matched[threadIdx.x] = 0;
+    }
+
     for (uint32_t tile = tile_base + blockIdx.x; tile < tile_end; tile += gridDim.x) {
         uint32_t tile_lo = tile * TILE_DOCS;
         uint32_t tile_hi = min(tile_lo + TILE_DOCS, q.num_docs);
@@ -270,13 +290,6 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
         if (NC)
             for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS)
                 sc_bits_m[i] = 0;
-        if (NA && q.n_aggs) {
-            for (uint32_t i = threadIdx.x; i < AGG_LDS_BUCKETS; i += TILE_THREADS) {
-                sc_agg_hist[i] = 0;
-                sc_agg_terms[i] = 0;
-            }
-            if (threadIdx.x < 4) sc_agg_matched[threadIdx.x] = 0;
-        }
         __syncthreads();
 
         // ---- should terms: score / count union
@@ -404,26 +417,6 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                 else if (lane_id() == 32) sc_bits_m[wbase + 1] = uint32_t(mb >> 32);
             }
         }
-        // ---- flush LDS agg arrays + matched counters (per-tile, once)
-        if (NA) {
-            __syncthreads();
-            for (uint32_t ai = 0; ai < q.n_aggs; ++ai) {
-                const AggDev& a = aggs[ai];
-                if (a.lds_slot > 1) continue;
-                uint32_t* src = a.lds_slot == 1 ? sc_agg_terms : sc_agg_hist;
-                for (uint32_t i = threadIdx.x; i < a.n_buckets; i += TILE_THREADS)
-                    if (src[i])
-                        atomicAdd((unsigned long long*)(q.results + a.counts_out) + i,
-                                  (unsigned long long)src[i]);
-            }
-            __syncthreads();
-            if (threadIdx.x < q.n_aggs && threadIdx.x < 4) {
-                const AggDev& a = aggs[threadIdx.x];
-                if (a.kind == AGGD_TERMS && sc_agg_matched[threadIdx.x])
-                    atomicAdd((unsigned long long*)(q.results + a.matched_out),
-                              (unsigned long long)sc_agg_matched[threadIdx.x]);
-            }
-        }
         // ---- per-tile count + phase B of collection (one global atomic per
         // TILE reserves contiguous cand space; ranks from an LDS bitset
         // prefix-scan; writes come out doc-ordered => coalesced)
@@ -470,6 +463,26 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                                     sc_wave_base[2] + sc_wave_base[3];
         }
         __syncthreads();
+    }
+
+    // ---- flush LDS agg arrays + matched counters (once per workgroup)
+    if (NA && q.n_aggs) {
+        __syncthreads();
+        for (uint32_t ai = 0; ai < q.n_aggs; ++ai) {
+            const AggDev& a = aggs[ai];
+            if (a.lds_slot > 1) continue;
+            uint32_t* src = a.lds_slot == 1 ? sc_agg_terms : sc_agg_hist;
+            for (uint32_t i = threadIdx.x; i < a.n_buckets; i += TILE_THREADS)
+                if (src[i])
+                    atomicAdd((unsigned long long*)(q.results + a.counts_out) + i,
+                              (unsigned long long)src[i]);
+        }
+        if (threadIdx.x < q.n_aggs && threadIdx.x < 4) {
+            const AggDev& a = aggs[threadIdx.x];
+            if (a.kind == AGGD_TERMS && sc_agg_matched[threadIdx.x])
+                atomicAdd((unsigned long long*)(q.results + a.matched_out),
+                          (unsigned long long)sc_agg_matched[threadIdx.x]);
+        }
     }
 }
 

@@ -724,7 +724,6 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         q.cand_cap = cand_cap;
         q.hist_off = r_hist;
 
-        uint32_t grid = n_tiles;  // one workgroup per tile; gfx950 has 256 CUs
         const char* kname = !fq.terms.empty() ? "union_bm25"
                             : do_aggs         ? "column_agg"
                                               : "range_filter";
@@ -734,6 +733,9 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         bool ns = n_should > 0 || (fq.scoring && !terms.empty());
         bool nb = n_must > 0 || n_must_not > 0;
         bool na = do_aggs && !ap.devs.empty();
+        // agg workloads: capped grid so the once-per-workgroup LDS flush
+        // stays cheap (each WG owns several tiles); otherwise one WG per tile
+        uint32_t grid = na ? std::min<uint32_t>(n_tiles, 2048) : n_tiles;
         HIP_CHECK(hipEventRecord(ctx->ev_start, ctx->stream));
         launch_leaf_tile(ns, nb, na, collect, dim3(grid), ctx->stream, q, 0u,
                          n_tiles, 1u);
