@@ -97,7 +97,17 @@ struct QueryDev {
     uint32_t scoring;       // 1 = BM25 scores wanted
     uint32_t match_all;     // 1 = base matches everything (no should clauses)
     uint32_t collect_hits;  // 1 = write candidates
-    uint32_t sort_asc;      // 1 = ascending _score sort (flip candidate keys)
+    uint32_t sort_asc;      // 1 = ascending primary sort (flip candidate keys)
+    // primary sort-key source for candidate records (collector.rs:403-414
+    // sort-key extraction). narrow records (8B: key32|doc) serve _score /
+    // doc-order sorts; wide records (16B: u64 key, u32 score_bits|u32 doc)
+    // serve fast-field sorts and two-field sorts — exact order is
+    // re-established on host over the survivors.
+    uint32_t wide_cand;     // 1 = 16B candidate records
+    uint32_t sort_src;      // 0 none/doc, 1 _score, 2 u64/ord column, 3 i64 column
+    uint32_t sort_width;    // column width (1/2/4/8), sort_src>=2
+    uint64_t sort_values_off;
+    uint64_t sort_nulls_off;  // 0 = non-nullable (missing -> None, sorts last)
     uint64_t terms_off;     // scratch offsets of descriptor arrays
     uint64_t preds_off;
     uint64_t aggs_off;

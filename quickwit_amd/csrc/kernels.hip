@@ -213,6 +213,33 @@ __device__ __forceinline__ uint64_t agg_ord(const QueryDev& q, uint64_t values_o
     }
 }
 
+// primary sort key of a matched doc for WIDE candidate records: the same
+// monotonic u64 maps the host comparator uses (sortkey.h); None (missing
+// fast-field value) maps to 0 so it sorts last under either order — exact
+// ordering among boundary ties is re-established on host (sorting.md:14-26).
+__device__ __forceinline__ uint64_t wide_sort_key(const QueryDev& q, uint32_t d,
+                                                  float sc) {
+    if (q.sort_src == 0) return 0;  // unknown sort field: None for every doc
+    if (q.sort_nulls_off) {
+        const uint64_t* nulls = (const uint64_t*)(q.split + q.sort_nulls_off);
+        if (!((nulls[d >> 6] >> (d & 63)) & 1)) return 0;
+    }
+    uint64_t S = 0;
+    if (q.sort_src == 1) {
+        S = uint64_t(f32_sortable(sc)) << 32;
+    } else {
+        const uint8_t* col = q.split + q.sort_values_off;
+        switch (q.sort_width) {
+            case 1: S = col[d]; break;
+            case 2: S = ((const uint16_t*)col)[d]; break;
+            case 4: S = ((const uint32_t*)col)[d]; break;
+            default: S = ((const uint64_t*)col)[d]; break;
+        }
+        if (q.sort_src == 3) S ^= (1ull << 63);  // i64_to_u64
+    }
+    return q.sort_asc ? ~S : S;
+}
+
 // histogram bucket index relative to base_index, or -1 when out of range /
 // null. int_fast: exact 64-bit floor-division (host proved it bit-equal to
 // the oracle's floor((v-offset)/interval) double path for this column's
@@ -446,11 +473,18 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                 uint64_t rank = sc_word_pref[li >> 5] +
                                 __popc(bits & ((1u << (li & 31)) - 1u));
                 float sc = (NS && q.scoring) ? sc_score[li] : 0.f;
-                uint32_t kh = q.scoring ? f32_sortable(sc) : 0u;
-                if (q.sort_asc) kh = ~kh;
-                uint32_t kl = q.sort_asc ? ~d : d;
-                if (base + rank < q.cand_cap)
+                if (base + rank >= q.cand_cap) continue;
+                if (q.wide_cand) {
+                    uint64_t key = wide_sort_key(q, d, sc);
+                    cand[2 * (base + rank)] = key;
+                    cand[2 * (base + rank) + 1] =
+                        (uint64_t(__float_as_uint(sc)) << 32) | d;
+                } else {
+                    uint32_t kh = q.scoring ? f32_sortable(sc) : 0u;
+                    if (q.sort_asc) kh = ~kh;
+                    uint32_t kl = q.sort_asc ? ~d : d;
                     cand[base + rank] = (uint64_t(kh) << 32) | kl;
+                }
             }
         } else if (do_count) {
             __syncthreads();
@@ -538,6 +572,50 @@ extern "C" __global__ void k_cand_hist(const uint64_t* cand, uint32_t n, uint64_
     __syncthreads();
     for (uint32_t i = threadIdx.x; i < TOPK_BINS; i += blockDim.x)
         if (lh[i]) atomicAdd(&hist[i], lh[i]);
+}
+
+// wide-record (16B {u64 key, u64 aux}) variants of the selection passes
+extern "C" __global__ void k_cand_hist_w(const uint64_t* cand, uint32_t n,
+                                         uint64_t prefix, uint32_t prefix_bits,
+                                         uint32_t* hist) {
+    __shared__ uint32_t lh[TOPK_BINS];
+    for (uint32_t i = threadIdx.x; i < TOPK_BINS; i += blockDim.x) lh[i] = 0;
+    __syncthreads();
+    uint32_t shift = 64 - prefix_bits - 12;
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * blockDim.x) {
+        uint64_t k = cand[2 * i];
+        if (prefix_bits && (k >> (64 - prefix_bits)) != prefix) continue;
+        atomicAdd(&lh[(k >> shift) & (TOPK_BINS - 1)], 1u);
+    }
+    __syncthreads();
+    for (uint32_t i = threadIdx.x; i < TOPK_BINS; i += blockDim.x)
+        if (lh[i]) atomicAdd(&hist[i], lh[i]);
+}
+
+extern "C" __global__ void k_cand_compact_w(const uint64_t* cand, uint32_t n,
+                                            uint64_t floor_key, uint64_t ceil_key,
+                                            uint64_t* out, uint32_t* out_count,
+                                            uint32_t cap) {
+    for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += gridDim.x * blockDim.x) {
+        uint64_t k = cand[2 * i];
+        bool take = k >= floor_key && k <= ceil_key;
+        uint64_t mask = __ballot(take);
+        uint32_t nw = __popcll(mask);
+        if (!nw) continue;
+        uint32_t leader = __ffsll((unsigned long long)mask) - 1;
+        uint32_t base;
+        if (lane_id() == leader) base = atomicAdd(out_count, nw);
+        base = __shfl(base, leader, 64);
+        if (take) {
+            uint32_t off = base + __popcll(mask & ((1ull << lane_id()) - 1ull));
+            if (off < cap) {
+                out[2 * off] = k;
+                out[2 * off + 1] = cand[2 * i + 1];
+            }
+        }
+    }
 }
 
 // compact candidates with key >= floor_key into out (bounded by cap)

@@ -507,10 +507,45 @@ struct SplitResult {
     std::string error;
 };
 
-struct SortPlan {
-    bool by_score = false;  // else doc-id order
-    int order1 = 1, order2 = 1;
+// sort specs, mirroring the oracle's (collector.rs:403-414 sort-key
+// extraction; sorting.md None-last + GlobalDocId tie-break semantics)
+struct SortSpec {
+    enum Comp { DOC_ID, SCORE, FAST_FIELD } comp = DOC_ID;
+    const FastFieldView* ff = nullptr;  // null: unknown field -> None values
+    int order = 1;                      // 0 asc 1 desc
 };
+
+static pb::SortByValue sort_value_of(const SortSpec& s, uint32_t doc, float score) {
+    pb::SortByValue v;
+    switch (s.comp) {
+        case SortSpec::DOC_ID:
+            break;
+        case SortSpec::SCORE:
+            v.kind = pb::SortByValue::F64;
+            v.f64 = double(score);
+            break;
+        case SortSpec::FAST_FIELD: {
+            const FastFieldView* f = s.ff;
+            if (!f || !f->present(doc)) break;
+            if (f->type == FastFieldView::U64) {
+                v.kind = pb::SortByValue::U64;
+                v.u64 = f->u64(doc);
+            } else if (f->type == FastFieldView::DATETIME) {
+                v.kind = pb::SortByValue::I64;
+                v.i64 = f->i64(doc) * 1000000;  // ms -> ns (sorting.md default
+                                                // output unix_timestamp_nanos)
+            } else if (f->type == FastFieldView::STR) {
+                v.kind = pb::SortByValue::U64;
+                v.u64 = f->ord(doc);
+            } else {
+                v.kind = pb::SortByValue::I64;
+                v.i64 = f->i64(doc);
+            }
+            break;
+        }
+    }
+    return v;
+}
 
 static void record_kernel_time(qw_ctx* ctx, const char* name, float ms) {
     KernelTimer& t = ctx->timers[name];
@@ -545,20 +580,31 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         plan = std::move(b);
     }
 
-    SortPlan sp;
-    if (!req.sort_fields.empty()) {
-        sp.order1 = req.sort_fields[0].sort_order;
-        if (req.sort_fields.size() > 1) sp.order2 = req.sort_fields[1].sort_order;
-        if (req.sort_fields[0].field_name == "_score") sp.by_score = true;
-        else
-            throw std::runtime_error(
-                "sort by fast field not supported on GPU path in r1 (use _score or "
-                "default doc order)");
-        if (req.sort_fields.size() > 1)
-            throw std::runtime_error("two sort fields not supported on GPU path in r1");
+    if (req.sort_fields.size() > 2)
+        throw std::runtime_error("more than two sort fields (search.proto:269)");
+    std::vector<SortSpec> specs;
+    for (auto& sf : req.sort_fields) {
+        SortSpec s;
+        s.order = sf.sort_order;
+        if (sf.field_name == "_score") s.comp = SortSpec::SCORE;
+        else {
+            s.comp = SortSpec::FAST_FIELD;
+            s.ff = sv.fast_field(sf.field_name);
+        }
+        specs.push_back(s);
     }
+    bool scoring = false;
+    for (auto& s : specs) scoring |= s.comp == SortSpec::SCORE;
+    int order1 = specs.empty() ? 1 : specs[0].order;
+    int order2 = specs.size() > 1 ? specs[1].order : 1;
+    // narrow (8B) candidate records embed the f32 _score key + doc tiebreak:
+    // exact device-side selection. fast-field or two-field sorts use wide
+    // (16B) records selected by primary key only; exact order over the
+    // survivors is re-established on host (kernels.hip wide_sort_key).
+    bool wide = (!specs.empty() && specs[0].comp == SortSpec::FAST_FIELD) ||
+                specs.size() > 1;
 
-    FlatQuery fq = flatten(sv, plan, sp.by_score);
+    FlatQuery fq = flatten(sv, plan, scoring);
     uint64_t leaf_max_hits = req.max_hits + req.start_offset;
 
     if (fq.match_none) {
@@ -586,9 +632,11 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
 
     uint32_t n_tiles = (sv.num_docs + TILE_DOCS - 1) / TILE_DOCS;
     bool pure_match_all = fq.match_all && fq.preds.empty() && fq.terms.empty();
-    bool need_kernel = !pure_match_all || do_aggs;
+    // hits trivially enumerable only under doc-id order (leaf.rs default)
+    bool trivial_hits = pure_match_all && specs.empty();
     // candidate collection: needed unless hits are trivially enumerable
-    bool collect = leaf_max_hits > 0 && !pure_match_all && !fq.match_none;
+    bool collect = leaf_max_hits > 0 && !trivial_hits && !fq.match_none;
+    bool need_kernel = !pure_match_all || do_aggs || collect;
 
     // ---- scratch assembly (descriptors + ktabs + block ranges), one H2D
     std::vector<TermDev> terms(fq.terms.size());
@@ -649,7 +697,8 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         ap = plan_aggs(sv, *req.aggregation_request, ctx->agg_bucket_limit, r_agg);
     size_t r_cand = r_agg + ((ap.out_bytes + 63) & ~size_t(63));
     size_t cand_cap = collect ? sv.num_docs : 0;
-    size_t results_bytes = r_cand + cand_cap * 8;
+    size_t cand_rec = wide ? 16 : 8;
+    size_t results_bytes = r_cand + cand_cap * cand_rec;
 
     size_t off_ktabs = off_aggs + ap.devs.size() * sizeof(AggDev);
     size_t off_ranges = off_ktabs + ktabs.size() * 4;
@@ -713,7 +762,30 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         q.scoring = fq.scoring ? 1 : 0;
         q.match_all = fq.match_all ? 1 : 0;
         q.collect_hits = collect ? 1 : 0;
-        q.sort_asc = (sp.by_score && sp.order1 == 0) ? 1 : 0;
+        q.sort_asc = (!specs.empty() && specs[0].comp != SortSpec::DOC_ID &&
+                      order1 == 0)
+                         ? 1
+                         : 0;
+        q.wide_cand = wide ? 1 : 0;
+        if (wide) {
+            const SortSpec& s0 = specs[0];
+            if (s0.comp == SortSpec::SCORE) {
+                q.sort_src = 1;
+            } else if (s0.ff) {
+                const FastFieldView* f = s0.ff;
+                q.sort_values_off = f->values.off;
+                q.sort_nulls_off = f->nullable ? f->nulls.off : 0;
+                if (f->type == FastFieldView::STR) {
+                    q.sort_src = 2;
+                    q.sort_width = uint32_t(f->ord_width);
+                } else {
+                    q.sort_src = f->type == FastFieldView::U64 ? 2 : 3;
+                    q.sort_width = 8;
+                }
+            } else {
+                q.sort_src = 0;  // unknown field: every key is None
+            }
+        }
         q.terms_off = off_terms;
         q.preds_off = off_preds;
         q.aggs_off = off_aggs;
@@ -764,6 +836,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             uint64_t* d_cand = (uint64_t*)(ctx->d_results.p + r_cand);
             uint32_t* d_hist = (uint32_t*)(ctx->d_results.p + r_hist);
             uint32_t* d_scount = (uint32_t*)(ctx->d_results.p + r_cand_count) + 1;
+            uint32_t rw = wide ? 2 : 1;  // u64 words per candidate record
             uint64_t prefix = 0;
             uint32_t prefix_bits = 0;
             uint64_t survivors = cand_n, above = 0, Krem = K;
@@ -779,8 +852,14 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                 // grid small enough that the per-WG 4096-bin LDS flush (global
                 // atomics ∝ grid) stays cheap, big enough to fill the chip
                 uint32_t hgrid = std::min<uint32_t>(512, (cand_n + 4095) / 4096);
-                hipLaunchKernelGGL(k_cand_hist, dim3(hgrid), dim3(256), 0, ctx->stream,
-                                   d_cand, cand_n, prefix, prefix_bits, d_hist);
+                if (wide)
+                    hipLaunchKernelGGL(k_cand_hist_w, dim3(hgrid), dim3(256), 0,
+                                       ctx->stream, d_cand, cand_n, prefix,
+                                       prefix_bits, d_hist);
+                else
+                    hipLaunchKernelGGL(k_cand_hist, dim3(hgrid), dim3(256), 0,
+                                       ctx->stream, d_cand, cand_n, prefix,
+                                       prefix_bits, d_hist);
                 HIP_CHECK(hipMemcpyAsync(hist.data(), d_hist, TOPK_BINS * 4,
                                          hipMemcpyDeviceToHost, ctx->stream));
                 HIP_CHECK(hipStreamSynchronize(ctx->stream));
@@ -802,24 +881,33 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             }
             // #keys >= floor_key is known EXACTLY from the histogram walk, so
             // no round trip for the compact count: one async D2H of survivors
-            ctx->d_survivors.ensure(survivors * 8 + 8);
+            ctx->d_survivors.ensure(survivors * 8 * rw + 16);
             HIP_CHECK(hipMemsetAsync(d_scount, 0, 4, ctx->stream));
             uint32_t cgrid = std::min<uint32_t>(2048, (cand_n + 255) / 256);
-            hipLaunchKernelGGL(k_cand_compact, dim3(cgrid), dim3(256), 0, ctx->stream,
-                               d_cand, cand_n, floor_key, ~0ull,
-                               (uint64_t*)ctx->d_survivors.p, d_scount,
-                               uint32_t(survivors));
-            top_keys.resize(survivors);
+            if (wide)
+                hipLaunchKernelGGL(k_cand_compact_w, dim3(cgrid), dim3(256), 0,
+                                   ctx->stream, d_cand, cand_n, floor_key, ~0ull,
+                                   (uint64_t*)ctx->d_survivors.p, d_scount,
+                                   uint32_t(survivors));
+            else
+                hipLaunchKernelGGL(k_cand_compact, dim3(cgrid), dim3(256), 0,
+                                   ctx->stream, d_cand, cand_n, floor_key, ~0ull,
+                                   (uint64_t*)ctx->d_survivors.p, d_scount,
+                                   uint32_t(survivors));
+            top_keys.resize(survivors * rw);
             HIP_CHECK(hipMemcpyAsync(top_keys.data(), ctx->d_survivors.p,
-                                     survivors * 8, hipMemcpyDeviceToHost,
+                                     survivors * 8 * rw, hipMemcpyDeviceToHost,
                                      ctx->stream));
             HIP_CHECK(hipEventRecord(ctx->ev_stop, ctx->stream));
             HIP_CHECK(hipStreamSynchronize(ctx->stream));
             float tms = 0;
             HIP_CHECK(hipEventElapsedTime(&tms, ctx->ev_start, ctx->ev_stop));
             record_kernel_time(ctx, "topk_select", tms);
-            std::sort(top_keys.begin(), top_keys.end(), std::greater<uint64_t>());
-            if (top_keys.size() > K) top_keys.resize(K);
+            if (!wide) {
+                std::sort(top_keys.begin(), top_keys.end(),
+                          std::greater<uint64_t>());
+                if (top_keys.size() > K) top_keys.resize(K);
+            }
         }
     } else {
         matched = sv.num_docs;  // pure match_all, no aggs
@@ -834,23 +922,44 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         h.split_id = sv.split_id;
         h.segment_ord = 0;
         h.doc_id = doc;
-        if (sp.by_score) {
-            h.sort_value.kind = pb::SortByValue::F64;
-            h.sort_value.f64 = double(score);
-        }
+        if (!specs.empty()) h.sort_value = sort_value_of(specs[0], doc, score);
+        if (specs.size() > 1) h.sort_value2 = sort_value_of(specs[1], doc, score);
         return h;
     };
     if (leaf_max_hits > 0) {
-        if (pure_match_all) {
+        if (trivial_hits) {
             // hits enumerable without the kernel: doc-id order (desc default)
             uint64_t k = std::min<uint64_t>(leaf_max_hits, sv.num_docs);
             for (uint64_t i = 0; i < k; ++i) {
-                uint32_t doc = sp.order1 == 1 ? uint32_t(sv.num_docs - 1 - i)
-                                              : uint32_t(i);
+                uint32_t doc = order1 == 1 ? uint32_t(sv.num_docs - 1 - i)
+                                           : uint32_t(i);
                 out.hits.push_back(mk_hit(doc, 0.f));
             }
+        } else if (wide) {
+            // survivors carry (selection key, score|doc); the exact reference
+            // order ((sort_value, sort_value2, GlobalDocId), sorting.md:14-26)
+            // is re-established here over the survivor set
+            size_t n = top_keys.size() / 2;
+            out.hits.reserve(n);
+            for (size_t i = 0; i < n; ++i) {
+                uint64_t aux = top_keys[2 * i + 1];
+                uint32_t doc = uint32_t(aux);
+                float score = 0.f;
+                if (fq.scoring) {
+                    uint32_t b = uint32_t(aux >> 32);
+                    memcpy(&score, &b, 4);
+                }
+                out.hits.push_back(mk_hit(doc, score));
+            }
+            auto cmp = [&](const pb::PartialHit& a, const pb::PartialHit& b) {
+                return hit_before(a, b, order1, order2);
+            };
+            size_t k = std::min<size_t>(leaf_max_hits, out.hits.size());
+            std::partial_sort(out.hits.begin(), out.hits.begin() + k,
+                              out.hits.end(), cmp);
+            out.hits.resize(k);
         } else {
-            bool asc = sp.by_score && sp.order1 == 0;
+            bool asc = !specs.empty() && order1 == 0;
             for (uint64_t key : top_keys) {
                 uint32_t kh = uint32_t(key >> 32), kl = uint32_t(key);
                 if (asc) {

@@ -154,3 +154,29 @@ def test_unknown_field_term(ragged):
     assert len(got.get("failed_splits", [])) == 1
     assert len(exp.get("failed_splits", [])) == 1
     assert "unknown field" in got["failed_splits"][0]["error"]
+
+
+@pytest.mark.parametrize("order", [0, 1])
+def test_sort_by_nullable_field_none_last(ragged, order):
+    # doc without tenant_id sorts LAST under either order (sorting.md None
+    # handling); exact id+value order must match the oracle
+    got, exp = ragged({"type": "match_all"}, max_hits=10, sort_fields=[
+        {"field_name": "tenant_id", "sort_order": order}])
+    def key(h):
+        v = h.get("sort_value", {})
+        return (hid(h), v.get("u64"), v.get("i64"), "kind" in v or None)
+    assert got.get("num_hits", 0) == exp.get("num_hits", 0)
+    assert [key(h) for h in got.get("partial_hits", [])] == \
+           [key(h) for h in exp.get("partial_hits", [])]
+
+
+@pytest.mark.parametrize("order", [0, 1])
+def test_sort_by_nullable_timestamp(ragged, order):
+    got, exp = ragged({"type": "term", "field": "severity_text",
+                       "value": "ERROR"}, max_hits=10, sort_fields=[
+        {"field_name": "timestamp", "sort_order": order}])
+    def key(h):
+        v = h.get("sort_value", {})
+        return (hid(h), v.get("u64"), v.get("i64"))
+    assert [key(h) for h in got.get("partial_hits", [])] == \
+           [key(h) for h in exp.get("partial_hits", [])]

@@ -344,3 +344,93 @@ def test_agg_golden_scenarios_on_gpu(case_name):
     resp = s.leaf_search(req)
     got = s.finalize_agg_json(resp["intermediate_aggregation_result"], aggs)
     approx_json(got, case["expected"], case_name)
+
+
+# ------------------------------------------------- sort by fast field (wide
+# candidate records; collector.rs:403-414 sort-key extraction + sorting.md
+# tie-breaks). Integer keys are deterministic => order must match EXACTLY.
+def sv_of(h):
+    v = h.get("sort_value", {})
+    return (v.get("u64"), v.get("i64"), v.get("f64"), v.get("boolean"))
+
+
+def assert_exact(got, exp):
+    assert got.get("num_hits", 0) == exp.get("num_hits", 0)
+    g, e = got.get("partial_hits", []), exp.get("partial_hits", [])
+    assert [(hid(h), sv_of(h)) for h in g] == [(hid(h), sv_of(h)) for h in e]
+
+
+TERM_Q = {"type": "term", "field": "severity_text", "value": "WARN"}
+
+
+@pytest.mark.parametrize("order", [0, 1])
+def test_sort_by_timestamp(searchers, order):
+    got, exp = run_both(searchers, TERM_Q, max_hits=25, sort_fields=[
+        {"field_name": "timestamp", "sort_order": order}])
+    assert_exact(got, exp)
+
+
+@pytest.mark.parametrize("order", [0, 1])
+def test_sort_by_timestamp_match_all(searchers, order):
+    got, exp = run_both(searchers, {"type": "match_all"}, max_hits=15,
+                        sort_fields=[{"field_name": "timestamp",
+                                      "sort_order": order}])
+    assert_exact(got, exp)
+
+
+def test_sort_by_u64_fast_field(searchers):
+    q = {"type": "bool",
+         "must": [{"type": "term", "field": "severity_text", "value": "ERROR"}]}
+    got, exp = run_both(searchers, q, max_hits=30, sort_fields=[
+        {"field_name": "tenant_id", "sort_order": 1}])
+    assert_exact(got, exp)
+
+
+def test_sort_by_str_ord(searchers):
+    got, exp = run_both(searchers, TERM_Q, max_hits=20, sort_fields=[
+        {"field_name": "tenant_name", "sort_order": 0}])
+    assert_exact(got, exp)
+
+
+def test_sort_by_unknown_field_is_none(searchers):
+    # unknown sort field: every key None -> GlobalDocId tie-break only
+    got, exp = run_both(searchers, TERM_Q, max_hits=10, sort_fields=[
+        {"field_name": "no_such_field", "sort_order": 1}])
+    assert_exact(got, exp)
+
+
+def test_sort_two_fields_ts_then_tenant(searchers):
+    # timestamps collide at second granularity -> sort_value2 + doc tie-breaks
+    got, exp = run_both(searchers, {"type": "match_all"}, max_hits=40,
+                        sort_fields=[
+                            {"field_name": "timestamp", "sort_order": 1},
+                            {"field_name": "tenant_id", "sort_order": 0}])
+    assert_exact(got, exp)
+
+
+def test_sort_field_then_score(searchers):
+    q = {"type": "bool", "should": [
+        {"type": "term", "field": "body", "value": "w%05d" % i}
+        for i in range(3)]}
+    got, exp = run_both(searchers, q, max_hits=20, sort_fields=[
+        {"field_name": "tenant_name", "sort_order": 1},
+        {"field_name": "_score", "sort_order": 1}])
+    # primary keys integer-exact; secondary f32 scores may tie within REL
+    assert got.get("num_hits", 0) == exp.get("num_hits", 0)
+    g, e = got.get("partial_hits", []), exp.get("partial_hits", [])
+    assert [hid(h) for h in g] == [hid(h) for h in e]
+    assert [sv_of(h) for h in g] == [sv_of(h) for h in e]
+
+
+def test_sort_score_then_timestamp(searchers):
+    q = {"type": "bool", "should": [
+        {"type": "term", "field": "body", "value": "w%05d" % i}
+        for i in range(3)]}
+    got, exp = run_both(searchers, q, max_hits=20, sort_fields=[
+        {"field_name": "_score", "sort_order": 1},
+        {"field_name": "timestamp", "sort_order": 1}])
+    g, e = got.get("partial_hits", []), exp.get("partial_hits", [])
+    assert len(g) == len(e)
+    for gh, eh in zip(g, e):
+        gs, es = hscore(gh), hscore(eh)
+        assert math.isclose(gs, es, rel_tol=REL, abs_tol=1e-9), (gs, es)
