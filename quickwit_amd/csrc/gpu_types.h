@@ -108,6 +108,11 @@ struct QueryDev {
     uint32_t sort_width;    // column width (1/2/4/8), sort_src>=2
     uint64_t sort_values_off;
     uint64_t sort_nulls_off;  // 0 = non-nullable (missing -> None, sorts last)
+    // 1 = straight-line agg path: agg[0] is an int_fast LDS histogram over a
+    // non-nullable 8B column with no sub-aggs (bucket bounds derived from the
+    // column min/max => no range check), optional agg[1] LDS terms over a
+    // non-nullable ord column. Descriptors hoisted to registers.
+    uint32_t agg_fast;
     uint64_t terms_off;     // scratch offsets of descriptor arrays
     uint64_t preds_off;
     uint64_t aggs_off;

@@ -54,6 +54,25 @@ __device__ __forceinline__ uint64_t extract_bits(const uint32_t* base, uint64_t 
     return v & ((1ull << w) - 1ull);  // w <= 32
 }
 
+// extract the adjacent pair (j, j+1) of w-bit fields in one 64-bit window
+// (valid for w <= 16: window holds >= 33 bits past any 32-bit alignment)
+__device__ __forceinline__ void extract_bits_pair(const uint32_t* base, uint32_t j,
+                                                  uint32_t w, uint32_t* g0,
+                                                  uint32_t* g1) {
+    if (w <= 16) {
+        uint64_t bitpos = uint64_t(j) * w;
+        uint64_t word = bitpos >> 5;
+        uint32_t sh = uint32_t(bitpos & 31);
+        uint64_t v = (uint64_t(base[word]) | (uint64_t(base[word + 1]) << 32)) >> sh;
+        uint32_t mask = (1u << w) - 1u;
+        *g0 = uint32_t(v) & mask;
+        *g1 = uint32_t(v >> w) & mask;
+    } else {
+        *g0 = uint32_t(extract_bits(base, uint64_t(j) * w, w));
+        *g1 = uint32_t(extract_bits(base, uint64_t(j + 1) * w, w));
+    }
+}
+
 __device__ __forceinline__ uint32_t wave_incl_scan_u32(uint32_t v) {
     #pragma unroll
     for (int d = 1; d < 64; d <<= 1) {
@@ -123,8 +142,8 @@ __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t t
         uint32_t w = e.id_bits;
         // lane handles elements 2*lane and 2*lane+1
         uint32_t j0 = 2 * lane, j1 = 2 * lane + 1;
-        uint32_t g0 = uint32_t(extract_bits(idbase, uint64_t(j0) * w, w));
-        uint32_t g1 = uint32_t(extract_bits(idbase, uint64_t(j1) * w, w));
+        uint32_t g0, g1;
+        extract_bits_pair(idbase, j0, w, &g0, &g1);
         uint32_t incl = wave_incl_scan_u32(g0 + g1);
         uint32_t doc1 = e.first_doc + incl;
         uint32_t doc0 = doc1 - g1;
@@ -134,10 +153,9 @@ __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t t
             if (e.tf_bits) {
                 const uint32_t* tfbase =
                     idbase + 2 * ((128u * e.id_bits + 63u) / 64u);
-                tf0 = uint32_t(extract_bits(tfbase, uint64_t(j0) * e.tf_bits,
-                                            e.tf_bits)) + 1u;
-                tf1 = uint32_t(extract_bits(tfbase, uint64_t(j1) * e.tf_bits,
-                                            e.tf_bits)) + 1u;
+                extract_bits_pair(tfbase, j0, e.tf_bits, &tf0, &tf1);
+                ++tf0;
+                ++tf1;
             }
             // BM25: W * tf / (tf + K[normid]) — same op order as the oracle
             float K0 = ktab[norms ? norms[min(doc0, q.num_docs - 1)] : 1];
@@ -299,6 +317,26 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
         }
         if (threadIdx.x < 4) sc_agg_matched[threadIdx.x] = 0;
     }
+    // agg_fast: hoist the (uniform) descriptor fields into registers once
+    const int64_t* af_col = nullptr;
+    const uint8_t* af_tcol = nullptr;
+    int64_t af_ioff = 0, af_ivl = 1, af_base = 0;
+    double af_inv = 0;
+    uint32_t af_twidth = 0;
+    bool af_terms = false;
+    if (NA && q.agg_fast) {
+        const AggDev& a0 = aggs[0];
+        af_col = (const int64_t*)(q.split + a0.values_off);
+        af_ioff = a0.i_offset;
+        af_ivl = a0.i_interval;
+        af_inv = a0.inv_interval;
+        af_base = a0.base_index;
+        if (q.n_aggs > 1) {
+            af_terms = true;
+            af_tcol = q.split + aggs[1].values_off;
+            af_twidth = aggs[1].value_width;
+        }
+    }
 
     for (uint32_t tile = tile_base + blockIdx.x; tile < tile_end; tile += gridDim.x) {
         uint32_t tile_lo = tile * TILE_DOCS;
@@ -386,7 +424,24 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
 
             if (m) {
                 ++local_count;
-                if (NA)
+                if (NA && q.agg_fast) {
+                    // straight-line: exact floor-div histogram bucket (bounds
+                    // cover the column's [min,max] => no range check) +
+                    // optional ord-count table, both LDS
+                    int64_t num = af_col[d] - af_ioff;
+                    int64_t idx = int64_t(floor(double(num) * af_inv));
+                    if (idx * af_ivl > num) --idx;
+                    else if ((idx + 1) * af_ivl <= num) ++idx;
+                    atomicAdd(&sc_agg_hist[uint32_t(idx - af_base)], 1u);
+                    if (af_terms) {
+                        uint64_t o = af_twidth == 2
+                                         ? ((const uint16_t*)af_tcol)[d]
+                                         : (af_twidth == 1
+                                                ? af_tcol[d]
+                                                : ((const uint32_t*)af_tcol)[d]);
+                        atomicAdd(&sc_agg_terms[o], 1u);
+                    }
+                } else if (NA)
                     for (uint32_t ai = 0; ai < q.n_aggs; ++ai) {
                         const AggDev& a = aggs[ai];
                         if (a.nulls_off) {
@@ -401,11 +456,16 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                             else
                                 atomicAdd((unsigned long long*)(q.results + a.counts_out) + o,
                                           1ull);
-                            if (ai < 4)  // per-tile LDS count, flushed below
-                                atomicAdd(&sc_agg_matched[ai], 1u);
-                            else
-                                atomicAdd((unsigned long long*)(q.results + a.matched_out),
-                                          1ull);
+                            // docs-with-value == matched docs for a
+                            // non-nullable column: host uses num_hits instead
+                            if (a.nulls_off) {
+                                if (ai < 4)  // LDS count, flushed at the end
+                                    atomicAdd(&sc_agg_matched[ai], 1u);
+                                else
+                                    atomicAdd((unsigned long long*)(q.results +
+                                                                    a.matched_out),
+                                              1ull);
+                            }
                         } else {
                             int64_t idx = histo_bucket(q, a, d);
                             if (idx < 0 || idx >= int64_t(a.n_buckets)) continue;

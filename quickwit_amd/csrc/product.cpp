@@ -808,6 +808,16 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         // agg workloads: capped grid so the once-per-workgroup LDS flush
         // stays cheap (each WG owns several tiles); otherwise one WG per tile
         uint32_t grid = na ? std::min<uint32_t>(n_tiles, 2048) : n_tiles;
+        // straight-line agg path (kernels.hip): histo[0] int_fast LDS
+        // non-nullable no-subs (+ optional terms[1] LDS non-nullable)
+        if (na && !ap.devs.empty() && ap.devs[0].kind == AGGD_HISTO &&
+            ap.devs[0].lds_slot == 0 && ap.devs[0].int_fast &&
+            ap.devs[0].n_sub == 0 && ap.devs[0].nulls_off == 0 &&
+            ap.devs.size() <= 2 &&
+            (ap.devs.size() < 2 ||
+             (ap.devs[1].kind == AGGD_TERMS && ap.devs[1].lds_slot == 1 &&
+              ap.devs[1].nulls_off == 0 && ap.devs[1].n_buckets > 0)))
+            q.agg_fast = 1;
         HIP_CHECK(hipEventRecord(ctx->ev_start, ctx->stream));
         launch_leaf_tile(ns, nb, na, collect, dim3(grid), ctx->stream, q, 0u,
                          n_tiles, 1u);
@@ -995,9 +1005,14 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             if (d.kind == AggDef::TERMS) {
                 r.kind = 3;
                 if (a.n_buckets) {
-                    uint64_t m = 0;
-                    memcpy(&m, agg_out.data() + (a.matched_out - r_agg), 8);
-                    r.terms_matched_docs = m;
+                    if (a.nulls_off) {
+                        uint64_t m = 0;
+                        memcpy(&m, agg_out.data() + (a.matched_out - r_agg), 8);
+                        r.terms_matched_docs = m;
+                    } else {
+                        // non-nullable column: every matched doc has a value
+                        r.terms_matched_docs = matched;
+                    }
                     for (uint32_t o = 0; o < a.n_buckets; ++o)
                         if (counts[o])
                             r.term_counts.emplace_back(f->dict_entry(o), counts[o]);
