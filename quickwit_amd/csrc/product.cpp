@@ -1054,6 +1054,97 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
     return out;
 }
 
+// ------------------------------------------------------------ pruning
+// Restates CanSplitDoBetter (leaf.rs:1337-1553), simplify_search_request
+// (leaf.rs:1666-1710) and the request-level part of rewrite_request
+// (leaf.rs:977-990, remove_redundant_timestamp_range :1106-1211). The
+// FindTraceIdsAggregation variant (Jaeger traces) is out of scope.
+// Missing split timestamps read as 0 (prost's Option accessor default, the
+// exact behavior of split.timestamp_start()/timestamp_end() in the reference).
+static int64_t split_ts_start(const pb::SplitIdAndFooterOffsets& s) {
+    return s.timestamp_start ? *s.timestamp_start : 0;
+}
+static int64_t split_ts_end(const pb::SplitIdAndFooterOffsets& s) {
+    return s.timestamp_end ? *s.timestamp_end : 0;
+}
+
+static int64_t div_ceil_i64(int64_t lhs, int64_t rhs) {  // quickwit-common lib.rs:210
+    int64_t d = lhs / rhs, r = lhs % rhs;
+    return ((r > 0 && rhs > 0) || (r < 0 && rhs < 0)) ? d + 1 : d;
+}
+
+struct SplitFilter {
+    enum Kind { UNINFORMATIVE, SPLIT_ID_HIGHER, TS_HIGHER, TS_LOWER } kind =
+        UNINFORMATIVE;
+    bool has = false;     // a worst-of-top-K hit has been recorded
+    std::string wid;      // SPLIT_ID_HIGHER: worst split id
+    int64_t wts = 0;      // TS_*: worst timestamp in seconds
+
+    static SplitFilter from_request(const pb::SearchRequest& req,
+                                    const std::string& ts_field) {
+        SplitFilter f;
+        if (req.sort_fields.empty()) {
+            f.kind = SPLIT_ID_HIGHER;
+        } else if (!ts_field.empty() && req.sort_fields[0].field_name == ts_field) {
+            f.kind = req.sort_fields[0].sort_order == 1 ? TS_HIGHER : TS_LOWER;
+        }
+        return f;
+    }
+
+    bool can_be_better(const pb::SplitIdAndFooterOffsets& s) const {
+        if (!has) return true;
+        switch (kind) {
+            case SPLIT_ID_HIGHER: return s.split_id >= wid;
+            case TS_HIGHER: return split_ts_end(s) >= wts;
+            case TS_LOWER: return split_ts_start(s) <= wts;
+            default: return true;
+        }
+    }
+
+    // record the worst of the (now full) top-K (leaf.rs:1516-1542)
+    void record_new_worst_hit(const pb::PartialHit& h) {
+        switch (kind) {
+            case SPLIT_ID_HIGHER:
+                has = true;
+                wid = h.split_id;
+                break;
+            case TS_HIGHER:
+                if (h.sort_value.kind == pb::SortByValue::I64) {
+                    has = true;
+                    wts = div_ceil_i64(h.sort_value.i64, 1000000000);  // round UP
+                }
+                break;
+            case TS_LOWER:
+                if (h.sort_value.kind == pb::SortByValue::I64) {
+                    has = true;
+                    wts = h.sort_value.i64 / 1000000000;  // truncate (Rust /)
+                }
+                break;
+            default: break;
+        }
+    }
+};
+
+static bool is_simple_all_query(const pb::SearchRequest& r) {  // leaf.rs:1315
+    if (r.aggregation_request) return false;
+    if (r.search_after) return false;
+    if (r.start_timestamp || r.end_timestamp) return false;
+    try {
+        mj::ValuePtr v = mj::parse(r.query_ast);
+        const mj::Value* t = v->get("type");
+        return t && t->kind == mj::Value::STR && t->s == "match_all";
+    } catch (...) {
+        return false;
+    }
+}
+
+static void disable_search_request_hits(pb::SearchRequest& r) {  // leaf.rs:1705
+    r.max_hits = 0;
+    r.start_offset = 0;
+    r.sort_fields.clear();
+    r.search_after.reset();
+}
+
 }  // namespace qw
 
 // ============================================================== C ABI
@@ -1157,7 +1248,12 @@ int32_t qw_leaf_search(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
         if (lreq.doc_mappers.empty()) throw std::runtime_error("missing doc_mapper");
         Schema schema = Schema::parse(lreq.doc_mappers[0]);
 
-        std::vector<std::pair<const DeviceSplit*, std::string>> tasks;
+        struct Task {
+            pb::SplitIdAndFooterOffsets so;
+            const DeviceSplit* ds;
+            pb::SearchRequest req;  // per-split request (may be demoted)
+        };
+        std::vector<Task> tasks;
         for (auto& lr : lreq.leaf_requests)
             for (auto& so : lr.split_offsets) {
                 auto it = ctx->splits.find(so.split_id);
@@ -1165,26 +1261,102 @@ int32_t qw_leaf_search(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
                     set_err(ctx, "unknown split: " + so.split_id);
                     return QW_ERR_NOT_FOUND;
                 }
-                tasks.push_back({it->second.get(), so.split_id});
+                tasks.push_back({so, it->second.get(), req});
             }
 
+        // ---- pruning: split ordering + upfront count-only demotion
+        // (CanSplitDoBetter::optimize, leaf.rs:1403-1514)
+        SplitFilter filter = SplitFilter::from_request(req, schema.timestamp_field);
+        switch (filter.kind) {
+            case SplitFilter::SPLIT_ID_HIGHER:
+                std::sort(tasks.begin(), tasks.end(), [](const Task& a, const Task& b) {
+                    return b.so.split_id < a.so.split_id;
+                });
+                break;
+            case SplitFilter::TS_HIGHER:
+                std::sort(tasks.begin(), tasks.end(), [](const Task& a, const Task& b) {
+                    return split_ts_end(b.so) < split_ts_end(a.so);
+                });
+                break;
+            case SplitFilter::TS_LOWER:
+                std::sort(tasks.begin(), tasks.end(), [](const Task& a, const Task& b) {
+                    return split_ts_start(a.so) < split_ts_start(b.so);
+                });
+                break;
+            default: break;
+        }
+        if (is_simple_all_query(req)) {
+            uint64_t num_requested = req.start_offset + req.max_hits;
+            size_t mrs = 0;  // number of splits guaranteed to fill the top-K
+            uint64_t psum = 0;
+            for (const Task& t : tasks) {
+                psum += t.so.num_docs;
+                if (psum >= num_requested) break;
+                ++mrs;
+            }
+            mrs += 1;
+            if (filter.kind == SplitFilter::SPLIT_ID_HIGHER) {
+                for (size_t i = mrs; i < tasks.size(); ++i)
+                    disable_search_request_hits(tasks[i].req);
+            } else if (filter.kind == SplitFilter::TS_LOWER) {
+                int64_t biggest_end = INT64_MIN;
+                for (size_t i = 0; i < std::min(mrs, tasks.size()); ++i)
+                    biggest_end = std::max(biggest_end, split_ts_end(tasks[i].so));
+                for (size_t i = mrs; i < tasks.size(); ++i)
+                    if (split_ts_start(tasks[i].so) > biggest_end)
+                        disable_search_request_hits(tasks[i].req);
+            } else if (filter.kind == SplitFilter::TS_HIGHER) {
+                int64_t smallest_start = INT64_MAX;
+                for (size_t i = 0; i < std::min(mrs, tasks.size()); ++i)
+                    smallest_start = std::min(smallest_start, split_ts_start(tasks[i].so));
+                for (size_t i = mrs; i < tasks.size(); ++i)
+                    if (split_ts_end(tasks[i].so) < smallest_start)
+                        disable_search_request_hits(tasks[i].req);
+            }
+        }
+        // per-split rewrite (leaf.rs:977-990): drop sort fields on count-only
+        // requests; drop request-level ts bounds covered by the split range
+        for (Task& t : tasks) {
+            if (t.req.max_hits == 0) t.req.sort_fields.clear();
+            if (!schema.timestamp_field.empty()) {
+                if (t.req.start_timestamp && t.so.timestamp_start &&
+                    *t.req.start_timestamp <= *t.so.timestamp_start)
+                    t.req.start_timestamp.reset();
+                if (t.req.end_timestamp && t.so.timestamp_end &&
+                    *t.req.end_timestamp >= *t.so.timestamp_end + 1)
+                    t.req.end_timestamp.reset();
+            }
+        }
+
         pb::LeafSearchResponse resp;
-        resp.num_attempted_splits = tasks.size();
         IntermediateAggResults merged_aggs;
         bool any_aggs = false;
         std::vector<pb::PartialHit> all_hits;
         pb::LeafResourceStats rstats;
         rstats.search_pool_cpu_threads = 1;  // one host thread per GPU
         uint64_t worst = 0;
-        for (auto& [ds, sid] : tasks) {
+        int order1 = req.sort_fields.empty() ? 1 : req.sort_fields[0].sort_order;
+        int order2 = req.sort_fields.size() > 1 ? req.sort_fields[1].sort_order : 1;
+        auto cmp = [&](const pb::PartialHit& a, const pb::PartialHit& b) {
+            return hit_before(a, b, order1, order2);
+        };
+        uint64_t leaf_max = req.max_hits + req.start_offset;
+        for (Task& t : tasks) {
+            // second simplify pass against the live filter state
+            // (simplify_search_request, leaf.rs:1666-1703)
+            if (!filter.can_be_better(t.so)) disable_search_request_hits(t.req);
+            if (t.req.max_hits == 0 && !t.req.aggregation_request &&
+                t.req.count_hits != 0 /* != CountHits::CountAll */)
+                continue;  // pruned before warmup — not attempted
+            resp.num_attempted_splits++;
             SplitResult r;
             try {
-                r = search_split_gpu(ctx, *ds, req, schema);
+                r = search_split_gpu(ctx, *t.ds, t.req, schema);
             } catch (const std::exception& e) {
                 // per-split failure is data, not an exception (leaf.rs:2143)
                 pb::SplitSearchError se;
                 se.error = e.what();
-                se.split_id = sid;
+                se.split_id = t.so.split_id;
                 se.retryable_error = true;
                 resp.failed_splits.push_back(std::move(se));
                 continue;
@@ -1198,8 +1370,17 @@ int32_t qw_leaf_search(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
                     any_aggs = true;
                 } else merged_aggs.merge(r.aggs);
             }
+            // incremental top-K + worst-hit feedback (leaf.rs:2310-2317)
+            if (leaf_max > 0 && all_hits.size() >= leaf_max) {
+                size_t k = std::min<size_t>(leaf_max, all_hits.size());
+                std::partial_sort(all_hits.begin(), all_hits.begin() + k,
+                                  all_hits.end(), cmp);
+                all_hits.resize(k);
+                if (all_hits.size() == leaf_max)
+                    filter.record_new_worst_hit(all_hits.back());
+            }
             pb::SplitResourceStats ss;
-            ss.split_num_docs = ds->view.num_docs;
+            ss.split_num_docs = t.ds->view.num_docs;
             ss.matched_num_docs = r.num_hits;
             ss.cpu_search_microsecs = r.micros;  // GPU wall (cpu_search analog)
             rstats.localexec_num_splits++;
@@ -1212,12 +1393,7 @@ int32_t qw_leaf_search(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
                 rstats.split_resources_worst = ss;
             }
         }
-        int order1 = req.sort_fields.empty() ? 1 : req.sort_fields[0].sort_order;
-        int order2 = req.sort_fields.size() > 1 ? req.sort_fields[1].sort_order : 1;
-        auto cmp = [&](const pb::PartialHit& a, const pb::PartialHit& b) {
-            return hit_before(a, b, order1, order2);
-        };
-        size_t k = std::min<size_t>(req.max_hits + req.start_offset, all_hits.size());
+        size_t k = std::min<size_t>(leaf_max, all_hits.size());
         std::partial_sort(all_hits.begin(), all_hits.begin() + k, all_hits.end(), cmp);
         all_hits.resize(k);
         resp.partial_hits = std::move(all_hits);
