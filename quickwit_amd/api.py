@@ -170,6 +170,9 @@ def make_leaf_request(query_ast: dict | str, schema: dict, splits: list,
         sreq["end_timestamp"] = end_timestamp
     offsets = []
     for s in splits:
+        if isinstance(s, dict):  # full SplitIdAndFooterOffsets (ts metadata
+            offsets.append(s)    # drives CanSplitDoBetter pruning)
+            continue
         sid, ndocs = s[0], s[1]
         e = {"split_id": sid, "num_docs": ndocs}
         if len(s) > 2:

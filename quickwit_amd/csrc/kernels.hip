@@ -133,52 +133,93 @@ __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t t
         (const float*)(q.scratch + q.ktabs_off) + 256 * t.ktab_idx;
     uint32_t wave = threadIdx.x >> 6;
     uint32_t lane = lane_id();
-    // decode one 128-doc block within a wave: funnel-shift gap extraction, a
-    // 64-lane shuffle prefix-sum for the deltas, fused BM25 FMA
-    auto do_block = [&](uint32_t blk) {
-        SkipEntryDev e = skip[blk];
-        if (e.first_doc >= tile_hi || e.last_doc < tile_lo) return;
+    // one 128-doc block per 32-lane HALF-wave (4 elements/sub-lane): twice
+    // the blocks in flight per wave and a 5-step scan instead of 6 — the
+    // dependent shuffle chain is the decode critical path
+    uint32_t half = lane >> 5;   // which block of the pair this lane works on
+    uint32_t sl = lane & 31;     // sub-lane within the half-wave
+    for (uint32_t blk0 = blo + 2 * wave; blk0 < bhi; blk0 += 2 * (TILE_THREADS / 64)) {
+        uint32_t blk = blk0 + half;
+        bool live = blk < bhi;
+        SkipEntryDev e = live ? skip[blk] : SkipEntryDev{};
+        live = live && !(e.first_doc >= tile_hi || e.last_doc < tile_lo);
+        uint32_t g0 = 0, g1 = 0, g2 = 0, g3 = 0;
         const uint32_t* idbase = payload + e.word_off;
         uint32_t w = e.id_bits;
-        // lane handles elements 2*lane and 2*lane+1
-        uint32_t j0 = 2 * lane, j1 = 2 * lane + 1;
-        uint32_t g0, g1;
-        extract_bits_pair(idbase, j0, w, &g0, &g1);
-        uint32_t incl = wave_incl_scan_u32(g0 + g1);
-        uint32_t doc1 = e.first_doc + incl;
+        uint32_t j0 = 4 * sl;
+        if (live) {
+            if (w <= 8) {
+                // 4 fields fit one 64-bit window past any 32-bit alignment
+                uint64_t bitpos = uint64_t(j0) * w;
+                uint32_t sh = uint32_t(bitpos & 31);
+                uint64_t word = bitpos >> 5;
+                uint64_t v = (uint64_t(idbase[word]) |
+                              (uint64_t(idbase[word + 1]) << 32)) >> sh;
+                uint32_t mask = (1u << w) - 1u;
+                g0 = uint32_t(v) & mask;
+                g1 = uint32_t(v >> w) & mask;
+                g2 = uint32_t(v >> (2 * w)) & mask;
+                g3 = uint32_t(v >> (3 * w)) & mask;
+            } else {
+                extract_bits_pair(idbase, j0, w, &g0, &g1);
+                extract_bits_pair(idbase, j0 + 2, w, &g2, &g3);
+            }
+        }
+        // inclusive 32-lane scan of the per-lane gap sums
+        uint32_t sum = g0 + g1 + g2 + g3;
+        #pragma unroll
+        for (int d = 1; d < 32; d <<= 1) {
+            uint32_t n = __shfl_up(sum, d, 32);
+            if (sl >= uint32_t(d)) sum += n;
+        }
+        if (!live) continue;
+        uint32_t doc3 = e.first_doc + sum;
+        uint32_t doc2 = doc3 - g3;
+        uint32_t doc1 = doc2 - g2;
         uint32_t doc0 = doc1 - g1;
-        float s0 = 1.0f, s1 = 1.0f;
+        float s0 = 1.f, s1 = 1.f, s2 = 1.f, s3 = 1.f;
         if (SCORE && scoring) {
-            uint32_t tf0 = 1, tf1 = 1;
+            uint32_t tf0 = 1, tf1 = 1, tf2 = 1, tf3 = 1;
             if (e.tf_bits) {
-                const uint32_t* tfbase =
-                    idbase + 2 * ((128u * e.id_bits + 63u) / 64u);
-                extract_bits_pair(tfbase, j0, e.tf_bits, &tf0, &tf1);
-                ++tf0;
-                ++tf1;
+                const uint32_t* tfbase = idbase + 2 * ((128u * e.id_bits + 63u) / 64u);
+                uint32_t tw = e.tf_bits;
+                if (tw <= 8) {
+                    uint64_t bitpos = uint64_t(j0) * tw;
+                    uint32_t sh = uint32_t(bitpos & 31);
+                    uint64_t word = bitpos >> 5;
+                    uint64_t v = (uint64_t(tfbase[word]) |
+                                  (uint64_t(tfbase[word + 1]) << 32)) >> sh;
+                    uint32_t mask = (1u << tw) - 1u;
+                    tf0 = (uint32_t(v) & mask) + 1u;
+                    tf1 = (uint32_t(v >> tw) & mask) + 1u;
+                    tf2 = (uint32_t(v >> (2 * tw)) & mask) + 1u;
+                    tf3 = (uint32_t(v >> (3 * tw)) & mask) + 1u;
+                } else {
+                    extract_bits_pair(tfbase, j0, tw, &tf0, &tf1);
+                    extract_bits_pair(tfbase, j0 + 2, tw, &tf2, &tf3);
+                    ++tf0; ++tf1; ++tf2; ++tf3;
+                }
             }
             // BM25: W * tf / (tf + K[normid]) — same op order as the oracle
             float K0 = ktab[norms ? norms[min(doc0, q.num_docs - 1)] : 1];
             float K1 = ktab[norms ? norms[min(doc1, q.num_docs - 1)] : 1];
+            float K2 = ktab[norms ? norms[min(doc2, q.num_docs - 1)] : 1];
+            float K3 = ktab[norms ? norms[min(doc3, q.num_docs - 1)] : 1];
             s0 = t.weight * (float(tf0) / (float(tf0) + K0));
             s1 = t.weight * (float(tf1) / (float(tf1) + K1));
+            s2 = t.weight * (float(tf2) / (float(tf2) + K2));
+            s3 = t.weight * (float(tf3) / (float(tf3) + K3));
         }
-        if (j0 < e.count && doc0 >= tile_lo && doc0 < tile_hi) {
-            uint32_t li = doc0 - tile_lo;
-            if (SCORE) atomicAdd(&score[li], s0);
-            else atomicOr(&bitset[li >> 5], 1u << (li & 31));
+        #pragma unroll
+        for (int el = 0; el < 4; ++el) {
+            uint32_t doc = el == 0 ? doc0 : el == 1 ? doc1 : el == 2 ? doc2 : doc3;
+            float s = el == 0 ? s0 : el == 1 ? s1 : el == 2 ? s2 : s3;
+            if (j0 + el < e.count && doc >= tile_lo && doc < tile_hi) {
+                uint32_t li = doc - tile_lo;
+                if (SCORE) atomicAdd(&score[li], s);
+                else atomicOr(&bitset[li >> 5], 1u << (li & 31));
+            }
         }
-        if (j1 < e.count && doc1 >= tile_lo && doc1 < tile_hi) {
-            uint32_t li = doc1 - tile_lo;
-            if (SCORE) atomicAdd(&score[li], s1);
-            else atomicOr(&bitset[li >> 5], 1u << (li & 31));
-        }
-    };
-    // two blocks per wave-iteration: the prefix-scan chains of the pair are
-    // independent, so their shuffle/latency chains overlap (ILP)
-    for (uint32_t blk = blo + 2 * wave; blk < bhi; blk += 2 * (TILE_THREADS / 64)) {
-        do_block(blk);
-        if (blk + 1 < bhi) do_block(blk + 1);
     }
 }
 
@@ -395,6 +436,28 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                                                 nullptr, sc_bits_not, false);
                 __syncthreads();
             }
+        }
+
+        // ---- pure-aggregation tile (match_all, no predicates, no
+        // collection): tight loop, every doc matches, count is the tile size
+        if (NA && !NS && !NB && !NC && q.agg_fast && q.match_all && !q.n_preds) {
+            #pragma unroll 4
+            for (uint32_t d = tile_lo + threadIdx.x; d < tile_hi; d += TILE_THREADS) {
+                int64_t num = af_col[d] - af_ioff;
+                int64_t idx = int64_t(floor(double(num) * af_inv));
+                if (idx * af_ivl > num) --idx;
+                else if ((idx + 1) * af_ivl <= num) ++idx;
+                atomicAdd(&sc_agg_hist[uint32_t(idx - af_base)], 1u);
+                if (af_terms) {
+                    uint64_t o = af_twidth == 2 ? ((const uint16_t*)af_tcol)[d]
+                                 : (af_twidth == 1 ? af_tcol[d]
+                                                   : ((const uint32_t*)af_tcol)[d]);
+                    atomicAdd(&sc_agg_terms[o], 1u);
+                }
+            }
+            if (do_count && threadIdx.x == 0) tile_counts[tile] = tile_hi - tile_lo;
+            __syncthreads();
+            continue;
         }
 
         // ---- epilogue: matched test, count, candidates, aggregations

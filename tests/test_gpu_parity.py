@@ -434,3 +434,71 @@ def test_sort_score_then_timestamp(searchers):
     for gh, eh in zip(g, e):
         gs, es = hscore(gh), hscore(eh)
         assert math.isclose(gs, es, rel_tol=REL, abs_tol=1e-9), (gs, es)
+
+
+# ------------------------------------------------- multi-split + pruning
+# (CanSplitDoBetter, leaf.rs:1337-1553). Pruning must be invisible in the
+# response: the oracle searches every split un-pruned; identical results.
+@pytest.fixture(scope="module")
+def multi():
+    gpu, cpu = GpuSearcher(device=0), OracleSearcher()
+    metas = []
+    for i in range(3):
+        nd = 4000 + 700 * i
+        data = splitgen.generate_split(i, nd, seed=7)
+        sid = f"synthetic-7-{i:04d}"
+        gpu.add_split(sid, data)
+        cpu.add_split(sid, data)
+        # split time ranges from the generated data (seconds, inclusive)
+        from quickwit_amd import splitread
+        sp = splitread.Split(data)
+        ts, _ = sp.fast_column("timestamp")
+        metas.append({"split_id": sid, "num_docs": nd,
+                      "timestamp_start": int(ts.min() // 1000),
+                      "timestamp_end": int(ts.max() // 1000)})
+
+    def run(query, **kw):
+        req = make_leaf_request(query, SCHEMA, metas, **kw)
+        return gpu.leaf_search(req), cpu.leaf_search(req)
+    return run
+
+
+def test_multi_split_default_order(multi):
+    # no sort fields -> SplitIdHigher: later (lower-id) splits demoted to
+    # count-only once the top-K is full; hits + num_hits must be unchanged
+    got, exp = multi({"type": "term", "field": "severity_text",
+                      "value": "INFO"}, max_hits=7)
+    assert got.get("num_hits", 0) == exp.get("num_hits", 0)
+    assert [hid(h) for h in got.get("partial_hits", [])] == \
+           [hid(h) for h in exp.get("partial_hits", [])]
+
+
+def test_multi_split_match_all_upfront_demotion(multi):
+    # simple-all query -> optimize() demotes all but min_required_splits
+    got, exp = multi({"type": "match_all"}, max_hits=5)
+    assert got.get("num_hits", 0) == exp.get("num_hits", 0)
+    assert [hid(h) for h in got.get("partial_hits", [])] == \
+           [hid(h) for h in exp.get("partial_hits", [])]
+
+
+@pytest.mark.parametrize("order", [0, 1])
+def test_multi_split_sort_timestamp_pruning(multi, order):
+    # SplitTimestampHigher/Lower with second-granularity worst-hit feedback
+    got, exp = multi({"type": "term", "field": "severity_text",
+                      "value": "ERROR"}, max_hits=9, sort_fields=[
+        {"field_name": "timestamp", "sort_order": order}])
+    assert got.get("num_hits", 0) == exp.get("num_hits", 0)
+    g = [(hid(h), sv_of(h)) for h in got.get("partial_hits", [])]
+    e = [(hid(h), sv_of(h)) for h in exp.get("partial_hits", [])]
+    assert g == e
+
+
+def test_multi_split_ts_range_rewrite(multi):
+    # request-level ts bounds covered by a split's range are dropped per
+    # split (remove_redundant_timestamp_range); results must be identical
+    got, exp = multi({"type": "term", "field": "severity_text",
+                      "value": "WARN"}, max_hits=10,
+                     start_timestamp=1, end_timestamp=2**31)
+    assert got.get("num_hits", 0) == exp.get("num_hits", 0)
+    assert [hid(h) for h in got.get("partial_hits", [])] == \
+           [hid(h) for h in exp.get("partial_hits", [])]
