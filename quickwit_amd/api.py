@@ -110,6 +110,10 @@ class GpuSearcher(_BaseSearcher):
     def __init__(self, device: int = 0, config: dict | None = None):
         cfg = dict(config or {})
         cfg.setdefault("device", device)
+        # benches/tests run the partial-result cache COLD by default
+        # (SURVEY §8a "bypass for benchmarking"); the C-ABI default mirrors
+        # the reference's SearcherConfig 64 MB. Pass a capacity to enable.
+        cfg.setdefault("partial_request_cache_capacity", 0)
         super().__init__(create_arg=json.dumps(cfg).encode())
 
     def finalize_agg_json(self, blob: bytes, agg_request: dict) -> dict:

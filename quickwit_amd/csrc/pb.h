@@ -418,6 +418,7 @@ struct SplitResourceStats {  // search.proto:383
 };
 
 struct LeafResourceStats {  // search.proto:425
+    uint64_t partial_result_cache_num_splits = 0, partial_result_cache_num_docs = 0;
     uint64_t localexec_num_splits = 0, localexec_num_docs = 0;
     SplitResourceStats split_resources_worst, split_resources_sum;
     std::optional<uint64_t> min_wait_for_search_permit_microsecs,
@@ -426,6 +427,8 @@ struct LeafResourceStats {  // search.proto:425
     uint64_t search_pool_cpu_threads = 0;
     std::string encode() const {
         Writer w;
+        w.u64_field(1, partial_result_cache_num_splits);
+        w.u64_field(2, partial_result_cache_num_docs);
         w.u64_field(3, localexec_num_splits);
         w.u64_field(4, localexec_num_docs);
         w.msg_field(5, split_resources_worst.encode());
@@ -442,7 +445,9 @@ struct LeafResourceStats {  // search.proto:425
         LeafResourceStats s;
         while (!r.done()) {
             uint32_t no, wt = r.read_tag(&no);
-            if (no == 3 && wt == 0) s.localexec_num_splits = r.varint();
+            if (no == 1 && wt == 0) s.partial_result_cache_num_splits = r.varint();
+            else if (no == 2 && wt == 0) s.partial_result_cache_num_docs = r.varint();
+            else if (no == 3 && wt == 0) s.localexec_num_splits = r.varint();
             else if (no == 4 && wt == 0) s.localexec_num_docs = r.varint();
             else if (no == 5 && wt == 2) {
                 std::string b = r.bytes();

@@ -24,6 +24,7 @@
 #include <cstdint>
 #include <cstdlib>
 #include <cstring>
+#include <list>
 #include <map>
 #include <memory>
 #include <stdexcept>
@@ -80,6 +81,112 @@ struct DevBuf {  // grow-only device scratch
     }
 };
 
+// LeafSearchCache restatement (quickwit-search/src/leaf_cache.rs): memoizes
+// per-(split, canonical request) LeafSearchResponse bytes in a byte-capacity
+// LRU (MemorySizedCache semantics). The key canonicalizes the request by
+// clearing the timestamp bounds and forcing count_hits = COUNT_ALL
+// (leaf_cache.rs:86-115) and appends the merged (request ∩ split) half-open
+// time range (leaf_cache.rs:118-180).
+struct LeafCache {
+    size_t capacity = 64ull << 20;  // SearcherConfig partial_request_cache_capacity
+    size_t used = 0;
+    std::list<std::pair<std::string, std::string>> lru;  // front = most recent
+    std::map<std::string, std::list<std::pair<std::string, std::string>>::iterator> idx;
+
+    bool get(const std::string& key, std::string* out) {
+        auto it = idx.find(key);
+        if (it == idx.end()) return false;
+        lru.splice(lru.begin(), lru, it->second);
+        *out = it->second->second;
+        return true;
+    }
+    void put(const std::string& key, std::string val) {
+        if (!capacity || key.size() + val.size() > capacity) return;
+        auto it = idx.find(key);
+        if (it != idx.end()) {
+            used -= it->second->first.size() + it->second->second.size();
+            lru.erase(it->second);
+            idx.erase(it);
+        }
+        used += key.size() + val.size();
+        lru.emplace_front(key, std::move(val));
+        idx[key] = lru.begin();
+        while (used > capacity && !lru.empty()) {
+            auto& back = lru.back();
+            used -= back.first.size() + back.second.size();
+            idx.erase(back.first);
+            lru.pop_back();
+        }
+    }
+    void remove_split(const std::string& split_id) {
+        std::string prefix = split_id + '\0';
+        for (auto it = lru.begin(); it != lru.end();) {
+            if (it->first.compare(0, prefix.size(), prefix) == 0) {
+                used -= it->first.size() + it->second.size();
+                idx.erase(it->first);
+                it = lru.erase(it);
+            } else ++it;
+        }
+    }
+};
+
+// half-open [start, end) second range; has_end=false means unbounded
+// (leaf_cache.rs HalfOpenRange, incl. the empty-range normalization)
+struct HalfOpen {
+    int64_t start = INT64_MIN;
+    bool has_end = false;
+    int64_t end = 0;
+    void normalize() {
+        if (has_end && end <= start) {
+            start = 0;
+            end = 0;
+            has_end = true;
+        }
+    }
+};
+
+static std::string leaf_cache_key(const pb::SplitIdAndFooterOffsets& so,
+                                  const pb::SearchRequest& req) {
+    // request time range: [start included, end excluded)
+    HalfOpen r;
+    r.start = req.start_timestamp ? *req.start_timestamp : INT64_MIN;
+    if (req.end_timestamp) {
+        r.has_end = true;
+        r.end = *req.end_timestamp;
+    }
+    r.normalize();
+    // split time range: [start, end] inclusive -> [start, end+1)
+    HalfOpen s;
+    s.start = so.timestamp_start ? *so.timestamp_start : INT64_MIN;
+    if (so.timestamp_end && *so.timestamp_end != INT64_MAX) {
+        s.has_end = true;
+        s.end = *so.timestamp_end + 1;
+    }
+    s.normalize();
+    // intersection
+    HalfOpen m;
+    m.start = std::max(r.start, s.start);
+    if (r.has_end || s.has_end) {
+        m.has_end = true;
+        m.end = r.has_end && s.has_end ? std::min(r.end, s.end)
+                                       : (r.has_end ? r.end : s.end);
+    }
+    m.normalize();
+    pb::SearchRequest canon = req;
+    canon.start_timestamp.reset();
+    canon.end_timestamp.reset();
+    canon.count_hits = 0;  // CountHits::CountAll (leaf_cache.rs:108)
+    std::string key = so.split_id;
+    key += '\0';
+    char rng[17];
+    memcpy(rng, &m.start, 8);
+    memcpy(rng + 8, &m.end, 8);
+    rng[16] = m.has_end ? 1 : 0;
+    key.append(rng, 17);
+    key += canon.encode();
+    return key;
+}
+
 }  // namespace qw
 
 struct qw_ctx {
@@ -87,6 +194,7 @@ struct qw_ctx {
     bool device_ready = false;
     int64_t agg_bucket_limit = 65000;
     std::map<std::string, std::unique_ptr<qw::DeviceSplit>> splits;
+    qw::LeafCache leaf_cache;
     std::string last_error;
     hipStream_t stream = nullptr;
     hipEvent_t ev_start = nullptr, ev_stop = nullptr;
@@ -1173,6 +1281,8 @@ qw_ctx* qw_ctx_create(const char* config_json) {
             if (const mj::Value* d = cfg->get("device")) ctx->device = int(d->as_i64());
             if (const mj::Value* b = cfg->get("aggregation_bucket_limit"))
                 ctx->agg_bucket_limit = b->as_i64();
+            if (const mj::Value* c = cfg->get("partial_request_cache_capacity"))
+                ctx->leaf_cache.capacity = size_t(c->as_i64());
         }
         return ctx;
     } catch (const std::exception& e) {
@@ -1229,6 +1339,7 @@ int32_t qw_ctx_remove_split(qw_ctx* ctx, const char* split_id) {
     if (it == ctx->splits.end()) return QW_ERR_NOT_FOUND;
     if (it->second->d_image) (void)hipFree(it->second->d_image);
     ctx->splits.erase(it);
+    ctx->leaf_cache.remove_split(split_id);
     return QW_OK;
 }
 
@@ -1349,6 +1460,41 @@ int32_t qw_leaf_search(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
                 t.req.count_hits != 0 /* != CountHits::CountAll */)
                 continue;  // pruned before warmup — not attempted
             resp.num_attempted_splits++;
+            // partial-result cache probe (leaf.rs:667 / leaf_cache.rs)
+            std::string ckey;
+            bool cacheable = ctx->leaf_cache.capacity > 0;
+            if (cacheable) {
+                ckey = leaf_cache_key(t.so, t.req);
+                std::string hit;
+                if (ctx->leaf_cache.get(ckey, &hit)) {
+                    pb::LeafSearchResponse cr = pb::LeafSearchResponse::decode(
+                        (const uint8_t*)hit.data(), hit.size());
+                    resp.num_successful_splits++;
+                    resp.num_hits += cr.num_hits;
+                    for (auto& h : cr.partial_hits) all_hits.push_back(std::move(h));
+                    if (cr.intermediate_aggregation_result) {
+                        IntermediateAggResults ir = IntermediateAggResults::decode(
+                            (const uint8_t*)cr.intermediate_aggregation_result->data(),
+                            cr.intermediate_aggregation_result->size());
+                        if (!any_aggs) {
+                            merged_aggs = std::move(ir);
+                            any_aggs = true;
+                        } else merged_aggs.merge(ir);
+                    }
+                    // cache-hit stats (leaf_cache.rs:73-79 overwrite-on-put)
+                    rstats.partial_result_cache_num_splits += 1;
+                    rstats.partial_result_cache_num_docs += t.so.num_docs;
+                    if (leaf_max > 0 && all_hits.size() >= leaf_max) {
+                        size_t k = std::min<size_t>(leaf_max, all_hits.size());
+                        std::partial_sort(all_hits.begin(), all_hits.begin() + k,
+                                          all_hits.end(), cmp);
+                        all_hits.resize(k);
+                        if (all_hits.size() == leaf_max)
+                            filter.record_new_worst_hit(all_hits.back());
+                    }
+                    continue;
+                }
+            }
             SplitResult r;
             try {
                 r = search_split_gpu(ctx, *t.ds, t.req, schema);
@@ -1363,6 +1509,13 @@ int32_t qw_leaf_search(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
             }
             resp.num_successful_splits++;
             resp.num_hits += r.num_hits;
+            if (cacheable) {  // cache fill on success (leaf.rs:966)
+                pb::LeafSearchResponse cv;
+                cv.num_hits = r.num_hits;
+                cv.partial_hits = r.hits;
+                if (r.has_aggs) cv.intermediate_aggregation_result = r.aggs.encode();
+                ctx->leaf_cache.put(ckey, cv.encode());
+            }
             for (auto& h : r.hits) all_hits.push_back(std::move(h));
             if (r.has_aggs) {
                 if (!any_aggs) {
@@ -1449,6 +1602,10 @@ int32_t qw_merge_leaf_responses(const uint8_t* search_request_pb,
             }
             if (r.resource_stats) {
                 if (!m.resource_stats) m.resource_stats = pb::LeafResourceStats{};
+                m.resource_stats->partial_result_cache_num_splits +=
+                    r.resource_stats->partial_result_cache_num_splits;
+                m.resource_stats->partial_result_cache_num_docs +=
+                    r.resource_stats->partial_result_cache_num_docs;
                 m.resource_stats->localexec_num_splits +=
                     r.resource_stats->localexec_num_splits;
                 m.resource_stats->localexec_num_docs +=

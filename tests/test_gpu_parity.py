@@ -502,3 +502,43 @@ def test_multi_split_ts_range_rewrite(multi):
     assert got.get("num_hits", 0) == exp.get("num_hits", 0)
     assert [hid(h) for h in got.get("partial_hits", [])] == \
            [hid(h) for h in exp.get("partial_hits", [])]
+
+
+# ------------------------------------------------- leaf search result cache
+# (LeafSearchCache, leaf_cache.rs: per-(split, canonical request) memoization)
+def test_leaf_cache_hit_identical_and_stats():
+    data = splitgen.generate_split(0, 20_000, seed=42)
+    gpu = GpuSearcher(device=0,
+                      config={"partial_request_cache_capacity": 1 << 20})
+    gpu.add_split(SID, data)
+    req = make_leaf_request(TERM_Q, SCHEMA, [(SID, 20_000)], max_hits=10)
+    r1 = gpu.leaf_search(req)
+    r2 = gpu.leaf_search(req)
+    assert r1.get("num_hits") == r2.get("num_hits")
+    assert [hid(h) for h in r1.get("partial_hits", [])] == \
+           [hid(h) for h in r2.get("partial_hits", [])]
+    st = r2["resource_stats"]
+    assert st.get("partial_result_cache_num_splits", 0) == 1
+    assert st.get("partial_result_cache_num_docs", 0) == 20_000
+    # request-level ts bounds are canonicalized into the merged time range:
+    # bounds wider than the split's range hit the same entry
+    r3 = gpu.leaf_search(make_leaf_request(
+        TERM_Q, SCHEMA, [{"split_id": SID, "num_docs": 20_000,
+                          "timestamp_start": 1, "timestamp_end": 2**31}],
+        max_hits=10, start_timestamp=0, end_timestamp=2**32))
+    assert r3["resource_stats"].get("partial_result_cache_num_splits", 0) == 1
+    # a different query misses
+    r4 = gpu.leaf_search(make_leaf_request(
+        {"type": "term", "field": "severity_text", "value": "ERROR"},
+        SCHEMA, [(SID, 20_000)], max_hits=10))
+    assert r4["resource_stats"].get("partial_result_cache_num_splits", 0) == 0
+
+
+def test_leaf_cache_disabled_by_default():
+    data = splitgen.generate_split(0, 20_000, seed=42)
+    gpu = GpuSearcher(device=0)
+    gpu.add_split(SID, data)
+    req = make_leaf_request(TERM_Q, SCHEMA, [(SID, 20_000)], max_hits=10)
+    gpu.leaf_search(req)
+    r2 = gpu.leaf_search(req)
+    assert r2["resource_stats"].get("partial_result_cache_num_splits", 0) == 0
