@@ -230,8 +230,30 @@ struct SplitSearcher {
                 return eval_presence(n);
             case PlanNode::BOOL:
                 return eval_bool(n);
+            case PlanNode::WILDCARD:
+                return eval_wildcard(n);
         }
         return Match{};
+    }
+
+    // union of every dict term matching the glob (wildcard_query.rs ->
+    // AutomatonQuery over the term dictionary; const score, see qast.h)
+    Match eval_wildcard(const PlanNode& n) const {
+        Match m;
+        const TextFieldView* f = sv.text_field(n.field);
+        if (!f) return m;
+        std::vector<uint32_t> docs;
+        for (uint32_t t = 0; t < f->num_terms; ++t) {
+            const char* s = (const char*)f->h_term_bytes + f->h_term_offsets[t];
+            size_t sl = f->h_term_offsets[t + 1] - f->h_term_offsets[t];
+            if (!glob_match(s, sl, n.value.data(), n.value.size(), n.ci)) continue;
+            Postings p = decode_term(*f, t);
+            docs.insert(docs.end(), p.docs.begin(), p.docs.end());
+        }
+        std::sort(docs.begin(), docs.end());
+        docs.erase(std::unique(docs.begin(), docs.end()), docs.end());
+        m.docs = std::move(docs);
+        return m;
     }
 
     Match eval_term(const PlanNode& n) const {
@@ -569,6 +591,10 @@ static SplitResult search_split(const SplitView& sv, const pb::SearchRequest& re
     bool scoring = false;
     for (auto& s : specs) scoring |= s.comp == SortSpec::SCORE;
 
+    if (scoring && plan_has_const_score(plan))
+        throw std::runtime_error(
+            "term_set/wildcard under _score sorting not supported (const-score "
+            "semantics, qast.h)");
     SplitSearcher searcher{sv, schema, scoring};
     Match m = searcher.eval(plan);
     out.num_hits = m.size(sv.num_docs);

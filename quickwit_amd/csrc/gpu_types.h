@@ -17,6 +17,8 @@ enum TermRole : uint32_t { ROLE_SHOULD = 0, ROLE_MUST = 1, ROLE_MUST_NOT = 2 };
 
 struct TermDev {
     uint32_t role;
+    uint32_t grp;          // must-group: terms in one group OR together, the
+                           // groups AND (wildcard/term_set under must)
     uint32_t n_blocks;
     uint64_t skip_off;     // byte offset of this term's first SkipEntry
     uint64_t payload_off;  // byte offset of the FIELD's payload section
@@ -89,7 +91,7 @@ struct QueryDev {
     uint32_t num_docs;
     uint32_t n_tiles;
     uint32_t n_terms;
-    uint32_t n_must;      // count of ROLE_MUST terms
+    uint32_t n_must;      // count of must GROUPS (see TermDev::grp)
     uint32_t n_must_not;  // count of ROLE_MUST_NOT terms
     uint32_t n_preds;
     uint32_t n_aggs;

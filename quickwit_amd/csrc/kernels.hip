@@ -406,24 +406,27 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                                            sc_score, nullptr, q.scoring);
         __syncthreads();
 
-        // ---- must terms: intersection via per-term bitset
+        // ---- must groups: union the group's terms into a temp bitset,
+        // AND the groups (q.n_must = group count; a group is one plain term
+        // or a wildcard/term_set expansion — OR within, AND across)
         if (NB) {
-            for (uint32_t t = 0, mi = 0; t < q.n_terms; ++t) {
-                if (terms[t].role != ROLE_MUST) continue;
+            for (uint32_t g = 0; g < q.n_must; ++g) {
                 // temp bitset lives in bits_not while must_nots are not yet done
                 uint32_t* tmp = sc_bits_not;
                 for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS)
                     tmp[i] = 0;
                 __syncthreads();
-                decode_term_tile<false>(q, terms[t], tile, tile_lo, tile_hi, nullptr,
-                                        tmp, false);
-                if (NS && q.scoring)  // musts contribute to the score too
-                    decode_term_tile<true>(q, terms[t], tile, tile_lo, tile_hi,
-                                           sc_score, nullptr, true);
+                for (uint32_t t = 0; t < q.n_terms; ++t) {
+                    if (terms[t].role != ROLE_MUST || terms[t].grp != g) continue;
+                    decode_term_tile<false>(q, terms[t], tile, tile_lo, tile_hi,
+                                            nullptr, tmp, false);
+                    if (NS && q.scoring && terms[t].weight != 0.f)
+                        decode_term_tile<true>(q, terms[t], tile, tile_lo, tile_hi,
+                                               sc_score, nullptr, true);
+                }
                 __syncthreads();
                 for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS)
-                    sc_bits_acc[i] = mi == 0 ? tmp[i] : (sc_bits_acc[i] & tmp[i]);
-                ++mi;
+                    sc_bits_acc[i] = g == 0 ? tmp[i] : (sc_bits_acc[i] & tmp[i]);
                 __syncthreads();
             }
             if (q.n_must_not) {

@@ -234,8 +234,10 @@ struct FlatQuery {
         int64_t tid;
         uint32_t role;
         float weight;  // BM25 W (0 for unscored filter terms)
+        uint32_t grp = 0;  // must-group (OR within a group, AND across)
     };
     std::vector<FTerm> terms;
+    uint32_t n_must_groups = 0;
     std::vector<PredDev> preds;
     std::vector<const TextFieldView*> ktab_fields;  // ktab_idx -> field
 
@@ -292,13 +294,14 @@ static void add_range_pred(FlatQuery& fq, const SplitView& sv, const PlanNode& n
 
 static void add_term(FlatQuery& fq, const SplitView& sv, const std::string& field,
                      const std::string& value, uint32_t role, float boost,
-                     bool scored) {
+                     bool scored, uint32_t grp = 0, bool group_member = false) {
     const TextFieldView* f = sv.text_field(field);
     int64_t tid = f ? f->find_term(value.data(), value.size()) : -1;
     if (tid < 0) {
         // absent term: MUST -> split matches nothing; SHOULD/MUST_NOT -> skip
-        // (skipped shoulds can never satisfy msm — handled by caller count)
-        if (role == ROLE_MUST) fq.match_none = true;
+        // (skipped shoulds can never satisfy msm — handled by caller count).
+        // A member of a must OR-GROUP only weakens the group when absent.
+        if (role == ROLE_MUST && !group_member) fq.match_none = true;
         return;
     }
     FlatQuery::FTerm t;
@@ -306,7 +309,32 @@ static void add_term(FlatQuery& fq, const SplitView& sv, const std::string& fiel
     t.tid = tid;
     t.role = role;
     t.weight = scored ? bm25_weight(sv, *f, tid, boost) : 0.f;
+    t.grp = grp;
     fq.terms.push_back(t);
+}
+
+// expand a wildcard over the field's term dictionary (qast.h glob_match;
+// wildcard_query.rs -> AutomatonQuery). Returns the number of matched terms.
+static size_t add_wildcard(FlatQuery& fq, const SplitView& sv, const PlanNode& n,
+                           uint32_t role, uint32_t grp) {
+    const TextFieldView* f = sv.text_field(n.field);
+    if (!f) return 0;
+    size_t cnt = 0;
+    for (uint32_t t = 0; t < f->num_terms; ++t) {
+        const char* s = (const char*)f->h_term_bytes + f->h_term_offsets[t];
+        size_t sl = f->h_term_offsets[t + 1] - f->h_term_offsets[t];
+        if (!glob_match(s, sl, n.value.data(), n.value.size(), n.ci)) continue;
+        FlatQuery::FTerm ft;
+        ft.f = f;
+        ft.tid = t;
+        ft.role = role;
+        ft.weight = 0.f;  // const-score node; scored plans were rejected
+        ft.grp = grp;
+        fq.terms.push_back(ft);
+        if (++cnt > 4096)
+            throw std::runtime_error("wildcard expands to >4096 terms (GPU r1 limit)");
+    }
+    return cnt;
 }
 
 static bool is_pure_should_terms(const PlanNode& n) {
@@ -337,7 +365,12 @@ static void add_positive(FlatQuery& fq, const SplitView& sv, const PlanNode& c,
                          bool scored, float boost) {
     switch (c.kind) {
         case PlanNode::TERM:
-            add_term(fq, sv, c.field, c.value, ROLE_MUST, boost * c.boost, scored);
+            add_term(fq, sv, c.field, c.value, ROLE_MUST, boost * c.boost, scored,
+                     fq.n_must_groups++);
+            break;
+        case PlanNode::WILDCARD:
+            if (add_wildcard(fq, sv, c, ROLE_MUST, fq.n_must_groups++) == 0)
+                fq.match_none = true;  // must clause matching nothing
             break;
         case PlanNode::RANGE:
         case PlanNode::FIELD_PRESENCE:
@@ -352,10 +385,17 @@ static void add_positive(FlatQuery& fq, const SplitView& sv, const PlanNode& c,
             if (is_pure_must_terms(c)) {
                 for (auto& m : c.must)
                     add_term(fq, sv, m.field, m.value, ROLE_MUST,
-                             boost * c.boost * m.boost, scored);
-            } else if (is_pure_should_terms(c) && c.should.size() == 1) {
-                add_term(fq, sv, c.should[0].field, c.should[0].value, ROLE_MUST,
-                         boost * c.boost * c.should[0].boost, scored);
+                             boost * c.boost * m.boost, scored,
+                             fq.n_must_groups++);
+            } else if (is_pure_should_terms(c)) {
+                // OR-of-terms under must/filter: one must-GROUP (the kernel
+                // ORs the group's bitsets, then ANDs the groups)
+                uint32_t grp = fq.n_must_groups++;
+                size_t before = fq.terms.size();
+                for (auto& s : c.should)
+                    add_term(fq, sv, s.field, s.value, ROLE_MUST,
+                             boost * c.boost * s.boost, scored, grp, true);
+                if (fq.terms.size() == before) fq.match_none = true;
             } else {
                 throw std::runtime_error(
                     "nested boolean inside must/filter not flattenable (GPU r1)");
@@ -385,6 +425,9 @@ static void flatten_bool(FlatQuery& fq, const SplitView& sv, const PlanNode& n,
                 add_term(fq, sv, t.field, t.value, ROLE_SHOULD,
                          boost * c.boost * t.boost, fq.scoring);
             ++should_clauses;
+        } else if (c.kind == PlanNode::WILDCARD) {
+            add_wildcard(fq, sv, c, ROLE_SHOULD, 0);
+            ++should_clauses;
         } else if (c.kind == PlanNode::MATCH_NONE) {
             // contributes nothing
         } else {
@@ -397,6 +440,8 @@ static void flatten_bool(FlatQuery& fq, const SplitView& sv, const PlanNode& n,
             add_term(fq, sv, c.field, c.value, ROLE_MUST_NOT, 1.f, false);
         else if (c.kind == PlanNode::RANGE || c.kind == PlanNode::FIELD_PRESENCE)
             add_range_pred(fq, sv, c, true);
+        else if (c.kind == PlanNode::WILDCARD)
+            add_wildcard(fq, sv, c, ROLE_MUST_NOT, 0);
         else if (is_pure_should_terms(c))
             // not(a or b) = not a and not b: must_not is a union-then-subtract
             for (auto& t : c.should)
@@ -444,6 +489,10 @@ static void flatten_bool(FlatQuery& fq, const SplitView& sv, const PlanNode& n,
 static FlatQuery flatten(const SplitView& sv, const PlanNode& plan, bool scoring) {
     FlatQuery fq;
     fq.scoring = scoring;
+    if (scoring && plan_has_const_score(plan))
+        throw std::runtime_error(
+            "term_set/wildcard under _score sorting not supported (const-score "
+            "semantics, qast.h)");
     switch (plan.kind) {
         case PlanNode::MATCH_ALL:
             fq.match_all = true;
@@ -460,6 +509,11 @@ static FlatQuery flatten(const SplitView& sv, const PlanNode& plan, bool scoring
         case PlanNode::FIELD_PRESENCE:
             fq.match_all = true;
             add_range_pred(fq, sv, plan, false);
+            break;
+        case PlanNode::WILDCARD:
+            add_wildcard(fq, sv, plan, ROLE_SHOULD, 0);
+            if (fq.terms.empty()) fq.match_none = true;
+            fq.msm = 1;
             break;
         case PlanNode::BOOL:
             flatten_bool(fq, sv, plan, 1.f);
@@ -770,6 +824,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         d.payload_off = t.f->payload.off;
         d.norms_off = t.f->has_norms ? t.f->fieldnorms.off : 0;
         d.weight = t.weight;
+        d.grp = t.grp;
         d.ktab_idx = fq.scoring ? ktab_idx_of(t.f) : 0;
         n_must += t.role == ROLE_MUST;
         n_must_not += t.role == ROLE_MUST_NOT;
@@ -862,7 +917,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         q.num_docs = sv.num_docs;
         q.n_tiles = n_tiles;
         q.n_terms = uint32_t(terms.size());
-        q.n_must = n_must;
+        q.n_must = fq.n_must_groups;  // group count (TermDev::grp)
         q.n_must_not = n_must_not;
         q.n_preds = uint32_t(fq.preds.size());
         q.n_aggs = uint32_t(ap.devs.size());

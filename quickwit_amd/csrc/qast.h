@@ -106,17 +106,56 @@ struct Bound {
 };
 
 struct PlanNode {
-    enum Kind { MATCH_ALL, MATCH_NONE, TERM, BOOL, RANGE, FIELD_PRESENCE } kind = MATCH_ALL;
-    // TERM
+    enum Kind {
+        MATCH_ALL, MATCH_NONE, TERM, BOOL, RANGE, FIELD_PRESENCE, WILDCARD
+    } kind = MATCH_ALL;
+    // TERM / WILDCARD (value = glob pattern: '*' any run, '?' one char)
     std::string field;
     std::string value;
+    bool ci = false;  // WILDCARD case_insensitive
     // BOOL
     std::vector<PlanNode> must, must_not, should, filter;
     int64_t minimum_should_match = -1;  // -1 = unset
     // RANGE / FIELD_PRESENCE
     Bound lo, hi;
     float boost = 1.0f;
+    // term_set / wildcard match the union's DOC SET but score const 1.0
+    // (tantivy TermSetQuery / AutomatonQuery const scorer) — the build
+    // rejects these under _score sorting instead of mis-scoring them
+    bool const_score = false;
 };
+
+// does any node require const-score semantics (term_set / wildcard)?
+inline bool plan_has_const_score(const PlanNode& n) {
+    if (n.const_score) return true;
+    for (auto* v : {&n.must, &n.must_not, &n.should, &n.filter})
+        for (auto& c : *v)
+            if (plan_has_const_score(c)) return true;
+    return false;
+}
+
+// glob match with '*' (any run) and '?' (single byte), iterative backtracking
+inline bool glob_match(const char* s, size_t sn, const char* p, size_t pn, bool ci) {
+    size_t si = 0, pi = 0, star = size_t(-1), mark = 0;
+    auto eq = [&](char a, char b) {
+        return ci ? std::tolower((unsigned char)a) == std::tolower((unsigned char)b)
+                  : a == b;
+    };
+    while (si < sn) {
+        if (pi < pn && (p[pi] == '?' || eq(p[pi], s[si]))) {
+            ++si;
+            ++pi;
+        } else if (pi < pn && p[pi] == '*') {
+            star = pi++;
+            mark = si;
+        } else if (star != size_t(-1)) {
+            pi = star + 1;
+            si = ++mark;
+        } else return false;
+    }
+    while (pi < pn && p[pi] == '*') ++pi;
+    return pi == pn;
+}
 
 inline int64_t parse_datetime_ms(const mj::Value* lit);
 
@@ -292,6 +331,51 @@ inline PlanNode build_plan(const mj::Value* ast, const Schema& schema) {
         PlanNode inner = build_plan(ast->at("underlying"), schema);
         inner.boost *= float(ast->at("boost")->num());
         return inner;
+    } else if (ty == "term_set") {
+        // TermSetQuery "matches the same document set as a union of the
+        // equivalent TermQueries", untokenized values
+        // (query_ast/term_set_query.rs:25-80); values arrive as a sorted set
+        n.kind = PlanNode::BOOL;
+        n.const_score = true;
+        const mj::Value* tpf = ast->at("terms_per_field");
+        for (auto& kv : tpf->obj) {
+            if (!schema.field(kv.first))
+                throw std::runtime_error("term_set on unknown field: " + kv.first);
+            for (auto& v : kv.second->arr) {
+                PlanNode t;
+                t.kind = PlanNode::TERM;
+                t.field = kv.first;
+                t.value = v->s;
+                n.should.push_back(std::move(t));
+            }
+        }
+        if (n.should.empty()) n.kind = PlanNode::MATCH_NONE;
+    } else if (ty == "cache") {
+        // CacheNode wraps without changing semantics (cache_node.rs:33-37);
+        // the result-level memoization lives in LeafSearchCache host-side
+        return build_plan(ast->at("inner"), schema);
+    } else if (ty == "wildcard") {
+        n.kind = PlanNode::WILDCARD;
+        n.const_score = true;
+        n.field = ast->at("field")->s;
+        n.value = ast->at("value")->s;
+        const mj::Value* ci = ast->get("case_insensitive");
+        n.ci = ci && ci->b;
+        const SchemaField* f = schema.field(n.field);
+        if (!f) {
+            const mj::Value* len = ast->get("lenient");
+            if (len && len->b) {
+                n.kind = PlanNode::MATCH_NONE;  // lenient: missing field OK
+                return n;
+            }
+            throw std::runtime_error("wildcard on unknown field: " + n.field);
+        }
+        // the reference tokenizes the pattern's literal segments with the
+        // field's tokenizer (wildcard_query.rs:111-160): for the default
+        // (lowercasing) analyzer that lowercases them
+        if (f->tokenizer != "raw")
+            std::transform(n.value.begin(), n.value.end(), n.value.begin(),
+                           [](unsigned char c) { return char(std::tolower(c)); });
     } else if (ty == "user_input") {
         std::vector<std::string> dfs;
         const mj::Value* d = ast->get("default_fields");

@@ -542,3 +542,89 @@ def test_leaf_cache_disabled_by_default():
     gpu.leaf_search(req)
     r2 = gpu.leaf_search(req)
     assert r2["resource_stats"].get("partial_result_cache_num_splits", 0) == 0
+
+
+# ------------------------------------------------- term_set / wildcard / cache
+# (query_ast/term_set_query.rs, wildcard_query.rs, cache_node.rs)
+def test_term_set_union(searchers):
+    q = {"type": "term_set", "terms_per_field": {
+        "severity_text": ["WARN", "FATAL"],
+        "body": ["w00003"]}}
+    got, exp = run_both(searchers, q, max_hits=20)
+    assert_hits_equal(got, exp, scored=False)
+
+
+def test_term_set_inside_filter(searchers):
+    q = {"type": "bool",
+         "must": [{"type": "term", "field": "severity_text", "value": "INFO"}],
+         "filter": [{"type": "term_set",
+                     "terms_per_field": {"body": ["w00001", "w00002"]}}]}
+    got, exp = run_both(searchers, q, max_hits=20)
+    assert_hits_equal(got, exp, scored=False)
+
+
+def test_cache_node_is_transparent(searchers):
+    inner = {"type": "term", "field": "severity_text", "value": "ERROR"}
+    got, exp = run_both(searchers, {"type": "cache", "inner": inner},
+                        max_hits=15)
+    got2, _ = run_both(searchers, inner, max_hits=15)
+    assert_hits_equal(got, exp, scored=False)
+    assert [hid(h) for h in got.get("partial_hits", [])] == \
+           [hid(h) for h in got2.get("partial_hits", [])]
+
+
+def test_wildcard_should(searchers):
+    q = {"type": "wildcard", "field": "body", "value": "w0000*"}
+    got, exp = run_both(searchers, q, max_hits=25)
+    assert_hits_equal(got, exp, scored=False)
+
+
+def test_wildcard_question_mark(searchers):
+    q = {"type": "wildcard", "field": "body", "value": "w?0001"}
+    got, exp = run_both(searchers, q, max_hits=25)
+    assert_hits_equal(got, exp, scored=False)
+
+
+def test_wildcard_under_must_with_range(searchers):
+    q = {"type": "bool",
+         "must": [{"type": "wildcard", "field": "severity_text",
+                   "value": "ERR*"}],
+         "filter": [{"type": "range", "field": "tenant_id",
+                     "lower_bound": {"included": 0},
+                     "upper_bound": {"excluded": 500}}]}
+    got, exp = run_both(searchers, q, max_hits=20)
+    assert_hits_equal(got, exp, scored=False)
+
+
+def test_wildcard_must_not(searchers):
+    q = {"type": "bool",
+         "must": [{"type": "term", "field": "severity_text", "value": "WARN"}],
+         "must_not": [{"type": "wildcard", "field": "body", "value": "w0001*"}]}
+    got, exp = run_both(searchers, q, max_hits=20)
+    assert_hits_equal(got, exp, scored=False)
+
+
+def test_or_of_terms_under_must(searchers):
+    # OR-of-terms as a must clause -> one must-GROUP on the GPU
+    q = {"type": "bool",
+         "must": [{"type": "bool", "should": [
+             {"type": "term", "field": "severity_text", "value": "WARN"},
+             {"type": "term", "field": "severity_text", "value": "ERROR"}]},
+             {"type": "bool", "should": [
+                 {"type": "term", "field": "body", "value": "w00000"},
+                 {"type": "term", "field": "body", "value": "w00009"}]}]}
+    got, exp = run_both(searchers, q, max_hits=30)
+    assert_hits_equal(got, exp, scored=False)
+
+
+def test_wildcard_scored_rejected(searchers):
+    gpu, cpu = searchers
+    req = make_leaf_request({"type": "wildcard", "field": "body",
+                             "value": "w*"}, SCHEMA, [(SID, NDOCS)],
+                            max_hits=5,
+                            sort_fields=[{"field_name": "_score",
+                                          "sort_order": 1}])
+    for s in (gpu, cpu):
+        resp = s.leaf_search(req)
+        fs = resp.get("failed_splits", [])
+        assert fs and "const-score" in fs[0]["error"]
