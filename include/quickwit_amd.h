@@ -77,6 +77,17 @@ int32_t qw_leaf_search(qw_ctx* ctx, const uint8_t* leaf_search_request_pb,
 
 void qw_buf_free(qw_buf* buf);
 
+/* leaf_list_terms (quickwit-search/src/list_terms.rs:211-322):
+ * LeafListTermsRequest protobuf in (search.proto:732), LeafListTermsResponse
+ * out (search.proto:745). Per split: sorted term-dictionary range scan over
+ * [start_key, end_key) limited to max_hits, then k-merge + dedup across
+ * splits. Term bytes are the RAW term value bytes (declared deviation: the
+ * reference prefixes tantivy's serialized-Term header, an engine-internal
+ * encoding with no analog here — DESIGN.md §7). Host-only (term dictionaries
+ * stay on CPU); splits must have been added. */
+int32_t qw_leaf_list_terms(qw_ctx* ctx, const uint8_t* leaf_list_terms_request_pb,
+                           size_t len, qw_buf* response_pb_out);
+
 /* Root-side helpers replacing the reference's tantivy-side merge
  * (collector.rs:832-861 merge_fruits / aggregation merge): merge N
  * LeafSearchResponse protobufs (top-K with reference tie-breaks + QAGG1

@@ -383,6 +383,43 @@ struct SplitSearchError {  // search.proto:350
     }
 };
 
+struct ListTermsRequest {  // search.proto:700
+    std::vector<std::string> index_id_patterns;      // 1
+    std::string field;                               // 3
+    std::optional<int64_t> start_timestamp;          // 4
+    std::optional<int64_t> end_timestamp;            // 5
+    std::optional<uint64_t> max_hits;                // 6
+    std::optional<std::string> start_key, end_key;   // 7 incl, 8 excl
+    std::string encode() const {
+        Writer w;
+        for (auto& s : index_id_patterns) w.str_field(1, s, true);
+        w.str_field(3, field);
+        if (start_timestamp) w.i64_field(4, *start_timestamp, true);
+        if (end_timestamp) w.i64_field(5, *end_timestamp, true);
+        if (max_hits) w.u64_field(6, *max_hits, true);
+        if (start_key) w.str_field(7, *start_key, true);
+        if (end_key) w.str_field(8, *end_key, true);
+        return w.out;
+    }
+    static ListTermsRequest decode(Reader r) {
+        ListTermsRequest q;
+        while (!r.done()) {
+            uint32_t no, wt = r.read_tag(&no);
+            switch (no) {
+                case 1: q.index_id_patterns.push_back(r.bytes()); break;
+                case 3: q.field = r.bytes(); break;
+                case 4: q.start_timestamp = int64_t(r.varint()); break;
+                case 5: q.end_timestamp = int64_t(r.varint()); break;
+                case 6: q.max_hits = r.varint(); break;
+                case 7: q.start_key = r.bytes(); break;
+                case 8: q.end_key = r.bytes(); break;
+                default: r.skip(wt);
+            }
+        }
+        return q;
+    }
+};
+
 struct SplitResourceStats {  // search.proto:383
     uint64_t split_num_docs = 0, input_memory_bytes = 0, download_num_bytes = 0,
              download_num_requests = 0, matched_num_docs = 0,
@@ -510,6 +547,68 @@ struct LeafSearchResponse {  // search.proto:618
             } else r.skip(wt);
         }
         return v;
+    }
+};
+
+struct LeafListTermsRequest {  // search.proto:732
+    ListTermsRequest list_terms_request;              // 1
+    std::vector<SplitIdAndFooterOffsets> split_offsets;  // 2
+    std::string index_uri;                            // 3
+    std::string encode() const {
+        Writer w;
+        w.msg_field(1, list_terms_request.encode());
+        for (auto& s : split_offsets) w.msg_field(2, s.encode());
+        w.str_field(3, index_uri);
+        return w.out;
+    }
+    static LeafListTermsRequest decode(const uint8_t* data, size_t n) {
+        Reader r(data, n);
+        LeafListTermsRequest q;
+        while (!r.done()) {
+            uint32_t no, wt = r.read_tag(&no);
+            if (no == 1 && wt == 2) {
+                std::string b = r.bytes();
+                q.list_terms_request = ListTermsRequest::decode(
+                    Reader((const uint8_t*)b.data(), b.size()));
+            } else if (no == 2 && wt == 2) {
+                std::string b = r.bytes();
+                q.split_offsets.push_back(SplitIdAndFooterOffsets::decode(
+                    Reader((const uint8_t*)b.data(), b.size())));
+            } else if (no == 3 && wt == 2) q.index_uri = r.bytes();
+            else r.skip(wt);
+        }
+        return q;
+    }
+};
+
+struct LeafListTermsResponse {  // search.proto:745
+    uint64_t num_hits = 0;                       // 1
+    std::vector<std::string> terms;              // 2
+    std::vector<SplitSearchError> failed_splits; // 3
+    uint64_t num_attempted_splits = 0;           // 4
+    std::string encode() const {
+        Writer w;
+        w.u64_field(1, num_hits);
+        for (auto& t : terms) w.bytes_field(2, t);
+        for (auto& f : failed_splits) w.msg_field(3, f.encode());
+        w.u64_field(4, num_attempted_splits);
+        return w.out;
+    }
+    static LeafListTermsResponse decode(const uint8_t* data, size_t n) {
+        Reader r(data, n);
+        LeafListTermsResponse q;
+        while (!r.done()) {
+            uint32_t no, wt = r.read_tag(&no);
+            if (no == 1 && wt == 0) q.num_hits = r.varint();
+            else if (no == 2 && wt == 2) q.terms.push_back(r.bytes());
+            else if (no == 3 && wt == 2) {
+                std::string b = r.bytes();
+                q.failed_splits.push_back(SplitSearchError::decode(
+                    Reader((const uint8_t*)b.data(), b.size())));
+            } else if (no == 4 && wt == 0) q.num_attempted_splits = r.varint();
+            else r.skip(wt);
+        }
+        return q;
     }
 };
 

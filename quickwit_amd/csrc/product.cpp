@@ -1619,6 +1619,74 @@ int32_t qw_leaf_search(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
     }
 }
 
+int32_t qw_leaf_list_terms(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
+                           qw_buf* out) {
+    using namespace qw;
+    try {
+        pb::LeafListTermsRequest lreq = pb::LeafListTermsRequest::decode(req_pb, req_len);
+        const pb::ListTermsRequest& req = lreq.list_terms_request;
+        pb::LeafListTermsResponse resp;
+        // per split: sorted dict range scan, limited (list_terms.rs:246-290)
+        std::vector<std::vector<std::string>> per_split;
+        for (auto& so : lreq.split_offsets) {
+            auto it = ctx->splits.find(so.split_id);
+            if (it == ctx->splits.end()) {
+                pb::SplitSearchError se;
+                se.split_id = so.split_id;
+                se.error = "unknown split: " + so.split_id;
+                se.retryable_error = true;
+                resp.failed_splits.push_back(std::move(se));
+                continue;
+            }
+            const SplitView& sv = it->second->view;
+            const TextFieldView* f = sv.text_field(req.field);
+            if (!f) {
+                pb::SplitSearchError se;
+                se.split_id = so.split_id;
+                se.error = "couldn't get field named \"" + req.field +
+                           "\" from schema to list terms";
+                se.retryable_error = false;
+                resp.failed_splits.push_back(std::move(se));
+                continue;
+            }
+            resp.num_attempted_splits++;
+            std::vector<std::string> terms;
+            for (uint32_t t = 0; t < f->num_terms; ++t) {
+                const char* s = (const char*)f->h_term_bytes + f->h_term_offsets[t];
+                size_t sl = f->h_term_offsets[t + 1] - f->h_term_offsets[t];
+                std::string term(s, sl);
+                if (req.start_key && term < *req.start_key) continue;  // ge
+                if (req.end_key && term >= *req.end_key) break;        // lt
+                terms.push_back(std::move(term));
+                if (req.max_hits && terms.size() >= *req.max_hits) break;
+            }
+            per_split.push_back(std::move(terms));
+        }
+        // k-merge + dedup + global limit (list_terms.rs:291-300,392-400)
+        std::vector<size_t> pos(per_split.size(), 0);
+        while (true) {
+            if (req.max_hits && resp.terms.size() >= *req.max_hits) break;
+            const std::string* best = nullptr;
+            for (size_t i = 0; i < per_split.size(); ++i)
+                if (pos[i] < per_split[i].size() &&
+                    (!best || per_split[i][pos[i]] < *best))
+                    best = &per_split[i][pos[i]];
+            if (!best) break;
+            std::string v = *best;
+            for (size_t i = 0; i < per_split.size(); ++i)
+                while (pos[i] < per_split[i].size() && per_split[i][pos[i]] == v)
+                    ++pos[i];
+            resp.terms.push_back(std::move(v));
+        }
+        resp.num_hits = resp.terms.size();
+        fill_buf(out, resp.encode());
+        return QW_OK;
+    } catch (const std::exception& e) {
+        set_err(ctx, e.what());
+        return QW_ERR_INTERNAL;
+    }
+}
+
 void qw_buf_free(qw_buf* buf) {
     if (!buf) return;
     free(buf->data);

@@ -72,6 +72,26 @@ SCHEMAS = {
         8: ("doc_mappers", "*str"),
         9: ("index_uris", "*str"),
     },
+    "ListTermsRequest": {      # search.proto:700
+        1: ("index_id_patterns", "*str"),
+        3: ("field", "str"),
+        4: ("start_timestamp", "i64"),
+        5: ("end_timestamp", "i64"),
+        6: ("max_hits", "u64"),
+        7: ("start_key", "bytes"),
+        8: ("end_key", "bytes"),
+    },
+    "LeafListTermsRequest": {  # search.proto:732
+        1: ("list_terms_request", "msg:ListTermsRequest"),
+        2: ("split_offsets", "*msg:SplitIdAndFooterOffsets"),
+        3: ("index_uri", "str"),
+    },
+    "LeafListTermsResponse": {  # search.proto:745
+        1: ("num_hits", "u64"),
+        2: ("terms", "*bytes"),
+        3: ("failed_splits", "*msg:SplitSearchError"),
+        4: ("num_attempted_splits", "u64"),
+    },
     "SplitSearchError": {
         1: ("error", "str"),
         2: ("split_id", "str"),
@@ -197,6 +217,11 @@ def encode(msg_name: str, d: dict) -> bytes:
 
 
 _OPTIONAL = {
+    ("ListTermsRequest", "start_timestamp"),
+    ("ListTermsRequest", "end_timestamp"),
+    ("ListTermsRequest", "max_hits"),
+    ("ListTermsRequest", "start_key"),
+    ("ListTermsRequest", "end_key"),
     ("SearchRequest", "start_timestamp"),
     ("SearchRequest", "end_timestamp"),
     ("SearchRequest", "aggregation_request"),

@@ -628,3 +628,47 @@ def test_wildcard_scored_rejected(searchers):
         resp = s.leaf_search(req)
         fs = resp.get("failed_splits", [])
         assert fs and "const-score" in fs[0]["error"]
+
+
+# ------------------------------------------------- leaf_list_terms
+# (list_terms.rs:211-322: per-split dict range scan, k-merge + dedup).
+# Checked against an independent Python walk of the same dictionaries.
+def test_leaf_list_terms(searchers):
+    gpu, _ = searchers
+    from quickwit_amd import splitread
+    data = splitgen.generate_split(0, NDOCS, seed=42)
+    sp = splitread.Split(data)
+    vocab = sp.terms("body")
+
+    def expect(start=None, end=None, limit=None):
+        terms = [t for t in vocab
+                 if (start is None or t >= start) and (end is None or t < end)]
+        return terms[:limit] if limit is not None else terms
+
+    def run(**kw):
+        ltr = {"index_id_patterns": ["bench-index"], "field": "body"}
+        for k in ("start_key", "end_key"):
+            if k in kw and kw[k] is not None:
+                ltr[k] = kw[k].encode()
+        if kw.get("max_hits") is not None:
+            ltr["max_hits"] = kw["max_hits"]
+        resp = gpu.leaf_list_terms({
+            "list_terms_request": ltr,
+            "split_offsets": [{"split_id": SID, "num_docs": NDOCS}]})
+        return [t.decode() for t in resp.get("terms", [])]
+
+    assert run(max_hits=50) == [t for t in expect(limit=50)]
+    assert run(start_key="w00100", end_key="w00200", max_hits=1000) == \
+        expect(start="w00100", end="w00200", limit=1000)
+    got_all = run()
+    assert got_all == expect()
+
+
+def test_leaf_list_terms_missing_field(searchers):
+    gpu, _ = searchers
+    resp = gpu.leaf_list_terms({
+        "list_terms_request": {"index_id_patterns": ["x"], "field": "nope"},
+        "split_offsets": [{"split_id": SID, "num_docs": NDOCS}]})
+    fs = resp.get("failed_splits", [])
+    assert fs and "couldn't get field" in fs[0]["error"]
+    assert resp.get("num_attempted_splits", 0) == 0
