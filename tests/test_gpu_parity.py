@@ -511,7 +511,9 @@ def test_leaf_cache_hit_identical_and_stats():
     gpu = GpuSearcher(device=0,
                       config={"partial_request_cache_capacity": 1 << 20})
     gpu.add_split(SID, data)
-    req = make_leaf_request(TERM_Q, SCHEMA, [(SID, 20_000)], max_hits=10)
+    meta = {"split_id": SID, "num_docs": 20_000,
+            "timestamp_start": 1, "timestamp_end": 2**31}
+    req = make_leaf_request(TERM_Q, SCHEMA, [meta], max_hits=10)
     r1 = gpu.leaf_search(req)
     r2 = gpu.leaf_search(req)
     assert r1.get("num_hits") == r2.get("num_hits")
@@ -523,8 +525,7 @@ def test_leaf_cache_hit_identical_and_stats():
     # request-level ts bounds are canonicalized into the merged time range:
     # bounds wider than the split's range hit the same entry
     r3 = gpu.leaf_search(make_leaf_request(
-        TERM_Q, SCHEMA, [{"split_id": SID, "num_docs": 20_000,
-                          "timestamp_start": 1, "timestamp_end": 2**31}],
+        TERM_Q, SCHEMA, [meta],
         max_hits=10, start_timestamp=0, end_timestamp=2**32))
     assert r3["resource_stats"].get("partial_result_cache_num_splits", 0) == 1
     # a different query misses
