@@ -77,6 +77,16 @@ int32_t qw_leaf_search(qw_ctx* ctx, const uint8_t* leaf_search_request_pb,
 
 void qw_buf_free(qw_buf* buf);
 
+/* fetch_docs phase 2 (quickwit-search/src/fetch_docs.rs; root.rs:903):
+ * FetchDocsRequest protobuf in (search.proto:674), FetchDocsResponse out
+ * (:695). Fetches the stored documents for the given PartialHits from the
+ * splits' row store (zlib blocks + doc->block index);
+ * leaf_json is the ingested document's canonical JSON. Snippets
+ * (snippet_request) are not produced in round 1. Hits are returned grouped
+ * by (split_id, doc_id) — the root matches them by partial_hit. */
+int32_t qw_fetch_docs(qw_ctx* ctx, const uint8_t* fetch_docs_request_pb,
+                      size_t len, qw_buf* response_pb_out);
+
 /* leaf_list_terms (quickwit-search/src/list_terms.rs:211-322):
  * LeafListTermsRequest protobuf in (search.proto:732), LeafListTermsResponse
  * out (search.proto:745). Per split: sorted term-dictionary range scan over

@@ -128,6 +128,19 @@ class GpuSearcher(_BaseSearcher):
         self._lib.qw_buf_free(ctypes.byref(buf))
         return json.loads(out)
 
+    def fetch_docs(self, req: dict) -> dict:
+        fn = self._lib.qw_fetch_docs
+        fn.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_size_t,
+                       ctypes.POINTER(_Buf)]
+        req_pb = proto.encode("FetchDocsRequest", req)
+        buf = _Buf()
+        rc = fn(self._ctx, req_pb, len(req_pb), ctypes.byref(buf))
+        if rc != 0:
+            raise RuntimeError(f"fetch_docs failed ({rc}): {self._err()}")
+        out = ctypes.string_at(buf.data, buf.len)
+        self._lib.qw_buf_free(ctypes.byref(buf))
+        return proto.decode("FetchDocsResponse", out)
+
     def leaf_list_terms(self, req: dict) -> dict:
         fn = self._lib.qw_leaf_list_terms
         fn.argtypes = [ctypes.c_void_p, ctypes.c_char_p, ctypes.c_size_t,

@@ -550,6 +550,89 @@ struct LeafSearchResponse {  // search.proto:618
     }
 };
 
+struct FetchDocsRequest {  // search.proto:674
+    std::vector<PartialHit> partial_hits;              // 1
+    std::vector<SplitIdAndFooterOffsets> split_offsets;  // 3
+    std::string index_uri;                             // 4
+    std::string doc_mapper;                            // 6
+    std::string encode() const {
+        Writer w;
+        for (auto& h : partial_hits) w.msg_field(1, h.encode());
+        for (auto& s : split_offsets) w.msg_field(3, s.encode());
+        w.str_field(4, index_uri);
+        w.str_field(6, doc_mapper);
+        return w.out;
+    }
+    static FetchDocsRequest decode(const uint8_t* data, size_t n) {
+        Reader r(data, n);
+        FetchDocsRequest q;
+        while (!r.done()) {
+            uint32_t no, wt = r.read_tag(&no);
+            if (no == 1 && wt == 2) {
+                std::string b = r.bytes();
+                q.partial_hits.push_back(
+                    PartialHit::decode(Reader((const uint8_t*)b.data(), b.size())));
+            } else if (no == 3 && wt == 2) {
+                std::string b = r.bytes();
+                q.split_offsets.push_back(SplitIdAndFooterOffsets::decode(
+                    Reader((const uint8_t*)b.data(), b.size())));
+            } else if (no == 4 && wt == 2) q.index_uri = r.bytes();
+            else if (no == 6 && wt == 2) q.doc_mapper = r.bytes();
+            else r.skip(wt);
+        }
+        return q;
+    }
+};
+
+struct LeafHit {  // search.proto:553
+    std::string leaf_json;                      // 1
+    PartialHit partial_hit;                     // 2
+    std::optional<std::string> leaf_snippet_json;  // 3
+    std::string encode() const {
+        Writer w;
+        w.str_field(1, leaf_json);
+        w.msg_field(2, partial_hit.encode());
+        if (leaf_snippet_json) w.str_field(3, *leaf_snippet_json, true);
+        return w.out;
+    }
+    static LeafHit decode(Reader r) {
+        LeafHit h;
+        while (!r.done()) {
+            uint32_t no, wt = r.read_tag(&no);
+            if (no == 1 && wt == 2) h.leaf_json = r.bytes();
+            else if (no == 2 && wt == 2) {
+                std::string b = r.bytes();
+                h.partial_hit =
+                    PartialHit::decode(Reader((const uint8_t*)b.data(), b.size()));
+            } else if (no == 3 && wt == 2) h.leaf_snippet_json = r.bytes();
+            else r.skip(wt);
+        }
+        return h;
+    }
+};
+
+struct FetchDocsResponse {  // search.proto:695
+    std::vector<LeafHit> hits;  // 1
+    std::string encode() const {
+        Writer w;
+        for (auto& h : hits) w.msg_field(1, h.encode());
+        return w.out;
+    }
+    static FetchDocsResponse decode(const uint8_t* data, size_t n) {
+        Reader r(data, n);
+        FetchDocsResponse q;
+        while (!r.done()) {
+            uint32_t no, wt = r.read_tag(&no);
+            if (no == 1 && wt == 2) {
+                std::string b = r.bytes();
+                q.hits.push_back(
+                    LeafHit::decode(Reader((const uint8_t*)b.data(), b.size())));
+            } else r.skip(wt);
+        }
+        return q;
+    }
+};
+
 struct LeafListTermsRequest {  // search.proto:732
     ListTermsRequest list_terms_request;              // 1
     std::vector<SplitIdAndFooterOffsets> split_offsets;  // 2

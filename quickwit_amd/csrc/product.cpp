@@ -17,6 +17,7 @@
 // (DESIGN.md §1). The CPU restatement lives in oracle/ (test infra only).
 // ============================================================================
 #include <hip/hip_runtime.h>
+#include <zlib.h>
 
 #include <algorithm>
 #include <chrono>
@@ -1615,6 +1616,90 @@ int32_t qw_leaf_search(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
         if (msg.find("not supported") != std::string::npos ||
             msg.find("not flattenable") != std::string::npos)
             return QW_ERR_INVALID_QUERY;
+        return QW_ERR_INTERNAL;
+    }
+}
+
+int32_t qw_fetch_docs(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
+                      qw_buf* out) {
+    using namespace qw;
+    try {
+        pb::FetchDocsRequest freq = pb::FetchDocsRequest::decode(req_pb, req_len);
+        // group hits by split (fetch_docs.rs:56-60 sorts GlobalDocAddress);
+        // one block decompression serves consecutive docs of the same block
+        std::vector<size_t> order(freq.partial_hits.size());
+        for (size_t i = 0; i < order.size(); ++i) order[i] = i;
+        std::sort(order.begin(), order.end(), [&](size_t a, size_t b) {
+            const pb::PartialHit& ha = freq.partial_hits[a];
+            const pb::PartialHit& hb = freq.partial_hits[b];
+            if (ha.split_id != hb.split_id) return ha.split_id < hb.split_id;
+            return ha.doc_id < hb.doc_id;
+        });
+        pb::FetchDocsResponse resp;
+        std::string cur_split;
+        const DeviceSplit* ds = nullptr;
+        std::vector<uint8_t> block;       // decompressed block cache
+        int64_t cached_block = -1;
+        for (size_t oi : order) {
+            const pb::PartialHit& h = freq.partial_hits[oi];
+            if (h.split_id != cur_split) {
+                auto it = ctx->splits.find(h.split_id);
+                if (it == ctx->splits.end())
+                    throw std::runtime_error("unknown split: " + h.split_id);
+                ds = it->second.get();
+                cur_split = h.split_id;
+                cached_block = -1;
+            }
+            const DocStoreView& d = ds->view.docstore;
+            if (!d.present)
+                throw std::runtime_error("split has no docstore: " + h.split_id);
+            if (h.doc_id >= ds->view.num_docs)
+                throw std::runtime_error("doc id out of range");
+            // binary search the block whose [first, next_first) covers doc
+            uint32_t lo = 0, hi = d.n_blocks;
+            while (lo + 1 < hi) {
+                uint32_t mid = (lo + hi) / 2;
+                if (d.firsts[mid] <= h.doc_id) lo = mid;
+                else hi = mid;
+            }
+            if (int64_t(lo) != cached_block) {
+                uint32_t nb = d.firsts[lo + 1] - d.firsts[lo];
+                // uncompressed size upper bound: grow until inflate fits
+                uLongf dst_len = uLongf(
+                    std::max<uint64_t>(64 * 1024, (d.offs[lo + 1] - d.offs[lo]) * 8));
+                for (;;) {
+                    block.resize(dst_len);
+                    uLongf got = dst_len;
+                    int rc = uncompress(block.data(), &got, d.blocks + d.offs[lo],
+                                        uLong(d.offs[lo + 1] - d.offs[lo]));
+                    if (rc == Z_OK) {
+                        block.resize(got);
+                        break;
+                    }
+                    if (rc != Z_BUF_ERROR)
+                        throw std::runtime_error("docstore block inflate failed");
+                    dst_len *= 2;
+                }
+                if (block.size() < nb * 4)
+                    throw std::runtime_error("docstore block too short");
+                cached_block = int64_t(lo);
+            }
+            uint32_t nb = d.firsts[lo + 1] - d.firsts[lo];
+            const uint32_t* lens = (const uint32_t*)block.data();
+            uint32_t local = h.doc_id - d.firsts[lo];
+            uint64_t off = nb * 4ull;
+            for (uint32_t i = 0; i < local; ++i) off += lens[i];
+            if (off + lens[local] > block.size())
+                throw std::runtime_error("docstore block corrupt");
+            pb::LeafHit lh;
+            lh.leaf_json.assign((const char*)block.data() + off, lens[local]);
+            lh.partial_hit = h;
+            resp.hits.push_back(std::move(lh));
+        }
+        fill_buf(out, resp.encode());
+        return QW_OK;
+    } catch (const std::exception& e) {
+        set_err(ctx, e.what());
         return QW_ERR_INTERNAL;
     }
 }

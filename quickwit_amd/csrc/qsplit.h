@@ -97,12 +97,23 @@ struct FastFieldView {
     }
 };
 
+// optional row store (fetch_docs phase 2): zlib blocks + block index
+// (u32 first_doc[n+1] then u64 block_off[n+1]) — splitgen._build_docstore
+struct DocStoreView {
+    bool present = false;
+    uint32_t n_blocks = 0;
+    const uint32_t* firsts = nullptr;  // [n_blocks+1], last = num_docs
+    const uint64_t* offs = nullptr;    // [n_blocks+1] into blocks
+    const uint8_t* blocks = nullptr;
+};
+
 struct SplitView {
     std::string split_id;
     uint32_t num_docs = 0;
     std::string timestamp_field;
     std::vector<TextFieldView> text_fields;
     std::vector<FastFieldView> fast_fields;
+    DocStoreView docstore;
     const uint8_t* base = nullptr;  // host copy of the file image
     size_t file_len = 0;
 
@@ -141,6 +152,16 @@ struct SplitView {
         num_docs = uint32_t(meta->at("num_docs")->as_i64());
         const mj::Value* tsf = meta->get("timestamp_field");
         if (tsf && !tsf->is_null()) timestamp_field = tsf->s;
+        if (const mj::Value* ds = meta->get("docstore")) {
+            docstore.present = true;
+            docstore.n_blocks = uint32_t(ds->at("n_blocks")->as_i64());
+            Section idx = sec(ds, "doc_index");
+            Section blk = sec(ds, "doc_blocks");
+            docstore.firsts = (const uint32_t*)(data + idx.off);
+            docstore.offs = (const uint64_t*)(data + idx.off +
+                                              (docstore.n_blocks + 1) * 4);
+            docstore.blocks = data + blk.off;
+        }
         for (auto& fv : meta->at("fields")->arr) {
             const mj::Value* f = fv.get();
             std::string ty = f->at("type")->s;

@@ -72,6 +72,20 @@ SCHEMAS = {
         8: ("doc_mappers", "*str"),
         9: ("index_uris", "*str"),
     },
+    "FetchDocsRequest": {      # search.proto:674
+        1: ("partial_hits", "*msg:PartialHit"),
+        3: ("split_offsets", "*msg:SplitIdAndFooterOffsets"),
+        4: ("index_uri", "str"),
+        6: ("doc_mapper", "str"),
+    },
+    "LeafHit": {               # search.proto:553
+        1: ("leaf_json", "str"),
+        2: ("partial_hit", "msg:PartialHit"),
+        3: ("leaf_snippet_json", "str"),
+    },
+    "FetchDocsResponse": {     # search.proto:695
+        1: ("hits", "*msg:LeafHit"),
+    },
     "ListTermsRequest": {      # search.proto:700
         1: ("index_id_patterns", "*str"),
         3: ("field", "str"),
@@ -217,6 +231,7 @@ def encode(msg_name: str, d: dict) -> bytes:
 
 
 _OPTIONAL = {
+    ("LeafHit", "leaf_snippet_json"),
     ("ListTermsRequest", "start_timestamp"),
     ("ListTermsRequest", "end_timestamp"),
     ("ListTermsRequest", "max_hits"),

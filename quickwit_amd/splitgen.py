@@ -284,9 +284,10 @@ class SplitWriter:
     """Document-at-a-time writer for small corpora (golden tests).
     Bulk columnar path: build_split_from_columns."""
 
-    def __init__(self, schema: dict, split_id: str):
+    def __init__(self, schema: dict, split_id: str, store_docs: bool = True):
         self.schema = schema
         self.split_id = split_id
+        self.store_docs = store_docs  # write the docstore (fetch_docs phase 2)
         self.docs = []
 
     def add_documents(self, docs):
@@ -311,7 +312,8 @@ class SplitWriter:
                         v = parse_datetime_ms(v)
                     vals.append(v if v is not None else 0)
                 fast_inputs[name] = (vals, np.array(pres, dtype=bool))
-        return _assemble(self.schema, self.split_id, num_docs, text_inputs, fast_inputs)
+        return _assemble(self.schema, self.split_id, num_docs, text_inputs,
+                         fast_inputs, docs=self.docs if self.store_docs else None)
 
 
 def _postings_from_tokens(per_doc_tokens):
@@ -338,7 +340,47 @@ def _postings_from_tokens(per_doc_tokens):
     return vocab, term_ids, doc_ids, tfs, lengths
 
 
-def _assemble(schema, split_id, num_docs, text_inputs, fast_inputs, precomputed_text=None):
+DOCSTORE_BLOCK_BYTES = 16 * 1024  # uncompressed block target (tantivy-shaped)
+
+
+def _build_docstore(sec, docs):
+    """Row store: zlib-compressed blocks of canonical-JSON docs + block index
+    (u32 first_doc[n+1], u64 block_off[n+1]). The on-disk encoding is ours
+    (tantivy's docstore byte layout is unpinned); the reference
+    shape is the same: compressed blocks + doc->block checkpoints."""
+    import zlib
+    doc_jsons = [json.dumps(d, sort_keys=True, separators=(",", ":")).encode()
+                 for d in docs]
+    firsts, offs, blobs = [], [], []
+    pos = 0
+    i = 0
+    while i < len(doc_jsons):
+        j = i
+        size = 0
+        while j < len(doc_jsons) and (j == i or size < DOCSTORE_BLOCK_BYTES):
+            size += len(doc_jsons[j]) + 4
+            j += 1
+        block = np.array([len(b) for b in doc_jsons[i:j]], dtype="<u4").tobytes()
+        block += b"".join(doc_jsons[i:j])
+        comp = zlib.compress(block, 6)
+        firsts.append(i)
+        offs.append(pos)
+        blobs.append(comp)
+        pos += len(comp)
+        i = j
+    firsts.append(len(doc_jsons))
+    offs.append(pos)
+    index = (np.array(firsts, dtype="<u4").tobytes() +
+             np.array(offs, dtype="<u8").tobytes())
+    return {
+        "n_blocks": len(blobs),
+        "doc_index": sec.add(np.frombuffer(index, dtype=np.uint8)),
+        "doc_blocks": sec.add(np.frombuffer(b"".join(blobs) or b"\0", dtype=np.uint8)),
+    }
+
+
+def _assemble(schema, split_id, num_docs, text_inputs, fast_inputs, precomputed_text=None,
+              docs=None):
     sec = _Sections()
     fields_meta = []
     for fspec in schema["fields"]:
@@ -365,6 +407,8 @@ def _assemble(schema, split_id, num_docs, text_inputs, fast_inputs, precomputed_
         "timestamp_field": schema.get("timestamp_field"),
         "fields": fields_meta,
     }
+    if docs is not None:
+        meta["docstore"] = _build_docstore(sec, docs)
     meta_b = json.dumps(meta, sort_keys=True).encode("utf-8")
     meta_off = sec.pos
     sec.chunks.append(meta_b)

@@ -673,3 +673,65 @@ def test_leaf_list_terms_missing_field(searchers):
     fs = resp.get("failed_splits", [])
     assert fs and "couldn't get field" in fs[0]["error"]
     assert resp.get("num_attempted_splits", 0) == 0
+
+
+# ------------------------------------------------- fetch_docs phase 2
+# (fetch_docs.rs; root.rs:903): search -> fetch the top-K docs' stored JSON
+def test_fetch_docs_roundtrip():
+    import random
+    rng = random.Random(3)
+    docs = []
+    for i in range(3000):
+        docs.append({
+            "timestamp": 1700000000 + rng.randrange(100000),
+            "tenant_id": rng.randrange(50),
+            "severity_text": rng.choice(["INFO", "ERROR", "WARN"]),
+            "body": " ".join("tok%03d" % rng.randrange(200)
+                             for _ in range(rng.randrange(3, 20)))})
+    w = splitgen.SplitWriter(SCHEMA, "fd-split")
+    w.add_documents(docs)
+    gpu = GpuSearcher(device=0)
+    gpu.add_split("fd-split", w.finalize())
+    req = make_leaf_request({"type": "term", "field": "severity_text",
+                             "value": "ERROR"}, SCHEMA,
+                            [("fd-split", len(docs))], max_hits=25)
+    resp = gpu.leaf_search(req)
+    hits = resp["partial_hits"]
+    assert hits
+    fresp = gpu.fetch_docs({
+        "partial_hits": hits,
+        "split_offsets": [{"split_id": "fd-split", "num_docs": len(docs)}],
+        "doc_mapper": json.dumps(SCHEMA)})
+    got = fresp["hits"]
+    assert len(got) == len(hits)
+    for lh in got:
+        doc_id = lh["partial_hit"].get("doc_id", 0)
+        stored = json.loads(lh["leaf_json"])
+        assert stored == docs[doc_id]
+        assert stored["severity_text"] == "ERROR"
+
+
+def test_fetch_docs_multi_split_and_errors():
+    w1 = splitgen.SplitWriter(SCHEMA, "fd-a")
+    w1.add_documents([{"timestamp": 1, "tenant_id": 1,
+                       "severity_text": "INFO", "body": "aa"}] * 10)
+    w2 = splitgen.SplitWriter(SCHEMA, "fd-b")
+    w2.add_documents([{"timestamp": 2, "tenant_id": 2,
+                       "severity_text": "WARN", "body": "bb"}] * 10)
+    gpu = GpuSearcher(device=0)
+    gpu.add_split("fd-a", w1.finalize())
+    gpu.add_split("fd-b", w2.finalize())
+    fresp = gpu.fetch_docs({
+        "partial_hits": [{"split_id": "fd-b", "doc_id": 3},
+                         {"split_id": "fd-a", "doc_id": 7},
+                         {"split_id": "fd-a"}],  # doc_id 0 (proto3 default)
+        "split_offsets": [{"split_id": "fd-a", "num_docs": 10},
+                          {"split_id": "fd-b", "num_docs": 10}],
+        "doc_mapper": json.dumps(SCHEMA)})
+    hits = fresp["hits"]
+    assert [(h["partial_hit"].get("split_id"), h["partial_hit"].get("doc_id", 0))
+            for h in hits] == [("fd-a", 0), ("fd-a", 7), ("fd-b", 3)]
+    assert json.loads(hits[2]["leaf_json"])["severity_text"] == "WARN"
+    with pytest.raises(RuntimeError, match="unknown split"):
+        gpu.fetch_docs({"partial_hits": [{"split_id": "nope", "doc_id": 0}],
+                        "split_offsets": [], "doc_mapper": "{}"})
