@@ -86,7 +86,7 @@ def test_bulk_matches_doc_writer():
     bulk = splitgen.build_split_from_columns(
         schema, "x", n, {"body": mat}, {"body": vocab}, {})
     docs = [{"body": " ".join(vocab[t] for t in row)} for row in mat]
-    w = splitgen.SplitWriter(schema, "x")
+    w = splitgen.SplitWriter(schema, "x", store_docs=False)
     w.add_documents(docs)
     slow = w.finalize()
     assert bulk == slow
