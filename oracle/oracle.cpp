@@ -613,10 +613,11 @@ static SplitResult search_split(const SplitView& sv, const pb::SearchRequest& re
             return h;
         };
         std::vector<pb::PartialHit> cand;
+        bool doc_order_shortcut = specs.empty() && !req.search_after;
         if (m.all) {
             // no clause work: hits are just doc ids; default order doc desc
             cand.reserve(std::min<uint64_t>(sv.num_docs, leaf_max_hits * 4 + 16));
-            if (specs.empty()) {
+            if (doc_order_shortcut) {
                 // doc-id sort: take head/tail directly
                 uint64_t k = std::min<uint64_t>(leaf_max_hits, sv.num_docs);
                 for (uint64_t i = 0; i < k; ++i) {
@@ -627,7 +628,7 @@ static SplitResult search_split(const SplitView& sv, const pb::SearchRequest& re
                 for (uint32_t doc = 0; doc < sv.num_docs; ++doc)
                     cand.push_back(mk_hit(doc, 0.f));
             }
-        } else if (specs.empty()) {
+        } else if (doc_order_shortcut) {
             uint64_t k = std::min<uint64_t>(leaf_max_hits, m.docs.size());
             for (uint64_t i = 0; i < k; ++i) {
                 uint32_t doc = order1 == 1 ? m.docs[m.docs.size() - 1 - i] : m.docs[i];
@@ -637,6 +638,16 @@ static SplitResult search_split(const SplitView& sv, const pb::SearchRequest& re
             cand.reserve(m.docs.size());
             for (size_t i = 0; i < m.docs.size(); ++i)
                 cand.push_back(mk_hit(m.docs[i], m.scores.empty() ? 0.f : m.scores[i]));
+        }
+        // search_after: keep only hits strictly after the cursor
+        // (top_k_collector.rs:663-700; num_hits counting is unaffected)
+        if (req.search_after) {
+            const pb::PartialHit& c = *req.search_after;
+            cand.erase(std::remove_if(cand.begin(), cand.end(),
+                                      [&](const pb::PartialHit& h) {
+                                          return !after_cursor(h, c, order1, order2);
+                                      }),
+                       cand.end());
         }
         auto cmp = [&](const pb::PartialHit& a, const pb::PartialHit& b) {
             return hit_before(a, b, order1, order2);

@@ -25,6 +25,7 @@
 #include <cstdint>
 #include <cstdlib>
 #include <cstring>
+#include <functional>
 #include <list>
 #include <map>
 #include <memory>
@@ -199,7 +200,7 @@ struct qw_ctx {
     std::string last_error;
     hipStream_t stream = nullptr;
     hipEvent_t ev_start = nullptr, ev_stop = nullptr;
-    qw::DevBuf d_scratch, d_results, d_survivors;
+    qw::DevBuf d_scratch, d_results, d_survivors, d_cand2;
     std::map<std::string, qw::KernelTimer> timers;
 };
 
@@ -710,6 +711,12 @@ static pb::SortByValue sort_value_of(const SortSpec& s, uint32_t doc, float scor
     return v;
 }
 
+static uint32_t f32_sortable_h(float f) {
+    uint32_t b;
+    memcpy(&b, &f, 4);
+    return (b & 0x80000000u) ? ~b : (b | 0x80000000u);
+}
+
 static void record_kernel_time(qw_ctx* ctx, const char* name, float ms) {
     KernelTimer& t = ctx->timers[name];
     t.total_ms += ms;
@@ -796,7 +803,9 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
     uint32_t n_tiles = (sv.num_docs + TILE_DOCS - 1) / TILE_DOCS;
     bool pure_match_all = fq.match_all && fq.preds.empty() && fq.terms.empty();
     // hits trivially enumerable only under doc-id order (leaf.rs default)
-    bool trivial_hits = pure_match_all && specs.empty();
+    const pb::PartialHit* after =
+        req.search_after ? &*req.search_after : nullptr;
+    bool trivial_hits = pure_match_all && specs.empty() && !after;
     // candidate collection: needed unless hits are trivially enumerable
     bool collect = leaf_max_hits > 0 && !trivial_hits && !fq.match_none;
     bool need_kernel = !pure_match_all || do_aggs || collect;
@@ -900,6 +909,8 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
 
     uint64_t matched = 0;
     std::vector<uint64_t> top_keys;  // survivors, sorted best-first
+    std::function<void(uint64_t)> rerun_select;  // search_after retry hook
+    uint64_t sel_band_n = 0, sel_kwant = 0;
 
     if (need_kernel) {
         ctx->d_scratch.ensure(scratch_bytes);
@@ -1006,11 +1017,71 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         // ---- top-K selection over candidates (device histogram refinement
         // + host exact sort of the survivors; top_k_collector.rs semantics)
         if (collect && cand_n > 0) {
-            uint64_t K = std::min<uint64_t>(leaf_max_hits, cand_n);
+            uint32_t rw = wide ? 2 : 1;  // u64 words per candidate record
             uint64_t* d_cand = (uint64_t*)(ctx->d_results.p + r_cand);
             uint32_t* d_hist = (uint32_t*)(ctx->d_results.p + r_hist);
             uint32_t* d_scount = (uint32_t*)(ctx->d_results.p + r_cand_count) + 1;
+            // search_after: pre-compact to the cursor's primary-value band
+            // (inclusive upper bound in device-key space; ties and the
+            // (v2, split, seg, doc) chain are settled by the exact host
+            // filter at hit building, with a growing-K retry on shortfall)
+            if (after) {
+                uint64_t ceil_key = ~0ull;
+                SortKey k1 = sort_key_of(after->sort_value);
+                if (wide) {
+                    if (!k1.has) ceil_key = 0;  // cursor in the None region
+                    else {
+                        uint64_t S = k1.key;
+                        if (!specs.empty() && specs[0].comp == SortSpec::SCORE) {
+                            uint32_t kh = f32_sortable_h(float(u64_to_f64(S)));
+                            S = (uint64_t(kh) << 32) | 0xFFFFFFFFull;
+                        } else if (!specs.empty() && specs[0].ff &&
+                                   specs[0].ff->type == FastFieldView::DATETIME) {
+                            int64_t ns = u64_to_i64(S);  // cursor carries ns
+                            int64_t ms = ns / 1000000 - ((ns % 1000000) < 0 ? 1 : 0);
+                            S = i64_to_u64(ms);
+                        }
+                        // I64: same i64_to_u64 map; U64/STR ord: identity
+                        ceil_key = (order1 == 0) ? ~S : S;
+                    }
+                } else {
+                    uint32_t kh = 0;
+                    if (fq.scoring && k1.has)
+                        kh = f32_sortable_h(float(u64_to_f64(k1.key)));
+                    bool nasc = !specs.empty() &&
+                                specs[0].comp != SortSpec::DOC_ID && order1 == 0;
+                    if (nasc) kh = ~kh;
+                    ceil_key = (uint64_t(kh) << 32) | 0xFFFFFFFFull;
+                }
+                ctx->d_cand2.ensure(size_t(cand_n) * 8 * rw + 16);
+                uint32_t* d_c2n = (uint32_t*)(ctx->d_results.p + r_cand_count) + 2;
+                HIP_CHECK(hipMemsetAsync(d_c2n, 0, 4, ctx->stream));
+                uint32_t pgrid = std::min<uint32_t>(2048, (cand_n + 255) / 256);
+                if (wide)
+                    hipLaunchKernelGGL(k_cand_compact_w, dim3(pgrid), dim3(256), 0,
+                                       ctx->stream, d_cand, cand_n, 0ull, ceil_key,
+                                       (uint64_t*)ctx->d_cand2.p, d_c2n, cand_n);
+                else
+                    hipLaunchKernelGGL(k_cand_compact, dim3(pgrid), dim3(256), 0,
+                                       ctx->stream, d_cand, cand_n, 0ull, ceil_key,
+                                       (uint64_t*)ctx->d_cand2.p, d_c2n, cand_n);
+                uint32_t n2 = 0;
+                HIP_CHECK(hipMemcpyAsync(&n2, d_c2n, 4, hipMemcpyDeviceToHost,
+                                         ctx->stream));
+                HIP_CHECK(hipStreamSynchronize(ctx->stream));
+                d_cand = (uint64_t*)ctx->d_cand2.p;
+                cand_n = n2;
+            }
+        }
+        if (collect && cand_n > 0) {
+            uint64_t* d_cand = after ? (uint64_t*)ctx->d_cand2.p
+                                     : (uint64_t*)(ctx->d_results.p + r_cand);
+            uint32_t* d_hist = (uint32_t*)(ctx->d_results.p + r_hist);
+            uint32_t* d_scount = (uint32_t*)(ctx->d_results.p + r_cand_count) + 1;
             uint32_t rw = wide ? 2 : 1;  // u64 words per candidate record
+            sel_band_n = cand_n;
+            rerun_select = [&, d_cand, rw](uint64_t Kwant) {
+            uint64_t K = std::min<uint64_t>(Kwant, cand_n);
             uint64_t prefix = 0;
             uint32_t prefix_bits = 0;
             uint64_t survivors = cand_n, above = 0, Krem = K;
@@ -1080,8 +1151,12 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             if (!wide) {
                 std::sort(top_keys.begin(), top_keys.end(),
                           std::greater<uint64_t>());
-                if (top_keys.size() > K) top_keys.resize(K);
+                if (top_keys.size() > K && !after) top_keys.resize(K);
             }
+            };  // rerun_select
+            sel_kwant = std::min<uint64_t>(
+                leaf_max_hits + (after ? 1024 : 0), cand_n);
+            rerun_select(sel_kwant);
         }
     } else {
         matched = sv.num_docs;  // pure match_all, no aggs
@@ -1109,43 +1184,68 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                                            : uint32_t(i);
                 out.hits.push_back(mk_hit(doc, 0.f));
             }
-        } else if (wide) {
-            // survivors carry (selection key, score|doc); the exact reference
-            // order ((sort_value, sort_value2, GlobalDocId), sorting.md:14-26)
-            // is re-established here over the survivor set
-            size_t n = top_keys.size() / 2;
-            out.hits.reserve(n);
-            for (size_t i = 0; i < n; ++i) {
-                uint64_t aux = top_keys[2 * i + 1];
-                uint32_t doc = uint32_t(aux);
-                float score = 0.f;
-                if (fq.scoring) {
-                    uint32_t b = uint32_t(aux >> 32);
-                    memcpy(&score, &b, 4);
-                }
-                out.hits.push_back(mk_hit(doc, score));
-            }
-            auto cmp = [&](const pb::PartialHit& a, const pb::PartialHit& b) {
-                return hit_before(a, b, order1, order2);
-            };
-            size_t k = std::min<size_t>(leaf_max_hits, out.hits.size());
-            std::partial_sort(out.hits.begin(), out.hits.begin() + k,
-                              out.hits.end(), cmp);
-            out.hits.resize(k);
         } else {
-            bool asc = !specs.empty() && order1 == 0;
-            for (uint64_t key : top_keys) {
-                uint32_t kh = uint32_t(key >> 32), kl = uint32_t(key);
-                if (asc) {
-                    kh = ~kh;
-                    kl = ~kl;
+            // build hits from the survivors; with search_after, the exact
+            // cursor filter runs before truncation and the selection widens
+            // (growing K) until the page is full or the band is exhausted
+            for (;;) {
+                out.hits.clear();
+                if (wide) {
+                    // survivors carry (selection key, score|doc); the exact
+                    // reference order ((sort_value, sort_value2, GlobalDocId),
+                    // sorting.md:14-26) is re-established over the survivors
+                    size_t n = top_keys.size() / 2;
+                    out.hits.reserve(n);
+                    for (size_t i = 0; i < n; ++i) {
+                        uint64_t aux = top_keys[2 * i + 1];
+                        uint32_t doc = uint32_t(aux);
+                        float score = 0.f;
+                        if (fq.scoring) {
+                            uint32_t b = uint32_t(aux >> 32);
+                            memcpy(&score, &b, 4);
+                        }
+                        out.hits.push_back(mk_hit(doc, score));
+                    }
+                } else {
+                    bool asc = !specs.empty() && order1 == 0;
+                    for (uint64_t key : top_keys) {
+                        uint32_t kh = uint32_t(key >> 32), kl = uint32_t(key);
+                        if (asc) {
+                            kh = ~kh;
+                            kl = ~kl;
+                        }
+                        float score = 0.f;
+                        if (fq.scoring) {
+                            uint32_t b =
+                                (kh & 0x80000000u) ? (kh & ~0x80000000u) : ~kh;
+                            memcpy(&score, &b, 4);
+                        }
+                        out.hits.push_back(mk_hit(kl, score));
+                    }
                 }
-                float score = 0.f;
-                if (fq.scoring) {
-                    uint32_t b = (kh & 0x80000000u) ? (kh & ~0x80000000u) : ~kh;
-                    memcpy(&score, &b, 4);
+                if (after) {
+                    const pb::PartialHit& c = *after;
+                    out.hits.erase(
+                        std::remove_if(out.hits.begin(), out.hits.end(),
+                                       [&](const pb::PartialHit& h) {
+                                           return !after_cursor(h, c, order1,
+                                                                order2);
+                                       }),
+                        out.hits.end());
                 }
-                out.hits.push_back(mk_hit(kl, score));
+                auto cmp = [&](const pb::PartialHit& a, const pb::PartialHit& b) {
+                    return hit_before(a, b, order1, order2);
+                };
+                size_t k = std::min<size_t>(leaf_max_hits, out.hits.size());
+                std::partial_sort(out.hits.begin(), out.hits.begin() + k,
+                                  out.hits.end(), cmp);
+                out.hits.resize(k);
+                if (!after || !rerun_select ||
+                    out.hits.size() >= std::min<uint64_t>(leaf_max_hits, sel_band_n) ||
+                    sel_kwant >= sel_band_n)
+                    break;
+                sel_kwant = std::min<uint64_t>(sel_band_n, sel_kwant * 8);
+                rerun_select(sel_kwant);
             }
         }
     }

@@ -75,4 +75,40 @@ inline bool hit_before(const pb::PartialHit& a, const pb::PartialHit& b, int ord
     return false;
 }
 
+// search_after filter (top_k_collector.rs:663-700 + SearchAfterSegment
+// :821-872): a hit is kept iff the comparison chain
+// (sort_value, sort_value2, then — only when the cursor carries a doc
+// address — split_id, segment_ord, doc_id) is strictly Less under
+// SortOrder::compare (Desc: natural, Asc: reversed; Some > None).
+inline bool after_cursor(const pb::PartialHit& h, const pb::PartialHit& c,
+                         int order1, int order2) {
+    auto cmpo = [](const SortKey& a, const SortKey& b, int order) -> int {
+        if (a.has && b.has) {
+            if (a.key == b.key) return 0;
+            int n = a.key > b.key ? 1 : -1;
+            return order == 1 ? n : -n;
+        }
+        if (a.has) return 1;   // (Some, None) -> Greater
+        if (b.has) return -1;  // (None, Some) -> Less
+        return 0;
+    };
+    int r = cmpo(sort_key_of(h.sort_value), sort_key_of(c.sort_value), order1);
+    if (r) return r < 0;
+    r = cmpo(sort_key_of(h.sort_value2), sort_key_of(c.sort_value2), order2);
+    if (r) return r < 0;
+    if (c.split_id.empty()) return false;  // equal values, no doc tiebreak
+    int cs = h.split_id.compare(c.split_id);
+    if (order1 != 1) cs = -cs;
+    if (cs) return cs < 0;
+    if (h.segment_ord != c.segment_ord) {
+        bool n = h.segment_ord > c.segment_ord;
+        return order1 == 1 ? !n : n;
+    }
+    if (h.doc_id != c.doc_id) {
+        bool n = h.doc_id > c.doc_id;
+        return order1 == 1 ? !n : n;
+    }
+    return false;  // identical to the cursor: excluded (not inclusive)
+}
+
 }  // namespace qw

@@ -735,3 +735,80 @@ def test_fetch_docs_multi_split_and_errors():
     with pytest.raises(RuntimeError, match="unknown split"):
         gpu.fetch_docs({"partial_hits": [{"split_id": "nope", "doc_id": 0}],
                         "split_offsets": [], "doc_mapper": "{}"})
+
+
+# ------------------------------------------------- search_after pagination
+# (top_k_collector.rs:663-700 generic filter + SearchAfterSegment :821-872)
+def paginate(searchers, query, sort_fields, page, npages, **kw):
+    gpu, cpu = searchers
+    g_pages, e_pages = [], []
+    g_after = e_after = None
+    for _ in range(npages):
+        greq = make_leaf_request(query, SCHEMA, [(SID, NDOCS)], max_hits=page,
+                                 sort_fields=sort_fields, **kw)
+        ereq = json.loads(json.dumps(greq))
+        if g_after is not None:
+            greq["search_request"]["search_after"] = g_after
+            ereq["search_request"]["search_after"] = e_after
+        got = gpu.leaf_search(greq)
+        exp = cpu.leaf_search(ereq)
+        gh = got.get("partial_hits", [])
+        eh = exp.get("partial_hits", [])
+        g_pages.append(gh)
+        e_pages.append(eh)
+        if not gh or not eh:
+            break
+        g_after, e_after = gh[-1], eh[-1]
+    return g_pages, e_pages
+
+
+def test_search_after_timestamp_pages(searchers):
+    g, e = paginate(searchers, TERM_Q,
+                    [{"field_name": "timestamp", "sort_order": 1}], 20, 4)
+    for gp, ep in zip(g, e):
+        assert [(hid(h), sv_of(h)) for h in gp] == \
+               [(hid(h), sv_of(h)) for h in ep]
+    # pages must not overlap
+    seen = [hid(h) for p in g for h in p]
+    assert len(seen) == len(set(seen))
+
+
+def test_search_after_score_pages(searchers):
+    q = {"type": "bool", "should": [
+        {"type": "term", "field": "body", "value": "w%05d" % i}
+        for i in range(3)]}
+    g, e = paginate(searchers, q,
+                    [{"field_name": "_score", "sort_order": 1}], 15, 3)
+    for gp, ep in zip(g, e):
+        assert [hid(h) for h in gp] == [hid(h) for h in ep]
+        for gh, eh in zip(gp, ep):
+            assert math.isclose(hscore(gh), hscore(eh), rel_tol=REL,
+                                abs_tol=1e-9)
+    seen = [hid(h) for p in g for h in p]
+    assert len(seen) == len(set(seen))
+
+
+def test_search_after_doc_order_pages(searchers):
+    g, e = paginate(searchers, TERM_Q, None, 25, 3)
+    for gp, ep in zip(g, e):
+        assert [hid(h) for h in gp] == [hid(h) for h in ep]
+    seen = [hid(h) for p in g for p_h in [p] for h in p_h]
+    assert len(seen) == len(set(seen))
+
+
+def test_search_after_value_only_cursor(searchers):
+    # cursor without a doc address: equal sort values are EXCLUDED
+    gpu, cpu = searchers
+    req = make_leaf_request(TERM_Q, SCHEMA, [(SID, NDOCS)], max_hits=10,
+                            sort_fields=[{"field_name": "tenant_id",
+                                          "sort_order": 1}])
+    first = gpu.leaf_search(req)["partial_hits"]
+    cursor = {"sort_value": first[-1]["sort_value"]}  # no split/doc
+    req2 = make_leaf_request(TERM_Q, SCHEMA, [(SID, NDOCS)], max_hits=10,
+                             sort_fields=[{"field_name": "tenant_id",
+                                           "sort_order": 1}])
+    req2["search_request"]["search_after"] = cursor
+    got = gpu.leaf_search(req2)
+    exp = cpu.leaf_search(req2)
+    assert [(hid(h), sv_of(h)) for h in got.get("partial_hits", [])] == \
+           [(hid(h), sv_of(h)) for h in exp.get("partial_hits", [])]
