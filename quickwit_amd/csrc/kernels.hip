@@ -165,15 +165,21 @@ __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t t
                 extract_bits_pair(idbase, j0 + 2, w, &g2, &g3);
             }
         }
-        // inclusive 32-lane scan of the per-lane gap sums
+        // inclusive scan of the per-lane gap sums WITHIN the 32-element
+        // segment only (8 sub-lanes): the block's v2 anchor words
+        // [pad, doc31, doc63, doc95] sit just before the gaps and give the
+        // segment base, cutting the dependent shuffle chain from 5 to 3
         uint32_t sum = g0 + g1 + g2 + g3;
         #pragma unroll
-        for (int d = 1; d < 32; d <<= 1) {
-            uint32_t n = __shfl_up(sum, d, 32);
-            if (sl >= uint32_t(d)) sum += n;
+        for (int d = 1; d < 8; d <<= 1) {
+            uint32_t n = __shfl_up(sum, d, 8);
+            if ((sl & 7u) >= uint32_t(d)) sum += n;
         }
         if (!live) continue;
-        uint32_t doc3 = e.first_doc + sum;
+        uint32_t seg = sl >> 3;
+        uint32_t base = seg == 0 ? e.first_doc
+                                 : ((const uint32_t*)idbase)[int(seg) - 4];
+        uint32_t doc3 = base + sum;
         uint32_t doc2 = doc3 - g3;
         uint32_t doc1 = doc2 - g2;
         uint32_t doc0 = doc1 - g1;
@@ -683,11 +689,14 @@ inline uint32_t launch_leaf_tile(bool ns, bool nb, bool na, bool nc, dim3 grid,
 // ----------------------------------------------------- top-K selection passes
 // histogram of the top 12 bits (after `shift`) of candidates matching
 // (key >> prefix_shift) == prefix
-extern "C" __global__ void k_cand_hist(const uint64_t* cand, uint32_t n, uint64_t prefix,
-                                       uint32_t prefix_bits, uint32_t* hist) {
+extern "C" __global__ void k_cand_hist(const uint64_t* cand, const uint32_t* n_ptr,
+                                       uint64_t prefix, uint32_t prefix_bits,
+                                       uint32_t* hist) {
     __shared__ uint32_t lh[TOPK_BINS];
     for (uint32_t i = threadIdx.x; i < TOPK_BINS; i += blockDim.x) lh[i] = 0;
     __syncthreads();
+    uint32_t n = *n_ptr;  // device-side count: pass 0 is enqueued without a
+                          // host round trip for the candidate count
     uint32_t shift = 64 - prefix_bits - 12;
     for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
          i += gridDim.x * blockDim.x) {
@@ -701,12 +710,13 @@ extern "C" __global__ void k_cand_hist(const uint64_t* cand, uint32_t n, uint64_
 }
 
 // wide-record (16B {u64 key, u64 aux}) variants of the selection passes
-extern "C" __global__ void k_cand_hist_w(const uint64_t* cand, uint32_t n,
+extern "C" __global__ void k_cand_hist_w(const uint64_t* cand, const uint32_t* n_ptr,
                                          uint64_t prefix, uint32_t prefix_bits,
                                          uint32_t* hist) {
     __shared__ uint32_t lh[TOPK_BINS];
     for (uint32_t i = threadIdx.x; i < TOPK_BINS; i += blockDim.x) lh[i] = 0;
     __syncthreads();
+    uint32_t n = *n_ptr;
     uint32_t shift = 64 - prefix_bits - 12;
     for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
          i += gridDim.x * blockDim.x) {

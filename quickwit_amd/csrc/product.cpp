@@ -911,6 +911,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
     std::vector<uint64_t> top_keys;  // survivors, sorted best-first
     std::function<void(uint64_t)> rerun_select;  // search_after retry hook
     uint64_t sel_band_n = 0, sel_kwant = 0;
+    std::vector<uint32_t> hist0(TOPK_BINS);  // prefetched selection pass 0
 
     if (need_kernel) {
         ctx->d_scratch.ensure(scratch_bytes);
@@ -998,6 +999,26 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                          n_tiles, 1u);
         HIP_CHECK(hipEventRecord(ctx->ev_stop, ctx->stream));
         HIP_CHECK(hipGetLastError());
+
+        // enqueue top-K histogram pass 0 behind the main kernel (count read
+        // device-side) so its result arrives with the same synchronize
+        bool hist0_valid = false;
+        if (collect && !after) {
+            uint64_t* d_cand0 = (uint64_t*)(ctx->d_results.p + r_cand);
+            uint32_t* d_hist0 = (uint32_t*)(ctx->d_results.p + r_hist);
+            uint32_t* d_n0 = (uint32_t*)(ctx->d_results.p + r_cand_count);
+            HIP_CHECK(hipMemsetAsync(d_hist0, 0, TOPK_BINS * 4, ctx->stream));
+            uint32_t hgrid = std::min<uint32_t>(512, (n_tiles * 16 + 4095) / 4096);
+            if (wide)
+                hipLaunchKernelGGL(k_cand_hist_w, dim3(hgrid), dim3(256), 0,
+                                   ctx->stream, d_cand0, d_n0, 0ull, 0u, d_hist0);
+            else
+                hipLaunchKernelGGL(k_cand_hist, dim3(hgrid), dim3(256), 0,
+                                   ctx->stream, d_cand0, d_n0, 0ull, 0u, d_hist0);
+            HIP_CHECK(hipMemcpyAsync(hist0.data(), d_hist0, TOPK_BINS * 4,
+                                     hipMemcpyDeviceToHost, ctx->stream));
+            hist0_valid = true;
+        }
 
         // ---- download counts
         std::vector<uint32_t> tile_counts(n_tiles);
@@ -1093,21 +1114,28 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     if (prefix_bits == 0) floor_key = 0;
                     break;
                 }
-                HIP_CHECK(hipMemsetAsync(d_hist, 0, TOPK_BINS * 4, ctx->stream));
-                // grid small enough that the per-WG 4096-bin LDS flush (global
-                // atomics ∝ grid) stays cheap, big enough to fill the chip
-                uint32_t hgrid = std::min<uint32_t>(512, (cand_n + 4095) / 4096);
-                if (wide)
-                    hipLaunchKernelGGL(k_cand_hist_w, dim3(hgrid), dim3(256), 0,
-                                       ctx->stream, d_cand, cand_n, prefix,
-                                       prefix_bits, d_hist);
-                else
-                    hipLaunchKernelGGL(k_cand_hist, dim3(hgrid), dim3(256), 0,
-                                       ctx->stream, d_cand, cand_n, prefix,
-                                       prefix_bits, d_hist);
-                HIP_CHECK(hipMemcpyAsync(hist.data(), d_hist, TOPK_BINS * 4,
-                                         hipMemcpyDeviceToHost, ctx->stream));
-                HIP_CHECK(hipStreamSynchronize(ctx->stream));
+                if (prefix_bits == 0 && hist0_valid) {
+                    hist = hist0;  // prefetched with the main kernel's sync
+                } else {
+                    HIP_CHECK(hipMemsetAsync(d_hist, 0, TOPK_BINS * 4, ctx->stream));
+                    // grid small enough that the per-WG 4096-bin LDS flush
+                    // (global atomics ∝ grid) stays cheap, yet fills the chip
+                    uint32_t hgrid = std::min<uint32_t>(512, (cand_n + 4095) / 4096);
+                    uint32_t* d_n = after
+                        ? (uint32_t*)(ctx->d_results.p + r_cand_count) + 2
+                        : (uint32_t*)(ctx->d_results.p + r_cand_count);
+                    if (wide)
+                        hipLaunchKernelGGL(k_cand_hist_w, dim3(hgrid), dim3(256), 0,
+                                           ctx->stream, d_cand, d_n, prefix,
+                                           prefix_bits, d_hist);
+                    else
+                        hipLaunchKernelGGL(k_cand_hist, dim3(hgrid), dim3(256), 0,
+                                           ctx->stream, d_cand, d_n, prefix,
+                                           prefix_bits, d_hist);
+                    HIP_CHECK(hipMemcpyAsync(hist.data(), d_hist, TOPK_BINS * 4,
+                                             hipMemcpyDeviceToHost, ctx->stream));
+                    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+                }
                 uint64_t cum = 0;
                 int b = TOPK_BINS - 1;
                 for (; b >= 0; --b) {
@@ -1478,6 +1506,10 @@ int32_t qw_ctx_add_split(qw_ctx* ctx, const char* split_id, const uint8_t* data,
         auto ds = std::make_unique<DeviceSplit>();
         ds->host.assign(data, data + len);
         ds->view.parse(ds->host.data(), len);
+        if (ds->view.version < 2)
+            throw std::runtime_error(
+                "QWA1 v1 container lacks posting segment anchors (regenerate "
+                "the split with the current writer)");
         ds->len = len;
         HIP_CHECK(hipSetDevice(ctx->device));
         HIP_CHECK(hipMalloc(&ds->d_image, len + 64));  // +64: decode overread pad

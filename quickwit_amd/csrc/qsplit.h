@@ -109,6 +109,7 @@ struct DocStoreView {
 
 struct SplitView {
     std::string split_id;
+    uint32_t version = 1;
     uint32_t num_docs = 0;
     std::string timestamp_field;
     std::vector<TextFieldView> text_fields;
@@ -149,6 +150,7 @@ struct SplitView {
         if (meta_off + meta_len > len) throw std::runtime_error("bad QWA1 footer");
         mj::ValuePtr meta = mj::parse((const char*)data + meta_off, meta_len);
         split_id = meta->at("split_id")->s;
+        if (const mj::Value* v = meta->get("version")) version = uint32_t(v->as_i64());
         num_docs = uint32_t(meta->at("num_docs")->as_i64());
         const mj::Value* tsf = meta->get("timestamp_field");
         if (tsf && !tsf->is_null()) timestamp_field = tsf->s;
