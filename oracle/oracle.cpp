@@ -232,6 +232,10 @@ struct SplitSearcher {
                 return eval_bool(n);
             case PlanNode::WILDCARD:
                 return eval_wildcard(n);
+            case PlanNode::CACHE:
+                // semantics-transparent (cache_node.rs); memoization is a
+                // product-side device-bitmap concern
+                return eval(n.cache_inner.at(0));
         }
         return Match{};
     }

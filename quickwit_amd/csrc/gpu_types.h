@@ -33,6 +33,7 @@ enum PredType : uint32_t {
     PRED_RANGE_U64 = 0,
     PRED_RANGE_I64 = 1,  // i64 / datetime(ms)
     PRED_PRESENCE = 2,
+    PRED_BITSET = 3,     // device-resident HitSet (predicate cache, §8f.2)
 };
 enum PredFlags : uint32_t {
     PRED_NEGATED = 1,       // must_not predicate
@@ -49,6 +50,8 @@ struct PredDev {
     uint64_t values_off;   // column byte offset (0 for pure presence w/o col)
     uint64_t nulls_off;    // null bitmap byte offset (0 = non-nullable)
     uint32_t value_width;  // 8 for u64/i64; 1/2/4 for str ords
+    uint64_t abs_bitmap;   // PRED_BITSET: absolute device VA of u32 words
+                           // laid out tile-major (TILE_DOCS/32 words per tile)
 };
 
 enum AggKindDev : uint32_t { AGGD_HISTO = 0, AGGD_TERMS = 1 };
@@ -115,6 +118,8 @@ struct QueryDev {
     // column min/max => no range check), optional agg[1] LDS terms over a
     // non-nullable ord column. Descriptors hoisted to registers.
     uint32_t agg_fast;
+    uint64_t bitmap_out;    // absolute device VA: store each tile's match
+                            // bitset words here (predicate-cache fill); 0=off
     uint64_t terms_off;     // scratch offsets of descriptor arrays
     uint64_t preds_off;
     uint64_t aggs_off;

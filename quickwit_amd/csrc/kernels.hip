@@ -236,6 +236,11 @@ __device__ __forceinline__ bool eval_pred(const QueryDev& q, const PredDev& p,
         const uint64_t* nulls = (const uint64_t*)(q.split + p.nulls_off);
         ok = (nulls[doc >> 6] >> (doc & 63)) & 1;
     }
+    if (p.type == PRED_BITSET) {
+        const uint32_t* bm = (const uint32_t*)p.abs_bitmap;
+        ok = (bm[doc >> 5] >> (doc & 31)) & 1;
+        return (p.flags & PRED_NEGATED) ? !ok : ok;
+    }
     if (ok && p.type != PRED_PRESENCE) {
         if (p.type == PRED_RANGE_U64) {
             uint64_t v = ((const uint64_t*)(q.split + p.values_off))[doc];
@@ -579,6 +584,11 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
         // ---- per-tile count + phase B of collection (one global atomic per
         // TILE reserves contiguous cand space; ranks from an LDS bitset
         // prefix-scan; writes come out doc-ordered => coalesced)
+        if (NC && q.bitmap_out) {
+            __syncthreads();  // bits_m final
+            ((uint32_t*)q.bitmap_out)[uint64_t(tile) * (TILE_DOCS / 32) +
+                                      threadIdx.x] = sc_bits_m[threadIdx.x];
+        }
         if (NC) {
             __syncthreads();  // bits_m/score final
             uint32_t w = threadIdx.x;  // one 32-doc word per thread (256 words)

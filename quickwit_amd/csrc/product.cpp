@@ -189,6 +189,51 @@ static std::string leaf_cache_key(const pb::SplitIdAndFooterOffsets& so,
     return key;
 }
 
+// device HitSet bitmaps memoized per (split, subquery fingerprint) — the
+// MI355X-native PredicateCache (cache_node.rs:202-486).
+// Count-capped LRU; a 100M-doc bitmap is 12.5 MB of HBM.
+struct HitsetCache {
+    size_t cap = 256;
+    std::list<std::pair<std::string, uint8_t*>> lru;
+    std::map<std::string, std::list<std::pair<std::string, uint8_t*>>::iterator> idx;
+    uint64_t hits = 0, misses = 0;
+
+    uint8_t* get(const std::string& key) {
+        auto it = idx.find(key);
+        if (it == idx.end()) {
+            ++misses;
+            return nullptr;
+        }
+        ++hits;
+        lru.splice(lru.begin(), lru, it->second);
+        return it->second->second;
+    }
+    void put(const std::string& key, uint8_t* bm) {
+        lru.emplace_front(key, bm);
+        idx[key] = lru.begin();
+        while (lru.size() > cap) {
+            (void)hipFree(lru.back().second);
+            idx.erase(lru.back().first);
+            lru.pop_back();
+        }
+    }
+    void remove_split(const std::string& split_id) {
+        std::string prefix = split_id + '\0';
+        for (auto it = lru.begin(); it != lru.end();) {
+            if (it->first.compare(0, prefix.size(), prefix) == 0) {
+                (void)hipFree(it->second);
+                idx.erase(it->first);
+                it = lru.erase(it);
+            } else ++it;
+        }
+    }
+    void clear() {
+        for (auto& kv : lru) (void)hipFree(kv.second);
+        lru.clear();
+        idx.clear();
+    }
+};
+
 }  // namespace qw
 
 struct qw_ctx {
@@ -197,6 +242,7 @@ struct qw_ctx {
     int64_t agg_bucket_limit = 65000;
     std::map<std::string, std::unique_ptr<qw::DeviceSplit>> splits;
     qw::LeafCache leaf_cache;
+    qw::HitsetCache hitsets;
     std::string last_error;
     hipStream_t stream = nullptr;
     hipEvent_t ev_start = nullptr, ev_stop = nullptr;
@@ -240,6 +286,9 @@ struct FlatQuery {
     };
     std::vector<FTerm> terms;
     uint32_t n_must_groups = 0;
+    // CACHE nodes in filter position: resolved to device HitSet bitmaps
+    // (PRED_BITSET) by the caller (predicate cache)
+    std::vector<const PlanNode*> cache_nodes;
     std::vector<PredDev> preds;
     std::vector<const TextFieldView*> ktab_fields;  // ktab_idx -> field
 
@@ -339,6 +388,12 @@ static size_t add_wildcard(FlatQuery& fq, const SplitView& sv, const PlanNode& n
     return cnt;
 }
 
+// strip semantics-transparent cache wrappers (contexts that cannot use the
+// bitmap: should/must_not/scored root)
+static const PlanNode& uncache(const PlanNode& c) {
+    return c.kind == PlanNode::CACHE ? uncache(c.cache_inner.front()) : c;
+}
+
 static bool is_pure_should_terms(const PlanNode& n) {
     if (n.kind != PlanNode::BOOL) return false;
     if (!n.must.empty() || !n.must_not.empty() || !n.filter.empty()) return false;
@@ -366,6 +421,9 @@ static void flatten_bool(FlatQuery& fq, const SplitView& sv, const PlanNode& n,
 static void add_positive(FlatQuery& fq, const SplitView& sv, const PlanNode& c,
                          bool scored, float boost) {
     switch (c.kind) {
+        case PlanNode::CACHE:
+            fq.cache_nodes.push_back(&c);
+            break;
         case PlanNode::TERM:
             add_term(fq, sv, c.field, c.value, ROLE_MUST, boost * c.boost, scored,
                      fq.n_must_groups++);
@@ -415,7 +473,8 @@ static void flatten_bool(FlatQuery& fq, const SplitView& sv, const PlanNode& n,
 
     // shoulds: flatten TERM / pure-OR bool children into SHOULD terms
     size_t should_clauses = 0;
-    for (auto& c : n.should) {
+    for (auto& c_raw : n.should) {
+        const PlanNode& c = uncache(c_raw);
         if (c.kind == PlanNode::TERM) {
             add_term(fq, sv, c.field, c.value, ROLE_SHOULD, boost * c.boost,
                      fq.scoring);
@@ -437,7 +496,8 @@ static void flatten_bool(FlatQuery& fq, const SplitView& sv, const PlanNode& n,
         }
     }
 
-    for (auto& c : n.must_not) {
+    for (auto& c_raw : n.must_not) {
+        const PlanNode& c = uncache(c_raw);
         if (c.kind == PlanNode::TERM)
             add_term(fq, sv, c.field, c.value, ROLE_MUST_NOT, 1.f, false);
         else if (c.kind == PlanNode::RANGE || c.kind == PlanNode::FIELD_PRESENCE)
@@ -516,6 +576,14 @@ static FlatQuery flatten(const SplitView& sv, const PlanNode& plan, bool scoring
             add_wildcard(fq, sv, plan, ROLE_SHOULD, 0);
             if (fq.terms.empty()) fq.match_none = true;
             fq.msm = 1;
+            break;
+        case PlanNode::CACHE:
+            if (!scoring) {  // root cache: match_all base + cached bitmap
+                fq.match_all = true;
+                fq.cache_nodes.push_back(&plan);
+            } else {
+                return flatten(sv, uncache(plan), scoring);
+            }
             break;
         case PlanNode::BOOL:
             flatten_bool(fq, sv, plan, 1.f);
@@ -711,6 +779,124 @@ static pb::SortByValue sort_value_of(const SortSpec& s, uint32_t doc, float scor
     return v;
 }
 
+// Evaluate a CACHE node's inner subtree into a device HitSet bitmap
+// (tile-major u32 words) and memoize it per (split, fingerprint). The
+// assembly mirrors the hot-path one minus scoring/aggs/top-K: the query
+// shape becomes an unscored flattened boolean whose per-tile match bitset
+// is stored via QueryDev::bitmap_out.
+static uint8_t* resolve_hitset(qw_ctx* ctx, const DeviceSplit& ds,
+                               const PlanNode& inner, const Schema& schema) {
+    const SplitView& sv = ds.view;
+    std::string key = sv.split_id;
+    key += '\0';
+    plan_fingerprint(inner, key);
+    if (uint8_t* bm = ctx->hitsets.get(key)) return bm;
+
+    uint32_t n_tiles = (sv.num_docs + TILE_DOCS - 1) / TILE_DOCS;
+    size_t bm_bytes = size_t(n_tiles) * (TILE_DOCS / 32) * 4;
+    uint8_t* bm = nullptr;
+    HIP_CHECK(hipMalloc(&bm, bm_bytes));
+
+    FlatQuery fq = flatten(sv, inner, false);
+    for (const PlanNode* cn : fq.cache_nodes) {  // nested cache nodes
+        PredDev p{};
+        p.type = PRED_BITSET;
+        p.abs_bitmap =
+            (uint64_t)resolve_hitset(ctx, ds, cn->cache_inner.front(), schema);
+        fq.preds.push_back(p);
+    }
+    if (fq.match_none) {
+        HIP_CHECK(hipMemsetAsync(bm, 0, bm_bytes, ctx->stream));
+        HIP_CHECK(hipStreamSynchronize(ctx->stream));
+        ctx->hitsets.put(key, bm);
+        return bm;
+    }
+
+    // scratch: TermDev[] | PredDev[] | zero ktab | per-term block ranges
+    std::vector<TermDev> terms(fq.terms.size());
+    uint32_t n_must_not = 0;
+    for (size_t i = 0; i < fq.terms.size(); ++i) {
+        const FlatQuery::FTerm& t = fq.terms[i];
+        TermDev& d = terms[i];
+        d.role = t.role;
+        d.grp = t.grp;
+        d.n_blocks = t.f->h_n_blocks[t.tid];
+        d.skip_off = t.f->skip.off + t.f->h_skip_off[t.tid];
+        d.payload_off = t.f->payload.off;
+        d.norms_off = 0;
+        d.weight = 0.f;
+        d.ktab_idx = 0;
+        n_must_not += t.role == ROLE_MUST_NOT;
+    }
+    size_t off_terms = 0;
+    size_t off_preds = off_terms + terms.size() * sizeof(TermDev);
+    size_t off_ktabs = off_preds + fq.preds.size() * sizeof(PredDev);
+    size_t off_ranges = (off_ktabs + 256 * 4 + 63) & ~size_t(63);
+    size_t ranges_bytes = fq.terms.size() * 2ull * n_tiles * 4;
+    std::vector<uint32_t> ranges(ranges_bytes / 4, 0);
+    for (size_t i = 0; i < fq.terms.size(); ++i) {
+        const FlatQuery::FTerm& t = fq.terms[i];
+        const SkipEntry* sk = t.f->h_skip + t.f->h_skip_off[t.tid] / 16;
+        uint32_t nb = t.f->h_n_blocks[t.tid];
+        uint32_t* lo = ranges.data() + (i * 2ull * n_tiles);
+        uint32_t* hi = lo + n_tiles;
+        uint32_t b_lo = 0, b_hi = 0;
+        for (uint32_t tile = 0; tile < n_tiles; ++tile) {
+            uint32_t tlo = tile * TILE_DOCS;
+            uint32_t thi = std::min<uint32_t>(tlo + TILE_DOCS, sv.num_docs);
+            while (b_lo < nb && sk[b_lo].last_doc < tlo) ++b_lo;
+            if (b_hi < b_lo) b_hi = b_lo;
+            while (b_hi < nb && sk[b_hi].first_doc < thi) ++b_hi;
+            lo[tile] = b_lo;
+            hi[tile] = b_hi;
+        }
+        terms[i].ranges_off = off_ranges + i * 2ull * n_tiles * 4;
+    }
+    size_t scratch_bytes = off_ranges + ranges_bytes;
+    std::vector<uint8_t> scratch(scratch_bytes, 0);
+    memcpy(scratch.data() + off_terms, terms.data(), terms.size() * sizeof(TermDev));
+    memcpy(scratch.data() + off_preds, fq.preds.data(),
+           fq.preds.size() * sizeof(PredDev));
+    memcpy(scratch.data() + off_ranges, ranges.data(), ranges_bytes);
+    ctx->d_scratch.ensure(scratch_bytes);
+    ctx->d_results.ensure(64);
+    HIP_CHECK(hipMemcpyAsync(ctx->d_scratch.p, scratch.data(), scratch_bytes,
+                             hipMemcpyHostToDevice, ctx->stream));
+    HIP_CHECK(hipMemsetAsync(ctx->d_results.p, 0, 64, ctx->stream));
+
+    uint32_t n_should = 0;
+    for (auto& t : fq.terms) n_should += t.role == ROLE_SHOULD;
+    QueryDev q{};
+    q.split = ds.d_image;
+    q.scratch = ctx->d_scratch.p;
+    q.results = ctx->d_results.p;
+    q.num_docs = sv.num_docs;
+    q.n_tiles = n_tiles;
+    q.n_terms = uint32_t(terms.size());
+    q.n_must = fq.n_must_groups;
+    q.n_must_not = n_must_not;
+    q.n_preds = uint32_t(fq.preds.size());
+    q.msm = fq.msm;
+    q.match_all = fq.match_all ? 1 : 0;
+    q.collect_hits = 1;
+    q.terms_off = off_terms;
+    q.preds_off = off_preds;
+    q.ktabs_off = off_ktabs;
+    q.tile_counts_off = 0;  // unused (do_count = 0)
+    q.cand_count_off = 0;   // phase-B reservation counter only
+    q.cand_off = 0;
+    q.cand_cap = 0;  // no candidate writes, bitmap only
+    q.bitmap_out = (uint64_t)bm;
+    bool ns = n_should > 0;
+    bool nb = fq.n_must_groups > 0 || n_must_not > 0;
+    launch_leaf_tile(ns, nb, false, true, dim3(n_tiles), ctx->stream, q, 0u,
+                     n_tiles, 0u);
+    HIP_CHECK(hipGetLastError());
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    ctx->hitsets.put(key, bm);
+    return bm;
+}
+
 static uint32_t f32_sortable_h(float f) {
     uint32_t b;
     memcpy(&b, &f, 4);
@@ -727,6 +913,17 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                                     const pb::SearchRequest& req, const Schema& schema) {
     SplitResult out;
     auto t0 = std::chrono::steady_clock::now();
+    static const bool timing = getenv("QW_TIMING") != nullptr;
+    auto tprev = t0;
+    auto mark = [&](const char* what) {
+        if (!timing) return;
+        auto now = std::chrono::steady_clock::now();
+        fprintf(stderr, "[qw_timing] %-18s %8.1f us\n", what,
+                std::chrono::duration_cast<std::chrono::nanoseconds>(now - tprev)
+                        .count() /
+                    1e3);
+        tprev = now;
+    };
     const SplitView& sv = ds.view;
 
     PlanNode plan = parse_query_ast(req.query_ast, schema);
@@ -775,6 +972,15 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                 specs.size() > 1;
 
     FlatQuery fq = flatten(sv, plan, scoring);
+    // CACHE nodes in filter position -> device HitSet bitmaps (PRED_BITSET)
+    for (const PlanNode* cn : fq.cache_nodes) {
+        PredDev p{};
+        p.type = PRED_BITSET;
+        p.abs_bitmap =
+            (uint64_t)resolve_hitset(ctx, const_cast<DeviceSplit&>(ds),
+                                     cn->cache_inner.front(), schema);
+        fq.preds.push_back(p);
+    }
     uint64_t leaf_max_hits = req.max_hits + req.start_offset;
 
     if (fq.match_none) {
@@ -908,6 +1114,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
     memcpy(scratch.data() + off_ranges, ranges.data(), ranges_bytes);
 
     uint64_t matched = 0;
+    mark("plan+assembly");
     std::vector<uint64_t> top_keys;  // survivors, sorted best-first
     std::function<void(uint64_t)> rerun_select;  // search_after retry hook
     uint64_t sel_band_n = 0, sel_kwant = 0;
@@ -1030,6 +1237,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             HIP_CHECK(hipMemcpyAsync(&cand_n, ctx->d_results.p + r_cand_count, 4,
                                      hipMemcpyDeviceToHost, ctx->stream));
         HIP_CHECK(hipStreamSynchronize(ctx->stream));
+        mark("main+pass0+sync");
         float ms = 0;
         HIP_CHECK(hipEventElapsedTime(&ms, ctx->ev_start, ctx->ev_stop));
         record_kernel_time(ctx, kname, ms);
@@ -1185,6 +1393,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             sel_kwant = std::min<uint64_t>(
                 leaf_max_hits + (after ? 1024 : 0), cand_n);
             rerun_select(sel_kwant);
+            mark("topk_select");
         }
     } else {
         matched = sv.num_docs;  // pure match_all, no aggs
@@ -1482,6 +1691,7 @@ void qw_ctx_free(qw_ctx* ctx) {
         (void)hipSetDevice(ctx->device);
         for (auto& kv : ctx->splits)
             if (kv.second->d_image) (void)hipFree(kv.second->d_image);
+        ctx->hitsets.clear();
         if (ctx->ev_start) (void)hipEventDestroy(ctx->ev_start);
         if (ctx->ev_stop) (void)hipEventDestroy(ctx->ev_stop);
         if (ctx->stream) (void)hipStreamDestroy(ctx->stream);
@@ -1528,6 +1738,7 @@ int32_t qw_ctx_remove_split(qw_ctx* ctx, const char* split_id) {
     if (it->second->d_image) (void)hipFree(it->second->d_image);
     ctx->splits.erase(it);
     ctx->leaf_cache.remove_split(split_id);
+    ctx->hitsets.remove_split(split_id);
     return QW_OK;
 }
 

@@ -22,6 +22,7 @@
 #include <algorithm>
 #include <cctype>
 #include <cstdint>
+#include <cstdio>
 #include <memory>
 #include <stdexcept>
 #include <string>
@@ -107,7 +108,8 @@ struct Bound {
 
 struct PlanNode {
     enum Kind {
-        MATCH_ALL, MATCH_NONE, TERM, BOOL, RANGE, FIELD_PRESENCE, WILDCARD
+        MATCH_ALL, MATCH_NONE, TERM, BOOL, RANGE, FIELD_PRESENCE, WILDCARD,
+        CACHE
     } kind = MATCH_ALL;
     // TERM / WILDCARD (value = glob pattern: '*' any run, '?' one char)
     std::string field;
@@ -123,12 +125,37 @@ struct PlanNode {
     // (tantivy TermSetQuery / AutomatonQuery const scorer) — the build
     // rejects these under _score sorting instead of mis-scoring them
     bool const_score = false;
+    // CACHE: the wrapped subtree (cache_node.rs CacheNode.inner); evaluated
+    // via a device-resident HitSet bitmap when in filter position
+    std::vector<PlanNode> cache_inner;  // size <= 1
 };
+
+// stable fingerprint of a plan subtree — the (split, subquery) key of the
+// device HitSet cache (the analog of CacheNode's PredicateCache key,
+// cache_node.rs:202-486)
+inline void plan_fingerprint(const PlanNode& n, std::string& out) {
+    out += char('A' + int(n.kind));
+    out += n.field;
+    out += '\x1f';
+    out += n.value;
+    out += '\x1f';
+    char buf[96];
+    snprintf(buf, sizeof buf, "%d:%lld:%d:%lld:%lld:%.9g:%d|",
+             int(n.lo.kind), (long long)n.lo.ival, int(n.hi.kind),
+             (long long)n.hi.ival, (long long)n.minimum_should_match,
+             double(n.boost), int(n.ci));
+    out += buf;
+    for (auto* v : {&n.must, &n.must_not, &n.should, &n.filter, &n.cache_inner}) {
+        out += '[';
+        for (auto& c : *v) plan_fingerprint(c, out);
+        out += ']';
+    }
+}
 
 // does any node require const-score semantics (term_set / wildcard)?
 inline bool plan_has_const_score(const PlanNode& n) {
     if (n.const_score) return true;
-    for (auto* v : {&n.must, &n.must_not, &n.should, &n.filter})
+    for (auto* v : {&n.must, &n.must_not, &n.should, &n.filter, &n.cache_inner})
         for (auto& c : *v)
             if (plan_has_const_score(c)) return true;
     return false;
@@ -351,9 +378,11 @@ inline PlanNode build_plan(const mj::Value* ast, const Schema& schema) {
         }
         if (n.should.empty()) n.kind = PlanNode::MATCH_NONE;
     } else if (ty == "cache") {
-        // CacheNode wraps without changing semantics (cache_node.rs:33-37);
-        // the result-level memoization lives in LeafSearchCache host-side
-        return build_plan(ast->at("inner"), schema);
+        // CacheNode (cache_node.rs:33-37): semantics-transparent wrapper;
+        // the product memoizes the inner subtree's HitSet as a device
+        // bitmap when the node sits in filter position
+        n.kind = PlanNode::CACHE;
+        n.cache_inner.push_back(build_plan(ast->at("inner"), schema));
     } else if (ty == "wildcard") {
         n.kind = PlanNode::WILDCARD;
         n.const_score = true;

@@ -812,3 +812,46 @@ def test_search_after_value_only_cursor(searchers):
     exp = cpu.leaf_search(req2)
     assert [(hid(h), sv_of(h)) for h in got.get("partial_hits", [])] == \
            [(hid(h), sv_of(h)) for h in exp.get("partial_hits", [])]
+
+
+# ------------------------------------------------- predicate cache (§8f.2)
+# cache nodes in filter position become device-resident HitSet bitmaps
+def test_cache_node_filter_bitmap_parity(searchers):
+    gpu, cpu = searchers
+    q = {"type": "bool",
+         "must": [{"type": "term", "field": "severity_text", "value": "INFO"}],
+         "filter": [{"type": "cache", "inner":
+                     {"type": "range", "field": "tenant_id",
+                      "lower_bound": {"included": 50},
+                      "upper_bound": {"excluded": 400}}}]}
+    got, exp = run_both(searchers, q, max_hits=20)
+    assert_hits_equal(got, exp, scored=False)
+    # second call reuses the bitmap: only the main kernel launches
+    gpu.kernel_stats_reset()
+    req = make_leaf_request(q, SCHEMA, [(SID, NDOCS)], max_hits=20)
+    gpu.leaf_search(req)
+    ms, n = gpu.kernel_stats("union_bm25")
+    assert n == 1  # one main launch; no bitmap rebuild
+
+
+def test_cache_node_root_and_nested(searchers):
+    q = {"type": "cache", "inner":
+         {"type": "bool",
+          "must": [{"type": "term", "field": "severity_text",
+                    "value": "WARN"}],
+          "filter": [{"type": "cache", "inner":
+                      {"type": "range", "field": "tenant_id",
+                       "lower_bound": {"included": 0},
+                       "upper_bound": {"excluded": 100}}}]}}
+    got, exp = run_both(searchers, q, max_hits=15)
+    assert_hits_equal(got, exp, scored=False)
+
+
+def test_cache_node_scored_transparent(searchers):
+    # scoring context: cache wrappers are transparent, BM25 unchanged
+    q = {"type": "cache", "inner": {"type": "bool", "should": [
+        {"type": "term", "field": "body", "value": "w%05d" % i}
+        for i in range(3)]}}
+    got, exp = run_both(searchers, q, max_hits=10, sort_fields=[
+        {"field_name": "_score", "sort_order": 1}])
+    assert_hits_equal(got, exp, scored=True)
