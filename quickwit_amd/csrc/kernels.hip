@@ -708,12 +708,22 @@ extern "C" __global__ void k_cand_hist(const uint64_t* cand, const uint32_t* n_p
     uint32_t n = *n_ptr;  // device-side count: pass 0 is enqueued without a
                           // host round trip for the candidate count
     uint32_t shift = 64 - prefix_bits - 12;
+    // per-thread run cache: score keys concentrate into few bins, so
+    // consecutive samples often repeat a bin — batch the LDS atomics
+    uint32_t last_bin = 0xFFFFFFFFu, acc = 0;
     for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
          i += gridDim.x * blockDim.x) {
         uint64_t k = cand[i];
         if (prefix_bits && (k >> (64 - prefix_bits)) != prefix) continue;
-        atomicAdd(&lh[(k >> shift) & (TOPK_BINS - 1)], 1u);
+        uint32_t b = uint32_t(k >> shift) & (TOPK_BINS - 1);
+        if (b == last_bin) ++acc;
+        else {
+            if (acc) atomicAdd(&lh[last_bin], acc);
+            last_bin = b;
+            acc = 1;
+        }
     }
+    if (acc) atomicAdd(&lh[last_bin], acc);
     __syncthreads();
     for (uint32_t i = threadIdx.x; i < TOPK_BINS; i += blockDim.x)
         if (lh[i]) atomicAdd(&hist[i], lh[i]);
@@ -728,12 +738,20 @@ extern "C" __global__ void k_cand_hist_w(const uint64_t* cand, const uint32_t* n
     __syncthreads();
     uint32_t n = *n_ptr;
     uint32_t shift = 64 - prefix_bits - 12;
+    uint32_t last_bin = 0xFFFFFFFFu, acc = 0;
     for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
          i += gridDim.x * blockDim.x) {
         uint64_t k = cand[2 * i];
         if (prefix_bits && (k >> (64 - prefix_bits)) != prefix) continue;
-        atomicAdd(&lh[(k >> shift) & (TOPK_BINS - 1)], 1u);
+        uint32_t b = uint32_t(k >> shift) & (TOPK_BINS - 1);
+        if (b == last_bin) ++acc;
+        else {
+            if (acc) atomicAdd(&lh[last_bin], acc);
+            last_bin = b;
+            acc = 1;
+        }
     }
+    if (acc) atomicAdd(&lh[last_bin], acc);
     __syncthreads();
     for (uint32_t i = threadIdx.x; i < TOPK_BINS; i += blockDim.x)
         if (lh[i]) atomicAdd(&hist[i], lh[i]);

@@ -1215,7 +1215,10 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             uint32_t* d_hist0 = (uint32_t*)(ctx->d_results.p + r_hist);
             uint32_t* d_n0 = (uint32_t*)(ctx->d_results.p + r_cand_count);
             HIP_CHECK(hipMemsetAsync(d_hist0, 0, TOPK_BINS * 4, ctx->stream));
-            uint32_t hgrid = std::min<uint32_t>(512, (n_tiles * 16 + 4095) / 4096);
+            // candidate count is device-side only at this point: size the
+            // grid for the worst case (every doc a candidate)
+            uint32_t hgrid =
+                std::min<uint32_t>(512, (sv.num_docs + 4095) / 4096);
             if (wide)
                 hipLaunchKernelGGL(k_cand_hist_w, dim3(hgrid), dim3(256), 0,
                                    ctx->stream, d_cand0, d_n0, 0ull, 0u, d_hist0);
