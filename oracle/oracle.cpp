@@ -501,7 +501,9 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
                     for (uint32_t doc : m.docs) visit(doc);
                 for (uint32_t o = 0; o < f->cardinality; ++o)
                     if (counts[o]) a.term_counts.emplace_back(f->dict_entry(o), counts[o]);
-                // dict order is lexicographic => term_counts sorted by key
+                // dict order is lexicographic => term_counts sorted by key;
+                // per-split split_size truncation + error bound (qagg_format.h)
+                truncate_terms_split(a, effective_split_size(d.size, d.split_size));
             }
             out.aggs.push_back(std::move(a));
             continue;

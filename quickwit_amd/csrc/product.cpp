@@ -983,7 +983,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
     }
     uint64_t leaf_max_hits = req.max_hits + req.start_offset;
 
-    if (fq.match_none) {
+    if (fq.match_none || sv.num_docs == 0) {
         if (req.aggregation_request) {
             // empty-but-shaped blob, like the oracle over an empty match set
             for (const AggDef& d : parse_agg_request(*req.aggregation_request)) {
@@ -1520,6 +1520,8 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     for (uint32_t o = 0; o < a.n_buckets; ++o)
                         if (counts[o])
                             r.term_counts.emplace_back(f->dict_entry(o), counts[o]);
+                    truncate_terms_split(
+                        r, effective_split_size(d.size, d.split_size));
                 }
             } else {
                 r.kind = d.kind == AggDef::DATE_HISTOGRAM ? 1 : 2;

@@ -58,7 +58,35 @@ struct AggResult {
     std::vector<AggBucket> buckets;                            // histos (sorted by key)
     std::vector<std::pair<std::string, uint64_t>> term_counts; // terms (sorted by key)
     uint64_t terms_matched_docs = 0;
+    // sum over truncated splits of the last-included term count — the ES
+    // doc_count_error_upper_bound semantics the golden scenario pins
+    // (aggregations/0001 split_size=1 case)
+    uint64_t terms_error_bound = 0;
 };
+
+// per-split terms truncation (tantivy terms agg split_size): keep the top
+// `split_size` entries by (count desc, key asc); when anything is dropped,
+// the last included count joins the error bound. Restores key order after.
+inline void truncate_terms_split(AggResult& r, int64_t split_size) {
+    if (split_size < 1) split_size = 1;
+    // strictly fewer terms than the cap: the split reported everything and
+    // contributes no error. A FULL list (== cap, even untruncated) could be
+    // hiding terms up to the last included count — the ES semantics the
+    // golden scenario pins (split_size=5 over a 5-term split -> error 1).
+    if (r.term_counts.size() < uint64_t(split_size)) return;
+    std::stable_sort(r.term_counts.begin(), r.term_counts.end(),
+                     [](const auto& x, const auto& y) {
+                         if (x.second != y.second) return x.second > y.second;
+                         return x.first < y.first;
+                     });
+    r.terms_error_bound += r.term_counts[size_t(split_size) - 1].second;
+    r.term_counts.resize(size_t(split_size));
+    std::sort(r.term_counts.begin(), r.term_counts.end());
+}
+
+inline int64_t effective_split_size(uint32_t size, int64_t split_size) {
+    return split_size >= 0 ? split_size : int64_t(size) * 3 / 2 + 10;
+}
 
 struct IntermediateAggResults {
     std::vector<AggResult> aggs;
@@ -67,7 +95,7 @@ struct IntermediateAggResults {
         std::string o;
         auto put = [&](const void* p, size_t n) { o.append((const char*)p, n); };
         uint32_t magic = 0x31474151;
-        uint16_t ver = 1, n = uint16_t(aggs.size());
+        uint16_t ver = 2, n = uint16_t(aggs.size());
         put(&magic, 4);
         put(&ver, 2);
         put(&n, 2);
@@ -85,6 +113,7 @@ struct IntermediateAggResults {
             }
             if (a.kind == 3) {
                 put(&a.terms_matched_docs, 8);
+                put(&a.terms_error_bound, 8);
                 uint32_t ne = uint32_t(a.term_counts.size());
                 put(&ne, 4);
                 for (auto& kv : a.term_counts) {
@@ -129,7 +158,7 @@ struct IntermediateAggResults {
         get(&magic, 4);
         get(&ver, 2);
         get(&n, 2);
-        if (magic != 0x31474151 || ver != 1) throw std::runtime_error("QAGG1: bad header");
+        if (magic != 0x31474151 || ver != 2) throw std::runtime_error("QAGG1: bad header");
         for (int i = 0; i < n; ++i) {
             AggResult a;
             uint16_t nl;
@@ -149,6 +178,7 @@ struct IntermediateAggResults {
             }
             if (a.kind == 3) {
                 get(&a.terms_matched_docs, 8);
+                get(&a.terms_error_bound, 8);
                 uint32_t ne;
                 get(&ne, 4);
                 a.term_counts.reserve(ne);
@@ -220,6 +250,7 @@ struct IntermediateAggResults {
                 }
                 a.term_counts = std::move(merged);
                 a.terms_matched_docs += b.terms_matched_docs;
+                a.terms_error_bound += b.terms_error_bound;
             } else {
                 std::vector<AggBucket> merged;
                 merged.reserve(a.buckets.size() + b.buckets.size());
@@ -336,7 +367,14 @@ inline std::string finalize_aggs_json(const IntermediateAggResults& ir,
                 mj::escape_to(o, ordered[b].first);
                 o += "}";
             }
-            o += "],\"doc_count_error_upper_bound\":0,\"sum_other_doc_count\":";
+            o += "],\"doc_count_error_upper_bound\":";
+            {
+                char ebuf[24];
+                snprintf(ebuf, sizeof ebuf, "%llu",
+                         (unsigned long long)a.terms_error_bound);
+                o += ebuf;
+            }
+            o += ",\"sum_other_doc_count\":";
             char buf[24];
             snprintf(buf, sizeof buf, "%llu",
                      (unsigned long long)(a.terms_matched_docs - shown_docs));

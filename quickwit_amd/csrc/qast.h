@@ -524,6 +524,8 @@ struct AggDef {
     bool has_bounds = false;
     double bmin = 0, bmax = 0;
     uint32_t size = 10;          // terms
+    int64_t split_size = -1;     // terms: per-split truncation; -1 = default
+                                 // (size*3/2+10, the ES shard_size default)
     int64_t min_doc_count = -1;  // -1 = default (0 for histos, 1 for terms)
     std::vector<MetricAgg> sub;
 };
@@ -579,6 +581,14 @@ inline std::vector<AggDef> parse_agg_request(const std::string& json) {
             a.kind = AggDef::TERMS;
             a.field = spec->at("field")->s;
             if (const mj::Value* sz = spec->get("size")) a.size = uint32_t(sz->as_i64());
+            // one segment per split in QWA1 => split/segment/shard size
+            // aliases all bound the same per-split truncation
+            if (const mj::Value* ss = spec->get("split_size"))
+                a.split_size = ss->as_i64();
+            else if (const mj::Value* sg = spec->get("segment_size"))
+                a.split_size = sg->as_i64();
+            else if (const mj::Value* sh = spec->get("shard_size"))
+                a.split_size = sh->as_i64();
             if (const mj::Value* mdc = spec->get("min_doc_count"))
                 a.min_doc_count = mdc->as_i64();
         } else {

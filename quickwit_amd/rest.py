@@ -1,0 +1,262 @@
+"""Minimal HTTP shim over the C-ABI — enough of Quickwit's REST surface to
+replay the reference's rest-api-tests golden scenarios unmodified
+(quickwit/rest-api-tests/run_tests.py semantics; SURVEY.md §4 calls these
+"the main cross-implementation parity tool"). Serving at scale is OUT OF
+SCOPE; this exists so the engine-agnostic scenario suites can
+drive the engine end-to-end over HTTP.
+
+Endpoints (subset the suites touch):
+  POST   /api/v1/indexes            create index (quickwit index config)
+  DELETE /api/v1/indexes/{id}
+  POST   /api/v1/{id}/ingest        ndjson body; one split per call
+  GET/POST /api/v1/_elastic/{id}/_search   ES body {query, aggs, size, sort}
+
+Dynamic mapping (doc_mapping.mode == "dynamic"): unmapped fields are
+inferred from the ingested corpus — str -> text(default) + fast str column,
+non-negative int -> u64 fast, int -> i64 fast; floats/arrays/objects are
+skipped (multi-valued and f64 fast columns are not in round 1). The engine
+behind the shim is either the product (GpuSearcher) or the oracle
+(OracleSearcher) — the replay asserts identical golden JSON on both.
+"""
+import json
+import time
+
+from . import splitgen
+from .api import make_leaf_request
+
+
+# -------------------------------------------------- ES query DSL -> QueryAst
+# (quickwit-query/src/elastic_query_dsl/ subset the scenario suites use)
+def es_query_to_ast(q):
+    if q is None:
+        return {"type": "match_all"}
+    if "match_all" in q:
+        return {"type": "match_all"}
+    if "match_none" in q:
+        return {"type": "match_none"}
+    if "term" in q:
+        [(field, body)] = q["term"].items()
+        value = body["value"] if isinstance(body, dict) else body
+        return {"type": "term", "field": field, "value": str(value)}
+    if "terms" in q:
+        [(field, values)] = [(k, v) for k, v in q["terms"].items()
+                             if k != "boost"]
+        return {"type": "term_set",
+                "terms_per_field": {field: sorted(str(v) for v in values)}}
+    if "exists" in q:
+        return {"type": "field_presence", "field": q["exists"]["field"]}
+    if "range" in q:
+        [(field, body)] = q["range"].items()
+        node = {"type": "range", "field": field}
+        if "gte" in body:
+            node["lower_bound"] = {"included": body["gte"]}
+        elif "gt" in body:
+            node["lower_bound"] = {"excluded": body["gt"]}
+        if "lte" in body:
+            node["upper_bound"] = {"included": body["lte"]}
+        elif "lt" in body:
+            node["upper_bound"] = {"excluded": body["lt"]}
+        return node
+    if "match" in q:
+        [(field, body)] = q["match"].items()
+        text = body["query"] if isinstance(body, dict) else body
+        op = (body.get("operator", "or") if isinstance(body, dict) else "or")
+        return {"type": "full_text", "field": field, "text": str(text),
+                "params": {"mode": {"type": "bool", "operator": op.lower()}}}
+    if "bool" in q:
+        out = {"type": "bool"}
+        for clause in ("must", "must_not", "should", "filter"):
+            if clause in q["bool"]:
+                items = q["bool"][clause]
+                if isinstance(items, dict):
+                    items = [items]
+                out[clause] = [es_query_to_ast(i) for i in items]
+        if "minimum_should_match" in q["bool"]:
+            out["minimum_should_match"] = int(q["bool"]["minimum_should_match"])
+        return out
+    if "query_string" in q:
+        return {"type": "user_input", "user_text": q["query_string"]["query"],
+                "default_fields": q["query_string"].get("fields")}
+    raise ValueError(f"unsupported es query: {list(q)}")
+
+
+def _infer_dynamic_fields(explicit_names, doc_batches):
+    """Dynamic-mode field inference over the whole corpus (stable across
+    splits). Returns a list of schema field dicts."""
+    seen = {}
+    for docs in doc_batches:
+        for d in docs:
+            for k, v in d.items():
+                if k in explicit_names or v is None:
+                    continue
+                kinds = seen.setdefault(k, set())
+                if isinstance(v, bool):
+                    kinds.add("skip")
+                elif isinstance(v, int):
+                    kinds.add("int" if v >= 0 else "negint")
+                elif isinstance(v, float):
+                    kinds.add("skip")  # f64 fast columns: later round
+                elif isinstance(v, str):
+                    kinds.add("str")
+                else:
+                    kinds.add("skip")  # arrays/objects: multi-valued, later
+    fields = []
+    for name in sorted(seen):
+        kinds = seen[name]
+        if "skip" in kinds:
+            continue
+        if kinds <= {"int"}:
+            fields.append({"name": name, "type": "u64", "fast": True})
+        elif kinds <= {"int", "negint"}:
+            fields.append({"name": name, "type": "i64", "fast": True})
+        elif kinds == {"str"}:
+            fields.append({"name": name, "type": "str", "fast": True})
+    return fields
+
+
+class Index:
+    def __init__(self, index_id, config, searcher_factory):
+        self.index_id = index_id
+        self.config = config
+        self.searcher_factory = searcher_factory
+        self.batches = []          # raw docs, one list per ingest call
+        self.searcher = None
+        self.schema = None
+        self.splits = []           # (split_id, num_docs)
+
+    def _explicit_fields(self):
+        fields = []
+        ts_field = None
+        dm = self.config.get("doc_mapping", {})
+        for fm in dm.get("field_mappings", []):
+            t = fm["type"]
+            name = fm["name"]
+            if t == "datetime":
+                fields.append({"name": name, "type": "datetime",
+                               "fast": bool(fm.get("fast", False))})
+                if dm.get("timestamp_field") == name:
+                    ts_field = name
+            elif t in ("u64", "i64"):
+                fields.append({"name": name, "type": t,
+                               "fast": bool(fm.get("fast", False))})
+            elif t == "text":
+                f = {"name": name, "type": "text",
+                     "tokenizer": fm.get("tokenizer", "default")}
+                if fm.get("record") == "freq":
+                    f["record"] = "freq"
+                fields.append(f)
+        return fields, ts_field
+
+    def rebuild(self):
+        fields, ts_field = self._explicit_fields()
+        explicit = {f["name"] for f in fields}
+        if self.config.get("doc_mapping", {}).get("mode") == "dynamic":
+            fields = fields + _infer_dynamic_fields(explicit, self.batches)
+        self.schema = {"timestamp_field": ts_field, "fields": fields,
+                       "default_search_fields":
+                           self.config.get("search_settings", {})
+                               .get("default_search_fields", [])}
+        self.searcher = self.searcher_factory()
+        self.splits = []
+        batches = self.batches or [[]]  # empty index: one 0-doc split so
+        for i, docs in enumerate(batches):  # aggs return shaped empties
+            sid = f"{self.index_id}-{i:04d}"
+            w = splitgen.SplitWriter(self.schema, sid)
+            w.add_documents(docs)
+            self.searcher.add_split(sid, w.finalize())
+            self.splits.append((sid, len(docs)))
+
+    def source_doc(self, split_id, doc_id):
+        i = int(split_id.rsplit("-", 1)[1])
+        return self.batches[i][doc_id] if self.batches else None
+
+
+def create_app(searcher_factory):
+    from fastapi import FastAPI, Request, Response
+
+    app = FastAPI()
+    indexes = {}
+
+    @app.post("/api/v1/indexes")
+    @app.post("/api/v1/indexes/")
+    async def create_index(request: Request):
+        cfg = await request.json()
+        iid = cfg["index_id"]
+        idx = Index(iid, cfg, searcher_factory)
+        idx.rebuild()
+        indexes[iid] = idx
+        return {"index_config": {"index_id": iid}}
+
+    @app.delete("/api/v1/indexes/{iid}")
+    async def delete_index(iid: str, response: Response):
+        if iid not in indexes:
+            response.status_code = 404
+            return {"message": f"index {iid} not found"}
+        del indexes[iid]
+        return {"removed": iid}
+
+    @app.post("/api/v1/{iid}/ingest")
+    async def ingest(iid: str, request: Request, response: Response):
+        if iid not in indexes:
+            response.status_code = 404
+            return {"message": "index not found"}
+        body = (await request.body()).decode()
+        docs = [json.loads(line) for line in body.splitlines() if line.strip()]
+        idx = indexes[iid]
+        idx.batches.append(docs)
+        idx.rebuild()  # commit=force semantics: searchable immediately
+        return {"num_docs_for_processing": len(docs)}
+
+    @app.api_route("/api/v1/_elastic/{iid}/_search", methods=["GET", "POST"])
+    async def es_search(iid: str, request: Request, response: Response):
+        if iid not in indexes:
+            response.status_code = 404
+            return {"message": "index not found"}
+        idx = indexes[iid]
+        body = {}
+        raw = await request.body()
+        if raw:
+            body = json.loads(raw)
+        t0 = time.perf_counter()
+        ast = es_query_to_ast(body.get("query"))
+        aggs = body.get("aggs") or body.get("aggregations")
+        size = int(body.get("size", 10))
+        sort_fields = []
+        for s in body.get("sort", []):
+            if isinstance(s, str):
+                field, order = s, "asc"
+            else:
+                [(field, so)] = s.items()
+                order = so.get("order", "asc") if isinstance(so, dict) else so
+            if field == "_score" and order == "desc":
+                sort_fields.append({"field_name": "_score", "sort_order": 1})
+            else:
+                sort_fields.append({"field_name": field,
+                                    "sort_order": 1 if order == "desc" else 0})
+        req = make_leaf_request(ast, idx.schema, idx.splits, max_hits=size,
+                                sort_fields=sort_fields or None,
+                                aggregation=aggs)
+        resp = idx.searcher.leaf_search(req)
+        took_ms = int((time.perf_counter() - t0) * 1e3)
+        hits = []
+        for h in resp.get("partial_hits", []):
+            doc_id = h.get("doc_id", 0)
+            sid = h.get("split_id", "")
+            src = idx.source_doc(sid, doc_id)
+            hit = {"_index": iid, "_id": f"{sid}:{doc_id}", "_source": src}
+            sv = h.get("sort_value", {})
+            if "f64" in sv:
+                hit["_score"] = sv["f64"]
+            hits.append(hit)
+        out = {"took": took_ms, "timed_out": False,
+               "hits": {"total": {"value": resp.get("num_hits", 0),
+                                  "relation": "eq"},
+                        "max_score": None, "hits": hits}}
+        if aggs is not None and "intermediate_aggregation_result" in resp:
+            out["aggregations"] = idx.searcher.finalize_agg_json(
+                resp["intermediate_aggregation_result"], aggs)
+        elif aggs is not None:
+            out["aggregations"] = {}
+        return out
+
+    return app

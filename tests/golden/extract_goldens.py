@@ -117,6 +117,43 @@ def agg_golden():
     }
 
 
+def rest_scenarios():
+    """Full step lists of selected rest-api-tests suites, replayed by
+    tests/test_rest_scenarios.py against the HTTP shim (quickwit_amd/rest.py)
+    with run_tests.py's checking semantics (ported in tests/rest_replay.py)."""
+    import yaml
+
+    base = os.path.join(REF, "rest-api-tests/scenarii")
+    suites = {
+        "aggregations": ["_setup.quickwit.yaml", "0001-aggregations.yaml",
+                         "_teardown.quickwit.yaml"],
+    }
+    keep = {"method", "endpoint", "params", "json", "ndjson", "expected",
+            "status_code"}
+    out = {}
+    for suite, files in suites.items():
+        steps = []
+        for fn in files:
+            for step in yaml.safe_load_all(open(os.path.join(base, suite, fn))):
+                if not isinstance(step, dict) or "method" not in step:
+                    continue
+                engines = step.get("engines")
+                if engines and "quickwit" not in engines:
+                    continue
+                s = {k: v for k, v in step.items() if k in keep}
+                if isinstance(s["method"], list):
+                    s["method"] = s["method"][0]
+                if "status_code" in step and step["status_code"] is None:
+                    s["status_code"] = None
+                steps.append(s)
+        out[suite] = steps
+    return {
+        "source": "quickwit/rest-api-tests/scenarii/ (run_tests.py driver "
+                  "semantics ported to tests/rest_replay.py)",
+        "suites": out,
+    }
+
+
 def main():
     if not os.path.isdir(REF):
         sys.exit("reference checkout not present; fixtures must already be committed")
@@ -126,7 +163,10 @@ def main():
     with open(os.path.join(OUT, "aggregations.json"), "w") as f:
         json.dump(agg_golden(), f, indent=1, sort_keys=True)
         f.write("\n")
-    print("wrote bm25_sort.json, aggregations.json")
+    with open(os.path.join(OUT, "rest_scenarios.json"), "w") as f:
+        json.dump(rest_scenarios(), f, indent=1, sort_keys=True)
+        f.write("\n")
+    print("wrote bm25_sort.json, aggregations.json, rest_scenarios.json")
 
 
 if __name__ == "__main__":
