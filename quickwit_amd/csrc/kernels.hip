@@ -216,15 +216,34 @@ __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t t
             s2 = t.weight * (float(tf2) / (float(tf2) + K2));
             s3 = t.weight * (float(tf3) / (float(tf3) + K3));
         }
-        #pragma unroll
-        for (int el = 0; el < 4; ++el) {
-            uint32_t doc = el == 0 ? doc0 : el == 1 ? doc1 : el == 2 ? doc2 : doc3;
-            float s = el == 0 ? s0 : el == 1 ? s1 : el == 2 ? s2 : s3;
-            if (j0 + el < e.count && doc >= tile_lo && doc < tile_hi) {
-                uint32_t li = doc - tile_lo;
-                if (SCORE) atomicAdd(&score[li], s);
-                else atomicOr(&bitset[li >> 5], 1u << (li & 31));
+        if (SCORE) {
+            #pragma unroll
+            for (int el = 0; el < 4; ++el) {
+                uint32_t doc = el == 0 ? doc0 : el == 1 ? doc1 : el == 2 ? doc2 : doc3;
+                float s = el == 0 ? s0 : el == 1 ? s1 : el == 2 ? s2 : s3;
+                if (j0 + el < e.count && doc >= tile_lo && doc < tile_hi)
+                    atomicAdd(&score[doc - tile_lo], s);
             }
+        } else {
+            // bitset role: a lane's 4 consecutive docs often share a 32-doc
+            // word on dense terms (gaps of 2-3) — merge the bits locally so
+            // the same-address LDS atomicOr serialization drops ~4x
+            uint32_t pw = 0xFFFFFFFFu, pm = 0;
+            #pragma unroll
+            for (int el = 0; el < 4; ++el) {
+                uint32_t doc = el == 0 ? doc0 : el == 1 ? doc1 : el == 2 ? doc2 : doc3;
+                if (j0 + el < e.count && doc >= tile_lo && doc < tile_hi) {
+                    uint32_t li = doc - tile_lo;
+                    uint32_t wd = li >> 5;
+                    if (wd == pw) pm |= 1u << (li & 31);
+                    else {
+                        if (pm) atomicOr(&bitset[pw], pm);
+                        pw = wd;
+                        pm = 1u << (li & 31);
+                    }
+                }
+            }
+            if (pm) atomicOr(&bitset[pw], pm);
         }
     }
 }
