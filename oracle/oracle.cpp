@@ -311,6 +311,16 @@ struct SplitSearcher {
             }
             return true;
         }
+        if (f.type == FastFieldView::F64) {
+            double v = f.f64(d);
+            if (lo.kind != Bound::UNBOUNDED &&
+                (lo.kind == Bound::INCLUDED ? v < lo.fval : v <= lo.fval))
+                return false;
+            if (hi.kind != Bound::UNBOUNDED &&
+                (hi.kind == Bound::INCLUDED ? v > hi.fval : v >= hi.fval))
+                return false;
+            return true;
+        }
         int64_t v = f.i64(d);
         if (lo.kind != Bound::UNBOUNDED &&
             (lo.kind == Bound::INCLUDED ? v < lo.ival : v <= lo.ival))
@@ -406,7 +416,12 @@ struct SplitSearcher {
             } else {
                 base = union_n(shoulds, msm, sv.num_docs);
             }
-            if (shoulds.empty()) base = Match{};
+            if (shoulds.empty()) {
+                base = Match{};
+                // must_not with no positive clause: implicit match_all
+                // (tantivy_query_ast.rs:310-322 pushes match_all)
+                if (!n.must_not.empty()) base.all = true;
+            }
         }
         if (!n.must_not.empty()) {
             std::vector<Match> nots;
@@ -456,6 +471,9 @@ static pb::SortByValue sort_value_of(const SortSpec& s, const SplitView& sv, uin
             } else if (f->type == FastFieldView::STR) {
                 v.kind = pb::SortByValue::U64;
                 v.u64 = f->ord(doc);
+            } else if (f->type == FastFieldView::F64) {
+                v.kind = pb::SortByValue::F64;
+                v.f64 = f->f64(doc);
             } else {
                 v.kind = pb::SortByValue::I64;
                 v.i64 = f->i64(doc);
@@ -483,6 +501,7 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
             switch (sf->type) {
                 case FastFieldView::U64: *v = double(sf->u64(doc)); return true;
                 case FastFieldView::STR: return false;
+                case FastFieldView::F64: *v = sf->f64(doc); return true;
                 default: *v = double(sf->i64(doc)); return true;
             }
         };
@@ -515,6 +534,7 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
             double v;
             if (f->type == FastFieldView::U64) v = double(f->u64(doc));
             else if (f->type == FastFieldView::STR) return;
+            else if (f->type == FastFieldView::F64) v = f->f64(doc);
             else v = double(f->i64(doc));
             int64_t bi = int64_t(std::floor((v - d.offset) / d.interval));
             AggBucket& b = buckets[bi];

@@ -34,6 +34,7 @@ enum PredType : uint32_t {
     PRED_RANGE_I64 = 1,  // i64 / datetime(ms)
     PRED_PRESENCE = 2,
     PRED_BITSET = 3,     // device-resident HitSet (predicate cache, §8f.2)
+    PRED_RANGE_F64 = 4,  // f64 column; lo/hi carry bit-cast doubles
 };
 enum PredFlags : uint32_t {
     PRED_NEGATED = 1,       // must_not predicate
@@ -74,7 +75,7 @@ struct AggDev {
     uint64_t values_off;    // column byte offset
     uint64_t nulls_off;
     uint32_t value_width;   // 8, or ord width for terms
-    uint32_t value_is_i64;  // 1 = signed column
+    uint32_t value_is_i64;  // 0 = u64, 1 = i64/datetime, 2 = f64
     uint64_t counts_out;    // byte offset into result scratch: u64[n_buckets]
     uint64_t matched_out;   // terms: u64 counter of docs-with-value
     // one optional stats sub-agg set per bucket: {u64 cnt, f64 sum, u64 min_s,
@@ -109,7 +110,8 @@ struct QueryDev {
     // serve fast-field sorts and two-field sorts — exact order is
     // re-established on host over the survivors.
     uint32_t wide_cand;     // 1 = 16B candidate records
-    uint32_t sort_src;      // 0 none/doc, 1 _score, 2 u64/ord column, 3 i64 column
+    uint32_t sort_src;      // 0 none/doc, 1 _score, 2 u64/ord column,
+                            // 3 i64 column, 4 f64 column
     uint32_t sort_width;    // column width (1/2/4/8), sort_src>=2
     uint64_t sort_values_off;
     uint64_t sort_nulls_off;  // 0 = non-nullable (missing -> None, sorts last)

@@ -260,6 +260,15 @@ __device__ __forceinline__ bool eval_pred(const QueryDev& q, const PredDev& p,
         ok = (bm[doc >> 5] >> (doc & 31)) & 1;
         return (p.flags & PRED_NEGATED) ? !ok : ok;
     }
+    if (ok && p.type == PRED_RANGE_F64) {
+        double v = ((const double*)(q.split + p.values_off))[doc];
+        double lo = __longlong_as_double(p.lo), hi = __longlong_as_double(p.hi);
+        if (p.flags & PRED_LO_INCLUDED) ok &= v >= lo;
+        if (p.flags & PRED_LO_EXCLUDED) ok &= v > lo;
+        if (p.flags & PRED_HI_INCLUDED) ok &= v <= hi;
+        if (p.flags & PRED_HI_EXCLUDED) ok &= v < hi;
+        return (p.flags & PRED_NEGATED) ? !ok : ok;
+    }
     if (ok && p.type != PRED_PRESENCE) {
         if (p.type == PRED_RANGE_U64) {
             uint64_t v = ((const uint64_t*)(q.split + p.values_off))[doc];
@@ -287,8 +296,9 @@ __device__ __forceinline__ double agg_value(const QueryDev& q, uint64_t values_o
         case 2: return double(((const uint16_t*)col)[doc]);
         case 4: return double(((const uint32_t*)col)[doc]);
         default:
-            return is_i64 ? double(((const int64_t*)col)[doc])
-                          : double(((const uint64_t*)col)[doc]);
+            return is_i64 == 2 ? ((const double*)col)[doc]
+                   : is_i64   ? double(((const int64_t*)col)[doc])
+                              : double(((const uint64_t*)col)[doc]);
     }
 }
 
@@ -325,6 +335,7 @@ __device__ __forceinline__ uint64_t wide_sort_key(const QueryDev& q, uint32_t d,
             default: S = ((const uint64_t*)col)[d]; break;
         }
         if (q.sort_src == 3) S ^= (1ull << 63);  // i64_to_u64
+        else if (q.sort_src == 4) S = f64_sortable(__longlong_as_double(S));
     }
     return q.sort_asc ? ~S : S;
 }

@@ -318,7 +318,9 @@ static void add_range_pred(FlatQuery& fq, const SplitView& sv, const PlanNode& n
         fq.match_none = true;
         return;
     }
-    p.type = f->type == FastFieldView::U64 ? PRED_RANGE_U64 : PRED_RANGE_I64;
+    p.type = f->type == FastFieldView::U64   ? PRED_RANGE_U64
+             : f->type == FastFieldView::F64 ? PRED_RANGE_F64
+                                             : PRED_RANGE_I64;
     p.flags = negated ? PRED_NEGATED : 0;
     if (n.kind == PlanNode::FIELD_PRESENCE) {
         if (!f->nullable && !negated) return;  // non-nullable: always present
@@ -332,8 +334,13 @@ static void add_range_pred(FlatQuery& fq, const SplitView& sv, const PlanNode& n
         if (n.lo.kind == Bound::EXCLUDED) p.flags |= PRED_LO_EXCLUDED;
         if (n.hi.kind == Bound::INCLUDED) p.flags |= PRED_HI_INCLUDED;
         if (n.hi.kind == Bound::EXCLUDED) p.flags |= PRED_HI_EXCLUDED;
-        p.lo = n.lo.ival;
-        p.hi = n.hi.ival;
+        if (f->type == FastFieldView::F64) {
+            memcpy(&p.lo, &n.lo.fval, 8);  // bit-cast double bounds
+            memcpy(&p.hi, &n.hi.fval, 8);
+        } else {
+            p.lo = n.lo.ival;
+            p.hi = n.hi.ival;
+        }
         if (f->type == FastFieldView::STR)
             throw std::runtime_error("range over str fast field not supported");
     }
@@ -526,7 +533,14 @@ static void flatten_bool(FlatQuery& fq, const SplitView& sv, const PlanNode& n,
         return;
     }
     if (!has_req && n.should.empty()) {
-        fq.match_none = true;  // no positive clause matches nothing
+        if (n.must_not.empty()) {
+            fq.match_none = true;  // no clause at all matches nothing
+            return;
+        }
+        // must_not only: implicit match_all base
+        // (tantivy_query_ast.rs:310-322 pushes match_all)
+        fq.match_all = true;
+        fq.msm = 0;
         return;
     }
     fq.msm = uint32_t(msm);
@@ -633,11 +647,16 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
                 a.values_off = f->values.off;
                 a.nulls_off = f->nullable ? f->nulls.off : 0;
                 a.value_width = 8;
-                a.value_is_i64 = f->type != FastFieldView::U64;
+                a.value_is_i64 = f->type == FastFieldView::U64   ? 0
+                                 : f->type == FastFieldView::F64 ? 2
+                                                                 : 1;
                 double mn = double(f->min_value), mx = double(f->max_value);
                 if (f->type == FastFieldView::U64) {
                     mn = double(uint64_t(f->min_value));
                     mx = double(uint64_t(f->max_value));
+                } else if (f->type == FastFieldView::F64) {
+                    mn = f->fmin;
+                    mx = f->fmax;
                 }
                 int64_t b0 = int64_t(std::floor((mn - d.offset) / d.interval));
                 int64_t b1 = int64_t(std::floor((mx - d.offset) / d.interval));
@@ -653,8 +672,9 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
                     double iv = d.interval, of = d.offset;
                     bool integral = iv >= 1.0 && std::floor(iv) == iv &&
                                     std::floor(of) == of;
-                    bool u64_ok = f->type != FastFieldView::U64 ||
-                                  uint64_t(f->max_value) < (1ull << 63);
+                    bool u64_ok = (f->type != FastFieldView::U64 ||
+                                   uint64_t(f->max_value) < (1ull << 63)) &&
+                                  f->type != FastFieldView::F64;
                     double span = std::max(std::fabs(mn - of), std::fabs(mx - of));
                     if (integral && u64_ok && span < 9.0e15 && iv < 9.0e15 &&
                         std::fabs(of) < 9.0e15 && std::fabs(mn) < 9.0e15 &&
@@ -679,7 +699,9 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
                         a.sub_values_off[si] = sf->values.off;
                         a.sub_nulls_off[si] = sf->nullable ? sf->nulls.off : 0;
                         a.sub_width[si] = 8;
-                        a.sub_is_i64[si] = sf->type != FastFieldView::U64;
+                        a.sub_is_i64[si] = sf->type == FastFieldView::U64   ? 0
+                                           : sf->type == FastFieldView::F64 ? 2
+                                                                            : 1;
                     } else {
                         a.sub_values_off[si] = 0;  // missing column: no values
                         a.sub_width[si] = 0;
@@ -769,6 +791,9 @@ static pb::SortByValue sort_value_of(const SortSpec& s, uint32_t doc, float scor
             } else if (f->type == FastFieldView::STR) {
                 v.kind = pb::SortByValue::U64;
                 v.u64 = f->ord(doc);
+            } else if (f->type == FastFieldView::F64) {
+                v.kind = pb::SortByValue::F64;
+                v.f64 = f->f64(doc);
             } else {
                 v.kind = pb::SortByValue::I64;
                 v.i64 = f->i64(doc);
@@ -1162,7 +1187,9 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     q.sort_src = 2;
                     q.sort_width = uint32_t(f->ord_width);
                 } else {
-                    q.sort_src = f->type == FastFieldView::U64 ? 2 : 3;
+                    q.sort_src = f->type == FastFieldView::U64   ? 2
+                                 : f->type == FastFieldView::F64 ? 4
+                                                                 : 3;
                     q.sort_width = 8;
                 }
             } else {

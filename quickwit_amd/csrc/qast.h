@@ -211,12 +211,14 @@ inline Bound parse_bound(const mj::Value* v, const SchemaField& f) {
         b.ival = parse_datetime_ms(lit);
     } else if (lit->kind == mj::Value::INT) {
         b.ival = lit->i;
+        b.fval = double(lit->i);
     } else if (lit->kind == mj::Value::DBL) {
         b.from_f64 = true;
         b.fval = lit->d;
         b.ival = int64_t(lit->d);
     } else if (lit->kind == mj::Value::STR) {
         b.ival = strtoll(lit->s.c_str(), nullptr, 10);
+        b.fval = atof(lit->s.c_str());
     } else throw std::runtime_error("bad range literal");
     return b;
 }

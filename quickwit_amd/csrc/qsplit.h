@@ -68,11 +68,12 @@ struct TextFieldView {
 
 struct FastFieldView {
     std::string name;
-    enum Type { U64, I64, DATETIME, STR } type = U64;
+    enum Type { U64, I64, DATETIME, STR, F64 } type = U64;
     bool nullable = false;
     uint32_t cardinality = 0;  // str
     int ord_width = 0;         // str: 1/2/4
     int64_t min_value = 0, max_value = 0;
+    double fmin = 0, fmax = 0;  // F64
     Section values, nulls, dict_offsets, dict_bytes;
     const void* h_values = nullptr;
     const uint64_t* h_nulls = nullptr;  // bit d set = doc d has a value
@@ -91,6 +92,7 @@ struct FastFieldView {
     }
     int64_t i64(uint32_t doc) const { return ((const int64_t*)h_values)[doc]; }
     uint64_t u64(uint32_t doc) const { return ((const uint64_t*)h_values)[doc]; }
+    double f64(uint32_t doc) const { return ((const double*)h_values)[doc]; }
     std::string dict_entry(uint64_t o) const {
         return std::string((const char*)h_dict_bytes + h_dict_offsets[o],
                            h_dict_offsets[o + 1] - h_dict_offsets[o]);
@@ -201,7 +203,8 @@ struct SplitView {
                 ff.type = ty == "u64"   ? FastFieldView::U64
                           : ty == "i64" ? FastFieldView::I64
                           : ty == "datetime" ? FastFieldView::DATETIME
-                                             : FastFieldView::STR;
+                          : ty == "f64" ? FastFieldView::F64
+                                        : FastFieldView::STR;
                 ff.nullable = f->at("nullable")->b;
                 ff.values = sec(s, "values");
                 if (ff.nullable) ff.nulls = sec(s, "nulls");
@@ -212,6 +215,11 @@ struct SplitView {
                     ff.dict_bytes = sec(s, "dict_bytes");
                     ff.h_dict_offsets = (const uint32_t*)(data + ff.dict_offsets.off);
                     ff.h_dict_bytes = data + ff.dict_bytes.off;
+                } else if (ff.type == FastFieldView::F64) {
+                    const mj::Value* mn = f->get("min_value");
+                    const mj::Value* mx = f->get("max_value");
+                    if (mn) ff.fmin = mn->num();
+                    if (mx) ff.fmax = mx->num();
                 } else {
                     const mj::Value* mn = f->get("min_value");
                     const mj::Value* mx = f->get("max_value");

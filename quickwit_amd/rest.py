@@ -27,7 +27,24 @@ from .api import make_leaf_request
 
 # -------------------------------------------------- ES query DSL -> QueryAst
 # (quickwit-query/src/elastic_query_dsl/ subset the scenario suites use)
-def es_query_to_ast(q):
+def es_query_to_ast(q, schema=None):
+    def numeric_field(name):
+        for f in (schema or {}).get("fields", []):
+            if f["name"] == name and f["type"] in ("u64", "i64", "f64"):
+                return True
+        return False
+
+    def point_or_term(field, value):
+        # term/match over a numeric fast-only column is a point range
+        # (the reference's FastFieldRangeQuery translation for dynamic
+        # fast fields)
+        if numeric_field(field):
+            v = value if isinstance(value, (int, float)) else float(value)
+            return {"type": "range", "field": field,
+                    "lower_bound": {"included": v},
+                    "upper_bound": {"included": v}}
+        return None
+
     if q is None:
         return {"type": "match_all"}
     if "match_all" in q:
@@ -37,6 +54,9 @@ def es_query_to_ast(q):
     if "term" in q:
         [(field, body)] = q["term"].items()
         value = body["value"] if isinstance(body, dict) else body
+        pt = point_or_term(field, value)
+        if pt:
+            return pt
         return {"type": "term", "field": field, "value": str(value)}
     if "terms" in q:
         [(field, values)] = [(k, v) for k, v in q["terms"].items()
@@ -60,6 +80,9 @@ def es_query_to_ast(q):
     if "match" in q:
         [(field, body)] = q["match"].items()
         text = body["query"] if isinstance(body, dict) else body
+        pt = point_or_term(field, text)
+        if pt:
+            return pt
         op = (body.get("operator", "or") if isinstance(body, dict) else "or")
         return {"type": "full_text", "field": field, "text": str(text),
                 "params": {"mode": {"type": "bool", "operator": op.lower()}}}
@@ -70,7 +93,7 @@ def es_query_to_ast(q):
                 items = q["bool"][clause]
                 if isinstance(items, dict):
                     items = [items]
-                out[clause] = [es_query_to_ast(i) for i in items]
+                out[clause] = [es_query_to_ast(i, schema) for i in items]
         if "minimum_should_match" in q["bool"]:
             out["minimum_should_match"] = int(q["bool"]["minimum_should_match"])
         return out
@@ -95,7 +118,7 @@ def _infer_dynamic_fields(explicit_names, doc_batches):
                 elif isinstance(v, int):
                     kinds.add("int" if v >= 0 else "negint")
                 elif isinstance(v, float):
-                    kinds.add("skip")  # f64 fast columns: later round
+                    kinds.add("float")
                 elif isinstance(v, str):
                     kinds.add("str")
                 else:
@@ -109,6 +132,8 @@ def _infer_dynamic_fields(explicit_names, doc_batches):
             fields.append({"name": name, "type": "u64", "fast": True})
         elif kinds <= {"int", "negint"}:
             fields.append({"name": name, "type": "i64", "fast": True})
+        elif kinds <= {"int", "negint", "float"}:
+            fields.append({"name": name, "type": "f64", "fast": True})
         elif kinds == {"str"}:
             fields.append({"name": name, "type": "str", "fast": True})
     return fields
@@ -218,7 +243,7 @@ def create_app(searcher_factory):
         if raw:
             body = json.loads(raw)
         t0 = time.perf_counter()
-        ast = es_query_to_ast(body.get("query"))
+        ast = es_query_to_ast(body.get("query"), idx.schema)
         aggs = body.get("aggs") or body.get("aggregations")
         size = int(body.get("size", 10))
         sort_fields = []

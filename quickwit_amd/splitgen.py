@@ -258,7 +258,8 @@ def _build_fast_field(sec, fspec, num_docs, values, present):
         meta["sec"]["dict_offsets"] = sec.add(dict_offsets)
         meta["sec"]["dict_bytes"] = sec.add(np.frombuffer(dict_bytes, dtype=np.uint8))
     else:
-        dt = np.uint64 if ftype == "u64" else np.int64  # i64/datetime -> i64
+        dt = (np.uint64 if ftype == "u64"
+              else np.float64 if ftype == "f64" else np.int64)
         col = np.zeros(num_docs, dtype=dt)
         vv = np.asarray(
             [v if p else 0 for v, p in zip(values, present)]
@@ -268,8 +269,12 @@ def _build_fast_field(sec, fspec, num_docs, values, present):
         )
         col[:] = vv
         pv = col[present]
-        meta["min_value"] = int(pv.min()) if len(pv) else 0
-        meta["max_value"] = int(pv.max()) if len(pv) else 0
+        if ftype == "f64":
+            meta["min_value"] = float(pv.min()) if len(pv) else 0.0
+            meta["max_value"] = float(pv.max()) if len(pv) else 0.0
+        else:
+            meta["min_value"] = int(pv.min()) if len(pv) else 0
+            meta["max_value"] = int(pv.max()) if len(pv) else 0
         meta["sec"]["values"] = sec.add(col)
     if nullable:
         bits = np.zeros((num_docs + 63) // 64, dtype=np.uint64)

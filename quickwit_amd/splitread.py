@@ -81,6 +81,8 @@ class Split:
             vals = self._sec(field, "values", {1: "u1", 2: "<u2", 4: "<u4"}[w])
         elif f["type"] == "u64":
             vals = self._sec(field, "values", "<u8")
+        elif f["type"] == "f64":
+            vals = self._sec(field, "values", "<f8")
         else:  # i64 / datetime
             vals = self._sec(field, "values", "<i8")
         present = None

@@ -127,6 +127,8 @@ def rest_scenarios():
     suites = {
         "aggregations": ["_setup.quickwit.yaml", "0001-aggregations.yaml",
                          "_teardown.quickwit.yaml"],
+        "sort_orders": ["_setup.quickwit.yaml", "0001-sort-elasticapi.yaml",
+                        "_teardown.quickwit.yaml"],
     }
     keep = {"method", "endpoint", "params", "json", "ndjson", "expected",
             "status_code"}
@@ -141,6 +143,26 @@ def rest_scenarios():
                 if engines and "quickwit" not in engines:
                     continue
                 s = {k: v for k, v in step.items() if k in keep}
+                if "shuffle_ndjson" in step:
+                    # split distribution must not affect results: resolve the
+                    # randomized distribution ONCE at extraction (seed 42,
+                    # same bucketing as run_tests.py distribute_items)
+                    import random
+                    rng = random.Random(42)
+                    nb = rng.randint(step.get("min_splits", 1),
+                                     step.get("max_splits", 5))
+                    buckets = [[] for _ in range(nb)]
+                    for item in step["shuffle_ndjson"]:
+                        buckets[rng.randint(0, nb - 1)].append(item)
+                    for b in buckets:
+                        if not b:
+                            continue
+                        sb = dict(s)
+                        sb["ndjson"] = b
+                        if isinstance(sb["method"], list):
+                            sb["method"] = sb["method"][0]
+                        steps.append(sb)
+                    continue
                 if isinstance(s["method"], list):
                     s["method"] = s["method"][0]
                 if "status_code" in step and step["status_code"] is None:

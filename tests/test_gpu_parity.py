@@ -855,3 +855,63 @@ def test_cache_node_scored_transparent(searchers):
     got, exp = run_both(searchers, q, max_hits=10, sort_fields=[
         {"field_name": "_score", "sort_order": 1}])
     assert_hits_equal(got, exp, scored=True)
+
+
+def test_must_not_only_bool(searchers):
+    # bool with only must_not: implicit match_all base
+    # (tantivy_query_ast.rs:310-322)
+    q = {"type": "bool", "must_not": [
+        {"type": "term", "field": "severity_text", "value": "INFO"}]}
+    got, exp = run_both(searchers, q, max_hits=15)
+    assert_hits_equal(got, exp, scored=False)
+
+
+def test_f64_column_sort_range_agg():
+    # dynamic-style f64 fast column: range pred, sort, histogram + stats
+    docs = [{"timestamp": 1700000000 + i, "severity_text": "INFO",
+             "body": "x", "tenant_id": i % 5} for i in range(500)]
+    import random
+    rng = random.Random(11)
+    for d in docs:
+        if rng.random() < 0.9:
+            d["score_f"] = round(rng.uniform(-50, 50), 3)
+    schema = {"timestamp_field": "timestamp", "fields":
+              splitgen.HDFS_SCHEMA["fields"] +
+              [{"name": "score_f", "type": "f64", "fast": True}]}
+    w = splitgen.SplitWriter(schema, "f64s")
+    w.add_documents(docs)
+    data = w.finalize()
+    gpu, cpu = GpuSearcher(device=0), OracleSearcher()
+    gpu.add_split("f64s", data)
+    cpu.add_split("f64s", data)
+
+    def both(q, **kw):
+        req = make_leaf_request(q, schema, [("f64s", len(docs))], **kw)
+        return gpu.leaf_search(req), cpu.leaf_search(req)
+
+    rq = {"type": "range", "field": "score_f",
+          "lower_bound": {"included": -10.5}, "upper_bound": {"excluded": 20.25}}
+    got, exp = both(rq, max_hits=30)
+    assert got.get("num_hits", 0) == exp.get("num_hits", 0)
+    assert [hid(h) for h in got.get("partial_hits", [])] == \
+           [hid(h) for h in exp.get("partial_hits", [])]
+
+    for order in (0, 1):
+        got, exp = both({"type": "match_all"}, max_hits=20, sort_fields=[
+            {"field_name": "score_f", "sort_order": order}])
+        assert [(hid(h), sv_of(h)) for h in got.get("partial_hits", [])] == \
+               [(hid(h), sv_of(h)) for h in exp.get("partial_hits", [])]
+
+    aggs = {"h": {"histogram": {"field": "score_f", "interval": 10.0},
+                  "aggs": {"st": {"stats": {"field": "score_f"}}}}}
+    req = make_leaf_request({"type": "match_all"}, schema,
+                            [("f64s", len(docs))], max_hits=0, aggregation=aggs)
+    g = gpu.leaf_search(req)
+    e = cpu.leaf_search(req)
+    gj = gpu.finalize_agg_json(g["intermediate_aggregation_result"], aggs)
+    ej = cpu.finalize_agg_json(e["intermediate_aggregation_result"], aggs)
+    for gb, eb in zip(gj["h"]["buckets"], ej["h"]["buckets"]):
+        assert gb["doc_count"] == eb["doc_count"] and gb["key"] == eb["key"]
+        for k in ("count", "min", "max"):
+            assert gb["st"][k] == eb["st"][k], (k, gb, eb)
+        assert math.isclose(gb["st"]["sum"], eb["st"]["sum"], rel_tol=1e-12)

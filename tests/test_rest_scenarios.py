@@ -85,3 +85,25 @@ def test_rest_aggregations_suite_gpu():
     import __graft_entry__
     __graft_entry__.build()
     run_aggregations(lambda: GpuSearcher(device=0))
+
+
+def run_sort_orders(searcher_factory):
+    steps = load_suite("sort_orders")
+    client = make_client(searcher_factory)
+    ran, skipped = replay_suite(client, steps)
+    assert not skipped and ran == len(steps)
+
+
+def test_rest_sort_orders_suite_oracle():
+    from quickwit_amd.api import OracleSearcher
+    import __graft_entry__
+    __graft_entry__.build()
+    run_sort_orders(OracleSearcher)
+
+
+@pytest.mark.gpu
+def test_rest_sort_orders_suite_gpu():
+    from quickwit_amd.api import GpuSearcher
+    import __graft_entry__
+    __graft_entry__.build()
+    run_sort_orders(lambda: GpuSearcher(device=0))
