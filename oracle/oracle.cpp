@@ -669,11 +669,33 @@ static SplitResult search_split(const SplitView& sv, const pb::SearchRequest& re
         // (top_k_collector.rs:663-700; num_hits counting is unaffected)
         if (req.search_after) {
             const pb::PartialHit& c = *req.search_after;
-            cand.erase(std::remove_if(cand.begin(), cand.end(),
-                                      [&](const pb::PartialHit& h) {
-                                          return !after_cursor(h, c, order1, order2);
-                                      }),
-                       cand.end());
+            auto kind_of = [&](const SortSpec& s) {
+                if (s.comp == SortSpec::SCORE) return SortFieldKind::SCORE;
+                if (s.comp != SortSpec::FAST_FIELD || !s.ff)
+                    return SortFieldKind::NONE;
+                switch (s.ff->type) {
+                    case FastFieldView::U64: return SortFieldKind::U64;
+                    case FastFieldView::I64: return SortFieldKind::I64;
+                    case FastFieldView::DATETIME: return SortFieldKind::DATETIME;
+                    case FastFieldView::F64: return SortFieldKind::F64;
+                    default: return SortFieldKind::STR;
+                }
+            };
+            CursorKey k1 = specs.empty()
+                               ? CursorKey{}
+                               : convert_cursor_key(c.sort_value,
+                                                    kind_of(specs[0]), order1);
+            CursorKey k2 = specs.size() < 2
+                               ? CursorKey{}
+                               : convert_cursor_key(c.sort_value2,
+                                                    kind_of(specs[1]), order2);
+            if (!k1.disabled)
+                cand.erase(std::remove_if(cand.begin(), cand.end(),
+                                          [&](const pb::PartialHit& h) {
+                                              return !after_cursor(h, c, k1, k2,
+                                                                   order1, order2);
+                                          }),
+                           cand.end());
         }
         auto cmp = [&](const pb::PartialHit& a, const pb::PartialHit& b) {
             return hit_before(a, b, order1, order2);

@@ -1036,6 +1036,29 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
     // hits trivially enumerable only under doc-id order (leaf.rs default)
     const pb::PartialHit* after =
         req.search_after ? &*req.search_after : nullptr;
+    // typed cursor conversion into the sort fields' u64 domains
+    // (convert_to_u64_ff_val, collector.rs:214-340)
+    auto kind_of = [&](const SortSpec& s) {
+        if (s.comp == SortSpec::SCORE) return SortFieldKind::SCORE;
+        if (s.comp != SortSpec::FAST_FIELD || !s.ff) return SortFieldKind::NONE;
+        switch (s.ff->type) {
+            case FastFieldView::U64: return SortFieldKind::U64;
+            case FastFieldView::I64: return SortFieldKind::I64;
+            case FastFieldView::DATETIME: return SortFieldKind::DATETIME;
+            case FastFieldView::F64: return SortFieldKind::F64;
+            default: return SortFieldKind::STR;
+        }
+    };
+    CursorKey ck1, ck2;
+    if (after) {
+        ck1 = specs.empty() ? CursorKey{}
+                            : convert_cursor_key(after->sort_value,
+                                                 kind_of(specs[0]), order1);
+        ck2 = specs.size() < 2 ? CursorKey{}
+                               : convert_cursor_key(after->sort_value2,
+                                                    kind_of(specs[1]), order2);
+        if (ck1.disabled) after = nullptr;  // cursor before all values
+    }
     bool trivial_hits = pure_match_all && specs.empty() && !after;
     // candidate collection: needed unless hits are trivially enumerable
     bool collect = leaf_max_hits > 0 && !trivial_hits && !fq.match_none;
@@ -1286,7 +1309,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             // filter at hit building, with a growing-K retry on shortfall)
             if (after) {
                 uint64_t ceil_key = ~0ull;
-                SortKey k1 = sort_key_of(after->sort_value);
+                const SortKey& k1 = ck1.key;  // converted to the field domain
                 if (wide) {
                     if (!k1.has) ceil_key = 0;  // cursor in the None region
                     else {
@@ -1300,7 +1323,8 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                             int64_t ms = ns / 1000000 - ((ns % 1000000) < 0 ? 1 : 0);
                             S = i64_to_u64(ms);
                         }
-                        // I64: same i64_to_u64 map; U64/STR ord: identity
+                        // I64/F64: same u64 map as the device key;
+                        // U64/STR ord: identity
                         ceil_key = (order1 == 0) ? ~S : S;
                     }
                 } else {
@@ -1495,8 +1519,8 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     out.hits.erase(
                         std::remove_if(out.hits.begin(), out.hits.end(),
                                        [&](const pb::PartialHit& h) {
-                                           return !after_cursor(h, c, order1,
-                                                                order2);
+                                           return !after_cursor(h, c, ck1, ck2,
+                                                                order1, order2);
                                        }),
                         out.hits.end());
                 }

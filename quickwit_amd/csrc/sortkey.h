@@ -6,6 +6,8 @@
 //   - ties broken by GlobalDocId (split_id, segment_ord, doc_id) in the
 //     order of the first sort field (Desc by default).
 #pragma once
+#include <algorithm>
+#include <climits>
 #include <cstdint>
 #include <cstring>
 #include <string>
@@ -75,13 +77,105 @@ inline bool hit_before(const pb::PartialHit& a, const pb::PartialHit& b, int ord
     return false;
 }
 
+// field type tags for cursor conversion (SortFieldType analog)
+enum class SortFieldKind { NONE, SCORE, U64, I64, DATETIME, F64, STR };
+
+// converted search_after cursor value, in the target field's u64-mapped
+// domain (SearchAfterSegment::new + convert_to_u64_ff_val,
+// collector.rs:214-340). `disabled` = the cursor sorts before every possible
+// value in the requested order: the whole search_after filter is off.
+struct CursorKey {
+    bool disabled = false;
+    SortKey key;  // has=false: None (cursor past all valued docs)
+};
+
+inline CursorKey convert_cursor_key(const pb::SortByValue& v, SortFieldKind ft,
+                                    int order /*0 asc 1 desc*/) {
+    CursorKey out;
+    if (v.kind == pb::SortByValue::NONE) return out;  // None cursor value
+    const bool asc = order == 0;
+    auto some = [&](uint64_t k) { out.key = {true, k}; };
+    switch (ft) {
+        case SortFieldKind::NONE:
+            return out;  // unknown sort field: values are all None
+        case SortFieldKind::SCORE:
+            some(sort_key_of(v).key);
+            return out;
+        case SortFieldKind::STR:
+            // ord cursor arrives as U64 from our own responses
+            some(sort_key_of(v).key);
+            return out;
+        case SortFieldKind::U64:
+            switch (v.kind) {
+                case pb::SortByValue::U64: some(v.u64); break;
+                case pb::SortByValue::I64:
+                    if (v.i64 < 0 && asc) out.disabled = true;
+                    else if (v.i64 < 0) some(0);  // desc: matches nothing after
+                    else some(uint64_t(v.i64));
+                    break;
+                case pb::SortByValue::F64:
+                    if ((v.f64 < 0.0 && asc) ||
+                        (v.f64 > 18446744073709551615.0 && !asc))
+                        out.disabled = true;
+                    else
+                        some(v.f64 < 0.0 ? 0
+                             : v.f64 >= 18446744073709551615.0
+                                 ? ~0ull
+                                 : uint64_t(v.f64));
+                    break;
+                default: some(v.boolean ? 1 : 0); break;
+            }
+            return out;
+        case SortFieldKind::I64:
+        case SortFieldKind::DATETIME:
+            switch (v.kind) {
+                case pb::SortByValue::I64: some(i64_to_u64(v.i64)); break;
+                case pb::SortByValue::U64: {
+                    uint64_t val = v.u64;
+                    if (!asc && val > uint64_t(INT64_MAX)) out.disabled = true;
+                    else
+                        some(i64_to_u64(int64_t(
+                            std::min<uint64_t>(val, uint64_t(INT64_MAX)))));
+                    break;
+                }
+                case pb::SortByValue::F64: {
+                    double val = v.f64;
+                    if ((val < -9.223372036854776e18 && asc) ||
+                        (val > 9.223372036854775e18 && !asc))
+                        out.disabled = true;
+                    else {
+                        int64_t vi = val <= -9.223372036854776e18 ? INT64_MIN
+                                     : val >= 9.223372036854775e18
+                                         ? INT64_MAX
+                                         : int64_t(val);
+                        some(i64_to_u64(vi));
+                    }
+                    break;
+                }
+                default: some(i64_to_u64(v.boolean ? 1 : 0)); break;
+            }
+            return out;
+        case SortFieldKind::F64:
+            switch (v.kind) {
+                case pb::SortByValue::F64: some(f64_to_u64(v.f64)); break;
+                case pb::SortByValue::U64: some(f64_to_u64(double(v.u64))); break;
+                case pb::SortByValue::I64: some(f64_to_u64(double(v.i64))); break;
+                default: some(f64_to_u64(v.boolean ? 1.0 : 0.0)); break;
+            }
+            return out;
+    }
+    return out;
+}
+
 // search_after filter (top_k_collector.rs:663-700 + SearchAfterSegment
 // :821-872): a hit is kept iff the comparison chain
 // (sort_value, sort_value2, then — only when the cursor carries a doc
 // address — split_id, segment_ord, doc_id) is strictly Less under
 // SortOrder::compare (Desc: natural, Asc: reversed; Some > None).
 inline bool after_cursor(const pb::PartialHit& h, const pb::PartialHit& c,
-                         int order1, int order2) {
+                         const CursorKey& k1, const CursorKey& k2, int order1,
+                         int order2) {
+    if (k1.disabled) return true;  // cursor before all values: filter off
     auto cmpo = [](const SortKey& a, const SortKey& b, int order) -> int {
         if (a.has && b.has) {
             if (a.key == b.key) return 0;
@@ -92,9 +186,9 @@ inline bool after_cursor(const pb::PartialHit& h, const pb::PartialHit& c,
         if (b.has) return -1;  // (None, Some) -> Less
         return 0;
     };
-    int r = cmpo(sort_key_of(h.sort_value), sort_key_of(c.sort_value), order1);
+    int r = cmpo(sort_key_of(h.sort_value), k1.key, order1);
     if (r) return r < 0;
-    r = cmpo(sort_key_of(h.sort_value2), sort_key_of(c.sort_value2), order2);
+    r = cmpo(sort_key_of(h.sort_value2), k2.disabled ? SortKey{} : k2.key, order2);
     if (r) return r < 0;
     if (c.split_id.empty()) return false;  // equal values, no doc tiebreak
     int cs = h.split_id.compare(c.split_id);

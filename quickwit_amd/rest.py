@@ -261,6 +261,24 @@ def create_app(searcher_factory):
         req = make_leaf_request(ast, idx.schema, idx.splits, max_hits=size,
                                 sort_fields=sort_fields or None,
                                 aggregation=aggs)
+        sa = body.get("search_after")
+        if sa:
+            # ES search_after literal array -> typed SortByValue cursor
+            # (SearchAfterSegment conversion happens engine-side)
+            def lit_to_sv(x):
+                if isinstance(x, bool):
+                    return {"boolean": x}
+                if isinstance(x, float):
+                    return {"f64": x}
+                if isinstance(x, int):
+                    return {"i64": x} if x < 0 else {"u64": x}
+                raise ValueError(f"unsupported search_after literal: {x!r}")
+            cursor = {}
+            if len(sa) >= 1:
+                cursor["sort_value"] = lit_to_sv(sa[0])
+            if len(sa) >= 2:
+                cursor["sort_value2"] = lit_to_sv(sa[1])
+            req["search_request"]["search_after"] = cursor
         resp = idx.searcher.leaf_search(req)
         took_ms = int((time.perf_counter() - t0) * 1e3)
         hits = []
@@ -270,8 +288,19 @@ def create_app(searcher_factory):
             src = idx.source_doc(sid, doc_id)
             hit = {"_index": iid, "_id": f"{sid}:{doc_id}", "_source": src}
             sv = h.get("sort_value", {})
-            if "f64" in sv:
+            if "f64" in sv and sort_fields and \
+                    sort_fields[0]["field_name"] == "_score":
                 hit["_score"] = sv["f64"]
+            if sort_fields:
+                def sv_raw(v):
+                    for k in ("u64", "i64", "f64", "boolean"):
+                        if k in v:
+                            return v[k]
+                    return None
+                srt = [sv_raw(h.get("sort_value", {}))]
+                if len(sort_fields) > 1:
+                    srt.append(sv_raw(h.get("sort_value2", {})))
+                hit["sort"] = srt
             hits.append(hit)
         out = {"took": took_ms, "timed_out": False,
                "hits": {"total": {"value": resp.get("num_hits", 0),

@@ -129,15 +129,37 @@ def rest_scenarios():
                          "_teardown.quickwit.yaml"],
         "sort_orders": ["_setup.quickwit.yaml", "0001-sort-elasticapi.yaml",
                         "_teardown.quickwit.yaml"],
+        "search_after": ["_setup.quickwit.yaml",
+                         "0001-search_after_edge_case.yaml",
+                         "_teardown.quickwit.yaml"],
     }
     keep = {"method", "endpoint", "params", "json", "ndjson", "expected",
             "status_code"}
     out = {}
     for suite, files in suites.items():
+        # suite context defaults (run_tests.py Visitor context stacking)
+        ctx = {}
+        ctx_path = os.path.join(base, suite, "_ctx.yaml")
+        if os.path.exists(ctx_path):
+            ctx = yaml.safe_load(open(ctx_path)) or {}
+        ctx_endpoint = ctx.get("endpoint")
+        api_root = ctx.get("api_root", "")
+        prefix = "_elastic/" if api_root.rstrip("/").endswith("_elastic") else ""
         steps = []
         for fn in files:
+            is_scenario = not fn.startswith("_")
             for step in yaml.safe_load_all(open(os.path.join(base, suite, fn))):
-                if not isinstance(step, dict) or "method" not in step:
+                if not isinstance(step, dict):
+                    continue
+                if is_scenario:  # numbered scenarios inherit the suite ctx
+                    if "method" not in step:
+                        step["method"] = ctx.get("method", "GET")
+                    if "endpoint" not in step and ctx_endpoint:
+                        step["endpoint"] = ctx_endpoint
+                    if "endpoint" in step and prefix and \
+                            not step["endpoint"].startswith("_elastic"):
+                        step["endpoint"] = prefix + step["endpoint"]
+                if "method" not in step:
                     continue
                 engines = step.get("engines")
                 if engines and "quickwit" not in engines:

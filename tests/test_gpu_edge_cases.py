@@ -97,11 +97,13 @@ def test_field_presence(ragged):
     assert got.get("num_hits") == 3
 
 
-def test_must_not_only_matches_nothing(ragged):
+def test_must_not_only_implicit_match_all(ragged):
+    # bool with only must_not: implicit match_all base
+    # (tantivy_query_ast.rs:310-322) — docs 1,2 contain "hello", 0,3 don't
     got, exp = ragged({"type": "bool", "must_not": [
         {"type": "term", "field": "body", "value": "hello"}]}, max_hits=10)
     same_hits(got, exp)
-    assert got.get("num_hits", 0) == 0
+    assert got.get("num_hits", 0) == 2
 
 
 def test_msm_exceeds_clause_count(ragged):
