@@ -69,6 +69,8 @@ struct AggDev {
     // the double path — see histo_bucket in kernels.hip)
     uint32_t int_fast;
     uint32_t lds_slot;      // 0 = agg_hist, 1 = agg_terms, 0xFF = global atomics
+    uint32_t lds_rep;       // 1 or 2: interleaved LDS histogram copies
+                            // (lane-parity split halves same-bucket conflicts)
     int64_t i_interval;
     int64_t i_offset;
     double inv_interval;    // 1.0 / i_interval

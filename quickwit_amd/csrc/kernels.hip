@@ -404,7 +404,7 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
     const uint8_t* af_tcol = nullptr;
     int64_t af_ioff = 0, af_ivl = 1, af_base = 0;
     double af_inv = 0;
-    uint32_t af_twidth = 0;
+    uint32_t af_twidth = 0, af_hrep = 0, af_trep = 0, af_par = 0;
     bool af_terms = false;
     if (NA && q.agg_fast) {
         const AggDev& a0 = aggs[0];
@@ -413,10 +413,13 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
         af_ivl = a0.i_interval;
         af_inv = a0.inv_interval;
         af_base = a0.base_index;
+        af_hrep = a0.lds_rep;
+        af_par = lane_id() & 1u;  // copy index for 2-way replication
         if (q.n_aggs > 1) {
             af_terms = true;
             af_tcol = q.split + aggs[1].values_off;
             af_twidth = aggs[1].value_width;
+            af_trep = aggs[1].lds_rep;
         }
     }
 
@@ -491,12 +494,13 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                 int64_t idx = int64_t(floor(double(num) * af_inv));
                 if (idx * af_ivl > num) --idx;
                 else if ((idx + 1) * af_ivl <= num) ++idx;
-                atomicAdd(&sc_agg_hist[uint32_t(idx - af_base)], 1u);
+                uint32_t hb = uint32_t(idx - af_base);
+                atomicAdd(&sc_agg_hist[af_hrep == 2 ? hb * 2 + af_par : hb], 1u);
                 if (af_terms) {
                     uint64_t o = af_twidth == 2 ? ((const uint16_t*)af_tcol)[d]
                                  : (af_twidth == 1 ? af_tcol[d]
                                                    : ((const uint32_t*)af_tcol)[d]);
-                    atomicAdd(&sc_agg_terms[o], 1u);
+                    atomicAdd(&sc_agg_terms[af_trep == 2 ? o * 2 + af_par : o], 1u);
                 }
             }
             if (do_count && threadIdx.x == 0) tile_counts[tile] = tile_hi - tile_lo;
@@ -559,7 +563,10 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                             if (!a.n_buckets) continue;  // missing/non-str column
                             uint64_t o = agg_ord(q, a.values_off, a.value_width, d);
                             if (a.lds_slot == 1)
-                                atomicAdd(&sc_agg_terms[o], 1u);
+                                atomicAdd(&sc_agg_terms[a.lds_rep == 2
+                                                            ? o * 2 + (lane_id() & 1u)
+                                                            : o],
+                                          1u);
                             else
                                 atomicAdd((unsigned long long*)(q.results + a.counts_out) + o,
                                           1ull);
@@ -577,7 +584,11 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                             int64_t idx = histo_bucket(q, a, d);
                             if (idx < 0 || idx >= int64_t(a.n_buckets)) continue;
                             if (a.lds_slot == 0)
-                                atomicAdd(&sc_agg_hist[uint32_t(idx)], 1u);
+                                atomicAdd(&sc_agg_hist[a.lds_rep == 2
+                                                           ? uint32_t(idx) * 2 +
+                                                                 (lane_id() & 1u)
+                                                           : uint32_t(idx)],
+                                          1u);
                             else
                                 atomicAdd((unsigned long long*)(q.results + a.counts_out) +
                                               idx, 1ull);
@@ -678,10 +689,12 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
             const AggDev& a = aggs[ai];
             if (a.lds_slot > 1) continue;
             uint32_t* src = a.lds_slot == 1 ? sc_agg_terms : sc_agg_hist;
-            for (uint32_t i = threadIdx.x; i < a.n_buckets; i += TILE_THREADS)
-                if (src[i])
+            for (uint32_t i = threadIdx.x; i < a.n_buckets; i += TILE_THREADS) {
+                uint32_t v = a.lds_rep == 2 ? src[2 * i] + src[2 * i + 1] : src[i];
+                if (v)
                     atomicAdd((unsigned long long*)(q.results + a.counts_out) + i,
-                              (unsigned long long)src[i]);
+                              (unsigned long long)v);
+            }
         }
         if (threadIdx.x < q.n_aggs && threadIdx.x < 4) {
             const AggDev& a = aggs[threadIdx.x];

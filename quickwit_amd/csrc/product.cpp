@@ -725,6 +725,7 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
     // table; everything else accumulates with global atomics.
     bool slot0 = false, slot1 = false;
     for (AggDev& a : ap.devs) {
+        a.lds_rep = 1;
         if (a.kind == AGGD_HISTO && !slot0 && a.n_buckets &&
             a.n_buckets <= AGG_LDS_BUCKETS) {
             a.lds_slot = 0;
@@ -734,6 +735,9 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
             a.lds_slot = 1;
             slot1 = true;
         }
+        // two interleaved copies when they fit the LDS array: lane-parity
+        // split halves same-bucket atomic serialization (Zipf ords)
+        if (a.lds_slot <= 1 && a.n_buckets * 2 <= AGG_LDS_BUCKETS) a.lds_rep = 2;
     }
     ap.out_bytes = off - out_base;
     // init pattern: zeros except sub min slots = f64_sortable(+inf) pattern max
