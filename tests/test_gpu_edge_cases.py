@@ -99,11 +99,12 @@ def test_field_presence(ragged):
 
 def test_must_not_only_implicit_match_all(ragged):
     # bool with only must_not: implicit match_all base
-    # (tantivy_query_ast.rs:310-322) — docs 1,2 contain "hello", 0,3 don't
+    # (tantivy_query_ast.rs:310-322) — docs 1,2,3 contain "hello"; only doc 0
+    # (empty body) survives
     got, exp = ragged({"type": "bool", "must_not": [
         {"type": "term", "field": "body", "value": "hello"}]}, max_hits=10)
     same_hits(got, exp)
-    assert got.get("num_hits", 0) == 2
+    assert got.get("num_hits", 0) == 1
 
 
 def test_msm_exceeds_clause_count(ragged):
