@@ -321,6 +321,17 @@ struct SplitSearcher {
                 return false;
             return true;
         }
+        if (f.type == FastFieldView::STR) {
+            // lexicographic bounds over the sorted ord dictionary
+            std::string v = f.dict_entry(f.ord(d));
+            if (lo.kind != Bound::UNBOUNDED &&
+                (lo.kind == Bound::INCLUDED ? v < lo.sval : v <= lo.sval))
+                return false;
+            if (hi.kind != Bound::UNBOUNDED &&
+                (hi.kind == Bound::INCLUDED ? v > hi.sval : v >= hi.sval))
+                return false;
+            return true;
+        }
         int64_t v = f.i64(d);
         if (lo.kind != Bound::UNBOUNDED &&
             (lo.kind == Bound::INCLUDED ? v < lo.ival : v <= lo.ival))

@@ -271,7 +271,14 @@ __device__ __forceinline__ bool eval_pred(const QueryDev& q, const PredDev& p,
     }
     if (ok && p.type != PRED_PRESENCE) {
         if (p.type == PRED_RANGE_U64) {
-            uint64_t v = ((const uint64_t*)(q.split + p.values_off))[doc];
+            const uint8_t* col = q.split + p.values_off;
+            uint64_t v;
+            switch (p.value_width) {  // str ord columns are 1/2/4 wide
+                case 1: v = col[doc]; break;
+                case 2: v = ((const uint16_t*)col)[doc]; break;
+                case 4: v = ((const uint32_t*)col)[doc]; break;
+                default: v = ((const uint64_t*)col)[doc]; break;
+            }
             if (p.flags & PRED_LO_INCLUDED) ok &= v >= uint64_t(p.lo);
             if (p.flags & PRED_LO_EXCLUDED) ok &= v > uint64_t(p.lo);
             if (p.flags & PRED_HI_INCLUDED) ok &= v <= uint64_t(p.hi);

@@ -337,16 +337,28 @@ static void add_range_pred(FlatQuery& fq, const SplitView& sv, const PlanNode& n
         if (f->type == FastFieldView::F64) {
             memcpy(&p.lo, &n.lo.fval, 8);  // bit-cast double bounds
             memcpy(&p.hi, &n.hi.fval, 8);
+        } else if (f->type == FastFieldView::STR) {
+            // lexicographic bounds -> ord space [lo, hi) over the sorted dict
+            p.type = PRED_RANGE_U64;
+            p.flags &= PRED_NEGATED;  // rebuild the bound flags below
+            if (n.lo.kind != Bound::UNBOUNDED) {
+                p.flags |= PRED_LO_INCLUDED;
+                p.lo = int64_t(
+                    f->str_bound_ord(n.lo.sval, n.lo.kind == Bound::EXCLUDED));
+            }
+            if (n.hi.kind != Bound::UNBOUNDED) {
+                p.flags |= PRED_HI_EXCLUDED;
+                p.hi = int64_t(
+                    f->str_bound_ord(n.hi.sval, n.hi.kind == Bound::INCLUDED));
+            }
         } else {
             p.lo = n.lo.ival;
             p.hi = n.hi.ival;
         }
-        if (f->type == FastFieldView::STR)
-            throw std::runtime_error("range over str fast field not supported");
     }
     p.values_off = f->values.off;
     p.nulls_off = f->nullable ? f->nulls.off : 0;
-    p.value_width = 8;
+    p.value_width = f->type == FastFieldView::STR ? uint32_t(f->ord_width) : 8;
     fq.preds.push_back(p);
 }
 

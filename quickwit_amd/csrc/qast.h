@@ -104,6 +104,7 @@ struct Bound {
     int64_t ival = 0;    // canonical value for u64/i64/datetime(ms) columns
     bool from_f64 = false;
     double fval = 0;
+    std::string sval;    // str columns: lexicographic bound (ord-mapped)
 };
 
 struct PlanNode {
@@ -207,6 +208,12 @@ inline Bound parse_bound(const mj::Value* v, const SchemaField& f) {
         b.kind = Bound::EXCLUDED;
         lit = exc;
     } else throw std::runtime_error("bad range bound");
+    if (f.type == "str") {
+        if (lit->kind != mj::Value::STR)
+            throw std::runtime_error("str range bound must be a string");
+        b.sval = lit->s;
+        return b;
+    }
     if (f.type == "datetime") {
         b.ival = parse_datetime_ms(lit);
     } else if (lit->kind == mj::Value::INT) {
@@ -259,47 +266,341 @@ inline PlanNode full_text_plan(const std::string& field, const std::string& text
 // or bare tokens over default fields, implicit OR (the subset quickwit's
 // query parser covers for plain log searches; anything else -> error =
 // SearchError::InvalidQuery, never a silent wrong answer)
-inline PlanNode user_input_plan(const std::string& text,
-                                const std::vector<std::string>& default_fields,
-                                const Schema& schema) {
-    std::vector<PlanNode> clauses;
+// ------------------------------------------------ query-language parser
+// Restates the tantivy query grammar subset quickwit's user_input queries
+// use (tantivy-query-grammar 0.26, external — pinned by the qw_search_api
+// golden scenarios): clause sequences with +/-/NOT prefixes and AND/OR
+// connectors, parentheses, `*`, field:*, field:value, field:"quoted",
+// field:[a TO b] / {a TO b}, field:>=v >v <=v <v, field:IN [a b], ^boost.
+// Out of scope r1: regex //, slop ~N, multi-token phrases (need positions).
+struct QTok {
+    enum Kind { LPAREN, RPAREN, AND, OR, NOT, PLUS, MINUS, LIT } kind = LIT;
+    std::string field;   // LIT: optional field ("" = default fields)
+    std::string text;    // LIT: raw literal text (quotes stripped)
+    bool quoted = false;
+    bool range_ge = false, range_gt = false, range_le = false, range_lt = false;
+    bool bracket = false;  // [a TO b] / {a TO b}
+    bool lo_incl = true, hi_incl = true;
+    std::string lo, hi;    // bracket bounds ("*" = unbounded)
+    bool set = false;      // IN [..]
+    std::vector<std::string> set_vals;
+    float boost = 1.0f;
+};
+
+inline std::vector<QTok> qlex(const std::string& s) {
+    std::vector<QTok> out;
     size_t i = 0;
-    while (i < text.size()) {
-        while (i < text.size() && text[i] == ' ') ++i;
-        size_t j = text.find(' ', i);
-        if (j == std::string::npos) j = text.size();
-        if (j > i) {
-            std::string tok = text.substr(i, j - i);
-            if (tok.find('"') != std::string::npos || tok.find('(') != std::string::npos ||
-                tok == "AND" || tok == "OR" || tok == "NOT" || tok[0] == '-' ||
-                tok[0] == '+')
-                throw std::runtime_error("user_input syntax not supported: " + tok);
-            size_t c = tok.find(':');
-            if (c != std::string::npos) {
-                clauses.push_back(
-                    full_text_plan(tok.substr(0, c), tok.substr(c + 1), "or", schema));
+    auto read_quoted = [&](std::string* t) {
+        ++i;  // opening quote
+        while (i < s.size() && s[i] != '"') t->push_back(s[i++]);
+        if (i >= s.size()) throw std::runtime_error("unterminated quote");
+        ++i;
+    };
+    while (i < s.size()) {
+        char c = s[i];
+        if (c == ' ' || c == '\t') { ++i; continue; }
+        if (c == '(') { out.push_back({QTok::LPAREN}); ++i; continue; }
+        if (c == ')') {
+            QTok t{QTok::RPAREN};
+            ++i;
+            out.push_back(t);
+            continue;
+        }
+        if (c == '+') { out.push_back({QTok::PLUS}); ++i; continue; }
+        if (c == '-') { out.push_back({QTok::MINUS}); ++i; continue; }
+        // word or field:...
+        QTok t;
+        std::string word;
+        if (c == '"') {
+            read_quoted(&t.text);
+            t.quoted = true;
+        } else {
+            while (i < s.size() && s[i] != ' ' && s[i] != '\t' && s[i] != '(' &&
+                   s[i] != ')' && s[i] != ':' && s[i] != '^')
+                word.push_back(s[i++]);
+            if (i < s.size() && s[i] == ':') {
+                t.field = word;
+                ++i;
+                // value part
+                if (i < s.size() && s[i] == '"') {
+                    read_quoted(&t.text);
+                    t.quoted = true;
+                } else if (i < s.size() && (s[i] == '[' || s[i] == '{')) {
+                    t.bracket = true;
+                    t.lo_incl = s[i] == '[';
+                    ++i;
+                    auto piece = [&]() {
+                        std::string p;
+                        if (i < s.size() && s[i] == '"') read_quoted(&p);
+                        else
+                            while (i < s.size() && s[i] != ' ' && s[i] != ']' &&
+                                   s[i] != '}')
+                                p.push_back(s[i++]);
+                        return p;
+                    };
+                    t.lo = piece();
+                    while (i < s.size() && s[i] == ' ') ++i;
+                    std::string to = piece();
+                    if (to != "TO") throw std::runtime_error("range needs TO");
+                    while (i < s.size() && s[i] == ' ') ++i;
+                    t.hi = piece();
+                    if (i >= s.size() || (s[i] != ']' && s[i] != '}'))
+                        throw std::runtime_error("unterminated range");
+                    t.hi_incl = s[i] == ']';
+                    ++i;
+                } else if (i + 1 < s.size() && s[i] == '>' && s[i + 1] == '=') {
+                    t.range_ge = true;
+                    i += 2;
+                    while (i < s.size() && s[i] != ' ') t.text.push_back(s[i++]);
+                } else if (i + 1 < s.size() && s[i] == '<' && s[i + 1] == '=') {
+                    t.range_le = true;
+                    i += 2;
+                    while (i < s.size() && s[i] != ' ') t.text.push_back(s[i++]);
+                } else if (i < s.size() && s[i] == '>') {
+                    t.range_gt = true;
+                    ++i;
+                    while (i < s.size() && s[i] != ' ') t.text.push_back(s[i++]);
+                } else if (i < s.size() && s[i] == '<') {
+                    t.range_lt = true;
+                    ++i;
+                    while (i < s.size() && s[i] != ' ') t.text.push_back(s[i++]);
+                } else {
+                    // plain value or IN [..]
+                    std::string v;
+                    while (i < s.size() && s[i] != ' ' && s[i] != ')' && s[i] != '^')
+                        v.push_back(s[i++]);
+                    if (v == "IN") {
+                        while (i < s.size() && s[i] == ' ') ++i;
+                        if (i >= s.size() || s[i] != '[')
+                            throw std::runtime_error("IN needs [..]");
+                        ++i;
+                        t.set = true;
+                        std::string cur;
+                        while (i < s.size() && s[i] != ']') {
+                            if (s[i] == ' ') {
+                                if (!cur.empty()) t.set_vals.push_back(cur);
+                                cur.clear();
+                            } else cur.push_back(s[i]);
+                            ++i;
+                        }
+                        if (!cur.empty()) t.set_vals.push_back(cur);
+                        if (i >= s.size()) throw std::runtime_error("unterminated IN");
+                        ++i;
+                    } else t.text = v;
+                }
             } else {
-                const std::vector<std::string>& dfs =
-                    default_fields.empty() ? schema.default_search_fields : default_fields;
-                if (dfs.empty()) throw std::runtime_error("no default search fields");
-                PlanNode b;
-                b.kind = PlanNode::BOOL;
-                for (auto& df : dfs) b.should.push_back(full_text_plan(df, tok, "or", schema));
-                clauses.push_back(dfs.size() == 1 ? std::move(b.should[0]) : std::move(b));
+                t.text = word;
+                if (word == "AND") { out.push_back({QTok::AND}); continue; }
+                if (word == "OR") { out.push_back({QTok::OR}); continue; }
+                if (word == "NOT") { out.push_back({QTok::NOT}); continue; }
             }
         }
-        i = j + 1;
+        if (i < s.size() && s[i] == '^') {
+            ++i;
+            std::string b;
+            while (i < s.size() && s[i] != ' ') b.push_back(s[i++]);
+            t.boost = float(atof(b.c_str()));
+        }
+        out.push_back(std::move(t));
     }
-    if (clauses.empty()) {
+    return out;
+}
+
+inline PlanNode qtok_leaf(const QTok& t, const std::vector<std::string>& dfs,
+                          const Schema& schema);
+
+// string literal -> typed range bound for one schema field
+inline Bound qbound(const std::string& text, Bound::Kind kind, const SchemaField& f) {
+    Bound b;
+    if (text.empty() || text == "*") return b;
+    b.kind = kind;
+    if (f.type == "datetime") {
+        mj::Value v;
+        v.kind = mj::Value::STR;
+        v.s = text;
+        b.ival = parse_datetime_ms(&v);
+    } else if (f.type == "str") {
+        b.sval = text;
+    } else if (f.type == "f64") {
+        b.fval = atof(text.c_str());
+        b.ival = int64_t(b.fval);
+        b.from_f64 = true;
+    } else {
+        b.ival = strtoll(text.c_str(), nullptr, 10);
+        b.fval = double(b.ival);
+    }
+    return b;
+}
+
+// parse a clause sequence (until RPAREN/end). The occur model restated from
+// tantivy's grammar: leading +/-/NOT set Must/MustNot on the next leaf; an
+// AND connector promotes BOTH neighbors to Must, OR keeps Should.
+inline PlanNode qparse_clause(const std::vector<QTok>& toks, size_t* pos,
+                              const std::vector<std::string>& dfs,
+                              const Schema& schema) {
+    struct Item {
+        int occur = 0;  // 0 default(Should), 1 Must, 2 MustNot
+        PlanNode node;
+    };
+    std::vector<Item> items;
+    bool saw_and = false;
+    int pending = 0;
+    while (*pos < toks.size()) {
+        const QTok& t = toks[*pos];
+        if (t.kind == QTok::RPAREN) { ++(*pos); break; }
+        if (t.kind == QTok::AND) { saw_and = true; ++(*pos); continue; }
+        if (t.kind == QTok::OR) { ++(*pos); continue; }
+        if (t.kind == QTok::PLUS) { pending = 1; ++(*pos); continue; }
+        if (t.kind == QTok::MINUS || t.kind == QTok::NOT) {
+            pending = 2;
+            ++(*pos);
+            continue;
+        }
+        Item it;
+        it.occur = pending;
+        pending = 0;
+        if (t.kind == QTok::LPAREN) {
+            ++(*pos);
+            it.node = qparse_clause(toks, pos, dfs, schema);
+        } else {
+            it.node = qtok_leaf(t, dfs, schema);
+            ++(*pos);
+        }
+        items.push_back(std::move(it));
+    }
+    if (items.empty()) {
         PlanNode n;
         n.kind = PlanNode::MATCH_ALL;
         return n;
     }
-    if (clauses.size() == 1) return clauses[0];
+    if (items.size() == 1 && items[0].occur == 0) return std::move(items[0].node);
     PlanNode b;
     b.kind = PlanNode::BOOL;
-    b.should = std::move(clauses);
+    for (Item& it : items) {
+        int occur = it.occur ? it.occur : (saw_and ? 1 : 0);
+        if (occur == 1) b.must.push_back(std::move(it.node));
+        else if (occur == 2) b.must_not.push_back(std::move(it.node));
+        else b.should.push_back(std::move(it.node));
+    }
+    if (items.size() == 1 && items[0].occur == 2) {
+        // single NOT clause: implicit match_all handled downstream
+    }
     return b;
+}
+
+inline PlanNode qtok_leaf_impl(const QTok& t, const std::vector<std::string>& dfs_in,
+                               const Schema& schema) {
+    const std::vector<std::string>& dfs =
+        dfs_in.empty() ? schema.default_search_fields : dfs_in;
+    PlanNode n;
+    bool is_range = t.bracket || t.range_ge || t.range_gt || t.range_le || t.range_lt;
+    if (t.field.empty() && !t.quoted && t.text == "*") {
+        n.kind = PlanNode::MATCH_ALL;
+        return n;
+    }
+    if (!t.field.empty() && !t.quoted && !is_range && !t.set && t.text == "*") {
+        n.kind = PlanNode::FIELD_PRESENCE;  // field:* -> Exists
+        n.field = t.field;
+        n.boost = t.boost;
+        return n;
+    }
+    if (t.set) {
+        // field:IN [a b] -> TermSetQuery (user_input_query.rs:158-176)
+        n.kind = PlanNode::BOOL;
+        n.const_score = true;
+        std::vector<std::string> fields =
+            t.field.empty() ? dfs : std::vector<std::string>{t.field};
+        if (fields.empty()) throw std::runtime_error("set query needs a field");
+        for (auto& f : fields)
+            for (auto& v : t.set_vals) {
+                PlanNode c;
+                c.kind = PlanNode::TERM;
+                c.field = f;
+                c.value = v;
+                n.should.push_back(std::move(c));
+            }
+        if (n.should.empty()) n.kind = PlanNode::MATCH_NONE;
+        n.boost = t.boost;
+        return n;
+    }
+    if (is_range) {
+        if (t.field.empty())
+            throw std::runtime_error("range query without field is not supported");
+        const SchemaField* f = schema.field(t.field);
+        if (!f) throw std::runtime_error("range on unknown field: " + t.field);
+        n.kind = PlanNode::RANGE;
+        n.field = t.field;
+        if (t.bracket) {
+            n.lo = qbound(t.lo, t.lo_incl ? Bound::INCLUDED : Bound::EXCLUDED, *f);
+            n.hi = qbound(t.hi, t.hi_incl ? Bound::INCLUDED : Bound::EXCLUDED, *f);
+        } else {
+            if (t.range_ge) n.lo = qbound(t.text, Bound::INCLUDED, *f);
+            if (t.range_gt) n.lo = qbound(t.text, Bound::EXCLUDED, *f);
+            if (t.range_le) n.hi = qbound(t.text, Bound::INCLUDED, *f);
+            if (t.range_lt) n.hi = qbound(t.text, Bound::EXCLUDED, *f);
+        }
+        n.boost = t.boost;
+        return n;
+    }
+    if (!t.field.empty()) {
+        const SchemaField* f = schema.field(t.field);
+        if (f && f->type != "text") {
+            // term on a typed fast column -> point range (the reference's
+            // FastFieldRangeQuery translation for fast-only fields)
+            n.kind = PlanNode::RANGE;
+            n.field = t.field;
+            n.lo = qbound(t.text, Bound::INCLUDED, *f);
+            n.hi = qbound(t.text, Bound::INCLUDED, *f);
+            n.boost = t.boost;
+            return n;
+        }
+        if (t.quoted) {
+            // phrase: tokenize with the field's tokenizer; single-token
+            // phrases (e.g. raw tokenizer keeps spaces) are plain terms,
+            // true multi-token phrases need positions (later round)
+            std::vector<std::string> toks =
+                tokenize(t.text, f ? f->tokenizer : "default");
+            if (toks.empty()) {
+                n.kind = PlanNode::MATCH_NONE;
+                return n;
+            }
+            if (toks.size() > 1)
+                throw std::runtime_error(
+                    "phrase query needs positions (not in round 1)");
+            n.kind = PlanNode::TERM;
+            n.field = t.field;
+            n.value = toks[0];
+            n.boost = t.boost;
+            return n;
+        }
+        PlanNode p = full_text_plan(t.field, t.text, "or", schema);
+        p.boost *= t.boost;
+        return p;
+    }
+    if (dfs.empty()) throw std::runtime_error("no default search fields");
+    PlanNode b;
+    b.kind = PlanNode::BOOL;
+    for (auto& df : dfs) b.should.push_back(full_text_plan(df, t.text, "or", schema));
+    PlanNode r = dfs.size() == 1 ? std::move(b.should[0]) : std::move(b);
+    r.boost *= t.boost;
+    return r;
+}
+
+inline PlanNode qtok_leaf(const QTok& t, const std::vector<std::string>& dfs,
+                          const Schema& schema) {
+    return qtok_leaf_impl(t, dfs, schema);
+}
+
+inline PlanNode user_input_plan(const std::string& text,
+                                const std::vector<std::string>& default_fields,
+                                const Schema& schema) {
+    // tantivy query grammar (see QTok above); default occur = Should (the
+    // reference's BooleanOperand::Or default, user_input_query.rs:125-130)
+    std::vector<QTok> toks = qlex(text);
+    size_t pos = 0;
+    PlanNode n = qparse_clause(toks, &pos, default_fields, schema);
+    if (pos < toks.size()) throw std::runtime_error("unbalanced parentheses");
+    return n;
 }
 
 inline PlanNode build_plan(const mj::Value* ast, const Schema& schema) {
@@ -472,6 +773,13 @@ inline int64_t parse_datetime_ms(const mj::Value* lit) {
         if (endp && *endp == 0) {
             long long a = v < 0 ? -v : v;
             return a < 1000000000000LL ? v * 1000 : v;  // s vs ms, like splitgen
+        }
+        if (lit->s.find('/') != std::string::npos) {
+            // "YYYY/MM/DD" (quickwit-datetime lenient date format)
+            std::string iso = lit->s;
+            for (char& c : iso)
+                if (c == '/') c = '-';
+            return rfc3339_to_ms(iso);
         }
         return rfc3339_to_ms(lit->s);
     }

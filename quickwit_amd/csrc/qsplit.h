@@ -97,6 +97,19 @@ struct FastFieldView {
         return std::string((const char*)h_dict_bytes + h_dict_offsets[o],
                            h_dict_offsets[o + 1] - h_dict_offsets[o]);
     }
+    // first ord whose dict entry compares >= s (after=false) or > s
+    // (after=true); == cardinality when none
+    uint32_t str_bound_ord(const std::string& s, bool after) const {
+        uint32_t lo = 0, hi = cardinality;
+        while (lo < hi) {
+            uint32_t mid = (lo + hi) / 2;
+            std::string e = dict_entry(mid);
+            bool go_right = after ? (e <= s) : (e < s);
+            if (go_right) lo = mid + 1;
+            else hi = mid;
+        }
+        return lo;
+    }
 };
 
 // optional row store (fetch_docs phase 2): zlib blocks + block index

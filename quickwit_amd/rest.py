@@ -228,9 +228,48 @@ def create_app(searcher_factory):
         body = (await request.body()).decode()
         docs = [json.loads(line) for line in body.splitlines() if line.strip()]
         idx = indexes[iid]
+        # docs missing the configured timestamp field are rejected at ingest
+        # (the reference's doc processor requires it)
+        ts_field = idx.config.get("doc_mapping", {}).get("timestamp_field")
+        if ts_field:
+            docs = [d for d in docs if d.get(ts_field) is not None]
         idx.batches.append(docs)
         idx.rebuild()  # commit=force semantics: searchable immediately
         return {"num_docs_for_processing": len(docs)}
+
+    @app.api_route("/api/v1/{iid}/search", methods=["GET", "POST"])
+    async def qw_search(iid: str, request: Request, response: Response):
+        """Quickwit-native search API subset: ?query=...&start_timestamp=..
+        (query strings go through the engine's tantivy-grammar parser)."""
+        if iid not in indexes:
+            response.status_code = 404
+            return {"message": "index not found"}
+        idx = indexes[iid]
+        params = dict(request.query_params)
+        raw = await request.body()
+        if raw:
+            params.update(json.loads(raw))
+        t0 = time.perf_counter()
+        ast = {"type": "user_input", "user_text": params.get("query", "*"),
+               "default_fields": None}
+        req = make_leaf_request(
+            ast, idx.schema, idx.splits,
+            max_hits=int(params.get("max_hits", 20)),
+            start_timestamp=(int(params["start_timestamp"])
+                             if "start_timestamp" in params else None),
+            end_timestamp=(int(params["end_timestamp"])
+                           if "end_timestamp" in params else None))
+        resp = idx.searcher.leaf_search(req)
+        if resp.get("failed_splits"):
+            response.status_code = 400
+            return {"message": resp["failed_splits"][0].get("error", "")}
+        hits = []
+        for h in resp.get("partial_hits", []):
+            sid = h.get("split_id", "")
+            hits.append(idx.source_doc(sid, h.get("doc_id", 0)))
+        return {"num_hits": resp.get("num_hits", 0), "hits": hits,
+                "elapsed_time_micros": int((time.perf_counter() - t0) * 1e6),
+                "errors": []}
 
     @app.api_route("/api/v1/_elastic/{iid}/_search", methods=["GET", "POST"])
     async def es_search(iid: str, request: Request, response: Response):

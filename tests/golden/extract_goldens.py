@@ -132,6 +132,9 @@ def rest_scenarios():
         "search_after": ["_setup.quickwit.yaml",
                          "0001-search_after_edge_case.yaml",
                          "_teardown.quickwit.yaml"],
+        "qw_search_api": ["_setup.quickwit.yaml", "0001_ts_range.yaml",
+                          "0002_negative_search.yaml",
+                          "_teardown.quickwit.yaml"],
     }
     keep = {"method", "endpoint", "params", "json", "ndjson", "expected",
             "status_code"}

@@ -139,3 +139,25 @@ def test_rest_search_after_suite_gpu():
     import __graft_entry__
     __graft_entry__.build()
     run_search_after(lambda: GpuSearcher(device=0))
+
+
+def run_qw_search_api(searcher_factory):
+    steps = load_suite("qw_search_api")
+    client = make_client(searcher_factory)
+    ran, skipped = replay_suite(client, steps)
+    assert not skipped and ran == len(steps)
+
+
+def test_rest_qw_search_api_suite_oracle():
+    from quickwit_amd.api import OracleSearcher
+    import __graft_entry__
+    __graft_entry__.build()
+    run_qw_search_api(OracleSearcher)
+
+
+@pytest.mark.gpu
+def test_rest_qw_search_api_suite_gpu():
+    from quickwit_amd.api import GpuSearcher
+    import __graft_entry__
+    __graft_entry__.build()
+    run_qw_search_api(lambda: GpuSearcher(device=0))
