@@ -322,13 +322,20 @@ struct SplitSearcher {
             return true;
         }
         if (f.type == FastFieldView::STR) {
-            // lexicographic bounds over the sorted ord dictionary
+            // lexicographic bounds; lowercase-normalized columns fold bounds
+            auto fold = [&](const std::string& s) {
+                if (!f.lower_norm) return s;
+                std::string o = s;
+                for (char& c : o) c = char(std::tolower((unsigned char)c));
+                return o;
+            };
             std::string v = f.dict_entry(f.ord(d));
+            std::string ls = fold(lo.sval), hs = fold(hi.sval);
             if (lo.kind != Bound::UNBOUNDED &&
-                (lo.kind == Bound::INCLUDED ? v < lo.sval : v <= lo.sval))
+                (lo.kind == Bound::INCLUDED ? v < ls : v <= ls))
                 return false;
             if (hi.kind != Bound::UNBOUNDED &&
-                (hi.kind == Bound::INCLUDED ? v > hi.sval : v >= hi.sval))
+                (hi.kind == Bound::INCLUDED ? v > hs : v >= hs))
                 return false;
             return true;
         }
@@ -365,6 +372,12 @@ struct SplitSearcher {
     }
 
     Match eval_bool(const PlanNode& n) const {
+        if (n.must.empty() && n.filter.empty() && n.should.empty() &&
+            n.must_not.empty() && n.minimum_should_match <= 0) {
+            Match m;  // empty bool = match_all (tantivy_query_ast.rs:193)
+            m.all = true;
+            return m;
+        }
         std::vector<Match> req;
         for (auto& c : n.must) req.push_back(eval(c));
         std::vector<Match> filt;

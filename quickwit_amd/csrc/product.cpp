@@ -300,6 +300,12 @@ struct FlatQuery {
     }
 };
 
+static std::string fold_lower(const std::string& s) {
+    std::string o = s;
+    for (char& c : o) c = char(std::tolower((unsigned char)c));
+    return o;
+}
+
 static float bm25_weight(const SplitView& sv, const TextFieldView& f, int64_t tid,
                          float boost) {
     double N = double(sv.num_docs);
@@ -338,18 +344,21 @@ static void add_range_pred(FlatQuery& fq, const SplitView& sv, const PlanNode& n
             memcpy(&p.lo, &n.lo.fval, 8);  // bit-cast double bounds
             memcpy(&p.hi, &n.hi.fval, 8);
         } else if (f->type == FastFieldView::STR) {
-            // lexicographic bounds -> ord space [lo, hi) over the sorted dict
+            // lexicographic bounds -> ord space [lo, hi) over the sorted
+            // dict; a lowercase-normalized column folds the bounds too
+            std::string lo_s = f->lower_norm ? fold_lower(n.lo.sval) : n.lo.sval;
+            std::string hi_s = f->lower_norm ? fold_lower(n.hi.sval) : n.hi.sval;
             p.type = PRED_RANGE_U64;
             p.flags &= PRED_NEGATED;  // rebuild the bound flags below
             if (n.lo.kind != Bound::UNBOUNDED) {
                 p.flags |= PRED_LO_INCLUDED;
                 p.lo = int64_t(
-                    f->str_bound_ord(n.lo.sval, n.lo.kind == Bound::EXCLUDED));
+                    f->str_bound_ord(lo_s, n.lo.kind == Bound::EXCLUDED));
             }
             if (n.hi.kind != Bound::UNBOUNDED) {
                 p.flags |= PRED_HI_EXCLUDED;
                 p.hi = int64_t(
-                    f->str_bound_ord(n.hi.sval, n.hi.kind == Bound::INCLUDED));
+                    f->str_bound_ord(hi_s, n.hi.kind == Bound::INCLUDED));
             }
         } else {
             p.lo = n.lo.ival;
@@ -486,6 +495,12 @@ static void add_positive(FlatQuery& fq, const SplitView& sv, const PlanNode& c,
 static void flatten_bool(FlatQuery& fq, const SplitView& sv, const PlanNode& n,
                          float boost) {
     boost *= n.boost;
+    if (n.must.empty() && n.filter.empty() && n.should.empty() &&
+        n.must_not.empty() && n.minimum_should_match <= 0) {
+        // empty bool is match_all by ES convention (tantivy_query_ast.rs:193)
+        fq.match_all = true;
+        return;
+    }
     bool has_req = !n.must.empty() || !n.filter.empty();
     for (auto& c : n.must) add_positive(fq, sv, c, fq.scoring, boost);
     for (auto& c : n.filter) add_positive(fq, sv, c, false, boost);

@@ -208,7 +208,9 @@ inline Bound parse_bound(const mj::Value* v, const SchemaField& f) {
         b.kind = Bound::EXCLUDED;
         lit = exc;
     } else throw std::runtime_error("bad range bound");
-    if (f.type == "str") {
+    if (f.type == "str" || f.type == "text") {
+        // str fast columns (incl. the raw fast column of a text+fast
+        // field): lexicographic bound over the sorted ord dictionary
         if (lit->kind != mj::Value::STR)
             throw std::runtime_error("str range bound must be a string");
         b.sval = lit->s;
@@ -233,14 +235,16 @@ inline Bound parse_bound(const mj::Value* v, const SchemaField& f) {
 inline PlanNode build_plan(const mj::Value* ast, const Schema& schema);
 
 inline PlanNode full_text_plan(const std::string& field, const std::string& text,
-                               const std::string& op, const Schema& schema) {
+                               const std::string& op, const Schema& schema,
+                               bool zero_terms_all = false) {
     const SchemaField* f = schema.field(field);
     if (!f || f->type != "text")
         throw std::runtime_error("full_text on unknown/non-text field: " + field);
     std::vector<std::string> toks = tokenize(text, f->tokenizer);
     if (toks.empty()) {
         PlanNode n;
-        n.kind = PlanNode::MATCH_NONE;  // zero_terms_query default MatchNone
+        // zero_terms_query: none (default) | all (full_text_query.rs params)
+        n.kind = zero_terms_all ? PlanNode::MATCH_ALL : PlanNode::MATCH_NONE;
         return n;
     }
     if (toks.size() == 1) {
@@ -419,7 +423,7 @@ inline Bound qbound(const std::string& text, Bound::Kind kind, const SchemaField
         v.kind = mj::Value::STR;
         v.s = text;
         b.ival = parse_datetime_ms(&v);
-    } else if (f.type == "str") {
+    } else if (f.type == "str" || f.type == "text") {
         b.sval = text;
     } else if (f.type == "f64") {
         b.fval = atof(text.c_str());
@@ -618,8 +622,11 @@ inline PlanNode build_plan(const mj::Value* ast, const Schema& schema) {
             throw std::runtime_error("term on unknown field: " + n.field);
     } else if (ty == "full_text") {
         std::string op = "or";
+        bool zta = false;
         const mj::Value* params = ast->get("params");
         if (params) {
+            const mj::Value* z = params->get("zero_terms_query");
+            zta = z && z->s == "all";
             const mj::Value* mode = params->get("mode");
             if (mode) {
                 // FullTextMode serde: {"type":"bool","operator":"or"|"and"} or
@@ -632,7 +639,8 @@ inline PlanNode build_plan(const mj::Value* ast, const Schema& schema) {
                 std::transform(op.begin(), op.end(), op.begin(), ::tolower);
             }
         }
-        return full_text_plan(ast->at("field")->s, ast->at("text")->s, op, schema);
+        return full_text_plan(ast->at("field")->s, ast->at("text")->s, op, schema,
+                              zta);
     } else if (ty == "bool") {
         n.kind = PlanNode::BOOL;
         auto fill = [&](const char* key, std::vector<PlanNode>& out) {

@@ -70,6 +70,8 @@ struct FastFieldView {
     std::string name;
     enum Type { U64, I64, DATETIME, STR, F64 } type = U64;
     bool nullable = false;
+    bool lower_norm = false;   // str column written through a lowercase
+                               // normalizer: query bounds fold too
     uint32_t cardinality = 0;  // str
     int ord_width = 0;         // str: 1/2/4
     int64_t min_value = 0, max_value = 0;
@@ -222,6 +224,8 @@ struct SplitView {
                 ff.values = sec(s, "values");
                 if (ff.nullable) ff.nulls = sec(s, "nulls");
                 if (ff.type == FastFieldView::STR) {
+                    const mj::Value* nrm = f->get("normalizer");
+                    ff.lower_norm = nrm && nrm->s == "lowercase";
                     ff.cardinality = uint32_t(f->at("cardinality")->as_i64());
                     ff.ord_width = int(f->at("ord_width")->as_i64());
                     ff.dict_offsets = sec(s, "dict_offsets");

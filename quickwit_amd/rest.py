@@ -57,6 +57,9 @@ def es_query_to_ast(q, schema=None):
         pt = point_or_term(field, value)
         if pt:
             return pt
+        if isinstance(body, dict) and body.get("case_insensitive"):
+            # tokenized (lowercasing) index: fold the needle too
+            value = str(value).lower()
         return {"type": "term", "field": field, "value": str(value)}
     if "terms" in q:
         [(field, values)] = [(k, v) for k, v in q["terms"].items()
@@ -64,7 +67,17 @@ def es_query_to_ast(q, schema=None):
         return {"type": "term_set",
                 "terms_per_field": {field: sorted(str(v) for v in values)}}
     if "exists" in q:
-        return {"type": "field_presence", "field": q["exists"]["field"]}
+        body = q["exists"]
+        if not isinstance(body, dict):
+            raise ValueError("exists query takes an object with a `field` key")
+        field = body["field"]
+        sub = [f["name"] for f in (schema or {}).get("fields", [])
+               if f["name"] == field or f["name"].startswith(field + ".")]
+        if field not in sub and sub:
+            # object field: present iff any dotted subfield is present
+            return {"type": "bool", "should": [
+                {"type": "field_presence", "field": s} for s in sub]}
+        return {"type": "field_presence", "field": field}
     if "range" in q:
         [(field, body)] = q["range"].items()
         node = {"type": "range", "field": field}
@@ -84,23 +97,74 @@ def es_query_to_ast(q, schema=None):
         if pt:
             return pt
         op = (body.get("operator", "or") if isinstance(body, dict) else "or")
+        params = {"mode": {"type": "bool", "operator": op.lower()}}
+        if isinstance(body, dict) and "zero_terms_query" in body:
+            params["zero_terms_query"] = body["zero_terms_query"]
         return {"type": "full_text", "field": field, "text": str(text),
-                "params": {"mode": {"type": "bool", "operator": op.lower()}}}
+                "params": params}
+    if "wildcard" in q:
+        [(field, body)] = q["wildcard"].items()
+        value = body["value"] if isinstance(body, dict) else body
+        node = {"type": "wildcard", "field": field, "value": str(value)}
+        if isinstance(body, dict) and body.get("case_insensitive"):
+            node["case_insensitive"] = True
+        return node
     if "bool" in q:
         out = {"type": "bool"}
         for clause in ("must", "must_not", "should", "filter"):
-            if clause in q["bool"]:
-                items = q["bool"][clause]
-                if isinstance(items, dict):
-                    items = [items]
-                out[clause] = [es_query_to_ast(i, schema) for i in items]
-        if "minimum_should_match" in q["bool"]:
-            out["minimum_should_match"] = int(q["bool"]["minimum_should_match"])
+            items = q["bool"].get(clause)
+            if items is None:
+                continue
+            if isinstance(items, dict):
+                items = [items]
+            out[clause] = [es_query_to_ast(i, schema) for i in items]
+        msm = q["bool"].get("minimum_should_match")
+        if msm is not None:
+            n = len(out.get("should", []))
+            if isinstance(msm, str) and msm.endswith("%"):
+                pct = int(msm[:-1])
+                msm = (n * pct) // 100 if pct >= 0 else n + (n * pct) // 100
+            msm = int(msm)
+            if msm < 0:
+                msm = n + msm  # ES negative form: total minus |msm|
+            out["minimum_should_match"] = max(msm, 0)
         return out
     if "query_string" in q:
         return {"type": "user_input", "user_text": q["query_string"]["query"],
                 "default_fields": q["query_string"].get("fields")}
     raise ValueError(f"unsupported es query: {list(q)}")
+
+
+def flatten_doc(d, prefix=""):
+    """Nested objects -> dotted paths (dynamic_mapping expand_dots); arrays
+    of objects contribute list-valued dotted leaves (multi-valued text)."""
+    out = {}
+
+    def add(key, val):
+        if key in out:
+            prev = out[key]
+            if not isinstance(prev, list):
+                prev = [prev]
+                out[key] = prev
+            if isinstance(val, list):
+                prev.extend(val)
+            else:
+                prev.append(val)
+        else:
+            out[key] = val
+
+    for k, v in d.items():
+        key = prefix + k
+        if isinstance(v, dict):
+            for k2, v2 in flatten_doc(v, key + ".").items():
+                add(k2, v2)
+        elif isinstance(v, list) and v and all(isinstance(x, dict) for x in v):
+            for elem in v:
+                for k2, v2 in flatten_doc(elem, key + ".").items():
+                    add(k2, v2)
+        else:
+            add(key, v)
+    return out
 
 
 def _infer_dynamic_fields(explicit_names, doc_batches):
@@ -121,8 +185,10 @@ def _infer_dynamic_fields(explicit_names, doc_batches):
                     kinds.add("float")
                 elif isinstance(v, str):
                     kinds.add("str")
+                elif isinstance(v, list) and all(isinstance(x, str) for x in v):
+                    kinds.add("strlist")  # text-indexable, no fast column
                 else:
-                    kinds.add("skip")  # arrays/objects: multi-valued, later
+                    kinds.add("skip")  # mixed arrays/objects: later round
     fields = []
     for name in sorted(seen):
         kinds = seen[name]
@@ -135,7 +201,13 @@ def _infer_dynamic_fields(explicit_names, doc_batches):
         elif kinds <= {"int", "negint", "float"}:
             fields.append({"name": name, "type": "f64", "fast": True})
         elif kinds == {"str"}:
-            fields.append({"name": name, "type": "str", "fast": True})
+            # quickwit dynamic strings: tokenized text index + raw str fast
+            # column under the same name (dynamic_mapping tokenizer+fast)
+            fields.append({"name": name, "type": "text",
+                           "tokenizer": "default", "fast": True})
+        elif "strlist" in kinds and kinds <= {"str", "strlist"}:
+            fields.append({"name": name, "type": "text",
+                           "tokenizer": "default"})
     return fields
 
 
@@ -153,7 +225,23 @@ class Index:
         fields = []
         ts_field = None
         dm = self.config.get("doc_mapping", {})
-        for fm in dm.get("field_mappings", []):
+        fast_norm = None
+        dyn = dm.get("dynamic_mapping", {})
+        if isinstance(dyn.get("fast"), dict):
+            fast_norm = dyn["fast"].get("normalizer")
+        self.fast_normalizer = fast_norm
+        flat_mappings = []
+
+        def walk(fms, prefix=""):
+            for fm in fms:
+                if fm.get("type") == "object":
+                    walk(fm.get("field_mappings", []), prefix + fm["name"] + ".")
+                else:
+                    fm = dict(fm)
+                    fm["name"] = prefix + fm["name"]
+                    flat_mappings.append(fm)
+        walk(dm.get("field_mappings", []))
+        for fm in flat_mappings:
             t = fm["type"]
             name = fm["name"]
             if t == "datetime":
@@ -169,21 +257,29 @@ class Index:
                      "tokenizer": fm.get("tokenizer", "default")}
                 if fm.get("record") == "freq":
                     f["record"] = "freq"
+                if fm.get("fast"):
+                    f["fast"] = True
                 fields.append(f)
         return fields, ts_field
 
     def rebuild(self):
         fields, ts_field = self._explicit_fields()
         explicit = {f["name"] for f in fields}
+        flat_batches = [[flatten_doc(d) for d in b] for b in self.batches]
         if self.config.get("doc_mapping", {}).get("mode") == "dynamic":
-            fields = fields + _infer_dynamic_fields(explicit, self.batches)
+            inferred = _infer_dynamic_fields(explicit, flat_batches)
+            if self.fast_normalizer == "lowercase":
+                for f in inferred:
+                    if f.get("fast"):
+                        f["fast_normalizer"] = "lowercase"
+            fields = fields + inferred
         self.schema = {"timestamp_field": ts_field, "fields": fields,
                        "default_search_fields":
                            self.config.get("search_settings", {})
                                .get("default_search_fields", [])}
         self.searcher = self.searcher_factory()
         self.splits = []
-        batches = self.batches or [[]]  # empty index: one 0-doc split so
+        batches = flat_batches or [[]]  # empty index: one 0-doc split so
         for i, docs in enumerate(batches):  # aggs return shaped empties
             sid = f"{self.index_id}-{i:04d}"
             w = splitgen.SplitWriter(self.schema, sid)
@@ -271,6 +367,28 @@ def create_app(searcher_factory):
                 "elapsed_time_micros": int((time.perf_counter() - t0) * 1e6),
                 "errors": []}
 
+    @app.api_route("/api/v1/{iid}/_count", methods=["GET", "POST"])
+    @app.api_route("/api/v1/_elastic/{iid}/_count", methods=["GET", "POST"])
+    async def es_count(iid: str, request: Request, response: Response):
+        if iid not in indexes:
+            response.status_code = 404
+            return {"message": "index not found"}
+        idx = indexes[iid]
+        body = {}
+        raw = await request.body()
+        if raw:
+            body = json.loads(raw)
+        if "q" in request.query_params:
+            ast = {"type": "user_input",
+                   "user_text": request.query_params["q"],
+                   "default_fields": None}
+        else:
+            ast = es_query_to_ast(body.get("query"), idx.schema)
+        req = make_leaf_request(ast, idx.schema, idx.splits, max_hits=0)
+        resp = idx.searcher.leaf_search(req)
+        return {"count": resp.get("num_hits", 0)}
+
+    @app.api_route("/api/v1/{iid}/_search", methods=["GET", "POST"])
     @app.api_route("/api/v1/_elastic/{iid}/_search", methods=["GET", "POST"])
     async def es_search(iid: str, request: Request, response: Response):
         if iid not in indexes:
@@ -282,9 +400,13 @@ def create_app(searcher_factory):
         if raw:
             body = json.loads(raw)
         t0 = time.perf_counter()
-        ast = es_query_to_ast(body.get("query"), idx.schema)
+        try:
+            ast = es_query_to_ast(body.get("query"), idx.schema)
+        except ValueError as e:
+            response.status_code = 400
+            return {"message": str(e)}
         aggs = body.get("aggs") or body.get("aggregations")
-        size = int(body.get("size", 10))
+        size = int(request.query_params.get("size", body.get("size", 10)))
         sort_fields = []
         for s in body.get("sort", []):
             if isinstance(s, str):

@@ -288,7 +288,14 @@ def parse_datetime_ms(v) -> int:
     """Accept RFC3339 strings or unix seconds/ms ints (hdfs-logs uses
     unix_timestamp input; the agg golden corpus uses rfc3339)."""
     if isinstance(v, str):
-        d = _dt.datetime.fromisoformat(v.replace("Z", "+00:00"))
+        s = v.replace("Z", "+00:00")
+        # fromisoformat (3.10) only takes 3- or 6-digit fractions: normalize
+        import re
+        m = re.match(r"^(.*?\.)(\d+)(.*)$", s)
+        if m and len(m.group(2)) not in (3, 6):
+            frac = (m.group(2) + "000000")[:6]
+            s = m.group(1) + frac + m.group(3)
+        d = _dt.datetime.fromisoformat(s)
         return int(d.timestamp() * 1000)
     v = int(v)
     # unix seconds unless the magnitude says millis (quickwit's
@@ -317,8 +324,20 @@ class SplitWriter:
             name = fspec["name"]
             if fspec["type"] == "text":
                 tok = fspec.get("tokenizer", "default")
-                per_doc = [tokenize(d.get(name), tok) if name in d else [] for d in self.docs]
+                per_doc = [tokenize_value(d.get(name), tok)
+                           if d.get(name) is not None else [] for d in self.docs]
                 text_inputs[name] = per_doc
+                if fspec.get("fast", False):
+                    lower = fspec.get("fast_normalizer") == "lowercase"
+                    vals, pres = [], []
+                    for d in self.docs:
+                        v = d.get(name)
+                        # raw string value; arrays are indexed (text) but
+                        # have no fast value (multi-valued: later round)
+                        ok = isinstance(v, str)
+                        pres.append(ok)
+                        vals.append((v.lower() if lower else v) if ok else "")
+                    fast_inputs[name] = (vals, np.array(pres, dtype=bool))
             elif fspec.get("fast", False):
                 vals, pres = [], []
                 for d in self.docs:
@@ -411,6 +430,17 @@ def _assemble(schema, split_id, num_docs, text_inputs, fast_inputs, precomputed_
             fields_meta.append(
                 _build_text_field(sec, fspec, num_docs, term_ids, doc_ids, tfs, vocab, lengths)
             )
+            if fspec.get("fast", False):
+                # text field with fast=true (quickwit dynamic mapping): BOTH
+                # an inverted index (tokenized) and a raw-string str fast
+                # column under the same name — two meta entries, found by
+                # text_field()/fast_field() respectively
+                vals, pres = fast_inputs[name]
+                fmeta = _build_fast_field(
+                    sec, {"name": name, "type": "str"}, num_docs, vals, pres)
+                if fspec.get("fast_normalizer") == "lowercase":
+                    fmeta["normalizer"] = "lowercase"
+                fields_meta.append(fmeta)
         elif fspec.get("fast", False):
             vals, pres = fast_inputs[name]
             fields_meta.append(_build_fast_field(sec, fspec, num_docs, vals, pres))
@@ -489,6 +519,16 @@ def body_vocab():
 def zipf_probs(n, s=1.0):
     p = 1.0 / np.arange(1, n + 1) ** s
     return p / p.sum()
+
+
+def tokenize_value(v, tok):
+    if isinstance(v, list):
+        out = []
+        for item in v:
+            if isinstance(item, str):
+                out.extend(tokenize(item, tok))
+        return out
+    return tokenize(v, tok)
 
 
 def generate_split(split_ord: int, num_docs: int, seed: int = 42) -> bytes:

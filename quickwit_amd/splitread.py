@@ -18,10 +18,23 @@ class Split:
         meta_off, meta_len = np.frombuffer(data[-24:-8], dtype="<u8")
         self.meta = json.loads(data[int(meta_off) : int(meta_off + meta_len)])
         self.data = data
-        self.fields = {f["name"]: f for f in self.meta["fields"]}
+        # text+fast fields appear as TWO meta entries under one name
+        # (inverted index + str fast column); keep them separately
+        self.fields = {}
+        self.fast_meta = {}
+        for f in self.meta["fields"]:
+            if f["type"] == "text":
+                self.fields[f["name"]] = f
+            else:
+                self.fast_meta[f["name"]] = f
+                self.fields.setdefault(f["name"], f)
 
     def _sec(self, field, name, dtype):
-        off, ln = self.fields[field]["sec"][name]
+        meta = self.fields[field]
+        if name in ("values", "nulls", "dict_offsets", "dict_bytes") and \
+                field in self.fast_meta:
+            meta = self.fast_meta[field]
+        off, ln = meta["sec"][name]
         return np.frombuffer(self.data, dtype=dtype, count=ln // np.dtype(dtype).itemsize,
                              offset=off)
 
@@ -75,7 +88,7 @@ class Split:
 
     # ---- fast fields
     def fast_column(self, field):
-        f = self.fields[field]
+        f = self.fast_meta.get(field, self.fields[field])
         if f["type"] == "str":
             w = f["ord_width"]
             vals = self._sec(field, "values", {1: "u1", 2: "<u2", 4: "<u4"}[w])

@@ -52,7 +52,13 @@ def run_step(client, step):
     kwargs = {}
     if "params" in step:
         kwargs["params"] = step["params"]
-    if "ndjson" in step:
+    if "ndjson_file" in step:
+        import gzip
+        import os
+        path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                            "golden", step["ndjson_file"])
+        kwargs["content"] = gzip.open(path).read()
+    elif "ndjson" in step:
         kwargs["content"] = "\n".join(
             json.dumps(d) for d in step["ndjson"]) + "\n"
     elif "json" in step:

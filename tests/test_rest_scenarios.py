@@ -46,6 +46,11 @@ def skip_step(i, step):
         field = s if isinstance(s, str) else next(iter(s))
         if field in MIXED_TYPE_SORT_FIELDS:
             return f"mixed-type dynamic column {field}: later round"
+    q = body.get("query")
+    if isinstance(q, dict) and "range" in q:
+        [(rf, rb)] = q["range"].items()
+        if isinstance(rb, dict) and "format" in rb:
+            return "custom date format in range: later round"
     for kind, field in agg_kinds(body.get("aggs")):
         reason = UNSUPPORTED_AGGS.get(kind, f"unknown agg kind {kind}")
         if reason:
@@ -139,6 +144,30 @@ def test_rest_search_after_suite_gpu():
     import __graft_entry__
     __graft_entry__.build()
     run_search_after(lambda: GpuSearcher(device=0))
+
+
+def run_es_compatibility(searcher_factory):
+    steps = load_suite("es_compatibility")
+    client = make_client(searcher_factory)
+    ran, skipped = replay_suite(client, steps, skip_step)
+    assert ran >= 90, (ran, skipped)
+    for _, reason in skipped:
+        assert "later round" in reason, reason
+
+
+def test_rest_es_compatibility_suite_oracle():
+    from quickwit_amd.api import OracleSearcher
+    import __graft_entry__
+    __graft_entry__.build()
+    run_es_compatibility(OracleSearcher)
+
+
+@pytest.mark.gpu
+def test_rest_es_compatibility_suite_gpu():
+    from quickwit_amd.api import GpuSearcher
+    import __graft_entry__
+    __graft_entry__.build()
+    run_es_compatibility(lambda: GpuSearcher(device=0))
 
 
 def run_qw_search_api(searcher_factory):
