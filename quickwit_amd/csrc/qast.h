@@ -584,8 +584,19 @@ inline PlanNode qtok_leaf_impl(const QTok& t, const std::vector<std::string>& df
     if (dfs.empty()) throw std::runtime_error("no default search fields");
     PlanNode b;
     b.kind = PlanNode::BOOL;
-    for (auto& df : dfs) b.should.push_back(full_text_plan(df, t.text, "or", schema));
-    PlanNode r = dfs.size() == 1 ? std::move(b.should[0]) : std::move(b);
+    for (auto& df : dfs) {
+        // lenient over the default-field list: configured names may be
+        // dynamic fields absent from a given split's schema
+        const SchemaField* f = schema.field(df);
+        if (!f || f->type != "text") continue;
+        b.should.push_back(full_text_plan(df, t.text, "or", schema));
+    }
+    if (b.should.empty()) {
+        PlanNode none;
+        none.kind = PlanNode::MATCH_NONE;
+        return none;
+    }
+    PlanNode r = b.should.size() == 1 ? std::move(b.should[0]) : std::move(b);
     r.boost *= t.boost;
     return r;
 }

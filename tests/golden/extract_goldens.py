@@ -135,6 +135,12 @@ def rest_scenarios():
         "qw_search_api": ["_setup.quickwit.yaml", "0001_ts_range.yaml",
                           "0002_negative_search.yaml",
                           "_teardown.quickwit.yaml"],
+        "default_search_fields": ["_setup.quickwit.yaml",
+                                  "0001_default_fields.yaml",
+                                  "_teardown.quickwit.yaml"],
+        "multi_splits": ["_setup.quickwit.yaml",
+                         "0001-request-optimizations.yaml",
+                         "_teardown.quickwit.yaml"],
         # the in-scope slice of the big ES-compat suite (gharchive corpus)
         "es_compatibility": ["_setup.quickwit.yaml", "0001-noquery.yaml",
                              "0003-match.yaml", "0006-term_query.yaml",

@@ -190,3 +190,27 @@ def test_rest_qw_search_api_suite_gpu():
     import __graft_entry__
     __graft_entry__.build()
     run_qw_search_api(lambda: GpuSearcher(device=0))
+
+
+def run_simple_suite(name, searcher_factory):
+    steps = load_suite(name)
+    client = make_client(searcher_factory)
+    ran, skipped = replay_suite(client, steps)
+    assert not skipped and ran == len(steps)
+
+
+@pytest.mark.parametrize("suite", ["default_search_fields", "multi_splits"])
+def test_rest_simple_suites_oracle(suite):
+    from quickwit_amd.api import OracleSearcher
+    import __graft_entry__
+    __graft_entry__.build()
+    run_simple_suite(suite, OracleSearcher)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("suite", ["default_search_fields", "multi_splits"])
+def test_rest_simple_suites_gpu(suite):
+    from quickwit_amd.api import GpuSearcher
+    import __graft_entry__
+    __graft_entry__.build()
+    run_simple_suite(suite, lambda: GpuSearcher(device=0))

@@ -1,0 +1,19 @@
+import json, sys, os
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), "tests"))
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+from fastapi.testclient import TestClient
+from quickwit_amd.rest import create_app
+from quickwit_amd.api import GpuSearcher
+from rest_replay import run_step
+from test_rest_scenarios import skip_step
+steps = json.load(open('tests/golden/rest_scenarios.json'))['suites']['es_compatibility']
+c = TestClient(create_app(lambda: GpuSearcher(device=0)))
+for i, s in enumerate(steps):
+    if skip_step(i, s):
+        continue
+    try:
+        run_step(c, s)
+    except Exception as e:
+        print('FAIL', i, json.dumps(s.get('json', s.get('params', {})))[:160])
+        print('  ', str(e)[:250])
+print('done')
