@@ -746,6 +746,23 @@ inline uint32_t launch_leaf_tile(bool ns, bool nb, bool na, bool nc, dim3 grid,
     return 0;  // unreachable
 }
 
+// bitmap combine for the recursive boolean evaluator (product.cpp
+// bitmap_eval): dst op= src over u32 words
+extern "C" __global__ void k_bitmap_combine(uint32_t* dst, const uint32_t* src,
+                                            uint64_t words, uint32_t op) {
+    for (uint64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < words;
+         i += uint64_t(gridDim.x) * blockDim.x) {
+        uint32_t d = dst[i], s = src[i];
+        switch (op) {
+            case 0: d &= s; break;   // AND
+            case 1: d |= s; break;   // OR
+            case 2: d &= ~s; break;  // AND NOT
+            default: d = s; break;   // COPY
+        }
+        dst[i] = d;
+    }
+}
+
 // ----------------------------------------------------- top-K selection passes
 // histogram of the top 12 bits (after `shift`) of candidates matching
 // (key >> prefix_shift) == prefix

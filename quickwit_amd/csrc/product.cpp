@@ -550,7 +550,10 @@ static void flatten_bool(FlatQuery& fq, const SplitView& sv, const PlanNode& n,
     }
 
     int64_t msm = n.minimum_should_match;
-    if (msm < 0) msm = has_req ? 0 : 1;
+    // unset OR non-positive: shoulds are optional beside musts, but with no
+    // positive clause at least one should must match (bool_query.rs:34-45 —
+    // the es_compat msm=0/negative cases pin this)
+    if (msm <= 0) msm = has_req ? 0 : 1;
     bool nested_or = false;
     for (auto& c : n.should) nested_or |= c.kind != PlanNode::TERM;
     if (msm > 1 && nested_or)
@@ -841,6 +844,102 @@ static pb::SortByValue sort_value_of(const SortSpec& s, uint32_t doc, float scor
 // shape becomes an unscored flattened boolean whose per-tile match bitset
 // is stored via QueryDev::bitmap_out.
 static uint8_t* resolve_hitset(qw_ctx* ctx, const DeviceSplit& ds,
+                               const PlanNode& inner, const Schema& schema);
+
+// combine helper: dst op= src (op per k_bitmap_combine)
+static void bitmap_combine(qw_ctx* ctx, uint8_t* dst, const uint8_t* src,
+                           size_t bm_bytes, uint32_t op) {
+    uint64_t words = bm_bytes / 4;
+    uint32_t grid = uint32_t(std::min<uint64_t>(2048, (words + 255) / 256));
+    hipLaunchKernelGGL(k_bitmap_combine, dim3(grid), dim3(256), 0, ctx->stream,
+                       (uint32_t*)dst, (const uint32_t*)src, words, op);
+}
+
+// Recursive boolean evaluation over device bitmaps — the fallback for plan
+// shapes the single-pass flattened kernel cannot express (nested booleans
+// under must_not, presence/range in should position, …). Each sub-plan's
+// bitmap is memoized in the (split, fingerprint) hitset cache, so repeated
+// subtrees and repeated queries pay once. Unscored semantics only.
+static uint8_t* bitmap_eval(qw_ctx* ctx, const DeviceSplit& ds,
+                            const PlanNode& node, const Schema& schema) {
+    const SplitView& sv = ds.view;
+    try {
+        return resolve_hitset(ctx, ds, node, schema);  // flattenable fast path
+    } catch (const std::exception&) {
+        if (node.kind != PlanNode::BOOL) throw;
+    }
+    if (node.minimum_should_match > 1)
+        throw std::runtime_error(
+            "minimum_should_match>1 over nested clauses (GPU r1)");
+    uint32_t n_tiles = (sv.num_docs + TILE_DOCS - 1) / TILE_DOCS;
+    size_t bm_bytes = size_t(n_tiles) * (TILE_DOCS / 32) * 4;
+    std::string key = sv.split_id;
+    key += '\0';
+    plan_fingerprint(node, key);
+    uint8_t* bm = nullptr;
+    HIP_CHECK(hipMalloc(&bm, bm_bytes));
+    bool has_req = !node.must.empty() || !node.filter.empty();
+    bool first = true;
+    for (auto* clauses : {&node.must, &node.filter})
+        for (const PlanNode& c : *clauses) {
+            uint8_t* cb = bitmap_eval(ctx, ds, c, schema);
+            bitmap_combine(ctx, bm, cb, bm_bytes, first ? 3u : 0u);
+            first = false;
+        }
+    int64_t msm = node.minimum_should_match;
+    if (msm <= 0) msm = has_req ? 0 : 1;
+    if (!node.should.empty() && msm > 0) {
+        uint8_t* su = nullptr;
+        HIP_CHECK(hipMalloc(&su, bm_bytes));
+        bool sfirst = true;
+        for (const PlanNode& c : node.should) {
+            uint8_t* cb = bitmap_eval(ctx, ds, c, schema);
+            bitmap_combine(ctx, su, cb, bm_bytes, sfirst ? 3u : 1u);
+            sfirst = false;
+        }
+        if (first) {
+            bitmap_combine(ctx, bm, su, bm_bytes, 3u);
+            first = false;
+        } else bitmap_combine(ctx, bm, su, bm_bytes, 0u);
+        HIP_CHECK(hipStreamSynchronize(ctx->stream));
+        (void)hipFree(su);
+    }
+    if (first) {
+        // no positive clause: implicit match_all base (must_not-only)
+        HIP_CHECK(hipMemsetAsync(bm, 0xFF, bm_bytes, ctx->stream));
+        first = false;
+    }
+    for (const PlanNode& c : node.must_not) {
+        uint8_t* cb = bitmap_eval(ctx, ds, c, schema);
+        bitmap_combine(ctx, bm, cb, bm_bytes, 2u);
+    }
+    // mask the tail beyond num_docs (match_all memset covers padding bits)
+    HIP_CHECK(hipStreamSynchronize(ctx->stream));
+    {
+        uint32_t ndocs = sv.num_docs;
+        uint32_t full_words = ndocs / 32;
+        size_t total_words = bm_bytes / 4;
+        if (full_words < total_words) {
+            std::vector<uint32_t> tail(total_words - full_words, 0);
+            if (ndocs % 32)
+                tail[0] = 0;  // recompute partial word below
+            // read-modify host-side for the single partial word
+            if (ndocs % 32) {
+                uint32_t w = 0;
+                HIP_CHECK(hipMemcpy(&w, bm + full_words * 4, 4,
+                                    hipMemcpyDeviceToHost));
+                w &= (1u << (ndocs % 32)) - 1u;
+                tail[0] = w;
+            }
+            HIP_CHECK(hipMemcpy(bm + full_words * 4, tail.data(),
+                                tail.size() * 4, hipMemcpyHostToDevice));
+        }
+    }
+    ctx->hitsets.put(key, bm);
+    return bm;
+}
+
+static uint8_t* resolve_hitset(qw_ctx* ctx, const DeviceSplit& ds,
                                const PlanNode& inner, const Schema& schema) {
     const SplitView& sv = ds.view;
     std::string key = sv.split_id;
@@ -853,7 +952,13 @@ static uint8_t* resolve_hitset(qw_ctx* ctx, const DeviceSplit& ds,
     uint8_t* bm = nullptr;
     HIP_CHECK(hipMalloc(&bm, bm_bytes));
 
-    FlatQuery fq = flatten(sv, inner, false);
+    FlatQuery fq;
+    try {
+        fq = flatten(sv, inner, false);
+    } catch (...) {
+        (void)hipFree(bm);
+        throw;
+    }
     for (const PlanNode* cn : fq.cache_nodes) {  // nested cache nodes
         PredDev p{};
         p.type = PRED_BITSET;
@@ -1027,14 +1132,30 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
     bool wide = (!specs.empty() && specs[0].comp == SortSpec::FAST_FIELD) ||
                 specs.size() > 1;
 
-    FlatQuery fq = flatten(sv, plan, scoring);
+    FlatQuery fq;
+    try {
+        fq = flatten(sv, plan, scoring);
+    } catch (const std::exception& e) {
+        std::string msg = e.what();
+        if (scoring || msg.find("flatten") == std::string::npos) throw;
+        // unscored plan the single-pass kernel cannot express: evaluate the
+        // whole boolean recursively into a device bitmap and search
+        // match_all + PRED_BITSET (results identical; scores don't exist)
+        uint8_t* bmp = bitmap_eval(ctx, ds, plan, schema);
+        fq = FlatQuery{};
+        fq.scoring = false;
+        fq.match_all = true;
+        PredDev p{};
+        p.type = PRED_BITSET;
+        p.abs_bitmap = (uint64_t)bmp;
+        fq.preds.push_back(p);
+    }
     // CACHE nodes in filter position -> device HitSet bitmaps (PRED_BITSET)
     for (const PlanNode* cn : fq.cache_nodes) {
         PredDev p{};
         p.type = PRED_BITSET;
-        p.abs_bitmap =
-            (uint64_t)resolve_hitset(ctx, const_cast<DeviceSplit&>(ds),
-                                     cn->cache_inner.front(), schema);
+        p.abs_bitmap = (uint64_t)bitmap_eval(ctx, const_cast<DeviceSplit&>(ds),
+                                             cn->cache_inner.front(), schema);
         fq.preds.push_back(p);
     }
     uint64_t leaf_max_hits = req.max_hits + req.start_offset;

@@ -1,6 +1,8 @@
 import json, sys, os
-sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)), "tests"))
-sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+os.chdir(REPO)
+sys.path.insert(0, os.path.join(REPO, "tests"))
+sys.path.insert(0, REPO)
 from fastapi.testclient import TestClient
 from quickwit_amd.rest import create_app
 from quickwit_amd.api import GpuSearcher
