@@ -285,7 +285,17 @@ class Index:
             w = splitgen.SplitWriter(self.schema, sid)
             w.add_documents(docs)
             self.searcher.add_split(sid, w.finalize())
-            self.splits.append((sid, len(docs)))
+            entry = {"split_id": sid, "num_docs": len(docs)}
+            if ts_field and docs:
+                # split time-range metadata (seconds): drives the
+                # CanSplitDoBetter pruning exactly like the reference's
+                # SplitIdAndFooterOffsets
+                ts = [splitgen.parse_datetime_ms(d[ts_field]) // 1000
+                      for d in docs if d.get(ts_field) is not None]
+                if ts:
+                    entry["timestamp_start"] = min(ts)
+                    entry["timestamp_end"] = max(ts)
+            self.splits.append(entry)
 
     def source_doc(self, split_id, doc_id):
         i = int(split_id.rsplit("-", 1)[1])
