@@ -329,15 +329,23 @@ struct SplitSearcher {
                 for (char& c : o) c = char(std::tolower((unsigned char)c));
                 return o;
             };
-            std::string v = f.dict_entry(f.ord(d));
             std::string ls = fold(lo.sval), hs = fold(hi.sval);
-            if (lo.kind != Bound::UNBOUNDED &&
-                (lo.kind == Bound::INCLUDED ? v < ls : v <= ls))
+            auto in_range = [&](const std::string& v) {
+                if (lo.kind != Bound::UNBOUNDED &&
+                    (lo.kind == Bound::INCLUDED ? v < ls : v <= ls))
+                    return false;
+                if (hi.kind != Bound::UNBOUNDED &&
+                    (hi.kind == Bound::INCLUDED ? v > hs : v >= hs))
+                    return false;
+                return true;
+            };
+            if (f.multi) {  // any value in range matches
+                uint32_t n = f.n_vals(d);
+                for (uint32_t i = 0; i < n; ++i)
+                    if (in_range(f.dict_entry(f.ord_at(d, i)))) return true;
                 return false;
-            if (hi.kind != Bound::UNBOUNDED &&
-                (hi.kind == Bound::INCLUDED ? v > hs : v >= hs))
-                return false;
-            return true;
+            }
+            return in_range(f.dict_entry(f.ord(d)));
         }
         int64_t v = f.i64(d);
         if (lo.kind != Bound::UNBOUNDED &&
@@ -362,7 +370,7 @@ struct SplitSearcher {
         Match m;
         const FastFieldView* f = sv.fast_field(n.field);
         if (!f) return m;
-        if (!f->nullable) {
+        if (!f->nullable && !f->multi) {
             m.all = true;
             return m;
         }
@@ -535,8 +543,17 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
                 std::vector<uint64_t> counts(f->cardinality, 0);
                 auto visit = [&](uint32_t doc) {
                     if (!f->present(doc)) return;
-                    counts[f->ord(doc)]++;
-                    a.terms_matched_docs++;
+                    if (f->multi) {
+                        uint32_t n = f->n_vals(doc);
+                        for (uint32_t i = 0; i < n; ++i)
+                            counts[f->ord_at(doc, i)]++;  // distinct per doc
+                        // sum_other base counts VALUE instances (equals the
+                        // sum of all bucket doc_counts, like the reference)
+                        a.terms_matched_docs += n;
+                    } else {
+                        counts[f->ord(doc)]++;
+                        a.terms_matched_docs++;
+                    }
                 };
                 if (m.all)
                     for (uint32_t doc = 0; doc < sv.num_docs; ++doc) visit(doc);

@@ -53,6 +53,8 @@ struct PredDev {
     uint32_t value_width;  // 8 for u64/i64; 1/2/4 for str ords
     uint64_t abs_bitmap;   // PRED_BITSET: absolute device VA of u32 words
                            // laid out tile-major (TILE_DOCS/32 words per tile)
+    uint64_t offsets_off;  // multi-valued column: u32[num_docs+1] prefix into
+                           // values (any-value-matches semantics); 0 = single
 };
 
 enum AggKindDev : uint32_t { AGGD_HISTO = 0, AGGD_TERMS = 1 };
@@ -80,6 +82,7 @@ struct AggDev {
     uint32_t value_is_i64;  // 0 = u64, 1 = i64/datetime, 2 = f64
     uint64_t counts_out;    // byte offset into result scratch: u64[n_buckets]
     uint64_t matched_out;   // terms: u64 counter of docs-with-value
+    uint64_t offsets_off;   // terms over a multi-valued column: u32 prefix
     // one optional stats sub-agg set per bucket: {u64 cnt, f64 sum, u64 min_s,
     // u64 max_s (sortable-mapped)} × n_sub, laid out bucket-major
     uint32_t n_sub;

@@ -269,6 +269,28 @@ __device__ __forceinline__ bool eval_pred(const QueryDev& q, const PredDev& p,
         if (p.flags & PRED_HI_EXCLUDED) ok &= v < hi;
         return (p.flags & PRED_NEGATED) ? !ok : ok;
     }
+    if (p.offsets_off) {
+        // multi-valued str column: presence = any value; range = any match
+        const uint32_t* offs = (const uint32_t*)(q.split + p.offsets_off);
+        uint32_t s = offs[doc], e = offs[doc + 1];
+        if (p.type == PRED_PRESENCE) ok = e > s;
+        else {
+            const uint8_t* col = q.split + p.values_off;
+            ok = false;
+            for (uint32_t pos = s; pos < e && !ok; ++pos) {
+                uint64_t v = p.value_width == 1   ? col[pos]
+                             : p.value_width == 2 ? ((const uint16_t*)col)[pos]
+                                                  : ((const uint32_t*)col)[pos];
+                bool m2 = true;
+                if (p.flags & PRED_LO_INCLUDED) m2 &= v >= uint64_t(p.lo);
+                if (p.flags & PRED_LO_EXCLUDED) m2 &= v > uint64_t(p.lo);
+                if (p.flags & PRED_HI_INCLUDED) m2 &= v <= uint64_t(p.hi);
+                if (p.flags & PRED_HI_EXCLUDED) m2 &= v < uint64_t(p.hi);
+                ok |= m2;
+            }
+        }
+        return (p.flags & PRED_NEGATED) ? !ok : ok;
+    }
     if (ok && p.type != PRED_PRESENCE) {
         if (p.type == PRED_RANGE_U64) {
             const uint8_t* col = q.split + p.values_off;
@@ -568,6 +590,41 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                         }
                         if (a.kind == AGGD_TERMS) {
                             if (!a.n_buckets) continue;  // missing/non-str column
+                            if (a.offsets_off) {
+                                // multi-valued: one count per (doc, distinct
+                                // value); doc counted once in matched
+                                const uint32_t* offs =
+                                    (const uint32_t*)(q.split + a.offsets_off);
+                                uint32_t s = offs[d], e2 = offs[d + 1];
+                                if (s == e2) continue;  // no value
+                                const uint8_t* col = q.split + a.values_off;
+                                for (uint32_t pos = s; pos < e2; ++pos) {
+                                    uint64_t o =
+                                        a.value_width == 1 ? col[pos]
+                                        : a.value_width == 2
+                                            ? ((const uint16_t*)col)[pos]
+                                            : ((const uint32_t*)col)[pos];
+                                    if (a.lds_slot == 1)
+                                        atomicAdd(
+                                            &sc_agg_terms[a.lds_rep == 2
+                                                              ? o * 2 +
+                                                                    (lane_id() & 1u)
+                                                              : o],
+                                            1u);
+                                    else
+                                        atomicAdd((unsigned long long*)(q.results +
+                                                                        a.counts_out) +
+                                                      o,
+                                                  1ull);
+                                }
+                                if (ai < 4)
+                                    atomicAdd(&sc_agg_matched[ai], e2 - s);
+                                else
+                                    atomicAdd((unsigned long long*)(q.results +
+                                                                    a.matched_out),
+                                              (unsigned long long)(e2 - s));
+                                continue;
+                            }
                             uint64_t o = agg_ord(q, a.values_off, a.value_width, d);
                             if (a.lds_slot == 1)
                                 atomicAdd(&sc_agg_terms[a.lds_rep == 2

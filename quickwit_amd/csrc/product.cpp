@@ -329,12 +329,13 @@ static void add_range_pred(FlatQuery& fq, const SplitView& sv, const PlanNode& n
                                              : PRED_RANGE_I64;
     p.flags = negated ? PRED_NEGATED : 0;
     if (n.kind == PlanNode::FIELD_PRESENCE) {
-        if (!f->nullable && !negated) return;  // non-nullable: always present
-        if (!f->nullable && negated) {
+        if (!f->nullable && !f->multi && !negated) return;  // always present
+        if (!f->nullable && !f->multi && negated) {
             fq.match_none = true;
             return;
         }
         p.type = PRED_PRESENCE;
+        if (f->multi) p.offsets_off = f->value_offsets.off;
     } else {
         if (n.lo.kind == Bound::INCLUDED) p.flags |= PRED_LO_INCLUDED;
         if (n.lo.kind == Bound::EXCLUDED) p.flags |= PRED_LO_EXCLUDED;
@@ -345,7 +346,9 @@ static void add_range_pred(FlatQuery& fq, const SplitView& sv, const PlanNode& n
             memcpy(&p.hi, &n.hi.fval, 8);
         } else if (f->type == FastFieldView::STR) {
             // lexicographic bounds -> ord space [lo, hi) over the sorted
-            // dict; a lowercase-normalized column folds the bounds too
+            // dict; a lowercase-normalized column folds the bounds too;
+            // multi-valued columns match when ANY value is in range
+            if (f->multi) p.offsets_off = f->value_offsets.off;
             std::string lo_s = f->lower_norm ? fold_lower(n.lo.sval) : n.lo.sval;
             std::string hi_s = f->lower_norm ? fold_lower(n.hi.sval) : n.hi.sval;
             p.type = PRED_RANGE_U64;
@@ -666,6 +669,7 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
                 a.values_off = f->values.off;
                 a.nulls_off = f->nullable ? f->nulls.off : 0;
                 a.value_width = uint32_t(f->ord_width);
+                if (f->multi) a.offsets_off = f->value_offsets.off;
             } else {
                 a.n_buckets = 0;  // no such column: zero buckets
             }
@@ -1356,6 +1360,9 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                 q.sort_src = 1;
             } else if (s0.ff) {
                 const FastFieldView* f = s0.ff;
+                if (f->multi)
+                    throw std::runtime_error(
+                        "sort by a multi-valued fast field not supported");
                 q.sort_values_off = f->values.off;
                 q.sort_nulls_off = f->nullable ? f->nulls.off : 0;
                 if (f->type == FastFieldView::STR) {
@@ -1401,7 +1408,8 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             ap.devs.size() <= 2 &&
             (ap.devs.size() < 2 ||
              (ap.devs[1].kind == AGGD_TERMS && ap.devs[1].lds_slot == 1 &&
-              ap.devs[1].nulls_off == 0 && ap.devs[1].n_buckets > 0)))
+              ap.devs[1].nulls_off == 0 && ap.devs[1].offsets_off == 0 &&
+              ap.devs[1].n_buckets > 0)))
             q.agg_fast = 1;
         HIP_CHECK(hipEventRecord(ctx->ev_start, ctx->stream));
         launch_leaf_tile(ns, nb, na, collect, dim3(grid), ctx->stream, q, 0u,
@@ -1712,7 +1720,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             if (d.kind == AggDef::TERMS) {
                 r.kind = 3;
                 if (a.n_buckets) {
-                    if (a.nulls_off) {
+                    if (a.nulls_off || a.offsets_off) {
                         uint64_t m = 0;
                         memcpy(&m, agg_out.data() + (a.matched_out - r_agg), 8);
                         r.terms_matched_docs = m;

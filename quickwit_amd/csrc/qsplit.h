@@ -70,20 +70,34 @@ struct FastFieldView {
     std::string name;
     enum Type { U64, I64, DATETIME, STR, F64 } type = U64;
     bool nullable = false;
+    bool multi = false;        // multi-valued str: per-doc ord lists
     bool lower_norm = false;   // str column written through a lowercase
                                // normalizer: query bounds fold too
     uint32_t cardinality = 0;  // str
     int ord_width = 0;         // str: 1/2/4
     int64_t min_value = 0, max_value = 0;
     double fmin = 0, fmax = 0;  // F64
-    Section values, nulls, dict_offsets, dict_bytes;
+    Section values, nulls, dict_offsets, dict_bytes, value_offsets;
+    const uint32_t* h_val_offsets = nullptr;  // multi: [num_docs+1] prefix
     const void* h_values = nullptr;
     const uint64_t* h_nulls = nullptr;  // bit d set = doc d has a value
     const uint32_t* h_dict_offsets = nullptr;
     const uint8_t* h_dict_bytes = nullptr;
 
     bool present(uint32_t doc) const {
+        if (multi) return h_val_offsets[doc + 1] > h_val_offsets[doc];
         return !nullable || ((h_nulls[doc >> 6] >> (doc & 63)) & 1);
+    }
+    uint32_t n_vals(uint32_t doc) const {
+        return h_val_offsets[doc + 1] - h_val_offsets[doc];
+    }
+    uint64_t ord_at(uint32_t doc, uint32_t i) const {  // multi
+        uint32_t pos = h_val_offsets[doc] + i;
+        switch (ord_width) {
+            case 1: return ((const uint8_t*)h_values)[pos];
+            case 2: return ((const uint16_t*)h_values)[pos];
+            default: return ((const uint32_t*)h_values)[pos];
+        }
     }
     uint64_t ord(uint32_t doc) const {
         switch (ord_width) {
@@ -226,6 +240,13 @@ struct SplitView {
                 if (ff.type == FastFieldView::STR) {
                     const mj::Value* nrm = f->get("normalizer");
                     ff.lower_norm = nrm && nrm->s == "lowercase";
+                    const mj::Value* mv = f->get("multi");
+                    ff.multi = mv && mv->b;
+                    if (ff.multi) {
+                        ff.value_offsets = sec(s, "value_offsets");
+                        ff.h_val_offsets =
+                            (const uint32_t*)(data + ff.value_offsets.off);
+                    }
                     ff.cardinality = uint32_t(f->at("cardinality")->as_i64());
                     ff.ord_width = int(f->at("ord_width")->as_i64());
                     ff.dict_offsets = sec(s, "dict_offsets");

@@ -206,8 +206,11 @@ def _infer_dynamic_fields(explicit_names, doc_batches):
             fields.append({"name": name, "type": "text",
                            "tokenizer": "default", "fast": True})
         elif "strlist" in kinds and kinds <= {"str", "strlist"}:
+            # string arrays: tokenized text index + multi-valued str fast
+            # column (distinct sorted values per doc)
             fields.append({"name": name, "type": "text",
-                           "tokenizer": "default"})
+                           "tokenizer": "default", "fast": True,
+                           "multi": True})
     return fields
 
 

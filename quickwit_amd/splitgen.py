@@ -219,6 +219,47 @@ def _build_text_field(sec, fspec, num_docs, term_ids, doc_ids, tfs, vocab, doc_l
     return meta
 
 
+def _build_multi_str_fast(sec, fspec, num_docs, values):
+    """Multi-valued str fast column: per-doc SORTED DISTINCT ord lists.
+    values: list aligned to docs; each entry None | str | list[str].
+    Sections: value_offsets u32[num_docs+1] + ords (+ dict)."""
+    per_doc = []
+    allvals = set()
+    for v in values:
+        if v is None:
+            per_doc.append([])
+        elif isinstance(v, str):
+            per_doc.append([v])
+            allvals.add(v)
+        else:
+            vs = sorted({x for x in v if isinstance(x, str)})
+            per_doc.append(vs)
+            allvals.update(vs)
+    strs = sorted(allvals)
+    dct = {s: i for i, s in enumerate(strs)}
+    card = len(strs)
+    width = 1 if card <= 0xFF else (2 if card <= 0xFFFF else 4)
+    dt = {1: np.uint8, 2: np.uint16, 4: np.uint32}[width]
+    offsets = np.zeros(num_docs + 1, dtype=np.uint32)
+    np.cumsum([len(p) for p in per_doc], out=offsets[1:])
+    ords = np.zeros(int(offsets[-1]), dtype=dt)
+    pos = 0
+    for p in per_doc:
+        for s in p:
+            ords[pos] = dct[s]
+            pos += 1
+    dict_bytes = "".join(strs).encode("utf-8")
+    dict_offsets = np.zeros(card + 1, dtype=np.uint32)
+    np.cumsum([len(s.encode("utf-8")) for s in strs], out=dict_offsets[1:])
+    meta = {"name": fspec["name"], "type": "str", "multi": True,
+            "nullable": False, "cardinality": card, "ord_width": width,
+            "sec": {"value_offsets": sec.add(offsets),
+                    "values": sec.add(ords),
+                    "dict_offsets": sec.add(dict_offsets),
+                    "dict_bytes": sec.add(np.frombuffer(dict_bytes, dtype=np.uint8))}}
+    return meta
+
+
 def _build_fast_field(sec, fspec, num_docs, values, present):
     """values: list aligned to docs (None where absent) or np array.
     present: bool array or None (all present)."""
@@ -329,15 +370,26 @@ class SplitWriter:
                 text_inputs[name] = per_doc
                 if fspec.get("fast", False):
                     lower = fspec.get("fast_normalizer") == "lowercase"
-                    vals, pres = [], []
-                    for d in self.docs:
-                        v = d.get(name)
-                        # raw string value; arrays are indexed (text) but
-                        # have no fast value (multi-valued: later round)
-                        ok = isinstance(v, str)
-                        pres.append(ok)
-                        vals.append((v.lower() if lower else v) if ok else "")
-                    fast_inputs[name] = (vals, np.array(pres, dtype=bool))
+                    if fspec.get("multi"):
+                        vals = []
+                        for d in self.docs:
+                            v = d.get(name)
+                            if isinstance(v, str):
+                                v = [v]
+                            if isinstance(v, list):
+                                vals.append([x.lower() if lower else x
+                                             for x in v if isinstance(x, str)])
+                            else:
+                                vals.append(None)
+                        fast_inputs[name] = (vals, None)
+                    else:
+                        vals, pres = [], []
+                        for d in self.docs:
+                            v = d.get(name)
+                            ok = isinstance(v, str)
+                            pres.append(ok)
+                            vals.append((v.lower() if lower else v) if ok else "")
+                        fast_inputs[name] = (vals, np.array(pres, dtype=bool))
             elif fspec.get("fast", False):
                 vals, pres = [], []
                 for d in self.docs:
@@ -436,8 +488,12 @@ def _assemble(schema, split_id, num_docs, text_inputs, fast_inputs, precomputed_
                 # column under the same name — two meta entries, found by
                 # text_field()/fast_field() respectively
                 vals, pres = fast_inputs[name]
-                fmeta = _build_fast_field(
-                    sec, {"name": name, "type": "str"}, num_docs, vals, pres)
+                if fspec.get("multi"):
+                    fmeta = _build_multi_str_fast(
+                        sec, {"name": name}, num_docs, vals)
+                else:
+                    fmeta = _build_fast_field(
+                        sec, {"name": name, "type": "str"}, num_docs, vals, pres)
                 if fspec.get("fast_normalizer") == "lowercase":
                     fmeta["normalizer"] = "lowercase"
                 fields_meta.append(fmeta)

@@ -97,12 +97,6 @@ def agg_golden():
             "request": step["json"],
             "expected": step["expected"]["aggregations"],
         }
-    # terms_full also asserts the multi-valued "tags" field; strip it (multi-
-    # valued fast fields are a later round) but keep hosts + note the removal.
-    tf = cases["terms_full"]
-    tf["request"]["aggs"].pop("tags", None)
-    tf["expected"].pop("tags", None)
-    tf["note"] = "tags (multi-valued) case removed: multi-valued fast fields not in r1 scope"
     return {
         "source": "quickwit/rest-api-tests/scenarii/aggregations/{_setup.quickwit.yaml,0001-aggregations.yaml}",
         "schema": [
@@ -111,6 +105,8 @@ def agg_golden():
             {"name": "name", "type": "str", "fast": True, "nullable": True},
             {"name": "host", "type": "str", "fast": True, "nullable": True},
             {"name": "id", "type": "u64", "fast": True, "nullable": True},
+            {"name": "tags", "type": "text", "tokenizer": "default",
+             "fast": True, "multi": True},
         ],
         "splits": splits,
         "cases": cases,

@@ -328,7 +328,7 @@ def test_agg_golden_scenarios_on_gpu(case_name):
         g = json.load(f)
     type_map = {"str_fast": "str"}
     schema = {"timestamp_field": None, "fields": [
-        {"name": f["name"], "type": type_map.get(f["type"], f["type"]), "fast": True}
+        dict(f, type=type_map.get(f["type"], f["type"]), fast=True)
         for f in g["schema"]]}
     s = GpuSearcher(device=0)
     splits = []

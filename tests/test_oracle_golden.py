@@ -90,7 +90,7 @@ def es_query_to_ast(q):
 def agg_schema(g):
     type_map = {"str_fast": "str"}
     return {"timestamp_field": None, "fields": [
-        {"name": f["name"], "type": type_map.get(f["type"], f["type"]), "fast": True}
+        dict(f, type=type_map.get(f["type"], f["type"]), fast=True)
         for f in g["schema"]]}
 
 

@@ -27,7 +27,7 @@ UNSUPPORTED_AGGS = {"percentiles": "percentiles: later round",
                     "avg": None, "stats": None, "sum": None, "min": None,
                     "max": None, "value_count": None,
                     "date_histogram": None, "histogram": None, "terms": None}
-MULTI_VALUED_FIELDS = {"tags"}  # multi-valued fast columns: later round
+MULTI_VALUED_FIELDS = set()  # multi-valued str fast columns now supported
 NUMERIC_TERMS_FIELDS = {"high_prec_test"}  # terms over numeric fast
                                            # columns: later round
 
