@@ -369,7 +369,14 @@ struct SplitSearcher {
     Match eval_presence(const PlanNode& n) const {
         Match m;
         const FastFieldView* f = sv.fast_field(n.field);
-        if (!f) return m;
+        if (!f) {
+            // text-only field: present iff the fieldnorm byte is nonzero
+            const TextFieldView* tf = sv.text_field(n.field);
+            if (tf && tf->has_norms)
+                for (uint32_t d = 0; d < sv.num_docs; ++d)
+                    if (tf->h_fieldnorms[d]) m.docs.push_back(d);
+            return m;
+        }
         if (!f->nullable && !f->multi) {
             m.all = true;
             return m;

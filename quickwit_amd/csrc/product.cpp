@@ -318,6 +318,20 @@ static void add_range_pred(FlatQuery& fq, const SplitView& sv, const PlanNode& n
                            bool negated) {
     const FastFieldView* f = sv.fast_field(n.field);
     PredDev p{};
+    if (!f && n.kind == PlanNode::FIELD_PRESENCE) {
+        // presence over a text-only field: a doc has the field iff its
+        // fieldnorm byte is nonzero (>= 1 token; index_field_presence analog)
+        const TextFieldView* tf = sv.text_field(n.field);
+        if (tf && tf->has_norms) {
+            p.type = PRED_RANGE_U64;
+            p.flags = (negated ? PRED_NEGATED : 0) | PRED_LO_INCLUDED;
+            p.lo = 1;
+            p.values_off = tf->fieldnorms.off;
+            p.value_width = 1;
+            fq.preds.push_back(p);
+            return;
+        }
+    }
     if (!f) {
         // unknown fast field: matches nothing (oracle eval_range -> empty)
         if (negated) return;  // must_not nothing = no-op

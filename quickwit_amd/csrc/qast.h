@@ -334,12 +334,13 @@ inline std::vector<QTok> qlex(const std::string& s) {
                     t.lo_incl = s[i] == '[';
                     ++i;
                     auto piece = [&]() {
+                        // no quoting/escaping inside range bounds — the
+                        // reference grammar rejects those forms (0004 golden
+                        // cases expect 400)
                         std::string p;
-                        if (i < s.size() && s[i] == '"') read_quoted(&p);
-                        else
-                            while (i < s.size() && s[i] != ' ' && s[i] != ']' &&
-                                   s[i] != '}')
-                                p.push_back(s[i++]);
+                        while (i < s.size() && s[i] != ' ' && s[i] != ']' &&
+                               s[i] != '}')
+                            p.push_back(s[i++]);
                         return p;
                     };
                     t.lo = piece();
@@ -503,7 +504,25 @@ inline PlanNode qtok_leaf_impl(const QTok& t, const std::vector<std::string>& df
         return n;
     }
     if (!t.field.empty() && !t.quoted && !is_range && !t.set && t.text == "*") {
-        n.kind = PlanNode::FIELD_PRESENCE;  // field:* -> Exists
+        // field:* -> Exists; an object/json parent is present iff any dotted
+        // subfield is (index_field_presence semantics over expand_dots)
+        if (!schema.field(t.field)) {
+            PlanNode b;
+            b.kind = PlanNode::BOOL;
+            std::string prefix = t.field + ".";
+            for (auto& f : schema.fields)
+                if (f.name.compare(0, prefix.size(), prefix) == 0) {
+                    PlanNode c;
+                    c.kind = PlanNode::FIELD_PRESENCE;
+                    c.field = f.name;
+                    b.should.push_back(std::move(c));
+                }
+            if (!b.should.empty()) {
+                b.boost = t.boost;
+                return b;
+            }
+        }
+        n.kind = PlanNode::FIELD_PRESENCE;
         n.field = t.field;
         n.boost = t.boost;
         return n;

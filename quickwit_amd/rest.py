@@ -255,6 +255,10 @@ class Index:
             elif t in ("u64", "i64"):
                 fields.append({"name": name, "type": t,
                                "fast": bool(fm.get("fast", False))})
+            elif t == "text" and fm.get("indexed") is False and fm.get("fast"):
+                # fast-only text: raw-string str fast column, no inverted
+                # index (term queries become exact point ranges)
+                fields.append({"name": name, "type": "str", "fast": True})
             elif t == "text":
                 f = {"name": name, "type": "text",
                      "tokenizer": fm.get("tokenizer", "default")}
@@ -269,7 +273,8 @@ class Index:
         fields, ts_field = self._explicit_fields()
         explicit = {f["name"] for f in fields}
         flat_batches = [[flatten_doc(d) for d in b] for b in self.batches]
-        if self.config.get("doc_mapping", {}).get("mode") == "dynamic":
+        mode = self.config.get("doc_mapping", {}).get("mode", "dynamic")
+        if mode == "dynamic":  # quickwit's default mode
             inferred = _infer_dynamic_fields(explicit, flat_batches)
             if self.fast_normalizer == "lowercase":
                 for f in inferred:

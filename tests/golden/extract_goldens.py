@@ -130,6 +130,8 @@ def rest_scenarios():
                          "_teardown.quickwit.yaml"],
         "qw_search_api": ["_setup.quickwit.yaml", "0001_ts_range.yaml",
                           "0002_negative_search.yaml",
+                          "0003_exists_search.yaml", "0004_exact_string.yaml",
+                          "0005_fast_field_search.yaml",
                           "_teardown.quickwit.yaml"],
         "default_search_fields": ["_setup.quickwit.yaml",
                                   "0001_default_fields.yaml",
