@@ -575,6 +575,38 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
             out.aggs.push_back(std::move(a));
             continue;
         }
+        if (d.kind == AggDef::RANGE) {
+            a.kind = 4;
+            if (f && f->type != FastFieldView::STR && !f->multi) {
+                std::vector<uint64_t> counts(d.ranges.size(), 0);
+                auto rvisit = [&](uint32_t doc) {
+                    if (!f->present(doc)) return;
+                    double v;
+                    if (f->type == FastFieldView::U64) v = double(f->u64(doc));
+                    else if (f->type == FastFieldView::F64) v = f->f64(doc);
+                    else v = double(f->i64(doc));
+                    for (size_t ri = 0; ri < d.ranges.size(); ++ri) {
+                        const RangeSpec& r = d.ranges[ri];
+                        if (r.has_from && v < r.from) continue;
+                        if (r.has_to && v >= r.to) continue;
+                        counts[ri]++;
+                    }
+                };
+                if (m.all)
+                    for (uint32_t doc = 0; doc < sv.num_docs; ++doc) rvisit(doc);
+                else
+                    for (uint32_t doc : m.docs) rvisit(doc);
+                for (size_t ri = 0; ri < d.ranges.size(); ++ri) {
+                    if (!counts[ri]) continue;
+                    AggBucket b;
+                    b.key = double(ri);  // range index; finalize maps back
+                    b.doc_count = counts[ri];
+                    a.buckets.push_back(std::move(b));
+                }
+            }
+            out.aggs.push_back(std::move(a));
+            continue;
+        }
         a.kind = d.kind == AggDef::DATE_HISTOGRAM ? 1 : 2;
         std::map<int64_t, AggBucket> buckets;  // key quantized to bucket index
         auto visit = [&](uint32_t doc) {

@@ -57,7 +57,8 @@ struct PredDev {
                            // values (any-value-matches semantics); 0 = single
 };
 
-enum AggKindDev : uint32_t { AGGD_HISTO = 0, AGGD_TERMS = 1 };
+enum AggKindDev : uint32_t { AGGD_HISTO = 0, AGGD_TERMS = 1, AGGD_RANGE = 2 };
+constexpr uint32_t AGG_MAX_RANGES = 16;
 
 struct AggDev {
     uint32_t kind;
@@ -83,6 +84,10 @@ struct AggDev {
     uint64_t counts_out;    // byte offset into result scratch: u64[n_buckets]
     uint64_t matched_out;   // terms: u64 counter of docs-with-value
     uint64_t offsets_off;   // terms over a multi-valued column: u32 prefix
+    // AGGD_RANGE: [from, to) per range; NaN-free sentinels via has-masks
+    uint32_t n_ranges;
+    uint32_t r_has_from, r_has_to;  // bitmasks over ranges
+    double r_from[AGG_MAX_RANGES], r_to[AGG_MAX_RANGES];
     // one optional stats sub-agg set per bucket: {u64 cnt, f64 sum, u64 min_s,
     // u64 max_s (sortable-mapped)} × n_sub, laid out bucket-major
     uint32_t n_sub;

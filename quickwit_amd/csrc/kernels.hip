@@ -644,6 +644,20 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                                                                     a.matched_out),
                                               1ull);
                             }
+                        } else if (a.kind == AGGD_RANGE) {
+                            if (!a.n_ranges) continue;
+                            double v = agg_value(q, a.values_off, a.value_width,
+                                                 a.value_is_i64, d);
+                            for (uint32_t ri = 0; ri < a.n_ranges; ++ri) {
+                                bool in = true;
+                                if ((a.r_has_from >> ri) & 1) in &= v >= a.r_from[ri];
+                                if ((a.r_has_to >> ri) & 1) in &= v < a.r_to[ri];
+                                if (in)
+                                    atomicAdd((unsigned long long*)(q.results +
+                                                                    a.counts_out) +
+                                                  ri,
+                                              1ull);
+                            }
                         } else {
                             int64_t idx = histo_bucket(q, a, d);
                             if (idx < 0 || idx >= int64_t(a.n_buckets)) continue;

@@ -687,6 +687,26 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
             } else {
                 a.n_buckets = 0;  // no such column: zero buckets
             }
+        } else if (d.kind == AggDef::RANGE) {
+            a.kind = AGGD_RANGE;
+            if (f && f->type != FastFieldView::STR && !f->multi) {
+                if (d.ranges.size() > AGG_MAX_RANGES)
+                    throw std::runtime_error(">16 ranges in range aggregation");
+                a.n_buckets = uint32_t(d.ranges.size());
+                a.n_ranges = a.n_buckets;
+                a.values_off = f->values.off;
+                a.nulls_off = f->nullable ? f->nulls.off : 0;
+                a.value_width = 8;
+                a.value_is_i64 = f->type == FastFieldView::U64   ? 0
+                                 : f->type == FastFieldView::F64 ? 2
+                                                                 : 1;
+                for (size_t ri = 0; ri < d.ranges.size(); ++ri) {
+                    if (d.ranges[ri].has_from) a.r_has_from |= 1u << ri;
+                    if (d.ranges[ri].has_to) a.r_has_to |= 1u << ri;
+                    a.r_from[ri] = d.ranges[ri].from;
+                    a.r_to[ri] = d.ranges[ri].to;
+                }
+            }
         } else {
             a.kind = AGGD_HISTO;
             if (f && f->type != FastFieldView::STR) {
@@ -1185,6 +1205,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                 AggResult r;
                 r.name = d.name;
                 r.kind = d.kind == AggDef::TERMS      ? 3
+                         : d.kind == AggDef::RANGE     ? 4
                          : d.kind == AggDef::HISTOGRAM ? 2
                                                        : 1;
                 for (auto& s : d.sub) r.sub_names.push_back(s.name);
@@ -1747,6 +1768,15 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                             r.term_counts.emplace_back(f->dict_entry(o), counts[o]);
                     truncate_terms_split(
                         r, effective_split_size(d.size, d.split_size));
+                }
+            } else if (d.kind == AggDef::RANGE) {
+                r.kind = 4;
+                for (uint32_t ri = 0; ri < a.n_buckets; ++ri) {
+                    if (!counts[ri]) continue;
+                    AggBucket b;
+                    b.key = double(ri);  // range index; finalize maps back
+                    b.doc_count = counts[ri];
+                    r.buckets.push_back(std::move(b));
                 }
             } else {
                 r.kind = d.kind == AggDef::DATE_HISTOGRAM ? 1 : 2;

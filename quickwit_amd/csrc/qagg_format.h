@@ -344,6 +344,35 @@ inline std::string finalize_aggs_json(const IntermediateAggResults& ir,
         if (i) o += ",";
         mj::escape_to(o, d.name);
         o += ":";
+        if (d.kind == AggDef::RANGE) {
+            // buckets in the REQUEST's range order: {key, from?, to?,
+            // doc_count}; intermediate buckets carry the range INDEX as key
+            o += "{\"buckets\":[";
+            for (size_t ri = 0; ri < d.ranges.size(); ++ri) {
+                if (ri) o += ",";
+                uint64_t dc = 0;
+                for (const AggBucket& b : a.buckets)
+                    if (size_t(b.key) == ri) dc = b.doc_count;
+                const RangeSpec& r = d.ranges[ri];
+                o += "{\"doc_count\":";
+                char buf[32];
+                snprintf(buf, sizeof buf, "%llu", (unsigned long long)dc);
+                o += buf;
+                if (r.has_from) {
+                    o += ",\"from\":";
+                    mj::num_to(o, r.from);
+                }
+                o += ",\"key\":";
+                mj::escape_to(o, r.key);
+                if (r.has_to) {
+                    o += ",\"to\":";
+                    mj::num_to(o, r.to);
+                }
+                o += "}";
+            }
+            o += "]}";
+            continue;
+        }
         if (d.kind == AggDef::TERMS) {
             // order: doc_count desc, then key asc (ES/tantivy default);
             // truncate to size; sum_other = matched - shown

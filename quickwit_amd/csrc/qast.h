@@ -863,10 +863,17 @@ struct MetricAgg {
     std::string field;
 };
 
+struct RangeSpec {
+    std::string key;
+    bool has_from = false, has_to = false;
+    double from = 0, to = 0;  // [from, to) — ES range-agg semantics
+};
+
 struct AggDef {
     std::string name;
-    enum Kind { DATE_HISTOGRAM, HISTOGRAM, TERMS } kind = DATE_HISTOGRAM;
+    enum Kind { DATE_HISTOGRAM, HISTOGRAM, TERMS, RANGE } kind = DATE_HISTOGRAM;
     std::string field;
+    std::vector<RangeSpec> ranges;  // RANGE
     double interval = 0;  // ms for date_histogram; raw units for histogram
     double offset = 0;
     bool has_bounds = false;
@@ -924,6 +931,31 @@ inline std::vector<AggDef> parse_agg_request(const std::string& json) {
                 a.has_bounds = true;
                 a.bmin = eb->at("min")->num();
                 a.bmax = eb->at("max")->num();
+            }
+        } else if ((spec = body->get("range"))) {
+            a.kind = AggDef::RANGE;
+            a.field = spec->at("field")->s;
+            for (auto& rv : spec->at("ranges")->arr) {
+                RangeSpec r;
+                if (const mj::Value* f2 = rv->get("from")) {
+                    r.has_from = true;
+                    r.from = f2->num();
+                }
+                if (const mj::Value* t2 = rv->get("to")) {
+                    r.has_to = true;
+                    r.to = t2->num();
+                }
+                if (const mj::Value* k2 = rv->get("key")) r.key = k2->s;
+                else {
+                    char buf[64];
+                    auto fmt = [&](bool has, double v) {
+                        if (!has) return std::string("*");
+                        snprintf(buf, sizeof buf, "%g", v);
+                        return std::string(buf);
+                    };
+                    r.key = fmt(r.has_from, r.from) + "-" + fmt(r.has_to, r.to);
+                }
+                a.ranges.push_back(std::move(r));
             }
         } else if ((spec = body->get("terms"))) {
             a.kind = AggDef::TERMS;
