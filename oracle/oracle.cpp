@@ -575,6 +575,29 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
             out.aggs.push_back(std::move(a));
             continue;
         }
+        if (d.kind == AggDef::METRIC) {
+            a.kind = 5;
+            if (f && f->type != FastFieldView::STR && !f->multi) {
+                auto mvisit = [&](uint32_t doc) {
+                    if (!f->present(doc)) return;
+                    double v;
+                    if (f->type == FastFieldView::U64) v = double(f->u64(doc));
+                    else if (f->type == FastFieldView::F64) v = f->f64(doc);
+                    else v = double(f->i64(doc));
+                    a.metric.count++;
+                    a.metric.sum += v;
+                    a.metric.min = std::min(a.metric.min, v);
+                    a.metric.max = std::max(a.metric.max, v);
+                    a.metric.sum_sq += v * v;
+                };
+                if (m.all)
+                    for (uint32_t doc = 0; doc < sv.num_docs; ++doc) mvisit(doc);
+                else
+                    for (uint32_t doc : m.docs) mvisit(doc);
+            }
+            out.aggs.push_back(std::move(a));
+            continue;
+        }
         if (d.kind == AggDef::RANGE) {
             a.kind = 4;
             if (f && f->type != FastFieldView::STR && !f->multi) {
@@ -628,6 +651,7 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
                     sp.sum += sval;
                     sp.min = std::min(sp.min, sval);
                     sp.max = std::max(sp.max, sval);
+                    sp.sum_sq += sval * sval;
                 }
             }
         };

@@ -57,7 +57,9 @@ struct PredDev {
                            // values (any-value-matches semantics); 0 = single
 };
 
-enum AggKindDev : uint32_t { AGGD_HISTO = 0, AGGD_TERMS = 1, AGGD_RANGE = 2 };
+enum AggKindDev : uint32_t {
+    AGGD_HISTO = 0, AGGD_TERMS = 1, AGGD_RANGE = 2, AGGD_METRIC = 3
+};
 constexpr uint32_t AGG_MAX_RANGES = 16;
 
 struct AggDev {
@@ -89,7 +91,8 @@ struct AggDev {
     uint32_t r_has_from, r_has_to;  // bitmasks over ranges
     double r_from[AGG_MAX_RANGES], r_to[AGG_MAX_RANGES];
     // one optional stats sub-agg set per bucket: {u64 cnt, f64 sum, u64 min_s,
-    // u64 max_s (sortable-mapped)} × n_sub, laid out bucket-major
+    // u64 max_s (sortable-mapped), f64 sum_sq} × n_sub (40 B), bucket-major.
+    // AGGD_METRIC uses ONE such slot at counts_out.
     uint32_t n_sub;
     uint64_t sub_out;        // byte offset: n_buckets * n_sub * 32 bytes
     uint64_t sub_values_off[4];  // sub-agg source columns (<=4)

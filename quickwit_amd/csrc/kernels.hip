@@ -644,6 +644,18 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                                                                     a.matched_out),
                                               1ull);
                             }
+                        } else if (a.kind == AGGD_METRIC) {
+                            if (!a.values_off) continue;
+                            double v = agg_value(q, a.values_off, a.value_width,
+                                                 a.value_is_i64, d);
+                            uint8_t* slot = q.results + a.counts_out;
+                            atomicAdd((unsigned long long*)slot, 1ull);
+                            atomicAdd((double*)(slot + 8), v);
+                            atomicMin((unsigned long long*)(slot + 16),
+                                      (unsigned long long)f64_sortable(v));
+                            atomicMax((unsigned long long*)(slot + 24),
+                                      (unsigned long long)f64_sortable(v));
+                            atomicAdd((double*)(slot + 32), v * v);
                         } else if (a.kind == AGGD_RANGE) {
                             if (!a.n_ranges) continue;
                             double v = agg_value(q, a.values_off, a.value_width,
@@ -679,13 +691,14 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                                 double sv = agg_value(q, a.sub_values_off[si],
                                                       a.sub_width[si], a.sub_is_i64[si], d);
                                 uint8_t* slot = q.results + a.sub_out +
-                                                (uint64_t(idx) * a.n_sub + si) * 32;
+                                                (uint64_t(idx) * a.n_sub + si) * 40;
                                 atomicAdd((unsigned long long*)slot, 1ull);
                                 atomicAdd((double*)(slot + 8), sv);
                                 atomicMin((unsigned long long*)(slot + 16),
                                           (unsigned long long)f64_sortable(sv));
                                 atomicMax((unsigned long long*)(slot + 24),
                                           (unsigned long long)f64_sortable(sv));
+                                atomicAdd((double*)(slot + 32), sv * sv);
                             }
                         }
                     }

@@ -687,6 +687,17 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
             } else {
                 a.n_buckets = 0;  // no such column: zero buckets
             }
+        } else if (d.kind == AggDef::METRIC) {
+            a.kind = AGGD_METRIC;
+            if (f && f->type != FastFieldView::STR && !f->multi) {
+                a.n_buckets = 1;  // one 40B stats slot at counts_out
+                a.values_off = f->values.off;
+                a.nulls_off = f->nullable ? f->nulls.off : 0;
+                a.value_width = 8;
+                a.value_is_i64 = f->type == FastFieldView::U64   ? 0
+                                 : f->type == FastFieldView::F64 ? 2
+                                                                 : 1;
+            }
         } else if (d.kind == AggDef::RANGE) {
             a.kind = AGGD_RANGE;
             if (f && f->type != FastFieldView::STR && !f->multi) {
@@ -779,13 +790,14 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
                 a.n_buckets = 0;
             }
         }
-        // layout: counts u64[n_buckets] | matched u64 | subs n_buckets*n_sub*32
+        // layout: counts u64[n_buckets] (METRIC: one 40 B stats slot) |
+        // matched u64 | subs n_buckets*n_sub*40
         a.counts_out = off;
-        off += uint64_t(a.n_buckets) * 8;
+        off += a.kind == AGGD_METRIC ? 40 : uint64_t(a.n_buckets) * 8;
         a.matched_out = off;
         off += 8;
         a.sub_out = off;
-        off += uint64_t(a.n_buckets) * a.n_sub * 32;
+        off += uint64_t(a.n_buckets) * a.n_sub * 40;
         ap.devs.push_back(a);
     }
     // LDS slot assignment: the first histogram agg that fits gets the LDS
@@ -812,10 +824,14 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
     ap.init.assign(ap.out_bytes, 0);
     for (size_t i = 0; i < ap.devs.size(); ++i) {
         const AggDev& a = ap.devs[i];
+        uint64_t ones = ~0ull;
+        if (a.kind == AGGD_METRIC && a.n_buckets) {
+            uint64_t slot = a.counts_out - out_base;
+            memcpy(&ap.init[slot + 16], &ones, 8);  // min: max sortable
+        }
         for (uint64_t b = 0; b < a.n_buckets; ++b)
             for (uint32_t s = 0; s < a.n_sub; ++s) {
-                uint64_t slot = a.sub_out - out_base + (b * a.n_sub + s) * 32;
-                uint64_t ones = ~0ull;
+                uint64_t slot = a.sub_out - out_base + (b * a.n_sub + s) * 40;
                 memcpy(&ap.init[slot + 16], &ones, 8);  // min slot: max sortable
                 // max slot stays 0 (= most negative sortable)
             }
@@ -1206,6 +1222,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                 r.name = d.name;
                 r.kind = d.kind == AggDef::TERMS      ? 3
                          : d.kind == AggDef::RANGE     ? 4
+                         : d.kind == AggDef::METRIC    ? 5
                          : d.kind == AggDef::HISTOGRAM ? 2
                                                        : 1;
                 for (auto& s : d.sub) r.sub_names.push_back(s.name);
@@ -1769,6 +1786,23 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     truncate_terms_split(
                         r, effective_split_size(d.size, d.split_size));
                 }
+            } else if (d.kind == AggDef::METRIC) {
+                r.kind = 5;
+                if (a.n_buckets) {
+                    const uint8_t* slot = agg_out.data() + (a.counts_out - r_agg);
+                    StatsPayload sp2;
+                    uint64_t mn, mx;
+                    memcpy(&sp2.count, slot, 8);
+                    memcpy(&sp2.sum, slot + 8, 8);
+                    memcpy(&mn, slot + 16, 8);
+                    memcpy(&mx, slot + 24, 8);
+                    memcpy(&sp2.sum_sq, slot + 32, 8);
+                    if (sp2.count) {
+                        sp2.min = u64_to_f64(mn);
+                        sp2.max = u64_to_f64(mx);
+                    }
+                    r.metric = sp2;
+                }
             } else if (d.kind == AggDef::RANGE) {
                 r.kind = 4;
                 for (uint32_t ri = 0; ri < a.n_buckets; ++ri) {
@@ -1788,13 +1822,14 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     b.doc_count = counts[bi];
                     b.sub.resize(d.sub.size());
                     for (uint32_t s = 0; s < a.n_sub; ++s) {
-                        const uint8_t* slot = subs + (uint64_t(bi) * a.n_sub + s) * 32;
+                        const uint8_t* slot = subs + (uint64_t(bi) * a.n_sub + s) * 40;
                         StatsPayload sp2;
                         uint64_t mn, mx;
                         memcpy(&sp2.count, slot, 8);
                         memcpy(&sp2.sum, slot + 8, 8);
                         memcpy(&mn, slot + 16, 8);
                         memcpy(&mx, slot + 24, 8);
+                        memcpy(&sp2.sum_sq, slot + 32, 8);
                         if (sp2.count) {
                             sp2.min = u64_to_f64(mn);
                             sp2.max = u64_to_f64(mx);

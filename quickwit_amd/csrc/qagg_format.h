@@ -37,11 +37,13 @@ struct StatsPayload {
     double sum = 0;
     double min = std::numeric_limits<double>::infinity();
     double max = -std::numeric_limits<double>::infinity();
+    double sum_sq = 0;  // extended_stats (sum of squares / variance)
     void merge(const StatsPayload& o) {
         count += o.count;
         sum += o.sum;
         min = std::min(min, o.min);
         max = std::max(max, o.max);
+        sum_sq += o.sum_sq;
     }
 };
 
@@ -57,6 +59,7 @@ struct AggResult {
     std::vector<std::string> sub_names;
     std::vector<AggBucket> buckets;                            // histos (sorted by key)
     std::vector<std::pair<std::string, uint64_t>> term_counts; // terms (sorted by key)
+    StatsPayload metric;  // kind 5: top-level metric aggregation
     uint64_t terms_matched_docs = 0;
     // sum over truncated splits of the last-included term count — the ES
     // doc_count_error_upper_bound semantics the golden scenario pins
@@ -95,7 +98,7 @@ struct IntermediateAggResults {
         std::string o;
         auto put = [&](const void* p, size_t n) { o.append((const char*)p, n); };
         uint32_t magic = 0x31474151;
-        uint16_t ver = 2, n = uint16_t(aggs.size());
+        uint16_t ver = 3, n = uint16_t(aggs.size());
         put(&magic, 4);
         put(&ver, 2);
         put(&n, 2);
@@ -111,7 +114,13 @@ struct IntermediateAggResults {
                 put(&sl, 2);
                 put(s.data(), sl);
             }
-            if (a.kind == 3) {
+            if (a.kind == 5) {
+                put(&a.metric.count, 8);
+                put(&a.metric.sum, 8);
+                put(&a.metric.min, 8);
+                put(&a.metric.max, 8);
+                put(&a.metric.sum_sq, 8);
+            } else if (a.kind == 3) {
                 put(&a.terms_matched_docs, 8);
                 put(&a.terms_error_bound, 8);
                 uint32_t ne = uint32_t(a.term_counts.size());
@@ -135,6 +144,7 @@ struct IntermediateAggResults {
                         put(&sp.sum, 8);
                         put(&sp.min, 8);
                         put(&sp.max, 8);
+                        put(&sp.sum_sq, 8);
                     }
                 }
             }
@@ -158,7 +168,7 @@ struct IntermediateAggResults {
         get(&magic, 4);
         get(&ver, 2);
         get(&n, 2);
-        if (magic != 0x31474151 || ver != 2) throw std::runtime_error("QAGG1: bad header");
+        if (magic != 0x31474151 || ver != 3) throw std::runtime_error("QAGG1: bad header");
         for (int i = 0; i < n; ++i) {
             AggResult a;
             uint16_t nl;
@@ -176,7 +186,13 @@ struct IntermediateAggResults {
                 a.sub_names.emplace_back((const char*)p, sl);
                 p += sl;
             }
-            if (a.kind == 3) {
+            if (a.kind == 5) {
+                get(&a.metric.count, 8);
+                get(&a.metric.sum, 8);
+                get(&a.metric.min, 8);
+                get(&a.metric.max, 8);
+                get(&a.metric.sum_sq, 8);
+            } else if (a.kind == 3) {
                 get(&a.terms_matched_docs, 8);
                 get(&a.terms_error_bound, 8);
                 uint32_t ne;
@@ -206,6 +222,7 @@ struct IntermediateAggResults {
                         get(&sp.sum, 8);
                         get(&sp.min, 8);
                         get(&sp.max, 8);
+                        get(&sp.sum_sq, 8);
                     }
                     a.buckets.push_back(std::move(b));
                 }
@@ -228,7 +245,9 @@ struct IntermediateAggResults {
             const AggResult& b = o.aggs[i];
             if (a.name != b.name || a.kind != b.kind)
                 throw std::runtime_error("QAGG1: merge mismatch");
-            if (a.kind == 3) {
+            if (a.kind == 5) {
+                a.metric.merge(b.metric);
+            } else if (a.kind == 3) {
                 std::vector<std::pair<std::string, uint64_t>> merged;
                 merged.reserve(a.term_counts.size() + b.term_counts.size());
                 size_t x = 0, y = 0;
@@ -330,6 +349,60 @@ inline void stats_to_json(std::string& o, const MetricAgg& m, const StatsPayload
             num(double(s.count));
             o += "}";
             break;
+        case MetricAgg::EXTENDED: {
+            // ES extended_stats shape (variance = population variance)
+            double avg = empty ? 0 : s.sum / double(s.count);
+            double var = empty ? 0 : s.sum_sq / double(s.count) - avg * avg;
+            if (var < 0) var = 0;
+            double sd = sqrt(var);
+            double n = double(s.count);
+            double var_samp = s.count > 1
+                                  ? (s.sum_sq - s.sum * s.sum / n) / (n - 1.0)
+                                  : 0.0;
+            if (var_samp < 0) var_samp = 0;
+            o += "{\"avg\":";
+            if (empty) o += "null";
+            else num(avg);
+            o += ",\"count\":";
+            num(double(s.count));
+            o += ",\"max\":";
+            if (empty) o += "null";
+            else num(s.max);
+            o += ",\"min\":";
+            if (empty) o += "null";
+            else num(s.min);
+            o += ",\"std_deviation\":";
+            if (empty) o += "null";
+            else num(sd);
+            o += ",\"std_deviation_bounds\":{\"lower\":";
+            if (empty) o += "null";
+            else num(avg - 2 * sd);
+            o += ",\"upper\":";
+            if (empty) o += "null";
+            else num(avg + 2 * sd);
+            o += "},\"std_deviation_population\":";
+            if (empty) o += "null";
+            else num(sd);
+            o += ",\"std_deviation_sampling\":";
+            if (empty) o += "null";
+            else num(sqrt(var_samp));
+            o += ",\"sum\":";
+            num(s.sum);
+            o += ",\"sum_of_squares\":";
+            if (empty) o += "null";
+            else num(s.sum_sq);
+            o += ",\"variance\":";
+            if (empty) o += "null";
+            else num(var);
+            o += ",\"variance_population\":";
+            if (empty) o += "null";
+            else num(var);
+            o += ",\"variance_sampling\":";
+            if (empty) o += "null";
+            else num(var_samp);
+            o += "}";
+            break;
+        }
     }
 }
 
@@ -344,6 +417,10 @@ inline std::string finalize_aggs_json(const IntermediateAggResults& ir,
         if (i) o += ",";
         mj::escape_to(o, d.name);
         o += ":";
+        if (d.kind == AggDef::METRIC) {
+            stats_to_json(o, d.metric, a.metric);
+            continue;
+        }
         if (d.kind == AggDef::RANGE) {
             // buckets in the REQUEST's range order: {key, from?, to?,
             // doc_count}; intermediate buckets carry the range INDEX as key

@@ -859,7 +859,7 @@ inline std::string ms_to_rfc3339(int64_t ms) {
 // ---------------------------------------------------------------- agg plan
 struct MetricAgg {
     std::string name;
-    enum Kind { STATS, AVG, SUM, MIN, MAX, COUNT } kind = STATS;
+    enum Kind { STATS, AVG, SUM, MIN, MAX, COUNT, EXTENDED } kind = STATS;
     std::string field;
 };
 
@@ -871,9 +871,11 @@ struct RangeSpec {
 
 struct AggDef {
     std::string name;
-    enum Kind { DATE_HISTOGRAM, HISTOGRAM, TERMS, RANGE } kind = DATE_HISTOGRAM;
+    enum Kind { DATE_HISTOGRAM, HISTOGRAM, TERMS, RANGE, METRIC } kind =
+        DATE_HISTOGRAM;
     std::string field;
     std::vector<RangeSpec> ranges;  // RANGE
+    MetricAgg metric;               // METRIC (top-level stats/avg/... agg)
     double interval = 0;  // ms for date_histogram; raw units for histogram
     double offset = 0;
     bool has_bounds = false;
@@ -972,7 +974,23 @@ inline std::vector<AggDef> parse_agg_request(const std::string& json) {
             if (const mj::Value* mdc = spec->get("min_doc_count"))
                 a.min_doc_count = mdc->as_i64();
         } else {
-            throw std::runtime_error("aggregation not supported: " + a.name);
+            // top-level metric aggregations (stats/avg/sum/min/max/
+            // value_count/extended_stats over every matched doc)
+            const mj::Value* ms0 = nullptr;
+            MetricAgg::Kind mk = MetricAgg::STATS;
+            if ((ms0 = body->get("stats"))) mk = MetricAgg::STATS;
+            else if ((ms0 = body->get("extended_stats"))) mk = MetricAgg::EXTENDED;
+            else if ((ms0 = body->get("avg"))) mk = MetricAgg::AVG;
+            else if ((ms0 = body->get("sum"))) mk = MetricAgg::SUM;
+            else if ((ms0 = body->get("min"))) mk = MetricAgg::MIN;
+            else if ((ms0 = body->get("max"))) mk = MetricAgg::MAX;
+            else if ((ms0 = body->get("value_count"))) mk = MetricAgg::COUNT;
+            else throw std::runtime_error("aggregation not supported: " + a.name);
+            a.kind = AggDef::METRIC;
+            a.field = ms0->at("field")->s;
+            a.metric.kind = mk;
+            a.metric.name = a.name;
+            a.metric.field = a.field;
         }
         if (a.min_doc_count < 0) a.min_doc_count = (a.kind == AggDef::TERMS) ? 1 : 0;
         if (const mj::Value* subs = body->get("aggs")) {
@@ -982,6 +1000,7 @@ inline std::vector<AggDef> parse_agg_request(const std::string& json) {
                 const mj::Value* sb = skv.second.get();
                 const mj::Value* ms = nullptr;
                 if ((ms = sb->get("stats"))) m.kind = MetricAgg::STATS;
+                else if ((ms = sb->get("extended_stats"))) m.kind = MetricAgg::EXTENDED;
                 else if ((ms = sb->get("avg"))) m.kind = MetricAgg::AVG;
                 else if ((ms = sb->get("sum"))) m.kind = MetricAgg::SUM;
                 else if ((ms = sb->get("min"))) m.kind = MetricAgg::MIN;

@@ -915,3 +915,28 @@ def test_f64_column_sort_range_agg():
         for k in ("count", "min", "max"):
             assert gb["st"][k] == eb["st"][k], (k, gb, eb)
         assert math.isclose(gb["st"]["sum"], eb["st"]["sum"], rel_tol=1e-12)
+
+
+# ------------------------------------------------- metric aggregations
+def test_top_level_metric_aggs(searchers):
+    gpu, cpu = searchers
+    aggs = {"ts_stats": {"stats": {"field": "timestamp"}},
+            "t_ext": {"extended_stats": {"field": "tenant_id"}},
+            "t_avg": {"avg": {"field": "tenant_id"}},
+            "t_cnt": {"value_count": {"field": "tenant_id"}}}
+    req = make_leaf_request(TERM_Q, SCHEMA, [(SID, NDOCS)], max_hits=0,
+                            aggregation=aggs)
+    g = gpu.leaf_search(req)
+    e = cpu.leaf_search(req)
+    gj = gpu.finalize_agg_json(g["intermediate_aggregation_result"], aggs)
+    ej = cpu.finalize_agg_json(e["intermediate_aggregation_result"], aggs)
+    for name in aggs:
+        for k, ev in ej[name].items():
+            gv = gj[name][k]
+            if isinstance(ev, dict):
+                for k2 in ev:
+                    assert math.isclose(gv[k2], ev[k2], rel_tol=1e-9), (name, k, k2)
+            elif isinstance(ev, float):
+                assert math.isclose(gv, ev, rel_tol=1e-9), (name, k, gv, ev)
+            else:
+                assert gv == ev, (name, k, gv, ev)
