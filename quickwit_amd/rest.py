@@ -25,6 +25,48 @@ from . import splitgen
 from .api import make_leaf_request
 
 
+def _parse_java_date(fmt, s):
+    """Java/ES date-format pattern subset (yyyy MM dd HH mm ss S+, quoted
+    literals, arbitrary separators) -> epoch milliseconds (float: fractional
+    seconds keep sub-ms precision)."""
+    import datetime as _dt
+    fields = {"yyyy": 0, "MM": 1, "dd": 1, "HH": 0, "mm": 0, "ss": 0}
+    frac = 0.0
+    fi = si = 0
+    while fi < len(fmt):
+        matched = False
+        for tok in ("yyyy", "MM", "dd", "HH", "mm", "ss"):
+            if fmt.startswith(tok, fi):
+                fields[tok] = int(s[si:si + len(tok)])
+                si += len(tok)
+                fi += len(tok)
+                matched = True
+                break
+        if matched:
+            continue
+        if fmt[fi] == "S":
+            n = 0
+            while fi + n < len(fmt) and fmt[fi + n] == "S":
+                n += 1
+            frac = int(s[si:si + n]) / (10 ** n)
+            si += n
+            fi += n
+            continue
+        if fmt[fi] == "'":
+            j = fmt.index("'", fi + 1)
+            lit = fmt[fi + 1:j]
+            assert s.startswith(lit, si), (fmt, s, si)
+            si += len(lit)
+            fi = j + 1
+            continue
+        assert si < len(s) and s[si] == fmt[fi], (fmt, s, si)
+        si += 1
+        fi += 1
+    d = _dt.datetime(fields["yyyy"], fields["MM"], fields["dd"], fields["HH"],
+                     fields["mm"], fields["ss"], tzinfo=_dt.timezone.utc)
+    return d.timestamp() * 1000.0 + frac * 1000.0
+
+
 # -------------------------------------------------- ES query DSL -> QueryAst
 # (quickwit-query/src/elastic_query_dsl/ subset the scenario suites use)
 def es_query_to_ast(q, schema=None):
@@ -81,14 +123,25 @@ def es_query_to_ast(q, schema=None):
     if "range" in q:
         [(field, body)] = q["range"].items()
         node = {"type": "range", "field": field}
+        fmt = body.get("format") if isinstance(body, dict) else None
+
+        def conv(v):
+            if fmt is None:
+                return v
+            # custom date format (Java-style pattern subset): parse to epoch
+            # ms, truncated to the column's fast precision exactly like the
+            # reference truncates query bounds (es_compat 0007 pins this:
+            # gte == lte == ..001999 matches the ..001 document)
+            return int(_parse_java_date(fmt, str(v)))
+
         if "gte" in body:
-            node["lower_bound"] = {"included": body["gte"]}
+            node["lower_bound"] = {"included": conv(body["gte"])}
         elif "gt" in body:
-            node["lower_bound"] = {"excluded": body["gt"]}
+            node["lower_bound"] = {"excluded": conv(body["gt"])}
         if "lte" in body:
-            node["upper_bound"] = {"included": body["lte"]}
+            node["upper_bound"] = {"included": conv(body["lte"])}
         elif "lt" in body:
-            node["upper_bound"] = {"excluded": body["lt"]}
+            node["upper_bound"] = {"excluded": conv(body["lt"])}
         return node
     if "match" in q:
         [(field, body)] = q["match"].items()
