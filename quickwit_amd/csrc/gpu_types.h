@@ -133,6 +133,8 @@ struct QueryDev {
     // column min/max => no range check), optional agg[1] LDS terms over a
     // non-nullable ord column. Descriptors hoisted to registers.
     uint32_t agg_fast;
+    uint32_t agg_nt;        // experiment: nontemporal loads in the pure-agg
+                            // loop (QW_AGG_NT=1; streaming columns bypass L2)
     uint64_t bitmap_out;    // absolute device VA: store each tile's match
                             // bitset words here (predicate-cache fill); 0=off
     uint64_t terms_off;     // scratch offsets of descriptor arrays

@@ -519,7 +519,9 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
         if (NA && !NS && !NB && !NC && q.agg_fast && q.match_all && !q.n_preds) {
             #pragma unroll 4
             for (uint32_t d = tile_lo + threadIdx.x; d < tile_hi; d += TILE_THREADS) {
-                int64_t num = af_col[d] - af_ioff;
+                int64_t num = (q.agg_nt ? __builtin_nontemporal_load(&af_col[d])
+                                        : af_col[d]) -
+                              af_ioff;
                 int64_t idx = int64_t(floor(double(num) * af_inv));
                 if (idx * af_ivl > num) --idx;
                 else if ((idx + 1) * af_ivl <= num) ++idx;

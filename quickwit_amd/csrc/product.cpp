@@ -1463,6 +1463,8 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
               ap.devs[1].nulls_off == 0 && ap.devs[1].offsets_off == 0 &&
               ap.devs[1].n_buckets > 0)))
             q.agg_fast = 1;
+        static const bool agg_nt = getenv("QW_AGG_NT") != nullptr;
+        q.agg_nt = agg_nt ? 1 : 0;
         HIP_CHECK(hipEventRecord(ctx->ev_start, ctx->stream));
         launch_leaf_tile(ns, nb, na, collect, dim3(grid), ctx->stream, q, 0u,
                          n_tiles, 1u);
