@@ -940,3 +940,80 @@ def test_top_level_metric_aggs(searchers):
                 assert math.isclose(gv, ev, rel_tol=1e-9), (name, k, gv, ev)
             else:
                 assert gv == ev, (name, k, gv, ev)
+
+
+# ------------------------------------------------- nested boolean shapes
+# (the recursive device-bitmap evaluator fallback — product.cpp bitmap_eval)
+NESTED_SHAPES = [
+    # must_not over a conjunction
+    {"type": "bool", "must_not": [
+        {"type": "bool", "must": [
+            {"type": "term", "field": "severity_text", "value": "INFO"},
+            {"type": "term", "field": "body", "value": "w00001"}]}]},
+    # OR of ANDs
+    {"type": "bool", "should": [
+        {"type": "bool", "must": [
+            {"type": "term", "field": "severity_text", "value": "ERROR"},
+            {"type": "term", "field": "body", "value": "w00002"}]},
+        {"type": "bool", "must": [
+            {"type": "term", "field": "severity_text", "value": "WARN"},
+            {"type": "term", "field": "body", "value": "w00003"}]}]},
+    # presence and range in should position
+    {"type": "bool", "should": [
+        {"type": "field_presence", "field": "tenant_id"},
+        {"type": "range", "field": "tenant_id",
+         "lower_bound": {"included": 990}, "upper_bound": "Unbounded"}]},
+    # deep nesting with must_not inside should inside must
+    {"type": "bool", "must": [
+        {"type": "bool", "should": [
+            {"type": "bool", "must_not": [
+                {"type": "term", "field": "severity_text", "value": "DEBUG"}]},
+            {"type": "term", "field": "body", "value": "w00004"}]},
+        {"type": "range", "field": "tenant_id",
+         "lower_bound": {"included": 0}, "upper_bound": {"excluded": 800}}]},
+    # nested bool under filter with its own must_not
+    {"type": "bool",
+     "must": [{"type": "term", "field": "severity_text", "value": "INFO"}],
+     "filter": [{"type": "bool",
+                 "should": [{"type": "term", "field": "body", "value": "w00005"},
+                            {"type": "term", "field": "body", "value": "w00006"}],
+                 "must_not": [{"type": "term", "field": "body",
+                               "value": "w00007"}]}]},
+]
+
+
+@pytest.mark.parametrize("qi", range(len(NESTED_SHAPES)))
+def test_nested_boolean_bitmap_fallback(searchers, qi):
+    got, exp = run_both(searchers, NESTED_SHAPES[qi], max_hits=25)
+    assert_hits_equal(got, exp, scored=False)
+
+
+def test_nested_boolean_with_sort_and_agg(searchers):
+    q = NESTED_SHAPES[1]
+    got, exp = run_both(searchers, q, max_hits=20, sort_fields=[
+        {"field_name": "timestamp", "sort_order": 1}])
+    assert [(hid(h), sv_of(h)) for h in got.get("partial_hits", [])] == \
+           [(hid(h), sv_of(h)) for h in exp.get("partial_hits", [])]
+    gpu, cpu = searchers
+    aggs = {"pt": {"terms": {"field": "tenant_name", "size": 5}}}
+    req = make_leaf_request(q, SCHEMA, [(SID, NDOCS)], max_hits=0,
+                            aggregation=aggs)
+    g = gpu.leaf_search(req)
+    e = cpu.leaf_search(req)
+    gj = gpu.finalize_agg_json(g["intermediate_aggregation_result"], aggs)
+    ej = cpu.finalize_agg_json(e["intermediate_aggregation_result"], aggs)
+    assert gj == ej
+
+
+def test_bitmap_fallback_memoized(searchers):
+    # second identical nested query reuses the memoized subtree bitmaps:
+    # only the main search kernel launches
+    gpu, _ = searchers
+    q = NESTED_SHAPES[0]
+    req = make_leaf_request(q, SCHEMA, [(SID, NDOCS)], max_hits=5)
+    gpu.leaf_search(req)
+    gpu.kernel_stats_reset()
+    gpu.leaf_search(req)
+    ms, n = gpu.kernel_stats("union_bm25")
+    ms2, n2 = gpu.kernel_stats("range_filter")
+    assert n + n2 == 1, (n, n2)
