@@ -60,12 +60,32 @@ struct AggResult {
     std::vector<AggBucket> buckets;                            // histos (sorted by key)
     std::vector<std::pair<std::string, uint64_t>> term_counts; // terms (sorted by key)
     StatsPayload metric;  // kind 5: top-level metric aggregation
+    // terms key domain: 0 = str (dictionary term bytes); 1/2/3 = numeric fast
+    // column (u64 / i64-or-date / f64), keys stored as the value's 8-byte
+    // big-endian SORTABLE bits so the string-keyed merge/truncate machinery
+    // keeps numeric order (tantivy terms agg over numeric columns —
+    // aggregation/bucket/term_agg.rs keys by the column value)
+    uint8_t key_kind = 0;
     uint64_t terms_matched_docs = 0;
     // sum over truncated splits of the last-included term count — the ES
     // doc_count_error_upper_bound semantics the golden scenario pins
     // (aggregations/0001 split_size=1 case)
     uint64_t terms_error_bound = 0;
 };
+
+// numeric terms keys: 8-byte big-endian of the order-preserving sortable
+// bits (u64: identity; i64/date: sign-flip; f64: f64_to_u64 total order)
+inline std::string num_term_key(uint64_t sortable) {
+    char b[8];
+    for (int i = 0; i < 8; ++i) b[i] = char(sortable >> (56 - 8 * i));
+    return std::string(b, 8);
+}
+inline uint64_t num_term_key_bits(const std::string& k) {
+    uint64_t v = 0;
+    for (int i = 0; i < 8 && i < int(k.size()); ++i)
+        v = v << 8 | uint8_t(k[i]);
+    return v;
+}
 
 // per-split terms truncation (tantivy terms agg split_size): keep the top
 // `split_size` entries by (count desc, key asc); when anything is dropped,
