@@ -571,6 +571,37 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
                 // dict order is lexicographic => term_counts sorted by key;
                 // per-split split_size truncation + error bound (qagg_format.h)
                 truncate_terms_split(a, effective_split_size(d.size, d.split_size));
+            } else if (f && !f->multi) {
+                // terms over a numeric fast column: count by the value's
+                // order-preserving sortable bits, keys encoded big-endian so
+                // map order == numeric order (mirrors the product's
+                // AGGD_TERMS_NUM hash table; tantivy term_agg.rs keys by the
+                // column value)
+                if (sv.num_docs > (1u << 21))
+                    throw std::runtime_error(
+                        "terms aggregation over a numeric fast field on a "
+                        ">2M-doc split (r1 limit)");
+                a.key_kind = f->type == FastFieldView::U64   ? 1
+                             : f->type == FastFieldView::F64 ? 3
+                                                             : 2;
+                std::map<uint64_t, uint64_t> counts;
+                auto nvisit = [&](uint32_t doc) {
+                    if (!f->present(doc)) return;
+                    uint64_t v;
+                    if (f->type == FastFieldView::U64) v = f->u64(doc);
+                    else if (f->type == FastFieldView::F64)
+                        v = f64_to_u64(f->f64(doc));
+                    else v = i64_to_u64(f->i64(doc));
+                    counts[v]++;
+                    a.terms_matched_docs++;
+                };
+                if (m.all)
+                    for (uint32_t doc = 0; doc < sv.num_docs; ++doc) nvisit(doc);
+                else
+                    for (uint32_t doc : m.docs) nvisit(doc);
+                for (auto& kv : counts)
+                    a.term_counts.emplace_back(num_term_key(kv.first), kv.second);
+                truncate_terms_split(a, effective_split_size(d.size, d.split_size));
             }
             out.aggs.push_back(std::move(a));
             continue;

@@ -58,7 +58,12 @@ struct PredDev {
 };
 
 enum AggKindDev : uint32_t {
-    AGGD_HISTO = 0, AGGD_TERMS = 1, AGGD_RANGE = 2, AGGD_METRIC = 3
+    AGGD_HISTO = 0, AGGD_TERMS = 1, AGGD_RANGE = 2, AGGD_METRIC = 3,
+    // terms over a NUMERIC fast column: open-addressing [key,count] u64-pair
+    // hash table in the counts_out region (slots = (n_buckets-2)/2, power of
+    // two; trailing two words = count for the key ~0 sentinel value +
+    // overflow flag), keyed by the value's order-preserving sortable bits
+    AGGD_TERMS_NUM = 4
 };
 constexpr uint32_t AGG_MAX_RANGES = 16;
 

@@ -646,6 +646,51 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                                                                     a.matched_out),
                                               1ull);
                             }
+                        } else if (a.kind == AGGD_TERMS_NUM) {
+                            // numeric-column terms: global open-addressing
+                            // hash of the value's sortable bits (layout in
+                            // gpu_types.h). Null docs were skipped above.
+                            if (!a.n_buckets) continue;
+                            uint64_t raw =
+                                ((const uint64_t*)(q.split + a.values_off))[d];
+                            uint64_t v = raw;
+                            if (a.value_is_i64 == 1) v = raw ^ (1ull << 63);
+                            else if (a.value_is_i64 == 2)
+                                v = f64_sortable(__longlong_as_double(raw));
+                            unsigned long long* tab =
+                                (unsigned long long*)(q.results + a.counts_out);
+                            uint32_t slots = (a.n_buckets - 2) >> 1;
+                            const unsigned long long SENT = ~0ull;
+                            if (v == SENT) {
+                                atomicAdd(&tab[2 * slots], 1ull);
+                            } else {
+                                uint64_t h = v * 0x9E3779B97F4A7C15ull;
+                                h ^= h >> 32;
+                                uint32_t sl = uint32_t(h) & (slots - 1);
+                                bool done = false;
+                                for (uint32_t pr = 0; pr < slots; ++pr) {
+                                    unsigned long long cur = tab[2 * sl];
+                                    if (cur == SENT)
+                                        cur = atomicCAS(&tab[2 * sl], SENT,
+                                                        (unsigned long long)v);
+                                    if (cur == SENT ||
+                                        cur == (unsigned long long)v) {
+                                        atomicAdd(&tab[2 * sl + 1], 1ull);
+                                        done = true;
+                                        break;
+                                    }
+                                    sl = (sl + 1) & (slots - 1);
+                                }
+                                if (!done) tab[2 * slots + 1] = 1ull;  // overflow
+                            }
+                            if (a.nulls_off) {
+                                if (ai < 4)
+                                    atomicAdd(&sc_agg_matched[ai], 1u);
+                                else
+                                    atomicAdd((unsigned long long*)(q.results +
+                                                                    a.matched_out),
+                                              1ull);
+                            }
                         } else if (a.kind == AGGD_METRIC) {
                             if (!a.values_off) continue;
                             double v = agg_value(q, a.values_off, a.value_width,

@@ -684,6 +684,27 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
                 a.nulls_off = f->nullable ? f->nulls.off : 0;
                 a.value_width = uint32_t(f->ord_width);
                 if (f->multi) a.offsets_off = f->value_offsets.off;
+            } else if (f && !f->multi) {
+                // terms over a numeric fast column (tantivy term_agg keys by
+                // the column value): device hash table sized so distinct
+                // values can never fill it (<= num_docs). Bounded at 4M slots
+                // — beyond that the per-doc worst-case probe cost is a
+                // hazard, and no golden/benchmark path needs it.
+                if (sv.num_docs > (1u << 21))
+                    throw std::runtime_error(
+                        "terms aggregation over a numeric fast field on a "
+                        ">2M-doc split (r1 limit)");
+                a.kind = AGGD_TERMS_NUM;
+                uint64_t want = uint64_t(sv.num_docs) * 2;
+                uint32_t slots = 1024;
+                while (slots < want) slots <<= 1;
+                a.n_buckets = slots * 2 + 2;
+                a.values_off = f->values.off;
+                a.nulls_off = f->nullable ? f->nulls.off : 0;
+                a.value_width = 8;
+                a.value_is_i64 = f->type == FastFieldView::U64   ? 0
+                                 : f->type == FastFieldView::F64 ? 2
+                                                                 : 1;
             } else {
                 a.n_buckets = 0;  // no such column: zero buckets
             }
@@ -828,6 +849,13 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
         if (a.kind == AGGD_METRIC && a.n_buckets) {
             uint64_t slot = a.counts_out - out_base;
             memcpy(&ap.init[slot + 16], &ones, 8);  // min: max sortable
+        }
+        if (a.kind == AGGD_TERMS_NUM && a.n_buckets) {
+            // hash table: every key word starts at the empty sentinel ~0
+            uint32_t slots = (a.n_buckets - 2) >> 1;
+            for (uint32_t s2 = 0; s2 < slots; ++s2)
+                memcpy(&ap.init[a.counts_out - out_base + uint64_t(s2) * 16],
+                       &ones, 8);
         }
         for (uint64_t b = 0; b < a.n_buckets; ++b)
             for (uint32_t s = 0; s < a.n_sub; ++s) {
@@ -1782,9 +1810,28 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                         // non-nullable column: every matched doc has a value
                         r.terms_matched_docs = matched;
                     }
-                    for (uint32_t o = 0; o < a.n_buckets; ++o)
-                        if (counts[o])
-                            r.term_counts.emplace_back(f->dict_entry(o), counts[o]);
+                    if (a.kind == AGGD_TERMS_NUM) {
+                        r.key_kind = f->type == FastFieldView::U64   ? 1
+                                     : f->type == FastFieldView::F64 ? 3
+                                                                     : 2;
+                        uint32_t slots = (a.n_buckets - 2) >> 1;
+                        if (counts[2 * uint64_t(slots) + 1])
+                            throw std::runtime_error(
+                                "numeric terms hash table overflow");
+                        for (uint64_t s2 = 0; s2 < slots; ++s2) {
+                            uint64_t key = counts[2 * s2], c = counts[2 * s2 + 1];
+                            if (key != ~0ull && c)
+                                r.term_counts.emplace_back(num_term_key(key), c);
+                        }
+                        if (uint64_t sc2 = counts[2 * uint64_t(slots)])
+                            r.term_counts.emplace_back(num_term_key(~0ull), sc2);
+                        std::sort(r.term_counts.begin(), r.term_counts.end());
+                    } else {
+                        for (uint32_t o = 0; o < a.n_buckets; ++o)
+                            if (counts[o])
+                                r.term_counts.emplace_back(f->dict_entry(o),
+                                                           counts[o]);
+                    }
                     truncate_terms_split(
                         r, effective_split_size(d.size, d.split_size));
                 }

@@ -29,6 +29,7 @@
 #include <vector>
 
 #include "qast.h"
+#include "sortkey.h"  // u64_to_f64 / u64_to_i64 for numeric terms keys
 
 namespace qw {
 
@@ -118,7 +119,7 @@ struct IntermediateAggResults {
         std::string o;
         auto put = [&](const void* p, size_t n) { o.append((const char*)p, n); };
         uint32_t magic = 0x31474151;
-        uint16_t ver = 3, n = uint16_t(aggs.size());
+        uint16_t ver = 4, n = uint16_t(aggs.size());
         put(&magic, 4);
         put(&ver, 2);
         put(&n, 2);
@@ -141,6 +142,7 @@ struct IntermediateAggResults {
                 put(&a.metric.max, 8);
                 put(&a.metric.sum_sq, 8);
             } else if (a.kind == 3) {
+                put(&a.key_kind, 1);
                 put(&a.terms_matched_docs, 8);
                 put(&a.terms_error_bound, 8);
                 uint32_t ne = uint32_t(a.term_counts.size());
@@ -188,7 +190,7 @@ struct IntermediateAggResults {
         get(&magic, 4);
         get(&ver, 2);
         get(&n, 2);
-        if (magic != 0x31474151 || ver != 3) throw std::runtime_error("QAGG1: bad header");
+        if (magic != 0x31474151 || ver != 4) throw std::runtime_error("QAGG1: bad header");
         for (int i = 0; i < n; ++i) {
             AggResult a;
             uint16_t nl;
@@ -213,6 +215,7 @@ struct IntermediateAggResults {
                 get(&a.metric.max, 8);
                 get(&a.metric.sum_sq, 8);
             } else if (a.kind == 3) {
+                get(&a.key_kind, 1);
                 get(&a.terms_matched_docs, 8);
                 get(&a.terms_error_bound, 8);
                 uint32_t ne;
@@ -288,6 +291,8 @@ struct IntermediateAggResults {
                     }
                 }
                 a.term_counts = std::move(merged);
+                if (!a.key_kind) a.key_kind = b.key_kind;  // splits missing
+                                                           // the column
                 a.terms_matched_docs += b.terms_matched_docs;
                 a.terms_error_bound += b.terms_error_bound;
             } else {
@@ -490,7 +495,27 @@ inline std::string finalize_aggs_json(const IntermediateAggResults& ir,
                 o += buf;
                 shown_docs += ordered[b].second;
                 o += ",\"key\":";
-                mj::escape_to(o, ordered[b].first);
+                if (a.key_kind) {
+                    // numeric column key: decode the sortable big-endian
+                    // bits back to the column value; u64/i64 printed as JSON
+                    // integers to keep full 64-bit precision (the
+                    // high_prec_test golden pins this)
+                    uint64_t bits = num_term_key_bits(ordered[b].first);
+                    char nbuf[32];
+                    if (a.key_kind == 1)
+                        snprintf(nbuf, sizeof nbuf, "%llu",
+                                 (unsigned long long)bits);
+                    else if (a.key_kind == 2)
+                        snprintf(nbuf, sizeof nbuf, "%lld",
+                                 (long long)u64_to_i64(bits));
+                    else {
+                        mj::num_to(o, u64_to_f64(bits));
+                        nbuf[0] = 0;
+                    }
+                    o += nbuf;
+                } else {
+                    mj::escape_to(o, ordered[b].first);
+                }
                 o += "}";
             }
             o += "],\"doc_count_error_upper_bound\":";

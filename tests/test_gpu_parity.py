@@ -1017,3 +1017,78 @@ def test_bitmap_fallback_memoized(searchers):
     ms, n = gpu.kernel_stats("union_bm25")
     ms2, n2 = gpu.kernel_stats("range_filter")
     assert n + n2 == 1, (n, n2)
+
+
+# ------------------------------------------------- numeric terms aggregation
+def test_numeric_terms_aggs():
+    # terms keyed by numeric fast-column values (device hash table vs the
+    # oracle's map; keys must keep full 64-bit precision — the
+    # high_prec_test golden pins u64 values beyond 2^53)
+    import random
+    rng = random.Random(7)
+    docs = []
+    for i in range(3000):
+        d = {"timestamp": 1700000000 + i, "severity_text":
+             "INFO" if i % 3 else "ERROR", "body": "x",
+             "tenant_id": i % 5}
+        if rng.random() < 0.8:
+            d["big_u"] = 1769070189829214201 + rng.randrange(6)
+        d["neg_i"] = rng.choice([-40, -3, 0, 12])
+        d["val_f"] = rng.choice([-2.5, 0.0, 0.125, 3e9])
+        docs.append(d)
+    docs[17]["big_u"] = 2**64 - 1  # the hash-table sentinel value itself
+    schema = {"timestamp_field": "timestamp", "fields":
+              splitgen.HDFS_SCHEMA["fields"] +
+              [{"name": "big_u", "type": "u64", "fast": True},
+               {"name": "neg_i", "type": "i64", "fast": True},
+               {"name": "val_f", "type": "f64", "fast": True}]}
+    w = splitgen.SplitWriter(schema, "numt")
+    w.add_documents(docs)
+    data = w.finalize()
+    gpu, cpu = GpuSearcher(device=0), OracleSearcher()
+    gpu.add_split("numt", data)
+    cpu.add_split("numt", data)
+
+    aggs = {"by_u": {"terms": {"field": "big_u", "size": 10}},
+            "by_i": {"terms": {"field": "neg_i", "size": 10}},
+            "by_f": {"terms": {"field": "val_f", "size": 10}}}
+    for q in ({"type": "match_all"},
+              {"type": "term", "field": "severity_text", "value": "ERROR"}):
+        req = make_leaf_request(q, schema, [("numt", len(docs))], max_hits=0,
+                                aggregation=aggs)
+        g = gpu.leaf_search(req)
+        e = cpu.leaf_search(req)
+        gj = gpu.finalize_agg_json(g["intermediate_aggregation_result"], aggs)
+        ej = cpu.finalize_agg_json(e["intermediate_aggregation_result"], aggs)
+        assert gj == ej, (q, gj, ej)
+        keys = [b["key"] for b in gj["by_u"]["buckets"]]
+        assert all(isinstance(k, int) for k in keys)  # full u64 precision
+        if q["type"] == "match_all":
+            assert 2**64 - 1 in keys
+            assert sum(b["doc_count"] for b in gj["by_i"]["buckets"]) == \
+                len(docs)
+
+
+def test_numeric_terms_split_size_truncation():
+    # split_size truncation + doc_count_error_upper_bound over numeric keys
+    docs = [{"timestamp": 1700000000 + i, "severity_text": "INFO",
+             "body": "x", "tenant_id": i % 7} for i in range(700)]
+    schema = splitgen.HDFS_SCHEMA
+    w = splitgen.SplitWriter(schema, "numtr")
+    w.add_documents(docs)
+    data = w.finalize()
+    gpu, cpu = GpuSearcher(device=0), OracleSearcher()
+    gpu.add_split("numtr", data)
+    cpu.add_split("numtr", data)
+    aggs = {"t": {"terms": {"field": "tenant_id", "size": 3,
+                            "split_size": 4}}}
+    req = make_leaf_request({"type": "match_all"}, schema,
+                            [("numtr", len(docs))], max_hits=0,
+                            aggregation=aggs)
+    g = gpu.leaf_search(req)
+    e = cpu.leaf_search(req)
+    gj = gpu.finalize_agg_json(g["intermediate_aggregation_result"], aggs)
+    ej = cpu.finalize_agg_json(e["intermediate_aggregation_result"], aggs)
+    assert gj == ej
+    assert len(gj["t"]["buckets"]) == 3
+    assert gj["t"]["doc_count_error_upper_bound"] > 0

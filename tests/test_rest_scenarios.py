@@ -28,8 +28,7 @@ UNSUPPORTED_AGGS = {"percentiles": "percentiles: later round",
                     "max": None, "value_count": None,
                     "date_histogram": None, "histogram": None, "terms": None}
 MULTI_VALUED_FIELDS = set()  # multi-valued str fast columns now supported
-NUMERIC_TERMS_FIELDS = {"high_prec_test"}  # terms over numeric fast
-                                           # columns: later round
+NUMERIC_TERMS_FIELDS = set()  # terms over numeric fast columns supported
 
 
 MIXED_TYPE_SORT_FIELDS = {"mixed_type"}  # one dynamic field, several typed
