@@ -836,7 +836,8 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
         }
         if (threadIdx.x < q.n_aggs && threadIdx.x < 4) {
             const AggDev& a = aggs[threadIdx.x];
-            if (a.kind == AGGD_TERMS && sc_agg_matched[threadIdx.x])
+            if ((a.kind == AGGD_TERMS || a.kind == AGGD_TERMS_NUM) &&
+                sc_agg_matched[threadIdx.x])
                 atomicAdd((unsigned long long*)(q.results + a.matched_out),
                           (unsigned long long)sc_agg_matched[threadIdx.x]);
         }
