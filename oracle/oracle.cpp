@@ -629,6 +629,72 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
             out.aggs.push_back(std::move(a));
             continue;
         }
+        if (d.kind == AggDef::COMPOSITE) {
+            // composite: one canonical byte-encoded tuple per doc (null /
+            // str / f64 components — qagg_format.h comp_encode_*), counted
+            // in a map whose order IS composite-key order (mirrors the
+            // product's packed-u64 hash table + assembly decode)
+            a.kind = 3;
+            a.key_kind = 9;
+            bool all_present = true;
+            std::vector<const FastFieldView*> srcs;
+            for (const CompSource& cs : d.comp) {
+                const FastFieldView* sf = sv.fast_field(cs.field);
+                if (!sf) { all_present = false; continue; }
+                if (!cs.is_histo && sf->type != FastFieldView::STR)
+                    throw std::runtime_error(
+                        "composite terms source over a numeric column "
+                        "(r1 limit)");
+                if (cs.is_histo && sf->type == FastFieldView::STR)
+                    throw std::runtime_error(
+                        "composite histogram source over a str column");
+                if (sf->multi)
+                    throw std::runtime_error(
+                        "composite source over a multi-valued column "
+                        "(r1 limit)");
+                srcs.push_back(sf);
+            }
+            if (all_present) {
+                if (sv.num_docs > (1u << 21))
+                    throw std::runtime_error(
+                        "composite aggregation on a >2M-doc split (r1 limit)");
+                std::map<std::string, uint64_t> counts;
+                auto cvisit = [&](uint32_t doc) {
+                    std::string ck;
+                    for (size_t si = 0; si < d.comp.size(); ++si) {
+                        const CompSource& cs = d.comp[si];
+                        const FastFieldView* sf = srcs[si];
+                        if (!sf->present(doc)) {
+                            if (!cs.missing_bucket) return;
+                            comp_encode_null(ck);
+                        } else if (cs.is_histo) {
+                            double v;
+                            if (sf->type == FastFieldView::U64)
+                                v = double(sf->u64(doc));
+                            else if (sf->type == FastFieldView::F64)
+                                v = sf->f64(doc);
+                            else v = double(sf->i64(doc));
+                            double key = std::floor((v - cs.offset) /
+                                                    cs.interval) *
+                                             cs.interval +
+                                         cs.offset;
+                            comp_encode_f64(ck, key);
+                        } else {
+                            comp_encode_str(ck, sf->dict_entry(sf->ord(doc)));
+                        }
+                    }
+                    counts[ck]++;
+                };
+                if (m.all)
+                    for (uint32_t doc = 0; doc < sv.num_docs; ++doc) cvisit(doc);
+                else
+                    for (uint32_t doc : m.docs) cvisit(doc);
+                for (auto& kv : counts)
+                    a.term_counts.emplace_back(kv.first, kv.second);
+            }
+            out.aggs.push_back(std::move(a));
+            continue;
+        }
         if (d.kind == AggDef::RANGE) {
             a.kind = 4;
             if (f && f->type != FastFieldView::STR && !f->multi) {

@@ -63,7 +63,12 @@ enum AggKindDev : uint32_t {
     // hash table in the counts_out region (slots = (n_buckets-2)/2, power of
     // two; trailing two words = count for the key ~0 sentinel value +
     // overflow flag), keyed by the value's order-preserving sortable bits
-    AGGD_TERMS_NUM = 4
+    AGGD_TERMS_NUM = 4,
+    // composite aggregation: same hash-table layout, keyed by the per-source
+    // components packed MSB-first into <=63 bits (so the key can never hit
+    // the ~0 sentinel); per-source params in the c_* fields + the sub_*
+    // column arrays
+    AGGD_COMP = 5
 };
 constexpr uint32_t AGG_MAX_RANGES = 16;
 
@@ -100,10 +105,18 @@ struct AggDev {
     // AGGD_METRIC uses ONE such slot at counts_out.
     uint32_t n_sub;
     uint64_t sub_out;        // byte offset: n_buckets * n_sub * 32 bytes
-    uint64_t sub_values_off[4];  // sub-agg source columns (<=4)
+    uint64_t sub_values_off[4];  // sub-agg source columns (<=4);
+                                 // AGGD_COMP: the composite SOURCE columns
     uint64_t sub_nulls_off[4];
     uint32_t sub_width[4];
     uint32_t sub_is_i64[4];
+    // AGGD_COMP per-source packing: component = ord+miss (terms) or
+    // bucket_idx-base+miss (histogram), shifted into the 63-bit key
+    uint32_t c_shift[4];
+    uint32_t c_histo;        // bitmask: source i is a histogram
+    uint32_t c_missing;      // bitmask: missing_bucket (else null doc drops)
+    double c_interval[4], c_offset[4];
+    int64_t c_base[4];       // histogram: floor((col_min-offset)/interval)
 };
 
 struct QueryDev {

@@ -691,6 +691,63 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                                                                     a.matched_out),
                                               1ull);
                             }
+                        } else if (a.kind == AGGD_COMP) {
+                            // composite: pack per-source components into a
+                            // <=63-bit key (ordering == composite tuple asc),
+                            // count via the same hash table as TERMS_NUM
+                            if (!a.n_buckets) continue;
+                            uint64_t key = 0;
+                            bool okc = true;
+                            for (uint32_t si = 0; si < a.n_sub; ++si) {
+                                uint64_t comp = 0;
+                                bool present = true;
+                                if (a.sub_nulls_off[si]) {
+                                    const uint64_t* nu =
+                                        (const uint64_t*)(q.split +
+                                                          a.sub_nulls_off[si]);
+                                    present = (nu[d >> 6] >> (d & 63)) & 1;
+                                }
+                                uint64_t miss = (a.c_missing >> si) & 1;
+                                if (!present) {
+                                    if (!miss) { okc = false; break; }
+                                } else if ((a.c_histo >> si) & 1) {
+                                    double v = agg_value(q, a.sub_values_off[si],
+                                                         a.sub_width[si],
+                                                         a.sub_is_i64[si], d);
+                                    int64_t idx =
+                                        int64_t(floor((v - a.c_offset[si]) /
+                                                      a.c_interval[si])) -
+                                        a.c_base[si];
+                                    comp = uint64_t(idx) + miss;
+                                } else {
+                                    comp = agg_ord(q, a.sub_values_off[si],
+                                                   a.sub_width[si], d) + miss;
+                                }
+                                key |= comp << a.c_shift[si];
+                            }
+                            if (!okc) continue;
+                            unsigned long long* tab =
+                                (unsigned long long*)(q.results + a.counts_out);
+                            uint32_t slots = (a.n_buckets - 2) >> 1;
+                            const unsigned long long SENT = ~0ull;
+                            uint64_t h = key * 0x9E3779B97F4A7C15ull;
+                            h ^= h >> 32;
+                            uint32_t sl = uint32_t(h) & (slots - 1);
+                            bool done = false;
+                            for (uint32_t pr = 0; pr < slots; ++pr) {
+                                unsigned long long cur = tab[2 * sl];
+                                if (cur == SENT)
+                                    cur = atomicCAS(&tab[2 * sl], SENT,
+                                                    (unsigned long long)key);
+                                if (cur == SENT ||
+                                    cur == (unsigned long long)key) {
+                                    atomicAdd(&tab[2 * sl + 1], 1ull);
+                                    done = true;
+                                    break;
+                                }
+                                sl = (sl + 1) & (slots - 1);
+                            }
+                            if (!done) tab[2 * slots + 1] = 1ull;  // overflow
                         } else if (a.kind == AGGD_METRIC) {
                             if (!a.values_off) continue;
                             double v = agg_value(q, a.values_off, a.value_width,

@@ -662,9 +662,24 @@ struct AggPlan {
     std::vector<AggDef> defs;
     std::vector<AggDev> devs;            // one per def
     std::vector<const FastFieldView*> fields;
+    // composite aggs: per-source column views (null for other kinds)
+    std::vector<std::array<const FastFieldView*, 4>> comp_fields;
     size_t out_bytes = 0;                // total result bytes (counts + subs)
     std::vector<uint8_t> init;           // initial contents of the out region
 };
+
+static void col_min_max(const FastFieldView& f, double* mn, double* mx) {
+    if (f.type == FastFieldView::U64) {
+        *mn = double(uint64_t(f.min_value));
+        *mx = double(uint64_t(f.max_value));
+    } else if (f.type == FastFieldView::F64) {
+        *mn = f.fmin;
+        *mx = f.fmax;
+    } else {
+        *mn = double(f.min_value);
+        *mx = double(f.max_value);
+    }
+}
 
 static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
                          int64_t bucket_limit, uint64_t out_base) {
@@ -674,9 +689,79 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
     for (const AggDef& d : ap.defs) {
         const FastFieldView* f = sv.fast_field(d.field);
         ap.fields.push_back(f);
+        ap.comp_fields.push_back({nullptr, nullptr, nullptr, nullptr});
         AggDev a{};
         a.lds_slot = 0xFF;
-        if (d.kind == AggDef::TERMS) {
+        if (d.kind == AggDef::COMPOSITE) {
+            a.kind = AGGD_COMP;
+            bool all_present = true;
+            for (const CompSource& cs : d.comp) {
+                const FastFieldView* sf = sv.fast_field(cs.field);
+                if (!sf) { all_present = false; continue; }
+                if (!cs.is_histo && sf->type != FastFieldView::STR)
+                    throw std::runtime_error(
+                        "composite terms source over a numeric column "
+                        "(r1 limit)");
+                if (cs.is_histo && sf->type == FastFieldView::STR)
+                    throw std::runtime_error(
+                        "composite histogram source over a str column");
+                if (sf->multi)
+                    throw std::runtime_error(
+                        "composite source over a multi-valued column "
+                        "(r1 limit)");
+            }
+            if (all_present) {
+                if (sv.num_docs > (1u << 21))
+                    throw std::runtime_error(
+                        "composite aggregation on a >2M-doc split (r1 limit)");
+                // pack sources MSB-first; component value count per source
+                // bounds its bit width (comp_key_bits in qagg_format.h keeps
+                // assembly decode in lockstep)
+                uint32_t acc = 0;
+                for (int si = int(d.comp.size()) - 1; si >= 0; --si) {
+                    const CompSource& cs = d.comp[si];
+                    const FastFieldView* sf = sv.fast_field(cs.field);
+                    ap.comp_fields.back()[si] = sf;
+                    uint64_t miss = cs.missing_bucket ? 1 : 0;
+                    uint64_t n_values;
+                    if (cs.is_histo) {
+                        double mn, mx;
+                        col_min_max(*sf, &mn, &mx);
+                        int64_t b0 = int64_t(
+                            std::floor((mn - cs.offset) / cs.interval));
+                        int64_t b1 = int64_t(
+                            std::floor((mx - cs.offset) / cs.interval));
+                        a.c_base[si] = b0;
+                        a.c_interval[si] = cs.interval;
+                        a.c_offset[si] = cs.offset;
+                        a.c_histo |= 1u << si;
+                        n_values = uint64_t(b1 - b0 + 1) + miss;
+                    } else {
+                        n_values = uint64_t(sf->cardinality) + miss;
+                    }
+                    if (miss) a.c_missing |= 1u << si;
+                    a.sub_values_off[si] = sf->values.off;
+                    a.sub_nulls_off[si] = sf->nullable ? sf->nulls.off : 0;
+                    a.sub_width[si] = cs.is_histo ? 8 : uint32_t(sf->ord_width);
+                    a.sub_is_i64[si] =
+                        sf->type == FastFieldView::U64   ? 0
+                        : sf->type == FastFieldView::F64 ? 2
+                                                         : 1;
+                    a.c_shift[si] = acc;
+                    acc += comp_key_bits(n_values);
+                }
+                if (acc > 63)
+                    throw std::runtime_error(
+                        "composite key exceeds 63 bits (r1 limit)");
+                a.n_sub = uint32_t(d.comp.size());
+                uint64_t want = uint64_t(sv.num_docs) * 2;
+                uint32_t slots = 1024;
+                while (slots < want) slots <<= 1;
+                a.n_buckets = slots * 2 + 2;
+            } else {
+                a.n_buckets = 0;  // a source column is absent: empty result
+            }
+        } else if (d.kind == AggDef::TERMS) {
             a.kind = AGGD_TERMS;
             if (f && f->type == FastFieldView::STR) {
                 a.n_buckets = f->cardinality;
@@ -818,7 +903,9 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
         a.matched_out = off;
         off += 8;
         a.sub_out = off;
-        off += uint64_t(a.n_buckets) * a.n_sub * 40;
+        // AGGD_COMP uses n_sub for its sources and n_buckets for the hash
+        // table words — no per-bucket stats region
+        if (a.kind != AGGD_COMP) off += uint64_t(a.n_buckets) * a.n_sub * 40;
         ap.devs.push_back(a);
     }
     // LDS slot assignment: the first histogram agg that fits gets the LDS
@@ -850,7 +937,7 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
             uint64_t slot = a.counts_out - out_base;
             memcpy(&ap.init[slot + 16], &ones, 8);  // min: max sortable
         }
-        if (a.kind == AGGD_TERMS_NUM && a.n_buckets) {
+        if ((a.kind == AGGD_TERMS_NUM || a.kind == AGGD_COMP) && a.n_buckets) {
             // hash table: every key word starts at the empty sentinel ~0
             uint32_t slots = (a.n_buckets - 2) >> 1;
             for (uint32_t s2 = 0; s2 < slots; ++s2)
@@ -1834,6 +1921,67 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     }
                     truncate_terms_split(
                         r, effective_split_size(d.size, d.split_size));
+                }
+            } else if (d.kind == AggDef::COMPOSITE) {
+                // decode packed 63-bit keys into the canonical per-source
+                // byte encoding (split-local ords -> term strings) so the
+                // cross-split QAGG merge compares composite tuples directly
+                r.kind = 3;
+                r.key_kind = 9;
+                if (a.n_buckets) {
+                    uint32_t slots = (a.n_buckets - 2) >> 1;
+                    if (counts[2 * uint64_t(slots) + 1])
+                        throw std::runtime_error(
+                            "composite hash table overflow");
+                    // recompute per-source widths exactly as plan_aggs did
+                    uint32_t width[4] = {0, 0, 0, 0};
+                    for (size_t si = 0; si < d.comp.size(); ++si) {
+                        const CompSource& cs = d.comp[si];
+                        const FastFieldView* sf = ap.comp_fields[i][si];
+                        uint64_t miss = cs.missing_bucket ? 1 : 0;
+                        uint64_t nv;
+                        if (cs.is_histo) {
+                            double mn, mx;
+                            col_min_max(*sf, &mn, &mx);
+                            int64_t b0 = int64_t(
+                                std::floor((mn - cs.offset) / cs.interval));
+                            int64_t b1 = int64_t(
+                                std::floor((mx - cs.offset) / cs.interval));
+                            nv = uint64_t(b1 - b0 + 1) + miss;
+                        } else {
+                            nv = uint64_t(sf->cardinality) + miss;
+                        }
+                        width[si] = comp_key_bits(nv);
+                    }
+                    auto decode_key = [&](uint64_t key) {
+                        std::string ck;
+                        for (size_t si = 0; si < d.comp.size(); ++si) {
+                            const CompSource& cs = d.comp[si];
+                            uint64_t comp = (key >> a.c_shift[si]) &
+                                            ((1ull << width[si]) - 1);
+                            uint64_t miss = cs.missing_bucket ? 1 : 0;
+                            if (miss && comp == 0) {
+                                comp_encode_null(ck);
+                            } else if (cs.is_histo) {
+                                double v = double(a.c_base[si] +
+                                                  int64_t(comp - miss)) *
+                                               cs.interval +
+                                           cs.offset;
+                                comp_encode_f64(ck, v);
+                            } else {
+                                comp_encode_str(
+                                    ck, ap.comp_fields[i][si]->dict_entry(
+                                            uint32_t(comp - miss)));
+                            }
+                        }
+                        return ck;
+                    };
+                    for (uint64_t s2 = 0; s2 < slots; ++s2) {
+                        uint64_t key = counts[2 * s2], c = counts[2 * s2 + 1];
+                        if (key != ~0ull && c)
+                            r.term_counts.emplace_back(decode_key(key), c);
+                    }
+                    std::sort(r.term_counts.begin(), r.term_counts.end());
                 }
             } else if (d.kind == AggDef::METRIC) {
                 r.kind = 5;

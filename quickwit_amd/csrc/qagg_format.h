@@ -88,6 +88,28 @@ inline uint64_t num_term_key_bits(const std::string& k) {
     return v;
 }
 
+// composite-source component bit width for n distinct component values
+// (>=1 bit; shared by the device-key packer and the assembly decoder)
+inline uint32_t comp_key_bits(uint64_t n_values) {
+    uint32_t b = 1;
+    while ((1ull << b) < n_values) ++b;
+    return b;
+}
+
+// canonical order-preserving byte encoding of one composite key component
+// (concatenation of per-source encodings compares as the composite tuple:
+// null sorts first, str by bytes, numbers by sortable-bits order)
+inline void comp_encode_null(std::string& o) { o += '\x00'; }
+inline void comp_encode_str(std::string& o, const std::string& s) {
+    o += '\x01';
+    o += s;
+    o += '\x00';
+}
+inline void comp_encode_f64(std::string& o, double v) {
+    o += '\x04';
+    o += num_term_key(f64_to_u64(v));
+}
+
 // per-split terms truncation (tantivy terms agg split_size): keep the top
 // `split_size` entries by (count desc, key asc); when anything is dropped,
 // the last included count joins the error bound. Restores key order after.
@@ -473,6 +495,76 @@ inline std::string finalize_aggs_json(const IntermediateAggResults& ir,
                 o += "}";
             }
             o += "]}";
+            continue;
+        }
+        if (d.kind == AggDef::COMPOSITE) {
+            // buckets in composite-key order (the canonical byte encoding);
+            // `after` resumes strictly past the given tuple; page of `size`
+            // buckets + after_key = the last emitted key
+            std::string after;
+            if (d.has_after)
+                for (const CompSource& cs : d.comp) {
+                    if (cs.after_kind == 1) comp_encode_null(after);
+                    else if (cs.after_kind == 2) comp_encode_str(after, cs.after_s);
+                    else comp_encode_f64(after, cs.after_n);
+                }
+            auto key_obj = [&](const std::string& ck, std::string& out) {
+                out += "{";
+                size_t pos = 0;
+                for (size_t si = 0; si < d.comp.size(); ++si) {
+                    if (si) out += ",";
+                    mj::escape_to(out, d.comp[si].name);
+                    out += ":";
+                    uint8_t tag = uint8_t(ck[pos++]);
+                    if (tag == 0) {
+                        out += "null";
+                    } else if (tag == 1) {
+                        size_t end = ck.find('\0', pos);
+                        mj::escape_to(out, ck.substr(pos, end - pos));
+                        pos = end + 1;
+                    } else {
+                        uint64_t bits = num_term_key_bits(ck.substr(pos, 8));
+                        pos += 8;
+                        if (tag == 2) {
+                            char nb[32];
+                            snprintf(nb, sizeof nb, "%llu",
+                                     (unsigned long long)bits);
+                            out += nb;
+                        } else if (tag == 3) {
+                            char nb[32];
+                            snprintf(nb, sizeof nb, "%lld",
+                                     (long long)u64_to_i64(bits));
+                            out += nb;
+                        } else {
+                            mj::num_to(out, u64_to_f64(bits));
+                        }
+                    }
+                }
+                out += "}";
+            };
+            std::string buckets, last_key;
+            size_t shown = 0;
+            for (auto& kv : a.term_counts) {
+                if (d.has_after && kv.first <= after) continue;
+                if (shown == d.size) break;
+                if (shown) buckets += ",";
+                buckets += "{\"doc_count\":";
+                char buf[24];
+                snprintf(buf, sizeof buf, "%llu", (unsigned long long)kv.second);
+                buckets += buf;
+                buckets += ",\"key\":";
+                key_obj(kv.first, buckets);
+                buckets += "}";
+                last_key = kv.first;
+                ++shown;
+            }
+            o += "{";
+            if (shown) {
+                o += "\"after_key\":";
+                key_obj(last_key, o);
+                o += ",";
+            }
+            o += "\"buckets\":[" + buckets + "]}";
             continue;
         }
         if (d.kind == AggDef::TERMS) {

@@ -869,11 +869,29 @@ struct RangeSpec {
     double from = 0, to = 0;  // [from, to) — ES range-agg semantics
 };
 
+// one source of a composite aggregation (ES composite agg: a flattened
+// cross-product of per-source buckets, paginated by composite key;
+// tantivy supports terms + histogram sources)
+struct CompSource {
+    std::string name;
+    std::string field;
+    bool is_histo = false;
+    double interval = 0, offset = 0;  // histogram source
+    bool missing_bucket = false;      // terms source: null bucket, sorts first
+    // `after` component for this source (resume strictly after this key):
+    // kind 0 = none, 1 = null, 2 = str, 3 = number
+    int after_kind = 0;
+    std::string after_s;
+    double after_n = 0;
+};
+
 struct AggDef {
     std::string name;
-    enum Kind { DATE_HISTOGRAM, HISTOGRAM, TERMS, RANGE, METRIC } kind =
-        DATE_HISTOGRAM;
+    enum Kind { DATE_HISTOGRAM, HISTOGRAM, TERMS, RANGE, METRIC, COMPOSITE }
+        kind = DATE_HISTOGRAM;
     std::string field;
+    std::vector<CompSource> comp;   // COMPOSITE
+    bool has_after = false;         // COMPOSITE: `after` given
     std::vector<RangeSpec> ranges;  // RANGE
     MetricAgg metric;               // METRIC (top-level stats/avg/... agg)
     double interval = 0;  // ms for date_histogram; raw units for histogram
@@ -958,6 +976,67 @@ inline std::vector<AggDef> parse_agg_request(const std::string& json) {
                     r.key = fmt(r.has_from, r.from) + "-" + fmt(r.has_to, r.to);
                 }
                 a.ranges.push_back(std::move(r));
+            }
+        } else if ((spec = body->get("composite"))) {
+            // ES composite aggregation: sources = terms | histogram, paged
+            // by composite key asc with optional `after` cursor; keys typed
+            // "str:x" / "f64:1.5" in quickwit's after serialization
+            a.kind = AggDef::COMPOSITE;
+            a.size = 10;
+            if (const mj::Value* sz = spec->get("size"))
+                a.size = uint32_t(sz->as_i64());
+            for (auto& sv2 : spec->at("sources")->arr) {
+                if (sv2->obj.size() != 1)
+                    throw std::runtime_error("composite source must have one name");
+                CompSource cs;
+                cs.name = sv2->obj.begin()->first;
+                const mj::Value* sb = sv2->obj.begin()->second.get();
+                const mj::Value* ss2 = nullptr;
+                if ((ss2 = sb->get("terms"))) {
+                    cs.is_histo = false;
+                    if (const mj::Value* mb = ss2->get("missing_bucket"))
+                        cs.missing_bucket = mb->b;
+                } else if ((ss2 = sb->get("histogram"))) {
+                    cs.is_histo = true;
+                    cs.interval = ss2->at("interval")->num();
+                    if (const mj::Value* off = ss2->get("offset"))
+                        cs.offset = off->num();
+                } else {
+                    throw std::runtime_error(
+                        "composite source must be terms or histogram");
+                }
+                cs.field = ss2->at("field")->s;
+                a.comp.push_back(std::move(cs));
+            }
+            if (a.comp.empty() || a.comp.size() > 4)
+                throw std::runtime_error("composite: 1..4 sources supported");
+            if (const mj::Value* af = spec->get("after")) {
+                a.has_after = true;
+                for (CompSource& cs : a.comp) {
+                    const mj::Value* v = af->get(cs.name);
+                    if (!v) throw std::runtime_error(
+                        "composite after missing source " + cs.name);
+                    if (v->kind == mj::Value::NUL) cs.after_kind = 1;
+                    else if (v->kind == mj::Value::STR) {
+                        std::string s = v->s;
+                        // typed after-key serialization: str:/f64:/i64:/u64:
+                        if (s.rfind("str:", 0) == 0) {
+                            cs.after_kind = 2;
+                            cs.after_s = s.substr(4);
+                        } else if (s.rfind("f64:", 0) == 0 ||
+                                   s.rfind("i64:", 0) == 0 ||
+                                   s.rfind("u64:", 0) == 0) {
+                            cs.after_kind = 3;
+                            cs.after_n = atof(s.c_str() + 4);
+                        } else {
+                            cs.after_kind = 2;
+                            cs.after_s = s;
+                        }
+                    } else {
+                        cs.after_kind = 3;
+                        cs.after_n = v->num();
+                    }
+                }
             }
         } else if ((spec = body->get("terms"))) {
             a.kind = AggDef::TERMS;

@@ -1092,3 +1092,74 @@ def test_numeric_terms_split_size_truncation():
     assert gj == ej
     assert len(gj["t"]["buckets"]) == 3
     assert gj["t"]["doc_count_error_upper_bound"] > 0
+
+
+# ------------------------------------------------- composite aggregation
+def test_composite_agg_parity():
+    # terms + histogram sources with missing_bucket and after-key pagination
+    # (the aggregations scenario's host/name/response composite), two splits
+    # so the canonical-key merge crosses split-local ord spaces
+    import random
+    rng = random.Random(3)
+    hosts = [None, "10.0.0.1", "10.0.0.2", "10.0.0.3"]
+    names = ["ann", "bob", "cat", "dan", "eve"]
+    schema = {"timestamp_field": "timestamp", "fields":
+              splitgen.HDFS_SCHEMA["fields"] +
+              [{"name": "host", "type": "str", "fast": True},
+               {"name": "user", "type": "str", "fast": True},
+               {"name": "resp", "type": "i64", "fast": True}]}
+    gpu, cpu = GpuSearcher(device=0), OracleSearcher()
+    splits = []
+    for s in range(2):
+        docs = []
+        for i in range(400):
+            d = {"timestamp": 1700000000 + i, "severity_text": "INFO",
+                 "body": "x", "tenant_id": i % 3,
+                 "user": rng.choice(names), "resp": rng.choice([0, 30, 100, 120])}
+            h = rng.choice(hosts)
+            if h is not None:
+                d["host"] = h
+            docs.append(d)
+        w = splitgen.SplitWriter(schema, f"comp-{s}")
+        w.add_documents(docs)
+        data = w.finalize()
+        gpu.add_split(f"comp-{s}", data)
+        cpu.add_split(f"comp-{s}", data)
+        splits.append((f"comp-{s}", len(docs)))
+
+    def both(aggs):
+        req = make_leaf_request({"type": "match_all"}, schema, splits,
+                                max_hits=0, aggregation=aggs)
+        g = gpu.leaf_search(req)
+        e = cpu.leaf_search(req)
+        gj = gpu.finalize_agg_json(g["intermediate_aggregation_result"], aggs)
+        ej = cpu.finalize_agg_json(e["intermediate_aggregation_result"], aggs)
+        assert gj == ej, (gj, ej)
+        return gj
+
+    base = {"sources": [
+        {"h": {"terms": {"field": "host", "missing_bucket": True}}},
+        {"u": {"terms": {"field": "user"}}},
+        {"r": {"histogram": {"field": "resp", "interval": 50}}}]}
+    aggs = {"c": {"composite": dict(base, size=7)}}
+    page1 = both(aggs)["c"]
+    assert len(page1["buckets"]) == 7
+    assert page1["buckets"][0]["key"]["h"] is None  # missing bucket first
+    keys = [tuple((b["key"][k] is None, b["key"][k] or 0 if k == "r" else
+                   str(b["key"][k])) for k in ("h", "u", "r"))
+            for b in page1["buckets"]]
+    assert keys == sorted(keys)
+    # paginate with after (typed key serialization like quickwit's)
+    ak = page1["after_key"]
+    assert ak == page1["buckets"][-1]["key"]
+    after = {"h": "str:" + ak["h"] if ak["h"] is not None else None,
+             "u": "str:" + ak["u"], "r": f'f64:{ak["r"]}'}
+    aggs2 = {"c": {"composite": dict(base, size=500, after=after)}}
+    page2 = both(aggs2)["c"]
+    # pages partition the full bucket space
+    aggs_all = {"c": {"composite": dict(base, size=1000)}}
+    full = both(aggs_all)["c"]
+    got = [json.dumps(b) for b in page1["buckets"] + page2["buckets"]]
+    assert got == [json.dumps(b) for b in full["buckets"]]
+    total = sum(b["doc_count"] for b in full["buckets"])
+    assert total == 800  # every doc lands somewhere (h missing-bucketed)

@@ -22,7 +22,7 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 UNSUPPORTED_AGGS = {"percentiles": "percentiles: later round",
                     "cardinality": "cardinality (hll): later round",
                     "extended_stats": None,
-                    "composite": "composite agg: later round",
+                    "composite": None,
                     "range": None,
                     "avg": None, "stats": None, "sum": None, "min": None,
                     "max": None, "value_count": None,
