@@ -544,7 +544,7 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
                 default: *v = double(sf->i64(doc)); return true;
             }
         };
-        if (d.kind == AggDef::TERMS) {
+        if (d.kind == AggDef::TERMS || d.kind == AggDef::CARDINALITY) {
             a.kind = 3;
             if (f && f->type == FastFieldView::STR) {
                 std::vector<uint64_t> counts(f->cardinality, 0);
@@ -570,7 +570,9 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
                     if (counts[o]) a.term_counts.emplace_back(f->dict_entry(o), counts[o]);
                 // dict order is lexicographic => term_counts sorted by key;
                 // per-split split_size truncation + error bound (qagg_format.h)
-                truncate_terms_split(a, effective_split_size(d.size, d.split_size));
+                if (d.kind == AggDef::TERMS)
+                    truncate_terms_split(a,
+                                         effective_split_size(d.size, d.split_size));
             } else if (f && !f->multi) {
                 // terms over a numeric fast column: count by the value's
                 // order-preserving sortable bits, keys encoded big-endian so
@@ -601,7 +603,9 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
                     for (uint32_t doc : m.docs) nvisit(doc);
                 for (auto& kv : counts)
                     a.term_counts.emplace_back(num_term_key(kv.first), kv.second);
-                truncate_terms_split(a, effective_split_size(d.size, d.split_size));
+                if (d.kind == AggDef::TERMS)
+                    truncate_terms_split(a,
+                                         effective_split_size(d.size, d.split_size));
             }
             out.aggs.push_back(std::move(a));
             continue;

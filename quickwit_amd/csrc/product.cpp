@@ -761,7 +761,7 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
             } else {
                 a.n_buckets = 0;  // a source column is absent: empty result
             }
-        } else if (d.kind == AggDef::TERMS) {
+        } else if (d.kind == AggDef::TERMS || d.kind == AggDef::CARDINALITY) {
             a.kind = AGGD_TERMS;
             if (f && f->type == FastFieldView::STR) {
                 a.n_buckets = f->cardinality;
@@ -1886,7 +1886,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             for (auto& s : d.sub) r.sub_names.push_back(s.name);
             const uint8_t* base = agg_out.data() + (a.counts_out - r_agg);
             const uint64_t* counts = (const uint64_t*)base;
-            if (d.kind == AggDef::TERMS) {
+            if (d.kind == AggDef::TERMS || d.kind == AggDef::CARDINALITY) {
                 r.kind = 3;
                 if (a.n_buckets) {
                     if (a.nulls_off || a.offsets_off) {
@@ -1919,8 +1919,9 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                                 r.term_counts.emplace_back(f->dict_entry(o),
                                                            counts[o]);
                     }
-                    truncate_terms_split(
-                        r, effective_split_size(d.size, d.split_size));
+                    if (d.kind == AggDef::TERMS)
+                        truncate_terms_split(
+                            r, effective_split_size(d.size, d.split_size));
                 }
             } else if (d.kind == AggDef::COMPOSITE) {
                 // decode packed 63-bit keys into the canonical per-source

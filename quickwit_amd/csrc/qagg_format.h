@@ -497,6 +497,14 @@ inline std::string finalize_aggs_json(const IntermediateAggResults& ir,
             o += "]}";
             continue;
         }
+        if (d.kind == AggDef::CARDINALITY) {
+            // exact distinct count of the merged key set (deviation from the
+            // reference's HLL++ sketch — DESIGN.md §7)
+            o += "{\"value\":";
+            mj::num_to(o, double(a.term_counts.size()));
+            o += "}";
+            continue;
+        }
         if (d.kind == AggDef::COMPOSITE) {
             // buckets in composite-key order (the canonical byte encoding);
             // `after` resumes strictly past the given tuple; page of `size`

@@ -887,8 +887,8 @@ struct CompSource {
 
 struct AggDef {
     std::string name;
-    enum Kind { DATE_HISTOGRAM, HISTOGRAM, TERMS, RANGE, METRIC, COMPOSITE }
-        kind = DATE_HISTOGRAM;
+    enum Kind { DATE_HISTOGRAM, HISTOGRAM, TERMS, RANGE, METRIC, COMPOSITE,
+                CARDINALITY } kind = DATE_HISTOGRAM;
     std::string field;
     std::vector<CompSource> comp;   // COMPOSITE
     bool has_after = false;         // COMPOSITE: `after` given
@@ -1038,6 +1038,14 @@ inline std::vector<AggDef> parse_agg_request(const std::string& json) {
                     }
                 }
             }
+        } else if ((spec = body->get("cardinality"))) {
+            // exact distinct count via the terms machinery (merged distinct
+            // key set). DEVIATION from the reference's HyperLogLog++ sketch:
+            // exact answers, matching ES/golden outputs wherever the sketch
+            // is exact (small cardinalities); bounded by the per-split
+            // distinct-set guards. DESIGN.md §7.
+            a.kind = AggDef::CARDINALITY;
+            a.field = spec->at("field")->s;
         } else if ((spec = body->get("terms"))) {
             a.kind = AggDef::TERMS;
             a.field = spec->at("field")->s;

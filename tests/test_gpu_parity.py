@@ -1163,3 +1163,47 @@ def test_composite_agg_parity():
     assert got == [json.dumps(b) for b in full["buckets"]]
     total = sum(b["doc_count"] for b in full["buckets"])
     assert total == 800  # every doc lands somewhere (h missing-bucketed)
+
+
+def test_cardinality_aggs():
+    # exact distinct counts over str / numeric / nullable columns (declared
+    # deviation from the reference's HLL++ sketch: exact values, which equal
+    # the golden's outputs at scenario cardinalities)
+    import random
+    rng = random.Random(13)
+    docs = []
+    for i in range(2500):
+        d = {"timestamp": 1700000000 + (i % 97), "severity_text":
+             rng.choice(["INFO", "WARN", "ERROR"]), "body": "x",
+             "tenant_id": i % 41}
+        if rng.random() < 0.7:
+            d["opt_u"] = rng.randrange(29)
+        docs.append(d)
+    schema = {"timestamp_field": "timestamp", "fields":
+              splitgen.HDFS_SCHEMA["fields"] +
+              [{"name": "opt_u", "type": "u64", "fast": True}]}
+    gpu, cpu = GpuSearcher(device=0), OracleSearcher()
+    splits = []
+    for s in range(2):
+        w = splitgen.SplitWriter(schema, f"card-{s}")
+        w.add_documents(docs[s::2])
+        data = w.finalize()
+        gpu.add_split(f"card-{s}", data)
+        cpu.add_split(f"card-{s}", data)
+        splits.append((f"card-{s}", len(docs[s::2])))
+    aggs = {"sev": {"cardinality": {"field": "severity_text"}},
+            "ten": {"cardinality": {"field": "tenant_id"}},
+            "opt": {"cardinality": {"field": "opt_u"}},
+            "ts": {"cardinality": {"field": "timestamp"}}}
+    for q in ({"type": "match_all"},
+              {"type": "term", "field": "severity_text", "value": "WARN"}):
+        req = make_leaf_request(q, schema, splits, max_hits=0,
+                                aggregation=aggs)
+        g = gpu.leaf_search(req)
+        e = cpu.leaf_search(req)
+        gj = gpu.finalize_agg_json(g["intermediate_aggregation_result"], aggs)
+        ej = cpu.finalize_agg_json(e["intermediate_aggregation_result"], aggs)
+        assert gj == ej, (q, gj, ej)
+        if q["type"] == "match_all":
+            assert gj == {"sev": {"value": 3.0}, "ten": {"value": 41.0},
+                          "opt": {"value": 29.0}, "ts": {"value": 97.0}}
