@@ -610,6 +610,29 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
             out.aggs.push_back(std::move(a));
             continue;
         }
+        if (d.kind == AggDef::METRIC && d.metric.kind == MetricAgg::PERCENTILES) {
+            a.kind = 6;
+            if (f && f->type != FastFieldView::STR && !f->multi) {
+                auto pvisit = [&](uint32_t doc) {
+                    if (!f->present(doc)) return;
+                    double v;
+                    if (f->type == FastFieldView::U64) v = double(f->u64(doc));
+                    else if (f->type == FastFieldView::F64) v = f->f64(doc);
+                    else v = double(f->i64(doc));
+                    if (v < 0)
+                        throw std::runtime_error(
+                            "percentiles over negative values (r1 limit)");
+                    if (v < PERC_MIN_VALUE) a.sketch.zero++;
+                    else a.sketch.counts[perc_key_for(v)]++;
+                };
+                if (m.all)
+                    for (uint32_t doc = 0; doc < sv.num_docs; ++doc) pvisit(doc);
+                else
+                    for (uint32_t doc : m.docs) pvisit(doc);
+            }
+            out.aggs.push_back(std::move(a));
+            continue;
+        }
         if (d.kind == AggDef::METRIC) {
             a.kind = 5;
             if (f && f->type != FastFieldView::STR && !f->multi) {
@@ -732,6 +755,8 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
             continue;
         }
         a.kind = d.kind == AggDef::DATE_HISTOGRAM ? 1 : 2;
+        for (auto& s : d.sub)
+            a.sub_kinds.push_back(s.kind == MetricAgg::PERCENTILES ? 1 : 0);
         std::map<int64_t, AggBucket> buckets;  // key quantized to bucket index
         auto visit = [&](uint32_t doc) {
             if (!f || !f->present(doc)) return;
@@ -743,10 +768,22 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
             int64_t bi = int64_t(std::floor((v - d.offset) / d.interval));
             AggBucket& b = buckets[bi];
             b.doc_count++;
-            if (!d.sub.empty() && b.sub.empty()) b.sub.resize(d.sub.size());
+            if (!d.sub.empty() && b.sub.empty()) {
+                b.sub.resize(d.sub.size());
+                b.psub.resize(d.sub.size());
+            }
             for (size_t si = 0; si < d.sub.size(); ++si) {
                 double sval;
                 if (sub_value(si, doc, &sval)) {
+                    if (d.sub[si].kind == MetricAgg::PERCENTILES) {
+                        if (sval < 0)
+                            throw std::runtime_error(
+                                "percentiles over negative values (r1 limit)");
+                        SketchPayload& pp = b.psub[si];
+                        if (sval < PERC_MIN_VALUE) pp.zero++;
+                        else pp.counts[perc_key_for(sval)]++;
+                        continue;
+                    }
                     StatsPayload& sp = b.sub[si];
                     sp.count++;
                     sp.sum += sval;
@@ -764,6 +801,7 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
             AggBucket b = kv.second;
             b.key = double(kv.first) * d.interval + d.offset;
             if (b.sub.empty() && !d.sub.empty()) b.sub.resize(d.sub.size());
+            if (b.psub.empty() && !d.sub.empty()) b.psub.resize(d.sub.size());
             a.buckets.push_back(std::move(b));
         }
         out.aggs.push_back(std::move(a));

@@ -68,7 +68,10 @@ enum AggKindDev : uint32_t {
     // components packed MSB-first into <=63 bits (so the key can never hit
     // the ~0 sentinel); per-source params in the c_* fields + the sub_*
     // column arrays
-    AGGD_COMP = 5
+    AGGD_COMP = 5,
+    // top-level percentiles: DDSketch key counts, (p_n_keys+1) u64 words at
+    // counts_out (word 0 = zero bucket, word j = key p_k_lo+j-1)
+    AGGD_PERC = 6
 };
 constexpr uint32_t AGG_MAX_RANGES = 16;
 
@@ -117,6 +120,15 @@ struct AggDev {
     uint32_t c_missing;      // bitmask: missing_bucket (else null doc drops)
     double c_interval[4], c_offset[4];
     int64_t c_base[4];       // histogram: floor((col_min-offset)/interval)
+    // percentiles (DDSketch restatement): sub index carrying a sketch
+    // (0xFF = none), sketch region, key range, and the scratch offset of the
+    // gamma^k boundary doubles the kernel binary-searches (host-computed so
+    // bucketing is bit-identical to the oracle)
+    uint32_t p_si;
+    uint64_t p_out;
+    int32_t p_k_lo;
+    uint32_t p_n_keys;
+    uint64_t p_bound_off;
 };
 
 struct QueryDev {

@@ -331,6 +331,22 @@ __device__ __forceinline__ double agg_value(const QueryDev& q, uint64_t values_o
     }
 }
 
+// DDSketch slot for value v: 0 = zero bucket (v < 1e-9), else 1 + index of
+// the first boundary >= v in the host-computed gamma^k table (identical
+// doubles to the oracle's perc_key_for => bit-identical bucketing)
+__device__ __forceinline__ uint32_t perc_slot(const QueryDev& q, const AggDev& a,
+                                              double v) {
+    if (v < 1e-9) return 0;
+    const double* B = (const double*)(q.scratch + a.p_bound_off);
+    uint32_t lo = 0, hi = a.p_n_keys - 1;
+    while (lo < hi) {
+        uint32_t mid = (lo + hi) >> 1;
+        if (v <= B[mid]) hi = mid;
+        else lo = mid + 1;
+    }
+    return lo + 1;
+}
+
 __device__ __forceinline__ uint64_t agg_ord(const QueryDev& q, uint64_t values_off,
                                             uint32_t width, uint32_t doc) {
     const uint8_t* col = q.split + values_off;
@@ -748,6 +764,15 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                                 sl = (sl + 1) & (slots - 1);
                             }
                             if (!done) tab[2 * slots + 1] = 1ull;  // overflow
+                        } else if (a.kind == AGGD_PERC) {
+                            if (!a.n_buckets) continue;
+                            double v = agg_value(q, a.values_off, a.value_width,
+                                                 a.value_is_i64, d);
+                            uint32_t sj = perc_slot(q, a, v);
+                            atomicAdd((unsigned long long*)(q.results +
+                                                            a.counts_out) +
+                                          sj,
+                                      1ull);
                         } else if (a.kind == AGGD_METRIC) {
                             if (!a.values_off) continue;
                             double v = agg_value(q, a.values_off, a.value_width,
@@ -794,6 +819,19 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                                 }
                                 double sv = agg_value(q, a.sub_values_off[si],
                                                       a.sub_width[si], a.sub_is_i64[si], d);
+                                if (si == a.p_si) {
+                                    // percentiles sub: DDSketch key count
+                                    // (word 0 = zero bucket, binary search
+                                    // over the host boundary table)
+                                    uint32_t sj = perc_slot(q, a, sv);
+                                    atomicAdd((unsigned long long*)(q.results +
+                                                                    a.p_out) +
+                                                  uint64_t(idx) *
+                                                      (a.p_n_keys + 1) +
+                                                  sj,
+                                              1ull);
+                                    continue;
+                                }
                                 uint8_t* slot = q.results + a.sub_out +
                                                 (uint64_t(idx) * a.n_sub + si) * 40;
                                 atomicAdd((unsigned long long*)slot, 1ull);

@@ -664,6 +664,9 @@ struct AggPlan {
     std::vector<const FastFieldView*> fields;
     // composite aggs: per-source column views (null for other kinds)
     std::vector<std::array<const FastFieldView*, 4>> comp_fields;
+    // percentiles: per-agg gamma^k boundary doubles (appended to scratch;
+    // empty for aggs without a sketch)
+    std::vector<std::vector<double>> pbounds;
     size_t out_bytes = 0;                // total result bytes (counts + subs)
     std::vector<uint8_t> init;           // initial contents of the out region
 };
@@ -690,8 +693,10 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
         const FastFieldView* f = sv.fast_field(d.field);
         ap.fields.push_back(f);
         ap.comp_fields.push_back({nullptr, nullptr, nullptr, nullptr});
+        ap.pbounds.emplace_back();
         AggDev a{};
         a.lds_slot = 0xFF;
+        a.p_si = 0xFF;
         if (d.kind == AggDef::COMPOSITE) {
             a.kind = AGGD_COMP;
             bool all_present = true;
@@ -793,6 +798,31 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
             } else {
                 a.n_buckets = 0;  // no such column: zero buckets
             }
+        } else if (d.kind == AggDef::METRIC &&
+                   d.metric.kind == MetricAgg::PERCENTILES) {
+            a.kind = AGGD_PERC;
+            if (f && f->type != FastFieldView::STR && !f->multi) {
+                double mn, mx;
+                col_min_max(*f, &mn, &mx);
+                if (mn < 0)
+                    throw std::runtime_error(
+                        "percentiles over negative values (r1 limit)");
+                int32_t k_lo = 0, k_hi = 0;
+                if (mx >= PERC_MIN_VALUE) {
+                    k_lo = perc_key_for(std::max(mn, PERC_MIN_VALUE));
+                    k_hi = perc_key_for(mx);
+                }
+                a.p_k_lo = k_lo;
+                a.p_n_keys = uint32_t(k_hi - k_lo + 1);
+                a.n_buckets = a.p_n_keys + 1;  // + zero bucket word
+                a.values_off = f->values.off;
+                a.nulls_off = f->nullable ? f->nulls.off : 0;
+                a.value_width = 8;
+                a.value_is_i64 = f->type == FastFieldView::U64   ? 0
+                                 : f->type == FastFieldView::F64 ? 2
+                                                                 : 1;
+                perc_boundaries(k_lo, k_hi, ap.pbounds.back());
+            }
         } else if (d.kind == AggDef::METRIC) {
             a.kind = AGGD_METRIC;
             if (f && f->type != FastFieldView::STR && !f->multi) {
@@ -887,6 +917,33 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
                         a.sub_is_i64[si] = sf->type == FastFieldView::U64   ? 0
                                            : sf->type == FastFieldView::F64 ? 2
                                                                             : 1;
+                        if (d.sub[si].kind == MetricAgg::PERCENTILES) {
+                            if (a.p_si != 0xFF)
+                                throw std::runtime_error(
+                                    ">1 percentiles sub-aggregation "
+                                    "(r1 limit)");
+                            double mn, mx;
+                            col_min_max(*sf, &mn, &mx);
+                            if (mn < 0)
+                                throw std::runtime_error(
+                                    "percentiles over negative values "
+                                    "(r1 limit)");
+                            int32_t k_lo = 0, k_hi = 0;
+                            if (mx >= PERC_MIN_VALUE) {
+                                k_lo = perc_key_for(
+                                    std::max(mn, PERC_MIN_VALUE));
+                                k_hi = perc_key_for(mx);
+                            }
+                            a.p_si = uint32_t(si);
+                            a.p_k_lo = k_lo;
+                            a.p_n_keys = uint32_t(k_hi - k_lo + 1);
+                            if (uint64_t(a.n_buckets) * (a.p_n_keys + 1) >
+                                (1ull << 25))
+                                throw std::runtime_error(
+                                    "percentiles sketch region too large "
+                                    "(r1 limit)");
+                            perc_boundaries(k_lo, k_hi, ap.pbounds.back());
+                        }
                     } else {
                         a.sub_values_off[si] = 0;  // missing column: no values
                         a.sub_width[si] = 0;
@@ -906,6 +963,10 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
         // AGGD_COMP uses n_sub for its sources and n_buckets for the hash
         // table words — no per-bucket stats region
         if (a.kind != AGGD_COMP) off += uint64_t(a.n_buckets) * a.n_sub * 40;
+        if (a.p_si != 0xFF) {
+            a.p_out = off;
+            off += uint64_t(a.n_buckets) * (a.p_n_keys + 1) * 8;
+        }
         ap.devs.push_back(a);
     }
     // LDS slot assignment: the first histogram agg that fits gets the LDS
@@ -1454,6 +1515,13 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
     size_t off_ranges = off_ktabs + ktabs.size() * 4;
     off_ranges = (off_ranges + 63) & ~size_t(63);
     size_t scratch_bytes = off_ranges + ranges_bytes;
+    // percentiles boundary tables (gamma^k doubles the kernel searches)
+    scratch_bytes = (scratch_bytes + 7) & ~size_t(7);
+    for (size_t i = 0; i < ap.devs.size(); ++i)
+        if (!ap.pbounds[i].empty()) {
+            ap.devs[i].p_bound_off = scratch_bytes;
+            scratch_bytes += ap.pbounds[i].size() * 8;
+        }
 
     // block ranges per term per tile (host two-pointer walk over skip entries)
     std::vector<uint32_t> ranges(ranges_bytes / 4, 0);
@@ -1483,6 +1551,10 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
     memcpy(scratch.data() + off_aggs, ap.devs.data(), ap.devs.size() * sizeof(AggDev));
     memcpy(scratch.data() + off_ktabs, ktabs.data(), ktabs.size() * 4);
     memcpy(scratch.data() + off_ranges, ranges.data(), ranges_bytes);
+    for (size_t i = 0; i < ap.devs.size(); ++i)
+        if (!ap.pbounds[i].empty())
+            memcpy(scratch.data() + ap.devs[i].p_bound_off,
+                   ap.pbounds[i].data(), ap.pbounds[i].size() * 8);
 
     uint64_t matched = 0;
     mark("plan+assembly");
@@ -1984,6 +2056,16 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     }
                     std::sort(r.term_counts.begin(), r.term_counts.end());
                 }
+            } else if (d.kind == AggDef::METRIC &&
+                       d.metric.kind == MetricAgg::PERCENTILES) {
+                r.kind = 6;
+                if (a.n_buckets) {
+                    r.sketch.zero = counts[0];
+                    for (uint32_t j = 1; j < a.n_buckets; ++j)
+                        if (counts[j])
+                            r.sketch.counts[a.p_k_lo + int32_t(j) - 1] =
+                                counts[j];
+                }
             } else if (d.kind == AggDef::METRIC) {
                 r.kind = 5;
                 if (a.n_buckets) {
@@ -2012,14 +2094,32 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                 }
             } else {
                 r.kind = d.kind == AggDef::DATE_HISTOGRAM ? 1 : 2;
+                for (auto& sdef : d.sub)
+                    r.sub_kinds.push_back(
+                        sdef.kind == MetricAgg::PERCENTILES ? 1 : 0);
                 const uint8_t* subs = agg_out.data() + (a.sub_out - r_agg);
+                const uint64_t* pwords =
+                    a.p_si != 0xFF
+                        ? (const uint64_t*)(agg_out.data() + (a.p_out - r_agg))
+                        : nullptr;
                 for (uint32_t bi = 0; bi < a.n_buckets; ++bi) {
                     if (!counts[bi]) continue;
                     AggBucket b;
                     b.key = double(a.base_index + int64_t(bi)) * d.interval + d.offset;
                     b.doc_count = counts[bi];
                     b.sub.resize(d.sub.size());
+                    b.psub.resize(d.sub.size());
+                    if (pwords) {
+                        const uint64_t* pw =
+                            pwords + uint64_t(bi) * (a.p_n_keys + 1);
+                        SketchPayload& pp = b.psub[a.p_si];
+                        pp.zero = pw[0];
+                        for (uint32_t j = 1; j <= a.p_n_keys; ++j)
+                            if (pw[j])
+                                pp.counts[a.p_k_lo + int32_t(j) - 1] = pw[j];
+                    }
                     for (uint32_t s = 0; s < a.n_sub; ++s) {
+                        if (s == a.p_si) continue;
                         const uint8_t* slot = subs + (uint64_t(bi) * a.n_sub + s) * 40;
                         StatsPayload sp2;
                         uint64_t mn, mx;

@@ -20,6 +20,7 @@
 //        u16 key_len, key bytes, u64 doc_count   (sorted by key bytes)
 #pragma once
 #include <algorithm>
+#include <cmath>
 #include <cstdint>
 #include <cstring>
 #include <limits>
@@ -48,10 +49,70 @@ struct StatsPayload {
     }
 };
 
+// restated DDSketch store (sketches-ddsketch defaults: alpha=0.01,
+// min_value=1e-9) — tantivy's percentiles aggregation; the bucket->value
+// mapping (k = min k with v <= gamma^k, reported value 2*gamma^k/(gamma+1),
+// rank = q*(n-1), first cumulative > rank) is pinned by the aggregations
+// golden's expected p85 values. Negative values: rejected (r1 limit).
+constexpr double PERC_ALPHA = 0.01;
+constexpr double PERC_MIN_VALUE = 1e-9;
+inline double perc_gamma() { return (1.0 + PERC_ALPHA) / (1.0 - PERC_ALPHA); }
+
+struct SketchPayload {
+    uint64_t zero = 0;                     // values below min_value
+    std::map<int32_t, uint64_t> counts;    // key k -> count
+    void merge(const SketchPayload& o) {
+        zero += o.zero;
+        for (auto& kv : o.counts) counts[kv.first] += kv.second;
+    }
+    uint64_t total() const {
+        uint64_t n = zero;
+        for (auto& kv : counts) n += kv.second;
+        return n;
+    }
+    // quantile q in [0,1]
+    double value_at(double q) const {
+        uint64_t n = total();
+        if (!n) return 0;
+        double rank = q * double(n - 1);
+        uint64_t cum = zero;
+        if (zero && double(cum) > rank) return 0.0;
+        double g = perc_gamma();
+        for (auto& kv : counts) {
+            cum += kv.second;
+            if (double(cum) > rank)
+                return 2.0 * pow(g, double(kv.first)) / (g + 1.0);
+        }
+        return counts.empty() ? 0.0
+                              : 2.0 * pow(g, double(counts.rbegin()->first)) /
+                                    (g + 1.0);
+    }
+};
+
+// the per-key upper boundaries gamma^(k_lo..k_hi), computed ONCE host-side
+// and shared verbatim with the device kernel so value->key bucketing is a
+// binary search over identical doubles on both engines (no libm divergence)
+inline void perc_boundaries(int32_t k_lo, int32_t k_hi,
+                            std::vector<double>& out) {
+    double g = perc_gamma();
+    out.resize(size_t(k_hi - k_lo + 1));
+    for (int32_t k = k_lo; k <= k_hi; ++k)
+        out[size_t(k - k_lo)] = pow(g, double(k));
+}
+// smallest k with v <= gamma^k (host; used for column-range planning)
+inline int32_t perc_key_for(double v) {
+    double g = perc_gamma();
+    int32_t k = int32_t(ceil(log(v) / log(g)));
+    while (pow(g, double(k)) < v) ++k;
+    while (k > INT32_MIN && pow(g, double(k - 1)) >= v) --k;
+    return k;
+}
+
 struct AggBucket {
     double key = 0;
     uint64_t doc_count = 0;
     std::vector<StatsPayload> sub;
+    std::vector<SketchPayload> psub;  // parallel: only sub_kinds[s]==1 slots
 };
 
 struct AggResult {
@@ -61,6 +122,9 @@ struct AggResult {
     std::vector<AggBucket> buckets;                            // histos (sorted by key)
     std::vector<std::pair<std::string, uint64_t>> term_counts; // terms (sorted by key)
     StatsPayload metric;  // kind 5: top-level metric aggregation
+    SketchPayload sketch;  // kind 6: top-level percentiles
+    // per sub-agg payload kind: 0 = stats (40 B), 1 = percentiles sketch
+    std::vector<uint8_t> sub_kinds;
     // terms key domain: 0 = str (dictionary term bytes); 1/2/3 = numeric fast
     // column (u64 / i64-or-date / f64), keys stored as the value's 8-byte
     // big-endian SORTABLE bits so the string-keyed merge/truncate machinery
@@ -152,12 +216,25 @@ struct IntermediateAggResults {
             put(&a.kind, 1);
             uint16_t ns = uint16_t(a.sub_names.size());
             put(&ns, 2);
-            for (auto& s : a.sub_names) {
-                uint16_t sl = uint16_t(s.size());
+            for (size_t si = 0; si < a.sub_names.size(); ++si) {
+                uint16_t sl = uint16_t(a.sub_names[si].size());
                 put(&sl, 2);
-                put(s.data(), sl);
+                put(a.sub_names[si].data(), sl);
+                uint8_t sk = si < a.sub_kinds.size() ? a.sub_kinds[si] : 0;
+                put(&sk, 1);
             }
-            if (a.kind == 5) {
+            auto put_sketch = [&](const SketchPayload& sp) {
+                put(&sp.zero, 8);
+                uint32_t ne = uint32_t(sp.counts.size());
+                put(&ne, 4);
+                for (auto& kv : sp.counts) {
+                    put(&kv.first, 4);
+                    put(&kv.second, 8);
+                }
+            };
+            if (a.kind == 6) {
+                put_sketch(a.sketch);
+            } else if (a.kind == 5) {
                 put(&a.metric.count, 8);
                 put(&a.metric.sum, 8);
                 put(&a.metric.min, 8);
@@ -182,6 +259,11 @@ struct IntermediateAggResults {
                     put(&b.key, 8);
                     put(&b.doc_count, 8);
                     for (size_t s = 0; s < a.sub_names.size(); ++s) {
+                        if (s < a.sub_kinds.size() && a.sub_kinds[s] == 1) {
+                            put_sketch(s < b.psub.size() ? b.psub[s]
+                                                         : SketchPayload{});
+                            continue;
+                        }
                         const StatsPayload& sp =
                             s < b.sub.size() ? b.sub[s] : StatsPayload{};
                         put(&sp.count, 8);
@@ -229,8 +311,25 @@ struct IntermediateAggResults {
                 need(sl);
                 a.sub_names.emplace_back((const char*)p, sl);
                 p += sl;
+                uint8_t sk;
+                get(&sk, 1);
+                a.sub_kinds.push_back(sk);
             }
-            if (a.kind == 5) {
+            auto get_sketch = [&](SketchPayload& sp) {
+                get(&sp.zero, 8);
+                uint32_t ne;
+                get(&ne, 4);
+                for (uint32_t e2 = 0; e2 < ne; ++e2) {
+                    int32_t k;
+                    uint64_t c;
+                    get(&k, 4);
+                    get(&c, 8);
+                    sp.counts[k] = c;
+                }
+            };
+            if (a.kind == 6) {
+                get_sketch(a.sketch);
+            } else if (a.kind == 5) {
                 get(&a.metric.count, 8);
                 get(&a.metric.sum, 8);
                 get(&a.metric.min, 8);
@@ -262,7 +361,13 @@ struct IntermediateAggResults {
                     get(&b.key, 8);
                     get(&b.doc_count, 8);
                     b.sub.resize(a.sub_names.size());
-                    for (auto& sp : b.sub) {
+                    b.psub.resize(a.sub_names.size());
+                    for (size_t s = 0; s < a.sub_names.size(); ++s) {
+                        if (s < a.sub_kinds.size() && a.sub_kinds[s] == 1) {
+                            get_sketch(b.psub[s]);
+                            continue;
+                        }
+                        StatsPayload& sp = b.sub[s];
                         get(&sp.count, 8);
                         get(&sp.sum, 8);
                         get(&sp.min, 8);
@@ -290,7 +395,9 @@ struct IntermediateAggResults {
             const AggResult& b = o.aggs[i];
             if (a.name != b.name || a.kind != b.kind)
                 throw std::runtime_error("QAGG1: merge mismatch");
-            if (a.kind == 5) {
+            if (a.kind == 6) {
+                a.sketch.merge(b.sketch);
+            } else if (a.kind == 5) {
                 a.metric.merge(b.metric);
             } else if (a.kind == 3) {
                 std::vector<std::pair<std::string, uint64_t>> merged;
@@ -334,6 +441,10 @@ struct IntermediateAggResults {
                         m.sub.resize(a.sub_names.size());
                         for (size_t s = 0; s < m.sub.size() && s < n.sub.size(); ++s)
                             m.sub[s].merge(n.sub[s]);
+                        m.psub.resize(a.sub_names.size());
+                        for (size_t s = 0; s < m.psub.size() && s < n.psub.size();
+                             ++s)
+                            m.psub[s].merge(n.psub[s]);
                         merged.push_back(std::move(m));
                     }
                 }
@@ -453,6 +564,52 @@ inline void stats_to_json(std::string& o, const MetricAgg& m, const StatsPayload
     }
 }
 
+inline void percentiles_to_json(std::string& o, const MetricAgg& m,
+                                const SketchPayload& s) {
+    bool empty = s.total() == 0;
+    auto pkey = [&](double p, bool as_str) {
+        char buf[40];
+        if (p == floor(p)) snprintf(buf, sizeof buf, "%.1f", p);
+        else {
+            std::string t;
+            mj::num_to(t, p);
+            snprintf(buf, sizeof buf, "%s", t.c_str());
+        }
+        if (as_str) {
+            std::string q = "\"";
+            q += buf;
+            q += "\"";
+            return q;
+        }
+        return std::string(buf);
+    };
+    o += "{\"values\":";
+    if (m.keyed) {
+        o += "{";
+        for (size_t i = 0; i < m.percents.size(); ++i) {
+            if (i) o += ",";
+            o += pkey(m.percents[i], true);
+            o += ":";
+            if (empty) o += "null";
+            else mj::num_to(o, s.value_at(m.percents[i] / 100.0));
+        }
+        o += "}";
+    } else {
+        o += "[";
+        for (size_t i = 0; i < m.percents.size(); ++i) {
+            if (i) o += ",";
+            o += "{\"key\":";
+            o += pkey(m.percents[i], false);
+            o += ",\"value\":";
+            if (empty) o += "null";
+            else mj::num_to(o, s.value_at(m.percents[i] / 100.0));
+            o += "}";
+        }
+        o += "]";
+    }
+    o += "}";
+}
+
 inline std::string finalize_aggs_json(const IntermediateAggResults& ir,
                                       const std::vector<AggDef>& defs) {
     if (ir.aggs.size() != defs.size())
@@ -465,7 +622,10 @@ inline std::string finalize_aggs_json(const IntermediateAggResults& ir,
         mj::escape_to(o, d.name);
         o += ":";
         if (d.kind == AggDef::METRIC) {
-            stats_to_json(o, d.metric, a.metric);
+            if (d.metric.kind == MetricAgg::PERCENTILES)
+                percentiles_to_json(o, d.metric, a.sketch);
+            else
+                stats_to_json(o, d.metric, a.metric);
             continue;
         }
         if (d.kind == AggDef::RANGE) {
@@ -683,9 +843,15 @@ inline std::string finalize_aggs_json(const IntermediateAggResults& ir,
                         o += ",";
                         mj::escape_to(o, d.sub[s].name);
                         o += ":";
-                        StatsPayload sp;
-                        if (b && s < b->sub.size()) sp = b->sub[s];
-                        stats_to_json(o, d.sub[s], sp);
+                        if (d.sub[s].kind == MetricAgg::PERCENTILES) {
+                            SketchPayload pp;
+                            if (b && s < b->psub.size()) pp = b->psub[s];
+                            percentiles_to_json(o, d.sub[s], pp);
+                        } else {
+                            StatsPayload sp;
+                            if (b && s < b->sub.size()) sp = b->sub[s];
+                            stats_to_json(o, d.sub[s], sp);
+                        }
                     }
                     o += "}";
                 }

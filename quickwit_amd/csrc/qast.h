@@ -859,8 +859,14 @@ inline std::string ms_to_rfc3339(int64_t ms) {
 // ---------------------------------------------------------------- agg plan
 struct MetricAgg {
     std::string name;
-    enum Kind { STATS, AVG, SUM, MIN, MAX, COUNT, EXTENDED } kind = STATS;
+    enum Kind { STATS, AVG, SUM, MIN, MAX, COUNT, EXTENDED, PERCENTILES }
+        kind = STATS;
     std::string field;
+    // PERCENTILES (restated DDSketch, alpha=0.01 — the sketches-ddsketch
+    // config tantivy's percentiles agg uses; fit pinned by the
+    // aggregations golden's expected values): requested percents + keyed
+    std::vector<double> percents{1, 5, 25, 50, 75, 95, 99};
+    bool keyed = true;
 };
 
 struct RangeSpec {
@@ -1072,12 +1078,21 @@ inline std::vector<AggDef> parse_agg_request(const std::string& json) {
             else if ((ms0 = body->get("min"))) mk = MetricAgg::MIN;
             else if ((ms0 = body->get("max"))) mk = MetricAgg::MAX;
             else if ((ms0 = body->get("value_count"))) mk = MetricAgg::COUNT;
+            else if ((ms0 = body->get("percentiles"))) mk = MetricAgg::PERCENTILES;
             else throw std::runtime_error("aggregation not supported: " + a.name);
             a.kind = AggDef::METRIC;
             a.field = ms0->at("field")->s;
             a.metric.kind = mk;
             a.metric.name = a.name;
             a.metric.field = a.field;
+            if (mk == MetricAgg::PERCENTILES) {
+                if (const mj::Value* ps = ms0->get("percents")) {
+                    a.metric.percents.clear();
+                    for (auto& pv : ps->arr) a.metric.percents.push_back(pv->num());
+                }
+                if (const mj::Value* kd = ms0->get("keyed"))
+                    a.metric.keyed = kd->b;
+            }
         }
         if (a.min_doc_count < 0) a.min_doc_count = (a.kind == AggDef::TERMS) ? 1 : 0;
         if (const mj::Value* subs = body->get("aggs")) {
@@ -1093,8 +1108,16 @@ inline std::vector<AggDef> parse_agg_request(const std::string& json) {
                 else if ((ms = sb->get("min"))) m.kind = MetricAgg::MIN;
                 else if ((ms = sb->get("max"))) m.kind = MetricAgg::MAX;
                 else if ((ms = sb->get("value_count"))) m.kind = MetricAgg::COUNT;
+                else if ((ms = sb->get("percentiles"))) m.kind = MetricAgg::PERCENTILES;
                 else throw std::runtime_error("sub-aggregation not supported: " + m.name);
                 m.field = ms->at("field")->s;
+                if (m.kind == MetricAgg::PERCENTILES) {
+                    if (const mj::Value* ps = ms->get("percents")) {
+                        m.percents.clear();
+                        for (auto& pv : ps->arr) m.percents.push_back(pv->num());
+                    }
+                    if (const mj::Value* kd = ms->get("keyed")) m.keyed = kd->b;
+                }
                 a.sub.push_back(std::move(m));
             }
         }
