@@ -1207,3 +1207,58 @@ def test_cardinality_aggs():
         if q["type"] == "match_all":
             assert gj == {"sev": {"value": 3.0}, "ten": {"value": 41.0},
                           "opt": {"value": 29.0}, "ts": {"value": 97.0}}
+
+
+# ------------------------------------------------- percentiles aggregation
+def test_percentiles_aggs():
+    # DDSketch restatement: sub-agg under date_histogram (the golden's shape)
+    # and top-level, vs the oracle; bucketing must be bit-identical (shared
+    # boundary tables), so exact JSON equality is required
+    import random
+    rng = random.Random(5)
+    docs = []
+    for i in range(4000):
+        d = {"timestamp": (1700000000 + i % 7200) * 1000,
+             "severity_text": "INFO" if i % 4 else "ERROR", "body": "x",
+             "tenant_id": i % 3}
+        if rng.random() < 0.9:
+            d["lat"] = round(rng.uniform(0.0, 5000.0), 3)
+        docs.append(d)
+    schema = {"timestamp_field": None, "fields":
+              [{"name": "timestamp", "type": "i64", "fast": True},
+               {"name": "severity_text", "type": "text", "tokenizer": "raw",
+                "fast": True},
+               {"name": "body", "type": "text"},
+               {"name": "tenant_id", "type": "u64", "fast": True},
+               {"name": "lat", "type": "f64", "fast": True}]}
+    gpu, cpu = GpuSearcher(device=0), OracleSearcher()
+    splits = []
+    for s in range(2):
+        w = splitgen.SplitWriter(schema, f"perc-{s}")
+        w.add_documents(docs[s::2])
+        data = w.finalize()
+        gpu.add_split(f"perc-{s}", data)
+        cpu.add_split(f"perc-{s}", data)
+        splits.append((f"perc-{s}", len(docs[s::2])))
+
+    aggs = {"per_hour": {"date_histogram": {"field": "timestamp",
+                         "fixed_interval": "3600000ms"},
+                         "aggs": {"lat_p": {"percentiles": {"field": "lat",
+                                  "percents": [50, 85, 99],
+                                  "keyed": False}}}},
+            "lat_top": {"percentiles": {"field": "lat"}}}
+    for q in ({"type": "match_all"},
+              {"type": "term", "field": "severity_text", "value": "ERROR"}):
+        req = make_leaf_request(q, schema, splits, max_hits=0,
+                                aggregation=aggs)
+        g = gpu.leaf_search(req)
+        e = cpu.leaf_search(req)
+        gj = gpu.finalize_agg_json(g["intermediate_aggregation_result"], aggs)
+        ej = cpu.finalize_agg_json(e["intermediate_aggregation_result"], aggs)
+        assert gj == ej, (q, gj, ej)
+    # sanity: p50 < p85 < p99 and all within the value range
+    b0 = gj["per_hour"]["buckets"][0]["lat_p"]["values"]
+    vals = [v["value"] for v in b0]
+    assert vals == sorted(vals) and 0 <= vals[0] <= 5100
+    top = gj["lat_top"]["values"]
+    assert set(top) == {"1.0", "5.0", "25.0", "50.0", "75.0", "95.0", "99.0"}
