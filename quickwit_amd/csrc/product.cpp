@@ -1005,8 +1005,11 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
                 memcpy(&ap.init[a.counts_out - out_base + uint64_t(s2) * 16],
                        &ones, 8);
         }
+        // AGGD_COMP reuses n_sub for its sources and has NO stats sub region
+        if (a.kind == AGGD_COMP) continue;
         for (uint64_t b = 0; b < a.n_buckets; ++b)
             for (uint32_t s = 0; s < a.n_sub; ++s) {
+                if (s == a.p_si) continue;  // percentiles slot: stays zero
                 uint64_t slot = a.sub_out - out_base + (b * a.n_sub + s) * 40;
                 memcpy(&ap.init[slot + 16], &ones, 8);  // min slot: max sortable
                 // max slot stays 0 (= most negative sortable)
