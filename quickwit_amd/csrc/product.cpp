@@ -1399,12 +1399,20 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             for (const AggDef& d : parse_agg_request(*req.aggregation_request)) {
                 AggResult r;
                 r.name = d.name;
-                r.kind = d.kind == AggDef::TERMS      ? 3
-                         : d.kind == AggDef::RANGE     ? 4
-                         : d.kind == AggDef::METRIC    ? 5
+                r.kind = d.kind == AggDef::TERMS         ? 3
+                         : d.kind == AggDef::CARDINALITY ? 3
+                         : d.kind == AggDef::COMPOSITE   ? 3
+                         : d.kind == AggDef::RANGE       ? 4
+                         : d.kind == AggDef::METRIC
+                             ? (d.metric.kind == MetricAgg::PERCENTILES ? 6 : 5)
                          : d.kind == AggDef::HISTOGRAM ? 2
                                                        : 1;
-                for (auto& s : d.sub) r.sub_names.push_back(s.name);
+                if (d.kind == AggDef::COMPOSITE) r.key_kind = 9;
+                for (auto& s : d.sub) {
+                    r.sub_names.push_back(s.name);
+                    r.sub_kinds.push_back(
+                        s.kind == MetricAgg::PERCENTILES ? 1 : 0);
+                }
                 out.aggs.aggs.push_back(std::move(r));
             }
             out.has_aggs = true;
