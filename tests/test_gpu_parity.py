@@ -1175,13 +1175,15 @@ def test_cardinality_aggs():
     for i in range(2500):
         d = {"timestamp": 1700000000 + (i % 97), "severity_text":
              rng.choice(["INFO", "WARN", "ERROR"]), "body": "x",
-             "tenant_id": i % 41}
+             "tenant_id": i % 41,
+             "svc": rng.choice(["api", "ingest", "janitor"])}
         if rng.random() < 0.7:
             d["opt_u"] = rng.randrange(29)
         docs.append(d)
     schema = {"timestamp_field": "timestamp", "fields":
               splitgen.HDFS_SCHEMA["fields"] +
-              [{"name": "opt_u", "type": "u64", "fast": True}]}
+              [{"name": "opt_u", "type": "u64", "fast": True},
+               {"name": "svc", "type": "str", "fast": True}]}
     gpu, cpu = GpuSearcher(device=0), OracleSearcher()
     splits = []
     for s in range(2):
@@ -1191,7 +1193,7 @@ def test_cardinality_aggs():
         gpu.add_split(f"card-{s}", data)
         cpu.add_split(f"card-{s}", data)
         splits.append((f"card-{s}", len(docs[s::2])))
-    aggs = {"sev": {"cardinality": {"field": "severity_text"}},
+    aggs = {"sev": {"cardinality": {"field": "svc"}},
             "ten": {"cardinality": {"field": "tenant_id"}},
             "opt": {"cardinality": {"field": "opt_u"}},
             "ts": {"cardinality": {"field": "timestamp"}}}
