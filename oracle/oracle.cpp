@@ -546,19 +546,48 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
         };
         if (d.kind == AggDef::TERMS || d.kind == AggDef::CARDINALITY) {
             a.kind = 3;
+            bool with_subs = d.kind == AggDef::TERMS && !d.sub.empty();
+            if (with_subs) {
+                for (auto& s : d.sub) {
+                    if (s.kind == MetricAgg::PERCENTILES)
+                        throw std::runtime_error(
+                            "percentiles under terms (r1 limit)");
+                    a.sub_names.push_back(s.name);
+                    a.sub_kinds.push_back(0);
+                }
+            }
             if (f && f->type == FastFieldView::STR) {
                 std::vector<uint64_t> counts(f->cardinality, 0);
+                std::vector<std::vector<StatsPayload>> subs(
+                    with_subs ? f->cardinality : 0);
+                auto acc_subs = [&](uint32_t ord, uint32_t doc) {
+                    if (!with_subs) return;
+                    if (subs[ord].empty()) subs[ord].resize(d.sub.size());
+                    for (size_t si = 0; si < d.sub.size(); ++si) {
+                        double sval;
+                        if (!sub_value(si, doc, &sval)) continue;
+                        StatsPayload& sp = subs[ord][si];
+                        sp.count++;
+                        sp.sum += sval;
+                        sp.min = std::min(sp.min, sval);
+                        sp.max = std::max(sp.max, sval);
+                        sp.sum_sq += sval * sval;
+                    }
+                };
                 auto visit = [&](uint32_t doc) {
                     if (!f->present(doc)) return;
                     if (f->multi) {
                         uint32_t n = f->n_vals(doc);
-                        for (uint32_t i = 0; i < n; ++i)
+                        for (uint32_t i = 0; i < n; ++i) {
                             counts[f->ord_at(doc, i)]++;  // distinct per doc
+                            acc_subs(f->ord_at(doc, i), doc);
+                        }
                         // sum_other base counts VALUE instances (equals the
                         // sum of all bucket doc_counts, like the reference)
                         a.terms_matched_docs += n;
                     } else {
                         counts[f->ord(doc)]++;
+                        acc_subs(f->ord(doc), doc);
                         a.terms_matched_docs++;
                     }
                 };
@@ -567,7 +596,14 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
                 else
                     for (uint32_t doc : m.docs) visit(doc);
                 for (uint32_t o = 0; o < f->cardinality; ++o)
-                    if (counts[o]) a.term_counts.emplace_back(f->dict_entry(o), counts[o]);
+                    if (counts[o]) {
+                        a.term_counts.emplace_back(f->dict_entry(o), counts[o]);
+                        if (with_subs)
+                            a.term_subs.push_back(
+                                subs[o].empty()
+                                    ? std::vector<StatsPayload>(d.sub.size())
+                                    : std::move(subs[o]));
+                    }
                 // dict order is lexicographic => term_counts sorted by key;
                 // per-split split_size truncation + error bound (qagg_format.h)
                 if (d.kind == AggDef::TERMS)

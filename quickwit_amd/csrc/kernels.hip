@@ -331,6 +331,32 @@ __device__ __forceinline__ double agg_value(const QueryDev& q, uint64_t values_o
     }
 }
 
+// accumulate one value into a 40B stats slot {count,sum,min_s,max_s,sum_sq}
+__device__ __forceinline__ void stats_slot_add(uint8_t* slot, double v) {
+    atomicAdd((unsigned long long*)slot, 1ull);
+    atomicAdd((double*)(slot + 8), v);
+    atomicMin((unsigned long long*)(slot + 16),
+              (unsigned long long)f64_sortable(v));
+    atomicMax((unsigned long long*)(slot + 24),
+              (unsigned long long)f64_sortable(v));
+    atomicAdd((double*)(slot + 32), v * v);
+}
+
+// per-term stats sub-aggs: bucket = the term ord (mirrors the histo subs)
+__device__ __forceinline__ void terms_subs_add(const QueryDev& q, const AggDev& a,
+                                               uint64_t ord, uint32_t d) {
+    for (uint32_t si = 0; si < a.n_sub; ++si) {
+        if (!a.sub_width[si]) continue;  // missing sub column
+        if (a.sub_nulls_off[si]) {
+            const uint64_t* sn = (const uint64_t*)(q.split + a.sub_nulls_off[si]);
+            if (!((sn[d >> 6] >> (d & 63)) & 1)) continue;
+        }
+        double sv = agg_value(q, a.sub_values_off[si], a.sub_width[si],
+                              a.sub_is_i64[si], d);
+        stats_slot_add(q.results + a.sub_out + (ord * a.n_sub + si) * 40, sv);
+    }
+}
+
 // DDSketch slot for value v: 0 = zero bucket (v < 1e-9), else 1 + index of
 // the first boundary >= v in the host-computed gamma^k table (identical
 // doubles to the oracle's perc_key_for => bit-identical bucketing)
@@ -634,6 +660,7 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                                                                         a.counts_out) +
                                                       o,
                                                   1ull);
+                                    if (a.n_sub) terms_subs_add(q, a, o, d);
                                 }
                                 if (ai < 4)
                                     atomicAdd(&sc_agg_matched[ai], e2 - s);
@@ -652,6 +679,7 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                             else
                                 atomicAdd((unsigned long long*)(q.results + a.counts_out) + o,
                                           1ull);
+                            if (a.n_sub) terms_subs_add(q, a, o, d);
                             // docs-with-value == matched docs for a
                             // non-nullable column: host uses num_hits instead
                             if (a.nulls_off) {
@@ -812,6 +840,7 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                                 atomicAdd((unsigned long long*)(q.results + a.counts_out) +
                                               idx, 1ull);
                             for (uint32_t si = 0; si < a.n_sub; ++si) {
+                                if (!a.sub_width[si]) continue;  // missing col
                                 if (a.sub_nulls_off[si]) {
                                     const uint64_t* sn =
                                         (const uint64_t*)(q.split + a.sub_nulls_off[si]);

@@ -768,12 +768,40 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
             }
         } else if (d.kind == AggDef::TERMS || d.kind == AggDef::CARDINALITY) {
             a.kind = AGGD_TERMS;
+            if (d.kind == AggDef::TERMS)
+                for (auto& s : d.sub)
+                    if (s.kind == MetricAgg::PERCENTILES)
+                        throw std::runtime_error(
+                            "percentiles under terms (r1 limit)");
             if (f && f->type == FastFieldView::STR) {
                 a.n_buckets = f->cardinality;
                 a.values_off = f->values.off;
                 a.nulls_off = f->nullable ? f->nulls.off : 0;
                 a.value_width = uint32_t(f->ord_width);
                 if (f->multi) a.offsets_off = f->value_offsets.off;
+                if (d.kind == AggDef::TERMS && !d.sub.empty()) {
+                    a.n_sub = uint32_t(std::min<size_t>(d.sub.size(), 4));
+                    if (d.sub.size() > 4)
+                        throw std::runtime_error(
+                            ">4 metric sub-aggregations");
+                    for (size_t si = 0; si < d.sub.size(); ++si) {
+                        const FastFieldView* sf =
+                            sv.fast_field(d.sub[si].field);
+                        if (sf && sf->type != FastFieldView::STR) {
+                            a.sub_values_off[si] = sf->values.off;
+                            a.sub_nulls_off[si] =
+                                sf->nullable ? sf->nulls.off : 0;
+                            a.sub_width[si] = 8;
+                            a.sub_is_i64[si] =
+                                sf->type == FastFieldView::U64   ? 0
+                                : sf->type == FastFieldView::F64 ? 2
+                                                                 : 1;
+                        } else {
+                            a.sub_values_off[si] = 0;
+                            a.sub_width[si] = 0;
+                        }
+                    }
+                }
             } else if (f && !f->multi) {
                 // terms over a numeric fast column (tantivy term_agg keys by
                 // the column value): device hash table sized so distinct
@@ -1997,10 +2025,34 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                             r.term_counts.emplace_back(num_term_key(~0ull), sc2);
                         std::sort(r.term_counts.begin(), r.term_counts.end());
                     } else {
+                        const uint8_t* subs =
+                            agg_out.data() + (a.sub_out - r_agg);
                         for (uint32_t o = 0; o < a.n_buckets; ++o)
-                            if (counts[o])
+                            if (counts[o]) {
                                 r.term_counts.emplace_back(f->dict_entry(o),
                                                            counts[o]);
+                                if (a.n_sub) {
+                                    std::vector<StatsPayload> ts(d.sub.size());
+                                    for (uint32_t s = 0; s < a.n_sub; ++s) {
+                                        const uint8_t* slot =
+                                            subs +
+                                            (uint64_t(o) * a.n_sub + s) * 40;
+                                        StatsPayload sp2;
+                                        uint64_t mn, mx;
+                                        memcpy(&sp2.count, slot, 8);
+                                        memcpy(&sp2.sum, slot + 8, 8);
+                                        memcpy(&mn, slot + 16, 8);
+                                        memcpy(&mx, slot + 24, 8);
+                                        memcpy(&sp2.sum_sq, slot + 32, 8);
+                                        if (sp2.count) {
+                                            sp2.min = u64_to_f64(mn);
+                                            sp2.max = u64_to_f64(mx);
+                                        }
+                                        ts[s] = sp2;
+                                    }
+                                    r.term_subs.push_back(std::move(ts));
+                                }
+                            }
                     }
                     if (d.kind == AggDef::TERMS)
                         truncate_terms_split(

@@ -131,6 +131,9 @@ struct AggResult {
     // keeps numeric order (tantivy terms agg over numeric columns —
     // aggregation/bucket/term_agg.rs keys by the column value)
     uint8_t key_kind = 0;
+    // per-term stats sub-aggs, parallel to term_counts (empty = no subs);
+    // co-permuted by truncate_terms_split and key-merged with the counts
+    std::vector<std::vector<StatsPayload>> term_subs;
     uint64_t terms_matched_docs = 0;
     // sum over truncated splits of the last-included term count — the ES
     // doc_count_error_upper_bound semantics the golden scenario pins
@@ -184,14 +187,26 @@ inline void truncate_terms_split(AggResult& r, int64_t split_size) {
     // hiding terms up to the last included count — the ES semantics the
     // golden scenario pins (split_size=5 over a 5-term split -> error 1).
     if (r.term_counts.size() < uint64_t(split_size)) return;
-    std::stable_sort(r.term_counts.begin(), r.term_counts.end(),
-                     [](const auto& x, const auto& y) {
-                         if (x.second != y.second) return x.second > y.second;
-                         return x.first < y.first;
-                     });
-    r.terms_error_bound += r.term_counts[size_t(split_size) - 1].second;
-    r.term_counts.resize(size_t(split_size));
-    std::sort(r.term_counts.begin(), r.term_counts.end());
+    auto by_count_then_key = [&](size_t x, size_t y) {
+        if (r.term_counts[x].second != r.term_counts[y].second)
+            return r.term_counts[x].second > r.term_counts[y].second;
+        return r.term_counts[x].first < r.term_counts[y].first;
+    };
+    std::vector<size_t> idx(r.term_counts.size());
+    for (size_t i = 0; i < idx.size(); ++i) idx[i] = i;
+    std::stable_sort(idx.begin(), idx.end(), by_count_then_key);
+    r.terms_error_bound += r.term_counts[idx[size_t(split_size) - 1]].second;
+    idx.resize(size_t(split_size));
+    std::sort(idx.begin(), idx.end());  // restore key order (indices ascend)
+    std::vector<std::pair<std::string, uint64_t>> tc;
+    std::vector<std::vector<StatsPayload>> ts;
+    tc.reserve(idx.size());
+    for (size_t i : idx) {
+        tc.push_back(std::move(r.term_counts[i]));
+        if (!r.term_subs.empty()) ts.push_back(std::move(r.term_subs[i]));
+    }
+    r.term_counts = std::move(tc);
+    r.term_subs = std::move(ts);
 }
 
 inline int64_t effective_split_size(uint32_t size, int64_t split_size) {
@@ -246,11 +261,23 @@ struct IntermediateAggResults {
                 put(&a.terms_error_bound, 8);
                 uint32_t ne = uint32_t(a.term_counts.size());
                 put(&ne, 4);
-                for (auto& kv : a.term_counts) {
+                for (size_t ei = 0; ei < a.term_counts.size(); ++ei) {
+                    auto& kv = a.term_counts[ei];
                     uint16_t kl = uint16_t(kv.first.size());
                     put(&kl, 2);
                     put(kv.first.data(), kl);
                     put(&kv.second, 8);
+                    for (size_t s = 0; s < a.sub_names.size(); ++s) {
+                        StatsPayload sp;
+                        if (ei < a.term_subs.size() &&
+                            s < a.term_subs[ei].size())
+                            sp = a.term_subs[ei][s];
+                        put(&sp.count, 8);
+                        put(&sp.sum, 8);
+                        put(&sp.min, 8);
+                        put(&sp.max, 8);
+                        put(&sp.sum_sq, 8);
+                    }
                 }
             } else {
                 uint32_t nb = uint32_t(a.buckets.size());
@@ -351,6 +378,17 @@ struct IntermediateAggResults {
                     uint64_t c;
                     get(&c, 8);
                     a.term_counts.emplace_back(std::move(k), c);
+                    if (!a.sub_names.empty()) {
+                        std::vector<StatsPayload> subs(a.sub_names.size());
+                        for (auto& sp : subs) {
+                            get(&sp.count, 8);
+                            get(&sp.sum, 8);
+                            get(&sp.min, 8);
+                            get(&sp.max, 8);
+                            get(&sp.sum_sq, 8);
+                        }
+                        a.term_subs.push_back(std::move(subs));
+                    }
                 }
             } else {
                 uint32_t nb;
@@ -400,18 +438,34 @@ struct IntermediateAggResults {
             } else if (a.kind == 5) {
                 a.metric.merge(b.metric);
             } else if (a.kind == 3) {
+                bool subs = !a.sub_names.empty() || !b.sub_names.empty();
+                auto sub_of = [](const AggResult& r, size_t i) {
+                    return i < r.term_subs.size()
+                               ? r.term_subs[i]
+                               : std::vector<StatsPayload>(r.sub_names.size());
+                };
                 std::vector<std::pair<std::string, uint64_t>> merged;
+                std::vector<std::vector<StatsPayload>> msubs;
                 merged.reserve(a.term_counts.size() + b.term_counts.size());
                 size_t x = 0, y = 0;
                 while (x < a.term_counts.size() || y < b.term_counts.size()) {
                     if (y >= b.term_counts.size() ||
                         (x < a.term_counts.size() &&
-                         a.term_counts[x].first < b.term_counts[y].first))
+                         a.term_counts[x].first < b.term_counts[y].first)) {
+                        if (subs) msubs.push_back(sub_of(a, x));
                         merged.push_back(a.term_counts[x++]);
-                    else if (x >= a.term_counts.size() ||
-                             b.term_counts[y].first < a.term_counts[x].first)
+                    } else if (x >= a.term_counts.size() ||
+                               b.term_counts[y].first < a.term_counts[x].first) {
+                        if (subs) msubs.push_back(sub_of(b, y));
                         merged.push_back(b.term_counts[y++]);
-                    else {
+                    } else {
+                        if (subs) {
+                            auto sa = sub_of(a, x), sb = sub_of(b, y);
+                            sa.resize(std::max(sa.size(), sb.size()));
+                            for (size_t s = 0; s < sb.size(); ++s)
+                                sa[s].merge(sb[s]);
+                            msubs.push_back(std::move(sa));
+                        }
                         merged.emplace_back(a.term_counts[x].first,
                                             a.term_counts[x].second +
                                                 b.term_counts[y].second);
@@ -420,6 +474,8 @@ struct IntermediateAggResults {
                     }
                 }
                 a.term_counts = std::move(merged);
+                a.term_subs = std::move(msubs);
+                if (a.sub_names.empty()) a.sub_names = b.sub_names;
                 if (!a.key_kind) a.key_kind = b.key_kind;  // splits missing
                                                            // the column
                 a.terms_matched_docs += b.terms_matched_docs;
@@ -738,29 +794,45 @@ inline std::string finalize_aggs_json(const IntermediateAggResults& ir,
         if (d.kind == AggDef::TERMS) {
             // order: doc_count desc, then key asc (ES/tantivy default);
             // truncate to size; sum_other = matched - shown
-            std::vector<std::pair<std::string, uint64_t>> ordered = a.term_counts;
+            std::vector<size_t> ordered(a.term_counts.size());
+            for (size_t i = 0; i < ordered.size(); ++i) ordered[i] = i;
             std::stable_sort(ordered.begin(), ordered.end(),
-                             [](const auto& x, const auto& y) {
-                                 if (x.second != y.second) return x.second > y.second;
-                                 return x.first < y.first;
+                             [&](size_t x, size_t y) {
+                                 if (a.term_counts[x].second !=
+                                     a.term_counts[y].second)
+                                     return a.term_counts[x].second >
+                                            a.term_counts[y].second;
+                                 return a.term_counts[x].first <
+                                        a.term_counts[y].first;
                              });
             uint64_t shown_docs = 0;
             size_t nshow = std::min(size_t(d.size), ordered.size());
             o += "{\"buckets\":[";
-            for (size_t b = 0; b < nshow; ++b) {
-                if (b) o += ",";
+            for (size_t bo = 0; bo < nshow; ++bo) {
+                size_t b = ordered[bo];
+                if (bo) o += ",";
                 o += "{\"doc_count\":";
                 char buf[24];
-                snprintf(buf, sizeof buf, "%llu", (unsigned long long)ordered[b].second);
+                snprintf(buf, sizeof buf, "%llu",
+                         (unsigned long long)a.term_counts[b].second);
                 o += buf;
-                shown_docs += ordered[b].second;
+                shown_docs += a.term_counts[b].second;
+                for (size_t s = 0; s < d.sub.size(); ++s) {
+                    o += ",";
+                    mj::escape_to(o, d.sub[s].name);
+                    o += ":";
+                    StatsPayload sp;
+                    if (b < a.term_subs.size() && s < a.term_subs[b].size())
+                        sp = a.term_subs[b][s];
+                    stats_to_json(o, d.sub[s], sp);
+                }
                 o += ",\"key\":";
                 if (a.key_kind) {
                     // numeric column key: decode the sortable big-endian
                     // bits back to the column value; u64/i64 printed as JSON
                     // integers to keep full 64-bit precision (the
                     // high_prec_test golden pins this)
-                    uint64_t bits = num_term_key_bits(ordered[b].first);
+                    uint64_t bits = num_term_key_bits(a.term_counts[b].first);
                     char nbuf[32];
                     if (a.key_kind == 1)
                         snprintf(nbuf, sizeof nbuf, "%llu",
@@ -774,7 +846,7 @@ inline std::string finalize_aggs_json(const IntermediateAggResults& ir,
                     }
                     o += nbuf;
                 } else {
-                    mj::escape_to(o, ordered[b].first);
+                    mj::escape_to(o, a.term_counts[b].first);
                 }
                 o += "}";
             }

@@ -1264,3 +1264,62 @@ def test_percentiles_aggs():
     assert vals == sorted(vals) and 0 <= vals[0] <= 5100
     top = gj["lat_top"]["values"]
     assert set(top) == {"1.0", "5.0", "25.0", "50.0", "75.0", "95.0", "99.0"}
+
+
+def test_terms_sub_aggregations():
+    # stats-family sub-aggs under terms buckets (per-ord 40B slots on device
+    # vs the oracle), incl. nullable sub columns and f64 sums
+    import random
+    rng = random.Random(9)
+    docs = []
+    for i in range(3000):
+        d = {"timestamp": 1700000000 + i, "severity_text":
+             "INFO" if i % 5 else "WARN", "body": "x", "tenant_id": i % 4,
+             "svc": rng.choice(["api", "ing", "jan", "ui"])}
+        if rng.random() < 0.8:
+            d["lat"] = round(rng.uniform(1, 100), 2)
+        docs.append(d)
+    schema = {"timestamp_field": "timestamp", "fields":
+              splitgen.HDFS_SCHEMA["fields"] +
+              [{"name": "svc", "type": "str", "fast": True},
+               {"name": "lat", "type": "f64", "fast": True}]}
+    gpu, cpu = GpuSearcher(device=0), OracleSearcher()
+    splits = []
+    for s in range(2):
+        w = splitgen.SplitWriter(schema, f"tsub-{s}")
+        w.add_documents(docs[s::2])
+        data = w.finalize()
+        gpu.add_split(f"tsub-{s}", data)
+        cpu.add_split(f"tsub-{s}", data)
+        splits.append((f"tsub-{s}", len(docs[s::2])))
+    aggs = {"by_svc": {"terms": {"field": "svc", "size": 3},
+                       "aggs": {"lat_stats": {"stats": {"field": "lat"}},
+                                "lat_ext": {"extended_stats":
+                                            {"field": "lat"}},
+                                "lat_cnt": {"value_count":
+                                            {"field": "lat"}}}}}
+    for q in ({"type": "match_all"},
+              {"type": "term", "field": "severity_text", "value": "WARN"}):
+        req = make_leaf_request(q, schema, splits, max_hits=0,
+                                aggregation=aggs)
+        g = gpu.leaf_search(req)
+        e = cpu.leaf_search(req)
+        gj = gpu.finalize_agg_json(g["intermediate_aggregation_result"], aggs)
+        ej = cpu.finalize_agg_json(e["intermediate_aggregation_result"], aggs)
+        for gb, eb in zip(gj["by_svc"]["buckets"], ej["by_svc"]["buckets"]):
+            assert gb["key"] == eb["key"]
+            assert gb["doc_count"] == eb["doc_count"]
+            for nm in ("lat_stats", "lat_ext"):
+                for k2, ev in eb[nm].items():
+                    gv = gb[nm][k2]
+                    if isinstance(ev, dict):
+                        for k3 in ev:
+                            assert math.isclose(gv[k3], ev[k3],
+                                                rel_tol=1e-9), (nm, k2, k3)
+                    elif isinstance(ev, float):
+                        assert math.isclose(gv, ev, rel_tol=1e-9), (nm, k2)
+                    else:
+                        assert gv == ev, (nm, k2)
+            assert gb["lat_cnt"] == eb["lat_cnt"]
+        assert gj["by_svc"]["sum_other_doc_count"] == \
+            ej["by_svc"]["sum_other_doc_count"]
