@@ -805,6 +805,15 @@ inline std::string finalize_aggs_json(const IntermediateAggResults& ir,
                                  return a.term_counts[x].first <
                                         a.term_counts[y].first;
                              });
+            // min_doc_count (ES default 1 for terms): drop below-threshold
+            // buckets BEFORE the size cut
+            int64_t mdc = d.min_doc_count < 0 ? 1 : d.min_doc_count;
+            ordered.erase(std::remove_if(ordered.begin(), ordered.end(),
+                                         [&](size_t i) {
+                                             return int64_t(a.term_counts[i]
+                                                                .second) < mdc;
+                                         }),
+                          ordered.end());
             uint64_t shown_docs = 0;
             size_t nshow = std::min(size_t(d.size), ordered.size());
             o += "{\"buckets\":[";
