@@ -1,0 +1,139 @@
+"""Cross-split aggregation merge equivalence (CPU): for every aggregation
+family, merging per-split leaf responses through the PRODUCT's ctx-less
+root-side merge (qw_merge_leaf_responses, the path bench.py's multi-GPU
+reduce uses) must finalize to exactly what a single multi-split leaf search
+produces. Covers the QAGG wire format round-trip (encode -> merge decode ->
+re-encode -> finalize) for terms+subs, numeric terms, cardinality,
+composite, percentiles and histograms.
+
+Reference semantics: collector.rs:832-861 merge_fruits + tantivy
+intermediate aggregation merge.
+"""
+import json
+import os
+import random
+import sys
+
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from quickwit_amd import proto, splitgen
+from quickwit_amd.api import OracleSearcher, make_leaf_request
+from quickwit_amd.merge import merge_leaf_responses
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+SCHEMA = {"timestamp_field": "timestamp", "fields":
+          splitgen.HDFS_SCHEMA["fields"] +
+          [{"name": "svc", "type": "str", "fast": True},
+           {"name": "lat", "type": "f64", "fast": True},
+           {"name": "code", "type": "u64", "fast": True}]}
+
+
+def corpus():
+    rng = random.Random(21)
+    docs = []
+    for i in range(1200):
+        d = {"timestamp": 1700000000 + i * 7, "severity_text":
+             "INFO" if i % 3 else "ERROR", "body": "x", "tenant_id": i % 5,
+             "svc": rng.choice(["api", "ing", "jan"]),
+             "code": rng.choice([200, 204, 404, 500, 1769070189829214201])}
+        if rng.random() < 0.85:
+            d["lat"] = round(rng.uniform(0.5, 900.0), 3)
+        docs.append(d)
+    return docs
+
+
+@pytest.fixture(scope="module")
+def setup():
+    import subprocess
+    subprocess.run(["make", "-s", "-C", os.path.join(REPO, "oracle")],
+                   check=True)
+    docs = corpus()
+    parts = [docs[0::3], docs[1::3], docs[2::3]]
+    datas = []
+    for i, part in enumerate(parts):
+        w = splitgen.SplitWriter(SCHEMA, f"m-{i}")
+        w.add_documents(part)
+        datas.append(w.finalize())
+    combined = OracleSearcher()
+    singles = []
+    for i, data in enumerate(datas):
+        combined.add_split(f"m-{i}", data)
+        s = OracleSearcher()
+        s.add_split(f"m-{i}", data)
+        singles.append(s)
+    splits = [(f"m-{i}", len(parts[i])) for i in range(3)]
+    return combined, singles, splits
+
+
+AGG_CASES = {
+    "histo_stats": {"h": {"date_histogram": {"field": "timestamp",
+                          "fixed_interval": "1000000ms"},
+                          "aggs": {"st": {"stats": {"field": "lat"}}}}},
+    "terms_subs": {"t": {"terms": {"field": "svc", "size": 10},
+                         "aggs": {"st": {"extended_stats": {"field": "lat"}},
+                                  "mx": {"max": {"field": "lat"}}}}},
+    "terms_numeric": {"n": {"terms": {"field": "code", "size": 10}}},
+    "terms_truncated": {"t": {"terms": {"field": "svc", "size": 2,
+                                        "split_size": 2}}},
+    "cardinality": {"c1": {"cardinality": {"field": "svc"}},
+                    "c2": {"cardinality": {"field": "code"}},
+                    "c3": {"cardinality": {"field": "lat"}}},
+    "composite": {"c": {"composite": {"size": 50, "sources": [
+        {"s": {"terms": {"field": "svc"}}},
+        {"r": {"histogram": {"field": "code", "interval": 100}}}]}}},
+    "percentiles": {"p": {"percentiles": {"field": "lat",
+                                          "percents": [50, 95]}},
+                    "h": {"date_histogram": {"field": "timestamp",
+                          "fixed_interval": "1000000ms"},
+                          "aggs": {"lp": {"percentiles": {"field": "lat",
+                                          "keyed": False}}}}},
+    "range_metric": {"r": {"range": {"field": "code", "ranges": [
+                          {"to": 300.0}, {"from": 300.0, "to": 600.0},
+                          {"from": 600.0}]}},
+                     "m": {"avg": {"field": "lat"}}},
+}
+
+
+@pytest.mark.parametrize("case", sorted(AGG_CASES))
+def test_merged_equals_combined(setup, case):
+    combined, singles, splits = setup
+    aggs = AGG_CASES[case]
+    for q in ({"type": "match_all"},
+              {"type": "term", "field": "severity_text", "value": "ERROR"}):
+        req_all = make_leaf_request(q, SCHEMA, splits, max_hits=5,
+                                    aggregation=aggs)
+        sreq_pb = proto.encode("SearchRequest", req_all["search_request"])
+        want = combined.leaf_search(req_all)
+        wj = combined.finalize_agg_json(
+            want["intermediate_aggregation_result"], aggs)
+
+        resps = []
+        for i, s in enumerate(singles):
+            req1 = make_leaf_request(q, SCHEMA, [splits[i]], max_hits=5,
+                                     aggregation=aggs)
+            resps.append(s.leaf_search_raw(proto.encode("LeafSearchRequest",
+                                                        req1)))
+        merged_pb = merge_leaf_responses(sreq_pb, resps)
+        merged = proto.decode("LeafSearchResponse", merged_pb)
+        assert merged.get("num_hits", 0) == want.get("num_hits", 0), case
+        # finalize the merged blob through the PRODUCT's ctx-less ABI
+        import ctypes
+
+        from quickwit_amd.merge import _Buf, _get_lib
+        lib = _get_lib()
+        lib.qw_finalize_agg_to_json.argtypes = [
+            ctypes.c_char_p, ctypes.c_size_t, ctypes.c_char_p,
+            ctypes.POINTER(_Buf)]
+        buf = _Buf()
+        blob = merged.get("intermediate_aggregation_result", b"")
+        rc = lib.qw_finalize_agg_to_json(blob, len(blob),
+                                         json.dumps(aggs).encode(),
+                                         ctypes.byref(buf))
+        assert rc == 0, case
+        out = ctypes.string_at(buf.data, buf.len)
+        lib.qw_buf_free(ctypes.byref(buf))
+        got = json.loads(out)
+        assert got == wj, (case, q, got, wj)
