@@ -607,8 +607,9 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
                 // dict order is lexicographic => term_counts sorted by key;
                 // per-split split_size truncation + error bound (qagg_format.h)
                 if (d.kind == AggDef::TERMS)
-                    truncate_terms_split(a,
-                                         effective_split_size(d.size, d.split_size));
+                    truncate_terms_split(
+                        a, effective_split_size(d.size, d.split_size),
+                        d.order_target, d.order_asc);
             } else if (f && !f->multi) {
                 // terms over a numeric fast column: count by the value's
                 // order-preserving sortable bits, keys encoded big-endian so
@@ -640,8 +641,9 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
                 for (auto& kv : counts)
                     a.term_counts.emplace_back(num_term_key(kv.first), kv.second);
                 if (d.kind == AggDef::TERMS)
-                    truncate_terms_split(a,
-                                         effective_split_size(d.size, d.split_size));
+                    truncate_terms_split(
+                        a, effective_split_size(d.size, d.split_size),
+                        d.order_target, d.order_asc);
             }
             out.aggs.push_back(std::move(a));
             continue;

@@ -180,8 +180,27 @@ inline void comp_encode_f64(std::string& o, double v) {
 // per-split terms truncation (tantivy terms agg split_size): keep the top
 // `split_size` entries by (count desc, key asc); when anything is dropped,
 // the last included count joins the error bound. Restores key order after.
-inline void truncate_terms_split(AggResult& r, int64_t split_size) {
+inline void truncate_terms_split(AggResult& r, int64_t split_size,
+                                 const std::string& order_target = "",
+                                 bool order_asc = false) {
     if (split_size < 1) split_size = 1;
+    if (order_target == "_key") {
+        // key order: the per-split prefix (or suffix) in key order is EXACT
+        // — no doc-count error contribution
+        if (r.term_counts.size() <= uint64_t(split_size)) return;
+        if (!order_asc) {
+            size_t drop = r.term_counts.size() - size_t(split_size);
+            r.term_counts.erase(r.term_counts.begin(),
+                                r.term_counts.begin() + drop);
+            if (!r.term_subs.empty())
+                r.term_subs.erase(r.term_subs.begin(),
+                                  r.term_subs.begin() + drop);
+        } else {
+            r.term_counts.resize(size_t(split_size));
+            if (!r.term_subs.empty()) r.term_subs.resize(size_t(split_size));
+        }
+        return;
+    }
     // strictly fewer terms than the cap: the split reported everything and
     // contributes no error. A FULL list (== cap, even untruncated) could be
     // hiding terms up to the last included count — the ES semantics the
@@ -793,15 +812,25 @@ inline std::string finalize_aggs_json(const IntermediateAggResults& ir,
         }
         if (d.kind == AggDef::TERMS) {
             // order: doc_count desc, then key asc (ES/tantivy default);
-            // truncate to size; sum_other = matched - shown
+            // explicit "order": _count asc/desc or _key asc/desc; truncate
+            // to size; sum_other = matched - shown
             std::vector<size_t> ordered(a.term_counts.size());
             for (size_t i = 0; i < ordered.size(); ++i) ordered[i] = i;
             std::stable_sort(ordered.begin(), ordered.end(),
                              [&](size_t x, size_t y) {
+                                 if (d.order_target == "_key")
+                                     return d.order_asc
+                                                ? a.term_counts[x].first <
+                                                      a.term_counts[y].first
+                                                : a.term_counts[y].first <
+                                                      a.term_counts[x].first;
                                  if (a.term_counts[x].second !=
                                      a.term_counts[y].second)
-                                     return a.term_counts[x].second >
-                                            a.term_counts[y].second;
+                                     return d.order_asc
+                                                ? a.term_counts[x].second <
+                                                      a.term_counts[y].second
+                                                : a.term_counts[x].second >
+                                                      a.term_counts[y].second;
                                  return a.term_counts[x].first <
                                         a.term_counts[y].first;
                              });

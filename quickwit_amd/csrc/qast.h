@@ -905,6 +905,10 @@ struct AggDef {
     bool has_bounds = false;
     double bmin = 0, bmax = 0;
     uint32_t size = 10;          // terms
+    // terms bucket order: "" = _count (default), "_key" = by term key;
+    // order by a sub-aggregation value: rejected (r1 limit)
+    std::string order_target;
+    bool order_asc = false;      // _count default desc; explicit order sets it
     int64_t split_size = -1;     // terms: per-split truncation; -1 = default
                                  // (size*3/2+10, the ES shard_size default)
     int64_t min_doc_count = -1;  // -1 = default (0 for histos, 1 for terms)
@@ -1066,6 +1070,27 @@ inline std::vector<AggDef> parse_agg_request(const std::string& json) {
                 a.split_size = sh->as_i64();
             if (const mj::Value* mdc = spec->get("min_doc_count"))
                 a.min_doc_count = mdc->as_i64();
+            if (const mj::Value* od = spec->get("order")) {
+                // ES: {"order": {"_key": "asc"}} (single criterion; arrays
+                // and sub-aggregation targets: later round)
+                const mj::Value* body2 = od;
+                if (od->kind == mj::Value::ARR) {
+                    if (od->arr.size() != 1)
+                        throw std::runtime_error(
+                            "terms order: multiple criteria (r1 limit)");
+                    body2 = od->arr[0].get();
+                }
+                if (body2->obj.size() != 1)
+                    throw std::runtime_error("terms order: one criterion");
+                const std::string& tgt = body2->obj.begin()->first;
+                const std::string dir = body2->obj.begin()->second->s;
+                if (tgt == "_count") a.order_target = "";
+                else if (tgt == "_key" || tgt == "_term") a.order_target = "_key";
+                else
+                    throw std::runtime_error(
+                        "terms order by sub-aggregation (r1 limit)");
+                a.order_asc = dir == "asc";
+            }
         } else {
             // top-level metric aggregations (stats/avg/sum/min/max/
             // value_count/extended_stats over every matched doc)

@@ -94,6 +94,14 @@ AGG_CASES = {
                           {"to": 300.0}, {"from": 300.0, "to": 600.0},
                           {"from": 600.0}]}},
                      "m": {"avg": {"field": "lat"}}},
+    "terms_key_asc": {"t": {"terms": {"field": "svc", "size": 2,
+                                      "split_size": 2,
+                                      "order": {"_key": "asc"}}}},
+    "terms_key_desc": {"t": {"terms": {"field": "svc", "size": 2,
+                                       "split_size": 2,
+                                       "order": {"_key": "desc"}}}},
+    "terms_count_asc": {"t": {"terms": {"field": "svc", "size": 10,
+                                        "order": {"_count": "asc"}}}},
 }
 
 
@@ -137,3 +145,46 @@ def test_merged_equals_combined(setup, case):
         lib.qw_buf_free(ctypes.byref(buf))
         got = json.loads(out)
         assert got == wj, (case, q, got, wj)
+
+
+def test_terms_order_key(setup):
+    combined, singles, splits = setup
+    for direction, keys in (("asc", ["api", "ing", "jan"]),
+                            ("desc", ["jan", "ing", "api"])):
+        aggs = {"t": {"terms": {"field": "svc", "size": 10,
+                                "order": {"_key": direction}}}}
+        req = make_leaf_request({"type": "match_all"}, SCHEMA, splits,
+                                max_hits=0, aggregation=aggs)
+        r = combined.leaf_search(req)
+        out = combined.finalize_agg_json(
+            r["intermediate_aggregation_result"], aggs)
+        assert [b["key"] for b in out["t"]["buckets"]] == keys
+
+
+def test_terms_order_count_asc(setup):
+    combined, singles, splits = setup
+    aggs = {"t": {"terms": {"field": "svc", "size": 10,
+                            "order": {"_count": "asc"}}}}
+    req = make_leaf_request({"type": "match_all"}, SCHEMA, splits,
+                            max_hits=0, aggregation=aggs)
+    r = combined.leaf_search(req)
+    out = combined.finalize_agg_json(r["intermediate_aggregation_result"],
+                                     aggs)
+    counts = [b["doc_count"] for b in out["t"]["buckets"]]
+    assert counts == sorted(counts)
+
+
+def test_terms_order_by_subagg_rejected(setup):
+    combined, singles, splits = setup
+    aggs = {"t": {"terms": {"field": "svc",
+                            "order": {"st.avg": "desc"},
+                            "aggs": {"st": {"stats": {"field": "lat"}}}}}}
+    req = make_leaf_request({"type": "match_all"}, SCHEMA, splits,
+                            max_hits=0, aggregation=aggs)
+    try:
+        r = combined.leaf_search(req)
+        # per-split failure reporting (leaf.rs incremental collector): every
+        # split must have failed with the r1-limit error
+        assert len(r.get("failed_splits", [])) == len(splits), r
+    except RuntimeError:
+        pass  # whole-request rejection is equally acceptable
