@@ -1323,3 +1323,21 @@ def test_terms_sub_aggregations():
             assert gb["lat_cnt"] == eb["lat_cnt"]
         assert gj["by_svc"]["sum_other_doc_count"] == \
             ej["by_svc"]["sum_other_doc_count"]
+
+
+def test_terms_order_key_on_gpu(searchers):
+    # order=_key flows through the product's per-split truncation (assembly)
+    gpu, cpu = searchers
+    for direction in ("asc", "desc"):
+        aggs = {"t": {"terms": {"field": "tenant_name", "size": 5,
+                                "split_size": 5,
+                                "order": {"_key": direction}}}}
+        req = make_leaf_request({"type": "match_all"}, SCHEMA, [(SID, NDOCS)],
+                                max_hits=0, aggregation=aggs)
+        g = gpu.leaf_search(req)
+        e = cpu.leaf_search(req)
+        gj = gpu.finalize_agg_json(g["intermediate_aggregation_result"], aggs)
+        ej = cpu.finalize_agg_json(e["intermediate_aggregation_result"], aggs)
+        assert gj == ej, (direction, gj, ej)
+        keys = [b["key"] for b in gj["t"]["buckets"]]
+        assert keys == sorted(keys, reverse=direction == "desc")
