@@ -209,3 +209,36 @@ def test_rest_simple_suites_gpu(suite):
     import __graft_entry__
     __graft_entry__.build()
     run_simple_suite(suite, lambda: GpuSearcher(device=0))
+
+
+def test_rest_terms_with_subaggs_oracle():
+    # terms buckets with stats subs through the ES _search surface
+    from quickwit_amd.api import OracleSearcher
+    import __graft_entry__
+    __graft_entry__.build()
+    client = make_client(OracleSearcher)
+    r = client.request("POST", "/api/v1/indexes", json={
+        "version": "0.8", "index_id": "tsub",
+        "doc_mapping": {"mode": "dynamic",
+                        "dynamic_mapping": {"fast": True,
+                                            "tokenizer": "default"}}})
+    assert r.status_code == 200, r.text
+    nd = "\n".join(
+        json.dumps({"svc": ["api", "ing", "jan"][i % 3],
+                    "lat": float(10 * (i % 7))}) for i in range(60)) + "\n"
+    r = client.request("POST", "/api/v1/tsub/ingest", content=nd)
+    assert r.status_code == 200, r.text
+    r = client.request("GET", "/api/v1/_elastic/tsub/_search", json={
+        "size": 0,
+        "aggs": {"by_svc": {"terms": {"field": "svc", "size": 10,
+                                      "order": {"_key": "asc"}},
+                            "aggs": {"lat_avg": {"avg": {"field": "lat"}},
+                                     "lat_st": {"stats": {"field": "lat"}}}}}})
+    assert r.status_code == 200, r.text
+    buckets = r.json()["aggregations"]["by_svc"]["buckets"]
+    assert [b["key"] for b in buckets] == ["api", "ing", "jan"]
+    assert sum(b["doc_count"] for b in buckets) == 60
+    for b in buckets:
+        assert b["lat_st"]["count"] == b["doc_count"]
+        assert abs(b["lat_avg"]["value"] -
+                   b["lat_st"]["sum"] / b["lat_st"]["count"]) < 1e-9
