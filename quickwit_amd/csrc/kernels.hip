@@ -616,14 +616,21 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                     int64_t idx = int64_t(floor(double(num) * af_inv));
                     if (idx * af_ivl > num) --idx;
                     else if ((idx + 1) * af_ivl <= num) ++idx;
-                    atomicAdd(&sc_agg_hist[uint32_t(idx - af_base)], 1u);
+                    uint32_t hb = uint32_t(idx - af_base);
+                    // same lds_rep mapping as the pure-agg tile loop above:
+                    // the end-of-kernel flush sums interleaved pairs when
+                    // lds_rep==2, so indices must match there too
+                    atomicAdd(&sc_agg_hist[af_hrep == 2 ? hb * 2 + af_par : hb],
+                              1u);
                     if (af_terms) {
                         uint64_t o = af_twidth == 2
                                          ? ((const uint16_t*)af_tcol)[d]
                                          : (af_twidth == 1
                                                 ? af_tcol[d]
                                                 : ((const uint32_t*)af_tcol)[d]);
-                        atomicAdd(&sc_agg_terms[o], 1u);
+                        atomicAdd(
+                            &sc_agg_terms[af_trep == 2 ? o * 2 + af_par : o],
+                            1u);
                     }
                 } else if (NA)
                     for (uint32_t ai = 0; ai < q.n_aggs; ++ai) {
@@ -635,8 +642,12 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                         if (a.kind == AGGD_TERMS) {
                             if (!a.n_buckets) continue;  // missing/non-str column
                             if (a.offsets_off) {
-                                // multi-valued: one count per (doc, distinct
-                                // value); doc counted once in matched
+                                // multi-valued: one count per (doc, value) in
+                                // both the per-bucket counts AND matched
+                                // (e2-s below) — matched must sum the same
+                                // units as the buckets so host-side
+                                // sum_other_doc_count = matched - Σ(kept
+                                // bucket counts) stays consistent
                                 const uint32_t* offs =
                                     (const uint32_t*)(q.split + a.offsets_off);
                                 uint32_t s = offs[d], e2 = offs[d + 1];

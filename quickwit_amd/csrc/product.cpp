@@ -193,11 +193,23 @@ static std::string leaf_cache_key(const pb::SplitIdAndFooterOffsets& so,
 // MI355X-native PredicateCache (cache_node.rs:202-486).
 // Count-capped LRU; a 100M-doc bitmap is 12.5 MB of HBM.
 struct HitsetCache {
+    // Count-capped LRU of device bitmaps. Entries touched by the CURRENT
+    // query (gen == cur_gen) are pinned: resolve_hitset/bitmap_eval return
+    // raw device pointers that the later main-kernel launch reads, so an
+    // eviction between resolve and launch would free in-use device memory.
+    // begin_query() unpins everything from previous queries.
+    struct Entry {
+        std::string key;
+        uint8_t* bm;
+        uint64_t gen;
+    };
     size_t cap = 256;
-    std::list<std::pair<std::string, uint8_t*>> lru;
-    std::map<std::string, std::list<std::pair<std::string, uint8_t*>>::iterator> idx;
+    uint64_t cur_gen = 0;
+    std::list<Entry> lru;
+    std::map<std::string, std::list<Entry>::iterator> idx;
     uint64_t hits = 0, misses = 0;
 
+    void begin_query() { ++cur_gen; }
     uint8_t* get(const std::string& key) {
         auto it = idx.find(key);
         if (it == idx.end()) {
@@ -205,30 +217,37 @@ struct HitsetCache {
             return nullptr;
         }
         ++hits;
+        it->second->gen = cur_gen;
         lru.splice(lru.begin(), lru, it->second);
-        return it->second->second;
+        return it->second->bm;
     }
     void put(const std::string& key, uint8_t* bm) {
-        lru.emplace_front(key, bm);
+        lru.push_front(Entry{key, bm, cur_gen});
         idx[key] = lru.begin();
-        while (lru.size() > cap) {
-            (void)hipFree(lru.back().second);
-            idx.erase(lru.back().first);
-            lru.pop_back();
+        // evict oldest UNPINNED entries; a query resolving > cap distinct
+        // subtrees temporarily overflows the cap rather than freeing memory
+        // the pending launch still references
+        for (auto it = std::prev(lru.end());
+             lru.size() > cap && it != lru.begin();) {
+            auto cur = it--;
+            if (cur->gen == cur_gen) continue;
+            (void)hipFree(cur->bm);
+            idx.erase(cur->key);
+            lru.erase(cur);
         }
     }
     void remove_split(const std::string& split_id) {
         std::string prefix = split_id + '\0';
         for (auto it = lru.begin(); it != lru.end();) {
-            if (it->first.compare(0, prefix.size(), prefix) == 0) {
-                (void)hipFree(it->second);
-                idx.erase(it->first);
+            if (it->key.compare(0, prefix.size(), prefix) == 0) {
+                (void)hipFree(it->bm);
+                idx.erase(it->key);
                 it = lru.erase(it);
             } else ++it;
         }
     }
     void clear() {
-        for (auto& kv : lru) (void)hipFree(kv.second);
+        for (auto& e : lru) (void)hipFree(e.bm);
         lru.clear();
         idx.clear();
     }
@@ -1347,6 +1366,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         tprev = now;
     };
     const SplitView& sv = ds.view;
+    ctx->hitsets.begin_query();  // unpin previous query's bitmap entries
 
     PlanNode plan = parse_query_ast(req.query_ast, schema);
     if ((req.start_timestamp || req.end_timestamp) && !schema.timestamp_field.empty()) {
@@ -1803,7 +1823,11 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             uint32_t* d_scount = (uint32_t*)(ctx->d_results.p + r_cand_count) + 1;
             uint32_t rw = wide ? 2 : 1;  // u64 words per candidate record
             sel_band_n = cand_n;
-            rerun_select = [&, d_cand, rw](uint64_t Kwant) {
+            // capture the block-local selection state BY VALUE: the lambda is
+            // re-invoked by the search_after growing-K retry (below) after
+            // this block's scope has ended
+            rerun_select = [&, d_cand, rw, cand_n, d_hist, d_scount,
+                            hist0_valid](uint64_t Kwant) {
             uint64_t K = std::min<uint64_t>(Kwant, cand_n);
             uint64_t prefix = 0;
             uint32_t prefix_bits = 0;

@@ -1341,3 +1341,64 @@ def test_terms_order_key_on_gpu(searchers):
         assert gj == ej, (direction, gj, ej)
         keys = [b["key"] for b in gj["t"]["buckets"]]
         assert keys == sorted(keys, reverse=direction == "desc")
+
+
+# ---------------------------------------------------------------- agg_fast
+# Regression: the agg_fast EPILOGUE branch (any query that is not pure
+# match_all-no-collect: filters, or match_all + hit collection) must apply
+# the same lds_rep bucket-index mapping as the pure-agg tile loop; when
+# n_buckets*2 <= AGG_LDS_BUCKETS the LDS histogram is 2-way replicated and
+# the end-of-kernel flush sums interleaved pairs. Bare aggs (no sub-aggs)
+# keep agg_fast=1; the per-hour histogram (~720 buckets) gets lds_rep=2.
+BARE_AGGS = {
+    "per_hour": {"date_histogram": {"field": "timestamp",
+                                    "fixed_interval": "3600000ms"}},
+    "per_tenant": {"terms": {"field": "tenant_name", "size": 10}},
+}
+
+
+def _bare_agg_both(searchers, query, max_hits=0):
+    gpu, cpu = searchers
+    req = make_leaf_request(query, SCHEMA, [(SID, NDOCS)],
+                            max_hits=max_hits, aggregation=BARE_AGGS,
+                            sort_fields=None)
+    g = gpu.leaf_search(req)
+    e = cpu.leaf_search(req)
+    gj = gpu.finalize_agg_json(g["intermediate_aggregation_result"], BARE_AGGS)
+    ej = cpu.finalize_agg_json(e["intermediate_aggregation_result"], BARE_AGGS)
+    assert g.get("num_hits", 0) == e.get("num_hits", 0)
+    return gj, ej
+
+
+def test_agg_fast_bare_histogram_under_term_filter(searchers):
+    q = {"type": "term", "field": "severity_text", "value": "INFO"}
+    gj, ej = _bare_agg_both(searchers, q)
+    assert gj == ej
+
+
+def test_agg_fast_bare_histogram_under_range_filter(searchers):
+    q = {"type": "range", "field": "tenant_id",
+         "lower_bound": {"included": 100}, "upper_bound": {"excluded": 600}}
+    gj, ej = _bare_agg_both(searchers, q)
+    assert gj == ej
+
+
+def test_agg_fast_bare_histogram_match_all_with_hits(searchers):
+    # match_all + max_hits>0 routes through the epilogue (collect), not the
+    # pure-agg tile loop
+    gj, ej = _bare_agg_both(searchers, {"type": "match_all"}, max_hits=10)
+    assert gj == ej
+
+
+def test_agg_fast_bare_histogram_only_under_filter(searchers):
+    # histogram WITHOUT the terms agg (af_terms=false leg)
+    gpu, cpu = searchers
+    aggs = {"per_hour": BARE_AGGS["per_hour"]}
+    q = {"type": "term", "field": "severity_text", "value": "ERROR"}
+    req = make_leaf_request(q, SCHEMA, [(SID, NDOCS)], max_hits=0,
+                            aggregation=aggs)
+    g = gpu.leaf_search(req)
+    e = cpu.leaf_search(req)
+    gj = gpu.finalize_agg_json(g["intermediate_aggregation_result"], aggs)
+    ej = cpu.finalize_agg_json(e["intermediate_aggregation_result"], aggs)
+    assert gj == ej
