@@ -1,9 +1,11 @@
 """Multi-rank merge path on CPU (gloo, world_size=2) — the exact exchange
 bench.py runs over RCCL on the 8-GPU node: each rank searches
-its own split, responses are all-gathered, rank 0 merges through
-qw_merge_leaf_responses. Oracle searchers stand in for the GPU path here
+its own split, then merge.distributed_merge exchanges PACKED TENSORS only
+(32 B top-K hit records allgather, dense histogram bucket arrays
+SUM/MIN/MAX-reduced, sparse aggs + stats as a sideband byte tensor) and
+rank 0 merges through qw_merge_leaf_responses. No pickled objects cross
+ranks on the data path. Oracle searchers stand in for the GPU path here
 (same protobuf surfaces); nccl/RCCL only swaps the transport."""
-import json
 import os
 
 import pytest
@@ -13,61 +15,139 @@ from quickwit_amd import proto, splitgen
 
 NDOCS = 5_000
 
+AGGS = {"per_hour": {"date_histogram": {"field": "timestamp",
+                                        "fixed_interval": "3600000ms"}},
+        "per_tenant": {"terms": {"field": "tenant_name", "size": 10}},
+        "stats_hour": {"date_histogram": {"field": "timestamp",
+                                          "fixed_interval": "7200000ms"},
+                       "aggs": {"ten": {"stats": {"field": "tenant_id"}}}}}
 
-def _rank_main(rank, world, q, result):
+CASES = {
+    "bm25": dict(
+        q={"type": "bool", "should": [
+            {"type": "term", "field": "body", "value": "w%05d" % i}
+            for i in range(3)]},
+        sort=[{"field_name": "_score", "sort_order": 1}], aggs=None),
+    "agg": dict(q={"type": "match_all"}, sort=None, aggs=AGGS),
+    "field_sort": dict(
+        q={"type": "term", "field": "severity_text", "value": "INFO"},
+        sort=[{"field_name": "timestamp", "sort_order": 1},
+              {"field_name": "tenant_id", "sort_order": 0}], aggs=None),
+}
+
+
+def _make_req(case, splits):
+    from quickwit_amd.api import make_leaf_request
+    c = CASES[case]
+    return make_leaf_request(c["q"], splitgen.HDFS_SCHEMA, splits, max_hits=15,
+                             sort_fields=c["sort"], aggregation=c["aggs"])
+
+
+def _rank_main(rank, world, case, port, result):
     import torch.distributed as dist
 
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-    os.environ.setdefault("MASTER_PORT", "29511")
+    os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
-    from quickwit_amd.api import OracleSearcher, make_leaf_request
-    from quickwit_amd.merge import merge_leaf_responses
+    from quickwit_amd.api import OracleSearcher
+    from quickwit_amd.merge import distributed_merge
 
     sid = f"synthetic-3-{rank:04d}"
     data = splitgen.generate_split(rank, NDOCS, seed=3)
     s = OracleSearcher()
     s.add_split(sid, data)
-    req = make_leaf_request(q, splitgen.HDFS_SCHEMA, [(sid, NDOCS)], max_hits=15,
-                            sort_fields=[{"field_name": "_score", "sort_order": 1}])
+    req = _make_req(case, [(sid, NDOCS)])
     resp_pb = s.leaf_search_raw(proto.encode("LeafSearchRequest", req))
 
-    gathered = [None] * world
-    dist.all_gather_object(gathered, resp_pb)
+    sreq_pb = proto.encode("SearchRequest", req["search_request"])
+    import torch
+    merged = distributed_merge(sreq_pb, resp_pb, [sid],
+                               device=torch.device("cpu"))
     if rank == 0:
-        sreq_pb = proto.encode("SearchRequest", req["search_request"])
-        merged = merge_leaf_responses(sreq_pb, gathered)
-        result.put(proto.decode("LeafSearchResponse", merged))
+        result.put(merged)
     dist.destroy_process_group()
 
 
-def test_two_rank_gloo_merge_equals_single_call():
+def _run_case(case, port):
     import __graft_entry__
     __graft_entry__.build()
-    q = {"type": "bool", "should": [
-        {"type": "term", "field": "body", "value": "w%05d" % i} for i in range(3)]}
-
     ctx = mp.get_context("spawn")
     result = ctx.Queue()
-    procs = [ctx.Process(target=_rank_main, args=(r, 2, q, result)) for r in range(2)]
+    procs = [ctx.Process(target=_rank_main, args=(r, 2, case, port, result))
+             for r in range(2)]
     for p in procs:
         p.start()
-    merged = result.get(timeout=180)
+    merged_pb = result.get(timeout=180)
     for p in procs:
         p.join(timeout=60)
         assert p.exitcode == 0
+    return proto.decode("LeafSearchResponse", merged_pb)
 
-    # reference: one oracle over both splits in a single call
-    from quickwit_amd.api import OracleSearcher, make_leaf_request
+
+def _single_call_expected(case):
+    from quickwit_amd.api import OracleSearcher
     both = OracleSearcher()
     splits = []
     for r in range(2):
         sid = f"synthetic-3-{r:04d}"
         both.add_split(sid, splitgen.generate_split(r, NDOCS, seed=3))
         splits.append((sid, NDOCS))
-    req = make_leaf_request(q, splitgen.HDFS_SCHEMA, splits, max_hits=15,
-                            sort_fields=[{"field_name": "_score", "sort_order": 1}])
-    expected = both.leaf_search(req)
+    return both, both.leaf_search(_make_req(case, splits))
 
+
+def test_two_rank_packed_merge_bm25_equals_single_call():
+    merged = _run_case("bm25", 29511)
+    _, expected = _single_call_expected("bm25")
     assert merged["num_hits"] == expected["num_hits"]
     assert ([(h["split_id"], h["doc_id"]) for h in merged["partial_hits"]] ==
             [(h["split_id"], h["doc_id"]) for h in expected["partial_hits"]])
+    for mh, eh in zip(merged["partial_hits"], expected["partial_hits"]):
+        assert mh["sort_value"] == eh["sort_value"]
+
+
+def test_two_rank_packed_merge_field_sort_equals_single_call():
+    # two-key sort: both sort_value and sort_value2 survive the 32B records
+    merged = _run_case("field_sort", 29512)
+    _, expected = _single_call_expected("field_sort")
+    assert merged["num_hits"] == expected["num_hits"]
+    assert ([(h["split_id"], h["doc_id"]) for h in merged["partial_hits"]] ==
+            [(h["split_id"], h["doc_id"]) for h in expected["partial_hits"]])
+    for mh, eh in zip(merged["partial_hits"], expected["partial_hits"]):
+        assert mh.get("sort_value") == eh.get("sort_value")
+        assert mh.get("sort_value2") == eh.get("sort_value2")
+
+
+def test_two_rank_packed_merge_aggs_equal_single_call():
+    # date_histogram rides the dense SUM/MIN/MAX reduce (incl. a stats sub);
+    # the terms agg rides the sideband + C-ABI merge
+    merged = _run_case("agg", 29513)
+    both, expected = _single_call_expected("agg")
+    gj = both.finalize_agg_json(merged["intermediate_aggregation_result"], AGGS)
+    ej = both.finalize_agg_json(expected["intermediate_aggregation_result"],
+                                AGGS)
+    assert merged["num_hits"] == expected["num_hits"]
+    assert gj == ej
+
+
+def test_qagg_blob_roundtrip():
+    # codec sanity without any ranks: parse+serialize is the identity on a
+    # real product blob
+    import __graft_entry__
+    __graft_entry__.build()
+    from quickwit_amd import qagg
+    from quickwit_amd.api import OracleSearcher
+
+    s = OracleSearcher()
+    s.add_split("s", splitgen.generate_split(0, NDOCS, seed=3))
+    resp = s.leaf_search(_make_req("agg", [("s", NDOCS)]))
+    blob = resp["intermediate_aggregation_result"]
+    entries = qagg.parse_blob(blob)
+    assert [e.name for e in entries] == ["per_hour", "per_tenant", "stats_hour"]
+    assert entries[0].dense_eligible and entries[2].dense_eligible
+    assert not entries[1].dense_eligible  # terms
+    assert qagg.serialize_blob(entries) == blob
+    # bucket re-encode (not raw passthrough) is also the identity
+    for e in entries:
+        if not e.dense_eligible:
+            e.buckets = None
+    assert qagg.serialize_blob(entries) == blob

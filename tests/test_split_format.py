@@ -121,3 +121,81 @@ def test_synthetic_generator_deterministic():
     ts, pres = s.fast_column("timestamp")
     assert pres is None
     assert ts.min() >= splitgen.T0_EPOCH_S * 1000
+
+
+def test_native_generator_matches_independent_readers():
+    """The OpenMP generator (csrc/splitgen_native.cpp) must be readable
+    identically by the independent numpy reader and the C++ oracle: doc ids
+    sorted, df/count/bm25 agreement, severity/tenant distributions sane."""
+    import math
+
+    from quickwit_amd.api import OracleSearcher, make_leaf_request
+    from quickwit_amd.fieldnorm import id_to_norm
+
+    if splitgen._native_gen() is None:
+        pytest.skip("libqwsplitgen.so not built")
+    N = 150_000
+    data = splitgen.generate_split(3, N, seed=7)
+    sp = Split(data)
+    assert sp.meta["num_docs"] == N
+    assert sp.meta["split_id"] == "synthetic-7-0003"
+
+    terms = ["w%05d" % i for i in (0, 9, 10, 11, 9999)]
+    posts = {t: sp.postings("body", t) for t in terms}
+    for t, (d, tf) in posts.items():
+        assert (np.diff(d.astype(np.int64)) > 0).all()
+        assert tf.min() >= 1
+    # df close to N*q_t (Bernoulli margin 5 sigma)
+    p = splitgen.zipf_probs(splitgen.BODY_VOCAB_SIZE)
+    for rank in (0, 9, 10, 11):
+        q = 1 - (1 - p[rank]) ** splitgen.BODY_TOKENS_PER_DOC
+        df = len(posts["w%05d" % rank][0])
+        assert abs(df - N * q) < 5 * math.sqrt(N * q * (1 - q)) + 10, rank
+
+    # numpy-reader BM25 vs the oracle on the same bytes
+    norms = sp.fieldnorms("body")
+    assert (norms == 10).all()
+    dl = float(id_to_norm(np.array([10]))[0])
+    avgdl = sp.fields["body"]["total_tokens"] / N
+    score = np.zeros(N, np.float64)
+    k1, b = 1.2, 0.75
+    for t in ["w%05d" % i for i in (9, 10, 11)]:
+        d, tf = posts[t]
+        idf = math.log(1 + (N - len(d) + 0.5) / (len(d) + 0.5))
+        K = np.float32(k1 * (1 - b + b * dl / avgdl))
+        tff = tf.astype(np.float32)
+        np.add.at(score, d, (np.float32(idf * (k1 + 1)) * tff / (tff + K)).astype(np.float32))
+    cpu = OracleSearcher()
+    cpu.add_split("s", data)
+    q3 = {"type": "bool", "should": [
+        {"type": "term", "field": "body", "value": "w%05d" % i} for i in (9, 10, 11)]}
+    req = make_leaf_request(q3, splitgen.HDFS_SCHEMA, [("s", N)], max_hits=5,
+                            sort_fields=[{"field_name": "_score", "sort_order": 1}])
+    r = cpu.leaf_search(req)
+    assert r["num_hits"] == int((score > 0).sum())
+    best = np.argsort(-score, kind="stable")[:5]
+    for h, nd in zip(r["partial_hits"], best):
+        assert math.isclose(h["sort_value"]["f64"], score[nd],
+                            rel_tol=1e-5, abs_tol=1e-9)
+
+    # severity postings partition the doc space
+    sev_dfs = sp._sec("severity_text", "doc_freq", "<u4")
+    assert int(sev_dfs.sum()) == N
+    info, _ = sp.postings("severity_text", "INFO")
+    r2 = cpu.leaf_search(make_leaf_request(
+        {"type": "term", "field": "severity_text", "value": "INFO"},
+        splitgen.HDFS_SCHEMA, [("s", N)], max_hits=0))
+    assert r2["num_hits"] == len(info)
+
+    # fast columns: range count via reader == oracle
+    ten, present = sp.fast_column("tenant_id")
+    assert present is None
+    r3 = cpu.leaf_search(make_leaf_request(
+        {"type": "range", "field": "tenant_id",
+         "lower_bound": {"included": 100}, "upper_bound": {"excluded": 300}},
+        splitgen.HDFS_SCHEMA, [("s", N)], max_hits=0))
+    assert r3["num_hits"] == int(((ten >= 100) & (ten < 300)).sum())
+
+    # determinism
+    assert splitgen.generate_split(3, 50_000, seed=7) == \
+        splitgen.generate_split(3, 50_000, seed=7)
