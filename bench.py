@@ -118,6 +118,39 @@ def make_workload(name, docs, max_hits):
             # profiles/ corroborates total traffic ~= these bytes + flush.
             "algo_bytes": lambda sp, nh, nt: nt * TILE_DOCS * (8 + 2),
         }
+    if name == "config5":
+        # BASELINE.json configs[4]: top-1000 BM25 + fast-field filter +
+        # terms agg (the 8-GPU slice is 1B docs / 64 splits; the 1-GPU slice
+        # is --docs 125000000 --splits 8 = 8 x 15.6M-doc splits)
+        q = {"type": "bool",
+             "should": [{"type": "term", "field": "body", "value": t}
+                        for t in BM25_TERMS],
+             "filter": [{"type": "range", "field": "tenant_id",
+                         "lower_bound": {"included": 100},
+                         "upper_bound": {"excluded": 900}}]}
+
+        def _algo(sp, nh, nt):
+            # postings of the 3 should terms + 8B tenant gather per
+            # union-matched doc (union size analytic from the per-term
+            # df under the generator's independence) + 1B norm + 16B wide
+            # candidate record per post-filter hit + tile counts
+            nd = nt * TILE_DOCS
+            prod = 1.0
+            for t in BM25_TERMS:
+                prod *= 1.0 - _doc_freq(sp, "body", t) / nd
+            union = int(nd * (1.0 - prod))
+            return (sum(_posting_bytes(sp, "body", t, nt) for t in BM25_TERMS)
+                    + union * (8 + 1) + nh * 16 + nt * 4)
+        return {
+            "query": q,
+            "sort": [{"field_name": "_score", "sort_order": 1}],
+            "aggregation": {"per_tenant": {"terms": {"field": "tenant_name",
+                                                     "size": 10}}},
+            "max_hits": 1000, "kernel": "union_bm25",
+            "label": f"top1000_bm25_filter_terms_{docs}",
+            "query_str": "3-term OR, tenant_id:[100..900), top-1000, terms agg",
+            "algo_bytes": _algo,
+        }
     raise SystemExit(f"unknown workload {name}")
 
 
@@ -149,7 +182,11 @@ def main():
                          "headline config of BASELINE.json's metric")
     ap.add_argument("--max-hits", type=int, default=10)
     ap.add_argument("--workload", default="bm25",
-                    choices=["bm25", "range", "agg"])
+                    choices=["bm25", "range", "agg", "config5"])
+    ap.add_argument("--splits", type=int, default=1,
+                    help="splits per GPU (docs are divided across them); "
+                         "config5's 1-GPU slice is --docs 125000000 "
+                         "--splits 8")
     ap.add_argument("--cpu-baseline-steps", type=int, default=2)
     ap.add_argument("--cpu-baseline-docs", type=int, default=0,
                     help="oracle sample size (0 = min(docs, 10M))")
@@ -176,15 +213,23 @@ def main():
     wl = make_workload(args.workload, args.docs, args.max_hits)
 
     t_gen = time.perf_counter()
-    split_bytes = cached_split(rank, args.docs)
-    sid = f"synthetic-42-{rank:04d}"
+    docs_per_split = args.docs // args.splits
+    searcher = GpuSearcher(device=local_rank)
+    split_set = []
+    split_bytes = None
+    for si in range(args.splits):
+        ordn = rank * 64 + si
+        data = cached_split(ordn, docs_per_split)
+        if split_bytes is None:
+            split_bytes = data  # algo-bytes accounting reads split 0
+        sid = f"synthetic-42-{ordn:04d}"
+        searcher.add_split(sid, data)
+        split_set.append((sid, docs_per_split))
+    sid = split_set[0][0]
     gen_s = time.perf_counter() - t_gen
 
-    searcher = GpuSearcher(device=local_rank)
-    searcher.add_split(sid, split_bytes)
-
     req = make_leaf_request(
-        wl["query"], splitgen.HDFS_SCHEMA, [(sid, args.docs)],
+        wl["query"], splitgen.HDFS_SCHEMA, split_set,
         max_hits=wl["max_hits"], sort_fields=wl["sort"],
         aggregation=wl["aggregation"])
     req_pb = proto.encode("LeafSearchRequest", req)
@@ -195,7 +240,8 @@ def main():
         if world > 1:
             # packed-tensor exchange over RCCL/xGMI: 32B hit
             # records allgather + dense-bucket sum-reduce + sideband bytes
-            merged = distributed_merge(sreq_pb, resp_pb, [sid])
+            merged = distributed_merge(sreq_pb, resp_pb,
+                                       [s for s, _ in split_set])
             return merged if rank == 0 else resp_pb
         return resp_pb
 
@@ -238,9 +284,9 @@ def main():
     roofline = None
     if launches:
         ms_per_launch = kms / launches
-        n_tiles = (args.docs + TILE_DOCS - 1) // TILE_DOCS
+        n_tiles = (docs_per_split + TILE_DOCS - 1) // TILE_DOCS
         sp = splitread.Split(split_bytes)
-        ab = int(wl["algo_bytes"](sp, per_split_hits, n_tiles))
+        ab = int(wl["algo_bytes"](sp, per_split_hits // args.splits, n_tiles))
         achieved = ab / (ms_per_launch / 1e3) / 1e9
         roofline = {"bound": "hbm", "achieved": round(achieved, 1),
                     "peak": HBM_PEAK_GBS, "unit": "GB/s",
@@ -309,12 +355,14 @@ def main():
         "config": {
             "workload": wl["label"],
             "query": wl["query_str"],
-            "docs_per_gpu": args.docs,
-            "splits_per_gpu": 1,
+            "docs_per_gpu": docs_per_split * args.splits,
+            "splits_per_gpu": args.splits,
             "max_hits": wl["max_hits"],
             "num_hits_per_split": per_split_hits,
             "parallelism": f"split-dp{world}",
             "gen_seconds": round(gen_s, 1),
+            "hbm_used_gb": round(searcher.memory_stats()[0] / 2**30, 2),
+            "hbm_budget_gb": round(searcher.memory_stats()[1] / 2**30, 1),
         },
         "roofline": roofline,
         "cpu_baseline": cpu_baseline,

@@ -46,6 +46,10 @@ enum qw_status {
     QW_ERR_NO_GPU = -5,           /* no HIP device: the product path never
                                      falls back to CPU */
     QW_ERR_BAD_SPLIT = -6,        /* malformed QWA1 container */
+    QW_ERR_OVER_MEMORY_BUDGET = -7, /* HBM budget refusal: the permit-provider
+                                     * memory gate (search_permit_provider.rs:
+                                     * 43-110) mapped to a synchronous C-ABI —
+                                     * refuse instead of queueing */
 };
 
 /* config_json mirrors the result-affecting subset of SearcherConfig
@@ -130,6 +134,20 @@ int32_t qw_kernel_stats(qw_ctx* ctx, const char* kernel_name,
                         double* total_ms, uint64_t* launches);
 void qw_kernel_stats_reset(qw_ctx* ctx);
 int32_t qw_ctx_device_sync(qw_ctx* ctx);
+
+/* HBM footprint accounting (SearchPermitProvider memory-budget analog):
+ * used = split images + scratch/result buffers + cached hitset bitmaps;
+ * budget = configured "hbm_memory_budget" or 97% of device free memory at
+ * first use; splits_bytes = resident split images alone. */
+int32_t qw_ctx_memory_stats(qw_ctx* ctx, uint64_t* used_bytes,
+                            uint64_t* budget_bytes, uint64_t* splits_bytes);
+
+/* Negative/absence term cache counters (leaf.rs:761-827): probes that
+ * short-circuited a split (hits), probes over required terms that found no
+ * cached absence (misses), and resident keys. Replaces the reference's
+ * prometheus-side SplitSearchOutcomeCounters for this cache. */
+int32_t qw_absence_cache_stats(qw_ctx* ctx, uint64_t* hits, uint64_t* misses,
+                               uint64_t* entries);
 
 /* Library version + build arch, e.g. "quickwit_amd 0.1 gfx950". */
 const char* qw_version(void);

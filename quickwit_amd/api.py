@@ -169,6 +169,28 @@ class GpuSearcher(_BaseSearcher):
         self._lib.qw_kernel_stats_reset.argtypes = [ctypes.c_void_p]
         self._lib.qw_kernel_stats_reset(self._ctx)
 
+    def memory_stats(self):
+        """(used_bytes, budget_bytes, splits_bytes) of the ctx HBM
+        accounting (SearchPermitProvider memory-budget analog)."""
+        fn = self._lib.qw_ctx_memory_stats
+        fn.argtypes = [ctypes.c_void_p] + [ctypes.POINTER(ctypes.c_uint64)] * 3
+        u, b, s = ctypes.c_uint64(), ctypes.c_uint64(), ctypes.c_uint64()
+        rc = fn(self._ctx, ctypes.byref(u), ctypes.byref(b), ctypes.byref(s))
+        if rc != 0:
+            raise RuntimeError(f"memory_stats failed: {rc}")
+        return u.value, b.value, s.value
+
+    def absence_cache_stats(self):
+        """(hits, misses, entries) of the negative term cache
+        (leaf.rs:761-827 analog)."""
+        fn = self._lib.qw_absence_cache_stats
+        fn.argtypes = [ctypes.c_void_p] + [ctypes.POINTER(ctypes.c_uint64)] * 3
+        h, m, e = ctypes.c_uint64(), ctypes.c_uint64(), ctypes.c_uint64()
+        rc = fn(self._ctx, ctypes.byref(h), ctypes.byref(m), ctypes.byref(e))
+        if rc != 0:
+            raise RuntimeError(f"absence_cache_stats failed: {rc}")
+        return h.value, m.value, e.value
+
     def device_sync(self):
         self._lib.qw_ctx_device_sync.argtypes = [ctypes.c_void_p]
         rc = self._lib.qw_ctx_device_sync(self._ctx)

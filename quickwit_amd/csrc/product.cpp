@@ -29,6 +29,7 @@
 #include <list>
 #include <map>
 #include <memory>
+#include <set>
 #include <stdexcept>
 #include <string>
 #include <vector>
@@ -202,9 +203,11 @@ struct HitsetCache {
         std::string key;
         uint8_t* bm;
         uint64_t gen;
+        uint64_t size;
     };
     size_t cap = 256;
     uint64_t cur_gen = 0;
+    uint64_t bytes = 0;  // device bytes held (feeds the ctx HBM accounting)
     std::list<Entry> lru;
     std::map<std::string, std::list<Entry>::iterator> idx;
     uint64_t hits = 0, misses = 0;
@@ -221,9 +224,10 @@ struct HitsetCache {
         lru.splice(lru.begin(), lru, it->second);
         return it->second->bm;
     }
-    void put(const std::string& key, uint8_t* bm) {
-        lru.push_front(Entry{key, bm, cur_gen});
+    void put(const std::string& key, uint8_t* bm, uint64_t size) {
+        lru.push_front(Entry{key, bm, cur_gen, size});
         idx[key] = lru.begin();
+        bytes += size;
         // evict oldest UNPINNED entries; a query resolving > cap distinct
         // subtrees temporarily overflows the cap rather than freeing memory
         // the pending launch still references
@@ -232,6 +236,7 @@ struct HitsetCache {
             auto cur = it--;
             if (cur->gen == cur_gen) continue;
             (void)hipFree(cur->bm);
+            bytes -= cur->size;
             idx.erase(cur->key);
             lru.erase(cur);
         }
@@ -241,6 +246,7 @@ struct HitsetCache {
         for (auto it = lru.begin(); it != lru.end();) {
             if (it->key.compare(0, prefix.size(), prefix) == 0) {
                 (void)hipFree(it->bm);
+                bytes -= it->size;
                 idx.erase(it->key);
                 it = lru.erase(it);
             } else ++it;
@@ -250,10 +256,36 @@ struct HitsetCache {
         for (auto& e : lru) (void)hipFree(e.bm);
         lru.clear();
         idx.clear();
+        bytes = 0;
     }
 };
 
 }  // namespace qw
+
+// negative/absence term cache (leaf.rs:761-827): (split, field, term) keys
+// proven absent from the split's term dictionary by an earlier search; a
+// probe hit short-circuits the whole split search (the reference aborts
+// during warmup, leaf.rs:315-320 / 472-477; key analog of
+// term_absence_cache_key, leaf.rs:641-652)
+struct AbsenceCache {
+    std::set<std::string> keys;
+    uint64_t hits = 0, misses = 0;
+    size_t cap = 1 << 20;
+    static std::string key(const std::string& split_id, const std::string& f,
+                           const std::string& t) {
+        std::string k = split_id;
+        k += '\0';
+        k += f;
+        k += '\x1f';
+        k += t;
+        return k;
+    }
+    void remove_split(const std::string& split_id) {
+        std::string lo = split_id + '\0';
+        std::string hi = split_id + '\x01';
+        keys.erase(keys.lower_bound(lo), keys.lower_bound(hi));
+    }
+};
 
 struct qw_ctx {
     int device = 0;
@@ -262,12 +294,43 @@ struct qw_ctx {
     std::map<std::string, std::unique_ptr<qw::DeviceSplit>> splits;
     qw::LeafCache leaf_cache;
     qw::HitsetCache hitsets;
+    AbsenceCache absence;
     std::string last_error;
     hipStream_t stream = nullptr;
     hipEvent_t ev_start = nullptr, ev_stop = nullptr;
     qw::DevBuf d_scratch, d_results, d_survivors, d_cand2;
     std::map<std::string, qw::KernelTimer> timers;
+    // HBM accounting (SearchPermitProvider memory-budget analog,
+    // search_permit_provider.rs:43-110): split images + scratch/result
+    // buffers in hbm_used, hitset bitmaps in hitsets.bytes. budget 0 =
+    // resolved from device free memory at first use; config key
+    // "hbm_memory_budget" overrides (bytes). Over-budget add_split refuses
+    // (QW_ERR_OVER_MEMORY_BUDGET); over-budget search scratch fails that
+    // split into failed_splits, like a permit that can never be granted.
+    uint64_t hbm_used = 0;
+    uint64_t hbm_budget = 0;
+    bool budget_configured = false;
 };
+
+static void qw_check_budget(qw_ctx* ctx, uint64_t need) {
+    uint64_t used = ctx->hbm_used + ctx->hitsets.bytes;
+    if (ctx->hbm_budget && used + need > ctx->hbm_budget)
+        throw std::runtime_error(
+            "HBM memory budget exceeded: used " + std::to_string(used) +
+            " + need " + std::to_string(need) + " > budget " +
+            std::to_string(ctx->hbm_budget) +
+            " (search_permit_provider analog; raise hbm_memory_budget or "
+            "remove splits)");
+}
+
+// budgeted grow of a ctx scratch DevBuf
+static void qw_ensure_acct(qw_ctx* ctx, qw::DevBuf& b, size_t n) {
+    if (n <= b.cap) return;
+    qw_check_budget(ctx, n - b.cap);
+    ctx->hbm_used -= b.cap;
+    b.ensure(n);
+    ctx->hbm_used += b.cap;
+}
 
 static void ctx_ensure_device(qw_ctx* ctx) {
     using namespace qw;
@@ -280,6 +343,13 @@ static void ctx_ensure_device(qw_ctx* ctx) {
     HIP_CHECK(hipStreamCreateWithFlags(&ctx->stream, hipStreamNonBlocking));
     HIP_CHECK(hipEventCreate(&ctx->ev_start));
     HIP_CHECK(hipEventCreate(&ctx->ev_stop));
+    if (!ctx->budget_configured) {
+        // default budget: 97% of the device's free HBM at first use
+        // (warmup_memory_budget analog, node_config defaults)
+        size_t free_b = 0, total_b = 0;
+        if (hipMemGetInfo(&free_b, &total_b) == hipSuccess && free_b)
+            ctx->hbm_budget = uint64_t(free_b) / 100 * 97;
+    }
     ctx->device_ready = true;
 }
 
@@ -296,6 +366,10 @@ struct FlatQuery {
     bool match_all = false;  // positive base = all docs (preds still filter)
     bool scoring = false;
     uint32_t msm = 0;  // >0: shoulds required (count >= msm)
+    // required (must/filter) single terms proven ABSENT from this split's
+    // term dict during flattening — feeds the negative term cache
+    // (leaf.rs:761-827 / term_absence_cache_key leaf.rs:641-652)
+    std::vector<std::pair<std::string, std::string>> absent_required;
     struct FTerm {
         const TextFieldView* f;
         int64_t tid;
@@ -416,7 +490,10 @@ static void add_term(FlatQuery& fq, const SplitView& sv, const std::string& fiel
         // absent term: MUST -> split matches nothing; SHOULD/MUST_NOT -> skip
         // (skipped shoulds can never satisfy msm — handled by caller count).
         // A member of a must OR-GROUP only weakens the group when absent.
-        if (role == ROLE_MUST && !group_member) fq.match_none = true;
+        if (role == ROLE_MUST && !group_member) {
+            fq.match_none = true;
+            fq.absent_required.emplace_back(field, value);
+        }
         return;
     }
     FlatQuery::FTerm t;
@@ -1157,6 +1234,7 @@ static uint8_t* bitmap_eval(qw_ctx* ctx, const DeviceSplit& ds,
     key += '\0';
     plan_fingerprint(node, key);
     uint8_t* bm = nullptr;
+    qw_check_budget(ctx, bm_bytes);
     HIP_CHECK(hipMalloc(&bm, bm_bytes));
     bool has_req = !node.must.empty() || !node.filter.empty();
     bool first = true;
@@ -1170,6 +1248,7 @@ static uint8_t* bitmap_eval(qw_ctx* ctx, const DeviceSplit& ds,
     if (msm <= 0) msm = has_req ? 0 : 1;
     if (!node.should.empty() && msm > 0) {
         uint8_t* su = nullptr;
+        qw_check_budget(ctx, bm_bytes);
         HIP_CHECK(hipMalloc(&su, bm_bytes));
         bool sfirst = true;
         for (const PlanNode& c : node.should) {
@@ -1215,7 +1294,7 @@ static uint8_t* bitmap_eval(qw_ctx* ctx, const DeviceSplit& ds,
                                 tail.size() * 4, hipMemcpyHostToDevice));
         }
     }
-    ctx->hitsets.put(key, bm);
+    ctx->hitsets.put(key, bm, bm_bytes);
     return bm;
 }
 
@@ -1230,6 +1309,7 @@ static uint8_t* resolve_hitset(qw_ctx* ctx, const DeviceSplit& ds,
     uint32_t n_tiles = (sv.num_docs + TILE_DOCS - 1) / TILE_DOCS;
     size_t bm_bytes = size_t(n_tiles) * (TILE_DOCS / 32) * 4;
     uint8_t* bm = nullptr;
+    qw_check_budget(ctx, bm_bytes);
     HIP_CHECK(hipMalloc(&bm, bm_bytes));
 
     FlatQuery fq;
@@ -1249,7 +1329,7 @@ static uint8_t* resolve_hitset(qw_ctx* ctx, const DeviceSplit& ds,
     if (fq.match_none) {
         HIP_CHECK(hipMemsetAsync(bm, 0, bm_bytes, ctx->stream));
         HIP_CHECK(hipStreamSynchronize(ctx->stream));
-        ctx->hitsets.put(key, bm);
+        ctx->hitsets.put(key, bm, bm_bytes);
         return bm;
     }
 
@@ -1299,8 +1379,8 @@ static uint8_t* resolve_hitset(qw_ctx* ctx, const DeviceSplit& ds,
     memcpy(scratch.data() + off_preds, fq.preds.data(),
            fq.preds.size() * sizeof(PredDev));
     memcpy(scratch.data() + off_ranges, ranges.data(), ranges_bytes);
-    ctx->d_scratch.ensure(scratch_bytes);
-    ctx->d_results.ensure(64);
+    qw_ensure_acct(ctx, ctx->d_scratch, scratch_bytes);
+    qw_ensure_acct(ctx, ctx->d_results, 64);
     HIP_CHECK(hipMemcpyAsync(ctx->d_scratch.p, scratch.data(), scratch_bytes,
                              hipMemcpyHostToDevice, ctx->stream));
     HIP_CHECK(hipMemsetAsync(ctx->d_results.p, 0, 64, ctx->stream));
@@ -1334,7 +1414,7 @@ static uint8_t* resolve_hitset(qw_ctx* ctx, const DeviceSplit& ds,
                      n_tiles, 0u);
     HIP_CHECK(hipGetLastError());
     HIP_CHECK(hipStreamSynchronize(ctx->stream));
-    ctx->hitsets.put(key, bm);
+    ctx->hitsets.put(key, bm, bm_bytes);
     return bm;
 }
 
@@ -1348,6 +1428,29 @@ static void record_kernel_time(qw_ctx* ctx, const char* name, float ms) {
     KernelTimer& t = ctx->timers[name];
     t.total_ms += ms;
     t.launches += 1;
+}
+
+// single terms in REQUIRED position (must/filter chains, unwrapping cache
+// nodes) — the candidates the absence cache can short-circuit on; restates
+// the reference's required-term extraction (quickwit-query required_terms
+// feeding leaf.rs:769-778)
+static void collect_required_terms(
+    const PlanNode& n, std::vector<std::pair<std::string, std::string>>& out) {
+    switch (n.kind) {
+        case PlanNode::TERM:
+            out.emplace_back(n.field, n.value);
+            break;
+        case PlanNode::BOOL:
+            for (const PlanNode& c : n.must) collect_required_terms(c, out);
+            for (const PlanNode& c : n.filter) collect_required_terms(c, out);
+            break;
+        case PlanNode::CACHE:
+            if (!n.cache_inner.empty())
+                collect_required_terms(n.cache_inner.front(), out);
+            break;
+        default:
+            break;
+    }
 }
 
 static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
@@ -1387,6 +1490,28 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         b.filter.push_back(std::move(ts));
         b.must.push_back(std::move(plan));
         plan = std::move(b);
+    }
+
+    // ---- negative/absence term cache probe (leaf.rs:761-827): a required
+    // term proven absent by an earlier search short-circuits this split —
+    // replace the plan with MATCH_NONE so the standard empty-response path
+    // (shaped agg blob, stats) runs with no kernel and no dict lookups
+    {
+        std::vector<std::pair<std::string, std::string>> reqs;
+        collect_required_terms(plan, reqs);
+        bool hit = false;
+        for (auto& ft : reqs)
+            if (ctx->absence.keys.count(
+                    AbsenceCache::key(sv.split_id, ft.first, ft.second))) {
+                hit = true;
+                break;
+            }
+        if (!reqs.empty()) (hit ? ctx->absence.hits : ctx->absence.misses)++;
+        if (hit) {
+            PlanNode none;
+            none.kind = PlanNode::MATCH_NONE;
+            plan = std::move(none);
+        }
     }
 
     if (req.sort_fields.size() > 2)
@@ -1431,6 +1556,12 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         p.abs_bitmap = (uint64_t)bmp;
         fq.preds.push_back(p);
     }
+    // populate the absence cache with required terms flattening proved
+    // absent (the next search for them skips this split up front)
+    for (auto& ab : fq.absent_required)
+        if (ctx->absence.keys.size() < ctx->absence.cap)
+            ctx->absence.keys.insert(
+                AbsenceCache::key(sv.split_id, ab.first, ab.second));
     // CACHE nodes in filter position -> device HitSet bitmaps (PRED_BITSET)
     for (const PlanNode* cn : fq.cache_nodes) {
         PredDev p{};
@@ -1623,8 +1754,8 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
     std::vector<uint32_t> hist0(TOPK_BINS);  // prefetched selection pass 0
 
     if (need_kernel) {
-        ctx->d_scratch.ensure(scratch_bytes);
-        ctx->d_results.ensure(results_bytes);
+        qw_ensure_acct(ctx, ctx->d_scratch, scratch_bytes);
+        qw_ensure_acct(ctx, ctx->d_results, results_bytes);
         HIP_CHECK(hipMemcpyAsync(ctx->d_scratch.p, scratch.data(), scratch_bytes,
                                  hipMemcpyHostToDevice, ctx->stream));
         HIP_CHECK(hipMemsetAsync(ctx->d_results.p, 0, r_agg, ctx->stream));
@@ -1796,7 +1927,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     if (nasc) kh = ~kh;
                     ceil_key = (uint64_t(kh) << 32) | 0xFFFFFFFFull;
                 }
-                ctx->d_cand2.ensure(size_t(cand_n) * 8 * rw + 16);
+                qw_ensure_acct(ctx, ctx->d_cand2, size_t(cand_n) * 8 * rw + 16);
                 uint32_t* d_c2n = (uint32_t*)(ctx->d_results.p + r_cand_count) + 2;
                 HIP_CHECK(hipMemsetAsync(d_c2n, 0, 4, ctx->stream));
                 uint32_t pgrid = std::min<uint32_t>(2048, (cand_n + 255) / 256);
@@ -1880,7 +2011,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             }
             // #keys >= floor_key is known EXACTLY from the histogram walk, so
             // no round trip for the compact count: one async D2H of survivors
-            ctx->d_survivors.ensure(survivors * 8 * rw + 16);
+            qw_ensure_acct(ctx, ctx->d_survivors, survivors * 8 * rw + 16);
             HIP_CHECK(hipMemsetAsync(d_scount, 0, 4, ctx->stream));
             uint32_t cgrid = std::min<uint32_t>(2048, (cand_n + 255) / 256);
             if (wide)
@@ -2357,6 +2488,10 @@ qw_ctx* qw_ctx_create(const char* config_json) {
                 ctx->agg_bucket_limit = b->as_i64();
             if (const mj::Value* c = cfg->get("partial_request_cache_capacity"))
                 ctx->leaf_cache.capacity = size_t(c->as_i64());
+            if (const mj::Value* mb = cfg->get("hbm_memory_budget")) {
+                ctx->hbm_budget = uint64_t(mb->as_i64());
+                ctx->budget_configured = true;
+            }
         }
         return ctx;
     } catch (const std::exception& e) {
@@ -2403,8 +2538,15 @@ int32_t qw_ctx_add_split(qw_ctx* ctx, const char* split_id, const uint8_t* data,
                 "the split with the current writer)");
         ds->len = len;
         HIP_CHECK(hipSetDevice(ctx->device));
+        try {
+            qw_check_budget(ctx, len + 64);
+        } catch (const std::exception& e) {
+            set_err(ctx, e.what());
+            return QW_ERR_OVER_MEMORY_BUDGET;
+        }
         HIP_CHECK(hipMalloc(&ds->d_image, len + 64));  // +64: decode overread pad
         HIP_CHECK(hipMemcpy(ds->d_image, data, len, hipMemcpyHostToDevice));
+        ctx->hbm_used += len + 64;
         ctx->splits[split_id] = std::move(ds);
         return QW_OK;
     } catch (const std::exception& e) {
@@ -2416,10 +2558,14 @@ int32_t qw_ctx_add_split(qw_ctx* ctx, const char* split_id, const uint8_t* data,
 int32_t qw_ctx_remove_split(qw_ctx* ctx, const char* split_id) {
     auto it = ctx->splits.find(split_id);
     if (it == ctx->splits.end()) return QW_ERR_NOT_FOUND;
-    if (it->second->d_image) (void)hipFree(it->second->d_image);
+    if (it->second->d_image) {
+        (void)hipFree(it->second->d_image);
+        ctx->hbm_used -= it->second->len + 64;
+    }
     ctx->splits.erase(it);
     ctx->leaf_cache.remove_split(split_id);
     ctx->hitsets.remove_split(split_id);
+    ctx->absence.remove_split(split_id);
     return QW_OK;
 }
 
@@ -2901,6 +3047,28 @@ int32_t qw_kernel_stats(qw_ctx* ctx, const char* kernel_name, double* total_ms,
 }
 
 void qw_kernel_stats_reset(qw_ctx* ctx) { ctx->timers.clear(); }
+
+int32_t qw_ctx_memory_stats(qw_ctx* ctx, uint64_t* used_bytes,
+                            uint64_t* budget_bytes, uint64_t* splits_bytes) {
+    if (!ctx) return QW_ERR_INVALID_ARGUMENT;
+    if (used_bytes) *used_bytes = ctx->hbm_used + ctx->hitsets.bytes;
+    if (budget_bytes) *budget_bytes = ctx->hbm_budget;
+    if (splits_bytes) {
+        uint64_t sb = 0;
+        for (auto& kv : ctx->splits) sb += kv.second->len + 64;
+        *splits_bytes = sb;
+    }
+    return QW_OK;
+}
+
+int32_t qw_absence_cache_stats(qw_ctx* ctx, uint64_t* hits, uint64_t* misses,
+                               uint64_t* entries) {
+    if (!ctx) return QW_ERR_INVALID_ARGUMENT;
+    if (hits) *hits = ctx->absence.hits;
+    if (misses) *misses = ctx->absence.misses;
+    if (entries) *entries = uint64_t(ctx->absence.keys.size());
+    return QW_OK;
+}
 
 int32_t qw_ctx_device_sync(qw_ctx* ctx) {
     try {

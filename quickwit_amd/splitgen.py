@@ -634,7 +634,11 @@ def generate_split(split_ord: int, num_docs: int, seed: int = 42) -> bytes:
         if rc != 0:
             raise RuntimeError(f"qw_gen_split failed: {rc}")
         try:
-            return ctypes.string_at(p, n.value)
+            # NOT ctypes.string_at: its size argument truncates to 32 bits
+            # (a 100M-doc split is ~3.6 GB)
+            view = (ctypes.c_char * n.value).from_address(
+                ctypes.addressof(p.contents))
+            return bytes(view)
         finally:
             lib.qw_gen_free(p)
     return generate_split_numpy(split_ord, num_docs, seed)

@@ -1402,3 +1402,97 @@ def test_agg_fast_bare_histogram_only_under_filter(searchers):
     gj = gpu.finalize_agg_json(g["intermediate_aggregation_result"], aggs)
     ej = cpu.finalize_agg_json(e["intermediate_aggregation_result"], aggs)
     assert gj == ej
+
+
+# ------------------------------------------------------- absence cache
+def test_absence_cache_hit_miss_and_parity():
+    """Negative term cache (leaf.rs:761-827): a required term proven absent
+    populates the cache; the next search for it short-circuits the split
+    (hit) with a response identical to the oracle's. Should-position absent
+    terms must NOT populate it."""
+    data = splitgen.generate_split(0, 30_000, seed=11)
+    gpu = GpuSearcher(device=0)
+    cpu = OracleSearcher()
+    gpu.add_split("abs-split", data)
+    cpu.add_split("abs-split", data)
+    assert gpu.absence_cache_stats() == (0, 0, 0)
+
+    q_absent = {"type": "bool", "must": [
+        {"type": "term", "field": "body", "value": "zz_not_a_term"},
+        {"type": "term", "field": "severity_text", "value": "INFO"}]}
+    req = make_leaf_request(q_absent, SCHEMA, [("abs-split", 30_000)],
+                            max_hits=5, aggregation={
+                                "h": {"date_histogram": {
+                                    "field": "timestamp",
+                                    "fixed_interval": "86400000ms"}}})
+    g1, e1 = gpu.leaf_search(req), cpu.leaf_search(req)
+    assert g1.get("num_hits", 0) == 0 == e1.get("num_hits", 0)
+    h, m, e = gpu.absence_cache_stats()
+    assert (h, m) == (0, 1) and e == 1  # miss recorded, absence learned
+
+    # second identical search: cache HIT short-circuits; response identical
+    # up to timing (resource_stats carries wall-clock microseconds)
+    def strip_stats(r):
+        return {k: v for k, v in r.items() if k != "resource_stats"}
+    g2 = gpu.leaf_search(req)
+    h2, m2, _ = gpu.absence_cache_stats()
+    assert (h2, m2) == (1, 1)
+    assert strip_stats(g2) == strip_stats(g1) == strip_stats(e1)
+
+    # a different query sharing the absent required term also hits
+    q2 = {"type": "bool", "filter": [
+        {"type": "term", "field": "body", "value": "zz_not_a_term"}]}
+    req2 = make_leaf_request(q2, SCHEMA, [("abs-split", 30_000)], max_hits=3)
+    g3 = gpu.leaf_search(req2)
+    assert gpu.absence_cache_stats()[0] == 2
+    assert g3.get("num_hits", 0) == 0 == \
+        cpu.leaf_search(req2).get("num_hits", 0)
+
+    # should-position absence is NOT required -> no new entries, no hit;
+    # present terms still match
+    q3 = {"type": "bool", "should": [
+        {"type": "term", "field": "body", "value": "zz_other_missing"},
+        {"type": "term", "field": "body", "value": "w00000"}]}
+    req3 = make_leaf_request(q3, SCHEMA, [("abs-split", 30_000)], max_hits=5,
+                             sort_fields=[{"field_name": "_score",
+                                           "sort_order": 1}])
+    g4, e4 = gpu.leaf_search(req3), cpu.leaf_search(req3)
+    assert g4["num_hits"] == e4["num_hits"] > 0
+    assert gpu.absence_cache_stats()[2] == 1  # unchanged
+
+    # removing the split drops its keys
+    gpu._lib.qw_ctx_remove_split.argtypes = [
+        __import__("ctypes").c_void_p, __import__("ctypes").c_char_p]
+    assert gpu._lib.qw_ctx_remove_split(gpu._ctx, b"abs-split") == 0
+    assert gpu.absence_cache_stats()[2] == 0
+
+
+# ------------------------------------------------- memory budget (permits)
+def test_memory_budget_accounting_and_refusal():
+    """HBM accounting (search_permit_provider.rs:43-110 analog): add_split
+    and search scratch are accounted against the ctx budget; an over-budget
+    add_split is refused with QW_ERR_OVER_MEMORY_BUDGET."""
+    data = splitgen.generate_split(0, 20_000, seed=5)
+    g = GpuSearcher(device=0)
+    g.add_split("m1", data)
+    u1, b1, s1 = g.memory_stats()
+    assert s1 == len(data) + 64
+    assert u1 >= s1
+    assert b1 > s1  # default budget resolved from device free memory
+    req = make_leaf_request({"type": "match_all"}, SCHEMA, [("m1", 20_000)],
+                            max_hits=5)
+    g.leaf_search(req)
+    u2, _, _ = g.memory_stats()
+    assert u2 >= u1  # search scratch accounted
+    # removing the split releases its accounted bytes
+    import ctypes
+    g._lib.qw_ctx_remove_split.argtypes = [ctypes.c_void_p, ctypes.c_char_p]
+    assert g._lib.qw_ctx_remove_split(g._ctx, b"m1") == 0
+    u3, _, s3 = g.memory_stats()
+    assert s3 == 0 and u3 == u2 - s1
+
+    # a ctx with a tiny configured budget refuses the split up front
+    g2 = GpuSearcher(device=0, config={"hbm_memory_budget": 4096})
+    with pytest.raises(RuntimeError, match="budget"):
+        g2.add_split("m1", data)
+    assert g2.memory_stats()[1] == 4096
