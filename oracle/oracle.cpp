@@ -1029,12 +1029,16 @@ void qw_oracle_buf_free(qw_oracle_buf* b) {
     b->len = 0;
 }
 
+// (the caller's split id is authoritative over the container's own id,
+// matching the product's qw_ctx_add_split — responses echo the id the
+// request named the split by)
 int qw_oracle_add_split(qw_oracle_ctx* ctx, const char* split_id, const uint8_t* data,
                         size_t len) {
     try {
         auto h = std::make_unique<qw_oracle_ctx::SplitHolder>();
         h->data.assign(data, data + len);
         h->view.parse(h->data.data(), h->data.size());
+        h->view.split_id = split_id;
         ctx->splits[split_id] = std::move(h);
         return 0;
     } catch (const std::exception& e) {

@@ -2532,6 +2532,10 @@ int32_t qw_ctx_add_split(qw_ctx* ctx, const char* split_id, const uint8_t* data,
         auto ds = std::make_unique<DeviceSplit>();
         ds->host.assign(data, data + len);
         ds->view.parse(ds->host.data(), len);
+        // the caller's split id is authoritative (the request names splits
+        // by SplitIdAndFooterOffsets.split_id and the response must echo
+        // it); the container's own id is only a default
+        ds->view.split_id = split_id;
         if (ds->view.version < 2)
             throw std::runtime_error(
                 "QWA1 v1 container lacks posting segment anchors (regenerate "
