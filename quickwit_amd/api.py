@@ -37,7 +37,13 @@ class _BaseSearcher:
     _libname = None
 
     def __init__(self, create_arg=None):
-        path = os.path.join(REPO_ROOT, self._libname)
+        # QW_PRODUCT_LIB: perf-experiment override for the product library
+        # path (e.g. a -DQW_TILE_DOCS variant built alongside the default)
+        override = os.environ.get("QW_PRODUCT_LIB")
+        if override and self._libname == "libquickwit_amd.so":
+            path = override
+        else:
+            path = os.path.join(REPO_ROOT, self._libname)
         if not os.path.exists(path):
             raise FileNotFoundError(
                 f"{path} not built — run __graft_entry__.build() first")

@@ -7,8 +7,16 @@
 
 namespace qw {
 
-constexpr uint32_t TILE_DOCS = 8192;   // docs per workgroup tile
+// docs per workgroup tile — build-time switch (-DQW_TILE_DOCS=4096) so the
+// LDS-footprint/occupancy tradeoff can be measured as whole-library variants
+#ifndef QW_TILE_DOCS
+#define QW_TILE_DOCS 8192
+#endif
+constexpr uint32_t TILE_DOCS = QW_TILE_DOCS;
 constexpr uint32_t TILE_THREADS = 256; // 4 waves
+// BM25 K tables staged to LDS when the query has at most this many scored
+// fields (1 KiB each)
+constexpr uint32_t KTAB_LDS_MAX = 4;
 constexpr uint32_t AGG_LDS_BUCKETS = 2048;
 constexpr uint32_t TOPK_BINS = 4096;
 
@@ -171,6 +179,14 @@ struct QueryDev {
     uint64_t preds_off;
     uint64_t aggs_off;
     uint64_t ktabs_off;     // scratch: n_fields * 256 f32 BM25 K tables
+    // LDS staging for the scoring decode path (cuts the per-posting critical
+    // path from two dependent global loads — norm byte, then K float — to
+    // two LDS reads). 0 disables; sections appended after the compile-time
+    // SmemMap, host adds the bytes to the dynamic-LDS size.
+    uint32_t n_ktabs;          // K tables to copy to LDS (0 = read global)
+    uint32_t _pad_ktab;
+    uint64_t norms_stage_off;  // common fieldnorms byte offset of all scored
+                               // terms (0 = heterogenous fields, read global)
     // result offsets (all in `results`)
     uint64_t tile_counts_off;  // u32[n_tiles]
     uint64_t cand_count_off;   // u32 (atomic)

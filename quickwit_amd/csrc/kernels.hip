@@ -122,15 +122,22 @@ constexpr uint32_t tile_lds_bytes(bool ns, bool nb, bool na, bool nc) {
 template <bool SCORE>
 __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t tile,
                                  uint32_t tile_lo, uint32_t tile_hi, float* score,
-                                 uint32_t* bitset, bool scoring) {
+                                 uint32_t* bitset, bool scoring,
+                                 const float* ktab_lds = nullptr,
+                                 const uint8_t* norms_lds = nullptr) {
     const uint32_t* ranges = (const uint32_t*)(q.scratch + t.ranges_off);
     uint32_t blo = ranges[tile], bhi = ranges[q.n_tiles + tile];
     if (blo >= bhi) return;
     const SkipEntryDev* __restrict__ skip = (const SkipEntryDev*)(q.split + t.skip_off);
     const uint32_t* __restrict__ payload = (const uint32_t*)(q.split + t.payload_off);
     const uint8_t* __restrict__ norms = t.norms_off ? q.split + t.norms_off : nullptr;
+    // LDS-staged fieldnorms for this tile (valid when the term's norms are
+    // the staged section) and K tables: per-posting lookups leave HBM
+    const bool nlds = norms_lds && t.norms_off &&
+                      t.norms_off == q.norms_stage_off;
     const float* __restrict__ ktab =
-        (const float*)(q.scratch + q.ktabs_off) + 256 * t.ktab_idx;
+        (ktab_lds ? ktab_lds : (const float*)(q.scratch + q.ktabs_off)) +
+        256 * t.ktab_idx;
     uint32_t wave = threadIdx.x >> 6;
     uint32_t lane = lane_id();
     // one 128-doc block per 32-lane HALF-wave (4 elements/sub-lane): twice
@@ -206,11 +213,20 @@ __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t t
                     ++tf0; ++tf1; ++tf2; ++tf3;
                 }
             }
-            // BM25: W * tf / (tf + K[normid]) — same op order as the oracle
-            float K0 = ktab[norms ? norms[min(doc0, q.num_docs - 1)] : 1];
-            float K1 = ktab[norms ? norms[min(doc1, q.num_docs - 1)] : 1];
-            float K2 = ktab[norms ? norms[min(doc2, q.num_docs - 1)] : 1];
-            float K3 = ktab[norms ? norms[min(doc3, q.num_docs - 1)] : 1];
+            // BM25: W * tf / (tf + K[normid]) — same op order as the oracle.
+            // Out-of-tile docs (boundary blocks) read a clamped index; their
+            // scores are discarded by the range guard below.
+            auto nid = [&](uint32_t doc) -> uint32_t {
+                if (nlds) {
+                    uint32_t li = doc - tile_lo;
+                    return norms_lds[li < TILE_DOCS ? li : 0];
+                }
+                return norms ? norms[min(doc, q.num_docs - 1)] : 1;
+            };
+            float K0 = ktab[nid(doc0)];
+            float K1 = ktab[nid(doc1)];
+            float K2 = ktab[nid(doc2)];
+            float K3 = ktab[nid(doc3)];
             s0 = t.weight * (float(tf0) / (float(tf0) + K0));
             s1 = t.weight * (float(tf1) / (float(tf1) + K1));
             s2 = t.weight * (float(tf2) / (float(tf2) + K2));
@@ -451,6 +467,25 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
     uint32_t* sc_agg_matched = (uint32_t*)(smem + M::tail_off);
     uint32_t* sc_wave_base = sc_agg_matched + 4;
     uint32_t* sc_cand_base = sc_wave_base + 4;
+    // runtime-optional LDS staging (appended after the compile-time map;
+    // host adds the bytes to the launch's dynamic-LDS size): per-tile
+    // fieldnorms u8[TILE_DOCS] + BM25 K tables f32[256*n_ktabs]
+    uint8_t* sc_norms = nullptr;
+    float* sc_ktabs = nullptr;
+    if (NS) {
+        uint32_t soff = M::total;
+        if (q.norms_stage_off) {
+            sc_norms = smem + soff;
+            soff += TILE_DOCS;
+        }
+        if (q.scoring && q.n_ktabs) {
+            sc_ktabs = (float*)(smem + soff);
+            const float* g = (const float*)(q.scratch + q.ktabs_off);
+            for (uint32_t i = threadIdx.x; i < 256u * q.n_ktabs;
+                 i += TILE_THREADS)
+                sc_ktabs[i] = g[i];
+        }
+    }
 
     const TermDev* __restrict__ terms = (const TermDev*)(q.scratch + q.terms_off);
     const PredDev* __restrict__ preds = (const PredDev*)(q.scratch + q.preds_off);
@@ -499,10 +534,18 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
         uint32_t tile_hi = min(tile_lo + TILE_DOCS, q.num_docs);
         bool have_should = q.msm > 0;  // shoulds REQUIRED (msm>=1); msm==0 =>
                                        // shoulds optional, only add scores
-        // ---- zero LDS
+        // ---- zero LDS (+ stage this tile's fieldnorms)
         if (NS)
             for (uint32_t i = threadIdx.x; i < TILE_DOCS; i += TILE_THREADS)
                 sc_score[i] = 0.f;
+        if (NS && sc_norms) {
+            const uint32_t* g32 =
+                (const uint32_t*)(q.split + q.norms_stage_off + tile_lo);
+            uint32_t* d32 = (uint32_t*)sc_norms;
+            uint32_t words = (tile_hi - tile_lo + 3) >> 2;  // split img has
+            for (uint32_t i = threadIdx.x; i < words; i += TILE_THREADS)
+                d32[i] = g32[i];                            // +64B pad
+        }
         if (NB)
             for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS) {
                 sc_bits_acc[i] = 0;
@@ -518,7 +561,8 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
             for (uint32_t t = 0; t < q.n_terms; ++t)
                 if (terms[t].role == ROLE_SHOULD)
                     decode_term_tile<true>(q, terms[t], tile, tile_lo, tile_hi,
-                                           sc_score, nullptr, q.scoring);
+                                           sc_score, nullptr, q.scoring,
+                                           sc_ktabs, sc_norms);
         __syncthreads();
 
         // ---- must groups: union the group's terms into a temp bitset,
@@ -537,7 +581,8 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                                             nullptr, tmp, false);
                     if (NS && q.scoring && terms[t].weight != 0.f)
                         decode_term_tile<true>(q, terms[t], tile, tile_lo, tile_hi,
-                                               sc_score, nullptr, true);
+                                               sc_score, nullptr, true,
+                                               sc_ktabs, sc_norms);
                 }
                 __syncthreads();
                 for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS)
@@ -986,6 +1031,10 @@ inline uint32_t launch_leaf_tile(bool ns, bool nb, bool na, bool nc, dim3 grid,
                                  uint32_t tile_base, uint32_t tile_end,
                                  uint32_t do_count) {
     uint32_t lds = tile_lds_bytes(ns, nb, na, nc);
+    // runtime-optional staging sections (norms tile + K tables): pay LDS
+    // only for what this query stages
+    if (ns && q.norms_stage_off) lds += TILE_DOCS;
+    if (ns && q.scoring && q.n_ktabs) lds += 1024 * q.n_ktabs;
     #define QW_CASE(NS_, NB_, NA_, NC_)                                           \
         if (ns == NS_ && nb == NB_ && na == NA_ && nc == NC_) {                   \
             hipLaunchKernelGGL((k_leaf_tile_t<NS_, NB_, NA_, NC_>), grid,         \

@@ -1811,6 +1811,26 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         q.preds_off = off_preds;
         q.aggs_off = off_aggs;
         q.ktabs_off = off_ktabs;
+        // LDS staging of fieldnorms + K tables for the scoring decode path
+        // (kernels.hip): enabled when every scored term's norms live in ONE
+        // section (single-field queries — the flagship shape) and the K
+        // tables fit the LDS budget. QW_NO_LDS_NORMS / QW_NO_LDS_KTAB are
+        // perf-experiment kill switches.
+        if (fq.scoring) {
+            static const bool no_norms = getenv("QW_NO_LDS_NORMS") != nullptr;
+            static const bool no_ktab = getenv("QW_NO_LDS_KTAB") != nullptr;
+            uint64_t common = 0;
+            bool uniform = true;
+            for (const TermDev& t : terms)
+                if (t.norms_off) {
+                    if (!common) common = t.norms_off;
+                    else if (common != t.norms_off) uniform = false;
+                }
+            q.norms_stage_off = (!no_norms && uniform) ? common : 0;
+            q.n_ktabs = (!no_ktab && ktab_fields.size() <= KTAB_LDS_MAX)
+                            ? uint32_t(ktab_fields.size())
+                            : 0;
+        }
         q.tile_counts_off = r_tile_counts;
         q.cand_count_off = r_cand_count;
         q.cand_off = r_cand;
