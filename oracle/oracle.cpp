@@ -913,7 +913,7 @@ static SplitResult search_split(const SplitView& sv, const pb::SearchRequest& re
         auto mk_hit = [&](uint32_t doc, float score) {
             pb::PartialHit h;
             h.split_id = sv.split_id;
-            h.segment_ord = 0;
+            h.segment_ord = sv.segment_ord;
             h.doc_id = doc;
             if (!specs.empty()) h.sort_value = sort_value_of(specs[0], sv, doc, score);
             if (specs.size() > 1) h.sort_value2 = sort_value_of(specs[1], sv, doc, score);
@@ -1009,7 +1009,8 @@ typedef struct qw_oracle_ctx qw_oracle_ctx;
 struct qw_oracle_ctx {
     struct SplitHolder {
         std::vector<uint8_t> data;
-        qw::SplitView view;
+        qw::SplitView view;                  // single-segment (or segment 0)
+        std::vector<qw::SplitView> seg_views;  // QWA2: one view per segment
     };
     std::map<std::string, std::unique_ptr<SplitHolder>> splits;
     std::string last_error;
@@ -1037,8 +1038,30 @@ int qw_oracle_add_split(qw_oracle_ctx* ctx, const char* split_id, const uint8_t*
     try {
         auto h = std::make_unique<qw_oracle_ctx::SplitHolder>();
         h->data.assign(data, data + len);
-        h->view.parse(h->data.data(), h->data.size());
-        h->view.split_id = split_id;
+        if (len >= 88 && memcmp(data, "QWAMDSP2", 8) == 0) {
+            // QWA2 multi-segment container (same layout the product reads)
+            uint64_t moff, mlen;
+            memcpy(&moff, h->data.data() + len - 24, 8);
+            memcpy(&mlen, h->data.data() + len - 16, 8);
+            mj::ValuePtr meta =
+                mj::parse((const char*)h->data.data() + moff, mlen);
+            uint32_t ord = 0;
+            for (auto& sgv : meta->at("segments")->arr) {
+                uint64_t off = uint64_t(sgv->at("off")->as_i64());
+                uint64_t slen = uint64_t(sgv->at("len")->as_i64());
+                qw::SplitView v;
+                v.parse(h->data.data() + off, slen);
+                v.split_id = split_id;
+                v.segment_ord = ord++;
+                h->seg_views.push_back(std::move(v));
+            }
+            if (h->seg_views.empty())
+                throw std::runtime_error("QWA2: no segments");
+            h->view = h->seg_views[0];
+        } else {
+            h->view.parse(h->data.data(), h->data.size());
+            h->view.split_id = split_id;
+        }
         ctx->splits[split_id] = std::move(h);
         return 0;
     } catch (const std::exception& e) {
@@ -1067,8 +1090,9 @@ int qw_oracle_leaf_search(qw_oracle_ctx* ctx, const uint8_t* req_pb, size_t req_
         Schema schema = Schema::parse(lreq.doc_mappers[0]);
 
         struct Task {
-            const qw_oracle_ctx::SplitHolder* holder;
+            const qw::SplitView* view;  // one task per (split, segment)
             std::string split_id;
+            bool first_seg;             // first segment of its split
         };
         std::vector<Task> tasks;
         for (auto& lr : lreq.leaf_requests)
@@ -1076,50 +1100,74 @@ int qw_oracle_leaf_search(qw_oracle_ctx* ctx, const uint8_t* req_pb, size_t req_
                 auto it = ctx->splits.find(so.split_id);
                 if (it == ctx->splits.end())
                     throw std::runtime_error("unknown split: " + so.split_id);
-                tasks.push_back({it->second.get(), so.split_id});
+                const auto* holder = it->second.get();
+                if (holder->seg_views.empty())
+                    tasks.push_back({&holder->view, so.split_id, true});
+                else {
+                    bool first = true;
+                    for (auto& v : holder->seg_views) {
+                        tasks.push_back({&v, so.split_id, first});
+                        first = false;
+                    }
+                }
             }
 
         std::vector<SplitResult> results(tasks.size());
 #pragma omp parallel for schedule(dynamic)
         for (size_t i = 0; i < tasks.size(); ++i) {
             try {
-                results[i] = search_split(tasks[i].holder->view, req, schema);
+                results[i] = search_split(*tasks[i].view, req, schema);
             } catch (const std::exception& e) {
                 results[i].error = e.what();
             }
         }
 
         pb::LeafSearchResponse resp;
-        resp.num_attempted_splits = tasks.size();
+        for (auto& t : tasks)
+            if (t.first_seg) resp.num_attempted_splits++;
         IntermediateAggResults merged_aggs;
         bool any_aggs = false;
         std::vector<pb::PartialHit> all_hits;
         pb::LeafResourceStats rstats;
         rstats.search_pool_cpu_threads = uint64_t(omp_get_max_threads());
         uint64_t worst_key = 0;
-        for (size_t i = 0; i < tasks.size(); ++i) {
-            SplitResult& r = results[i];
-            if (!r.error.empty()) {
+        for (size_t i = 0; i < tasks.size();) {
+            size_t j = i + 1;  // [i, j) = this split's segment tasks
+            while (j < tasks.size() && !tasks[j].first_seg) ++j;
+            std::string err;
+            uint64_t split_hits = 0, split_docs = 0, split_micros = 0;
+            for (size_t t = i; t < j; ++t) {
+                if (!results[t].error.empty() && err.empty())
+                    err = results[t].error;
+                split_hits += results[t].num_hits;
+                split_docs += tasks[t].view->num_docs;
+                split_micros += results[t].cpu_micros;
+            }
+            if (!err.empty()) {
                 pb::SplitSearchError e;
-                e.error = r.error;
+                e.error = err;
                 e.split_id = tasks[i].split_id;
                 e.retryable_error = true;
                 resp.failed_splits.push_back(std::move(e));
+                i = j;
                 continue;
             }
             resp.num_successful_splits++;
-            resp.num_hits += r.num_hits;
-            for (auto& h : r.hits) all_hits.push_back(std::move(h));
-            if (r.has_aggs) {
-                if (!any_aggs) {
-                    merged_aggs = std::move(r.aggs);
-                    any_aggs = true;
-                } else merged_aggs.merge(r.aggs);
+            resp.num_hits += split_hits;
+            for (size_t t = i; t < j; ++t) {
+                SplitResult& r = results[t];
+                for (auto& h : r.hits) all_hits.push_back(std::move(h));
+                if (r.has_aggs) {
+                    if (!any_aggs) {
+                        merged_aggs = std::move(r.aggs);
+                        any_aggs = true;
+                    } else merged_aggs.merge(r.aggs);
+                }
             }
             pb::SplitResourceStats ss;
-            ss.split_num_docs = tasks[i].holder->view.num_docs;
-            ss.matched_num_docs = r.num_hits;
-            ss.cpu_search_microsecs = r.cpu_micros;
+            ss.split_num_docs = split_docs;
+            ss.matched_num_docs = split_hits;
+            ss.cpu_search_microsecs = split_micros;
             rstats.localexec_num_splits++;
             rstats.localexec_num_docs += ss.split_num_docs;
             rstats.split_resources_sum.split_num_docs += ss.split_num_docs;
@@ -1129,6 +1177,7 @@ int qw_oracle_leaf_search(qw_oracle_ctx* ctx, const uint8_t* req_pb, size_t req_
                 worst_key = ss.cpu_search_microsecs;
                 rstats.split_resources_worst = ss;
             }
+            i = j;
         }
         // cross-split merge (merge_fruits semantics, collector.rs:832-861)
         int order1 = req.sort_fields.empty() ? 1 : req.sort_fields[0].sort_order;

@@ -58,9 +58,15 @@ namespace qw {
 
 struct DeviceSplit {
     std::vector<uint8_t> host;  // host copy: term-dict lookups + meta parse
-    SplitView view;             // views into `host`
-    uint8_t* d_image = nullptr; // full QWA1 file image resident in HBM
+    SplitView view;             // views into `host` (segment 0 for QWA2)
+    uint8_t* d_image = nullptr; // full file image resident in HBM
     size_t len = 0;
+    uint64_t total_docs = 0;    // across all segments
+    // QWA2 multi-segment container (collector.rs:475-594 per-segment
+    // collection): one child per segment; child host/view/d_image alias the
+    // parent allocation (owns_image=false)
+    std::vector<std::unique_ptr<DeviceSplit>> segs;
+    bool owns_image = true;
 };
 
 struct KernelTimer {
@@ -1150,6 +1156,9 @@ struct SplitResult {
     bool has_aggs = false;
     uint64_t micros = 0;
     std::string error;
+    // required terms this SEGMENT proved absent; the caller populates the
+    // absence cache only with terms absent from EVERY segment of the split
+    std::vector<std::pair<std::string, std::string>> absent_required;
 };
 
 // sort specs, mirroring the oracle's (collector.rs:403-414 sort-key
@@ -1232,6 +1241,8 @@ static uint8_t* bitmap_eval(qw_ctx* ctx, const DeviceSplit& ds,
     size_t bm_bytes = size_t(n_tiles) * (TILE_DOCS / 32) * 4;
     std::string key = sv.split_id;
     key += '\0';
+    key += std::to_string(sv.segment_ord);
+    key += '\x1e';
     plan_fingerprint(node, key);
     uint8_t* bm = nullptr;
     qw_check_budget(ctx, bm_bytes);
@@ -1303,6 +1314,8 @@ static uint8_t* resolve_hitset(qw_ctx* ctx, const DeviceSplit& ds,
     const SplitView& sv = ds.view;
     std::string key = sv.split_id;
     key += '\0';
+    key += std::to_string(sv.segment_ord);
+    key += '\x1e';
     plan_fingerprint(inner, key);
     if (uint8_t* bm = ctx->hitsets.get(key)) return bm;
 
@@ -1556,12 +1569,10 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         p.abs_bitmap = (uint64_t)bmp;
         fq.preds.push_back(p);
     }
-    // populate the absence cache with required terms flattening proved
-    // absent (the next search for them skips this split up front)
-    for (auto& ab : fq.absent_required)
-        if (ctx->absence.keys.size() < ctx->absence.cap)
-            ctx->absence.keys.insert(
-                AbsenceCache::key(sv.split_id, ab.first, ab.second));
+    // report required terms flattening proved absent; the caller inserts
+    // into the absence cache (for multi-segment splits only terms absent
+    // from EVERY segment qualify)
+    out.absent_required = fq.absent_required;
     // CACHE nodes in filter position -> device HitSet bitmaps (PRED_BITSET)
     for (const PlanNode* cn : fq.cache_nodes) {
         PredDev p{};
@@ -1817,8 +1828,13 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         // tables fit the LDS budget. QW_NO_LDS_NORMS / QW_NO_LDS_KTAB are
         // perf-experiment kill switches.
         if (fq.scoring) {
-            static const bool no_norms = getenv("QW_NO_LDS_NORMS") != nullptr;
-            static const bool no_ktab = getenv("QW_NO_LDS_KTAB") != nullptr;
+            // measured on MI355X at the 100M flagship: norms staging LOSES
+            // (the +8KB LDS drops occupancy 4->3 WGs/CU while the norm
+            // gathers already hit L2 — union covers ~25% of docs, ~16 hits
+            // per cacheline); ktab staging is noise. Both stay available as
+            // opt-in experiment switches (gpurun_out/r02_exp_variants.log).
+            static const bool want_norms = getenv("QW_LDS_NORMS") != nullptr;
+            static const bool want_ktab = getenv("QW_LDS_KTAB") != nullptr;
             uint64_t common = 0;
             bool uniform = true;
             for (const TermDev& t : terms)
@@ -1826,8 +1842,8 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     if (!common) common = t.norms_off;
                     else if (common != t.norms_off) uniform = false;
                 }
-            q.norms_stage_off = (!no_norms && uniform) ? common : 0;
-            q.n_ktabs = (!no_ktab && ktab_fields.size() <= KTAB_LDS_MAX)
+            q.norms_stage_off = (want_norms && uniform) ? common : 0;
+            q.n_ktabs = (want_ktab && ktab_fields.size() <= KTAB_LDS_MAX)
                             ? uint32_t(ktab_fields.size())
                             : 0;
         }
@@ -2075,7 +2091,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
     auto mk_hit = [&](uint32_t doc, float score) {
         pb::PartialHit h;
         h.split_id = sv.split_id;
-        h.segment_ord = 0;
+        h.segment_ord = sv.segment_ord;
         h.doc_id = doc;
         if (!specs.empty()) h.sort_value = sort_value_of(specs[0], doc, score);
         if (specs.size() > 1) h.sort_value2 = sort_value_of(specs[1], doc, score);
@@ -2155,6 +2171,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             }
         }
     }
+    mark("build_hits");
 
     // ---- aggregation download + assembly (QAGG1 intermediate)
     if (do_aggs) {
@@ -2380,6 +2397,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         }
         out.has_aggs = true;
     }
+    mark("aggs_assembly");
 
     out.micros = uint64_t(std::chrono::duration_cast<std::chrono::microseconds>(
                               std::chrono::steady_clock::now() - t0)
@@ -2394,6 +2412,57 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
 // FindTraceIdsAggregation variant (Jaeger traces) is out of scope.
 // Missing split timestamps read as 0 (prost's Option accessor default, the
 // exact behavior of split.timestamp_start()/timestamp_end() in the reference).
+// one split = one or more segments (collector.rs:475-594 per-segment
+// collection): run the per-segment search and merge within the split with
+// the exact (sort_value, sort_value2, split, segment_ord, doc) order
+static SplitResult search_device_split(qw_ctx* ctx, const DeviceSplit& ds,
+                                       const pb::SearchRequest& req,
+                                       const Schema& schema) {
+    using namespace qw;
+    std::vector<const DeviceSplit*> segs;
+    if (ds.segs.empty()) segs.push_back(&ds);
+    else
+        for (auto& sp : ds.segs) segs.push_back(sp.get());
+    SplitResult out;
+    std::map<std::pair<std::string, std::string>, size_t> absent;
+    for (const DeviceSplit* seg : segs) {
+        SplitResult r = search_split_gpu(ctx, *seg, req, schema);
+        out.num_hits += r.num_hits;
+        for (auto& h : r.hits) out.hits.push_back(std::move(h));
+        if (r.has_aggs) {
+            if (!out.has_aggs) {
+                out.aggs = std::move(r.aggs);
+                out.has_aggs = true;
+            } else out.aggs.merge(r.aggs);
+        }
+        out.micros += r.micros;
+        for (auto& ab : r.absent_required) absent[ab]++;
+    }
+    // absence cache: only terms absent from EVERY segment of the split
+    // qualify (the reference early-aborts when warmup finds the term in no
+    // segment, leaf.rs:315-320)
+    const std::string& sid = segs[0]->view.split_id;
+    for (auto& kv : absent)
+        if (kv.second == segs.size() &&
+            ctx->absence.keys.size() < ctx->absence.cap)
+            ctx->absence.keys.insert(
+                AbsenceCache::key(sid, kv.first.first, kv.first.second));
+    uint64_t leaf_max = req.max_hits + req.start_offset;
+    if (segs.size() > 1 && leaf_max > 0 && !out.hits.empty()) {
+        int order1 = req.sort_fields.empty() ? 1 : req.sort_fields[0].sort_order;
+        int order2 =
+            req.sort_fields.size() > 1 ? req.sort_fields[1].sort_order : 1;
+        size_t k = std::min<uint64_t>(leaf_max, out.hits.size());
+        std::partial_sort(out.hits.begin(), out.hits.begin() + k,
+                          out.hits.end(),
+                          [&](const pb::PartialHit& a, const pb::PartialHit& b) {
+                              return hit_before(a, b, order1, order2);
+                          });
+        out.hits.resize(k);
+    }
+    return out;
+}
+
 static int64_t split_ts_start(const pb::SplitIdAndFooterOffsets& s) {
     return s.timestamp_start ? *s.timestamp_start : 0;
 }
@@ -2551,15 +2620,49 @@ int32_t qw_ctx_add_split(qw_ctx* ctx, const char* split_id, const uint8_t* data,
     try {
         auto ds = std::make_unique<DeviceSplit>();
         ds->host.assign(data, data + len);
-        ds->view.parse(ds->host.data(), len);
-        // the caller's split id is authoritative (the request names splits
-        // by SplitIdAndFooterOffsets.split_id and the response must echo
-        // it); the container's own id is only a default
-        ds->view.split_id = split_id;
-        if (ds->view.version < 2)
-            throw std::runtime_error(
-                "QWA1 v1 container lacks posting segment anchors (regenerate "
-                "the split with the current writer)");
+        if (len >= 88 && memcmp(data, "QWAMDSP2", 8) == 0) {
+            // QWA2 multi-segment container: outer meta lists per-segment
+            // QWA1 images; children alias the parent host/device image
+            if (memcmp(data + len - 8, "QWA1FOOT", 8) != 0)
+                throw std::runtime_error("bad QWA2 footer");
+            uint64_t moff, mlen;
+            memcpy(&moff, data + len - 24, 8);
+            memcpy(&mlen, data + len - 16, 8);
+            if (moff + mlen > len) throw std::runtime_error("bad QWA2 footer");
+            mj::ValuePtr meta = mj::parse((const char*)data + moff, mlen);
+            uint32_t ord = 0;
+            for (auto& sgv : meta->at("segments")->arr) {
+                uint64_t off = uint64_t(sgv->at("off")->as_i64());
+                uint64_t slen = uint64_t(sgv->at("len")->as_i64());
+                if (off + slen > len)
+                    throw std::runtime_error("QWA2: segment out of bounds");
+                auto child = std::make_unique<DeviceSplit>();
+                child->owns_image = false;
+                child->view.parse(ds->host.data() + off, slen);
+                if (child->view.version < 2)
+                    throw std::runtime_error("QWA2 segment: v1 container");
+                child->view.split_id = split_id;
+                child->view.segment_ord = ord++;
+                child->len = slen;
+                child->total_docs = child->view.num_docs;
+                ds->total_docs += child->view.num_docs;
+                ds->segs.push_back(std::move(child));
+            }
+            if (ds->segs.empty())
+                throw std::runtime_error("QWA2: no segments");
+            ds->view = ds->segs[0]->view;  // pointers alias ds->host
+        } else {
+            ds->view.parse(ds->host.data(), len);
+            // the caller's split id is authoritative (the request names
+            // splits by SplitIdAndFooterOffsets.split_id and the response
+            // must echo it); the container's own id is only a default
+            ds->view.split_id = split_id;
+            if (ds->view.version < 2)
+                throw std::runtime_error(
+                    "QWA1 v1 container lacks posting segment anchors "
+                    "(regenerate the split with the current writer)");
+            ds->total_docs = ds->view.num_docs;
+        }
         ds->len = len;
         HIP_CHECK(hipSetDevice(ctx->device));
         try {
@@ -2571,6 +2674,21 @@ int32_t qw_ctx_add_split(qw_ctx* ctx, const char* split_id, const uint8_t* data,
         HIP_CHECK(hipMalloc(&ds->d_image, len + 64));  // +64: decode overread pad
         HIP_CHECK(hipMemcpy(ds->d_image, data, len, hipMemcpyHostToDevice));
         ctx->hbm_used += len + 64;
+        {
+            // children point into the parent device image (their views'
+            // offsets are relative to their own segment base)
+            size_t ci = 0;
+            uint64_t moff2, mlen2;
+            if (!ds->segs.empty()) {
+                mj::ValuePtr meta2;
+                memcpy(&moff2, data + len - 24, 8);
+                memcpy(&mlen2, data + len - 16, 8);
+                meta2 = mj::parse((const char*)data + moff2, mlen2);
+                for (auto& sgv : meta2->at("segments")->arr)
+                    ds->segs[ci++]->d_image =
+                        ds->d_image + uint64_t(sgv->at("off")->as_i64());
+            }
+        }
         ctx->splits[split_id] = std::move(ds);
         return QW_OK;
     } catch (const std::exception& e) {
@@ -2747,7 +2865,7 @@ int32_t qw_leaf_search(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
             }
             SplitResult r;
             try {
-                r = search_split_gpu(ctx, *t.ds, t.req, schema);
+                r = search_device_split(ctx, *t.ds, t.req, schema);
             } catch (const std::exception& e) {
                 // per-split failure is data, not an exception (leaf.rs:2143)
                 pb::SplitSearchError se;
@@ -2783,7 +2901,7 @@ int32_t qw_leaf_search(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
                     filter.record_new_worst_hit(all_hits.back());
             }
             pb::SplitResourceStats ss;
-            ss.split_num_docs = t.ds->view.num_docs;
+            ss.split_num_docs = t.ds->total_docs;
             ss.matched_num_docs = r.num_hits;
             ss.cpu_search_microsecs = r.micros;  // GPU wall (cpu_search analog)
             rstats.localexec_num_splits++;
@@ -2827,27 +2945,41 @@ int32_t qw_fetch_docs(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
             const pb::PartialHit& ha = freq.partial_hits[a];
             const pb::PartialHit& hb = freq.partial_hits[b];
             if (ha.split_id != hb.split_id) return ha.split_id < hb.split_id;
+            if (ha.segment_ord != hb.segment_ord)
+                return ha.segment_ord < hb.segment_ord;
             return ha.doc_id < hb.doc_id;
         });
         pb::FetchDocsResponse resp;
         std::string cur_split;
+        uint32_t cur_seg = ~0u;
         const DeviceSplit* ds = nullptr;
+        const SplitView* svp = nullptr;
         std::vector<uint8_t> block;       // decompressed block cache
         int64_t cached_block = -1;
         for (size_t oi : order) {
             const pb::PartialHit& h = freq.partial_hits[oi];
-            if (h.split_id != cur_split) {
+            if (h.split_id != cur_split || h.segment_ord != cur_seg) {
                 auto it = ctx->splits.find(h.split_id);
                 if (it == ctx->splits.end())
                     throw std::runtime_error("unknown split: " + h.split_id);
                 ds = it->second.get();
                 cur_split = h.split_id;
+                cur_seg = h.segment_ord;
+                if (ds->segs.empty()) {
+                    if (h.segment_ord != 0)
+                        throw std::runtime_error("bad segment ord");
+                    svp = &ds->view;
+                } else {
+                    if (h.segment_ord >= ds->segs.size())
+                        throw std::runtime_error("bad segment ord");
+                    svp = &ds->segs[h.segment_ord]->view;
+                }
                 cached_block = -1;
             }
-            const DocStoreView& d = ds->view.docstore;
+            const DocStoreView& d = svp->docstore;
             if (!d.present)
                 throw std::runtime_error("split has no docstore: " + h.split_id);
-            if (h.doc_id >= ds->view.num_docs)
+            if (h.doc_id >= svp->num_docs)
                 throw std::runtime_error("doc id out of range");
             // binary search the block whose [first, next_first) covers doc
             uint32_t lo = 0, hi = d.n_blocks;
@@ -2917,9 +3049,13 @@ int32_t qw_leaf_list_terms(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
                 resp.failed_splits.push_back(std::move(se));
                 continue;
             }
-            const SplitView& sv = it->second->view;
-            const TextFieldView* f = sv.text_field(req.field);
-            if (!f) {
+            const DeviceSplit* dsp = it->second.get();
+            std::vector<const SplitView*> views;
+            if (dsp->segs.empty()) views.push_back(&dsp->view);
+            else
+                for (auto& sg : dsp->segs) views.push_back(&sg->view);
+            const TextFieldView* f0 = views[0]->text_field(req.field);
+            if (!f0) {
                 pb::SplitSearchError se;
                 se.split_id = so.split_id;
                 se.error = "couldn't get field named \"" + req.field +
@@ -2929,17 +3065,23 @@ int32_t qw_leaf_list_terms(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
                 continue;
             }
             resp.num_attempted_splits++;
-            std::vector<std::string> terms;
-            for (uint32_t t = 0; t < f->num_terms; ++t) {
-                const char* s = (const char*)f->h_term_bytes + f->h_term_offsets[t];
-                size_t sl = f->h_term_offsets[t + 1] - f->h_term_offsets[t];
-                std::string term(s, sl);
-                if (req.start_key && term < *req.start_key) continue;  // ge
-                if (req.end_key && term >= *req.end_key) break;        // lt
-                terms.push_back(std::move(term));
-                if (req.max_hits && terms.size() >= *req.max_hits) break;
+            // each segment dict is sorted: one run per segment, k-merged +
+            // deduped below with the other splits' runs
+            for (const SplitView* svv : views) {
+                const TextFieldView* f = svv->text_field(req.field);
+                if (!f) continue;
+                std::vector<std::string> terms;
+                for (uint32_t t = 0; t < f->num_terms; ++t) {
+                    const char* s = (const char*)f->h_term_bytes + f->h_term_offsets[t];
+                    size_t sl = f->h_term_offsets[t + 1] - f->h_term_offsets[t];
+                    std::string term(s, sl);
+                    if (req.start_key && term < *req.start_key) continue;  // ge
+                    if (req.end_key && term >= *req.end_key) break;        // lt
+                    terms.push_back(std::move(term));
+                    if (req.max_hits && terms.size() >= *req.max_hits) break;
+                }
+                per_split.push_back(std::move(terms));
             }
-            per_split.push_back(std::move(terms));
         }
         // k-merge + dedup + global limit (list_terms.rs:291-300,392-400)
         std::vector<size_t> pos(per_split.size(), 0);

@@ -140,6 +140,8 @@ struct DocStoreView {
 
 struct SplitView {
     std::string split_id;
+    uint32_t segment_ord = 0;  // QWA2 multi-segment containers: this
+                               // segment's ordinal within the split
     uint32_t version = 1;
     uint32_t num_docs = 0;
     std::string timestamp_field;

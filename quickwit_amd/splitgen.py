@@ -675,3 +675,37 @@ def generate_split_numpy(split_ord: int, num_docs: int, seed: int = 42) -> bytes
             "tenant_name": (("ords", tenants, tenant_names), None),
         },
     )
+
+
+def concat_segments(segment_images: list, split_id: str) -> bytes:
+    """QWA2 multi-segment split container: outer magic "QWAMDSP2" +
+    per-segment complete QWA1 images (64-aligned) + outer meta JSON +
+    QWA1-shaped footer. The reference's splits carry multiple tantivy
+    segments (collector.rs:475-594 collects per segment); readers
+    (product qw_ctx_add_split, oracle, splitread.Split) run each segment
+    as its own doc-id space with PartialHit.segment_ord set."""
+    chunks = [b"QWAMDSP2", b"\0" * 56]
+    pos = 64
+    segs = []
+    total = 0
+    for img in segment_images:
+        assert img[:8] == MAGIC and img[-8:] == FOOTER_MAGIC, "not a QWA1 image"
+        moff, mlen = np.frombuffer(img[-24:-8], dtype="<u8")
+        ndocs = json.loads(img[int(moff):int(moff + mlen)])["num_docs"]
+        off = pos
+        chunks.append(img)
+        pos += len(img)
+        pad = (-pos) % ALIGN
+        if pad:
+            chunks.append(b"\0" * pad)
+            pos += pad
+        segs.append({"off": int(off), "len": len(img), "num_docs": int(ndocs)})
+        total += int(ndocs)
+    meta = json.dumps({"format": "QWA2", "version": 1, "split_id": split_id,
+                       "num_docs": total, "segments": segs},
+                      sort_keys=True).encode()
+    meta_off = pos
+    chunks.append(meta)
+    footer = np.array([meta_off, len(meta)], dtype="<u8").tobytes() + FOOTER_MAGIC
+    chunks.append(footer)
+    return b"".join(chunks)

@@ -127,3 +127,18 @@ def _unpack(payload_u32, word_off, bits, count):
             k += 1
         out[j] = v & ((np.uint64(1) << np.uint64(bits)) - np.uint64(1))
     return out
+
+
+class MultiSplit:
+    """QWA2 reader: .segments = list of Split, one per segment."""
+
+    def __init__(self, data: bytes):
+        assert data[:8] == b"QWAMDSP2", "not a QWA2 container"
+        assert data[-8:] == FOOTER_MAGIC
+        meta_off, meta_len = np.frombuffer(data[-24:-8], dtype="<u8")
+        self.meta = json.loads(data[int(meta_off): int(meta_off + meta_len)])
+        assert self.meta["format"] == "QWA2"
+        self.segments = [
+            Split(data[s["off"]: s["off"] + s["len"]])
+            for s in self.meta["segments"]
+        ]

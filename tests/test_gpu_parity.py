@@ -1496,3 +1496,120 @@ def test_memory_budget_accounting_and_refusal():
     with pytest.raises(RuntimeError, match="budget"):
         g2.add_split("m1", data)
     assert g2.memory_stats()[1] == 4096
+
+
+# ---------------------------------------------------- multi-segment (QWA2)
+def test_multi_segment_split_parity():
+    """QWA2 multi-segment split through the HIP path vs the oracle:
+    per-segment collection (collector.rs:475-594), (split, segment_ord,
+    doc) tie-breaks, agg merge across segments, absence-cache semantics
+    (a term must be absent from EVERY segment to be cached)."""
+    a = splitgen.generate_split(0, 30_000, seed=9)
+    b = splitgen.generate_split(1, 20_000, seed=9)
+    multi = splitgen.concat_segments([a, b], "s")
+    gpu = GpuSearcher(device=0)
+    cpu = OracleSearcher()
+    gpu.add_split("s", multi)
+    cpu.add_split("s", multi)
+    splits = [("s", 50_000)]
+
+    # scored top-K across segments
+    q = {"type": "bool", "should": [
+        {"type": "term", "field": "body", "value": "w%05d" % i}
+        for i in (9, 10, 11)]}
+    req = make_leaf_request(q, SCHEMA, splits, max_hits=20,
+                            sort_fields=[{"field_name": "_score",
+                                          "sort_order": 1}])
+    g, e = gpu.leaf_search(req), cpu.leaf_search(req)
+    assert_hits_equal(g, e, scored=True)
+    assert {h.get("segment_ord", 0) for h in g["partial_hits"]} == {0, 1}
+    for gh, eh in zip(g["partial_hits"], e["partial_hits"]):
+        assert gh.get("segment_ord", 0) == eh.get("segment_ord", 0)
+
+    # two-key field sort
+    req = make_leaf_request(
+        {"type": "term", "field": "severity_text", "value": "INFO"},
+        SCHEMA, splits, max_hits=25,
+        sort_fields=[{"field_name": "timestamp", "sort_order": 1},
+                     {"field_name": "tenant_id", "sort_order": 0}])
+    g, e = gpu.leaf_search(req), cpu.leaf_search(req)
+    assert g["num_hits"] == e["num_hits"]
+    assert [(h.get("segment_ord", 0), h.get("doc_id", 0))
+            for h in g["partial_hits"]] == \
+        [(h.get("segment_ord", 0), h.get("doc_id", 0))
+         for h in e["partial_hits"]]
+
+    # aggregations merge across segments
+    aggs = {"per_hour": {"date_histogram": {"field": "timestamp",
+                                            "fixed_interval": "3600000ms"}},
+            "per_tenant": {"terms": {"field": "tenant_name", "size": 10}}}
+    req = make_leaf_request({"type": "match_all"}, SCHEMA, splits,
+                            max_hits=0, aggregation=aggs)
+    g, e = gpu.leaf_search(req), cpu.leaf_search(req)
+    assert g["num_hits"] == e["num_hits"] == 50_000
+    gj = gpu.finalize_agg_json(g["intermediate_aggregation_result"], aggs)
+    ej = cpu.finalize_agg_json(e["intermediate_aggregation_result"], aggs)
+    assert gj == ej
+
+    # search_after pagination crossing a segment boundary
+    q = {"type": "bool", "should": [
+        {"type": "term", "field": "body", "value": "w%05d" % i}
+        for i in (9, 10, 11)]}
+    sort = [{"field_name": "_score", "sort_order": 1}]
+    seen = []
+    cursor = None
+    for _page in range(3):
+        req = make_leaf_request(q, SCHEMA, splits, max_hits=9,
+                                sort_fields=sort)
+        if cursor:
+            req["search_request"]["search_after"] = cursor
+        r = gpu.leaf_search(req)
+        hits = r.get("partial_hits", [])
+        if not hits:
+            break
+        seen.extend(hits)
+        cursor = hits[-1]
+    r_all = cpu.leaf_search(make_leaf_request(
+        q, SCHEMA, splits, max_hits=len(seen), sort_fields=sort))
+    assert [(h.get("segment_ord", 0), h.get("doc_id", 0)) for h in seen] == \
+        [(h.get("segment_ord", 0), h.get("doc_id", 0))
+         for h in r_all["partial_hits"]]
+
+
+def test_multi_segment_absence_cache_all_segments_rule():
+    """A required term present in one segment must NOT be cached as absent;
+    a term absent from every segment must."""
+    # craft two doc-writer segments: 'onlyseg0' appears only in segment 0
+    w0 = splitgen.SplitWriter(
+        {"timestamp_field": None,
+         "fields": [{"name": "body", "type": "text", "tokenizer": "default",
+                     "record": "freq", "fieldnorms": True}]}, "seg0",
+        store_docs=False)
+    w0.add_documents([{"body": "onlyseg0 shared"}, {"body": "shared"}])
+    w1 = splitgen.SplitWriter(
+        {"timestamp_field": None,
+         "fields": [{"name": "body", "type": "text", "tokenizer": "default",
+                     "record": "freq", "fieldnorms": True}]}, "seg1",
+        store_docs=False)
+    w1.add_documents([{"body": "shared word"}, {"body": "word"}])
+    multi = splitgen.concat_segments([w0.finalize(), w1.finalize()], "ms")
+    schema = {"timestamp_field": None, "fields": [
+        {"name": "body", "type": "text", "tokenizer": "default",
+         "record": "freq", "fieldnorms": True}]}
+    gpu = GpuSearcher(device=0)
+    gpu.add_split("ms", multi)
+
+    q1 = {"type": "bool", "must": [
+        {"type": "term", "field": "body", "value": "onlyseg0"}]}
+    r = gpu.leaf_search(make_leaf_request(q1, schema, [("ms", 4)], max_hits=5))
+    assert r["num_hits"] == 1
+    assert gpu.absence_cache_stats()[2] == 0  # present in seg0: not cached
+
+    q2 = {"type": "bool", "must": [
+        {"type": "term", "field": "body", "value": "nowhere"}]}
+    r = gpu.leaf_search(make_leaf_request(q2, schema, [("ms", 4)], max_hits=5))
+    assert r.get("num_hits", 0) == 0
+    assert gpu.absence_cache_stats()[2] == 1  # absent everywhere: cached
+    r2 = gpu.leaf_search(make_leaf_request(q2, schema, [("ms", 4)], max_hits=5))
+    assert r2.get("num_hits", 0) == 0
+    assert gpu.absence_cache_stats()[0] >= 1  # second probe hits
