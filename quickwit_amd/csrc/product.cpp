@@ -90,6 +90,23 @@ struct DevBuf {  // grow-only device scratch
     }
 };
 
+struct PinnedBuf {  // grow-only pinned host staging (fast H2D/D2H, no
+                    // per-call allocation)
+    uint8_t* p = nullptr;
+    size_t cap = 0;
+    void ensure(size_t n) {
+        if (n <= cap) return;
+        if (p) (void)hipHostFree(p);
+        p = nullptr;
+        cap = 0;
+        HIP_CHECK(hipHostMalloc((void**)&p, n));
+        cap = n;
+    }
+    ~PinnedBuf() {
+        if (p) (void)hipHostFree(p);
+    }
+};
+
 // LeafSearchCache restatement (quickwit-search/src/leaf_cache.rs): memoizes
 // per-(split, canonical request) LeafSearchResponse bytes in a byte-capacity
 // LRU (MemorySizedCache semantics). The key canonicalizes the request by
@@ -305,6 +322,7 @@ struct qw_ctx {
     hipStream_t stream = nullptr;
     hipEvent_t ev_start = nullptr, ev_stop = nullptr;
     qw::DevBuf d_scratch, d_results, d_survivors, d_cand2;
+    qw::PinnedBuf h_scratch, h_counts, h_surv, h_topk;  // pinned staging
     std::map<std::string, qw::KernelTimer> timers;
     // HBM accounting (SearchPermitProvider memory-budget analog,
     // search_permit_provider.rs:43-110): split images + scratch/result
@@ -1724,13 +1742,18 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             scratch_bytes += ap.pbounds[i].size() * 8;
         }
 
-    // block ranges per term per tile (host two-pointer walk over skip entries)
-    std::vector<uint32_t> ranges(ranges_bytes / 4, 0);
+    mark("plan");
+    // block ranges per term per tile (host two-pointer walk over skip
+    // entries), written straight into the PINNED staging buffer — no
+    // per-call zero-init/copy of the ~300KB scratch at 100M docs
+    ctx->h_scratch.ensure(scratch_bytes);
+    uint8_t* scratch = ctx->h_scratch.p;
     for (size_t i = 0; i < fq.terms.size(); ++i) {
         const FlatQuery::FTerm& t = fq.terms[i];
         const SkipEntry* sk = t.f->h_skip + t.f->h_skip_off[t.tid] / 16;
         uint32_t nb = t.f->h_n_blocks[t.tid];
-        uint32_t* lo = ranges.data() + (i * 2ull * n_tiles);
+        uint32_t* lo =
+            (uint32_t*)(scratch + off_ranges) + (i * 2ull * n_tiles);
         uint32_t* hi = lo + n_tiles;
         uint32_t b_lo = 0, b_hi = 0;
         for (uint32_t tile = 0; tile < n_tiles; ++tile) {
@@ -1744,30 +1767,32 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         }
         terms[i].ranges_off = off_ranges + i * 2ull * n_tiles * 4;
     }
+    mark("ranges");
 
-    std::vector<uint8_t> scratch(scratch_bytes, 0);
-    memcpy(scratch.data() + off_terms, terms.data(), terms.size() * sizeof(TermDev));
-    memcpy(scratch.data() + off_preds, fq.preds.data(),
+    memcpy(scratch + off_terms, terms.data(), terms.size() * sizeof(TermDev));
+    memcpy(scratch + off_preds, fq.preds.data(),
            fq.preds.size() * sizeof(PredDev));
-    memcpy(scratch.data() + off_aggs, ap.devs.data(), ap.devs.size() * sizeof(AggDev));
-    memcpy(scratch.data() + off_ktabs, ktabs.data(), ktabs.size() * 4);
-    memcpy(scratch.data() + off_ranges, ranges.data(), ranges_bytes);
+    memcpy(scratch + off_aggs, ap.devs.data(), ap.devs.size() * sizeof(AggDev));
+    memcpy(scratch + off_ktabs, ktabs.data(), ktabs.size() * 4);
     for (size_t i = 0; i < ap.devs.size(); ++i)
         if (!ap.pbounds[i].empty())
-            memcpy(scratch.data() + ap.devs[i].p_bound_off,
+            memcpy(scratch + ap.devs[i].p_bound_off,
                    ap.pbounds[i].data(), ap.pbounds[i].size() * 8);
 
     uint64_t matched = 0;
-    mark("plan+assembly");
+    mark("assembly");
     std::vector<uint64_t> top_keys;  // survivors, sorted best-first
     std::function<void(uint64_t)> rerun_select;  // search_after retry hook
     uint64_t sel_band_n = 0, sel_kwant = 0;
-    std::vector<uint32_t> hist0(TOPK_BINS);  // prefetched selection pass 0
+    // prefetched selection pass-0 histogram, in pinned staging (async D2H
+    // to pageable memory degrades to a blocking staged copy)
+    ctx->h_surv.ensure(TOPK_BINS * 4);
+    uint32_t* hist0 = (uint32_t*)ctx->h_surv.p;
 
     if (need_kernel) {
         qw_ensure_acct(ctx, ctx->d_scratch, scratch_bytes);
         qw_ensure_acct(ctx, ctx->d_results, results_bytes);
-        HIP_CHECK(hipMemcpyAsync(ctx->d_scratch.p, scratch.data(), scratch_bytes,
+        HIP_CHECK(hipMemcpyAsync(ctx->d_scratch.p, scratch, scratch_bytes,
                                  hipMemcpyHostToDevice, ctx->stream));
         HIP_CHECK(hipMemsetAsync(ctx->d_results.p, 0, r_agg, ctx->stream));
         if (do_aggs && ap.out_bytes)
@@ -1902,26 +1927,28 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             else
                 hipLaunchKernelGGL(k_cand_hist, dim3(hgrid), dim3(256), 0,
                                    ctx->stream, d_cand0, d_n0, 0ull, 0u, d_hist0);
-            HIP_CHECK(hipMemcpyAsync(hist0.data(), d_hist0, TOPK_BINS * 4,
+            HIP_CHECK(hipMemcpyAsync(hist0, d_hist0, TOPK_BINS * 4,
                                      hipMemcpyDeviceToHost, ctx->stream));
             hist0_valid = true;
         }
 
-        // ---- download counts
-        std::vector<uint32_t> tile_counts(n_tiles);
-        HIP_CHECK(hipMemcpyAsync(tile_counts.data(), ctx->d_results.p + r_tile_counts,
+        // ---- download counts (into pinned staging: no pageable D2H)
+        ctx->h_counts.ensure(size_t(n_tiles) * 4 + 64);
+        uint32_t* tile_counts = (uint32_t*)ctx->h_counts.p;
+        uint32_t* pc = (uint32_t*)(ctx->h_counts.p + size_t(n_tiles) * 4);
+        HIP_CHECK(hipMemcpyAsync(tile_counts, ctx->d_results.p + r_tile_counts,
                                  size_t(n_tiles) * 4, hipMemcpyDeviceToHost,
                                  ctx->stream));
-        uint32_t cand_n = 0;
         if (collect)
-            HIP_CHECK(hipMemcpyAsync(&cand_n, ctx->d_results.p + r_cand_count, 4,
+            HIP_CHECK(hipMemcpyAsync(pc, ctx->d_results.p + r_cand_count, 4,
                                      hipMemcpyDeviceToHost, ctx->stream));
         HIP_CHECK(hipStreamSynchronize(ctx->stream));
+        uint32_t cand_n = collect ? *pc : 0;
         mark("main+pass0+sync");
         float ms = 0;
         HIP_CHECK(hipEventElapsedTime(&ms, ctx->ev_start, ctx->ev_stop));
         record_kernel_time(ctx, kname, ms);
-        for (uint32_t c : tile_counts) matched += c;
+        for (uint32_t i = 0; i < n_tiles; ++i) matched += tile_counts[i];
 
         // ---- top-K selection over candidates (device histogram refinement
         // + host exact sort of the survivors; top_k_collector.rs semantics)
@@ -2008,7 +2035,8 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     break;
                 }
                 if (prefix_bits == 0 && hist0_valid) {
-                    hist = hist0;  // prefetched with the main kernel's sync
+                    memcpy(hist.data(), hist0,
+                           TOPK_BINS * 4);  // prefetched with the main sync
                 } else {
                     HIP_CHECK(hipMemsetAsync(d_hist, 0, TOPK_BINS * 4, ctx->stream));
                     // grid small enough that the per-WG 4096-bin LDS flush
@@ -2061,11 +2089,13 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                                    (uint64_t*)ctx->d_survivors.p, d_scount,
                                    uint32_t(survivors));
             top_keys.resize(survivors * rw);
-            HIP_CHECK(hipMemcpyAsync(top_keys.data(), ctx->d_survivors.p,
+            ctx->h_topk.ensure(survivors * 8 * rw + 16);
+            HIP_CHECK(hipMemcpyAsync(ctx->h_topk.p, ctx->d_survivors.p,
                                      survivors * 8 * rw, hipMemcpyDeviceToHost,
                                      ctx->stream));
             HIP_CHECK(hipEventRecord(ctx->ev_stop, ctx->stream));
             HIP_CHECK(hipStreamSynchronize(ctx->stream));
+            memcpy(top_keys.data(), ctx->h_topk.p, survivors * 8 * rw);
             float tms = 0;
             HIP_CHECK(hipEventElapsedTime(&tms, ctx->ev_start, ctx->ev_stop));
             record_kernel_time(ctx, "topk_select", tms);
