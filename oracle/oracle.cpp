@@ -361,6 +361,9 @@ struct SplitSearcher {
         Match m;
         const FastFieldView* f = sv.fast_field(n.field);
         if (!f) return m;
+        if (f->type == FastFieldView::MIXED)
+            throw std::runtime_error(
+                "range/term over a mixed-type dynamic column (r2 limit)");
         for (uint32_t d = 0; d < sv.num_docs; ++d)
             if (range_test(*f, d, n.lo, n.hi)) m.docs.push_back(d);
         return m;
@@ -500,7 +503,26 @@ static pb::SortByValue sort_value_of(const SortSpec& s, const SplitView& sv, uin
         case SortSpec::FAST_FIELD: {
             const FastFieldView* f = s.ff;
             if (!f || !f->present(doc)) break;
-            if (f->type == FastFieldView::U64) {
+            if (f->type == FastFieldView::MIXED) {
+                uint64_t r = f->mixed_raw(doc);
+                switch (f->mixed_tag(doc)) {
+                    case 1:
+                        v.kind = pb::SortByValue::I64;
+                        v.i64 = int64_t(r);
+                        break;
+                    case 2:
+                        v.kind = pb::SortByValue::F64;
+                        memcpy(&v.f64, &r, 8);
+                        break;
+                    case 3:
+                        v.kind = pb::SortByValue::BOOL;
+                        v.boolean = r != 0;
+                        break;
+                    default:
+                        v.kind = pb::SortByValue::U64;
+                        v.u64 = r;
+                }
+            } else if (f->type == FastFieldView::U64) {
                 v.kind = pb::SortByValue::U64;
                 v.u64 = f->u64(doc);
             } else if (f->type == FastFieldView::DATETIME) {
@@ -959,6 +981,7 @@ static SplitResult search_split(const SplitView& sv, const pb::SearchRequest& re
                     case FastFieldView::I64: return SortFieldKind::I64;
                     case FastFieldView::DATETIME: return SortFieldKind::DATETIME;
                     case FastFieldView::F64: return SortFieldKind::F64;
+                    case FastFieldView::MIXED: return SortFieldKind::MIXED;
                     default: return SortFieldKind::STR;
                 }
             };
@@ -971,12 +994,18 @@ static SplitResult search_split(const SplitView& sv, const pb::SearchRequest& re
                                : convert_cursor_key(c.sort_value2,
                                                     kind_of(specs[1]), order2);
             if (!k1.disabled)
-                cand.erase(std::remove_if(cand.begin(), cand.end(),
-                                          [&](const pb::PartialHit& h) {
-                                              return !after_cursor(h, c, k1, k2,
-                                                                   order1, order2);
-                                          }),
-                           cand.end());
+                cand.erase(
+                    std::remove_if(
+                        cand.begin(), cand.end(),
+                        [&](const pb::PartialHit& h) {
+                            return !after_cursor(
+                                h, c, k1, k2, order1, order2,
+                                !specs.empty() && kind_of(specs[0]) ==
+                                                      SortFieldKind::MIXED,
+                                specs.size() > 1 && kind_of(specs[1]) ==
+                                                        SortFieldKind::MIXED);
+                        }),
+                    cand.end());
         }
         auto cmp = [&](const pb::PartialHit& a, const pb::PartialHit& b) {
             return hit_before(a, b, order1, order2);

@@ -455,6 +455,10 @@ static void add_range_pred(FlatQuery& fq, const SplitView& sv, const PlanNode& n
         fq.match_none = true;
         return;
     }
+    if (f->type == FastFieldView::MIXED && n.kind != PlanNode::FIELD_PRESENCE)
+        throw std::runtime_error(
+            "range/term over a mixed-type dynamic column (r2 limit: sorting "
+            "and search_after are supported; typed predicates are not)");
     p.type = f->type == FastFieldView::U64   ? PRED_RANGE_U64
              : f->type == FastFieldView::F64 ? PRED_RANGE_F64
                                              : PRED_RANGE_I64;
@@ -1199,7 +1203,28 @@ static pb::SortByValue sort_value_of(const SortSpec& s, uint32_t doc, float scor
         case SortSpec::FAST_FIELD: {
             const FastFieldView* f = s.ff;
             if (!f || !f->present(doc)) break;
-            if (f->type == FastFieldView::U64) {
+            if (f->type == FastFieldView::MIXED) {
+                // echo the ORIGINAL typed value (true stays a bool, a u64
+                // beyond 2^63 stays exact) — tags/raw sections
+                uint64_t r = f->mixed_raw(doc);
+                switch (f->mixed_tag(doc)) {
+                    case 1:
+                        v.kind = pb::SortByValue::I64;
+                        v.i64 = int64_t(r);
+                        break;
+                    case 2:
+                        v.kind = pb::SortByValue::F64;
+                        memcpy(&v.f64, &r, 8);
+                        break;
+                    case 3:
+                        v.kind = pb::SortByValue::BOOL;
+                        v.boolean = r != 0;
+                        break;
+                    default:
+                        v.kind = pb::SortByValue::U64;
+                        v.u64 = r;
+                }
+            } else if (f->type == FastFieldView::U64) {
                 v.kind = pb::SortByValue::U64;
                 v.u64 = f->u64(doc);
             } else if (f->type == FastFieldView::DATETIME) {
@@ -1649,6 +1674,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             case FastFieldView::I64: return SortFieldKind::I64;
             case FastFieldView::DATETIME: return SortFieldKind::DATETIME;
             case FastFieldView::F64: return SortFieldKind::F64;
+            case FastFieldView::MIXED: return SortFieldKind::MIXED;
             default: return SortFieldKind::STR;
         }
     };
@@ -1834,9 +1860,13 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     q.sort_src = 2;
                     q.sort_width = uint32_t(f->ord_width);
                 } else {
-                    q.sort_src = f->type == FastFieldView::U64   ? 2
-                                 : f->type == FastFieldView::F64 ? 4
-                                                                 : 3;
+                    // MIXED columns store f64-monotonic u64 keys: ascending
+                    // u64 order IS the numeric order -> device treats them
+                    // like a u64 column
+                    q.sort_src = f->type == FastFieldView::U64    ? 2
+                                 : f->type == FastFieldView::MIXED ? 2
+                                 : f->type == FastFieldView::F64  ? 4
+                                                                  : 3;
                     q.sort_width = 8;
                 }
             } else {
@@ -2180,8 +2210,14 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     out.hits.erase(
                         std::remove_if(out.hits.begin(), out.hits.end(),
                                        [&](const pb::PartialHit& h) {
-                                           return !after_cursor(h, c, ck1, ck2,
-                                                                order1, order2);
+                                           return !after_cursor(
+                                               h, c, ck1, ck2, order1, order2,
+                                               !specs.empty() &&
+                                                   kind_of(specs[0]) ==
+                                                       SortFieldKind::MIXED,
+                                               specs.size() > 1 &&
+                                                   kind_of(specs[1]) ==
+                                                       SortFieldKind::MIXED);
                                        }),
                         out.hits.end());
                 }

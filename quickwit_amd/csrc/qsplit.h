@@ -68,7 +68,11 @@ struct TextFieldView {
 
 struct FastFieldView {
     std::string name;
-    enum Type { U64, I64, DATETIME, STR, F64 } type = U64;
+    // MIXED: a dynamic field whose docs carry several value types.
+    // `values` holds f64-monotonic u64 sort keys (bool as 0/1); `tags`
+    // (u8: 0=u64 1=i64 2=f64 3=bool) + `raw` (u64 bits) reproduce the
+    // original typed value per doc.
+    enum Type { U64, I64, DATETIME, STR, F64, MIXED } type = U64;
     bool nullable = false;
     bool multi = false;        // multi-valued str: per-doc ord lists
     bool lower_norm = false;   // str column written through a lowercase
@@ -78,6 +82,9 @@ struct FastFieldView {
     int64_t min_value = 0, max_value = 0;
     double fmin = 0, fmax = 0;  // F64
     Section values, nulls, dict_offsets, dict_bytes, value_offsets;
+    Section tags, raw;  // MIXED only
+    const uint8_t* h_tags = nullptr;
+    const uint64_t* h_raw = nullptr;
     const uint32_t* h_val_offsets = nullptr;  // multi: [num_docs+1] prefix
     const void* h_values = nullptr;
     const uint64_t* h_nulls = nullptr;  // bit d set = doc d has a value
@@ -107,6 +114,8 @@ struct FastFieldView {
         }
     }
     int64_t i64(uint32_t doc) const { return ((const int64_t*)h_values)[doc]; }
+    uint8_t mixed_tag(uint32_t doc) const { return h_tags[doc]; }
+    uint64_t mixed_raw(uint32_t doc) const { return h_raw[doc]; }
     uint64_t u64(uint32_t doc) const { return ((const uint64_t*)h_values)[doc]; }
     double f64(uint32_t doc) const { return ((const double*)h_values)[doc]; }
     std::string dict_entry(uint64_t o) const {
@@ -235,6 +244,7 @@ struct SplitView {
                           : ty == "i64" ? FastFieldView::I64
                           : ty == "datetime" ? FastFieldView::DATETIME
                           : ty == "f64" ? FastFieldView::F64
+                          : ty == "mixed" ? FastFieldView::MIXED
                                         : FastFieldView::STR;
                 ff.nullable = f->at("nullable")->b;
                 ff.values = sec(s, "values");
@@ -255,6 +265,11 @@ struct SplitView {
                     ff.dict_bytes = sec(s, "dict_bytes");
                     ff.h_dict_offsets = (const uint32_t*)(data + ff.dict_offsets.off);
                     ff.h_dict_bytes = data + ff.dict_bytes.off;
+                } else if (ff.type == FastFieldView::MIXED) {
+                    ff.tags = sec(s, "tags");
+                    ff.raw = sec(s, "raw");
+                    ff.h_tags = data + ff.tags.off;
+                    ff.h_raw = (const uint64_t*)(data + ff.raw.off);
                 } else if (ff.type == FastFieldView::F64) {
                     const mj::Value* mn = f->get("min_value");
                     const mj::Value* mx = f->get("max_value");

@@ -60,12 +60,50 @@ inline int cmp_key(const SortKey& a, const SortKey& b, int order) {
     return a_first ? -1 : 1;
 }
 
+// Cross-type numeric comparison for MIXED dynamic columns: hits from one
+// field may carry different SortByValue kinds (u64 beyond i64, i64, f64,
+// bool) — the reference's SortValue ordering compares these numerically
+// (bool as 0/1). Same-kind pairs keep the exact u64-mapped compare.
+inline int cmp_value(const pb::SortByValue& a, const pb::SortByValue& b,
+                     int order) {
+    if (a.kind == b.kind || a.kind == pb::SortByValue::NONE ||
+        b.kind == pb::SortByValue::NONE)
+        return cmp_key(sort_key_of(a), sort_key_of(b), order);
+    auto sign = [](const pb::SortByValue& v) -> int {
+        switch (v.kind) {
+            case pb::SortByValue::U64: return v.u64 > 0 ? 1 : 0;
+            case pb::SortByValue::I64: return v.i64 > 0 ? 1 : (v.i64 < 0 ? -1 : 0);
+            case pb::SortByValue::F64: return v.f64 > 0 ? 1 : (v.f64 < 0 ? -1 : 0);
+            default: return v.boolean ? 1 : 0;
+        }
+    };
+    int sa = sign(a), sb = sign(b);
+    int c;
+    if (sa != sb) c = sa < sb ? -1 : 1;
+    else {
+        // same sign, different kinds: exact for int-vs-int, via long double
+        // against floats (declared approximation beyond 2^53 — DESIGN §7)
+        auto mag = [](const pb::SortByValue& v) -> long double {
+            switch (v.kind) {
+                case pb::SortByValue::U64: return (long double)v.u64;
+                case pb::SortByValue::I64: return (long double)v.i64;
+                case pb::SortByValue::F64: return (long double)v.f64;
+                default: return v.boolean ? 1.0L : 0.0L;
+            }
+        };
+        long double ma = mag(a), mb = mag(b);
+        c = ma == mb ? 0 : (ma < mb ? -1 : 1);
+    }
+    if (c == 0) return 0;
+    return order == 1 ? -c : c;
+}
+
 // Full PartialHit comparator for leaf top-K and cross-split merge.
 inline bool hit_before(const pb::PartialHit& a, const pb::PartialHit& b, int order1,
                        int order2) {
-    int c = cmp_key(sort_key_of(a.sort_value), sort_key_of(b.sort_value), order1);
+    int c = cmp_value(a.sort_value, b.sort_value, order1);
     if (c) return c < 0;
-    c = cmp_key(sort_key_of(a.sort_value2), sort_key_of(b.sort_value2), order2);
+    c = cmp_value(a.sort_value2, b.sort_value2, order2);
     if (c) return c < 0;
     // GlobalDocId tie-break in the first field's order (sorting.md:14-17)
     int sc = a.split_id.compare(b.split_id);
@@ -77,8 +115,22 @@ inline bool hit_before(const pb::PartialHit& a, const pb::PartialHit& b, int ord
     return false;
 }
 
-// field type tags for cursor conversion (SortFieldType analog)
-enum class SortFieldKind { NONE, SCORE, U64, I64, DATETIME, F64, STR };
+// field type tags for cursor conversion (SortFieldType analog).
+// MIXED = a dynamic field whose docs carry several value types (u64/i64/
+// f64/bool): device keys and cursor conversion use the f64-monotonic map
+// over the numeric value (bool as 0/1); response values stay typed.
+enum class SortFieldKind { NONE, SCORE, U64, I64, DATETIME, F64, STR, MIXED };
+
+// typed SortByValue -> the MIXED column's f64-sortable key domain
+inline SortKey mixed_key_of(const pb::SortByValue& v) {
+    switch (v.kind) {
+        case pb::SortByValue::NONE: return {};
+        case pb::SortByValue::U64: return {true, f64_to_u64(double(v.u64))};
+        case pb::SortByValue::I64: return {true, f64_to_u64(double(v.i64))};
+        case pb::SortByValue::F64: return {true, f64_to_u64(v.f64)};
+        default: return {true, f64_to_u64(v.boolean ? 1.0 : 0.0)};
+    }
+}
 
 // converted search_after cursor value, in the target field's u64-mapped
 // domain (SearchAfterSegment::new + convert_to_u64_ff_val,
@@ -156,6 +208,7 @@ inline CursorKey convert_cursor_key(const pb::SortByValue& v, SortFieldKind ft,
             }
             return out;
         case SortFieldKind::F64:
+        case SortFieldKind::MIXED:
             switch (v.kind) {
                 case pb::SortByValue::F64: some(f64_to_u64(v.f64)); break;
                 case pb::SortByValue::U64: some(f64_to_u64(double(v.u64))); break;
@@ -174,7 +227,7 @@ inline CursorKey convert_cursor_key(const pb::SortByValue& v, SortFieldKind ft,
 // SortOrder::compare (Desc: natural, Asc: reversed; Some > None).
 inline bool after_cursor(const pb::PartialHit& h, const pb::PartialHit& c,
                          const CursorKey& k1, const CursorKey& k2, int order1,
-                         int order2) {
+                         int order2, bool mixed1 = false, bool mixed2 = false) {
     if (k1.disabled) return true;  // cursor before all values: filter off
     auto cmpo = [](const SortKey& a, const SortKey& b, int order) -> int {
         if (a.has && b.has) {
@@ -186,9 +239,13 @@ inline bool after_cursor(const pb::PartialHit& h, const pb::PartialHit& c,
         if (b.has) return -1;  // (None, Some) -> Less
         return 0;
     };
-    int r = cmpo(sort_key_of(h.sort_value), k1.key, order1);
+    // MIXED sort fields: compare in the f64-key domain on both sides (the
+    // cursor was converted there; hit values stay typed in the response)
+    int r = cmpo(mixed1 ? mixed_key_of(h.sort_value) : sort_key_of(h.sort_value),
+                 k1.key, order1);
     if (r) return r < 0;
-    r = cmpo(sort_key_of(h.sort_value2), k2.disabled ? SortKey{} : k2.key, order2);
+    r = cmpo(mixed2 ? mixed_key_of(h.sort_value2) : sort_key_of(h.sort_value2),
+             k2.disabled ? SortKey{} : k2.key, order2);
     if (r) return r < 0;
     if (c.split_id.empty()) return false;  // equal values, no doc tiebreak
     int cs = h.split_id.compare(c.split_id);

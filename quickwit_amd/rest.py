@@ -231,9 +231,12 @@ def _infer_dynamic_fields(explicit_names, doc_batches):
                     continue
                 kinds = seen.setdefault(k, set())
                 if isinstance(v, bool):
-                    kinds.add("skip")
+                    kinds.add("bool")
                 elif isinstance(v, int):
-                    kinds.add("int" if v >= 0 else "negint")
+                    if v > 0x7FFFFFFFFFFFFFFF:
+                        kinds.add("bigint")  # beyond i64: u64-only territory
+                    else:
+                        kinds.add("int" if v >= 0 else "negint")
                 elif isinstance(v, float):
                     kinds.add("float")
                 elif isinstance(v, str):
@@ -247,12 +250,17 @@ def _infer_dynamic_fields(explicit_names, doc_batches):
         kinds = seen[name]
         if "skip" in kinds:
             continue
-        if kinds <= {"int"}:
+        if kinds <= {"int", "bigint"}:
             fields.append({"name": name, "type": "u64", "fast": True})
         elif kinds <= {"int", "negint"}:
             fields.append({"name": name, "type": "i64", "fast": True})
         elif kinds <= {"int", "negint", "float"}:
             fields.append({"name": name, "type": "f64", "fast": True})
+        elif kinds <= {"int", "negint", "float", "bool", "bigint"}:
+            # one dynamic field, several typed columns in the reference
+            # (u64 beyond i64 + negatives/floats/bools): our MIXED column
+            # (typed values + f64-monotonic sort keys)
+            fields.append({"name": name, "type": "mixed", "fast": True})
         elif kinds == {"str"}:
             # quickwit dynamic strings: tokenized text index + raw str fast
             # column under the same name (dynamic_mapping tokenizer+fast)

@@ -298,6 +298,44 @@ def _build_fast_field(sec, fspec, num_docs, values, present):
         meta["sec"]["values"] = sec.add(ords)
         meta["sec"]["dict_offsets"] = sec.add(dict_offsets)
         meta["sec"]["dict_bytes"] = sec.add(np.frombuffer(dict_bytes, dtype=np.uint8))
+    elif ftype == "mixed":
+        # dynamic field with several value types per corpus (u64/i64/f64/
+        # bool): sections = f64-monotonic u64 sort keys (bool as 0/1) +
+        # per-doc type tag + raw typed bits. Order across types is numeric
+        # (the reference's SortValue ordering); exactness beyond 2^53 is a
+        # declared approximation.
+        import struct as _st
+
+        def _f64_sortable(d):
+            b = _st.unpack("<Q", _st.pack("<d", d))[0]
+            return (~b) & 0xFFFFFFFFFFFFFFFF if b >> 63 else b | (1 << 63)
+
+        keys = np.zeros(num_docs, dtype=np.uint64)
+        tags = np.zeros(num_docs, dtype=np.uint8)
+        raw = np.zeros(num_docs, dtype=np.uint64)
+        for d, (v, p) in enumerate(zip(values, present)):
+            if not p:
+                continue
+            if isinstance(v, bool):
+                tags[d] = 3
+                raw[d] = int(v)
+                num = 1.0 if v else 0.0
+            elif isinstance(v, int):
+                if v < 0:
+                    tags[d] = 1
+                    raw[d] = v & 0xFFFFFFFFFFFFFFFF
+                else:
+                    tags[d] = 0
+                    raw[d] = v
+                num = float(v)
+            else:
+                tags[d] = 2
+                raw[d] = _st.unpack("<Q", _st.pack("<d", float(v)))[0]
+                num = float(v)
+            keys[d] = _f64_sortable(num)
+        meta["sec"]["values"] = sec.add(keys)
+        meta["sec"]["tags"] = sec.add(tags)
+        meta["sec"]["raw"] = sec.add(raw)
     else:
         dt = (np.uint64 if ftype == "u64"
               else np.float64 if ftype == "f64" else np.int64)

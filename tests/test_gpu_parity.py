@@ -1613,3 +1613,51 @@ def test_multi_segment_absence_cache_all_segments_rule():
     r2 = gpu.leaf_search(make_leaf_request(q2, schema, [("ms", 4)], max_hits=5))
     assert r2.get("num_hits", 0) == 0
     assert gpu.absence_cache_stats()[0] >= 1  # second probe hits
+
+
+# ---------------------------------------------------- mixed-type columns
+def test_mixed_type_column_sort_parity():
+    """Mixed-type dynamic column (u64-beyond-i64 + i64 + f64 + bool):
+    numeric sort order across types, typed echo in sort_value, search_after
+    band through the f64-key domain — GPU vs oracle."""
+    schema = {"timestamp_field": None, "fields": [
+        {"name": "mx", "type": "mixed", "fast": True},
+        {"name": "body", "type": "text", "tokenizer": "default",
+         "record": "freq", "fieldnorms": True}]}
+    w = splitgen.SplitWriter(schema, "mx-split", store_docs=False)
+    w.add_documents([
+        {"mx": 18000000000000000000, "body": "a"},
+        {"mx": 0, "body": "a"},
+        {"mx": True, "body": "a"},
+        {"mx": 10.5, "body": "a"},
+        {"mx": -10, "body": "a"},
+        {"body": "a"},  # missing -> None, sorts last
+    ])
+    data = w.finalize()
+    gpu, cpu = GpuSearcher(device=0), OracleSearcher()
+    gpu.add_split("mx-split", data)
+    cpu.add_split("mx-split", data)
+    for order in (0, 1):
+        req = make_leaf_request({"type": "match_all"}, schema,
+                                [("mx-split", 6)], max_hits=10,
+                                sort_fields=[{"field_name": "mx",
+                                              "sort_order": order}])
+        g, e = gpu.leaf_search(req), cpu.leaf_search(req)
+        assert g["num_hits"] == e["num_hits"] == 6
+        assert g["partial_hits"] == e["partial_hits"], (order, g, e)
+        vals = [h.get("sort_value") for h in g["partial_hits"]]
+        if order == 0:  # asc: -10, 0, true, 10.5, 18e18, None
+            assert vals == [{"i64": -10}, {"u64": 0}, {"boolean": True},
+                            {"f64": 10.5}, {"u64": 18000000000000000000},
+                            None]
+    # search_after from a negative i64 cursor, ascending
+    req = make_leaf_request({"type": "match_all"}, schema, [("mx-split", 6)],
+                            max_hits=10,
+                            sort_fields=[{"field_name": "mx",
+                                          "sort_order": 0}])
+    req["search_request"]["search_after"] = {"sort_value": {"i64": -10}}
+    g, e = gpu.leaf_search(req), cpu.leaf_search(req)
+    assert g["partial_hits"] == e["partial_hits"]
+    assert [h.get("sort_value") for h in g["partial_hits"]][:4] == [
+        {"u64": 0}, {"boolean": True}, {"f64": 10.5},
+        {"u64": 18000000000000000000}]

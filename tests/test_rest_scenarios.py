@@ -31,8 +31,7 @@ MULTI_VALUED_FIELDS = set()  # multi-valued str fast columns now supported
 NUMERIC_TERMS_FIELDS = set()  # terms over numeric fast columns supported
 
 
-MIXED_TYPE_SORT_FIELDS = {"mixed_type"}  # one dynamic field, several typed
-                                         # columns (u64+f64+bool): later round
+MIXED_TYPE_SORT_FIELDS = set()  # mixed-type dynamic columns supported (r2)
 
 
 def skip_step(i, step):
