@@ -33,8 +33,10 @@ struct TermDev {
     uint64_t norms_off;    // byte offset of the field's fieldnorms (0 = none)
     float weight;          // BM25 W = idf*(1+k1)*boost (0 when not scoring)
     uint32_t ktab_idx;     // index into the query's K tables (per field)
-    uint64_t ranges_off;   // byte offset into query scratch: u32 lo[ntiles],
-                           // u32 hi[ntiles] block-index ranges for this term
+    uint64_t ranges_addr;  // ABSOLUTE device VA: u32 lo[ntiles], u32
+                           // hi[ntiles] block-index ranges for this term
+                           // (cached per (split, term) — the ranges depend
+                           // only on the split's static block structure)
 };
 
 enum PredType : uint32_t {

@@ -125,7 +125,7 @@ __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t t
                                  uint32_t* bitset, bool scoring,
                                  const float* ktab_lds = nullptr,
                                  const uint8_t* norms_lds = nullptr) {
-    const uint32_t* ranges = (const uint32_t*)(q.scratch + t.ranges_off);
+    const uint32_t* ranges = (const uint32_t*)t.ranges_addr;
     uint32_t blo = ranges[tile], bhi = ranges[q.n_tiles + tile];
     if (blo >= bhi) return;
     const SkipEntryDev* __restrict__ skip = (const SkipEntryDev*)(q.split + t.skip_off);

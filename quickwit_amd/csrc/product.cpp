@@ -67,6 +67,17 @@ struct DeviceSplit {
     // parent allocation (owns_image=false)
     std::vector<std::unique_ptr<DeviceSplit>> segs;
     bool owns_image = true;
+    // per-(field, term) device block-range tables (u32 lo[ntiles] +
+    // hi[ntiles]): they depend only on the split's static block structure,
+    // so the O(blocks+tiles) host walk (~180 us at 100M docs) and its
+    // 300 KB upload happen once per term, not per query
+    mutable std::map<std::pair<const void*, int64_t>, uint8_t*> range_cache;
+    mutable uint64_t range_cache_bytes = 0;
+    void free_range_cache() const {
+        for (auto& kv : range_cache) (void)hipFree(kv.second);
+        range_cache.clear();
+        range_cache_bytes = 0;
+    }
 };
 
 struct KernelTimer {
@@ -354,6 +365,45 @@ static void qw_ensure_acct(qw_ctx* ctx, qw::DevBuf& b, size_t n) {
     ctx->hbm_used -= b.cap;
     b.ensure(n);
     ctx->hbm_used += b.cap;
+}
+
+// device address of the (split, term) block-range table, computed and
+// uploaded on first use (DeviceSplit::range_cache)
+static uint64_t term_ranges_addr(qw_ctx* ctx, const qw::DeviceSplit& ds,
+                                 const qw::TextFieldView* f, int64_t tid,
+                                 uint32_t n_tiles, uint32_t num_docs) {
+    using namespace qw;
+    auto key = std::make_pair((const void*)f, tid);
+    auto it = ds.range_cache.find(key);
+    if (it != ds.range_cache.end()) return (uint64_t)it->second;
+    if (ds.range_cache.size() >= 1024) {  // wildcard-expansion backstop
+        ctx->hbm_used -= ds.range_cache_bytes;
+        ds.free_range_cache();
+    }
+    size_t bytes = 2ull * n_tiles * 4;
+    qw_check_budget(ctx, bytes);
+    std::vector<uint32_t> r(2ull * n_tiles);
+    const SkipEntry* sk = f->h_skip + f->h_skip_off[tid] / 16;
+    uint32_t nb = f->h_n_blocks[tid];
+    uint32_t* lo = r.data();
+    uint32_t* hi = lo + n_tiles;
+    uint32_t b_lo = 0, b_hi = 0;
+    for (uint32_t tile = 0; tile < n_tiles; ++tile) {
+        uint32_t tlo = tile * TILE_DOCS;
+        uint32_t thi = std::min<uint32_t>(tlo + TILE_DOCS, num_docs);
+        while (b_lo < nb && sk[b_lo].last_doc < tlo) ++b_lo;
+        if (b_hi < b_lo) b_hi = b_lo;
+        while (b_hi < nb && sk[b_hi].first_doc < thi) ++b_hi;
+        lo[tile] = b_lo;
+        hi[tile] = b_hi;
+    }
+    uint8_t* d = nullptr;
+    HIP_CHECK(hipMalloc(&d, bytes));
+    HIP_CHECK(hipMemcpy(d, r.data(), bytes, hipMemcpyHostToDevice));
+    ctx->hbm_used += bytes;
+    ds.range_cache.emplace(key, d);
+    ds.range_cache_bytes += bytes;
+    return (uint64_t)d;
 }
 
 static void ctx_ensure_device(qw_ctx* ctx) {
@@ -1408,33 +1458,16 @@ static uint8_t* resolve_hitset(qw_ctx* ctx, const DeviceSplit& ds,
     size_t off_terms = 0;
     size_t off_preds = off_terms + terms.size() * sizeof(TermDev);
     size_t off_ktabs = off_preds + fq.preds.size() * sizeof(PredDev);
-    size_t off_ranges = (off_ktabs + 256 * 4 + 63) & ~size_t(63);
-    size_t ranges_bytes = fq.terms.size() * 2ull * n_tiles * 4;
-    std::vector<uint32_t> ranges(ranges_bytes / 4, 0);
+    size_t scratch_bytes = off_ktabs + 256 * 4;
     for (size_t i = 0; i < fq.terms.size(); ++i) {
         const FlatQuery::FTerm& t = fq.terms[i];
-        const SkipEntry* sk = t.f->h_skip + t.f->h_skip_off[t.tid] / 16;
-        uint32_t nb = t.f->h_n_blocks[t.tid];
-        uint32_t* lo = ranges.data() + (i * 2ull * n_tiles);
-        uint32_t* hi = lo + n_tiles;
-        uint32_t b_lo = 0, b_hi = 0;
-        for (uint32_t tile = 0; tile < n_tiles; ++tile) {
-            uint32_t tlo = tile * TILE_DOCS;
-            uint32_t thi = std::min<uint32_t>(tlo + TILE_DOCS, sv.num_docs);
-            while (b_lo < nb && sk[b_lo].last_doc < tlo) ++b_lo;
-            if (b_hi < b_lo) b_hi = b_lo;
-            while (b_hi < nb && sk[b_hi].first_doc < thi) ++b_hi;
-            lo[tile] = b_lo;
-            hi[tile] = b_hi;
-        }
-        terms[i].ranges_off = off_ranges + i * 2ull * n_tiles * 4;
+        terms[i].ranges_addr =
+            term_ranges_addr(ctx, ds, t.f, t.tid, n_tiles, sv.num_docs);
     }
-    size_t scratch_bytes = off_ranges + ranges_bytes;
     std::vector<uint8_t> scratch(scratch_bytes, 0);
     memcpy(scratch.data() + off_terms, terms.data(), terms.size() * sizeof(TermDev));
     memcpy(scratch.data() + off_preds, fq.preds.data(),
            fq.preds.size() * sizeof(PredDev));
-    memcpy(scratch.data() + off_ranges, ranges.data(), ranges_bytes);
     qw_ensure_acct(ctx, ctx->d_scratch, scratch_bytes);
     qw_ensure_acct(ctx, ctx->d_results, 64);
     HIP_CHECK(hipMemcpyAsync(ctx->d_scratch.p, scratch.data(), scratch_bytes,
@@ -1696,8 +1729,6 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
     // ---- scratch assembly (descriptors + ktabs + block ranges), one H2D
     std::vector<TermDev> terms(fq.terms.size());
     uint32_t n_must = 0, n_must_not = 0;
-    size_t ranges_bytes = 0;
-    for (size_t i = 0; i < fq.terms.size(); ++i) ranges_bytes += 2ull * n_tiles * 4;
     std::vector<float> ktabs(fq.ktab_fields.size() * 0);  // filled below
 
     // K tables per field actually referenced by scored terms
@@ -1757,9 +1788,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
     size_t results_bytes = r_cand + cand_cap * cand_rec;
 
     size_t off_ktabs = off_aggs + ap.devs.size() * sizeof(AggDev);
-    size_t off_ranges = off_ktabs + ktabs.size() * 4;
-    off_ranges = (off_ranges + 63) & ~size_t(63);
-    size_t scratch_bytes = off_ranges + ranges_bytes;
+    size_t scratch_bytes = off_ktabs + ktabs.size() * 4;
     // percentiles boundary tables (gamma^k doubles the kernel searches)
     scratch_bytes = (scratch_bytes + 7) & ~size_t(7);
     for (size_t i = 0; i < ap.devs.size(); ++i)
@@ -1769,29 +1798,14 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         }
 
     mark("plan");
-    // block ranges per term per tile (host two-pointer walk over skip
-    // entries), written straight into the PINNED staging buffer — no
-    // per-call zero-init/copy of the ~300KB scratch at 100M docs
+    // per-term block ranges come from the split-level device cache
+    // (term_ranges_addr): first query pays the walk+upload, repeats don't
     ctx->h_scratch.ensure(scratch_bytes);
     uint8_t* scratch = ctx->h_scratch.p;
     for (size_t i = 0; i < fq.terms.size(); ++i) {
         const FlatQuery::FTerm& t = fq.terms[i];
-        const SkipEntry* sk = t.f->h_skip + t.f->h_skip_off[t.tid] / 16;
-        uint32_t nb = t.f->h_n_blocks[t.tid];
-        uint32_t* lo =
-            (uint32_t*)(scratch + off_ranges) + (i * 2ull * n_tiles);
-        uint32_t* hi = lo + n_tiles;
-        uint32_t b_lo = 0, b_hi = 0;
-        for (uint32_t tile = 0; tile < n_tiles; ++tile) {
-            uint32_t tlo = tile * TILE_DOCS;
-            uint32_t thi = std::min<uint32_t>(tlo + TILE_DOCS, sv.num_docs);
-            while (b_lo < nb && sk[b_lo].last_doc < tlo) ++b_lo;
-            if (b_hi < b_lo) b_hi = b_lo;
-            while (b_hi < nb && sk[b_hi].first_doc < thi) ++b_hi;
-            lo[tile] = b_lo;
-            hi[tile] = b_hi;
-        }
-        terms[i].ranges_off = off_ranges + i * 2ull * n_tiles * 4;
+        terms[i].ranges_addr =
+            term_ranges_addr(ctx, ds, t.f, t.tid, n_tiles, sv.num_docs);
     }
     mark("ranges");
 
@@ -2660,8 +2674,11 @@ void qw_ctx_free(qw_ctx* ctx) {
     if (!ctx) return;
     if (ctx->device_ready) {
         (void)hipSetDevice(ctx->device);
-        for (auto& kv : ctx->splits)
+        for (auto& kv : ctx->splits) {
+            kv.second->free_range_cache();
+            for (auto& sg : kv.second->segs) sg->free_range_cache();
             if (kv.second->d_image) (void)hipFree(kv.second->d_image);
+        }
         ctx->hitsets.clear();
         if (ctx->ev_start) (void)hipEventDestroy(ctx->ev_start);
         if (ctx->ev_stop) (void)hipEventDestroy(ctx->ev_stop);
@@ -2766,6 +2783,12 @@ int32_t qw_ctx_add_split(qw_ctx* ctx, const char* split_id, const uint8_t* data,
 int32_t qw_ctx_remove_split(qw_ctx* ctx, const char* split_id) {
     auto it = ctx->splits.find(split_id);
     if (it == ctx->splits.end()) return QW_ERR_NOT_FOUND;
+    ctx->hbm_used -= it->second->range_cache_bytes;
+    it->second->free_range_cache();
+    for (auto& sg : it->second->segs) {
+        ctx->hbm_used -= sg->range_cache_bytes;
+        sg->free_range_cache();
+    }
     if (it->second->d_image) {
         (void)hipFree(it->second->d_image);
         ctx->hbm_used -= it->second->len + 64;
