@@ -39,6 +39,7 @@
 #include <cstdint>
 #include <cstdlib>
 #include <cstring>
+#include <functional>
 #include <map>
 #include <memory>
 #include <stdexcept>
@@ -236,8 +237,79 @@ struct SplitSearcher {
                 // semantics-transparent (cache_node.rs); memoization is a
                 // product-side device-bitmap concern
                 return eval(n.cache_inner.at(0));
+            case PlanNode::PHRASE:
+                return eval_phrase(n);
         }
         return Match{};
+    }
+
+    // multi-token phrase, slop 0 (PhraseQuery semantics restated from
+    // full_text_query.rs:113-137's phrase mode): a doc matches iff some
+    // position p has token i at p+i for all i. Needs record: position.
+    // Unscored (const-score like wildcard/term_set — the product rejects
+    // phrases under _score sorting the same way).
+    Match eval_phrase(const PlanNode& n) const {
+        Match m;
+        const TextFieldView* f = sv.text_field(n.field);
+        if (!f) return m;
+        if (!f->has_positions)
+            throw std::runtime_error(
+                "phrase query needs record: position on field " + n.field);
+        size_t k = n.phrase_toks.size();
+        struct TP {
+            Postings p;
+            std::vector<uint32_t> pos_off;  // per posting into h_positions
+        };
+        std::vector<TP> tps(k);
+        for (size_t i = 0; i < k; ++i) {
+            const std::string& tok = n.phrase_toks[i];
+            int64_t tid = f->find_term(tok.data(), tok.size());
+            if (tid < 0) return m;  // absent token: phrase matches nothing
+            tps[i].p = decode_term(*f, tid);
+            uint32_t df = f->h_doc_freq[tid];
+            tps[i].pos_off.resize(df);
+            uint32_t first_blk = uint32_t(f->h_skip_off[tid] / 16);
+            uint32_t nblk = f->h_n_blocks[tid];
+            uint32_t out = 0;
+            for (uint32_t b = 0; b < nblk; ++b) {
+                uint32_t start = f->h_pos_start[first_blk + b];
+                uint32_t cnt = f->h_skip[first_blk + b].count;
+                for (uint32_t j = 0; j < cnt; ++j) {
+                    tps[i].pos_off[out] = start;
+                    start += tps[i].p.tfs[out];
+                    ++out;
+                }
+            }
+        }
+        // doc intersection driven by token 0, then the position chain
+        std::vector<size_t> idx(k, 0);
+        const uint32_t* P = f->h_positions;
+        for (size_t i0 = 0; i0 < tps[0].p.docs.size(); ++i0) {
+            uint32_t doc = tps[0].p.docs[i0];
+            bool all = true;
+            for (size_t i = 1; i < k && all; ++i) {
+                auto& d = tps[i].p.docs;
+                while (idx[i] < d.size() && d[idx[i]] < doc) ++idx[i];
+                all = idx[i] < d.size() && d[idx[i]] == doc;
+            }
+            if (!all) continue;
+            bool hit = false;
+            uint32_t o0 = tps[0].pos_off[i0], n0 = tps[0].p.tfs[i0];
+            for (uint32_t a = 0; a < n0 && !hit; ++a) {
+                uint32_t p0 = P[o0 + a];
+                bool chain = true;
+                for (size_t i = 1; i < k && chain; ++i) {
+                    uint32_t oi = tps[i].pos_off[idx[i]];
+                    uint32_t ni = tps[i].p.tfs[idx[i]];
+                    const uint32_t* lo2 = P + oi;
+                    const uint32_t* hi2 = lo2 + ni;
+                    chain = std::binary_search(lo2, hi2, p0 + uint32_t(i));
+                }
+                hit = chain;
+            }
+            if (hit) m.docs.push_back(doc);
+        }
+        return m;
     }
 
     // union of every dict term matching the glob (wildcard_query.rs ->
@@ -905,6 +977,18 @@ static SplitResult search_split(const SplitView& sv, const pb::SearchRequest& re
         plan = std::move(b);
     }
 
+    // phrases are const-score (like wildcard/term_set): reject them under
+    // _score sorting instead of mis-scoring — the product does the same
+    std::function<bool(const PlanNode&)> has_phrase =
+        [&](const PlanNode& pn) -> bool {
+        if (pn.kind == PlanNode::PHRASE) return true;
+        for (auto* v : {&pn.must, &pn.must_not, &pn.should, &pn.filter,
+                        &pn.cache_inner})
+            for (auto& c : *v)
+                if (has_phrase(c)) return true;
+        return false;
+    };
+
     // sort specs (max 2, like the reference — search.proto:269)
     std::vector<SortSpec> specs;
     for (auto& sf : req.sort_fields) {
@@ -919,6 +1003,9 @@ static SplitResult search_split(const SplitView& sv, const pb::SearchRequest& re
     }
     bool scoring = false;
     for (auto& s : specs) scoring |= s.comp == SortSpec::SCORE;
+    if (scoring && has_phrase(plan))
+        throw std::runtime_error(
+            "phrase under _score sorting (const-score semantics, r2 limit)");
 
     if (scoring && plan_has_const_score(plan))
         throw std::runtime_error(

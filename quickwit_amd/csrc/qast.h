@@ -110,7 +110,7 @@ struct Bound {
 struct PlanNode {
     enum Kind {
         MATCH_ALL, MATCH_NONE, TERM, BOOL, RANGE, FIELD_PRESENCE, WILDCARD,
-        CACHE
+        CACHE, PHRASE
     } kind = MATCH_ALL;
     // TERM / WILDCARD (value = glob pattern: '*' any run, '?' one char)
     std::string field;
@@ -129,6 +129,10 @@ struct PlanNode {
     // CACHE: the wrapped subtree (cache_node.rs CacheNode.inner); evaluated
     // via a device-resident HitSet bitmap when in filter position
     std::vector<PlanNode> cache_inner;  // size <= 1
+    // PHRASE: consecutive tokens (slop 0 — full_text_query.rs:113-137 phrase
+    // mode; needs record: position). Matching is unscored (const-score
+    // semantics like term_set/wildcard).
+    std::vector<std::string> phrase_toks;
 };
 
 // stable fingerprint of a plan subtree — the (split, subquery) key of the
@@ -140,6 +144,10 @@ inline void plan_fingerprint(const PlanNode& n, std::string& out) {
     out += '\x1f';
     out += n.value;
     out += '\x1f';
+    for (const std::string& t : n.phrase_toks) {
+        out += t;
+        out += '\x1c';
+    }
     char buf[96];
     snprintf(buf, sizeof buf, "%d:%lld:%d:%lld:%lld:%.9g:%d|",
              int(n.lo.kind), (long long)n.lo.ival, int(n.hi.kind),
@@ -264,6 +272,32 @@ inline PlanNode full_text_plan(const std::string& field, const std::string& text
         (op == "and" ? b.must : b.should).push_back(std::move(n));
     }
     return b;
+}
+
+// multi-token phrase: consecutive positions, slop 0 (PhraseQuery,
+// full_text_query.rs phrase mode). Single-token phrases degrade to TERM;
+// zero tokens follow zero_terms_query like full_text_plan.
+inline PlanNode phrase_plan(const std::string& field, const std::string& text,
+                            const Schema& schema, bool zero_terms_all = false) {
+    const SchemaField* f = schema.field(field);
+    if (!f || f->type != "text")
+        throw std::runtime_error("phrase on unknown/non-text field: " + field);
+    std::vector<std::string> toks = tokenize(text, f->tokenizer);
+    PlanNode n;
+    if (toks.empty()) {
+        n.kind = zero_terms_all ? PlanNode::MATCH_ALL : PlanNode::MATCH_NONE;
+        return n;
+    }
+    if (toks.size() == 1) {
+        n.kind = PlanNode::TERM;
+        n.field = field;
+        n.value = toks[0];
+        return n;
+    }
+    n.kind = PlanNode::PHRASE;
+    n.field = field;
+    n.phrase_toks = std::move(toks);
+    return n;
 }
 
 // minimal user_input support: whitespace-separated clauses of `field:token`
@@ -580,16 +614,20 @@ inline PlanNode qtok_leaf_impl(const QTok& t, const std::vector<std::string>& df
         if (t.quoted) {
             // phrase: tokenize with the field's tokenizer; single-token
             // phrases (e.g. raw tokenizer keeps spaces) are plain terms,
-            // true multi-token phrases need positions (later round)
+            // multi-token phrases match consecutive positions (slop 0)
             std::vector<std::string> toks =
                 tokenize(t.text, f ? f->tokenizer : "default");
             if (toks.empty()) {
                 n.kind = PlanNode::MATCH_NONE;
                 return n;
             }
-            if (toks.size() > 1)
-                throw std::runtime_error(
-                    "phrase query needs positions (not in round 1)");
+            if (toks.size() > 1) {
+                n.kind = PlanNode::PHRASE;
+                n.field = t.field;
+                n.phrase_toks = std::move(toks);
+                n.boost = t.boost;
+                return n;
+            }
             n.kind = PlanNode::TERM;
             n.field = t.field;
             n.value = toks[0];
@@ -659,9 +697,17 @@ inline PlanNode build_plan(const mj::Value* ast, const Schema& schema) {
             zta = z && z->s == "all";
             const mj::Value* mode = params->get("mode");
             if (mode) {
-                // FullTextMode serde: {"type":"bool","operator":"or"|"and"} or
-                // {"type":"phrase",...} (full_text_query.rs:172)
+                // FullTextMode serde: {"type":"bool","operator":"or"|"and"},
+                // {"type":"phrase","slop":N} (full_text_query.rs:113-137,172)
                 const mj::Value* mt = mode->get("type");
+                if (mt && mt->s == "phrase") {
+                    const mj::Value* slop = mode->get("slop");
+                    if (slop && slop->as_i64() != 0)
+                        throw std::runtime_error(
+                            "phrase slop > 0 not supported (r2 limit)");
+                    return phrase_plan(ast->at("field")->s,
+                                       ast->at("text")->s, schema, zta);
+                }
                 if (mt && mt->s != "bool" && mt->s != "bool_prefix")
                     throw std::runtime_error("full_text mode not supported: " + mt->s);
                 const mj::Value* o = mode->get("operator");

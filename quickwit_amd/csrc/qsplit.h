@@ -33,11 +33,12 @@ struct TextFieldView {
     std::string name;
     std::string tokenizer;  // "raw" | "default"
     bool record_freq = false;
+    bool has_positions = false;  // record: "position" (phrase queries)
     bool has_norms = false;
     uint64_t total_tokens = 0;
     uint32_t num_terms = 0;
     Section term_offsets, term_bytes, posting_off, doc_freq, n_blocks, skip_off, skip,
-        payload, fieldnorms;
+        payload, fieldnorms, pos_start, positions;
     // host views
     const uint32_t* h_term_offsets = nullptr;
     const uint8_t* h_term_bytes = nullptr;
@@ -48,6 +49,8 @@ struct TextFieldView {
     const SkipEntry* h_skip = nullptr;
     const uint32_t* h_payload = nullptr;
     const uint8_t* h_fieldnorms = nullptr;
+    const uint32_t* h_pos_start = nullptr;  // per block: cumulative tf
+    const uint32_t* h_positions = nullptr;  // per (posting, occurrence)
 
     // binary search the sorted term dictionary; -1 if absent
     int64_t find_term(const char* t, size_t tlen) const {
@@ -214,7 +217,8 @@ struct SplitView {
                 TextFieldView t;
                 t.name = f->at("name")->s;
                 t.tokenizer = f->at("tokenizer")->s;
-                t.record_freq = f->at("record")->s == "freq";
+                t.record_freq = f->at("record")->s != "basic";
+                t.has_positions = f->at("record")->s == "position";
                 t.has_norms = f->at("fieldnorms")->b;
                 t.total_tokens = uint64_t(f->at("total_tokens")->as_i64());
                 t.num_terms = uint32_t(f->at("num_terms")->as_i64());
@@ -227,6 +231,10 @@ struct SplitView {
                 t.skip = sec(s, "skip");
                 t.payload = sec(s, "payload");
                 if (t.has_norms) t.fieldnorms = sec(s, "fieldnorms");
+                if (t.has_positions) {
+                    t.pos_start = sec(s, "pos_start");
+                    t.positions = sec(s, "positions");
+                }
                 t.h_term_offsets = (const uint32_t*)(data + t.term_offsets.off);
                 t.h_term_bytes = data + t.term_bytes.off;
                 t.h_posting_off = (const uint64_t*)(data + t.posting_off.off);
@@ -236,6 +244,10 @@ struct SplitView {
                 t.h_skip = (const SkipEntry*)(data + t.skip.off);
                 t.h_payload = (const uint32_t*)(data + t.payload.off);
                 if (t.has_norms) t.h_fieldnorms = data + t.fieldnorms.off;
+                if (t.has_positions) {
+                    t.h_pos_start = (const uint32_t*)(data + t.pos_start.off);
+                    t.h_positions = (const uint32_t*)(data + t.positions.off);
+                }
                 text_fields.push_back(std::move(t));
             } else {
                 FastFieldView ff;

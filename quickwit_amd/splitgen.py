@@ -112,10 +112,14 @@ class _Sections:
         return [off, len(b)]
 
 
-def _build_text_field(sec, fspec, num_docs, term_ids, doc_ids, tfs, vocab, doc_lengths):
+def _build_text_field(sec, fspec, num_docs, term_ids, doc_ids, tfs, vocab, doc_lengths,
+                      positions=None):
     """term_ids/doc_ids/tfs: postings sorted by (term, doc). vocab: list of
-    term strings ordered by term id == sorted order. Returns meta dict."""
-    record_freq = fspec.get("record", "basic") == "freq"
+    term strings ordered by term id == sorted order. positions: u32 array
+    ordered (term, doc, occurrence) when record == "position" (phrase
+    queries). Returns meta dict."""
+    record = fspec.get("record", "basic")
+    record_freq = record in ("freq", "position")
     nterms = len(vocab)
     df = np.bincount(term_ids, minlength=nterms).astype(np.int64)
     term_start = np.concatenate([[0], np.cumsum(df)])
@@ -199,7 +203,7 @@ def _build_text_field(sec, fspec, num_docs, term_ids, doc_ids, tfs, vocab, doc_l
         "name": fspec["name"],
         "type": "text",
         "tokenizer": fspec.get("tokenizer", "default"),
-        "record": "freq" if record_freq else "basic",
+        "record": record if record_freq else "basic",
         "fieldnorms": norms is not None,
         "total_tokens": int(doc_lengths.sum()),
         "num_terms": nterms,
@@ -216,6 +220,18 @@ def _build_text_field(sec, fspec, num_docs, term_ids, doc_ids, tfs, vocab, doc_l
     }
     if norms is not None:
         meta["sec"]["fieldnorms"] = sec.add(norms)
+    if record == "position":
+        assert positions is not None and len(positions) == int(tfs.sum())
+        # per-block cumulative tf: posting j of block b starts at
+        # pos_start[b] + sum(tf of postings before j in the block)
+        cum = np.concatenate([[0], np.cumsum(tfs.astype(np.int64))])
+        if total_blocks:
+            pos_start = cum[blk_start].astype(np.uint32)
+        else:
+            pos_start = np.zeros(0, np.uint32)
+        meta["sec"]["pos_start"] = sec.add(pos_start)
+        meta["sec"]["positions"] = sec.add(
+            np.asarray(positions, dtype=np.uint32))
     return meta
 
 
@@ -443,26 +459,29 @@ class SplitWriter:
 
 def _postings_from_tokens(per_doc_tokens):
     """per_doc_tokens: list of token lists -> (vocab, term_ids, doc_ids, tfs)
-    sorted by (term, doc), plus doc token counts."""
+    sorted by (term, doc), plus doc token counts and the flat positions
+    array ((term, doc, occurrence) order — token index within the doc)."""
     vocab = sorted({t for toks in per_doc_tokens for t in toks})
     tid = {t: i for i, t in enumerate(vocab)}
-    tuples = []
+    occ = {}
     for d, toks in enumerate(per_doc_tokens):
-        counts = {}
-        for t in toks:
-            counts[t] = counts.get(t, 0) + 1
-        for t, c in counts.items():
-            tuples.append((tid[t], d, c))
-    tuples.sort()
-    if tuples:
-        arr = np.array(tuples, dtype=np.int64)
-        term_ids, doc_ids, tfs = arr[:, 0], arr[:, 1].astype(np.uint32), arr[:, 2].astype(np.uint32)
+        for p, t in enumerate(toks):
+            occ.setdefault((tid[t], d), []).append(p)
+    keys = sorted(occ)
+    if keys:
+        arr = np.array(keys, dtype=np.int64)
+        term_ids = arr[:, 0]
+        doc_ids = arr[:, 1].astype(np.uint32)
+        tfs = np.array([len(occ[k]) for k in keys], dtype=np.uint32)
+        positions = np.array([p for k in keys for p in occ[k]],
+                             dtype=np.uint32)
     else:
         term_ids = np.zeros(0, np.int64)
         doc_ids = np.zeros(0, np.uint32)
         tfs = np.zeros(0, np.uint32)
+        positions = np.zeros(0, np.uint32)
     lengths = np.array([len(t) for t in per_doc_tokens], dtype=np.int64)
-    return vocab, term_ids, doc_ids, tfs, lengths
+    return vocab, term_ids, doc_ids, tfs, lengths, positions
 
 
 DOCSTORE_BLOCK_BYTES = 16 * 1024  # uncompressed block target (tantivy-shaped)
@@ -511,14 +530,17 @@ def _assemble(schema, split_id, num_docs, text_inputs, fast_inputs, precomputed_
     for fspec in schema["fields"]:
         name = fspec["name"]
         if fspec["type"] == "text":
+            positions = None
             if precomputed_text and name in precomputed_text:
                 vocab, term_ids, doc_ids, tfs, lengths = precomputed_text[name]
+                assert fspec.get("record") != "position", \
+                    "positions need the document-at-a-time writer"
             else:
-                vocab, term_ids, doc_ids, tfs, lengths = _postings_from_tokens(
-                    text_inputs[name]
-                )
+                (vocab, term_ids, doc_ids, tfs, lengths,
+                 positions) = _postings_from_tokens(text_inputs[name])
             fields_meta.append(
-                _build_text_field(sec, fspec, num_docs, term_ids, doc_ids, tfs, vocab, lengths)
+                _build_text_field(sec, fspec, num_docs, term_ids, doc_ids,
+                                  tfs, vocab, lengths, positions=positions)
             )
             if fspec.get("fast", False):
                 # text field with fast=true (quickwit dynamic mapping): BOTH
