@@ -1189,3 +1189,158 @@ extern "C" __global__ void k_cand_compact(const uint64_t* cand, uint32_t n,
 }
 
 }  // namespace qw
+
+namespace qw {
+
+// ------------------------------------------------------------ phrase eval
+// Multi-token phrase (slop 0) into a device HitSet bitmap: one THREAD per
+// posting of token 0 (the driver term). Per posting: decode its doc id via
+// the block's 32-element anchors (<=32 gap extracts), its position-stream
+// offset (tf prefix scan over the block), then for each other token a
+// binary search over its skip entries + an in-block scan for the doc, and
+// finally the consecutive-position chain over the raw u32 position lists.
+// Phrases are rare-path (niche vs the flagship decode kernels) — scalar
+// per-thread work is the right shape; throughput is bounded by the driver
+// term's df, not the split size.
+struct PhraseTokDev {
+    uint64_t skip_off;       // byte off of this term's first SkipEntry
+    uint64_t payload_off;    // field payload byte off
+    uint64_t pos_start_off;  // byte off of pos_start u32 AT the term's first block
+    uint64_t positions_off;  // field positions byte off
+    uint32_t n_blocks;
+    uint32_t df;
+};
+constexpr uint32_t PHRASE_MAX_TOKS = 8;
+struct PhraseDev {
+    uint32_t n_toks;
+    uint32_t num_docs;
+    PhraseTokDev tok[PHRASE_MAX_TOKS];
+};
+
+__device__ __forceinline__ uint32_t phrase_gap(const uint32_t* gw, uint32_t j,
+                                               uint32_t w) {
+    return uint32_t(extract_bits(gw, uint64_t(j) * w, w));
+}
+
+// doc id of posting j in block e (anchors shorten the scan to <=32 gaps)
+__device__ static uint32_t phrase_doc_at(const uint32_t* payload,
+                                         const SkipEntryDev& e, uint32_t j) {
+    const uint32_t* gw = payload + e.word_off;
+    uint32_t seg = j >> 5;
+    // anchors live in the 2 u64 words before the gaps, u32 layout
+    // [pad, doc31, doc63, doc95]: anchor for segment s is (gw-4)[s]
+    uint32_t doc = seg == 0 ? e.first_doc : (gw - 4)[seg];
+    for (uint32_t i = seg == 0 ? 1 : seg * 32; i <= j; ++i)
+        doc += phrase_gap(gw, i, e.id_bits);
+    return doc;
+}
+
+// tf of posting j + sum of tfs of postings [0, j) in block e
+__device__ static void phrase_tf_prefix(const uint32_t* payload,
+                                        const SkipEntryDev& e, uint32_t j,
+                                        uint32_t* tf_out, uint32_t* pre_out) {
+    if (!e.tf_bits) {  // record without tf: tf = 1 everywhere
+        *tf_out = 1;
+        *pre_out = j;
+        return;
+    }
+    // tf fields start after the gap words (2*id_bits u64 = this many u32s)
+    const uint32_t* tfw = payload + e.word_off + ((128u * e.id_bits + 63u) / 64u) * 2;
+    uint32_t pre = 0;
+    for (uint32_t i = 0; i < j; ++i)
+        pre += phrase_gap(tfw, i, e.tf_bits) + 1u;
+    *tf_out = phrase_gap(tfw, j, e.tf_bits) + 1u;
+    *pre_out = pre;
+}
+
+extern "C" __global__ void k_phrase_bitmap(const uint8_t* split, PhraseDev p,
+                                           uint32_t* bitmap) {
+    const PhraseTokDev& t0 = p.tok[0];
+    const SkipEntryDev* skip0 = (const SkipEntryDev*)(split + t0.skip_off);
+    const uint32_t* pay0 = (const uint32_t*)(split + t0.payload_off);
+    const uint32_t* ps0 = (const uint32_t*)(split + t0.pos_start_off);
+    const uint32_t* P0 = (const uint32_t*)(split + t0.positions_off);
+    for (uint32_t t = blockIdx.x * blockDim.x + threadIdx.x; t < t0.df;
+         t += gridDim.x * blockDim.x) {
+        uint32_t b = t >> 7, j = t & 127u;
+        SkipEntryDev e0 = skip0[b];
+        if (j >= e0.count) continue;
+        uint32_t doc = phrase_doc_at(pay0, e0, j);
+        uint32_t tf0, pre0;
+        phrase_tf_prefix(pay0, e0, j, &tf0, &pre0);
+        uint32_t off0 = ps0[b] + pre0;
+
+        // locate `doc` in every other token; collect (pos_off, tf)
+        uint32_t offi[PHRASE_MAX_TOKS], ni[PHRASE_MAX_TOKS];
+        bool ok = true;
+        for (uint32_t i = 1; i < p.n_toks && ok; ++i) {
+            const PhraseTokDev& ti = p.tok[i];
+            const SkipEntryDev* skipi = (const SkipEntryDev*)(split + ti.skip_off);
+            // first block with last_doc >= doc
+            uint32_t lo = 0, hi = ti.n_blocks;
+            while (lo < hi) {
+                uint32_t mid = (lo + hi) >> 1;
+                if (skipi[mid].last_doc < doc) lo = mid + 1;
+                else hi = mid;
+            }
+            if (lo == ti.n_blocks || skipi[lo].first_doc > doc) {
+                ok = false;
+                break;
+            }
+            SkipEntryDev ei = skipi[lo];
+            const uint32_t* payi = (const uint32_t*)(split + ti.payload_off);
+            const uint32_t* gw = payi + ei.word_off;
+            uint32_t cur = ei.first_doc, pre = 0, jj = 0, tfj = 0;
+            bool found = false;
+            const uint32_t* tfw =
+                ei.tf_bits
+                    ? payi + ei.word_off + ((128u * ei.id_bits + 63u) / 64u) * 2
+                    : nullptr;
+            for (; jj < ei.count; ++jj) {
+                if (jj) cur += phrase_gap(gw, jj, ei.id_bits);
+                tfj = tfw ? phrase_gap(tfw, jj, ei.tf_bits) + 1u : 1u;
+                if (cur == doc) {
+                    found = true;
+                    break;
+                }
+                if (cur > doc) break;
+                pre += tfj;
+            }
+            if (!found) {
+                ok = false;
+                break;
+            }
+            offi[i] = ((const uint32_t*)(split + ti.pos_start_off))[lo] + pre;
+            ni[i] = tfj;
+        }
+        if (!ok) continue;
+
+        // consecutive-position chain: exists a in tok0 positions with
+        // (a + i) in tok_i positions for all i
+        bool hit = false;
+        for (uint32_t a = 0; a < tf0 && !hit; ++a) {
+            uint32_t pos0 = P0[off0 + a];
+            bool chain = true;
+            for (uint32_t i = 1; i < p.n_toks && chain; ++i) {
+                const uint32_t* Pi =
+                    (const uint32_t*)(split + p.tok[i].positions_off);
+                uint32_t lo = offi[i], hi = offi[i] + ni[i], want = pos0 + i;
+                chain = false;
+                while (lo < hi) {
+                    uint32_t mid = (lo + hi) >> 1;
+                    uint32_t v = Pi[mid];
+                    if (v == want) {
+                        chain = true;
+                        break;
+                    }
+                    if (v < want) lo = mid + 1;
+                    else hi = mid;
+                }
+            }
+            hit = chain;
+        }
+        if (hit) atomicOr(&bitmap[doc >> 5], 1u << (doc & 31u));
+    }
+}
+
+}  // namespace qw

@@ -680,6 +680,9 @@ static void add_positive(FlatQuery& fq, const SplitView& sv, const PlanNode& c,
                     "nested boolean inside must/filter not flattenable (GPU r1)");
             }
             break;
+        case PlanNode::PHRASE:
+            throw std::runtime_error(
+                "phrase clause not flattenable (device-bitmap path)");
     }
 }
 
@@ -788,8 +791,11 @@ static FlatQuery flatten(const SplitView& sv, const PlanNode& plan, bool scoring
     fq.scoring = scoring;
     if (scoring && plan_has_const_score(plan))
         throw std::runtime_error(
-            "term_set/wildcard under _score sorting not supported (const-score "
-            "semantics, qast.h)");
+            "term_set/wildcard/phrase under _score sorting not supported "
+            "(const-score semantics, qast.h)");
+    if (plan.kind == PlanNode::PHRASE)
+        throw std::runtime_error(
+            "phrase not flattenable (device-bitmap path)");
     switch (plan.kind) {
         case PlanNode::MATCH_ALL:
             fq.match_all = true;
@@ -1418,6 +1424,53 @@ static uint8_t* resolve_hitset(qw_ctx* ctx, const DeviceSplit& ds,
     qw_check_budget(ctx, bm_bytes);
     HIP_CHECK(hipMalloc(&bm, bm_bytes));
 
+    if (inner.kind == PlanNode::PHRASE) {
+        // multi-token phrase (slop 0, full_text_query.rs phrase mode):
+        // k_phrase_bitmap — one thread per posting of the driver token,
+        // consecutive-position chain over the field's position streams
+        const TextFieldView* f = sv.text_field(inner.field);
+        bool none = !f;
+        if (f && !f->has_positions) {
+            (void)hipFree(bm);
+            throw std::runtime_error(
+                "phrase query needs record: position on field " + inner.field);
+        }
+        if (inner.phrase_toks.size() > PHRASE_MAX_TOKS) {
+            (void)hipFree(bm);
+            throw std::runtime_error("phrase longer than 8 tokens (r2 limit)");
+        }
+        PhraseDev p{};
+        p.n_toks = uint32_t(inner.phrase_toks.size());
+        p.num_docs = sv.num_docs;
+        for (size_t i = 0; i < inner.phrase_toks.size() && !none; ++i) {
+            const std::string& tok = inner.phrase_toks[i];
+            int64_t tid = f->find_term(tok.data(), tok.size());
+            if (tid < 0) {
+                none = true;  // absent token: phrase matches nothing
+                break;
+            }
+            PhraseTokDev& td = p.tok[i];
+            td.skip_off = f->skip.off + f->h_skip_off[tid];
+            td.payload_off = f->payload.off;
+            td.pos_start_off =
+                f->pos_start.off + (f->h_skip_off[tid] / 16) * 4;
+            td.positions_off = f->positions.off;
+            td.n_blocks = f->h_n_blocks[tid];
+            td.df = f->h_doc_freq[tid];
+        }
+        HIP_CHECK(hipMemsetAsync(bm, 0, bm_bytes, ctx->stream));
+        if (!none && p.tok[0].df) {
+            uint32_t grid =
+                std::min<uint32_t>(2048, (p.tok[0].df + 255) / 256);
+            hipLaunchKernelGGL(k_phrase_bitmap, dim3(grid), dim3(256), 0,
+                               ctx->stream, ds.d_image, p, (uint32_t*)bm);
+        }
+        HIP_CHECK(hipStreamSynchronize(ctx->stream));
+        HIP_CHECK(hipGetLastError());
+        ctx->hitsets.put(key, bm, bm_bytes);
+        return bm;
+    }
+
     FlatQuery fq;
     try {
         fq = flatten(sv, inner, false);
@@ -1528,6 +1581,12 @@ static void collect_required_terms(
     switch (n.kind) {
         case PlanNode::TERM:
             out.emplace_back(n.field, n.value);
+            break;
+        case PlanNode::PHRASE:
+            // every phrase token is required (the phrase cannot match if
+            // any token is absent)
+            for (const std::string& t : n.phrase_toks)
+                out.emplace_back(n.field, t);
             break;
         case PlanNode::BOOL:
             for (const PlanNode& c : n.must) collect_required_terms(c, out);

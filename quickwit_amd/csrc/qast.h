@@ -70,7 +70,7 @@ struct Schema {
             const mj::Value* tok = fv->get("tokenizer");
             f.tokenizer = tok ? tok->s : "default";
             const mj::Value* rec = fv->get("record");
-            f.record_freq = rec && rec->s == "freq";
+            f.record_freq = rec && (rec->s == "freq" || rec->s == "position");
             const mj::Value* fast = fv->get("fast");
             f.fast = fast && fast->b;
             s.fields.push_back(std::move(f));
@@ -163,7 +163,7 @@ inline void plan_fingerprint(const PlanNode& n, std::string& out) {
 
 // does any node require const-score semantics (term_set / wildcard)?
 inline bool plan_has_const_score(const PlanNode& n) {
-    if (n.const_score) return true;
+    if (n.const_score || n.kind == PlanNode::PHRASE) return true;
     for (auto* v : {&n.must, &n.must_not, &n.should, &n.filter, &n.cache_inner})
         for (auto& c : *v)
             if (plan_has_const_score(c)) return true;

@@ -143,6 +143,14 @@ def es_query_to_ast(q, schema=None):
         elif "lt" in body:
             node["upper_bound"] = {"excluded": conv(body["lt"])}
         return node
+    if "match_phrase" in q:
+        [(field, body)] = q["match_phrase"].items()
+        text = body["query"] if isinstance(body, dict) else body
+        mode = {"type": "phrase"}
+        if isinstance(body, dict) and body.get("slop"):
+            mode["slop"] = body["slop"]
+        return {"type": "full_text", "field": field, "text": str(text),
+                "params": {"mode": mode}}
     if "match" in q:
         [(field, body)] = q["match"].items()
         text = body["query"] if isinstance(body, dict) else body
@@ -323,8 +331,8 @@ class Index:
             elif t == "text":
                 f = {"name": name, "type": "text",
                      "tokenizer": fm.get("tokenizer", "default")}
-                if fm.get("record") == "freq":
-                    f["record"] = "freq"
+                if fm.get("record") in ("freq", "position"):
+                    f["record"] = fm["record"]
                 if fm.get("fast"):
                     f["fast"] = True
                 fields.append(f)

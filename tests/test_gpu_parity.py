@@ -1661,3 +1661,58 @@ def test_mixed_type_column_sort_parity():
     assert [h.get("sort_value") for h in g["partial_hits"]][:4] == [
         {"u64": 0}, {"boolean": True}, {"f64": 10.5},
         {"u64": 18000000000000000000}]
+
+
+# --------------------------------------------------------------- phrases
+def test_phrase_parity_gpu():
+    """Multi-token phrases through k_phrase_bitmap (device HitSet path) vs
+    the oracle: bare phrase, under bool filter/must/must_not, repeated
+    tokens, absent token, multi-block postings."""
+    import random as _r
+    schema = {"timestamp_field": None, "fields": [
+        {"name": "body", "type": "text", "tokenizer": "default",
+         "record": "position", "fieldnorms": True},
+        {"name": "sev", "type": "u64", "fast": True}]}
+    rng = _r.Random(77)
+    vocab = ["alpha", "beta", "gamma", "delta", "eps", "zeta"]
+    docs = [{"body": " ".join(rng.choice(vocab)
+                              for _ in range(rng.randint(3, 12))),
+             "sev": rng.randint(0, 9)} for _ in range(3000)]
+    w = splitgen.SplitWriter(schema, "ph", store_docs=False)
+    w.add_documents(docs)
+    data = w.finalize()
+    gpu, cpu = GpuSearcher(device=0), OracleSearcher()
+    gpu.add_split("ph", data)
+    cpu.add_split("ph", data)
+
+    def phrase(toks):
+        return {"type": "full_text", "field": "body", "text": " ".join(toks),
+                "params": {"mode": {"type": "phrase"}}}
+
+    queries = [
+        phrase(["alpha", "beta"]),
+        phrase(["beta", "beta"]),
+        phrase(["gamma", "delta", "eps"]),
+        phrase(["eps", "eps", "eps"]),
+        phrase(["alpha", "missingtok"]),
+        {"type": "bool", "filter": [phrase(["alpha", "beta"])],
+         "must": [{"type": "term", "field": "body", "value": "gamma"}]},
+        {"type": "bool", "filter": [phrase(["alpha", "beta"])],
+         "must": [{"type": "range", "field": "sev",
+                   "lower_bound": {"included": 3},
+                   "upper_bound": {"excluded": 8}}]},
+        {"type": "bool",
+         "must": [{"type": "term", "field": "body", "value": "gamma"}],
+         "must_not": [phrase(["alpha", "beta"])]},
+    ]
+    for q in queries:
+        req = make_leaf_request(q, schema, [("ph", 3000)], max_hits=3000)
+        g, e = gpu.leaf_search(req), cpu.leaf_search(req)
+        assert g.get("num_hits", 0) == e.get("num_hits", 0), q
+        assert sorted(h.get("doc_id", 0) for h in g.get("partial_hits", [])) \
+            == sorted(h.get("doc_id", 0) for h in e.get("partial_hits", [])), q
+    # memoized: second run hits the bitmap cache and stays identical
+    req = make_leaf_request(queries[0], schema, [("ph", 3000)], max_hits=10)
+    r1 = gpu.leaf_search(req)
+    r2 = gpu.leaf_search(req)
+    assert r1.get("num_hits") == r2.get("num_hits")
