@@ -703,7 +703,7 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
                 if (d.kind == AggDef::TERMS)
                     truncate_terms_split(
                         a, effective_split_size(d.size, d.split_size),
-                        d.order_target, d.order_asc);
+                        d.order_target, d.order_asc, &d.sub);
             } else if (f && !f->multi) {
                 // terms over a numeric fast column: count by the value's
                 // order-preserving sortable bits, keys encoded big-endian so
@@ -737,7 +737,7 @@ static IntermediateAggResults collect_aggs(const std::vector<AggDef>& defs,
                 if (d.kind == AggDef::TERMS)
                     truncate_terms_split(
                         a, effective_split_size(d.size, d.split_size),
-                        d.order_target, d.order_asc);
+                        d.order_target, d.order_asc, &d.sub);
             }
             out.aggs.push_back(std::move(a));
             continue;

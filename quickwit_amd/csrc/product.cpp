@@ -2388,7 +2388,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     if (d.kind == AggDef::TERMS)
                         truncate_terms_split(
                             r, effective_split_size(d.size, d.split_size),
-                            d.order_target, d.order_asc);
+                            d.order_target, d.order_asc, &d.sub);
                 }
             } else if (d.kind == AggDef::COMPOSITE) {
                 // decode packed 63-bit keys into the canonical per-source

@@ -177,13 +177,85 @@ inline void comp_encode_f64(std::string& o, double v) {
     o += num_term_key(f64_to_u64(v));
 }
 
+// resolve a terms `order` target of the form "sub" or "sub.stat" to the
+// bucket's ordering value (tantivy terms order by sub-aggregation; ES
+// {"order": {"lat.avg": "desc"}}). Single-valued metric subs accept the
+// bare sub name; stats/extended_stats need the ".stat" selector.
+inline double terms_order_sub_value(const std::vector<MetricAgg>& subs,
+                                    const std::vector<std::vector<StatsPayload>>& term_subs,
+                                    const std::string& target, size_t bucket) {
+    std::string name = target, stat;
+    size_t dot = target.rfind('.');
+    if (dot != std::string::npos) {
+        name = target.substr(0, dot);
+        stat = target.substr(dot + 1);
+    }
+    for (size_t si = 0; si < subs.size(); ++si) {
+        if (subs[si].name != name) continue;
+        if (bucket >= term_subs.size() || si >= term_subs[bucket].size())
+            return 0.0;
+        const StatsPayload& sp = term_subs[bucket][si];
+        MetricAgg::Kind k = subs[si].kind;
+        if (stat.empty()) {
+            switch (k) {
+                case MetricAgg::AVG:
+                    return sp.count ? sp.sum / double(sp.count) : 0.0;
+                case MetricAgg::SUM: return sp.sum;
+                case MetricAgg::MIN: return sp.count ? sp.min : 0.0;
+                case MetricAgg::MAX: return sp.count ? sp.max : 0.0;
+                case MetricAgg::COUNT: return double(sp.count);
+                default:
+                    throw std::runtime_error(
+                        "terms order by multi-value sub '" + name +
+                        "' needs a .stat selector");
+            }
+        }
+        if (stat == "avg") return sp.count ? sp.sum / double(sp.count) : 0.0;
+        if (stat == "sum") return sp.sum;
+        if (stat == "min") return sp.count ? sp.min : 0.0;
+        if (stat == "max") return sp.count ? sp.max : 0.0;
+        if (stat == "count") return double(sp.count);
+        throw std::runtime_error("terms order: unknown stat '" + stat + "'");
+    }
+    throw std::runtime_error("terms order: unknown sub-aggregation '" + name +
+                             "'");
+}
+
 // per-split terms truncation (tantivy terms agg split_size): keep the top
 // `split_size` entries by (count desc, key asc); when anything is dropped,
 // the last included count joins the error bound. Restores key order after.
 inline void truncate_terms_split(AggResult& r, int64_t split_size,
                                  const std::string& order_target = "",
-                                 bool order_asc = false) {
+                                 bool order_asc = false,
+                                 const std::vector<MetricAgg>* subs = nullptr) {
     if (split_size < 1) split_size = 1;
+    if (!order_target.empty() && order_target != "_key" && subs) {
+        // order by a sub-aggregation value: shard-local top split_size by
+        // that value (ES/tantivy shard semantics; doc-count error bounds
+        // don't apply to sub-agg ordering)
+        if (r.term_counts.size() <= uint64_t(split_size)) return;
+        std::vector<size_t> idx(r.term_counts.size());
+        for (size_t i = 0; i < idx.size(); ++i) idx[i] = i;
+        std::stable_sort(idx.begin(), idx.end(), [&](size_t x, size_t y) {
+            double vx =
+                terms_order_sub_value(*subs, r.term_subs, order_target, x);
+            double vy =
+                terms_order_sub_value(*subs, r.term_subs, order_target, y);
+            if (vx != vy) return order_asc ? vx < vy : vx > vy;
+            return r.term_counts[x].first < r.term_counts[y].first;
+        });
+        idx.resize(size_t(split_size));
+        std::sort(idx.begin(), idx.end());  // restore key order
+        std::vector<std::pair<std::string, uint64_t>> tc;
+        std::vector<std::vector<StatsPayload>> ts;
+        for (size_t i : idx) {
+            tc.push_back(std::move(r.term_counts[i]));
+            if (!r.term_subs.empty()) ts.push_back(std::move(r.term_subs[i]));
+        }
+        r.term_counts = std::move(tc);
+        r.term_subs = std::move(ts);
+        return;
+    }
     if (order_target == "_key") {
         // key order: the per-split prefix (or suffix) in key order is EXACT
         // — no doc-count error contribution
@@ -823,24 +895,32 @@ inline std::string finalize_aggs_json(const IntermediateAggResults& ir,
             // to size; sum_other = matched - shown
             std::vector<size_t> ordered(a.term_counts.size());
             for (size_t i = 0; i < ordered.size(); ++i) ordered[i] = i;
-            std::stable_sort(ordered.begin(), ordered.end(),
-                             [&](size_t x, size_t y) {
-                                 if (d.order_target == "_key")
-                                     return d.order_asc
-                                                ? a.term_counts[x].first <
-                                                      a.term_counts[y].first
-                                                : a.term_counts[y].first <
-                                                      a.term_counts[x].first;
-                                 if (a.term_counts[x].second !=
-                                     a.term_counts[y].second)
-                                     return d.order_asc
-                                                ? a.term_counts[x].second <
-                                                      a.term_counts[y].second
-                                                : a.term_counts[x].second >
-                                                      a.term_counts[y].second;
-                                 return a.term_counts[x].first <
-                                        a.term_counts[y].first;
-                             });
+            bool sub_order = !d.order_target.empty() &&
+                             d.order_target != "_key";
+            std::stable_sort(
+                ordered.begin(), ordered.end(), [&](size_t x, size_t y) {
+                    if (d.order_target == "_key")
+                        return d.order_asc ? a.term_counts[x].first <
+                                                 a.term_counts[y].first
+                                           : a.term_counts[y].first <
+                                                 a.term_counts[x].first;
+                    if (sub_order) {
+                        double vx = terms_order_sub_value(
+                            d.sub, a.term_subs, d.order_target, x);
+                        double vy = terms_order_sub_value(
+                            d.sub, a.term_subs, d.order_target, y);
+                        if (vx != vy)
+                            return d.order_asc ? vx < vy : vx > vy;
+                        return a.term_counts[x].first <
+                               a.term_counts[y].first;
+                    }
+                    if (a.term_counts[x].second != a.term_counts[y].second)
+                        return d.order_asc ? a.term_counts[x].second <
+                                                 a.term_counts[y].second
+                                           : a.term_counts[x].second >
+                                                 a.term_counts[y].second;
+                    return a.term_counts[x].first < a.term_counts[y].first;
+                });
             // min_doc_count (ES default 1 for terms): drop below-threshold
             // buckets BEFORE the size cut
             int64_t mdc = d.min_doc_count < 0 ? 1 : d.min_doc_count;

@@ -1132,9 +1132,8 @@ inline std::vector<AggDef> parse_agg_request(const std::string& json) {
                 const std::string dir = body2->obj.begin()->second->s;
                 if (tgt == "_count") a.order_target = "";
                 else if (tgt == "_key" || tgt == "_term") a.order_target = "_key";
-                else
-                    throw std::runtime_error(
-                        "terms order by sub-aggregation (r1 limit)");
+                else a.order_target = tgt;  // sub-aggregation value path:
+                                            // "sub" or "sub.stat"
                 a.order_asc = dir == "asc";
             }
         } else {

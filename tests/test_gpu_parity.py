@@ -1716,3 +1716,22 @@ def test_phrase_parity_gpu():
     r1 = gpu.leaf_search(req)
     r2 = gpu.leaf_search(req)
     assert r1.get("num_hits") == r2.get("num_hits")
+
+
+def test_terms_order_by_subagg_on_gpu(searchers):
+    # terms order by a sub-aggregation value through the HIP path
+    gpu, cpu = searchers
+    for order in ({"st.avg": "desc"}, {"st.max": "asc"},
+                  {"avg_t": "desc"}):
+        aggs = {"t": {"terms": {"field": "tenant_name", "size": 8,
+                                "order": order},
+                      "aggs": {"st": {"stats": {"field": "tenant_id"}},
+                               "avg_t": {"avg": {"field": "tenant_id"}}}}}
+        req = make_leaf_request({"type": "match_all"}, SCHEMA, [(SID, NDOCS)],
+                                max_hits=0, aggregation=aggs)
+        g = gpu.leaf_search(req)
+        e = cpu.leaf_search(req)
+        gj = gpu.finalize_agg_json(g["intermediate_aggregation_result"], aggs)
+        ej = cpu.finalize_agg_json(e["intermediate_aggregation_result"], aggs)
+        assert [b["key"] for b in gj["t"]["buckets"]] == \
+            [b["key"] for b in ej["t"]["buckets"]], order
