@@ -52,7 +52,8 @@ def main():
         for k in ("union_bm25", "column_agg", "range_filter", "topk_select"):
             try:
                 ms, n = s.kernel_stats(k)
-                parts.append(f"{k}={ms/n:.3f}ms x{n}")
+                if n:
+                    parts.append(f"{k}={ms/n:.3f}ms x{n}")
             except KeyError:
                 pass
         print(f"{name:18s} wall {wall:6.3f} ms | " + "  ".join(parts),
