@@ -514,18 +514,25 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
     bool af_terms = false;
     if (NA && q.agg_fast) {
         const AggDev& a0 = aggs[0];
-        af_col = (const int64_t*)(q.split + a0.values_off);
-        af_ioff = a0.i_offset;
-        af_ivl = a0.i_interval;
-        af_inv = a0.inv_interval;
-        af_base = a0.base_index;
-        af_hrep = a0.lds_rep;
         af_par = lane_id() & 1u;  // copy index for 2-way replication
-        if (q.n_aggs > 1) {
+        if (a0.kind == AGGD_HISTO) {
+            af_col = (const int64_t*)(q.split + a0.values_off);
+            af_ioff = a0.i_offset;
+            af_ivl = a0.i_interval;
+            af_inv = a0.inv_interval;
+            af_base = a0.base_index;
+            af_hrep = a0.lds_rep;
+            if (q.n_aggs > 1) {
+                af_terms = true;
+                af_tcol = q.split + aggs[1].values_off;
+                af_twidth = aggs[1].value_width;
+                af_trep = aggs[1].lds_rep;
+            }
+        } else {  // terms-only straight-line path (af_col stays null)
             af_terms = true;
-            af_tcol = q.split + aggs[1].values_off;
-            af_twidth = aggs[1].value_width;
-            af_trep = aggs[1].lds_rep;
+            af_tcol = q.split + a0.values_off;
+            af_twidth = a0.value_width;
+            af_trep = a0.lds_rep;
         }
     }
 
@@ -606,14 +613,18 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
         if (NA && !NS && !NB && !NC && q.agg_fast && q.match_all && !q.n_preds) {
             #pragma unroll 4
             for (uint32_t d = tile_lo + threadIdx.x; d < tile_hi; d += TILE_THREADS) {
-                int64_t num = (q.agg_nt ? __builtin_nontemporal_load(&af_col[d])
-                                        : af_col[d]) -
-                              af_ioff;
-                int64_t idx = int64_t(floor(double(num) * af_inv));
-                if (idx * af_ivl > num) --idx;
-                else if ((idx + 1) * af_ivl <= num) ++idx;
-                uint32_t hb = uint32_t(idx - af_base);
-                atomicAdd(&sc_agg_hist[af_hrep == 2 ? hb * 2 + af_par : hb], 1u);
+                if (af_col) {
+                    int64_t num =
+                        (q.agg_nt ? __builtin_nontemporal_load(&af_col[d])
+                                  : af_col[d]) -
+                        af_ioff;
+                    int64_t idx = int64_t(floor(double(num) * af_inv));
+                    if (idx * af_ivl > num) --idx;
+                    else if ((idx + 1) * af_ivl <= num) ++idx;
+                    uint32_t hb = uint32_t(idx - af_base);
+                    atomicAdd(&sc_agg_hist[af_hrep == 2 ? hb * 2 + af_par : hb],
+                              1u);
+                }
                 if (af_terms) {
                     uint64_t o = af_twidth == 2 ? ((const uint16_t*)af_tcol)[d]
                                  : (af_twidth == 1 ? af_tcol[d]
@@ -657,16 +668,19 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                     // straight-line: exact floor-div histogram bucket (bounds
                     // cover the column's [min,max] => no range check) +
                     // optional ord-count table, both LDS
-                    int64_t num = af_col[d] - af_ioff;
-                    int64_t idx = int64_t(floor(double(num) * af_inv));
-                    if (idx * af_ivl > num) --idx;
-                    else if ((idx + 1) * af_ivl <= num) ++idx;
-                    uint32_t hb = uint32_t(idx - af_base);
-                    // same lds_rep mapping as the pure-agg tile loop above:
-                    // the end-of-kernel flush sums interleaved pairs when
-                    // lds_rep==2, so indices must match there too
-                    atomicAdd(&sc_agg_hist[af_hrep == 2 ? hb * 2 + af_par : hb],
-                              1u);
+                    if (af_col) {
+                        int64_t num = af_col[d] - af_ioff;
+                        int64_t idx = int64_t(floor(double(num) * af_inv));
+                        if (idx * af_ivl > num) --idx;
+                        else if ((idx + 1) * af_ivl <= num) ++idx;
+                        uint32_t hb = uint32_t(idx - af_base);
+                        // same lds_rep mapping as the pure-agg tile loop:
+                        // the end-of-kernel flush sums interleaved pairs
+                        // when lds_rep==2, so indices must match there too
+                        atomicAdd(
+                            &sc_agg_hist[af_hrep == 2 ? hb * 2 + af_par : hb],
+                            1u);
+                    }
                     if (af_terms) {
                         uint64_t o = af_twidth == 2
                                          ? ((const uint16_t*)af_tcol)[d]
@@ -1007,7 +1021,14 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
             const AggDev& a = aggs[ai];
             if (a.lds_slot > 1) continue;
             uint32_t* src = a.lds_slot == 1 ? sc_agg_terms : sc_agg_hist;
-            for (uint32_t i = threadIdx.x; i < a.n_buckets; i += TILE_THREADS) {
+            // stagger the bucket order per workgroup: concurrent WGs would
+            // otherwise sweep the buckets in lockstep and serialize on the
+            // same global words
+            uint32_t rot = (blockIdx.x * 131u) % (a.n_buckets ? a.n_buckets : 1u);
+            for (uint32_t i0 = threadIdx.x; i0 < a.n_buckets;
+                 i0 += TILE_THREADS) {
+                uint32_t i = i0 + rot;
+                if (i >= a.n_buckets) i -= a.n_buckets;
                 uint32_t v = a.lds_rep == 2 ? src[2 * i] + src[2 * i + 1] : src[i];
                 if (v)
                     atomicAdd((unsigned long long*)(q.results + a.counts_out) + i,

@@ -1991,8 +1991,14 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         bool nb = n_must > 0 || n_must_not > 0;
         bool na = do_aggs && !ap.devs.empty();
         // agg workloads: capped grid so the once-per-workgroup LDS flush
-        // stays cheap (each WG owns several tiles); otherwise one WG per tile
-        uint32_t grid = na ? std::min<uint32_t>(n_tiles, 2048) : n_tiles;
+        // stays cheap (each WG owns several tiles) — and scaled DOWN with
+        // the split size: the flush costs grid x buckets global atomics
+        // regardless of docs, which dominated small (config5-sized) splits.
+        // >=256 WGs x 4 waves still fills the chip's 1024 resident-wave slots.
+        uint32_t grid = n_tiles;
+        if (na)
+            grid = std::min<uint32_t>(
+                {n_tiles, 2048u, std::max<uint32_t>(256u, n_tiles / 8)});
         // straight-line agg path (kernels.hip): histo[0] int_fast LDS
         // non-nullable no-subs (+ optional terms[1] LDS non-nullable)
         if (na && !ap.devs.empty() && ap.devs[0].kind == AGGD_HISTO &&
@@ -2003,6 +2009,13 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
              (ap.devs[1].kind == AGGD_TERMS && ap.devs[1].lds_slot == 1 &&
               ap.devs[1].nulls_off == 0 && ap.devs[1].offsets_off == 0 &&
               ap.devs[1].n_buckets > 0)))
+            q.agg_fast = 1;
+        // terms-only straight-line path (config5's shape): one LDS terms
+        // agg, non-nullable single-valued ord column, no subs
+        if (na && ap.devs.size() == 1 && ap.devs[0].kind == AGGD_TERMS &&
+            ap.devs[0].lds_slot == 1 && ap.devs[0].n_sub == 0 &&
+            ap.devs[0].nulls_off == 0 && ap.devs[0].offsets_off == 0 &&
+            ap.devs[0].n_buckets > 0)
             q.agg_fast = 1;
         static const bool agg_nt = getenv("QW_AGG_NT") != nullptr;
         q.agg_nt = agg_nt ? 1 : 0;
