@@ -195,6 +195,13 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if world > 1 and "QW_GEN_THREADS" not in os.environ:
+        # torchrun defaults workers to OMP_NUM_THREADS=1, which would make
+        # each rank's native split generation single-threaded (minutes at
+        # 100M docs); give each rank an explicit share of the host cores
+        # (qw_gen_set_threads overrides the OpenMP env pin)
+        os.environ["QW_GEN_THREADS"] = str(
+            max(1, (os.cpu_count() or 8) // world))
 
     import torch
     dist = None

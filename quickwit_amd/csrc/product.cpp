@@ -1991,14 +1991,21 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         bool nb = n_must > 0 || n_must_not > 0;
         bool na = do_aggs && !ap.devs.empty();
         // agg workloads: capped grid so the once-per-workgroup LDS flush
-        // stays cheap (each WG owns several tiles) — and scaled DOWN with
-        // the split size: the flush costs grid x buckets global atomics
-        // regardless of docs, which dominated small (config5-sized) splits.
-        // >=256 WGs x 4 waves still fills the chip's 1024 resident-wave slots.
+        // stays cheap (each WG owns several tiles). For PURE column scans
+        // (no posting decode) the grid also scales DOWN with the split
+        // size — the flush costs grid x buckets global atomics regardless
+        // of docs, which dominated small splits (256 WGs x 4 waves still
+        // fill the chip's 1024 resident-wave slots). Decode-carrying
+        // kernels keep the wide grid: fewer WGs serialize the per-tile
+        // decode loops and cost more than the flush saves (measured,
+        // config5 ablation).
         uint32_t grid = n_tiles;
         if (na)
-            grid = std::min<uint32_t>(
-                {n_tiles, 2048u, std::max<uint32_t>(256u, n_tiles / 8)});
+            grid = fq.terms.empty()
+                       ? std::min<uint32_t>(
+                             {n_tiles, 2048u,
+                              std::max<uint32_t>(256u, n_tiles / 8)})
+                       : std::min<uint32_t>(n_tiles, 2048u);
         // straight-line agg path (kernels.hip): histo[0] int_fast LDS
         // non-nullable no-subs (+ optional terms[1] LDS non-nullable)
         if (na && !ap.devs.empty() && ap.devs[0].kind == AGGD_HISTO &&

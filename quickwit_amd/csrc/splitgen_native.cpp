@@ -580,6 +580,12 @@ int64_t qw_gen_split(int64_t split_ord, int64_t num_docs, int64_t seed,
 
 void qw_gen_free(uint8_t* p) { free(p); }
 
+// explicit worker count (overrides OMP_NUM_THREADS — torchrun pins workers
+// to 1 thread, which would serialize 100M-doc generation)
+void qw_gen_set_threads(int n) {
+    if (n > 0) omp_set_num_threads(n);
+}
+
 int64_t qw_gen_split_to_file(int64_t split_ord, int64_t num_docs, int64_t seed,
                              const char* path) {
     uint8_t* buf = nullptr;

@@ -669,6 +669,7 @@ def _native_gen():
                 ctypes.POINTER(ctypes.POINTER(ctypes.c_uint8)),
                 ctypes.POINTER(ctypes.c_size_t)]
             lib.qw_gen_free.argtypes = [ctypes.POINTER(ctypes.c_uint8)]
+            lib.qw_gen_set_threads.argtypes = [ctypes.c_int]
             _NATIVE_GEN = lib
     return _NATIVE_GEN or None
 
@@ -687,6 +688,10 @@ def generate_split(split_ord: int, num_docs: int, seed: int = 42) -> bytes:
     lib = _native_gen()
     if lib is not None:
         import ctypes
+        import os as _os
+        nt = _os.environ.get("QW_GEN_THREADS")
+        if nt:
+            lib.qw_gen_set_threads(int(nt))
         p = ctypes.POINTER(ctypes.c_uint8)()
         n = ctypes.c_size_t()
         rc = lib.qw_gen_split(split_ord, num_docs, seed,
