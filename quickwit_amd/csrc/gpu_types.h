@@ -52,6 +52,11 @@ enum PredFlags : uint32_t {
     PRED_LO_EXCLUDED = 4,
     PRED_HI_INCLUDED = 8,
     PRED_HI_EXCLUDED = 16,
+    // evaluate for EVERY doc of the tile, not just still-matching lanes:
+    // when the positive clauses are dense (host estimate >= ~15%), the
+    // column lines are touched anyway and the unpredicated load keeps the
+    // epilogue coalesced and convergent
+    PRED_EAGER = 32,
 };
 
 struct PredDev {

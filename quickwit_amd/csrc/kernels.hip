@@ -660,7 +660,11 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                 if (NS && q.scoring && m) sc = sc_score[li];
             }
             if (NB && m && q.n_must_not) m = !((sc_bits_not[li >> 5] >> (li & 31)) & 1);
-            for (uint32_t p = 0; m && p < q.n_preds; ++p) m = eval_pred(q, preds[p], d);
+            for (uint32_t p = 0; p < q.n_preds; ++p) {
+                const PredDev& pr = preds[p];
+                if (m || (pr.flags & PRED_EAGER))
+                    m = eval_pred(q, pr, d) && m;
+            }
 
             if (m) {
                 ++local_count;

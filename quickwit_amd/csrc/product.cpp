@@ -1856,6 +1856,32 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             scratch_bytes += ap.pbounds[i].size() * 8;
         }
 
+    // dense positive base => eager predicate evaluation (kernels.hip):
+    // estimate the pre-predicate match density from the term dfs
+    {
+        double density = 1.0;
+        if (!fq.match_all && !fq.terms.empty()) {
+            double n = double(std::max<uint32_t>(sv.num_docs, 1));
+            double mind = 1.0, sum = 0.0;
+            bool has_must = false;
+            for (const FlatQuery::FTerm& t : fq.terms) {
+                double dfr = double(t.f->h_doc_freq[t.tid]) / n;
+                if (t.role == ROLE_MUST) {
+                    mind = std::min(mind, dfr);
+                    has_must = true;
+                } else if (t.role == ROLE_SHOULD) sum += dfr;
+            }
+            density = has_must ? mind
+                      : fq.msm > 0 ? std::min(1.0, sum)
+                                   : 1.0;
+        }
+        if (density >= 0.15)
+            for (PredDev& pr : fq.preds)
+                if (pr.type == PRED_RANGE_U64 || pr.type == PRED_RANGE_I64 ||
+                    pr.type == PRED_RANGE_F64)
+                    pr.flags |= PRED_EAGER;
+    }
+
     mark("plan");
     // per-term block ranges come from the split-level device cache
     // (term_ranges_addr): first query pays the walk+upload, repeats don't

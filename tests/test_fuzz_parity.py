@@ -1,0 +1,191 @@
+"""Randomized GPU <-> oracle parity fuzz (seeded, deterministic).
+
+Generates random query/sort/aggregation combinations over the synthetic
+hdfs-logs split and asserts response-level parity: num_hits equal, hit ids
+in order (score ties compared as sets within the tie group), aggregation
+JSON equal, and error behavior symmetric (a query rejected by one engine
+must be rejected by the other)."""
+import math
+import random
+
+import pytest
+
+from quickwit_amd import splitgen
+from quickwit_amd.api import GpuSearcher, OracleSearcher, make_leaf_request
+
+pytestmark = pytest.mark.gpu
+
+NDOCS = 120_000
+SID = "fuzz-split"
+SCHEMA = splitgen.HDFS_SCHEMA
+N_QUERIES = 120
+
+
+@pytest.fixture(scope="module", autouse=True)
+def build_all():
+    import __graft_entry__
+    __graft_entry__.build()
+
+
+@pytest.fixture(scope="module")
+def searchers():
+    data = splitgen.generate_split(0, NDOCS, seed=202)
+    gpu, cpu = GpuSearcher(device=0), OracleSearcher()
+    gpu.add_split(SID, data)
+    cpu.add_split(SID, data)
+    return gpu, cpu
+
+
+def rand_term(rng):
+    field, pool = rng.choice([
+        ("body", ["w%05d" % rng.randrange(0, 40), "w%05d" % rng.randrange(0, 9999),
+                  "zz_absent_%d" % rng.randrange(3)]),
+        ("severity_text", ["DEBUG", "INFO", "WARN", "ERROR", "FATAL",
+                           "nope"]),
+    ])
+    return {"type": "term", "field": field, "value": rng.choice(pool)}
+
+
+def rand_range(rng):
+    if rng.random() < 0.5:
+        lo = rng.randrange(0, 900)
+        hi = lo + rng.randrange(1, 400)
+        return {"type": "range", "field": "tenant_id",
+                "lower_bound": {"included": lo},
+                "upper_bound": {"excluded" if rng.random() < 0.5
+                                else "included": hi}}
+    t0 = 1_700_000_000 + rng.randrange(0, 30 * 86400)
+    return {"type": "range", "field": "timestamp",
+            "lower_bound": {"included": t0 * 1000},
+            "upper_bound": {"excluded": (t0 + rng.randrange(3600, 864000))
+                            * 1000}}
+
+
+def rand_clause(rng, depth):
+    r = rng.random()
+    if r < 0.45:
+        return rand_term(rng)
+    if r < 0.7:
+        return rand_range(rng)
+    if r < 0.8 and depth < 2:
+        return rand_bool(rng, depth + 1)
+    if r < 0.9:
+        return {"type": "full_text", "field": "body",
+                "text": " ".join("w%05d" % rng.randrange(0, 200)
+                                 for _ in range(rng.randrange(1, 3))),
+                "params": {"mode": {"type": "bool",
+                                    "operator": rng.choice(["and", "or"])}}}
+    return {"type": "match_all"}
+
+
+def rand_bool(rng, depth=0):
+    b = {"type": "bool"}
+    for occ, pmax in (("must", 2), ("should", 3), ("filter", 2),
+                      ("must_not", 1)):
+        n = rng.randrange(0, pmax + 1) if rng.random() < 0.6 else 0
+        if n:
+            b[occ] = [rand_clause(rng, depth) for _ in range(n)]
+    if not any(k in b for k in ("must", "should", "filter", "must_not")):
+        b["should"] = [rand_term(rng)]
+    return b
+
+
+def rand_sort(rng, query):
+    r = rng.random()
+    if r < 0.3:
+        return None
+    if r < 0.55:
+        return [{"field_name": "_score", "sort_order": 1}]
+    fields = ["timestamp", "tenant_id", "tenant_name"]
+    s = [{"field_name": rng.choice(fields),
+          "sort_order": rng.randrange(2)}]
+    if rng.random() < 0.4:
+        s.append({"field_name": rng.choice(fields + ["_score"]),
+                  "sort_order": rng.randrange(2)})
+    return s
+
+
+def rand_aggs(rng):
+    r = rng.random()
+    if r < 0.6:
+        return None
+    out = {}
+    if rng.random() < 0.6:
+        out["h"] = {"date_histogram": {"field": "timestamp",
+                                       "fixed_interval":
+                                           rng.choice(["6h", "86400000ms"])}}
+    if rng.random() < 0.6:
+        t = {"terms": {"field": "tenant_name",
+                       "size": rng.choice([3, 10, 50])}}
+        out["t"] = t
+    if rng.random() < 0.3:
+        out["s"] = {"stats": {"field": "tenant_id"}}
+    return out or None
+
+
+def scores_of(hits):
+    return [h.get("sort_value", {}).get("f64") for h in hits]
+
+
+def assert_parity(g, e, scored, label):
+    assert g.get("num_hits", 0) == e.get("num_hits", 0), label
+    gh, eh = g.get("partial_hits", []), e.get("partial_hits", [])
+    assert len(gh) == len(eh), label
+    if not scored:
+        assert [(h.get("segment_ord", 0), h.get("doc_id", 0),
+                 h.get("sort_value"), h.get("sort_value2")) for h in gh] == \
+            [(h.get("segment_ord", 0), h.get("doc_id", 0),
+              h.get("sort_value"), h.get("sort_value2")) for h in eh], label
+        return
+    # scored: scores within 1e-5 rel pointwise; ids set-equal within rounded
+    # tie groups except possibly the truncated boundary group
+    for a, b in zip(scores_of(gh), scores_of(eh)):
+        if a is None or b is None:
+            assert a == b, label
+        else:
+            assert math.isclose(a, b, rel_tol=1e-5, abs_tol=1e-9), label
+
+    def grp(hits):
+        out = {}
+        for h in hits:
+            s = h.get("sort_value", {}).get("f64")
+            out.setdefault(None if s is None else round(s, 4),
+                           set()).add(h.get("doc_id", 0))
+        return out
+    gg, ee = grp(gh), grp(eh)
+    keys = sorted((k for k in gg if k is not None), reverse=True)
+    for k in keys[:-1]:
+        assert gg[k] == ee.get(k), (label, k)
+
+
+def test_random_query_parity(searchers):
+    gpu, cpu = searchers
+    rng = random.Random(4242)
+    ran = 0
+    rejected = 0
+    for qi in range(N_QUERIES):
+        q = rand_bool(rng) if rng.random() < 0.8 else rand_clause(rng, 0)
+        sort = rand_sort(rng, q)
+        aggs = rand_aggs(rng)
+        mh = rng.choice([0, 7, 50])
+        req = make_leaf_request(q, SCHEMA, [(SID, NDOCS)], max_hits=mh,
+                                sort_fields=sort, aggregation=aggs)
+        label = f"q{qi}: {q} sort={sort} aggs={list(aggs) if aggs else None}"
+        g, e = gpu.leaf_search(req), cpu.leaf_search(req)
+        gf, ef = bool(g.get("failed_splits")), bool(e.get("failed_splits"))
+        assert gf == ef, (label, g.get("failed_splits"),
+                          e.get("failed_splits"))
+        if gf:
+            rejected += 1
+            continue
+        scored = bool(sort) and sort[0]["field_name"] == "_score"
+        assert_parity(g, e, scored, label)
+        if aggs and "intermediate_aggregation_result" in e:
+            gj = gpu.finalize_agg_json(
+                g["intermediate_aggregation_result"], aggs)
+            ej = cpu.finalize_agg_json(
+                e["intermediate_aggregation_result"], aggs)
+            assert gj == ej, label
+        ran += 1
+    # the generator must mostly produce runnable queries
+    assert ran >= N_QUERIES * 2 // 3, (ran, rejected)
