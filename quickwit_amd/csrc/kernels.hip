@@ -145,28 +145,73 @@ __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t t
     // dependent shuffle chain is the decode critical path
     uint32_t half = lane >> 5;   // which block of the pair this lane works on
     uint32_t sl = lane & 31;     // sub-lane within the half-wave
-    for (uint32_t blk0 = blo + 2 * wave; blk0 < bhi; blk0 += 2 * (TILE_THREADS / 64)) {
+    uint32_t j0 = 4 * sl;
+    uint32_t seg = sl >> 3;
+    uint32_t stride2 = 2 * (TILE_THREADS / 64);
+
+    // 4 consecutive w-bit fields starting at bit `pos` of the 128-bit
+    // window (lo, hi); valid for w <= 16 (pos <= 31, so pos + 4w <= 95)
+    auto ext128 = [](uint64_t lo, uint64_t hi, uint32_t pos,
+                     uint32_t w) -> uint32_t {
+        uint64_t v;
+        if (pos < 64) {
+            v = lo >> pos;
+            if (pos) v |= hi << (64 - pos);
+        } else v = hi >> (pos - 64);
+        return uint32_t(v) & ((1u << w) - 1u);
+    };
+
+    // SOFTWARE PIPELINE: the per-block decode chain is three dependent
+    // global loads (skip entry -> payload words -> fieldnorm gather).
+    // Iteration i consumes the skip entry, segment anchor and gap/tf
+    // windows prefetched during iteration i-1, overlapping the two leading
+    // loads with the previous block's extract/scan/score tail.
+    SkipEntryDev e_pf{};
+    uint64_t g_lo = 0, g_hi = 0, t_lo = 0, t_hi = 0;
+    uint32_t base_pf = 0;
+    auto prefetch = [&](uint32_t blk) {
+        e_pf = skip[blk];
+        const uint32_t* idb = payload + e_pf.word_off;
+        uint32_t w = e_pf.id_bits;
+        base_pf = seg == 0 ? e_pf.first_doc : idb[int(seg) - 4];
+        if (w <= 16) {
+            uint64_t word = (uint64_t(j0) * w) >> 5;
+            g_lo = uint64_t(idb[word]) | (uint64_t(idb[word + 1]) << 32);
+            g_hi = uint64_t(idb[word + 2]) | (uint64_t(idb[word + 3]) << 32);
+        }
+        if (SCORE && e_pf.tf_bits) {
+            const uint32_t* tfb = idb + 2 * ((128u * w + 63u) / 64u);
+            uint32_t tw = e_pf.tf_bits;
+            if (tw <= 16) {
+                uint64_t word = (uint64_t(j0) * tw) >> 5;
+                t_lo = uint64_t(tfb[word]) | (uint64_t(tfb[word + 1]) << 32);
+                t_hi =
+                    uint64_t(tfb[word + 2]) | (uint64_t(tfb[word + 3]) << 32);
+            }
+        }
+    };
+    {
+        uint32_t first_blk = blo + 2 * wave + half;
+        if (first_blk < bhi) prefetch(first_blk);
+    }
+    for (uint32_t blk0 = blo + 2 * wave; blk0 < bhi; blk0 += stride2) {
         uint32_t blk = blk0 + half;
         bool live = blk < bhi;
-        SkipEntryDev e = live ? skip[blk] : SkipEntryDev{};
+        SkipEntryDev e = e_pf;
+        uint64_t glo = g_lo, ghi = g_hi, tlo = t_lo, thi = t_hi;
+        uint32_t abase = base_pf;
+        if (live && blk + stride2 < bhi) prefetch(blk + stride2);
         live = live && !(e.first_doc >= tile_hi || e.last_doc < tile_lo);
         uint32_t g0 = 0, g1 = 0, g2 = 0, g3 = 0;
         const uint32_t* idbase = payload + e.word_off;
         uint32_t w = e.id_bits;
-        uint32_t j0 = 4 * sl;
         if (live) {
-            if (w <= 8) {
-                // 4 fields fit one 64-bit window past any 32-bit alignment
-                uint64_t bitpos = uint64_t(j0) * w;
-                uint32_t sh = uint32_t(bitpos & 31);
-                uint64_t word = bitpos >> 5;
-                uint64_t v = (uint64_t(idbase[word]) |
-                              (uint64_t(idbase[word + 1]) << 32)) >> sh;
-                uint32_t mask = (1u << w) - 1u;
-                g0 = uint32_t(v) & mask;
-                g1 = uint32_t(v >> w) & mask;
-                g2 = uint32_t(v >> (2 * w)) & mask;
-                g3 = uint32_t(v >> (3 * w)) & mask;
+            if (w <= 16) {
+                uint32_t posb = (j0 * w) & 31u;
+                g0 = ext128(glo, ghi, posb, w);
+                g1 = ext128(glo, ghi, posb + w, w);
+                g2 = ext128(glo, ghi, posb + 2 * w, w);
+                g3 = ext128(glo, ghi, posb + 3 * w, w);
             } else {
                 extract_bits_pair(idbase, j0, w, &g0, &g1);
                 extract_bits_pair(idbase, j0 + 2, w, &g2, &g3);
@@ -183,9 +228,7 @@ __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t t
             if ((sl & 7u) >= uint32_t(d)) sum += n;
         }
         if (!live) continue;
-        uint32_t seg = sl >> 3;
-        uint32_t base = seg == 0 ? e.first_doc
-                                 : ((const uint32_t*)idbase)[int(seg) - 4];
+        uint32_t base = abase;  // prefetched segment anchor
         uint32_t doc3 = base + sum;
         uint32_t doc2 = doc3 - g3;
         uint32_t doc1 = doc2 - g2;
@@ -196,17 +239,12 @@ __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t t
             if (e.tf_bits) {
                 const uint32_t* tfbase = idbase + 2 * ((128u * e.id_bits + 63u) / 64u);
                 uint32_t tw = e.tf_bits;
-                if (tw <= 8) {
-                    uint64_t bitpos = uint64_t(j0) * tw;
-                    uint32_t sh = uint32_t(bitpos & 31);
-                    uint64_t word = bitpos >> 5;
-                    uint64_t v = (uint64_t(tfbase[word]) |
-                                  (uint64_t(tfbase[word + 1]) << 32)) >> sh;
-                    uint32_t mask = (1u << tw) - 1u;
-                    tf0 = (uint32_t(v) & mask) + 1u;
-                    tf1 = (uint32_t(v >> tw) & mask) + 1u;
-                    tf2 = (uint32_t(v >> (2 * tw)) & mask) + 1u;
-                    tf3 = (uint32_t(v >> (3 * tw)) & mask) + 1u;
+                if (tw <= 16) {
+                    uint32_t posb = (j0 * tw) & 31u;
+                    tf0 = ext128(tlo, thi, posb, tw) + 1u;
+                    tf1 = ext128(tlo, thi, posb + tw, tw) + 1u;
+                    tf2 = ext128(tlo, thi, posb + 2 * tw, tw) + 1u;
+                    tf3 = ext128(tlo, thi, posb + 3 * tw, tw) + 1u;
                 } else {
                     extract_bits_pair(tfbase, j0, tw, &tf0, &tf1);
                     extract_bits_pair(tfbase, j0 + 2, tw, &tf2, &tf3);
