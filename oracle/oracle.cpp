@@ -1200,7 +1200,10 @@ int qw_oracle_leaf_search(qw_oracle_ctx* ctx, const uint8_t* req_pb, size_t req_
     using namespace qw::oracle;
     try {
         pb::LeafSearchRequest lreq = pb::LeafSearchRequest::decode(req_pb, req_len);
-        const pb::SearchRequest& req = lreq.search_request;
+        pb::SearchRequest req = lreq.search_request;
+        // count-only rewrite (leaf.rs:977-990, mirrored by the product's
+        // per-split rewrite): no hits requested -> sort fields are dead
+        if (req.max_hits == 0) req.sort_fields.clear();
         // one doc mapper (schema json) per index; r1 handles one index per call
         if (lreq.doc_mappers.empty()) throw std::runtime_error("missing doc_mapper");
         Schema schema = Schema::parse(lreq.doc_mappers[0]);
