@@ -1023,24 +1023,33 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
             }
             __syncthreads();
             uint64_t base = *sc_cand_base;
-            for (uint32_t d = tile_lo + threadIdx.x; d < tile_hi; d += TILE_THREADS) {
-                uint32_t li = d - tile_lo;
-                uint32_t bits = sc_bits_m[li >> 5];
-                if (!((bits >> (li & 31)) & 1)) continue;
-                uint64_t rank = sc_word_pref[li >> 5] +
-                                __popc(bits & ((1u << (li & 31)) - 1u));
-                float sc = (NS && q.scoring) ? sc_score[li] : 0.f;
-                if (base + rank >= q.cand_cap) continue;
-                if (q.wide_cand) {
-                    uint64_t key = wide_sort_key(q, d, sc);
-                    cand[2 * (base + rank)] = key;
-                    cand[2 * (base + rank) + 1] =
-                        (uint64_t(__float_as_uint(sc)) << 32) | d;
-                } else {
-                    uint32_t kh = q.scoring ? f32_sortable(sc) : 0u;
-                    if (q.sort_asc) kh = ~kh;
-                    uint32_t kl = q.sort_asc ? ~d : d;
-                    cand[base + rank] = (uint64_t(kh) << 32) | kl;
+            // one 32-doc word per thread, iterating SET bits only (a full
+            // grid-stride re-walk of the tile cost ~as much as the decode
+            // itself at ~25% match density); per-thread writes are
+            // consecutive addresses, coalescing at cache-line granularity
+            {
+                uint32_t wd = threadIdx.x;
+                uint32_t bits = sc_bits_m[wd];
+                uint64_t rank = sc_word_pref[wd];
+                while (bits) {
+                    uint32_t b = __ffs(bits) - 1u;
+                    bits &= bits - 1u;
+                    uint32_t li = wd * 32u + b;
+                    uint32_t d = tile_lo + li;
+                    float sc = (NS && q.scoring) ? sc_score[li] : 0.f;
+                    uint64_t pos = base + rank++;
+                    if (pos >= q.cand_cap) break;
+                    if (q.wide_cand) {
+                        uint64_t key = wide_sort_key(q, d, sc);
+                        cand[2 * pos] = key;
+                        cand[2 * pos + 1] =
+                            (uint64_t(__float_as_uint(sc)) << 32) | d;
+                    } else {
+                        uint32_t kh = q.scoring ? f32_sortable(sc) : 0u;
+                        if (q.sort_asc) kh = ~kh;
+                        uint32_t kl = q.sort_asc ? ~d : d;
+                        cand[pos] = (uint64_t(kh) << 32) | kl;
+                    }
                 }
             }
         } else if (do_count) {
