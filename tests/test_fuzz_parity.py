@@ -202,3 +202,54 @@ def test_random_query_parity(searchers):
         ran += 1
     # the generator must mostly produce runnable queries
     assert ran >= N_QUERIES * 2 // 3, (ran, rejected)
+
+
+@pytest.fixture(scope="module")
+def multiseg_searchers():
+    a = splitgen.generate_split(0, 70_000, seed=303)
+    b = splitgen.generate_split(1, 50_000, seed=303)
+    multi = splitgen.concat_segments([a, b], SID)
+    gpu, cpu = GpuSearcher(device=0), OracleSearcher()
+    gpu.add_split(SID, multi)
+    cpu.add_split(SID, multi)
+    return gpu, cpu
+
+
+def test_random_query_parity_multi_segment(multiseg_searchers):
+    """The same randomized battery over a 2-segment QWA2 split: per-segment
+    collection, cross-segment merge and (split, segment, doc) tie-breaks."""
+    gpu, cpu = multiseg_searchers
+    rng = random.Random(777)
+    ran = 0
+    for qi in range(60):
+        q = rand_bool(rng) if rng.random() < 0.8 else rand_clause(rng, 0)
+        sort = rand_sort(rng, q)
+        aggs = rand_aggs(rng)
+        mh = rng.choice([0, 9, 40])
+        req = make_leaf_request(q, SCHEMA, [(SID, NDOCS)], max_hits=mh,
+                                sort_fields=sort, aggregation=aggs)
+        label = f"mseg q{qi}: {q} sort={sort}"
+        g, e = gpu.leaf_search(req), cpu.leaf_search(req)
+        gf, ef = bool(g.get("failed_splits")), bool(e.get("failed_splits"))
+        if gf and not ef:
+            continue  # declared product capability gap (see above)
+        assert gf == ef, (label, g.get("failed_splits"),
+                          e.get("failed_splits"))
+        if gf:
+            continue
+        scored = bool(sort) and sort[0]["field_name"] == "_score" and mh > 0
+        assert_parity(g, e, scored, label)
+        # segment ords must agree hit-by-hit on unscored orders
+        if not scored:
+            assert [h.get("segment_ord", 0)
+                    for h in g.get("partial_hits", [])] == \
+                [h.get("segment_ord", 0) for h in e.get("partial_hits", [])], \
+                label
+        if aggs and "intermediate_aggregation_result" in e:
+            gj = gpu.finalize_agg_json(
+                g["intermediate_aggregation_result"], aggs)
+            ej = cpu.finalize_agg_json(
+                e["intermediate_aggregation_result"], aggs)
+            assert gj == ej, label
+        ran += 1
+    assert ran >= 35, ran
