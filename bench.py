@@ -217,11 +217,18 @@ def main():
     from quickwit_amd.api import GpuSearcher, OracleSearcher, make_leaf_request
     from quickwit_amd.merge import distributed_merge
 
+    # QW_BENCH_ENGINE=oracle: CPU-only dry run of the FULL bench path (incl.
+    # the N>1 packed exchange over gloo) for tests — never the measured
+    # engine (tests/test_bench_distributed.py); the product run requires
+    # the GPU and fails loudly without one.
+    engine = os.environ.get("QW_BENCH_ENGINE", "gpu")
+
     wl = make_workload(args.workload, args.docs, args.max_hits)
 
     t_gen = time.perf_counter()
     docs_per_split = args.docs // args.splits
-    searcher = GpuSearcher(device=local_rank)
+    searcher = (OracleSearcher() if engine == "oracle"
+                else GpuSearcher(device=local_rank))
     split_set = []
     split_bytes = None
     for si in range(args.splits):
@@ -254,7 +261,8 @@ def main():
 
     for _ in range(args.warmup):
         last = one_step()
-    searcher.kernel_stats_reset()
+    if engine != "oracle":
+        searcher.kernel_stats_reset()
 
     if dist:
         dist.barrier()
@@ -287,7 +295,8 @@ def main():
     p50_ms = sorted(step_times)[len(step_times) // 2] * 1e3
 
     # roofline of the dominant kernel
-    kms, launches = searcher.kernel_stats(wl["kernel"])
+    kms, launches = (searcher.kernel_stats(wl["kernel"])
+                     if engine != "oracle" else (0.0, 0))
     roofline = None
     if launches:
         ms_per_launch = kms / launches
@@ -368,8 +377,10 @@ def main():
             "num_hits_per_split": per_split_hits,
             "parallelism": f"split-dp{world}",
             "gen_seconds": round(gen_s, 1),
-            "hbm_used_gb": round(searcher.memory_stats()[0] / 2**30, 2),
-            "hbm_budget_gb": round(searcher.memory_stats()[1] / 2**30, 1),
+            "hbm_used_gb": (round(searcher.memory_stats()[0] / 2**30, 2)
+                            if engine != "oracle" else None),
+            "hbm_budget_gb": (round(searcher.memory_stats()[1] / 2**30, 1)
+                              if engine != "oracle" else None),
         },
         "roofline": roofline,
         "cpu_baseline": cpu_baseline,
