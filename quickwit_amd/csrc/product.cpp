@@ -333,7 +333,7 @@ struct qw_ctx {
     hipStream_t stream = nullptr;
     hipEvent_t ev_start = nullptr, ev_stop = nullptr;
     qw::DevBuf d_scratch, d_results, d_survivors, d_cand2;
-    qw::PinnedBuf h_scratch, h_counts, h_surv, h_topk;  // pinned staging
+    qw::PinnedBuf h_scratch, h_counts, h_surv, h_topk, h_agg;  // pinned staging
     std::map<std::string, qw::KernelTimer> timers;
     // HBM accounting (SearchPermitProvider memory-budget analog,
     // search_permit_provider.rs:43-110): split images + scratch/result
@@ -2091,6 +2091,14 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         if (collect)
             HIP_CHECK(hipMemcpyAsync(pc, ctx->d_results.p + r_cand_count, 4,
                                      hipMemcpyDeviceToHost, ctx->stream));
+        if (do_aggs && ap.out_bytes) {
+            // aggregation results ride the same async batch + sync (a
+            // blocking copy after hit building cost ~40 us per split)
+            ctx->h_agg.ensure(ap.out_bytes + 64);
+            HIP_CHECK(hipMemcpyAsync(ctx->h_agg.p, ctx->d_results.p + r_agg,
+                                     ap.out_bytes, hipMemcpyDeviceToHost,
+                                     ctx->stream));
+        }
         HIP_CHECK(hipStreamSynchronize(ctx->stream));
         uint32_t cand_n = collect ? *pc : 0;
         mark("main+pass0+sync");
@@ -2360,11 +2368,9 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
 
     // ---- aggregation download + assembly (QAGG1 intermediate)
     if (do_aggs) {
-        std::vector<uint8_t> agg_out(ap.out_bytes);
-        if (need_kernel && ap.out_bytes) {
-            HIP_CHECK(hipMemcpy(agg_out.data(), ctx->d_results.p + r_agg, ap.out_bytes,
-                                hipMemcpyDeviceToHost));
-        }
+        // downloaded asynchronously with the counts batch above (pinned)
+        ctx->h_agg.ensure(ap.out_bytes + 64);
+        const uint8_t* agg_out_p = ctx->h_agg.p;
         for (size_t i = 0; i < ap.defs.size(); ++i) {
             const AggDef& d = ap.defs[i];
             const AggDev& a = ap.devs[i];
@@ -2372,14 +2378,14 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             AggResult r;
             r.name = d.name;
             for (auto& s : d.sub) r.sub_names.push_back(s.name);
-            const uint8_t* base = agg_out.data() + (a.counts_out - r_agg);
+            const uint8_t* base = agg_out_p + (a.counts_out - r_agg);
             const uint64_t* counts = (const uint64_t*)base;
             if (d.kind == AggDef::TERMS || d.kind == AggDef::CARDINALITY) {
                 r.kind = 3;
                 if (a.n_buckets) {
                     if (a.nulls_off || a.offsets_off) {
                         uint64_t m = 0;
-                        memcpy(&m, agg_out.data() + (a.matched_out - r_agg), 8);
+                        memcpy(&m, agg_out_p + (a.matched_out - r_agg), 8);
                         r.terms_matched_docs = m;
                     } else {
                         // non-nullable column: every matched doc has a value
@@ -2403,7 +2409,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                         std::sort(r.term_counts.begin(), r.term_counts.end());
                     } else {
                         const uint8_t* subs =
-                            agg_out.data() + (a.sub_out - r_agg);
+                            agg_out_p + (a.sub_out - r_agg);
                         for (uint32_t o = 0; o < a.n_buckets; ++o)
                             if (counts[o]) {
                                 r.term_counts.emplace_back(f->dict_entry(o),
@@ -2510,7 +2516,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             } else if (d.kind == AggDef::METRIC) {
                 r.kind = 5;
                 if (a.n_buckets) {
-                    const uint8_t* slot = agg_out.data() + (a.counts_out - r_agg);
+                    const uint8_t* slot = agg_out_p + (a.counts_out - r_agg);
                     StatsPayload sp2;
                     uint64_t mn, mx;
                     memcpy(&sp2.count, slot, 8);
@@ -2538,10 +2544,10 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                 for (auto& sdef : d.sub)
                     r.sub_kinds.push_back(
                         sdef.kind == MetricAgg::PERCENTILES ? 1 : 0);
-                const uint8_t* subs = agg_out.data() + (a.sub_out - r_agg);
+                const uint8_t* subs = agg_out_p + (a.sub_out - r_agg);
                 const uint64_t* pwords =
                     a.p_si != 0xFF
-                        ? (const uint64_t*)(agg_out.data() + (a.p_out - r_agg))
+                        ? (const uint64_t*)(agg_out_p + (a.p_out - r_agg))
                         : nullptr;
                 for (uint32_t bi = 0; bi < a.n_buckets; ++bi) {
                     if (!counts[bi]) continue;
