@@ -2187,7 +2187,10 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             std::vector<uint32_t> hist(TOPK_BINS);
             HIP_CHECK(hipEventRecord(ctx->ev_start, ctx->stream));
             while (true) {
-                if (survivors <= std::max<uint64_t>(4 * K, 65536) || prefix_bits >= 48) {
+                // refine until the band is small: the D2H + host sort of
+                // the survivors is on the critical path (a 65k-u64 band
+                // cost ~90 us at K=1000; one more 4096-bin pass is ~20 us)
+                if (survivors <= std::max<uint64_t>(4 * K, 16384) || prefix_bits >= 48) {
                     if (prefix_bits == 0) floor_key = 0;
                     break;
                 }
@@ -2299,6 +2302,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             // (growing K) until the page is full or the band is exhausted
             for (;;) {
                 out.hits.clear();
+                out.hits.reserve(top_keys.size() / (wide ? 2 : 1));
                 if (wide) {
                     // survivors carry (selection key, score|doc); the exact
                     // reference order ((sort_value, sort_value2, GlobalDocId),
