@@ -1735,3 +1735,33 @@ def test_terms_order_by_subagg_on_gpu(searchers):
         ej = cpu.finalize_agg_json(e["intermediate_aggregation_result"], aggs)
         assert [b["key"] for b in gj["t"]["buckets"]] == \
             [b["key"] for b in ej["t"]["buckets"]], order
+
+
+# ------------------------------------------------ per-split failure as data
+def test_per_split_failure_is_data_not_exception():
+    """One split failing must not fail the call: the reference reports it
+    inside failed_splits (leaf.rs:2143-2148) and the others still answer.
+    Here: a phrase query over one split WITH positions and one WITHOUT."""
+    schema = {"timestamp_field": None, "fields": [
+        {"name": "body", "type": "text", "tokenizer": "default",
+         "record": "position", "fieldnorms": True}]}
+    wa = splitgen.SplitWriter(schema, "sa", store_docs=False)
+    wa.add_documents([{"body": "alpha beta"}, {"body": "beta alpha"}])
+    schema_nf = {"timestamp_field": None, "fields": [
+        {"name": "body", "type": "text", "tokenizer": "default",
+         "record": "freq", "fieldnorms": True}]}
+    wb = splitgen.SplitWriter(schema_nf, "sb", store_docs=False)
+    wb.add_documents([{"body": "alpha beta"}])
+    gpu = GpuSearcher(device=0)
+    gpu.add_split("sa", wa.finalize())
+    gpu.add_split("sb", wb.finalize())
+    q = {"type": "full_text", "field": "body", "text": "alpha beta",
+         "params": {"mode": {"type": "phrase"}}}
+    r = gpu.leaf_search(make_leaf_request(q, schema, [("sa", 2), ("sb", 1)],
+                                          max_hits=5))
+    assert r["num_hits"] == 1  # sa's "alpha beta" doc
+    fails = r.get("failed_splits", [])
+    assert len(fails) == 1 and fails[0]["split_id"] == "sb"
+    assert "position" in fails[0]["error"]
+    assert r.get("num_successful_splits", 0) == 1
+    assert r.get("num_attempted_splits", 0) == 2
