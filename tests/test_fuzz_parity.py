@@ -1,11 +1,14 @@
 """Randomized GPU <-> oracle parity fuzz (seeded, deterministic).
 
+QW_FUZZ_N scales the single-split battery (soak runs).
+
 Generates random query/sort/aggregation combinations over the synthetic
 hdfs-logs split and asserts response-level parity: num_hits equal, hit ids
 in order (score ties compared as sets within the tie group), aggregation
 JSON equal, and error behavior symmetric (a query rejected by one engine
 must be rejected by the other)."""
 import math
+import os
 import random
 
 import pytest
@@ -18,7 +21,7 @@ pytestmark = pytest.mark.gpu
 NDOCS = 120_000
 SID = "fuzz-split"
 SCHEMA = splitgen.HDFS_SCHEMA
-N_QUERIES = 120
+N_QUERIES = int(os.environ.get("QW_FUZZ_N", "120"))
 
 
 @pytest.fixture(scope="module", autouse=True)
