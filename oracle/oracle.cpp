@@ -171,29 +171,59 @@ static Match subtract(Match a, const Match& b, uint32_t num_docs) {
 }
 
 // k-way union; keeps docs matched by >= msm clauses; sums scores of matching
-// scoring clauses (tantivy BufferedUnionScorer semantics)
+// scoring clauses (tantivy BufferedUnionScorer semantics). match_all
+// clauses count toward msm for EVERY doc and contribute no score (our
+// restatement scores match_all as 0, like the unscored AllQuery leg).
+static Match union_over(const std::vector<const Match*>& ms, size_t msm);
+
 static Match union_n(const std::vector<Match>& ms, size_t msm, uint32_t num_docs) {
-    (void)num_docs;
+    if (msm == 0) msm = 1;
+    size_t n_all = 0;
+    std::vector<const Match*> rest;
+    for (auto& m : ms) {
+        if (m.all) ++n_all;
+        else rest.push_back(&m);
+    }
+    if (n_all == 0) return union_over(rest, msm);
+    if (n_all >= msm) {
+        // every doc matches; keep per-doc scores of the non-all clauses
+        bool scored = false;
+        for (auto* m : rest) scored |= !m->scores.empty();
+        Match r;
+        if (!scored) {
+            r.all = true;
+            return r;
+        }
+        Match ru = union_over(rest, 1);
+        r.docs.resize(num_docs);
+        for (uint32_t d = 0; d < num_docs; ++d) r.docs[d] = d;
+        r.scores.assign(num_docs, 0.f);
+        for (size_t i = 0; i < ru.docs.size(); ++i)
+            r.scores[ru.docs[i]] =
+                ru.scores.empty() ? 0.f : ru.scores[i];
+        return r;
+    }
+    return union_over(rest, msm - n_all);
+}
+
+static Match union_over(const std::vector<const Match*>& msp, size_t msm) {
     Match r;
     if (msm == 0) msm = 1;
-    for (auto& m : ms)
-        if (m.all)
-            throw std::runtime_error("union over a match_all clause not supported here");
-    size_t k = ms.size();
+    size_t k = msp.size();
     std::vector<size_t> idx(k, 0);
     bool scored = false;
-    for (auto& m : ms) scored |= !m.scores.empty();
+    for (auto* m : msp) scored |= !m->scores.empty();
     while (true) {
         uint32_t best = UINT32_MAX;
         for (size_t c = 0; c < k; ++c) {
-            const Match& m = ms[c];
+            const Match& m = *msp[c];
             if (idx[c] < m.docs.size()) best = std::min(best, m.docs[idx[c]]);
         }
         if (best == UINT32_MAX) break;
         size_t cnt = 0;
         float sc = 0;
         for (size_t c = 0; c < k; ++c) {
-            const Match& m = ms[c];
+            const Match& m = *msp[c];
             if (idx[c] < m.docs.size() && m.docs[idx[c]] == best) {
                 ++cnt;
                 if (!m.scores.empty()) sc += m.scores[idx[c]];
@@ -516,20 +546,8 @@ struct SplitSearcher {
             }
         } else {
             size_t msm = n.minimum_should_match < 0 ? 1 : size_t(n.minimum_should_match);
-            bool any_all = false;
-            for (auto& s : shoulds) any_all |= s.all;
-            if (any_all) {
-                if (shoulds.size() == 1) base = shoulds[0];
-                else {
-                    // union with a match_all clause = match_all (+ scores of
-                    // the others); represent as all + no scores unless scoring
-                    base.all = true;
-                    if (scoring) throw std::runtime_error(
-                        "scored union with match_all clause not supported");
-                }
-            } else {
-                base = union_n(shoulds, msm, sv.num_docs);
-            }
+            base = union_n(shoulds, msm, sv.num_docs);  // handles match_all
+                                                        // clauses (see above)
             if (shoulds.empty()) {
                 base = Match{};
                 // must_not with no positive clause: implicit match_all
