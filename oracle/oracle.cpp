@@ -534,6 +534,14 @@ struct SplitSearcher {
                     // add should scores for docs already in base (merge-join)
                     Match su = union_n(shoulds, 1, sv.num_docs);
                     if (!su.scores.empty()) {
+                        if (base.all) {
+                            // materialize: an all-base (match_all filters)
+                            // has no doc list to carry scores on
+                            base.all = false;
+                            base.docs.resize(sv.num_docs);
+                            for (uint32_t d = 0; d < sv.num_docs; ++d)
+                                base.docs[d] = d;
+                        }
                         if (base.scores.empty()) base.scores.resize(base.docs.size(), 0.f);
                         size_t j = 0;
                         for (size_t i = 0; i < base.docs.size(); ++i) {
