@@ -1836,8 +1836,7 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
     size_t r_cand_count = r_tile_counts + size_t(n_tiles) * 4;
     r_cand_count = (r_cand_count + 63) & ~size_t(63);
     size_t r_hist = r_cand_count + 64;  // cand_count u32 + survivors count u32
-    size_t r_hist512 = r_hist + TOPK_BINS * 4;  // fused selection pass 0
-    size_t r_agg = r_hist512 + TOPK_FUSED_BINS * 4;
+    size_t r_agg = r_hist + TOPK_BINS * 4;
     r_agg = (r_agg + 63) & ~size_t(63);
 
     if (do_aggs)
@@ -2007,9 +2006,6 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         q.cand_off = r_cand;
         q.cand_cap = cand_cap;
         q.hist_off = r_hist;
-        // fused selection pass 0 (kernels.hip candidate-append): not for
-        // search_after (the band is re-compacted and re-histogrammed)
-        q.hist512_off = (collect && !after) ? r_hist512 : 0;
 
         const char* kname = !fq.terms.empty() ? "union_bm25"
                             : do_aggs         ? "column_agg"
@@ -2062,13 +2058,25 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
         HIP_CHECK(hipEventRecord(ctx->ev_stop, ctx->stream));
         HIP_CHECK(hipGetLastError());
 
-        // the coarse selection pass 0 is FUSED into the main kernel's
-        // candidate writes (TOPK_FUSED_BINS, kernels.hip): just ride the
-        // async D2H batch — no extra kernel pass over the candidate array
+        // enqueue top-K histogram pass 0 behind the main kernel (count read
+        // device-side) so its result arrives with the same synchronize
         bool hist0_valid = false;
         if (collect && !after) {
-            HIP_CHECK(hipMemcpyAsync(hist0, ctx->d_results.p + r_hist512,
-                                     TOPK_FUSED_BINS * 4,
+            uint64_t* d_cand0 = (uint64_t*)(ctx->d_results.p + r_cand);
+            uint32_t* d_hist0 = (uint32_t*)(ctx->d_results.p + r_hist);
+            uint32_t* d_n0 = (uint32_t*)(ctx->d_results.p + r_cand_count);
+            HIP_CHECK(hipMemsetAsync(d_hist0, 0, TOPK_BINS * 4, ctx->stream));
+            // candidate count is device-side only at this point: size the
+            // grid for the worst case (every doc a candidate)
+            uint32_t hgrid =
+                std::min<uint32_t>(512, (sv.num_docs + 4095) / 4096);
+            if (wide)
+                hipLaunchKernelGGL(k_cand_hist_w, dim3(hgrid), dim3(256), 0,
+                                   ctx->stream, d_cand0, d_n0, 0ull, 0u, d_hist0);
+            else
+                hipLaunchKernelGGL(k_cand_hist, dim3(hgrid), dim3(256), 0,
+                                   ctx->stream, d_cand0, d_n0, 0ull, 0u, d_hist0);
+            HIP_CHECK(hipMemcpyAsync(hist0, d_hist0, TOPK_BINS * 4,
                                      hipMemcpyDeviceToHost, ctx->stream));
             hist0_valid = true;
         }
