@@ -19,9 +19,6 @@ constexpr uint32_t TILE_THREADS = 256; // 4 waves
 constexpr uint32_t KTAB_LDS_MAX = 4;
 constexpr uint32_t AGG_LDS_BUCKETS = 2048;
 constexpr uint32_t TOPK_BINS = 4096;
-// coarse top-K histogram fused into the main kernel's candidate writes
-// (level-0 of the selection walk; 9 bits of the sort key)
-constexpr uint32_t TOPK_FUSED_BINS = 512;
 
 // role of a term in the flattened boolean
 enum TermRole : uint32_t { ROLE_SHOULD = 0, ROLE_MUST = 1, ROLE_MUST_NOT = 2 };
@@ -203,9 +200,6 @@ struct QueryDev {
     uint64_t cand_off;         // {u32 key, u32 doc}[cap]
     uint64_t cand_cap;
     uint64_t hist_off;         // u32[TOPK_BINS] histogram of candidate keys
-    uint64_t hist512_off;      // u32[TOPK_FUSED_BINS]: coarse histogram
-                               // accumulated during candidate writes (the
-                               // fused selection pass 0); 0 = disabled
 };
 
 }  // namespace qw

@@ -104,9 +104,7 @@ struct SmemMap {
     static constexpr uint32_t bits_end = bits_not_off + (NB ? TILE_DOCS / 8 : 0);
     static constexpr uint32_t bits_m_off = bits_end;               // u32[256]
     static constexpr uint32_t word_pref_off = bits_m_off + (NC ? TILE_DOCS / 8 : 0);
-    static constexpr uint32_t hist512_off = word_pref_off + (NC ? TILE_DOCS / 8 : 0);
-    static constexpr uint32_t coll_end =
-        hist512_off + (NC ? TOPK_FUSED_BINS * 4 : 0);
+    static constexpr uint32_t coll_end = word_pref_off + (NC ? TILE_DOCS / 8 : 0);
     static constexpr uint32_t agg_hist_off = coll_end;             // u32[2048] x2
     static constexpr uint32_t agg_terms_off = agg_hist_off + (NA ? AGG_LDS_BUCKETS * 4 : 0);
     static constexpr uint32_t agg_end = agg_terms_off + (NA ? AGG_LDS_BUCKETS * 4 : 0);
@@ -116,8 +114,7 @@ struct SmemMap {
 
 constexpr uint32_t tile_lds_bytes(bool ns, bool nb, bool na, bool nc) {
     return (ns ? TILE_DOCS * 4 : 0) + (nb ? 2 * TILE_DOCS / 8 : 0) +
-           (nc ? 2 * TILE_DOCS / 8 + TOPK_FUSED_BINS * 4 : 0) +
-           (na ? 2 * AGG_LDS_BUCKETS * 4 : 0) + 36;
+           (nc ? 2 * TILE_DOCS / 8 : 0) + (na ? 2 * AGG_LDS_BUCKETS * 4 : 0) + 36;
 }
 
 // decode every block of `t` overlapping the tile; accumulate into score[] /
@@ -503,7 +500,6 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
     uint32_t* sc_bits_not = (uint32_t*)(smem + M::bits_not_off);
     uint32_t* sc_bits_m = (uint32_t*)(smem + M::bits_m_off);
     uint32_t* sc_word_pref = (uint32_t*)(smem + M::word_pref_off);
-    uint32_t* sc_hist512 = (uint32_t*)(smem + M::hist512_off);
     uint32_t* sc_agg_hist = (uint32_t*)(smem + M::agg_hist_off);
     uint32_t* sc_agg_terms = (uint32_t*)(smem + M::agg_terms_off);
     uint32_t* sc_agg_matched = (uint32_t*)(smem + M::tail_off);
@@ -600,14 +596,9 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                 sc_bits_acc[i] = 0;
                 sc_bits_not[i] = 0;
             }
-        if (NC) {
+        if (NC)
             for (uint32_t i = threadIdx.x; i < TILE_DOCS / 32; i += TILE_THREADS)
                 sc_bits_m[i] = 0;
-            if (q.hist512_off)
-                for (uint32_t i = threadIdx.x; i < TOPK_FUSED_BINS;
-                     i += TILE_THREADS)
-                    sc_hist512[i] = 0;
-        }
         __syncthreads();
 
         // ---- should terms: score / count union
@@ -1048,9 +1039,8 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                     float sc = (NS && q.scoring) ? sc_score[li] : 0.f;
                     uint64_t pos = base + rank++;
                     if (pos >= q.cand_cap) break;
-                    uint64_t key;
                     if (q.wide_cand) {
-                        key = wide_sort_key(q, d, sc);
+                        uint64_t key = wide_sort_key(q, d, sc);
                         cand[2 * pos] = key;
                         cand[2 * pos + 1] =
                             (uint64_t(__float_as_uint(sc)) << 32) | d;
@@ -1058,27 +1048,8 @@ __global__ void __launch_bounds__(TILE_THREADS) k_leaf_tile_t(QueryDev q,
                         uint32_t kh = q.scoring ? f32_sortable(sc) : 0u;
                         if (q.sort_asc) kh = ~kh;
                         uint32_t kl = q.sort_asc ? ~d : d;
-                        key = (uint64_t(kh) << 32) | kl;
-                        cand[pos] = key;
+                        cand[pos] = (uint64_t(kh) << 32) | kl;
                     }
-                    // fused selection pass 0: coarse 9-bit key histogram
-                    if (q.hist512_off)
-                        atomicAdd(&sc_hist512[uint32_t(key >> 55)], 1u);
-                }
-            }
-            if (q.hist512_off) {
-                __syncthreads();  // all LDS hist adds visible
-                // per-tile flush, nonzero bins only (scores concentrate:
-                // ~tens of bins per tile), start rotated per WG so
-                // concurrent tiles don't serialize on the same words
-                uint32_t* gh = (uint32_t*)(q.results + q.hist512_off);
-                uint32_t rot = (blockIdx.x * 37u) % TOPK_FUSED_BINS;
-                for (uint32_t i0 = threadIdx.x; i0 < TOPK_FUSED_BINS;
-                     i0 += TILE_THREADS) {
-                    uint32_t i = i0 + rot;
-                    if (i >= TOPK_FUSED_BINS) i -= TOPK_FUSED_BINS;
-                    uint32_t v = sc_hist512[i];
-                    if (v) atomicAdd(&gh[i], v);
                 }
             }
         } else if (do_count) {

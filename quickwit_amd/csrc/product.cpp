@@ -2194,13 +2194,9 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     if (prefix_bits == 0) floor_key = 0;
                     break;
                 }
-                // level width: the fused pass-0 histogram is 9 bits wide
-                // (TOPK_FUSED_BINS), refinement levels 12 (TOPK_BINS)
-                uint32_t lvl_bits = 12;
                 if (prefix_bits == 0 && hist0_valid) {
-                    lvl_bits = 9;
                     memcpy(hist.data(), hist0,
-                           TOPK_FUSED_BINS * 4);  // fused with the main pass
+                           TOPK_BINS * 4);  // prefetched with the main sync
                 } else {
                     HIP_CHECK(hipMemsetAsync(d_hist, 0, TOPK_BINS * 4, ctx->stream));
                     // grid small enough that the per-WG 4096-bin LDS flush
@@ -2222,20 +2218,20 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
                     HIP_CHECK(hipStreamSynchronize(ctx->stream));
                 }
                 uint64_t cum = 0;
-                int b = int(1u << lvl_bits) - 1;
+                int b = TOPK_BINS - 1;
                 for (; b >= 0; --b) {
                     if (cum + hist[b] >= Krem) break;
                     cum += hist[b];
                 }
                 if (b < 0) b = 0;
-                uint32_t shift = 64 - prefix_bits - lvl_bits;
+                uint32_t shift = 64 - prefix_bits - 12;
                 floor_key = (prefix_bits ? (prefix << (64 - prefix_bits)) : 0) |
                             (uint64_t(uint32_t(b)) << shift);
                 above += cum;
                 survivors = above + hist[b];
                 Krem = Krem - cum;
-                prefix = (prefix << lvl_bits) | uint64_t(uint32_t(b));
-                prefix_bits += lvl_bits;
+                prefix = (prefix << 12) | uint64_t(uint32_t(b));
+                prefix_bits += 12;
             }
             // #keys >= floor_key is known EXACTLY from the histogram walk, so
             // no round trip for the compact count: one async D2H of survivors
