@@ -256,3 +256,59 @@ def test_random_query_parity_multi_segment(multiseg_searchers):
             assert gj == ej, label
         ran += 1
     assert ran >= 35, ran
+
+
+def test_random_phrase_parity():
+    """Random phrase battery over a positions-enabled corpus: the phrase
+    kernel (k_phrase_bitmap) vs the oracle (itself pinned against a python
+    brute force in test_phrase.py), alone and inside boolean contexts."""
+    schema = {"timestamp_field": None, "fields": [
+        {"name": "body", "type": "text", "tokenizer": "default",
+         "record": "position", "fieldnorms": True},
+        {"name": "num", "type": "u64", "fast": True}]}
+    rng = random.Random(int(os.environ.get("QW_FUZZ_SEED", "99")))
+    vocab = ["aa", "bb", "cc", "dd", "ee"]
+    docs = [{"body": " ".join(rng.choice(vocab)
+                              for _ in range(rng.randint(2, 15))),
+             "num": rng.randrange(100)} for _ in range(2500)]
+    w = splitgen.SplitWriter(schema, "pf", store_docs=False)
+    w.add_documents(docs)
+    data = w.finalize()
+    gpu, cpu = GpuSearcher(device=0), OracleSearcher()
+    gpu.add_split("pf", data)
+    cpu.add_split("pf", data)
+
+    def phrase(toks):
+        return {"type": "full_text", "field": "body", "text": " ".join(toks),
+                "params": {"mode": {"type": "phrase"}}}
+
+    for qi in range(40):
+        toks = [rng.choice(vocab + (["zz"] if rng.random() < 0.15 else []))
+                for _ in range(rng.randint(2, 5))]
+        p = phrase(toks)
+        r = rng.random()
+        if r < 0.4:
+            q = p
+        elif r < 0.65:
+            q = {"type": "bool", "filter": [p],
+                 "must": [{"type": "range", "field": "num",
+                           "lower_bound": {"included": rng.randrange(50)}}]}
+        elif r < 0.85:
+            q = {"type": "bool",
+                 "must": [{"type": "term", "field": "body",
+                           "value": rng.choice(vocab)}],
+                 "must_not": [p]}
+        else:
+            q = {"type": "bool", "filter": [p, phrase(
+                [rng.choice(vocab), rng.choice(vocab)])]}
+        req = make_leaf_request(q, schema, [("pf", len(docs))],
+                                max_hits=len(docs))
+        g, e = gpu.leaf_search(req), cpu.leaf_search(req)
+        assert bool(g.get("failed_splits")) == bool(e.get("failed_splits")), \
+            (qi, toks, q)
+        if g.get("failed_splits"):
+            continue
+        assert g.get("num_hits", 0) == e.get("num_hits", 0), (qi, toks, q)
+        assert sorted(h.get("doc_id", 0) for h in g.get("partial_hits", [])) \
+            == sorted(h.get("doc_id", 0)
+                      for h in e.get("partial_hits", [])), (qi, toks, q)
