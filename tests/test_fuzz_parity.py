@@ -163,7 +163,7 @@ def assert_parity(g, e, scored, label):
 
 def test_random_query_parity(searchers):
     gpu, cpu = searchers
-    rng = random.Random(4242)
+    rng = random.Random(4242 + int(os.environ.get("QW_FUZZ_SEED", "0")))
     ran = 0
     rejected = 0
     for qi in range(N_QUERIES):
@@ -222,7 +222,7 @@ def test_random_query_parity_multi_segment(multiseg_searchers):
     """The same randomized battery over a 2-segment QWA2 split: per-segment
     collection, cross-segment merge and (split, segment, doc) tie-breaks."""
     gpu, cpu = multiseg_searchers
-    rng = random.Random(777)
+    rng = random.Random(777 + int(os.environ.get("QW_FUZZ_SEED", "0")))
     ran = 0
     for qi in range(60):
         q = rand_bool(rng) if rng.random() < 0.8 else rand_clause(rng, 0)
