@@ -25,9 +25,12 @@ Rank 0 prints ONE JSON line with metric/value plus:
     box's host cores on a bounded sample (--cpu-baseline-docs), repeated
     until >=2 s of wall time so the rate is honestly measurable.
 
-Extra workloads (SURVEY §8d configs 3/4; NOT the default the driver runs):
-  --workload range  bool must severity_text:INFO + u64 tenant_id range
-  --workload agg    date_histogram(1h) + terms(tenant) under match_all
+Extra workloads (SURVEY §8d configs 3/4/5; NOT the default the driver runs):
+  --workload range    bool must severity_text:INFO + u64 tenant_id range
+  --workload agg      date_histogram(1h) + terms(tenant) under match_all
+                      (--splits 8 = config #4's per-GPU shape)
+  --workload config5  top-1000 BM25 + range filter + terms agg
+                      (--docs 125000000 --splits 8 = configs[4]'s 1-GPU slice)
 """
 import argparse
 import json
