@@ -7,17 +7,19 @@
 // produces the ES-shaped JSON the golden scenarios assert
 // (rest-api-tests/scenarii/aggregations/0001-aggregations.yaml).
 //
-// Layout (little-endian):
-//   u32 magic 'QAG1' (0x31474151), u16 version=1, u16 n_aggs
+// Layout v4 (little-endian; python restatement: quickwit_amd/qagg.py):
+//   u32 magic 'QAG1' (0x31474151), u16 version=4, u16 n_aggs
 //   per agg:
-//     u16 name_len, name bytes
-//     u8 kind: 1=date_histogram 2=histogram 3=terms
-//     u16 n_sub (metric sub-aggs); per sub: u16 name_len, name
-//     histos: u32 n_buckets; per bucket:
-//        f64 key, u64 doc_count, per sub {u64 count, f64 sum, f64 min, f64 max}
-//       (buckets sorted by key, sparse: only non-empty buckets pre-merge)
-//     terms: u64 matched_docs_with_value, u32 n_entries; per entry:
-//        u16 key_len, key bytes, u64 doc_count   (sorted by key bytes)
+//     u16 name_len + name, u8 kind, u16 n_sub,
+//       per sub: u16 name_len + name + u8 sub_kind (0=stats 1=sketch)
+//     kind 6 (percentiles): u64 zero, u32 ne, ne x (i32 key, u64 count)
+//     kind 5 (metric): 40 B stats (count,sum,min,max,sum_sq)
+//     kind 3 (terms/cardinality/composite): u8 key_kind, u64 matched,
+//        f64 error_bound, u32 ne; per entry: u16 key_len + key +
+//        u64 doc_count + n_sub x 40 B stats   (sorted by key bytes)
+//     else (1=date_histogram 2=histogram 4=range): u32 n_buckets; per
+//        bucket: f64 key + u64 doc_count + per sub (sketch if sub_kind==1
+//        else 40 B stats)   (sorted by key, sparse pre-merge)
 #pragma once
 #include <algorithm>
 #include <cmath>
