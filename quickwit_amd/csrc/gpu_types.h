@@ -180,6 +180,9 @@ struct QueryDev {
     uint32_t agg_fast;
     uint32_t agg_nt;        // experiment: nontemporal loads in the pure-agg
                             // loop (QW_AGG_NT=1; streaming columns bypass L2)
+    uint32_t nt_decode;     // experiment: nontemporal loads for the posting
+                            // skip/gap/tf streams (QW_NT_DECODE=1; they are
+                            // read once — keep L2 for the fieldnorm gathers)
     uint64_t bitmap_out;    // absolute device VA: store each tile's match
                             // bitset words here (predicate-cache fill); 0=off
     uint64_t terms_off;     // scratch offsets of descriptor arrays

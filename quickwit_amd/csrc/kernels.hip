@@ -169,6 +169,15 @@ __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t t
     SkipEntryDev e_pf{};
     uint64_t g_lo = 0, g_hi = 0, t_lo = 0, t_hi = 0;
     uint32_t base_pf = 0;
+    const bool ntd = q.nt_decode != 0;
+    auto ld2 = [&](const uint32_t* p2) -> uint64_t {
+        // posting streams are read once: optionally bypass L2 so it stays
+        // warm for the per-posting fieldnorm gathers (QW_NT_DECODE)
+        if (ntd)
+            return uint64_t(__builtin_nontemporal_load(p2)) |
+                   (uint64_t(__builtin_nontemporal_load(p2 + 1)) << 32);
+        return uint64_t(p2[0]) | (uint64_t(p2[1]) << 32);
+    };
     auto prefetch = [&](uint32_t blk) {
         e_pf = skip[blk];
         const uint32_t* idb = payload + e_pf.word_off;
@@ -176,17 +185,16 @@ __device__ void decode_term_tile(const QueryDev& q, const TermDev& t, uint32_t t
         base_pf = seg == 0 ? e_pf.first_doc : idb[int(seg) - 4];
         if (w <= 16) {
             uint64_t word = (uint64_t(j0) * w) >> 5;
-            g_lo = uint64_t(idb[word]) | (uint64_t(idb[word + 1]) << 32);
-            g_hi = uint64_t(idb[word + 2]) | (uint64_t(idb[word + 3]) << 32);
+            g_lo = ld2(idb + word);
+            g_hi = ld2(idb + word + 2);
         }
         if (SCORE && e_pf.tf_bits) {
             const uint32_t* tfb = idb + 2 * ((128u * w + 63u) / 64u);
             uint32_t tw = e_pf.tf_bits;
             if (tw <= 16) {
                 uint64_t word = (uint64_t(j0) * tw) >> 5;
-                t_lo = uint64_t(tfb[word]) | (uint64_t(tfb[word + 1]) << 32);
-                t_hi =
-                    uint64_t(tfb[word + 2]) | (uint64_t(tfb[word + 3]) << 32);
+                t_lo = ld2(tfb + word);
+                t_hi = ld2(tfb + word + 2);
             }
         }
     };

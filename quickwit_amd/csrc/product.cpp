@@ -2052,6 +2052,8 @@ static SplitResult search_split_gpu(qw_ctx* ctx, const DeviceSplit& ds,
             q.agg_fast = 1;
         static const bool agg_nt = getenv("QW_AGG_NT") != nullptr;
         q.agg_nt = agg_nt ? 1 : 0;
+        static const bool nt_decode = getenv("QW_NT_DECODE") != nullptr;
+        q.nt_decode = nt_decode ? 1 : 0;
         HIP_CHECK(hipEventRecord(ctx->ev_start, ctx->stream));
         launch_leaf_tile(ns, nb, na, collect, dim3(grid), ctx->stream, q, 0u,
                          n_tiles, 1u);
