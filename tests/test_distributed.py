@@ -151,3 +151,64 @@ def test_qagg_blob_roundtrip():
         if not e.dense_eligible:
             e.buckets = None
     assert qagg.serialize_blob(entries) == blob
+
+
+def _rank_main_multiseg(rank, world, port, result):
+    import torch.distributed as dist
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    import torch
+
+    from quickwit_amd.api import OracleSearcher
+    from quickwit_amd.merge import distributed_merge
+
+    # each rank owns a 2-segment QWA2 split
+    a = splitgen.generate_split(10 * rank, 4000, seed=5)
+    b = splitgen.generate_split(10 * rank + 1, 3000, seed=5)
+    sid = f"ms-{rank}"
+    s = OracleSearcher()
+    s.add_split(sid, splitgen.concat_segments([a, b], sid))
+    req = _make_req("bm25", [(sid, 7000)])
+    resp_pb = s.leaf_search_raw(proto.encode("LeafSearchRequest", req))
+    sreq_pb = proto.encode("SearchRequest", req["search_request"])
+    merged = distributed_merge(sreq_pb, resp_pb, [sid],
+                               device=torch.device("cpu"))
+    if rank == 0:
+        result.put(merged)
+    dist.destroy_process_group()
+
+
+def test_two_rank_packed_merge_multisegment_splits():
+    """Packed 32B hit records carry (split_ord, segment_ord) across ranks:
+    two ranks each holding a 2-segment QWA2 split must merge exactly like
+    one oracle call over all four segments' splits."""
+    import __graft_entry__
+    __graft_entry__.build()
+    ctx = mp.get_context("spawn")
+    result = ctx.Queue()
+    procs = [ctx.Process(target=_rank_main_multiseg,
+                         args=(r, 2, 29514, result)) for r in range(2)]
+    for p in procs:
+        p.start()
+    merged = proto.decode("LeafSearchResponse", result.get(timeout=180))
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    from quickwit_amd.api import OracleSearcher
+    both = OracleSearcher()
+    splits = []
+    for r in range(2):
+        a = splitgen.generate_split(10 * r, 4000, seed=5)
+        b = splitgen.generate_split(10 * r + 1, 3000, seed=5)
+        sid = f"ms-{r}"
+        both.add_split(sid, splitgen.concat_segments([a, b], sid))
+        splits.append((sid, 7000))
+    expected = both.leaf_search(_make_req("bm25", splits))
+    assert merged["num_hits"] == expected["num_hits"]
+    assert ([(h["split_id"], h.get("segment_ord", 0), h["doc_id"])
+             for h in merged["partial_hits"]] ==
+            [(h["split_id"], h.get("segment_ord", 0), h["doc_id"])
+             for h in expected["partial_hits"]])
