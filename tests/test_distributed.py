@@ -22,6 +22,15 @@ AGGS = {"per_hour": {"date_histogram": {"field": "timestamp",
                                           "fixed_interval": "7200000ms"},
                        "aggs": {"ten": {"stats": {"field": "tenant_id"}}}}}
 
+AGGS2 = {
+    "card_t": {"cardinality": {"field": "tenant_name"}},
+    "card_ts": {"cardinality": {"field": "timestamp"}},
+    "comp": {"composite": {"size": 200, "sources": [
+        {"t": {"terms": {"field": "tenant_name"}}},
+        {"i": {"histogram": {"field": "tenant_id", "interval": 5}}}]}},
+    "pct": {"percentiles": {"field": "tenant_id", "percents": [50, 95]}},
+}
+
 CASES = {
     "bm25": dict(
         q={"type": "bool", "should": [
@@ -29,6 +38,7 @@ CASES = {
             for i in range(3)]},
         sort=[{"field_name": "_score", "sort_order": 1}], aggs=None),
     "agg": dict(q={"type": "match_all"}, sort=None, aggs=AGGS),
+    "agg2": dict(q={"type": "match_all"}, sort=None, aggs=AGGS2),
     "field_sort": dict(
         q={"type": "term", "field": "severity_text", "value": "INFO"},
         sort=[{"field_name": "timestamp", "sort_order": 1},
@@ -127,6 +137,26 @@ def test_two_rank_packed_merge_aggs_equal_single_call():
                                 AGGS)
     assert merged["num_hits"] == expected["num_hits"]
     assert gj == ej
+
+
+def test_two_rank_packed_merge_sideband_aggs_equal_single_call():
+    """Cardinality (exact distinct-set union), composite (canonical-key
+    merge across split-local ord spaces) and percentiles (DDSketch bucket
+    add) are NOT dense-eligible: they cross ranks only through the sideband
+    byte tensor and the rank-0 C-ABI merge. Must equal one oracle call
+    over both splits."""
+    merged = _run_case("agg2", 29515)
+    both, expected = _single_call_expected("agg2")
+    gj = both.finalize_agg_json(merged["intermediate_aggregation_result"],
+                                AGGS2)
+    ej = both.finalize_agg_json(expected["intermediate_aggregation_result"],
+                                AGGS2)
+    assert merged["num_hits"] == expected["num_hits"]
+    assert gj == ej
+    # sanity on shape: all three kinds actually produced values
+    assert gj["card_t"]["value"] > 0
+    assert gj["comp"]["buckets"]
+    assert "50.0" in gj["pct"]["values"] or "50" in gj["pct"]["values"]
 
 
 def test_qagg_blob_roundtrip():
