@@ -147,6 +147,27 @@ def test_merged_equals_combined(setup, case):
         assert got == wj, (case, q, got, wj)
 
 
+@pytest.mark.parametrize("case", sorted(AGG_CASES))
+def test_qagg_codec_identity_every_family(setup, case):
+    """merge.py's structural QAGG codec (qagg.parse_blob/serialize_blob —
+    the dense/sideband splitter the RCCL exchange rides) must be the exact
+    identity on real product blobs of EVERY aggregation family, both as
+    raw passthrough and when buckets are force-re-encoded."""
+    from quickwit_amd import qagg
+    combined, _, splits = setup
+    aggs = AGG_CASES[case]
+    req = make_leaf_request({"type": "match_all"}, SCHEMA, splits,
+                            max_hits=0, aggregation=aggs)
+    blob = combined.leaf_search(req)["intermediate_aggregation_result"]
+    entries = qagg.parse_blob(blob)
+    assert entries, case
+    assert qagg.serialize_blob(entries) == blob, case
+    for e in entries:
+        if not e.dense_eligible:
+            e.buckets = None  # force bucket re-encode instead of passthrough
+    assert qagg.serialize_blob(entries) == blob, case
+
+
 def test_terms_order_key(setup):
     combined, singles, splits = setup
     for direction, keys in (("asc", ["api", "ing", "jan"]),
