@@ -210,6 +210,88 @@ def _rank_main_multiseg(rank, world, port, result):
     dist.destroy_process_group()
 
 
+PHRASE_SCHEMA = {"timestamp_field": None, "fields": [
+    {"name": "body", "type": "text", "tokenizer": "default",
+     "record": "position", "fieldnorms": True}]}
+PHRASE_SCHEMA_NOPOS = {"timestamp_field": None, "fields": [
+    {"name": "body", "type": "text", "tokenizer": "default",
+     "record": "freq", "fieldnorms": True}]}
+
+
+def _phrase_split(rank):
+    schema = PHRASE_SCHEMA if rank == 0 else PHRASE_SCHEMA_NOPOS
+    w = splitgen.SplitWriter(schema, f"pf-{rank}", store_docs=False)
+    w.add_documents([{"body": "alpha beta"}, {"body": "beta alpha"}])
+    return w.finalize()
+
+
+def _rank_main_failure(rank, world, port, result):
+    import torch.distributed as dist
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    import torch
+
+    from quickwit_amd.api import OracleSearcher, make_leaf_request
+    from quickwit_amd.merge import distributed_merge
+
+    sid = f"pf-{rank}"
+    s = OracleSearcher()
+    s.add_split(sid, _phrase_split(rank))
+    q = {"type": "full_text", "field": "body", "text": "alpha beta",
+         "params": {"mode": {"type": "phrase"}}}
+    req = make_leaf_request(q, PHRASE_SCHEMA, [(sid, 2)], max_hits=10)
+    resp_pb = s.leaf_search_raw(proto.encode("LeafSearchRequest", req))
+    sreq_pb = proto.encode("SearchRequest", req["search_request"])
+    merged = distributed_merge(sreq_pb, resp_pb, [sid],
+                               device=torch.device("cpu"))
+    if rank == 0:
+        result.put(merged)
+    dist.destroy_process_group()
+
+
+def test_two_rank_merge_carries_failed_splits_as_data():
+    """Per-split failures are DATA inside the response (leaf.rs:2143-2148),
+    and the packed exchange must carry them across ranks through the
+    sideband bytes: rank 1's positions-less split fails the phrase query,
+    rank 0's split answers — the merged response reports both, exactly
+    like one call over both splits."""
+    import __graft_entry__
+    __graft_entry__.build()
+    ctx = mp.get_context("spawn")
+    result = ctx.Queue()
+    procs = [ctx.Process(target=_rank_main_failure, args=(r, 2, 29516, result))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    merged = proto.decode("LeafSearchResponse", result.get(timeout=180))
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    from quickwit_amd.api import OracleSearcher, make_leaf_request
+    both = OracleSearcher()
+    for r in range(2):
+        both.add_split(f"pf-{r}", _phrase_split(r))
+    q = {"type": "full_text", "field": "body", "text": "alpha beta",
+         "params": {"mode": {"type": "phrase"}}}
+    expected = both.leaf_search(make_leaf_request(
+        q, PHRASE_SCHEMA, [("pf-0", 2), ("pf-1", 2)], max_hits=10))
+
+    assert merged.get("num_hits") == expected.get("num_hits") == 1
+    assert ([h["split_id"] for h in merged["partial_hits"]] ==
+            [h["split_id"] for h in expected["partial_hits"]] == ["pf-0"])
+    mf = merged.get("failed_splits", [])
+    ef = expected.get("failed_splits", [])
+    assert [f["split_id"] for f in mf] == [f["split_id"] for f in ef] == \
+        ["pf-1"]
+    assert merged.get("num_successful_splits") == \
+        expected.get("num_successful_splits")
+    assert merged.get("num_attempted_splits") == \
+        expected.get("num_attempted_splits")
+
+
 def test_two_rank_packed_merge_multisegment_splits():
     """Packed 32B hit records carry (split_ord, segment_ord) across ranks:
     two ranks each holding a 2-segment QWA2 split must merge exactly like
