@@ -148,6 +148,59 @@ def test_merged_equals_combined(setup, case):
 
 
 @pytest.mark.parametrize("case", sorted(AGG_CASES))
+def test_merge_is_input_order_independent(setup, case):
+    """merge_fruits' exact tie-breaks (sorting.md:14-26) make the merged
+    result a function of the SET of leaf responses, not their arrival
+    order — required for the multi-GPU allgather, where rank order is
+    arbitrary. Hits, counters and finalized aggs must be identical for
+    every permutation of the per-split responses; the one permitted
+    order effect is f64 summation non-associativity in accumulated
+    metrics (sum/avg wander by ~1 ulp — the reference's own merge adds
+    f64 in arrival order too), so floats compare at rel 1e-12."""
+    import itertools
+    import math
+
+    def eq(a, b):
+        if isinstance(a, float) or isinstance(b, float):
+            return (a == b or (isinstance(a, float) and isinstance(b, float)
+                               and math.isclose(a, b, rel_tol=1e-12)))
+        if isinstance(a, dict):
+            return (isinstance(b, dict) and a.keys() == b.keys()
+                    and all(eq(a[k], b[k]) for k in a))
+        if isinstance(a, list):
+            return (isinstance(b, list) and len(a) == len(b)
+                    and all(eq(x, y) for x, y in zip(a, b)))
+        return a == b
+    combined, singles, splits = setup
+    aggs = AGG_CASES[case]
+    req_all = make_leaf_request({"type": "match_all"}, SCHEMA, splits,
+                                max_hits=5, aggregation=aggs)
+    sreq_pb = proto.encode("SearchRequest", req_all["search_request"])
+    resps = []
+    for i, s in enumerate(singles):
+        req1 = make_leaf_request({"type": "match_all"}, SCHEMA, [splits[i]],
+                                 max_hits=5, aggregation=aggs)
+        resps.append(s.leaf_search_raw(proto.encode("LeafSearchRequest",
+                                                    req1)))
+    base = None
+    for perm in itertools.permutations(range(3)):
+        merged = proto.decode("LeafSearchResponse",
+                              merge_leaf_responses(sreq_pb,
+                                                   [resps[i] for i in perm]))
+        key = (merged.get("num_hits"),
+               [(h["split_id"], h.get("doc_id", 0))
+                for h in merged.get("partial_hits", [])],
+               merged.get("num_successful_splits"),
+               combined.finalize_agg_json(
+                   merged.get("intermediate_aggregation_result", b""), aggs))
+        if base is None:
+            base = key
+        else:
+            assert key[:3] == base[:3], (case, perm)
+            assert eq(key[3], base[3]), (case, perm, key[3], base[3])
+
+
+@pytest.mark.parametrize("case", sorted(AGG_CASES))
 def test_qagg_codec_identity_every_family(setup, case):
     """merge.py's structural QAGG codec (qagg.parse_blob/serialize_blob —
     the dense/sideband splitter the RCCL exchange rides) must be the exact
