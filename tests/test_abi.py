@@ -25,6 +25,35 @@ def build_all():
     __graft_entry__.build()
 
 
+def test_malformed_request_bytes_never_crash():
+    """§8b boundary contract: errors come back as status codes + messages,
+    never faults. The C++ pb decoder (csrc/pb.h, shared by product and
+    oracle) must survive truncated, corrupted and random request bytes —
+    returning an error or an empty result, in-process."""
+    import random
+
+    from quickwit_amd import proto, splitgen
+    from quickwit_amd.api import OracleSearcher, make_leaf_request
+    s = OracleSearcher()
+    s.add_split("s", splitgen.generate_split(0, 500, seed=1))
+    good = proto.encode("LeafSearchRequest", make_leaf_request(
+        {"type": "match_all"}, splitgen.HDFS_SCHEMA, [("s", 500)],
+        max_hits=5))
+    rng = random.Random(7)
+    cases = [b"", b"\xff", b"\xff" * 16, good[:-3], good[:7],
+             bytes(rng.randrange(256) for _ in range(64))]
+    for _ in range(120):
+        b = bytearray(good)
+        for _ in range(rng.randrange(1, 5)):
+            b[rng.randrange(len(b))] = rng.randrange(256)
+        cases.append(bytes(b))
+    for c in cases:
+        try:
+            s.leaf_search_raw(c)  # error status surfaces as an exception
+        except Exception:
+            pass
+
+
 def header_symbols():
     hdr = open(os.path.join(REPO, "include", "quickwit_amd.h")).read()
     return sorted(set(re.findall(r"\b(qw_[a-z_0-9]+)\s*\(", hdr)) - {"qw_buf"})
