@@ -54,6 +54,39 @@ def test_malformed_request_bytes_never_crash():
             pass
 
 
+def test_corrupted_split_containers_never_crash():
+    """qw_ctx_add_split must reject truncated/corrupted QWA1 images with a
+    status code (wild meta offsets must not drive pointer arithmetic);
+    a split that slips past validation may return garbage results but
+    must not fault."""
+    import random
+
+    from quickwit_amd import splitgen
+    from quickwit_amd.api import OracleSearcher, make_leaf_request
+    good = splitgen.generate_split(0, 500, seed=1)
+    rng = random.Random(11)
+    cases = [b"", b"QWAMDSP1", good[:100], good[:-10], b"\x00" * 200,
+             bytes(rng.randrange(256) for _ in range(512))]
+    for _ in range(80):
+        b = bytearray(good)
+        for _ in range(rng.randrange(1, 6)):
+            b[rng.randrange(len(b))] = rng.randrange(256)
+        cases.append(bytes(b))
+    for _ in range(40):  # footer bytes drive the meta offsets — hit them
+        b = bytearray(good)
+        b[len(b) - 24 + rng.randrange(16)] = rng.randrange(256)
+        cases.append(bytes(b))
+    req = make_leaf_request({"type": "match_all"}, splitgen.HDFS_SCHEMA,
+                            [("x", 500)], max_hits=5)
+    for c in cases:
+        s = OracleSearcher()
+        try:
+            s.add_split("x", c)
+            s.leaf_search(req)
+        except Exception:
+            pass
+
+
 def header_symbols():
     hdr = open(os.path.join(REPO, "include", "quickwit_amd.h")).read()
     return sorted(set(re.findall(r"\b(qw_[a-z_0-9]+)\s*\(", hdr)) - {"qw_buf"})
