@@ -87,6 +87,64 @@ def test_corrupted_split_containers_never_crash():
             pass
 
 
+def test_corrupted_merge_inputs_never_crash():
+    """The rank-0 merge path consumes response bytes and QAGG1 blobs that
+    crossed the wire: qw_merge_leaf_responses and qw_finalize_agg_to_json
+    must survive mutated and truncated inputs with a status code."""
+    import ctypes
+    import json
+    import random
+
+    from quickwit_amd import proto, splitgen
+    from quickwit_amd.api import OracleSearcher, make_leaf_request
+    from quickwit_amd.merge import _Buf, _get_lib, merge_leaf_responses
+
+    aggs = {"h": {"date_histogram": {"field": "timestamp",
+                                     "fixed_interval": "3600000ms"}},
+            "t": {"terms": {"field": "tenant_name", "size": 5},
+                  "aggs": {"st": {"stats": {"field": "tenant_id"}}}}}
+    s = OracleSearcher()
+    s.add_split("s", splitgen.generate_split(0, 2000, seed=3))
+    req = make_leaf_request({"type": "match_all"}, splitgen.HDFS_SCHEMA,
+                            [("s", 2000)], max_hits=5, aggregation=aggs)
+    resp_pb = s.leaf_search_raw(proto.encode("LeafSearchRequest", req))
+    sreq_pb = proto.encode("SearchRequest", req["search_request"])
+    rng = random.Random(17)
+
+    for _ in range(60):
+        b = bytearray(resp_pb)
+        for _ in range(rng.randrange(1, 5)):
+            b[rng.randrange(len(b))] = rng.randrange(256)
+        try:
+            merge_leaf_responses(sreq_pb, [resp_pb, bytes(b)])
+        except Exception:
+            pass
+    for cut in (0, 1, 7, len(resp_pb) // 2, len(resp_pb) - 2):
+        try:
+            merge_leaf_responses(sreq_pb, [resp_pb[:cut]])
+        except Exception:
+            pass
+
+    blob = proto.decode("LeafSearchResponse",
+                        resp_pb)["intermediate_aggregation_result"]
+    lib = _get_lib()
+    lib.qw_finalize_agg_to_json.argtypes = [
+        ctypes.c_char_p, ctypes.c_size_t, ctypes.c_char_p,
+        ctypes.POINTER(_Buf)]
+    muts = []
+    for _ in range(60):
+        b = bytearray(blob)
+        for _ in range(rng.randrange(1, 5)):
+            b[rng.randrange(len(b))] = rng.randrange(256)
+        muts.append(bytes(b))
+    muts += [blob[:c] for c in (0, 1, 5, len(blob) // 2, len(blob) - 1)]
+    for m in muts:
+        buf = _Buf()
+        if lib.qw_finalize_agg_to_json(m, len(m), json.dumps(aggs).encode(),
+                                       ctypes.byref(buf)) == 0:
+            lib.qw_buf_free(ctypes.byref(buf))
+
+
 def header_symbols():
     hdr = open(os.path.join(REPO, "include", "quickwit_amd.h")).read()
     return sorted(set(re.findall(r"\b(qw_[a-z_0-9]+)\s*\(", hdr)) - {"qw_buf"})
