@@ -28,6 +28,8 @@
 #include <string>
 #include <vector>
 
+#include "qw_unicode.h"
+
 #include "minijson.h"
 
 namespace qw {
@@ -79,6 +81,27 @@ struct Schema {
     }
 };
 
+// UTF-8-aware default tokenizer; alnum/lowercase from the GENERATED
+// shared tables (qw_unicode.h == quickwit_amd/unicode_tables.py) so the
+// query side agrees with the split writer on every codepoint. A malformed
+// UTF-8 byte is treated as a token separator.
+inline void qw_utf8_append(std::string& s, uint32_t cp) {
+    if (cp < 0x80) s.push_back(char(cp));
+    else if (cp < 0x800) {
+        s.push_back(char(0xC0 | (cp >> 6)));
+        s.push_back(char(0x80 | (cp & 0x3F)));
+    } else if (cp < 0x10000) {
+        s.push_back(char(0xE0 | (cp >> 12)));
+        s.push_back(char(0x80 | ((cp >> 6) & 0x3F)));
+        s.push_back(char(0x80 | (cp & 0x3F)));
+    } else {
+        s.push_back(char(0xF0 | (cp >> 18)));
+        s.push_back(char(0x80 | ((cp >> 12) & 0x3F)));
+        s.push_back(char(0x80 | ((cp >> 6) & 0x3F)));
+        s.push_back(char(0x80 | (cp & 0x3F)));
+    }
+}
+
 inline std::vector<std::string> tokenize(const std::string& text,
                                          const std::string& tokenizer) {
     if (tokenizer == "raw") {
@@ -87,14 +110,39 @@ inline std::vector<std::string> tokenize(const std::string& text,
     }
     std::vector<std::string> out;
     std::string cur;
-    for (unsigned char c : text) {
-        if (std::isalnum(c)) cur.push_back(char(std::tolower(c)));
-        else if (!cur.empty()) {
+    auto flush = [&]() {
+        if (!cur.empty()) {
             if (cur.size() <= 40) out.push_back(cur);
             cur.clear();
         }
+    };
+    size_t i = 0, n = text.size();
+    while (i < n) {
+        unsigned char c = (unsigned char)text[i];
+        if (c < 0x80) {
+            if (std::isalnum(c)) cur.push_back(char(std::tolower(c)));
+            else flush();
+            ++i;
+            continue;
+        }
+        // decode one UTF-8 sequence (2-4 bytes)
+        uint32_t cp = 0;
+        size_t len = 0;
+        if ((c & 0xE0) == 0xC0) { cp = c & 0x1F; len = 2; }
+        else if ((c & 0xF0) == 0xE0) { cp = c & 0x0F; len = 3; }
+        else if ((c & 0xF8) == 0xF0) { cp = c & 0x07; len = 4; }
+        bool ok = len > 0 && i + len <= n;
+        for (size_t k = 1; ok && k < len; ++k) {
+            unsigned char cc = (unsigned char)text[i + k];
+            if ((cc & 0xC0) != 0x80) ok = false;
+            else cp = (cp << 6) | (cc & 0x3F);
+        }
+        if (!ok) { flush(); ++i; continue; }
+        if (qw_is_alnum_cp(cp)) qw_utf8_append(cur, qw_lower_cp(cp));
+        else flush();
+        i += len;
     }
-    if (!cur.empty() && cur.size() <= 40) out.push_back(cur);
+    flush();
     return out;
 }
 

@@ -100,8 +100,10 @@ def es_query_to_ast(q, schema=None):
         if pt:
             return pt
         if isinstance(body, dict) and body.get("case_insensitive"):
-            # tokenized (lowercasing) index: fold the needle too
-            value = str(value).lower()
+            # tokenized (lowercasing) index: fold the needle too, with the
+            # same shared table the tokenizer uses
+            from .unicode_tables import lower_cp
+            value = "".join(chr(lower_cp(ord(c))) for c in str(value))
         return {"type": "term", "field": field, "value": str(value)}
     if "terms" in q:
         [(field, values)] = [(k, v) for k, v in q["terms"].items()

@@ -40,22 +40,35 @@ assert SKIP_DTYPE.itemsize == 16
 # ---------------------------------------------------------------- tokenizers
 def tokenize(text: str, tokenizer: str):
     """'raw' = whole string, verbatim (tantivy raw tokenizer). 'default' =
-    alphanumeric runs, lowercased, tokens longer than 40 chars dropped
-    (tantivy default = SimpleTokenizer + RemoveLongFilter(40) + LowerCaser)."""
+    alphanumeric runs, lowercased, tokens longer than 40 UTF-8 bytes
+    dropped (tantivy default = SimpleTokenizer + RemoveLongFilter(40) +
+    LowerCaser). Alnum/lowercase come from the GENERATED shared tables
+    (unicode_tables.py == csrc/qw_unicode.h) so the C++ query-side
+    tokenizer agrees with this writer on every codepoint."""
+    from .unicode_tables import is_alnum_cp, lower_cp
     if text is None:
         return []
     if tokenizer == "raw":
         return [text] if text else []
     toks, cur = [], []
     for ch in text:
-        if ch.isalnum():
-            cur.append(ch.lower())
-        elif cur:
+        cp = ord(ch)
+        if cp < 128:
+            if ("0" <= ch <= "9") or ("a" <= ch <= "z"):
+                cur.append(ch)
+                continue
+            if "A" <= ch <= "Z":
+                cur.append(chr(cp + 32))
+                continue
+        elif is_alnum_cp(cp):
+            cur.append(chr(lower_cp(cp)))
+            continue
+        if cur:
             toks.append("".join(cur))
             cur = []
     if cur:
         toks.append("".join(cur))
-    return [t for t in toks if len(t) <= 40]
+    return [t for t in toks if len(t.encode("utf-8")) <= 40]
 
 
 # ------------------------------------------------------------- bit packing
