@@ -102,6 +102,37 @@ inline void qw_utf8_append(std::string& s, uint32_t cp) {
     }
 }
 
+// UTF-8-aware lowercase with the same shared table (wildcard patterns,
+// case_insensitive folds); malformed bytes pass through unchanged
+inline std::string qw_utf8_lower(const std::string& text) {
+    std::string out;
+    out.reserve(text.size());
+    size_t i = 0, n = text.size();
+    while (i < n) {
+        unsigned char c = (unsigned char)text[i];
+        if (c < 0x80) {
+            out.push_back(char(std::tolower(c)));
+            ++i;
+            continue;
+        }
+        uint32_t cp = 0;
+        size_t len = 0;
+        if ((c & 0xE0) == 0xC0) { cp = c & 0x1F; len = 2; }
+        else if ((c & 0xF0) == 0xE0) { cp = c & 0x0F; len = 3; }
+        else if ((c & 0xF8) == 0xF0) { cp = c & 0x07; len = 4; }
+        bool ok = len > 0 && i + len <= n;
+        for (size_t k = 1; ok && k < len; ++k) {
+            unsigned char cc = (unsigned char)text[i + k];
+            if ((cc & 0xC0) != 0x80) ok = false;
+            else cp = (cp << 6) | (cc & 0x3F);
+        }
+        if (!ok) { out.push_back(char(c)); ++i; continue; }
+        qw_utf8_append(out, qw_lower_cp(cp));
+        i += len;
+    }
+    return out;
+}
+
 inline std::vector<std::string> tokenize(const std::string& text,
                                          const std::string& tokenizer) {
     if (tokenizer == "raw") {
@@ -218,24 +249,67 @@ inline bool plan_has_const_score(const PlanNode& n) {
     return false;
 }
 
-// glob match with '*' (any run) and '?' (single byte), iterative backtracking
+// decode one UTF-8 codepoint at byte index i (advances i); a malformed
+// lead byte yields a per-byte sentinel so it only equals itself
+inline uint32_t qw_utf8_next(const char* s, size_t n, size_t& i) {
+    unsigned char c = (unsigned char)s[i];
+    if (c < 0x80) { ++i; return c; }
+    uint32_t cp = 0;
+    size_t len = 0;
+    if ((c & 0xE0) == 0xC0) { cp = c & 0x1F; len = 2; }
+    else if ((c & 0xF0) == 0xE0) { cp = c & 0x0F; len = 3; }
+    else if ((c & 0xF8) == 0xF0) { cp = c & 0x07; len = 4; }
+    if (!len || i + len > n) { ++i; return 0x80000000u | c; }
+    for (size_t k = 1; k < len; ++k) {
+        unsigned char cc = (unsigned char)s[i + k];
+        if ((cc & 0xC0) != 0x80) { ++i; return 0x80000000u | c; }
+        cp = (cp << 6) | (cc & 0x3F);
+    }
+    i += len;
+    return cp;
+}
+
+// glob match with '*' (any run) and '?' (single CODEPOINT — the reference
+// compiles wildcards to a regex over chars, wildcard_query.rs), iterative
+// backtracking over UTF-8; case_insensitive folds through the shared
+// lowercase table on both sides
 inline bool glob_match(const char* s, size_t sn, const char* p, size_t pn, bool ci) {
-    size_t si = 0, pi = 0, star = size_t(-1), mark = 0;
-    auto eq = [&](char a, char b) {
-        return ci ? std::tolower((unsigned char)a) == std::tolower((unsigned char)b)
-                  : a == b;
+    auto fold = [&](uint32_t cp) -> uint32_t {
+        if (!ci) return cp;
+        if (cp < 0x80) return uint32_t(std::tolower(int(cp)));
+        return cp < 0x80000000u ? qw_lower_cp(cp) : cp;
     };
+    size_t si = 0, pi = 0, star_p = size_t(-1), star_s = 0;
     while (si < sn) {
-        if (pi < pn && (p[pi] == '?' || eq(p[pi], s[si]))) {
-            ++si;
-            ++pi;
-        } else if (pi < pn && p[pi] == '*') {
-            star = pi++;
-            mark = si;
-        } else if (star != size_t(-1)) {
-            pi = star + 1;
-            si = ++mark;
-        } else return false;
+        if (pi < pn && p[pi] == '*') {
+            star_p = ++pi;
+            star_s = si;
+            continue;
+        }
+        if (pi < pn) {
+            size_t pj = pi, sj = si;
+            if (p[pi] == '?') {
+                ++pj;
+                qw_utf8_next(s, sn, sj);
+                pi = pj;
+                si = sj;
+                continue;
+            }
+            uint32_t pc = qw_utf8_next(p, pn, pj);
+            uint32_t sc = qw_utf8_next(s, sn, sj);
+            if (fold(pc) == fold(sc)) {
+                pi = pj;
+                si = sj;
+                continue;
+            }
+        }
+        if (star_p != size_t(-1)) {
+            qw_utf8_next(s, sn, star_s);  // widen the '*' by one codepoint
+            si = star_s;
+            pi = star_p;
+            continue;
+        }
+        return false;
     }
     while (pi < pn && p[pi] == '*') ++pi;
     return pi == pn;
@@ -837,9 +911,7 @@ inline PlanNode build_plan(const mj::Value* ast, const Schema& schema) {
         // the reference tokenizes the pattern's literal segments with the
         // field's tokenizer (wildcard_query.rs:111-160): for the default
         // (lowercasing) analyzer that lowercases them
-        if (f->tokenizer != "raw")
-            std::transform(n.value.begin(), n.value.end(), n.value.begin(),
-                           [](unsigned char c) { return char(std::tolower(c)); });
+        if (f->tokenizer != "raw") n.value = qw_utf8_lower(n.value);
     } else if (ty == "user_input") {
         std::vector<std::string> dfs;
         const mj::Value* d = ast->get("default_fields");

@@ -92,6 +92,38 @@ def test_random_unicode_index_query_agreement():
                                            splitgen.tokenize(t, "default"))
 
 
+def test_unicode_wildcards():
+    """Wildcard patterns fold through the field's tokenizer chain like the
+    reference (wildcard_query.rs:111-160): default-analyzer fields
+    lowercase the pattern with the shared table; `?` consumes one
+    CODEPOINT (the reference compiles to a regex over chars); raw fields
+    fold only under case_insensitive."""
+    schema = {"timestamp_field": None, "fields": [
+        {"name": "body", "type": "text", "tokenizer": "default",
+         "record": "freq", "fieldnorms": True},
+        {"name": "tag", "type": "text", "tokenizer": "raw",
+         "record": "basic", "fieldnorms": False}]}
+    w = splitgen.SplitWriter(schema, "wc", store_docs=False)
+    w.add_documents([{"body": "hello héllo world wörld",
+                      "tag": "Héllo-World"}])
+    s = OracleSearcher()
+    s.add_split("wc", w.finalize())
+    cases = [
+        (dict(field="body", value="HÉL*"), 1),
+        (dict(field="body", value="WÖR*"), 1),
+        (dict(field="body", value="h?llo"), 1),   # ? matches é (one cp)
+        (dict(field="body", value="h??llo"), 0),  # é is ONE cp, not 2 bytes
+        (dict(field="tag", value="Héllo-*"), 1),  # raw: byte-exact
+        (dict(field="tag", value="héllo-*"), 0),
+        (dict(field="tag", value="HÉLLO-*", case_insensitive=True), 1),
+    ]
+    for kw, want in cases:
+        q = dict({"type": "wildcard"}, **kw)
+        r = s.leaf_search(make_leaf_request(q, schema, [("wc", 1)],
+                                            max_hits=5))
+        assert r.get("num_hits", 0) == want, kw
+
+
 @pytest.mark.gpu
 def test_unicode_phrase_parity_gpu():
     from quickwit_amd.api import GpuSearcher
