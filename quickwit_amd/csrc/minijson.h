@@ -56,6 +56,12 @@ class Parser {
   private:
     const char* p_;
     const char* end_;
+    // serde_json's default recursion limit is 128; the reference's
+    // query_ast parse errors on deeper input instead of overflowing the
+    // stack — mirror that (a pathological request must be a status code,
+    // never a fault; pinned by tests/test_abi.py)
+    static constexpr int kMaxDepth = 128;
+    int depth_ = 0;
     [[noreturn]] void fail(const char* msg) {
         throw std::runtime_error(std::string("json parse error: ") + msg);
     }
@@ -77,6 +83,12 @@ class Parser {
     ValuePtr value() {
         ws();
         char c = peek();
+        if ((c == '{' || c == '[') && ++depth_ > kMaxDepth)
+            fail("recursion limit exceeded");
+        struct DepthGuard {
+            int* d;
+            ~DepthGuard() { if (d) --*d; }
+        } guard{(c == '{' || c == '[') ? &depth_ : nullptr};
         auto v = std::make_shared<Value>();
         if (c == '{') {
             v->kind = Value::OBJ;

@@ -87,6 +87,36 @@ def test_corrupted_split_containers_never_crash():
             pass
 
 
+def test_deeply_nested_query_ast_is_error_not_stack_overflow():
+    """A pathological query_ast nested 200k levels deep must come back as
+    a per-split failure ('recursion limit exceeded', mirroring
+    serde_json's 128 default in the reference), not a stack-overflow
+    fault — which is exactly what it was before the minijson depth cap."""
+    import json
+
+    from quickwit_amd import proto, splitgen
+    from quickwit_amd.api import OracleSearcher, make_leaf_request
+    s = OracleSearcher()
+    s.add_split("d", splitgen.generate_split(0, 200, seed=1))
+    base = make_leaf_request({"type": "match_all"}, splitgen.HDFS_SCHEMA,
+                             [("d", 200)], max_hits=5)
+    for depth, ok in ((50, True), (200, False), (200_000, False)):
+        ast = ('{"type":"bool","must":[' * depth
+               + '{"type":"term","field":"body","value":"w00001"}'
+               + ']}' * depth)
+        req = json.loads(json.dumps(base))
+        req["search_request"]["query_ast"] = ast
+        resp = proto.decode(
+            "LeafSearchResponse",
+            s.leaf_search_raw(proto.encode("LeafSearchRequest", req)))
+        failed = resp.get("failed_splits", [])
+        if ok:
+            assert not failed and resp.get("num_hits", 0) > 0, depth
+        else:
+            assert failed and "recursion" in failed[0]["error"], (depth,
+                                                                  failed)
+
+
 def test_corrupted_merge_inputs_never_crash():
     """The rank-0 merge path consumes response bytes and QAGG1 blobs that
     crossed the wire: qw_merge_leaf_responses and qw_finalize_agg_to_json
