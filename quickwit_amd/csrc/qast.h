@@ -598,7 +598,12 @@ inline Bound qbound(const std::string& text, Bound::Kind kind, const SchemaField
 // AND connector promotes BOTH neighbors to Must, OR keeps Should.
 inline PlanNode qparse_clause(const std::vector<QTok>& toks, size_t* pos,
                               const std::vector<std::string>& dfs,
-                              const Schema& schema) {
+                              const Schema& schema, int depth = 0) {
+    // paren nesting capped like the JSON parser (a pathological query is
+    // an error, never a stack overflow; tests/test_abi.py)
+    if (depth > 128)
+        throw std::runtime_error(
+            "query grammar: recursion limit exceeded");
     struct Item {
         int occur = 0;  // 0 default(Should), 1 Must, 2 MustNot
         PlanNode node;
@@ -622,7 +627,7 @@ inline PlanNode qparse_clause(const std::vector<QTok>& toks, size_t* pos,
         pending = 0;
         if (t.kind == QTok::LPAREN) {
             ++(*pos);
-            it.node = qparse_clause(toks, pos, dfs, schema);
+            it.node = qparse_clause(toks, pos, dfs, schema, depth + 1);
         } else {
             it.node = qtok_leaf(t, dfs, schema);
             ++(*pos);

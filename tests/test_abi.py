@@ -117,6 +117,32 @@ def test_deeply_nested_query_ast_is_error_not_stack_overflow():
                                                                   failed)
 
 
+def test_deep_paren_nesting_in_query_grammar_is_error():
+    """Same stack-overflow class through the query-grammar parser:
+    200k nested parens in user_input must fail the split, not the
+    process."""
+    from quickwit_amd import proto, splitgen
+    from quickwit_amd.api import OracleSearcher, make_leaf_request
+    s = OracleSearcher()
+    s.add_split("g", splitgen.generate_split(0, 200, seed=1))
+    for depth, ok in ((100, True), (200_000, False)):
+        txt = "(" * depth + "body:w00001" + ")" * depth
+        q = {"type": "user_input", "user_text": txt,
+             "default_fields": ["body"]}
+        resp = proto.decode(
+            "LeafSearchResponse",
+            s.leaf_search_raw(proto.encode(
+                "LeafSearchRequest",
+                make_leaf_request(q, splitgen.HDFS_SCHEMA, [("g", 200)],
+                                  max_hits=5))))
+        failed = resp.get("failed_splits", [])
+        if ok:
+            assert not failed and resp.get("num_hits", 0) > 0, depth
+        else:
+            assert failed and "recursion" in failed[0]["error"], (depth,
+                                                                  failed)
+
+
 def test_corrupted_merge_inputs_never_crash():
     """The rank-0 merge path consumes response bytes and QAGG1 blobs that
     crossed the wire: qw_merge_leaf_responses and qw_finalize_agg_to_json
