@@ -1017,6 +1017,13 @@ inline std::string finalize_aggs_json(const IntermediateAggResults& ir,
             size_t bi = 0;
             if (hi >= lo && (!bs.empty() || d.has_bounds)) {
                 long long nb = (long long)llround((hi - lo) / d.interval);
+                // tantivy's AggregationLimits default bucket cap
+                // (leaf.rs:722-725 guard): a degenerate interval must be
+                // an error, not a near-infinite gap-fill loop
+                if (nb < 0 || nb >= 65000)
+                    throw std::runtime_error(
+                        "aggregation bucket limit (65000) exceeded in "
+                        "histogram gap fill");
                 for (long long k = 0; k <= nb; ++k) {
                     double key = lo + double(k) * d.interval;
                     const AggBucket* b = nullptr;

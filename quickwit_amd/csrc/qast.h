@@ -1115,6 +1115,9 @@ inline std::vector<AggDef> parse_agg_request(const std::string& json) {
             const mj::Value* fi = spec->get("fixed_interval");
             if (!fi) throw std::runtime_error("date_histogram requires fixed_interval");
             a.interval = parse_duration_ms(fi->s);
+            if (!(a.interval > 0))
+                throw std::runtime_error(
+                    "date_histogram fixed_interval must be > 0");
             if (const mj::Value* off = spec->get("offset"))
                 a.offset = off->kind == mj::Value::STR ? parse_duration_ms(off->s)
                                                        : off->num();
@@ -1127,6 +1130,8 @@ inline std::vector<AggDef> parse_agg_request(const std::string& json) {
             a.kind = AggDef::HISTOGRAM;
             a.field = spec->at("field")->s;
             a.interval = spec->at("interval")->num();
+            if (!(a.interval > 0))
+                throw std::runtime_error("histogram interval must be > 0");
             if (const mj::Value* off = spec->get("offset")) a.offset = off->num();
             if (const mj::Value* eb = spec->get("extended_bounds")) {
                 a.has_bounds = true;

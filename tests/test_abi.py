@@ -143,6 +143,47 @@ def test_deep_paren_nesting_in_query_grammar_is_error():
                                                                   failed)
 
 
+def test_degenerate_histogram_intervals_error_not_hang():
+    """interval <= 0 is rejected at parse (tantivy errors on it) and a
+    pathologically tiny interval trips the 65000-bucket gap-fill cap
+    (tantivy's AggregationLimits default) at finalize — before the fix,
+    interval=1e-12 looped over ~1e15 empty buckets."""
+    from quickwit_amd import proto, splitgen
+    from quickwit_amd.api import OracleSearcher, make_leaf_request
+    s = OracleSearcher()
+    s.add_split("a", splitgen.generate_split(0, 500, seed=1))
+
+    def leaf(aggs):
+        req = make_leaf_request({"type": "match_all"}, splitgen.HDFS_SCHEMA,
+                                [("a", 500)], max_hits=0, aggregation=aggs)
+        return proto.decode(
+            "LeafSearchResponse",
+            s.leaf_search_raw(proto.encode("LeafSearchRequest", req)))
+
+    for aggs in (
+        {"h": {"date_histogram": {"field": "timestamp",
+                                  "fixed_interval": "0ms"}}},
+        {"h": {"date_histogram": {"field": "timestamp",
+                                  "fixed_interval": "-5ms"}}},
+        {"h": {"histogram": {"field": "tenant_id", "interval": 0}}},
+    ):
+        failed = leaf(aggs).get("failed_splits", [])
+        assert failed and "interval" in failed[0]["error"], aggs
+
+    tiny = {"h": {"histogram": {"field": "tenant_id", "interval": 1e-12}}}
+    resp = leaf(tiny)
+    assert not resp.get("failed_splits")  # sparse collection is fine
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError):   # gap-fill cap trips at finalize
+        s.finalize_agg_json(resp["intermediate_aggregation_result"], tiny)
+
+    ok = {"h": {"date_histogram": {"field": "timestamp",
+                                   "fixed_interval": "3600000ms"}}}
+    resp = leaf(ok)
+    j = s.finalize_agg_json(resp["intermediate_aggregation_result"], ok)
+    assert len(j["h"]["buckets"]) > 0  # normal path unaffected
+
+
 def test_corrupted_merge_inputs_never_crash():
     """The rank-0 merge path consumes response bytes and QAGG1 blobs that
     crossed the wire: qw_merge_leaf_responses and qw_finalize_agg_to_json
