@@ -1016,6 +1016,9 @@ static SplitResult search_split(const SplitView& sv, const pb::SearchRequest& re
     };
 
     // sort specs (max 2, like the reference — search.proto:269)
+    if (req.sort_fields.size() > 2)
+        throw std::runtime_error(
+            "more than two sort fields (search.proto:269)");
     std::vector<SortSpec> specs;
     for (auto& sf : req.sort_fields) {
         SortSpec s;
