@@ -1243,11 +1243,15 @@ int qw_oracle_leaf_search(qw_oracle_ctx* ctx, const uint8_t* req_pb, size_t req_
             bool first_seg;             // first segment of its split
         };
         std::vector<Task> tasks;
+        std::vector<std::string> pre_failed;  // unknown splits -> data
         for (auto& lr : lreq.leaf_requests)
             for (auto& so : lr.split_offsets) {
                 auto it = ctx->splits.find(so.split_id);
-                if (it == ctx->splits.end())
-                    throw std::runtime_error("unknown split: " + so.split_id);
+                if (it == ctx->splits.end()) {
+                    // per-split failure as data (leaf.rs:2143-2148)
+                    pre_failed.push_back(so.split_id);
+                    continue;
+                }
                 const auto* holder = it->second.get();
                 if (holder->seg_views.empty())
                     tasks.push_back({&holder->view, so.split_id, true});
@@ -1273,6 +1277,14 @@ int qw_oracle_leaf_search(qw_oracle_ctx* ctx, const uint8_t* req_pb, size_t req_
         pb::LeafSearchResponse resp;
         for (auto& t : tasks)
             if (t.first_seg) resp.num_attempted_splits++;
+        for (auto& sid : pre_failed) {
+            resp.num_attempted_splits++;
+            pb::SplitSearchError se;
+            se.error = "unknown split: " + sid;
+            se.split_id = sid;
+            se.retryable_error = true;
+            resp.failed_splits.push_back(std::move(se));
+        }
         IntermediateAggResults merged_aggs;
         bool any_aggs = false;
         std::vector<pb::PartialHit> all_hits;

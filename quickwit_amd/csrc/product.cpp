@@ -2939,12 +2939,20 @@ int32_t qw_leaf_search(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
             pb::SearchRequest req;  // per-split request (may be demoted)
         };
         std::vector<Task> tasks;
+        std::vector<pb::SplitSearchError> pre_failed;
         for (auto& lr : lreq.leaf_requests)
             for (auto& so : lr.split_offsets) {
                 auto it = ctx->splits.find(so.split_id);
                 if (it == ctx->splits.end()) {
-                    set_err(ctx, "unknown split: " + so.split_id);
-                    return QW_ERR_NOT_FOUND;
+                    // a split that cannot be opened is a per-split failure
+                    // inside the response (leaf.rs:2143-2148), not a
+                    // whole-call error — the other splits still answer
+                    pb::SplitSearchError se;
+                    se.error = "unknown split: " + so.split_id;
+                    se.split_id = so.split_id;
+                    se.retryable_error = true;
+                    pre_failed.push_back(std::move(se));
+                    continue;
                 }
                 tasks.push_back({so, it->second.get(), req});
             }
@@ -3014,6 +3022,10 @@ int32_t qw_leaf_search(qw_ctx* ctx, const uint8_t* req_pb, size_t req_len,
         }
 
         pb::LeafSearchResponse resp;
+        for (auto& se : pre_failed) {
+            resp.num_attempted_splits++;
+            resp.failed_splits.push_back(std::move(se));
+        }
         IntermediateAggResults merged_aggs;
         bool any_aggs = false;
         std::vector<pb::PartialHit> all_hits;

@@ -204,6 +204,27 @@ def test_more_than_two_sort_fields_rejected():
     assert failed and "two sort fields" in failed[0]["error"]
 
 
+def test_unknown_split_is_per_split_failure_not_call_error():
+    """A split the ctx cannot open is a failed_splits entry inside the
+    response (leaf.rs:2143-2148) and the other splits still answer —
+    previously it failed the whole call with QW_ERR_NOT_FOUND."""
+    from quickwit_amd import proto, splitgen
+    from quickwit_amd.api import OracleSearcher, make_leaf_request
+    s = OracleSearcher()
+    s.add_split("a", splitgen.generate_split(0, 500, seed=1))
+    req = make_leaf_request({"type": "match_all"}, splitgen.HDFS_SCHEMA,
+                            [("a", 500), ("ghost", 100)], max_hits=3)
+    resp = proto.decode(
+        "LeafSearchResponse",
+        s.leaf_search_raw(proto.encode("LeafSearchRequest", req)))
+    assert resp["num_hits"] == 500
+    assert resp["num_attempted_splits"] == 2
+    assert resp["num_successful_splits"] == 1
+    failed = resp.get("failed_splits", [])
+    assert [f["split_id"] for f in failed] == ["ghost"]
+    assert "unknown split" in failed[0]["error"]
+
+
 def test_corrupted_merge_inputs_never_crash():
     """The rank-0 merge path consumes response bytes and QAGG1 blobs that
     crossed the wire: qw_merge_leaf_responses and qw_finalize_agg_to_json
