@@ -1134,6 +1134,11 @@ static SplitResult search_split(const SplitView& sv, const pb::SearchRequest& re
 
     if (req.aggregation_request) {
         std::vector<AggDef> defs = parse_agg_request(*req.aggregation_request);
+        auto agg_col_type = [&](const std::string& name) -> int {
+            const FastFieldView* cf = sv.fast_field(name);
+            return cf ? int(cf->type) : -1;
+        };
+        validate_agg_fields(defs, agg_col_type);
         out.aggs = collect_aggs(defs, sv, m);
         out.has_aggs = true;
     }

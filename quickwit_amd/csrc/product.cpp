@@ -868,6 +868,13 @@ static AggPlan plan_aggs(const SplitView& sv, const std::string& agg_json,
                          int64_t bucket_limit, uint64_t out_base) {
     AggPlan ap;
     ap.defs = parse_agg_request(agg_json);
+    {
+        auto agg_col_type = [&](const std::string& name) -> int {
+            const FastFieldView* cf = sv.fast_field(name);
+            return cf ? int(cf->type) : -1;
+        };
+        validate_agg_fields(ap.defs, agg_col_type);
+    }
     uint64_t off = out_base;
     for (const AggDef& d : ap.defs) {
         const FastFieldView* f = sv.fast_field(d.field);

@@ -1101,6 +1101,67 @@ inline double parse_duration_ms(const std::string& s) {
     throw std::runtime_error("bad interval: " + s);
 }
 
+// Validate aggregation targets against the split's fast columns:
+// wrong-typed PRESENT columns are errors (tantivy rejects e.g.
+// date_histogram over a text column); absent columns are allowed and
+// contribute empty results. col_type(name) returns
+// the FastFieldView::Type as int, or -1 when the field has no fast column.
+// Type codes: 0=U64 1=I64 2=DATETIME 3=STR 4=F64 5=MIXED.
+template <class ColType>
+inline void validate_agg_fields(const std::vector<AggDef>& defs,
+                                ColType&& col_type) {
+    // A column absent from this split is NOT an error: tantivy opens
+    // columns optionally and an empty column contributes empty results
+    // (the aggregations golden aggregates on fields some splits lack).
+    // Only a PRESENT column of an incompatible type is an error.
+    auto require = [&](const std::string& field) -> int {
+        return col_type(field);  // -1 = absent, allowed everywhere
+    };
+    auto metric_ok = [&](const MetricAgg& m) {
+        int t = require(m.field);
+        // value_count works on any column; other metrics need numbers
+        if (t >= 0 && m.kind != MetricAgg::COUNT && t == 3 /*STR*/)
+            throw std::runtime_error(
+                "metric aggregation on non-numeric field: " + m.field);
+    };
+    for (const AggDef& d : defs) {
+        switch (d.kind) {
+            case AggDef::DATE_HISTOGRAM: {
+                int t = require(d.field);
+                if (t >= 0 && t != 2 /*DATETIME*/)
+                    throw std::runtime_error(
+                        "date_histogram on non-datetime field: " + d.field);
+                break;
+            }
+            case AggDef::HISTOGRAM:
+            case AggDef::RANGE: {
+                int t = require(d.field);
+                if (t == 3 /*STR*/)
+                    throw std::runtime_error(
+                        "histogram/range on non-numeric field: " + d.field);
+                break;
+            }
+            case AggDef::TERMS:
+            case AggDef::CARDINALITY:
+                require(d.field);
+                break;
+            case AggDef::METRIC:
+                metric_ok(d.metric);
+                break;
+            case AggDef::COMPOSITE:
+                for (const auto& cs : d.comp) {
+                    int t = require(cs.field);
+                    if (cs.is_histo && t == 3 /*STR*/)
+                        throw std::runtime_error(
+                            "composite histogram source on non-numeric "
+                            "field: " + cs.field);
+                }
+                break;
+        }
+        for (const MetricAgg& m : d.sub) metric_ok(m);
+    }
+}
+
 inline std::vector<AggDef> parse_agg_request(const std::string& json) {
     std::vector<AggDef> out;
     mj::ValuePtr root = mj::parse(json);

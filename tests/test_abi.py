@@ -225,6 +225,40 @@ def test_unknown_split_is_per_split_failure_not_call_error():
     assert "unknown split" in failed[0]["error"]
 
 
+def test_agg_field_type_mismatch_is_error_absent_is_empty():
+    """tantivy semantics: an aggregation over a PRESENT column of an
+    incompatible type is an error (date_histogram on u64, stats on str);
+    a column ABSENT from the split contributes empty results (columns
+    open optionally — the aggregations golden relies on this)."""
+    from quickwit_amd import proto, splitgen
+    from quickwit_amd.api import OracleSearcher, make_leaf_request
+    s = OracleSearcher()
+    s.add_split("a", splitgen.generate_split(0, 500, seed=1))
+
+    def failed(aggs):
+        req = make_leaf_request({"type": "match_all"}, splitgen.HDFS_SCHEMA,
+                                [("a", 500)], max_hits=0, aggregation=aggs)
+        resp = proto.decode(
+            "LeafSearchResponse",
+            s.leaf_search_raw(proto.encode("LeafSearchRequest", req)))
+        return [f["error"] for f in resp.get("failed_splits", [])]
+
+    assert "non-datetime" in failed(
+        {"h": {"date_histogram": {"field": "tenant_id",
+                                  "fixed_interval": "1000ms"}}})[0]
+    assert "non-numeric" in failed(
+        {"m": {"stats": {"field": "tenant_name"}}})[0]
+    assert "non-numeric" in failed(
+        {"h": {"histogram": {"field": "tenant_name", "interval": 5}}})[0]
+    # absent columns: empty results, no failure
+    assert failed({"h": {"date_histogram": {"field": "nope",
+                                            "fixed_interval": "1000ms"}}}) \
+        == []
+    assert failed({"m": {"stats": {"field": "nope"}}}) == []
+    # value_count works on any column type
+    assert failed({"m": {"value_count": {"field": "tenant_name"}}}) == []
+
+
 def test_corrupted_merge_inputs_never_crash():
     """The rank-0 merge path consumes response bytes and QAGG1 blobs that
     crossed the wire: qw_merge_leaf_responses and qw_finalize_agg_to_json
