@@ -320,7 +320,8 @@ inline int64_t parse_datetime_ms(const mj::Value* lit);
 // RFC3339 (subset: YYYY-MM-DD['T'HH:MM[:SS[.fff]]][Z|±hh:mm]) -> epoch ms
 inline int64_t rfc3339_to_ms(const std::string& s);
 
-inline Bound parse_bound(const mj::Value* v, const SchemaField& f) {
+inline Bound parse_bound(const mj::Value* v, const SchemaField& f,
+                         bool is_upper = false) {
     Bound b;
     if (!v) return b;
     // serde's built-in std::ops::Bound impl: {"Included": lit} /
@@ -359,6 +360,22 @@ inline Bound parse_bound(const mj::Value* v, const SchemaField& f) {
         b.ival = strtoll(lit->s.c_str(), nullptr, 10);
         b.fval = atof(lit->s.c_str());
     } else throw std::runtime_error("bad range literal");
+    if (f.type == "u64" && (b.ival < 0 || (b.from_f64 && b.fval < 0))) {
+        // negative bound against an unsigned column (tantivy coerces at
+        // the column's domain edge): lower clamps to 0 (all docs satisfy
+        // it), upper becomes the empty range x < 0
+        if (is_upper) {
+            b.kind = Bound::EXCLUDED;
+            b.ival = 0;
+            b.fval = 0;
+            b.from_f64 = false;
+        } else {
+            b.kind = Bound::INCLUDED;
+            b.ival = 0;
+            b.fval = 0;
+            b.from_f64 = false;
+        }
+    }
     return b;
 }
 
@@ -813,8 +830,51 @@ inline PlanNode build_plan(const mj::Value* ast, const Schema& schema) {
         n.kind = PlanNode::TERM;
         n.field = ast->at("field")->s;
         n.value = ast->at("value")->s;
-        if (!schema.field(n.field))
+        const SchemaField* tf = schema.field(n.field);
+        if (!tf)
             throw std::runtime_error("term on unknown field: " + n.field);
+        if (tf->type != "text") {
+            // fast-only column: term = equality over the fast field (the
+            // reference routes term queries on fast-only fields through
+            // the fast-field range machinery; our REST shim's
+            // point_or_term does the same)
+            PlanNode r;
+            r.kind = PlanNode::RANGE;
+            r.field = n.field;
+            Bound b;
+            b.kind = Bound::INCLUDED;
+            if (tf->type == "str") {
+                b.sval = n.value;
+            } else if (tf->type == "datetime") {
+                mj::Value lit;
+                lit.kind = mj::Value::STR;
+                lit.s = n.value;
+                b.ival = parse_datetime_ms(&lit);
+            } else {
+                char* endp = nullptr;
+                errno = 0;
+                if (tf->type == "f64") {
+                    b.from_f64 = true;
+                    b.fval = strtod(n.value.c_str(), &endp);
+                    b.ival = int64_t(b.fval);
+                } else {
+                    long long v = strtoll(n.value.c_str(), &endp, 10);
+                    b.ival = v;
+                    b.fval = double(v);
+                }
+                if (errno != 0 || !endp || *endp != 0)
+                    throw std::runtime_error(
+                        "invalid term value for numeric field " + n.field);
+                if (tf->type == "u64" && b.ival < 0) {
+                    PlanNode none;
+                    none.kind = PlanNode::MATCH_NONE;
+                    return none;
+                }
+            }
+            r.lo = b;
+            r.hi = b;
+            return r;
+        }
     } else if (ty == "full_text") {
         std::string op = "or";
         bool zta = false;
@@ -863,8 +923,8 @@ inline PlanNode build_plan(const mj::Value* ast, const Schema& schema) {
         const SchemaField* f = schema.field(n.field);
         if (!f || !f->fast)
             throw std::runtime_error("range on non-fast field: " + n.field);
-        n.lo = parse_bound(ast->get("lower_bound"), *f);
-        n.hi = parse_bound(ast->get("upper_bound"), *f);
+        n.lo = parse_bound(ast->get("lower_bound"), *f, false);
+        n.hi = parse_bound(ast->get("upper_bound"), *f, true);
     } else if (ty == "field_presence") {
         n.kind = PlanNode::FIELD_PRESENCE;
         n.field = ast->at("field")->s;

@@ -259,6 +259,48 @@ def test_agg_field_type_mismatch_is_error_absent_is_empty():
     assert failed({"m": {"value_count": {"field": "tenant_name"}}}) == []
 
 
+def test_term_on_fast_only_column_and_negative_u64_bounds():
+    """Two query-plan parity fixes pinned: (1) a term query on a fast-only
+    (non-text) column is equality over the fast field — previously it
+    silently matched nothing; (2) a negative range bound against a u64
+    column clamps at the domain edge (lower -> 0, upper -> empty) instead
+    of wrapping to a huge unsigned value."""
+    from quickwit_amd import proto, splitgen
+    from quickwit_amd.api import OracleSearcher, make_leaf_request
+    s = OracleSearcher()
+    s.add_split("a", splitgen.generate_split(0, 500, seed=1))
+
+    def hits(q):
+        req = make_leaf_request(q, splitgen.HDFS_SCHEMA, [("a", 500)],
+                                max_hits=3)
+        resp = proto.decode(
+            "LeafSearchResponse",
+            s.leaf_search_raw(proto.encode("LeafSearchRequest", req)))
+        return (resp.get("num_hits", 0),
+                [f["error"] for f in resp.get("failed_splits", [])])
+
+    n_0_100, _ = hits({"type": "range", "field": "tenant_id",
+                       "lower_bound": {"included": 0},
+                       "upper_bound": {"included": 100}})
+    n_neg, _ = hits({"type": "range", "field": "tenant_id",
+                     "lower_bound": {"included": -50},
+                     "upper_bound": {"included": 100}})
+    assert n_neg == n_0_100 > 0
+    assert hits({"type": "range", "field": "tenant_id",
+                 "upper_bound": {"included": -5}})[0] == 0
+
+    n_range, _ = hits({"type": "range", "field": "tenant_id",
+                       "lower_bound": {"included": 17},
+                       "upper_bound": {"included": 17}})
+    n_term, _ = hits({"type": "term", "field": "tenant_id", "value": "17"})
+    assert n_term == n_range > 0
+    assert hits({"type": "term", "field": "tenant_name",
+                 "value": "t0001"})[0] > 0
+    assert hits({"type": "term", "field": "tenant_id", "value": "-3"})[0] == 0
+    _, failed = hits({"type": "term", "field": "tenant_id", "value": "abc"})
+    assert failed and "invalid term value" in failed[0]
+
+
 def test_corrupted_merge_inputs_never_crash():
     """The rank-0 merge path consumes response bytes and QAGG1 blobs that
     crossed the wire: qw_merge_leaf_responses and qw_finalize_agg_to_json
