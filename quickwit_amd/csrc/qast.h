@@ -379,6 +379,59 @@ inline Bound parse_bound(const mj::Value* v, const SchemaField& f,
     return b;
 }
 
+// TERM leaf over any schema field: text fields use the inverted index;
+// fast-only columns become an equality range over the fast field (the
+// reference routes such terms through the fast-field machinery).
+inline PlanNode term_leaf_plan(const std::string& field,
+                               const std::string& value,
+                               const Schema& schema) {
+    const SchemaField* tf = schema.field(field);
+    if (!tf) throw std::runtime_error("term on unknown field: " + field);
+    PlanNode n;
+    if (tf->type == "text") {
+        n.kind = PlanNode::TERM;
+        n.field = field;
+        n.value = value;
+        return n;
+    }
+    PlanNode r;
+    r.kind = PlanNode::RANGE;
+    r.field = field;
+    Bound b;
+    b.kind = Bound::INCLUDED;
+    if (tf->type == "str") {
+        b.sval = value;
+    } else if (tf->type == "datetime") {
+        mj::Value lit;
+        lit.kind = mj::Value::STR;
+        lit.s = value;
+        b.ival = parse_datetime_ms(&lit);
+    } else {
+        char* endp = nullptr;
+        errno = 0;
+        if (tf->type == "f64") {
+            b.from_f64 = true;
+            b.fval = strtod(value.c_str(), &endp);
+            b.ival = int64_t(b.fval);
+        } else {
+            long long v = strtoll(value.c_str(), &endp, 10);
+            b.ival = v;
+            b.fval = double(v);
+        }
+        if (errno != 0 || !endp || *endp != 0)
+            throw std::runtime_error(
+                "invalid term value for numeric field " + field);
+        if (tf->type == "u64" && b.ival < 0) {
+            PlanNode none;
+            none.kind = PlanNode::MATCH_NONE;
+            return none;
+        }
+    }
+    r.lo = b;
+    r.hi = b;
+    return r;
+}
+
 inline PlanNode build_plan(const mj::Value* ast, const Schema& schema);
 
 inline PlanNode full_text_plan(const std::string& field, const std::string& text,
@@ -827,54 +880,8 @@ inline PlanNode build_plan(const mj::Value* ast, const Schema& schema) {
     } else if (ty == "match_none") {
         n.kind = PlanNode::MATCH_NONE;
     } else if (ty == "term") {
-        n.kind = PlanNode::TERM;
-        n.field = ast->at("field")->s;
-        n.value = ast->at("value")->s;
-        const SchemaField* tf = schema.field(n.field);
-        if (!tf)
-            throw std::runtime_error("term on unknown field: " + n.field);
-        if (tf->type != "text") {
-            // fast-only column: term = equality over the fast field (the
-            // reference routes term queries on fast-only fields through
-            // the fast-field range machinery; our REST shim's
-            // point_or_term does the same)
-            PlanNode r;
-            r.kind = PlanNode::RANGE;
-            r.field = n.field;
-            Bound b;
-            b.kind = Bound::INCLUDED;
-            if (tf->type == "str") {
-                b.sval = n.value;
-            } else if (tf->type == "datetime") {
-                mj::Value lit;
-                lit.kind = mj::Value::STR;
-                lit.s = n.value;
-                b.ival = parse_datetime_ms(&lit);
-            } else {
-                char* endp = nullptr;
-                errno = 0;
-                if (tf->type == "f64") {
-                    b.from_f64 = true;
-                    b.fval = strtod(n.value.c_str(), &endp);
-                    b.ival = int64_t(b.fval);
-                } else {
-                    long long v = strtoll(n.value.c_str(), &endp, 10);
-                    b.ival = v;
-                    b.fval = double(v);
-                }
-                if (errno != 0 || !endp || *endp != 0)
-                    throw std::runtime_error(
-                        "invalid term value for numeric field " + n.field);
-                if (tf->type == "u64" && b.ival < 0) {
-                    PlanNode none;
-                    none.kind = PlanNode::MATCH_NONE;
-                    return none;
-                }
-            }
-            r.lo = b;
-            r.hi = b;
-            return r;
-        }
+        return term_leaf_plan(ast->at("field")->s, ast->at("value")->s,
+                              schema);
     } else if (ty == "full_text") {
         std::string op = "or";
         bool zta = false;
@@ -942,13 +949,8 @@ inline PlanNode build_plan(const mj::Value* ast, const Schema& schema) {
         for (auto& kv : tpf->obj) {
             if (!schema.field(kv.first))
                 throw std::runtime_error("term_set on unknown field: " + kv.first);
-            for (auto& v : kv.second->arr) {
-                PlanNode t;
-                t.kind = PlanNode::TERM;
-                t.field = kv.first;
-                t.value = v->s;
-                n.should.push_back(std::move(t));
-            }
+            for (auto& v : kv.second->arr)
+                n.should.push_back(term_leaf_plan(kv.first, v->s, schema));
         }
         if (n.should.empty()) n.kind = PlanNode::MATCH_NONE;
     } else if (ty == "cache") {
