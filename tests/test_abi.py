@@ -299,6 +299,13 @@ def test_term_on_fast_only_column_and_negative_u64_bounds():
     assert hits({"type": "term", "field": "tenant_id", "value": "-3"})[0] == 0
     _, failed = hits({"type": "term", "field": "tenant_id", "value": "abc"})
     assert failed and "invalid term value" in failed[0]
+    # term_set shares the same leaf helper
+    n_set, _ = hits({"type": "term_set", "terms_per_field":
+                     {"tenant_name": ["t0001", "t0002"]}})
+    n_bool, _ = hits({"type": "bool", "should": [
+        {"type": "term", "field": "tenant_name", "value": "t0001"},
+        {"type": "term", "field": "tenant_name", "value": "t0002"}]})
+    assert n_set == n_bool > 0
 
 
 def test_corrupted_merge_inputs_never_crash():
