@@ -45,6 +45,9 @@ def rand_term(rng):
                   "zz_absent_%d" % rng.randrange(3)]),
         ("severity_text", ["DEBUG", "INFO", "WARN", "ERROR", "FATAL",
                            "nope"]),
+        # fast-only columns: term = equality over the fast field (r2)
+        ("tenant_id", ["0", "17", "312", "999", "5000"]),
+        ("tenant_name", ["t0001", "t0042", "t0999", "absent"]),
     ])
     return {"type": "term", "field": field, "value": rng.choice(pool)}
 
