@@ -169,6 +169,25 @@ def distributed_merge(sreq_pb: bytes, resp_pb: bytes, split_ids: list,
             dense.append([i, e, params[e.name][0], params[e.name][1]])
         else:
             e.buckets = None  # opaque residual
+    # Structure agreement BEFORE the variable-shaped meta reduce: the dense
+    # list is derived from each rank's OWN blob, so a rank whose split
+    # failed (empty blob) would build a different-shaped tensor and either
+    # deadlock or crash the collective. One fixed-shape reduce detects any
+    # disagreement (count + order/sub-structure hash, min+max of both);
+    # on mismatch every rank falls back to the sideband path.
+    if params and dist.get_world_size() > 1:
+        sig = 0
+        for i, e, _iv, _ofs in dense:
+            sig = (sig * 1000003 + i * 131 + len(e.sub_kinds) * 7
+                   + 1) & 0x7FFFFFFF
+        t_sig = torch.tensor([len(dense), sig, -len(dense), -sig],
+                             dtype=torch.int64, device=device)
+        dist.all_reduce(t_sig, op=dist.ReduceOp.MAX)
+        mx_n, mx_s, mn_n, mn_s = (int(x) for x in t_sig.cpu())
+        if mx_n != -mn_n or mx_s != -mn_s:
+            for _i, e, _iv, _ofs in dense:
+                e.buckets = None  # opaque residual: rides the sideband
+            dense = []
     # per-agg index ranges; off-grid keys disqualify the agg (structural
     # eligibility is identical on every rank, so the fallback must be too:
     # decided from the request+blob grid check, then AND-reduced)

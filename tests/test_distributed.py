@@ -292,6 +292,69 @@ def test_two_rank_merge_carries_failed_splits_as_data():
         expected.get("num_attempted_splits")
 
 
+def _rank_main_agg_failure(rank, world, port, result):
+    import torch.distributed as dist
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    import torch
+
+    from quickwit_amd.api import OracleSearcher, make_leaf_request
+    from quickwit_amd.merge import distributed_merge
+
+    s = OracleSearcher()
+    s.add_split("af-0", splitgen.generate_split(0, NDOCS, seed=3))
+    # rank 1 queries a split its ctx does not hold: per-split failure,
+    # EMPTY agg blob — the dense reduce must not deadlock on the shape
+    # mismatch and falls back to the sideband for every rank
+    sid = "af-0" if rank == 0 else "af-missing"
+    req = make_leaf_request({"type": "match_all"}, splitgen.HDFS_SCHEMA,
+                            [(sid, NDOCS)], max_hits=5, aggregation=AGGS)
+    resp_pb = s.leaf_search_raw(proto.encode("LeafSearchRequest", req))
+    sreq_pb = proto.encode("SearchRequest", req["search_request"])
+    merged = distributed_merge(sreq_pb, resp_pb, [sid],
+                               device=torch.device("cpu"))
+    if rank == 0:
+        result.put(merged)
+    dist.destroy_process_group()
+
+
+def test_two_rank_merge_with_one_empty_agg_blob_does_not_deadlock():
+    """Rank 1's only split fails, so its response carries NO aggregation
+    blob while rank 0's does. The dense bucket reduce derives its tensor
+    shape from the local blob; without the structure-agreement reduce
+    this mismatched collective deadlocked. Merged result == rank 0's
+    aggs + rank 1's failure as data."""
+    import __graft_entry__
+    __graft_entry__.build()
+    ctx = mp.get_context("spawn")
+    result = ctx.Queue()
+    procs = [ctx.Process(target=_rank_main_agg_failure,
+                         args=(r, 2, 29517, result)) for r in range(2)]
+    for p in procs:
+        p.start()
+    merged = proto.decode("LeafSearchResponse", result.get(timeout=180))
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    from quickwit_amd.api import OracleSearcher, make_leaf_request
+    solo = OracleSearcher()
+    solo.add_split("af-0", splitgen.generate_split(0, NDOCS, seed=3))
+    expected = solo.leaf_search(make_leaf_request(
+        {"type": "match_all"}, splitgen.HDFS_SCHEMA, [("af-0", NDOCS)],
+        max_hits=5, aggregation=AGGS))
+    assert merged["num_hits"] == expected["num_hits"]
+    assert [f["split_id"] for f in merged.get("failed_splits", [])] == \
+        ["af-missing"]
+    gj = solo.finalize_agg_json(merged["intermediate_aggregation_result"],
+                                AGGS)
+    ej = solo.finalize_agg_json(expected["intermediate_aggregation_result"],
+                                AGGS)
+    assert gj == ej
+
+
 def test_two_rank_packed_merge_multisegment_splits():
     """Packed 32B hit records carry (split_ord, segment_ord) across ranks:
     two ranks each holding a 2-segment QWA2 split must merge exactly like
