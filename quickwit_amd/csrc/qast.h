@@ -508,6 +508,7 @@ struct QTok {
     std::string field;   // LIT: optional field ("" = default fields)
     std::string text;    // LIT: raw literal text (quotes stripped)
     bool quoted = false;
+    bool prefix = false;  // quoted phrase followed by '*' (phrase prefix)
     bool range_ge = false, range_gt = false, range_le = false, range_lt = false;
     bool bracket = false;  // [a TO b] / {a TO b}
     bool lo_incl = true, hi_incl = true;
@@ -520,11 +521,18 @@ struct QTok {
 inline std::vector<QTok> qlex(const std::string& s) {
     std::vector<QTok> out;
     size_t i = 0;
-    auto read_quoted = [&](std::string* t) {
+    auto read_quoted = [&](std::string* t, char q) {
         ++i;  // opening quote
-        while (i < s.size() && s[i] != '"') t->push_back(s[i++]);
+        while (i < s.size() && s[i] != q) t->push_back(s[i++]);
         if (i >= s.size()) throw std::runtime_error("unterminated quote");
         ++i;
+    };
+    auto after_quote_star = [&]() {  // "phrase"* -> prefix
+        if (i < s.size() && s[i] == '*') {
+            ++i;
+            return true;
+        }
+        return false;
     };
     while (i < s.size()) {
         char c = s[i];
@@ -541,9 +549,10 @@ inline std::vector<QTok> qlex(const std::string& s) {
         // word or field:...
         QTok t;
         std::string word;
-        if (c == '"') {
-            read_quoted(&t.text);
+        if (c == '"' || c == '\'') {
+            read_quoted(&t.text, c);
             t.quoted = true;
+            t.prefix = after_quote_star();
         } else {
             while (i < s.size() && s[i] != ' ' && s[i] != '\t' && s[i] != '(' &&
                    s[i] != ')' && s[i] != ':' && s[i] != '^')
@@ -552,9 +561,10 @@ inline std::vector<QTok> qlex(const std::string& s) {
                 t.field = word;
                 ++i;
                 // value part
-                if (i < s.size() && s[i] == '"') {
-                    read_quoted(&t.text);
+                if (i < s.size() && (s[i] == '"' || s[i] == '\'')) {
+                    read_quoted(&t.text, s[i]);
                     t.quoted = true;
+                    t.prefix = after_quote_star();
                 } else if (i < s.size() && (s[i] == '[' || s[i] == '{')) {
                     t.bracket = true;
                     t.lo_incl = s[i] == '[';
@@ -816,6 +826,23 @@ inline PlanNode qtok_leaf_impl(const QTok& t, const std::vector<std::string>& df
                 tokenize(t.text, f ? f->tokenizer : "default");
             if (toks.empty()) {
                 n.kind = PlanNode::MATCH_NONE;
+                return n;
+            }
+            if (t.prefix) {
+                // "phrase"* phrase-prefix: supported where it reduces to a
+                // single-token prefix (e.g. raw tokenizer), as a wildcard
+                if (toks.size() != 1)
+                    throw std::runtime_error(
+                        "multi-token phrase prefix not supported");
+                if (toks[0].find('*') != std::string::npos ||
+                    toks[0].find('?') != std::string::npos)
+                    throw std::runtime_error(
+                        "phrase prefix with wildcard metachars");
+                n.kind = PlanNode::WILDCARD;
+                n.const_score = true;
+                n.field = t.field;
+                n.value = toks[0] + "*";
+                n.boost = t.boost;
                 return n;
             }
             if (toks.size() > 1) {
