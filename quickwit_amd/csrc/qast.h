@@ -858,9 +858,29 @@ inline PlanNode qtok_leaf_impl(const QTok& t, const std::vector<std::string>& df
             n.boost = t.boost;
             return n;
         }
-        PlanNode p = full_text_plan(t.field, t.text, "or", schema);
-        p.boost *= t.boost;
-        return p;
+        // unquoted value: a single grammar word; when the field's analyzer
+        // splits it into several tokens (1.5 -> [1,5], AB-CD -> [ab,cd])
+        // the reference's parser emits a slop-0 PhraseQuery, not an OR
+        {
+            std::vector<std::string> toks =
+                tokenize(t.text, f ? f->tokenizer : "default");
+            if (toks.empty()) {
+                n.kind = PlanNode::MATCH_NONE;
+                return n;
+            }
+            if (toks.size() > 1) {
+                n.kind = PlanNode::PHRASE;
+                n.field = t.field;
+                n.phrase_toks = std::move(toks);
+                n.boost = t.boost;
+                return n;
+            }
+            n.kind = PlanNode::TERM;
+            n.field = t.field;
+            n.value = toks[0];
+            n.boost = t.boost;
+            return n;
+        }
     }
     if (dfs.empty()) throw std::runtime_error("no default search fields");
     PlanNode b;

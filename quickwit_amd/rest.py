@@ -315,6 +315,26 @@ class Index:
                     fm["name"] = prefix + fm["name"]
                     flat_mappings.append(fm)
         walk(dm.get("field_mappings", []))
+        # concatenate fields (doc_mapping type "concatenate"): the listed
+        # source fields' leaf values, stringified, indexed under ONE text
+        # field with the CONCAT field's tokenizer (scenarii/concat_fields)
+        self.concat_specs = []
+        self.explicit_top = {fm["name"] for fm in flat_mappings}
+        for fm in flat_mappings:
+            if fm["type"] == "concatenate":
+                self.concat_specs.append({
+                    "name": fm["name"],
+                    "sources": set(fm.get("concatenate_fields", [])),
+                    "include_dynamic":
+                        bool(fm.get("include_dynamic_fields", False))})
+                cf = {"name": fm["name"], "type": "text",
+                      "tokenizer": fm.get("tokenizer", "default")}
+                if cf["tokenizer"] != "raw":
+                    # splitting analyzers need positions: unquoted values
+                    # that analyze to several tokens query as slop-0
+                    # phrases (1.5 -> "1 5")
+                    cf["record"] = "position"
+                fields.append(cf)
         for fm in flat_mappings:
             t = fm["type"]
             name = fm["name"]
@@ -344,6 +364,10 @@ class Index:
         fields, ts_field = self._explicit_fields()
         explicit = {f["name"] for f in fields}
         flat_batches = [[flatten_doc(d) for d in b] for b in self.batches]
+        if getattr(self, "concat_specs", None):
+            for b in flat_batches:
+                for d in b:
+                    self._add_concat_values(d)
         mode = self.config.get("doc_mapping", {}).get("mode", "dynamic")
         if mode == "dynamic":  # quickwit's default mode
             inferred = _infer_dynamic_fields(explicit, flat_batches)
@@ -375,6 +399,40 @@ class Index:
                     entry["timestamp_start"] = min(ts)
                     entry["timestamp_end"] = max(ts)
             self.splits.append(entry)
+
+    def _add_concat_values(self, flat):
+        def leaves_under(name):
+            vals = []
+            for k, v in flat.items():
+                if k == name or k.startswith(name + "."):
+                    vals.extend(v if isinstance(v, list) else [v])
+            return vals
+
+        def stringify(v):
+            if isinstance(v, bool):
+                return "true" if v else "false"
+            if isinstance(v, (int, float)):
+                return str(v)
+            return v if isinstance(v, str) else None
+
+        for spec in self.concat_specs:
+            out = []
+            for src in spec["sources"]:
+                for v in leaves_under(src):
+                    sv = stringify(v)
+                    if sv is not None:
+                        out.append(sv)
+            if spec["include_dynamic"]:
+                for k, v in flat.items():
+                    top = k.split(".", 1)[0]
+                    if top in self.explicit_top:
+                        continue
+                    for item in (v if isinstance(v, list) else [v]):
+                        sv = stringify(item)
+                        if sv is not None:
+                            out.append(sv)
+            if out:
+                flat[spec["name"]] = out
 
     def source_doc(self, split_id, doc_id):
         i = int(split_id.rsplit("-", 1)[1])

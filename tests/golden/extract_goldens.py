@@ -139,6 +139,9 @@ def rest_scenarios():
         "multi_splits": ["_setup.quickwit.yaml",
                          "0001-request-optimizations.yaml",
                          "_teardown.quickwit.yaml"],
+        "concat_fields": ["_setup.quickwit.yaml",
+                          "0001_concat_field.yaml",
+                          "_teardown.quickwit.yaml"],
         # the in-scope slice of the big ES-compat suite (gharchive corpus)
         "es_compatibility": ["_setup.quickwit.yaml", "0001-noquery.yaml",
                              "0003-match.yaml", "0006-term_query.yaml",

@@ -241,3 +241,25 @@ def test_rest_terms_with_subaggs_oracle():
         assert b["lat_st"]["count"] == b["doc_count"]
         assert abs(b["lat_avg"]["value"] -
                    b["lat_st"]["sum"] / b["lat_st"]["count"]) < 1e-9
+
+
+def run_concat(searcher_factory):
+    steps = load_suite("concat_fields")
+    client = make_client(searcher_factory)
+    ran, skipped = replay_suite(client, steps)
+    assert not skipped and ran == len(steps)
+
+
+def test_rest_concat_fields_suite_oracle():
+    from quickwit_amd.api import OracleSearcher
+    import __graft_entry__
+    __graft_entry__.build()
+    run_concat(OracleSearcher)
+
+
+@pytest.mark.gpu
+def test_rest_concat_fields_suite_gpu():
+    from quickwit_amd.api import GpuSearcher
+    import __graft_entry__
+    __graft_entry__.build()
+    run_concat(lambda: GpuSearcher(device=0))
