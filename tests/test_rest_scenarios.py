@@ -5,7 +5,8 @@ shim (quickwit_amd/rest.py) with run_tests.py's checking semantics
 reference's own (tests/golden/rest_scenarios.json, extracted by
 tests/golden/extract_goldens.py).
 
-As of round 2 every suite replays in full (219/219 steps, zero skips);
+As of round 2 every suite replays in full (251/251 steps over eight
+suites, zero skips);
 the skip machinery remains so a future regression reports a reason instead
 of a bare failure. Every step must match the reference byte-for-byte at
 run_tests.py's granularity.
