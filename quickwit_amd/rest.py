@@ -340,12 +340,16 @@ class Index:
             name = fm["name"]
             if t == "datetime":
                 fields.append({"name": name, "type": "datetime",
-                               "fast": bool(fm.get("fast", False))})
+                               "fast": bool(fm.get("fast", True))})
                 if dm.get("timestamp_field") == name:
                     ts_field = name
             elif t in ("u64", "i64"):
+                # numeric mappings are queryable without an explicit fast
+                # flag (tag_fields/0002 golden: u64 `tag` with no flags
+                # answers term queries); our engine's numeric term/range
+                # path is the fast column, so default fast = true
                 fields.append({"name": name, "type": t,
-                               "fast": bool(fm.get("fast", False))})
+                               "fast": bool(fm.get("fast", True))})
             elif t == "text" and fm.get("indexed") is False and fm.get("fast"):
                 # fast-only text: raw-string str fast column, no inverted
                 # index (term queries become exact point ranges)
@@ -450,6 +454,32 @@ def create_app(searcher_factory):
     async def create_index(request: Request):
         cfg = await request.json()
         iid = cfg["index_id"]
+        # tag_fields type validation (doc mapper: only raw-tokenizer text,
+        # u64 and i64 fields may be tags — scenarii/tag_fields/0001)
+        dm = cfg.get("doc_mapping", {})
+        tags = set(dm.get("tag_fields", []))
+        if tags:
+            by_name = {}
+
+            def walk(fms, prefix=""):
+                for fm in fms:
+                    if fm.get("type") == "object":
+                        walk(fm.get("field_mappings", []),
+                             prefix + fm["name"] + ".")
+                    else:
+                        by_name[prefix + fm["name"]] = fm
+            walk(dm.get("field_mappings", []))
+            for t in tags:
+                fm = by_name.get(t)
+                ok = fm is not None and (
+                    fm["type"] in ("u64", "i64") or
+                    (fm["type"] == "text" and
+                     fm.get("tokenizer", "default") == "raw"))
+                if not ok:
+                    from fastapi.responses import JSONResponse
+                    return JSONResponse(status_code=400, content={
+                        "message": f"tag field {t} must be raw text, u64 "
+                                   f"or i64"})
         idx = Index(iid, cfg, searcher_factory)
         idx.rebuild()
         indexes[iid] = idx

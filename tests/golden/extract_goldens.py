@@ -139,6 +139,8 @@ def rest_scenarios():
         "multi_splits": ["_setup.quickwit.yaml",
                          "0001-request-optimizations.yaml",
                          "_teardown.quickwit.yaml"],
+        "tag_fields": ["_setup.quickwit.yaml", "0001_allowed_types.yaml",
+                       "0002_negative_tags.yaml", "_teardown.quickwit.yaml"],
         "concat_fields": ["_setup.quickwit.yaml",
                           "0001_concat_field.yaml",
                           "_teardown.quickwit.yaml"],

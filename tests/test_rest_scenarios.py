@@ -5,7 +5,7 @@ shim (quickwit_amd/rest.py) with run_tests.py's checking semantics
 reference's own (tests/golden/rest_scenarios.json, extracted by
 tests/golden/extract_goldens.py).
 
-As of round 2 every suite replays in full (251/251 steps over eight
+As of round 2 every suite replays in full (272/272 steps over nine
 suites, zero skips);
 the skip machinery remains so a future regression reports a reason instead
 of a bare failure. Every step must match the reference byte-for-byte at
@@ -264,3 +264,25 @@ def test_rest_concat_fields_suite_gpu():
     import __graft_entry__
     __graft_entry__.build()
     run_concat(lambda: GpuSearcher(device=0))
+
+
+def run_tag_fields(searcher_factory):
+    steps = load_suite("tag_fields")
+    client = make_client(searcher_factory)
+    ran, skipped = replay_suite(client, steps)
+    assert not skipped and ran == len(steps)
+
+
+def test_rest_tag_fields_suite_oracle():
+    from quickwit_amd.api import OracleSearcher
+    import __graft_entry__
+    __graft_entry__.build()
+    run_tag_fields(OracleSearcher)
+
+
+@pytest.mark.gpu
+def test_rest_tag_fields_suite_gpu():
+    from quickwit_amd.api import GpuSearcher
+    import __graft_entry__
+    __graft_entry__.build()
+    run_tag_fields(lambda: GpuSearcher(device=0))
