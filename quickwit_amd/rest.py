@@ -165,6 +165,16 @@ def es_query_to_ast(q, schema=None):
             params["zero_terms_query"] = body["zero_terms_query"]
         return {"type": "full_text", "field": field, "text": str(text),
                 "params": params}
+    if "prefix" in q:
+        [(field, body)] = q["prefix"].items()
+        value = body["value"] if isinstance(body, dict) else body
+        value = str(value)
+        if "*" in value or "?" in value:
+            raise ValueError("prefix value with wildcard metachars")
+        node = {"type": "wildcard", "field": field, "value": value + "*"}
+        if isinstance(body, dict) and body.get("case_insensitive"):
+            node["case_insensitive"] = True
+        return node
     if "wildcard" in q:
         [(field, body)] = q["wildcard"].items()
         value = body["value"] if isinstance(body, dict) else body
@@ -272,16 +282,18 @@ def _infer_dynamic_fields(explicit_names, doc_batches):
             # (typed values + f64-monotonic sort keys)
             fields.append({"name": name, "type": "mixed", "fast": True})
         elif kinds == {"str"}:
-            # quickwit dynamic strings: tokenized text index + raw str fast
+            # quickwit dynamic strings: tokenized text index (with
+            # positions, record_field_value_positions) + raw str fast
             # column under the same name (dynamic_mapping tokenizer+fast)
             fields.append({"name": name, "type": "text",
-                           "tokenizer": "default", "fast": True})
+                           "tokenizer": "default", "fast": True,
+                           "record": "position"})
         elif "strlist" in kinds and kinds <= {"str", "strlist"}:
             # string arrays: tokenized text index + multi-valued str fast
             # column (distinct sorted values per doc)
             fields.append({"name": name, "type": "text",
                            "tokenizer": "default", "fast": True,
-                           "multi": True})
+                           "multi": True, "record": "position"})
     return fields
 
 
@@ -386,12 +398,15 @@ class Index:
                                .get("default_search_fields", [])}
         self.searcher = self.searcher_factory()
         self.splits = []
+        self.store_bytes = 0
         batches = flat_batches or [[]]  # empty index: one 0-doc split so
         for i, docs in enumerate(batches):  # aggs return shaped empties
             sid = f"{self.index_id}-{i:04d}"
             w = splitgen.SplitWriter(self.schema, sid)
             w.add_documents(docs)
-            self.searcher.add_split(sid, w.finalize())
+            data = w.finalize()
+            self.store_bytes = getattr(self, "store_bytes", 0) + len(data)
+            self.searcher.add_split(sid, data)
             entry = {"split_id": sid, "num_docs": len(docs)}
             if ts_field and docs:
                 # split time-range metadata (seconds): drives the
@@ -493,6 +508,35 @@ def create_app(searcher_factory):
         del indexes[iid]
         return {"removed": iid}
 
+    def _stats_payload(matching):
+        def one(idxs):
+            docs = sum(sp["num_docs"] for ix in idxs for sp in ix.splits)
+            size = sum(getattr(ix, "store_bytes", 0) for ix in idxs)
+            segs = sum(len(ix.splits) for ix in idxs)
+            half = {"docs": {"count": docs, "deleted": 0},
+                    "store": {"size_in_bytes": size},
+                    "segments": {"count": segs}}
+            return {"primaries": half, "total": half}
+        return {"_all": one(matching),
+                "indices": {ix.index_id: one([ix]) for ix in matching}}
+
+    def _match_indexes(pattern):
+        import fnmatch
+        return [ix for name, ix in indexes.items()
+                if fnmatch.fnmatchcase(name, pattern)]
+
+    @app.get("/api/v1/_stats")
+    async def stats_all():
+        return _stats_payload(list(indexes.values()))
+
+    @app.get("/api/v1/{pattern}/_stats")
+    async def stats_pattern(pattern: str, response: Response):
+        matching = _match_indexes(pattern)
+        if not matching:
+            response.status_code = 404
+            return {"message": "no matching index"}
+        return _stats_payload(matching)
+
     @app.post("/api/v1/{iid}/ingest")
     async def ingest(iid: str, request: Request, response: Response):
         if iid not in indexes:
@@ -578,14 +622,29 @@ def create_app(searcher_factory):
             body = json.loads(raw)
         t0 = time.perf_counter()
         try:
-            ast = es_query_to_ast(body.get("query"), idx.schema)
+            qparam = request.query_params.get("q")
+            if qparam is not None and "query" not in body:
+                ast = {"type": "user_input", "user_text": qparam,
+                       "default_fields": None} if qparam != "*" else                     {"type": "match_all"}
+            else:
+                ast = es_query_to_ast(body.get("query"), idx.schema)
         except ValueError as e:
             response.status_code = 400
             return {"message": str(e)}
         aggs = body.get("aggs") or body.get("aggregations")
         size = int(request.query_params.get("size", body.get("size", 10)))
+        sorts_in = body.get("sort", [])
+        if isinstance(sorts_in, dict):  # ES also accepts a single dict
+            sorts_in = [{k: v} for k, v in sorts_in.items()]
+        sparam = request.query_params.get("sort")
+        if sparam and not sorts_in:
+            # ES URL sort syntax: "field:desc,other:asc"
+            sorts_in = []
+            for part in sparam.split(","):
+                f, _, o = part.partition(":")
+                sorts_in.append({f: {"order": o or "asc"}})
         sort_fields = []
-        for s in body.get("sort", []):
+        for s in sorts_in:
             if isinstance(s, str):
                 field, order = s, "asc"
             else:
@@ -593,6 +652,13 @@ def create_app(searcher_factory):
                 order = so.get("order", "asc") if isinstance(so, dict) else so
             if field == "_score" and order == "desc":
                 sort_fields.append({"field_name": "_score", "sort_order": 1})
+            elif field == "_doc":
+                # ES _doc = internal doc order: realized as an absent sort
+                # column whose doc-id tie-break runs in the sort direction
+                # (0008-sort_by golden pins both directions)
+                sort_fields.append({"field_name": "_doc",
+                                    "sort_order": 1 if order == "desc"
+                                    else 0})
             else:
                 sort_fields.append({"field_name": field,
                                     "sort_order": 1 if order == "desc" else 0})
@@ -601,11 +667,41 @@ def create_app(searcher_factory):
                                 aggregation=aggs)
         sa = body.get("search_after")
         if sa:
+            # datetime sort keys travel as epoch NANOS on the engine's wire
+            # (quickwit's internal representation); the ES layer's default
+            # input format for datetime search_after literals is epoch
+            # millis (0018-search_after golden) unless the sort spec names
+            # epoch_nanos_int
+            def dt_scale(pos):
+                if pos >= len(sorts_in):
+                    return 1
+                spec = sorts_in[pos]
+                if not isinstance(spec, dict):
+                    return 1
+                [(fname, so)] = spec.items()
+                fmt = so.get("format") if isinstance(so, dict) else None
+                is_dt = any(f["name"] == fname and f["type"] == "datetime"
+                            for f in idx.schema["fields"])
+                if not is_dt:
+                    return 1
+                return 1 if fmt == "epoch_nanos_int" else 1_000_000
+            sa = [v * dt_scale(k) if isinstance(v, (int, float))
+                  and not isinstance(v, bool) else v
+                  for k, v in enumerate(sa)]
+            sa = [int(v) * dt_scale(k) if isinstance(v, str)
+                  and v.lstrip("-").isdigit() and dt_scale(k) != 1 else v
+                  for k, v in enumerate(sa)]
             # ES search_after literal array -> typed SortByValue cursor
             # (SearchAfterSegment conversion happens engine-side)
             def lit_to_sv(x):
                 if isinstance(x, bool):
                     return {"boolean": x}
+                if isinstance(x, str):
+                    # ES coerces string literals for numeric sort keys
+                    try:
+                        x = int(x)
+                    except ValueError:
+                        x = float(x)
                 if isinstance(x, float):
                     return {"f64": x}
                 if isinstance(x, int):

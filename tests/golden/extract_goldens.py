@@ -151,14 +151,23 @@ def rest_scenarios():
                              "0011-exists-query.yaml", "0015-terms-query.yaml",
                              "0019-count.yaml", "0029-wildcard.yaml",
                              "_teardown.quickwit.yaml"],
+        "es_compat_extra": ["_setup.quickwit.yaml", "0008-sort_by.yaml",
+                            "0013-phrase-query.yaml", "0018-search_after.yaml",
+                            "0020-stats.yaml",
+                            "0028-fast_only_field_query.yaml",
+                            "0030-prefix.yaml",
+                            "_teardown.quickwit.yaml"],
     }
     keep = {"method", "endpoint", "params", "json", "ndjson", "expected",
             "status_code", "ndjson_file"}
     out = {}
+    # suites whose files live in another suite's directory
+    dir_of = {"es_compat_extra": "es_compatibility"}
     for suite, files in suites.items():
+        sdir = dir_of.get(suite, suite)
         # suite context defaults (run_tests.py Visitor context stacking)
         ctx = {}
-        ctx_path = os.path.join(base, suite, "_ctx.yaml")
+        ctx_path = os.path.join(base, sdir, "_ctx.yaml")
         if os.path.exists(ctx_path):
             ctx = yaml.safe_load(open(ctx_path)) or {}
         ctx_endpoint = ctx.get("endpoint")
@@ -167,7 +176,7 @@ def rest_scenarios():
         steps = []
         for fn in files:
             is_scenario = not fn.startswith("_")
-            for step in yaml.safe_load_all(open(os.path.join(base, suite, fn))):
+            for step in yaml.safe_load_all(open(os.path.join(base, sdir, fn))):
                 if not isinstance(step, dict):
                     continue
                 if is_scenario:  # numbered scenarios inherit the suite ctx
@@ -187,7 +196,7 @@ def rest_scenarios():
                     # ES bulk gz corpus -> committed ndjson fixture + a plain
                     # ingest step per target index
                     import gzip
-                    raw = gzip.open(os.path.join(base, suite,
+                    raw = gzip.open(os.path.join(base, sdir,
                                                  step["body_from_file"])).read()
                     lines = [l for l in raw.decode().splitlines() if l.strip()]
                     by_index = {}
