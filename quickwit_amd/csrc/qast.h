@@ -401,6 +401,14 @@ inline PlanNode term_leaf_plan(const std::string& field,
     b.kind = Bound::INCLUDED;
     if (tf->type == "str") {
         b.sval = value;
+    } else if (tf->type == "bool") {
+        // doc-mapper bool: u64 0/1 column (splitgen stores bools that way)
+        if (value == "true") b.ival = 1;
+        else if (value == "false") b.ival = 0;
+        else
+            throw std::runtime_error(
+                "invalid term value for bool field " + field);
+        b.fval = double(b.ival);
     } else if (tf->type == "datetime") {
         mj::Value lit;
         lit.kind = mj::Value::STR;
@@ -508,6 +516,7 @@ struct QTok {
     std::string field;   // LIT: optional field ("" = default fields)
     std::string text;    // LIT: raw literal text (quotes stripped)
     bool quoted = false;
+    bool no_glob = false;  // backslash-escaped * or ? (literal, not glob)
     bool prefix = false;  // quoted phrase followed by '*' (phrase prefix)
     bool range_ge = false, range_gt = false, range_le = false, range_lt = false;
     bool bracket = false;  // [a TO b] / {a TO b}
@@ -554,9 +563,16 @@ inline std::vector<QTok> qlex(const std::string& s) {
             t.quoted = true;
             t.prefix = after_quote_star();
         } else {
+            bool esc = false;
             while (i < s.size() && s[i] != ' ' && s[i] != '\t' && s[i] != '(' &&
-                   s[i] != ')' && s[i] != ':' && s[i] != '^')
+                   s[i] != ')' && s[i] != ':' && s[i] != '^') {
+                if (s[i] == '\\' && i + 1 < s.size()) {
+                    esc = true;
+                    ++i;  // escaped char, taken verbatim
+                }
                 word.push_back(s[i++]);
+            }
+            t.no_glob = esc;
             if (i < s.size() && s[i] == ':') {
                 t.field = word;
                 ++i;
@@ -608,8 +624,13 @@ inline std::vector<QTok> qlex(const std::string& s) {
                 } else {
                     // plain value or IN [..]
                     std::string v;
-                    while (i < s.size() && s[i] != ' ' && s[i] != ')' && s[i] != '^')
+                    while (i < s.size() && s[i] != ' ' && s[i] != ')' && s[i] != '^') {
+                        if (s[i] == '\\' && i + 1 < s.size()) {
+                            t.no_glob = true;
+                            ++i;
+                        }
                         v.push_back(s[i++]);
+                    }
                     if (v == "IN") {
                         while (i < s.size() && s[i] == ' ') ++i;
                         if (i >= s.size() || s[i] != '[')
@@ -655,7 +676,12 @@ inline Bound qbound(const std::string& text, Bound::Kind kind, const SchemaField
     Bound b;
     if (text.empty() || text == "*") return b;
     b.kind = kind;
-    if (f.type == "datetime") {
+    if (f.type == "bool") {
+        if (text == "true") b.ival = 1;
+        else if (text == "false") b.ival = 0;
+        else throw std::runtime_error("invalid bool literal: " + text);
+        b.fval = double(b.ival);
+    } else if (f.type == "datetime") {
         mj::Value v;
         v.kind = mj::Value::STR;
         v.s = text;
@@ -690,12 +716,13 @@ inline PlanNode qparse_clause(const std::vector<QTok>& toks, size_t* pos,
     };
     std::vector<Item> items;
     bool saw_and = false;
+    bool dangling = false;  // connector awaiting its right operand
     int pending = 0;
     while (*pos < toks.size()) {
         const QTok& t = toks[*pos];
         if (t.kind == QTok::RPAREN) { ++(*pos); break; }
-        if (t.kind == QTok::AND) { saw_and = true; ++(*pos); continue; }
-        if (t.kind == QTok::OR) { ++(*pos); continue; }
+        if (t.kind == QTok::AND) { saw_and = true; dangling = true; ++(*pos); continue; }
+        if (t.kind == QTok::OR) { dangling = true; ++(*pos); continue; }
         if (t.kind == QTok::PLUS) { pending = 1; ++(*pos); continue; }
         if (t.kind == QTok::MINUS || t.kind == QTok::NOT) {
             pending = 2;
@@ -713,7 +740,11 @@ inline PlanNode qparse_clause(const std::vector<QTok>& toks, size_t* pos,
             ++(*pos);
         }
         items.push_back(std::move(it));
+        dangling = false;
     }
+    if (dangling)
+        throw std::runtime_error(
+            "query grammar: operator without right operand");
     if (items.empty()) {
         PlanNode n;
         n.kind = PlanNode::MATCH_ALL;
@@ -858,6 +889,18 @@ inline PlanNode qtok_leaf_impl(const QTok& t, const std::vector<std::string>& df
             n.boost = t.boost;
             return n;
         }
+        if (!t.no_glob && (t.text.find('*') != std::string::npos ||
+                           t.text.find('?') != std::string::npos)) {
+            // field:pat* -> wildcard over the term dictionary, pattern
+            // folded through the field's (non-raw) analyzer chain
+            n.kind = PlanNode::WILDCARD;
+            n.const_score = true;
+            n.field = t.field;
+            n.value = (f && f->tokenizer == "raw") ? t.text
+                                                   : qw_utf8_lower(t.text);
+            n.boost = t.boost;
+            return n;
+        }
         // unquoted value: a single grammar word; when the field's analyzer
         // splits it into several tokens (1.5 -> [1,5], AB-CD -> [ab,cd])
         // the reference's parser emits a slop-0 PhraseQuery, not an OR
@@ -885,11 +928,68 @@ inline PlanNode qtok_leaf_impl(const QTok& t, const std::vector<std::string>& df
     if (dfs.empty()) throw std::runtime_error("no default search fields");
     PlanNode b;
     b.kind = PlanNode::BOOL;
+    bool has_glob = !t.no_glob &&
+                    (t.text.find('*') != std::string::npos ||
+                     t.text.find('?') != std::string::npos);
+    // bare ">=N" / "<N" over the default fields: a range on each
+    // (query_string with a numeric default_field — es_compat golden)
+    {
+        Bound::Kind lk = Bound::UNBOUNDED, hk = Bound::UNBOUNDED;
+        size_t skip = 0;
+        if (t.text.rfind(">=", 0) == 0) { lk = Bound::INCLUDED; skip = 2; }
+        else if (t.text.rfind("<=", 0) == 0) { hk = Bound::INCLUDED; skip = 2; }
+        else if (!t.text.empty() && t.text[0] == '>') { lk = Bound::EXCLUDED; skip = 1; }
+        else if (!t.text.empty() && t.text[0] == '<') { hk = Bound::EXCLUDED; skip = 1; }
+        if (skip) {
+            std::string lit = t.text.substr(skip);
+            for (auto& df : dfs) {
+                const SchemaField* f = schema.field(df);
+                if (!f || f->type == "text") continue;
+                try {
+                    PlanNode r;
+                    r.kind = PlanNode::RANGE;
+                    r.field = df;
+                    if (lk != Bound::UNBOUNDED) r.lo = qbound(lit, lk, *f);
+                    else r.hi = qbound(lit, hk, *f);
+                    b.should.push_back(std::move(r));
+                } catch (const std::exception&) {
+                }
+            }
+            if (!b.should.empty()) {
+                PlanNode r = b.should.size() == 1 ? std::move(b.should[0])
+                                                  : std::move(b);
+                r.boost *= t.boost;
+                return r;
+            }
+            PlanNode none;
+            none.kind = PlanNode::MATCH_NONE;
+            return none;
+        }
+    }
     for (auto& df : dfs) {
         // lenient over the default-field list: configured names may be
         // dynamic fields absent from a given split's schema
         const SchemaField* f = schema.field(df);
-        if (!f || f->type != "text") continue;
+        if (!f) continue;
+        if (f->type != "text") {
+            // numeric/bool/str fast default field: equality when the
+            // literal parses for that type (lenient otherwise)
+            if (has_glob) continue;
+            try {
+                b.should.push_back(term_leaf_plan(df, t.text, schema));
+            } catch (const std::exception&) {
+            }
+            continue;
+        }
+        if (has_glob) {
+            PlanNode w;
+            w.kind = PlanNode::WILDCARD;
+            w.const_score = true;
+            w.field = df;
+            w.value = qw_utf8_lower(t.text);
+            b.should.push_back(std::move(w));
+            continue;
+        }
         b.should.push_back(full_text_plan(df, t.text, "or", schema));
     }
     if (b.should.empty()) {

@@ -145,6 +145,35 @@ def es_query_to_ast(q, schema=None):
         elif "lt" in body:
             node["upper_bound"] = {"excluded": conv(body["lt"])}
         return node
+    if "match_phrase_prefix" in q:
+        [(field, body)] = q["match_phrase_prefix"].items()
+        text = str(body["query"] if isinstance(body, dict) else body).strip()
+        if " " in text:
+            raise ValueError(
+                "multi-token match_phrase_prefix not supported")
+        # single analyzed token: phrase prefix == term prefix == wildcard
+        if "*" in text or "?" in text:
+            raise ValueError("phrase prefix with wildcard metachars")
+        return {"type": "wildcard", "field": field, "value": text + "*"}
+    if "match_bool_prefix" in q:
+        # every token is a term except the LAST, which matches as a prefix
+        [(field, body)] = q["match_bool_prefix"].items()
+        text = str(body["query"] if isinstance(body, dict) else body).strip()
+        op = (body.get("operator", "or") if isinstance(body, dict)
+              else "or").lower()
+        words = text.split()
+        if not words:
+            return {"type": "match_none"}
+        last = words[-1]
+        if "*" in last or "?" in last:
+            raise ValueError("bool prefix with wildcard metachars")
+        subs = [{"type": "full_text", "field": field, "text": w,
+                 "params": {"mode": {"type": "bool", "operator": "or"}}}
+                for w in words[:-1]]
+        subs.append({"type": "wildcard", "field": field,
+                     "value": last + "*"})
+        key = "must" if op == "and" else "should"
+        return {"type": "bool", key: subs}
     if "match_phrase" in q:
         [(field, body)] = q["match_phrase"].items()
         text = body["query"] if isinstance(body, dict) else body
@@ -203,8 +232,64 @@ def es_query_to_ast(q, schema=None):
             out["minimum_should_match"] = max(msm, 0)
         return out
     if "query_string" in q:
-        return {"type": "user_input", "user_text": q["query_string"]["query"],
-                "default_fields": q["query_string"].get("fields")}
+        body = q["query_string"]
+        fields = body.get("fields")
+        dfield = body.get("default_field")
+        lenient = bool(body.get("lenient", False))
+        if fields is not None and not isinstance(fields, list):
+            raise ValueError("query_string fields must be an array")
+        if fields and dfield:
+            raise ValueError(
+                "query_string cannot take both default_field and fields")
+        dfs = fields if fields else ([dfield] if dfield else None)
+        if dfs and schema is not None:
+            known = {f["name"] for f in schema.get("fields", [])}
+            missing = [f for f in dfs if f not in known]
+            if missing and not lenient:
+                raise ValueError(
+                    f"query_string over unknown fields {missing} "
+                    f"(lenient is false)")
+            dfs = [f for f in dfs if f in known]
+            if not dfs:
+                # every named field unknown under lenient: the query must
+                # STILL parse (a syntax error is 400 even then); bare
+                # tokens then match nothing via a field no split carries
+                dfs = ["__lenient_no_field__"]
+        return {"type": "user_input", "user_text": body["query"],
+                "default_fields": dfs}
+    if "multi_match" in q:
+        # unknown fields contribute nothing (ES semantics, regardless of
+        # lenient); slop and multi-token phrase_prefix are declared out
+        body = q["multi_match"]
+        fields = body.get("fields")
+        if isinstance(fields, str):  # ES accepts a single-string fields
+            fields = [fields]
+        if not fields:
+            raise ValueError("multi_match requires a non-empty fields list")
+        mtype = body.get("type", "best_fields")
+        if body.get("slop"):
+            raise ValueError("multi_match slop > 0 not supported")
+        if mtype == "phrase_prefix":
+            raise ValueError("multi_match phrase_prefix not supported")
+        known = {f["name"] for f in (schema or {}).get("fields", [])}
+        use = [f for f in fields if f in known] if schema else fields
+        subs = []
+        for f in use:
+            if mtype == "phrase":
+                subs.append({"type": "full_text", "field": f,
+                             "text": str(body["query"]),
+                             "params": {"mode": {"type": "phrase"}}})
+            else:
+                op = str(body.get("operator", "or")).lower()
+                subs.append({"type": "full_text", "field": f,
+                             "text": str(body["query"]),
+                             "params": {"mode": {"type": "bool",
+                                                 "operator": op}}})
+        if not subs:
+            return {"type": "match_none"}
+        if len(subs) == 1:
+            return subs[0]
+        return {"type": "bool", "should": subs}
     raise ValueError(f"unsupported es query: {list(q)}")
 
 
@@ -270,7 +355,11 @@ def _infer_dynamic_fields(explicit_names, doc_batches):
         kinds = seen[name]
         if "skip" in kinds:
             continue
-        if kinds <= {"int", "bigint"}:
+        if kinds == {"bool"}:
+            # boolean dynamic column: u64 0/1 storage, doc-mapper type
+            # bool so true/false term literals map to 1/0
+            fields.append({"name": name, "type": "bool", "fast": True})
+        elif kinds <= {"int", "bigint"}:
             fields.append({"name": name, "type": "u64", "fast": True})
         elif kinds <= {"int", "negint"}:
             fields.append({"name": name, "type": "i64", "fast": True})
@@ -623,7 +712,7 @@ def create_app(searcher_factory):
         t0 = time.perf_counter()
         try:
             qparam = request.query_params.get("q")
-            if qparam is not None and "query" not in body:
+            if qparam is not None:  # URL q wins over the body query (0008)
                 ast = {"type": "user_input", "user_text": qparam,
                        "default_fields": None} if qparam != "*" else                     {"type": "match_all"}
             else:
@@ -714,6 +803,16 @@ def create_app(searcher_factory):
                 cursor["sort_value2"] = lit_to_sv(sa[1])
             req["search_request"]["search_after"] = cursor
         resp = idx.searcher.leaf_search(req)
+        if ast.get("type") == "user_input":
+            # ES rejects an unparsable query_string with 400; engine-side
+            # the parse error is per-split data, so lift it back out when
+            # NO split could even parse the query
+            failed = resp.get("failed_splits", [])
+            if failed and not resp.get("num_successful_splits"):
+                err = failed[0].get("error", "")
+                if "grammar" in err or "parse" in err or "unknown" in err:
+                    response.status_code = 400
+                    return {"message": err}
         took_ms = int((time.perf_counter() - t0) * 1e3)
         hits = []
         for h in resp.get("partial_hits", []):

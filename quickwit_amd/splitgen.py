@@ -366,6 +366,11 @@ def _build_fast_field(sec, fspec, num_docs, values, present):
         meta["sec"]["tags"] = sec.add(tags)
         meta["sec"]["raw"] = sec.add(raw)
     else:
+        if ftype == "bool":
+            # doc-mapper bool: u64 0/1 storage; the query side maps
+            # true/false literals to 1/0 (SchemaField type stays bool)
+            meta["type"] = ftype = "u64"
+            values = [int(bool(v)) for v in values]
         dt = (np.uint64 if ftype == "u64"
               else np.float64 if ftype == "f64" else np.int64)
         col = np.zeros(num_docs, dtype=dt)
@@ -444,8 +449,12 @@ class SplitWriter:
                             if isinstance(v, str):
                                 v = [v]
                             if isinstance(v, list):
+                                # empty strings are absent from the fast
+                                # column (es_compat exists golden: a doc
+                                # whose only value is "" does not exist)
                                 vals.append([x.lower() if lower else x
-                                             for x in v if isinstance(x, str)])
+                                             for x in v
+                                             if isinstance(x, str) and x])
                             else:
                                 vals.append(None)
                         fast_inputs[name] = (vals, None)
@@ -453,7 +462,7 @@ class SplitWriter:
                         vals, pres = [], []
                         for d in self.docs:
                             v = d.get(name)
-                            ok = isinstance(v, str)
+                            ok = isinstance(v, str) and v != ""
                             pres.append(ok)
                             vals.append((v.lower() if lower else v) if ok else "")
                         fast_inputs[name] = (vals, np.array(pres, dtype=bool))

@@ -155,7 +155,13 @@ def rest_scenarios():
                             "0013-phrase-query.yaml", "0018-search_after.yaml",
                             "0020-stats.yaml",
                             "0028-fast_only_field_query.yaml",
-                            "0030-prefix.yaml",
+                            "0030-prefix.yaml", "0002-query_string.yaml",
+                            "0004-term_aggregations.yaml",
+                            "0005-query_string_query.yaml",
+                            "0010-match_phrase_prefix_query.yaml",
+                            "0014-multi-match-query.yaml",
+                            "0016-misc-query.yaml",
+                            "0017-match-bool-prefix-query.yaml",
                             "_teardown.quickwit.yaml"],
     }
     keep = {"method", "endpoint", "params", "json", "ndjson", "expected",

@@ -297,11 +297,27 @@ def skip_extra(i, step):
         return "global _stats totals depend on cross-suite session state"
     body = step.get("json")
     if isinstance(body, dict):
-        mp = body.get("query", {}).get("match_phrase")
+        qy = body.get("query", {})
+        mp = qy.get("match_phrase")
         if isinstance(mp, dict):
-            [(f, spec)] = mp.items()
+            [(_f, spec)] = mp.items()
             if isinstance(spec, dict) and spec.get("slop"):
                 return "phrase slop > 0: declared out (DESIGN.md §7)"
+        mm = qy.get("multi_match")
+        if isinstance(mm, dict):
+            if mm.get("slop"):
+                return "phrase slop > 0: declared out (DESIGN.md §7)"
+            if mm.get("type") == "phrase_prefix":
+                return "multi-token phrase prefix: declared out"
+        mpp = qy.get("match_phrase_prefix")
+        if isinstance(mpp, dict):
+            [(_f, spec)] = mpp.items()
+            text = str(spec["query"] if isinstance(spec, dict)
+                       else spec).strip()
+            if " " in text:
+                return "multi-token phrase prefix: declared out"
+        if "regexp" in qy:
+            return "regex queries: declared out (DESIGN.md §7)"
     return None
 
 
@@ -309,7 +325,8 @@ def run_es_compat_extra(searcher_factory):
     steps = load_suite("es_compat_extra")
     client = make_client(searcher_factory)
     ran, skipped = replay_suite(client, steps, skip_extra)
-    assert ran >= 52, (ran, skipped)
+    assert ran >= 104, (ran, skipped)
+    assert len(skipped) <= 13, skipped
     for _, reason in skipped:
         assert "declared" in reason or "session state" in reason, reason
     return ran, skipped
