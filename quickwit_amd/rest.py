@@ -597,6 +597,20 @@ def create_app(searcher_factory):
         del indexes[iid]
         return {"removed": iid}
 
+    @app.delete("/api/v1/_elastic/{targets}")
+    @app.delete("/api/v1/{targets}")
+    async def es_delete_indices(targets: str, request: Request,
+                                response: Response):
+        names = [t for t in targets.split(",") if t]
+        ignore = request.query_params.get("ignore_unavailable") == "true"
+        missing = [n for n in names if n not in indexes]
+        if missing and not ignore:
+            response.status_code = 404
+            return {"message": f"index(es) not found: {missing}"}
+        for n in names:
+            indexes.pop(n, None)
+        return {"acknowledged": True}
+
     def _stats_payload(matching):
         def one(idxs):
             docs = sum(sp["num_docs"] for ix in idxs for sp in ix.splits)
@@ -802,6 +816,13 @@ def create_app(searcher_factory):
             if len(sa) >= 2:
                 cursor["sort_value2"] = lit_to_sv(sa[1])
             req["search_request"]["search_after"] = cursor
+        def src_paths(name):
+            vals = []
+            for v in request.query_params.getlist(name):
+                vals.extend(p for p in v.split(",") if p)
+            return vals
+        src_inc = src_paths("_source_includes")
+        src_exc = src_paths("_source_excludes")
         resp = idx.searcher.leaf_search(req)
         if ast.get("type") == "user_input":
             # ES rejects an unparsable query_string with 400; engine-side
@@ -819,6 +840,8 @@ def create_app(searcher_factory):
             doc_id = h.get("doc_id", 0)
             sid = h.get("split_id", "")
             src = idx.source_doc(sid, doc_id)
+            if src_inc or src_exc:
+                src = _filter_source(src, src_inc, src_exc)
             hit = {"_index": iid, "_id": f"{sid}:{doc_id}", "_source": src}
             sv = h.get("sort_value", {})
             if "f64" in sv and sort_fields and \
@@ -847,3 +870,46 @@ def create_app(searcher_factory):
         return out
 
     return app
+
+
+def _filter_source(src, includes, excludes):
+    """ES _source_includes/_source_excludes tree filtering over dotted
+    paths (0022-source golden): includes keep matching subtrees, then
+    excludes drop theirs."""
+    def keep(tree, path):
+        if not isinstance(tree, dict) or not path:
+            return tree
+        head, _, rest = path.partition(".")
+        if head not in tree:
+            return None
+        sub = keep(tree[head], rest) if rest else tree[head]
+        return None if sub is None else {head: sub}
+
+    def merge(a, b):
+        for k, v in b.items():
+            if k in a and isinstance(a[k], dict) and isinstance(v, dict):
+                merge(a[k], v)
+            else:
+                a[k] = v
+        return a
+
+    out = src
+    if includes:
+        out = {}
+        for p in includes:
+            kept = keep(src, p)
+            if kept:
+                merge(out, kept)
+    if excludes:
+        import copy
+        out = copy.deepcopy(out)
+        for p in excludes:
+            parts = p.split(".")
+            node = out
+            for part in parts[:-1]:
+                node = node.get(part) if isinstance(node, dict) else None
+                if node is None:
+                    break
+            if isinstance(node, dict):
+                node.pop(parts[-1], None)
+    return out

@@ -162,6 +162,8 @@ def rest_scenarios():
                             "0014-multi-match-query.yaml",
                             "0016-misc-query.yaml",
                             "0017-match-bool-prefix-query.yaml",
+                            "0022-source.yaml",
+                            "0024-delete_indices.yaml",
                             "_teardown.quickwit.yaml"],
     }
     keep = {"method", "endpoint", "params", "json", "ndjson", "expected",

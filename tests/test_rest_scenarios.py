@@ -325,7 +325,7 @@ def run_es_compat_extra(searcher_factory):
     steps = load_suite("es_compat_extra")
     client = make_client(searcher_factory)
     ran, skipped = replay_suite(client, steps, skip_extra)
-    assert ran >= 104, (ran, skipped)
+    assert ran >= 114, (ran, skipped)
     assert len(skipped) <= 13, skipped
     for _, reason in skipped:
         assert "declared" in reason or "session state" in reason, reason
