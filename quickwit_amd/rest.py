@@ -12,11 +12,14 @@ Endpoints (subset the suites touch):
   GET/POST /api/v1/_elastic/{id}/_search   ES body {query, aggs, size, sort}
 
 Dynamic mapping (doc_mapping.mode == "dynamic"): unmapped fields are
-inferred from the ingested corpus — str -> text(default) + fast str column,
-non-negative int -> u64 fast, int -> i64 fast; floats/arrays/objects are
-skipped (multi-valued and f64 fast columns are not in round 1). The engine
-behind the shim is either the product (GpuSearcher) or the oracle
-(OracleSearcher) — the replay asserts identical golden JSON on both.
+inferred from the ingested corpus — str -> text(default, positions) + fast
+str column, string arrays -> text + multi-valued str fast, non-negative
+int -> u64 fast, int -> i64 fast, floats -> f64 fast, bools -> u64 0/1
+with doc-mapper type bool, mixed numeric types -> MIXED column. Explicit
+mappings cover text/str/u64/i64/f64/datetime/bool/json/object/concatenate
+and tag_fields validation. The engine behind the shim is either the
+product (GpuSearcher) or the oracle (OracleSearcher) — the replay asserts
+identical golden JSON on both.
 """
 import json
 import time
