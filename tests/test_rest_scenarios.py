@@ -5,8 +5,10 @@ shim (quickwit_amd/rest.py) with run_tests.py's checking semantics
 reference's own (tests/golden/rest_scenarios.json, extracted by
 tests/golden/extract_goldens.py).
 
-As of round 2 every suite replays in full (272/272 steps over nine
-suites, zero skips);
+As of round 2 the nine core suites replay in full (272/272 steps, zero
+skips) and a tenth extra ES-compat slice replays 114/127 with 13
+declared skips (slop>0, multi-token phrase prefix, regex, cross-suite
+_stats state);
 the skip machinery remains so a future regression reports a reason instead
 of a bare failure. Every step must match the reference byte-for-byte at
 run_tests.py's granularity.
