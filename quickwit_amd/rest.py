@@ -737,6 +737,14 @@ def create_app(searcher_factory):
         except ValueError as e:
             response.status_code = 400
             return {"message": str(e)}
+        extra = request.query_params.get("extra_filters")
+        if extra:
+            # comma-separated query-string filters ANDed onto the query
+            # (es_compat 0023: permissions-style extra filters)
+            ast = {"type": "bool", "must": [ast], "filter": [
+                {"type": "user_input", "user_text": part,
+                 "default_fields": None}
+                for part in extra.split(",") if part]}
         aggs = body.get("aggs") or body.get("aggregations")
         size = int(request.query_params.get("size", body.get("size", 10)))
         sorts_in = body.get("sort", [])

@@ -163,6 +163,7 @@ def rest_scenarios():
                             "0016-misc-query.yaml",
                             "0017-match-bool-prefix-query.yaml",
                             "0022-source.yaml",
+                            "0023-extra_filters.yaml",
                             "0024-delete_indices.yaml",
                             "_teardown.quickwit.yaml"],
     }
