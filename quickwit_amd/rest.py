@@ -790,6 +790,8 @@ def create_app(searcher_factory):
                 if pos >= len(sorts_in):
                     return 1
                 spec = sorts_in[pos]
+                if isinstance(spec, str):  # plain "field" sort form
+                    spec = {spec: {}}
                 if not isinstance(spec, dict):
                     return 1
                 [(fname, so)] = spec.items()
